@@ -1,0 +1,203 @@
+/* sdb_gpu.h — C ABI of the MI355X-native SereneDB search/analytics hot path.
+ *
+ * This is the drop-in boundary of SURVEY.md §8(b): the entry points a
+ * maintainer of the reference would bind where its scan operator hands a
+ * prepared per-segment query to the execution hot loop:
+ *
+ *   sdb_gpu_execute_topk  replaces the per-segment collect loop
+ *     CollectSegmentTopK / RunTopKScan
+ *     (server/connector/duckdb_search_full_scan.cpp:1868-1943) and the
+ *     standalone driver irs::ExecuteTopK/ExecuteTopKWithCount
+ *     (libs/iresearch/include/iresearch/search/doc_collector.hpp:44-136),
+ *     i.e. DocIterator::Collect (index/iterators.hpp:297-298,400-428) over a
+ *     BlockDisjunction/Conjunction (search/block_disjunction.hpp:122-732,
+ *     search/conjunction.hpp:248-529) scored by BM25
+ *     (search/bm25.cpp:60-109,279-310) into an NthPartitionScoreCollector
+ *     (index/iterators.hpp:103-253).
+ *
+ *   sdb_gpu_scan_agg  replaces FullScanner::Scan + ColFilterChain predicate
+ *     narrowing (server/connector/full_scanner.h:40-90,
+ *     index/table_filter_iterator.hpp:104-312) feeding the external DuckDB
+ *     PhysicalHashAggregate (un-vendored; parity pinned at SQL-result level,
+ *     SURVEY.md §8c).
+ *
+ * Conventions: every function returns 0 on success or a negative SDB_ERR_*
+ * code; all buffers are caller-allocated; an opaque context owns the device,
+ * streams and workspace; one context is single-threaded, use one per thread.
+ * Hot entry points never allocate on the device after segment load. The
+ * implementation REQUIRES a GPU: calls fail loudly (SDB_ERR_NO_GPU) rather
+ * than fall back to any CPU path.
+ *
+ * ScoreDoc mirrors irs::ScoreDoc (index/iterators.hpp:93-99). Doc-id
+ * conventions follow utils/type_limits.hpp:41-47 (invalid=0, min=1,
+ * eof=0xFFFFFFFF).
+ */
+#ifndef SDB_GPU_H
+#define SDB_GPU_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#include "sdb_format.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define SDB_OK 0
+#define SDB_ERR_INVALID (-1)
+#define SDB_ERR_NO_GPU (-2)
+#define SDB_ERR_HIP (-3)
+#define SDB_ERR_OOM (-4)
+#define SDB_ERR_BAD_SEGMENT (-5)
+
+/* mirrors irs::ScoreDoc: {score_t score; doc_id_t doc; uint32_t segment_idx}
+ * (index/iterators.hpp:93-99) */
+typedef struct SdbScoreDoc {
+  float score;
+  uint32_t doc;
+  uint32_t segment_idx;
+} SdbScoreDoc;
+
+typedef struct SdbGpuCtx SdbGpuCtx;     /* opaque: device, stream, workspace */
+typedef struct SdbGpuSegment SdbGpuSegment; /* opaque: device-resident segment */
+
+/* ---- context ---- */
+int sdb_gpu_ctx_create(int device, SdbGpuCtx** out);
+int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx);
+/* version/build string (static) */
+const char* sdb_gpu_version(void);
+
+/* ---- segment residency ----
+ * Uploads a segment blob (the byte span of sdb_format.h) into HBM and keeps
+ * it resident — the analogue of the reference's mmap'd segment files
+ * (store/mmap_directory.hpp) with 288 GB HBM3E standing in for the page
+ * cache. `blob` must stay valid during the call only. */
+int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
+                         SdbGpuSegment** out);
+int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg);
+
+/* ---- query plan ----
+ * Terms are addressed by index into the segment's term table: the term-dict
+ * lookup (burst-trie .tm) is a per-query O(#terms) CPU step in the reference
+ * and stays on the host (SURVEY.md §2 "formats/index": out of scope).
+ * Per-term boost mirrors irs::Filter boost. min_match = 1 is an OR
+ * (BlockDisjunction), min_match = nterms an AND (Conjunction), anything
+ * between mirrors BlockDisjunction's min-match counting
+ * (block_disjunction.hpp:122-732 match counts). */
+typedef struct SdbTermRef {
+  uint32_t term_idx;
+  float boost;
+} SdbTermRef;
+
+typedef struct SdbQueryPlan {
+  const SdbTermRef* terms;
+  uint32_t nterms;
+  uint32_t min_match;
+  float k1; /* BM25 k (search/bm25.hpp:62: default 1.2) */
+  float b;  /* BM25 b (search/bm25.hpp:64: default 0.75) */
+  /* Optional GLOBAL BM25 stats for sharded (multi-rank) execution, where
+   * each device holds only its shard: the cross-rank stats merge is the
+   * PreparePhase barrier analogue (duckdb_search_full_scan.cpp:1359-1384)
+   * done over RCCL by the caller. Zero/NULL = derive from the segments
+   * passed to this call (single-node ExecuteTopKWithCount semantics). */
+  uint64_t g_docs_with_field;
+  uint64_t g_total_term_freq;
+  const uint64_t* g_docs_with_term; /* per plan term, or NULL */
+} SdbQueryPlan;
+
+/* Execute BM25 top-k over segments resident on this context's device.
+ *
+ * Mirrors ExecuteTopKWithCount (doc_collector.hpp:44-86): a two-phase
+ * stats-then-collect execution (PreparePhase analogue,
+ * duckdb_search_full_scan.cpp:1359-1384): BM25Stats idf/norm_const/
+ * norm_length are computed on the host in double then narrowed to f32
+ * exactly as bm25.cpp:288-306 does, then every segment is scored on the GPU
+ * and a global top-k is selected.
+ *
+ * Results: hits[0..*out_count) sorted by (score desc, segment asc, doc asc);
+ * *out_count = min(k, accepted); *total_matches = number of matching docs
+ * (TotalMatches analogue, index/iterators.hpp:133). Acceptance mirrors the
+ * collector: score > FLT_MIN (doc_collector.hpp:58 initial threshold).
+ * Determinism: scores are bit-exact reproductions of the reference fp32
+ * arithmetic with term-major merge order; ties at the k-th score resolve by
+ * (segment, doc) ascending — a deterministic refinement of the reference's
+ * unspecified nth_element tie order (DESIGN.md "Determinism").
+ *
+ * hits must have room for k entries. */
+int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                         uint32_t nsegs, const SdbQueryPlan* plan, uint32_t k,
+                         SdbScoreDoc* hits, uint32_t* out_count,
+                         uint64_t* total_matches);
+
+/* Raw postings-block decode of one term into caller buffers (docs+freqs,
+ * df entries each). Parity/diagnostic entry (mirrors
+ * FormatTraits128::ReadBlockDelta/ReadBlock, format_block_128.hpp:446-636);
+ * runs on the GPU. */
+int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
+                        uint32_t* docs, uint32_t* freqs);
+
+/* ---- columnar scan -> filter -> hash aggregate ---- */
+
+/* Column layout: dense device-resident columns (round 1: uncompressed i64 /
+ * f32; FoR/bitpack codecs are a planned next row). */
+typedef enum SdbColType {
+  SDB_COL_I64 = 0,
+  SDB_COL_F32 = 1,
+} SdbColType;
+
+typedef struct SdbColumnView {
+  const void* data; /* HOST pointer at load time */
+  uint64_t rows;
+  SdbColType type;
+} SdbColumnView;
+
+typedef struct SdbGpuTable SdbGpuTable;
+
+int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
+                       uint32_t ncols, uint64_t rows, SdbGpuTable** out);
+int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab);
+
+typedef enum SdbPredOp {
+  SDB_PRED_NONE = 0,
+  SDB_PRED_LT = 1,  /* col < v */
+  SDB_PRED_GE = 2,  /* col >= v */
+  SDB_PRED_BETWEEN = 3, /* lo <= col <= hi */
+} SdbPredOp;
+
+typedef struct SdbPredSpec {
+  uint32_t col;
+  SdbPredOp op;
+  int64_t ilo, ihi;
+  float flo, fhi;
+} SdbPredSpec;
+
+typedef enum SdbAggOp {
+  SDB_AGG_COUNT = 0,
+  SDB_AGG_SUM_I64 = 1, /* exact wrap-around i64 sum */
+  SDB_AGG_SUM_F64 = 2, /* f32 column summed in f64 */
+} SdbAggOp;
+
+typedef struct SdbAggSpec {
+  uint32_t col; /* ignored for COUNT */
+  SdbAggOp op;
+} SdbAggSpec;
+
+/* group_key column must be i64 with values in [0, ngroups).
+ * out layout: for each group g, naggs consecutive values; COUNT/SUM_I64
+ * results are int64 stored in the i64 field, SUM_F64 in the f64 field. */
+typedef struct SdbAggResult {
+  int64_t i64;
+  double f64;
+} SdbAggResult;
+
+int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
+                     uint32_t ngroups, const SdbPredSpec* preds,
+                     uint32_t npreds, const SdbAggSpec* aggs, uint32_t naggs,
+                     SdbAggResult* out /* ngroups*naggs */,
+                     uint64_t* rows_passed);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* SDB_GPU_H */

@@ -1,0 +1,292 @@
+"""pyoracle — ctypes bindings for the CPU oracle (oracle/liboracle.so) and
+the reference's own simdcomp (oracle/_ref/libsimdcomp_ref.so).
+
+TEST INFRASTRUCTURE ONLY: importable from tests/, __graft_entry__.smoke()
+and bench.py's cpu_baseline leg, and nowhere else (see sdb_oracle.c header).
+"""
+
+import ctypes as C
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+PU32 = C.POINTER(C.c_uint32)
+PU64 = C.POINTER(C.c_uint64)
+
+
+class OScoreDoc(C.Structure):
+    _fields_ = [
+        ("score", C.c_float),
+        ("doc", C.c_uint32),
+        ("seg", C.c_uint32),
+    ]
+
+
+class OSegBlob(C.Structure):
+    _fields_ = [("blob", C.c_void_p), ("size", C.c_uint64)]
+
+
+_lib = None
+_ref = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        path = os.path.join(_DIR, "liboracle.so")
+        if not os.path.exists(path):
+            subprocess.run(["make", "-C", _DIR, "liboracle.so"], check=True)
+        _lib = C.CDLL(path)
+    return _lib
+
+
+def ref_simdcomp():
+    """The reference's own simdcomp, or None when not built (GPU box without
+    /root/reference and without a prebuilt _ref)."""
+    global _ref
+    if _ref is None:
+        path = os.path.join(_DIR, "_ref", "libsimdcomp_ref.so")
+        if not os.path.exists(path):
+            subprocess.run(["make", "-C", _DIR, "ref"], check=False)
+        if os.path.exists(path):
+            _ref = C.CDLL(path)
+    return _ref
+
+
+def _u32arr(a):
+    return np.ascontiguousarray(a, dtype=np.uint32)
+
+
+def decode_doc_block(payload, length, prev):
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    consumed = lib().o_decode_doc_block(
+        buf.ctypes.data_as(C.POINTER(C.c_uint8)), C.c_uint32(length),
+        C.c_uint32(prev), out.ctypes.data_as(PU32))
+    return out[:length].copy(), consumed
+
+
+def decode_freq_block(payload, length):
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    consumed = lib().o_decode_freq_block(
+        buf.ctypes.data_as(C.POINTER(C.c_uint8)), C.c_uint32(length),
+        out.ctypes.data_as(PU32))
+    return out[:length].copy(), consumed
+
+
+def encode_doc_block(docs, prev):
+    docs = _u32arr(docs)
+    out = np.zeros(len(docs) * 5 + 16, dtype=np.uint8)
+    size = lib().o_encode_doc_block(
+        docs.ctypes.data_as(PU32), C.c_uint32(len(docs)), C.c_uint32(prev),
+        out.ctypes.data_as(C.POINTER(C.c_uint8)))
+    return bytes(out[:size])
+
+
+def encode_freq_block(freqs):
+    freqs = _u32arr(freqs)
+    out = np.zeros(len(freqs) * 5 + 16, dtype=np.uint8)
+    size = lib().o_encode_freq_block(
+        freqs.ctypes.data_as(PU32), C.c_uint32(len(freqs)),
+        out.ctypes.data_as(C.POINTER(C.c_uint8)))
+    return bytes(out[:size])
+
+
+def decode_term(blob, term_idx, df):
+    """Decode a whole term's postings from a segment blob via the oracle."""
+    buf = np.frombuffer(blob, dtype=np.uint8)
+
+    class _View(C.Structure):
+        _fields_ = [("hdr", C.c_void_p), ("terms", C.c_void_p),
+                    ("desc", C.c_void_p), ("norms", C.c_void_p),
+                    ("payload", C.c_void_p)]
+
+    v = _View()
+    rc = lib().o_segment_parse(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(blob)), C.byref(v))
+    assert rc == 0, rc
+    docs = np.zeros(df, dtype=np.uint32)
+    freqs = np.zeros(df, dtype=np.uint32)
+    rc = lib().o_decode_term(C.byref(v), C.c_uint32(term_idx),
+                             docs.ctypes.data_as(PU32),
+                             freqs.ctypes.data_as(PU32))
+    assert rc == 0, rc
+    return docs, freqs
+
+
+def bm25_stats(dwf, dwt, ttf, k=1.2, b=0.75):
+    idf = C.c_float(0)
+    nc = C.c_float(0)
+    nl = C.c_float(0)
+    lib().o_bm25_stats(C.c_uint64(dwf), C.c_uint64(dwt), C.c_uint64(ttf),
+                       C.c_float(k), C.c_float(b), C.byref(idf),
+                       C.byref(nc), C.byref(nl))
+    return idf.value, nc.value, nl.value
+
+
+def _mkblobs(blobs):
+    bufs = [np.frombuffer(b, dtype=np.uint8) for b in blobs]
+    arr = (OSegBlob * len(blobs))()
+    for i, b in enumerate(bufs):
+        arr[i].blob = b.ctypes.data_as(C.c_void_p).value
+        arr[i].size = len(b)
+    return arr, bufs  # keep bufs alive
+
+
+def _hits_to_np(hits, n):
+    res = np.zeros(n, dtype=[("score", "f4"), ("doc", "u4"),
+                             ("segment", "u4")])
+    for i in range(n):
+        res[i] = (hits[i].score, hits[i].doc, hits[i].seg)
+    return res
+
+
+def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
+                 global_stats=None):
+    """EXACT top-k (parity oracle). blobs: list of segment blob bytes."""
+    arr, keep = _mkblobs(blobs)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    g_dwf, g_ttf, g_dwt_ptr, keep2 = 0, 0, None, None
+    if global_stats is not None:
+        g_dwf, g_ttf, dwt = global_stats
+        keep2 = np.ascontiguousarray(dwt, dtype=np.uint64)
+        g_dwt_ptr = keep2.ctypes.data_as(PU64)
+    hits = (OScoreDoc * k)()
+    out_count = C.c_uint32(0)
+    total = C.c_uint64(0)
+    rc = lib().o_execute_topk(
+        arr, C.c_uint32(len(blobs)), ti.ctypes.data_as(PU32),
+        bo.ctypes.data_as(C.POINTER(C.c_float)), C.c_uint32(len(ti)),
+        C.c_uint32(min_match), C.c_float(k1), C.c_float(b),
+        C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf), C.c_uint32(k),
+        hits, C.byref(out_count), C.byref(total))
+    assert rc == 0, rc
+    del keep, keep2
+    return _hits_to_np(hits, out_count.value), total.value
+
+
+def execute_topk_mech(blobs, term_idx, boosts, k, min_match=1, k1=1.2,
+                      b=0.75, global_stats=None):
+    """2k-buffer/nth_element mechanics emulation (reference fixture
+    validation; single-thread timed CPU path)."""
+    arr, keep = _mkblobs(blobs)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    g_dwf, g_ttf, g_dwt_ptr, keep2 = 0, 0, None, None
+    if global_stats is not None:
+        g_dwf, g_ttf, dwt = global_stats
+        keep2 = np.ascontiguousarray(dwt, dtype=np.uint64)
+        g_dwt_ptr = keep2.ctypes.data_as(PU64)
+    hits = (OScoreDoc * (2 * k))()
+    out_count = C.c_uint32(0)
+    total = C.c_uint64(0)
+    rc = lib().o_execute_topk_mech(
+        arr, C.c_uint32(len(blobs)), ti.ctypes.data_as(PU32),
+        bo.ctypes.data_as(C.POINTER(C.c_float)), C.c_uint32(len(ti)),
+        C.c_uint32(min_match), C.c_float(k1), C.c_float(b),
+        C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf), C.c_uint32(k),
+        hits, C.byref(out_count), C.byref(total))
+    assert rc == 0, rc
+    del keep, keep2
+    return _hits_to_np(hits, out_count.value), total.value
+
+
+def execute_topk_mt(blob, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
+                    nthreads=0, global_stats=None):
+    """Multithreaded mechanics baseline over one segment (timed CPU leg)."""
+    if nthreads == 0:
+        nthreads = os.cpu_count() or 1
+    buf = np.frombuffer(blob, dtype=np.uint8)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    g_dwf, g_ttf, g_dwt_ptr, keep2 = 0, 0, None, None
+    if global_stats is not None:
+        g_dwf, g_ttf, dwt = global_stats
+        keep2 = np.ascontiguousarray(dwt, dtype=np.uint64)
+        g_dwt_ptr = keep2.ctypes.data_as(PU64)
+    hits = (OScoreDoc * k)()
+    out_count = C.c_uint32(0)
+    total = C.c_uint64(0)
+    rc = lib().o_execute_topk_mt(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(blob)),
+        ti.ctypes.data_as(PU32), bo.ctypes.data_as(C.POINTER(C.c_float)),
+        C.c_uint32(len(ti)), C.c_uint32(min_match), C.c_float(k1),
+        C.c_float(b), C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf),
+        C.c_uint32(k), C.c_uint32(nthreads), hits, C.byref(out_count),
+        C.byref(total))
+    assert rc == 0, rc
+    del keep2
+    return _hits_to_np(hits, out_count.value), total.value
+
+
+def scan_agg(keys, v1, v2, ngroups, pred_op=0, lo=0, hi=0):
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    v1 = np.ascontiguousarray(v1, dtype=np.int64)
+    v2 = np.ascontiguousarray(v2, dtype=np.float32)
+    cnt = np.zeros(ngroups, dtype=np.int64)
+    si = np.zeros(ngroups, dtype=np.int64)
+    sf = np.zeros(ngroups, dtype=np.float64)
+    passed = C.c_uint64(0)
+    PI64 = C.POINTER(C.c_int64)
+    rc = lib().o_scan_agg(
+        keys.ctypes.data_as(PI64), v1.ctypes.data_as(PI64),
+        v2.ctypes.data_as(C.POINTER(C.c_float)), C.c_uint64(len(keys)),
+        C.c_uint32(ngroups), C.c_int(pred_op), C.c_int64(lo), C.c_int64(hi),
+        cnt.ctypes.data_as(PI64), si.ctypes.data_as(PI64),
+        sf.ctypes.data_as(C.POINTER(C.c_double)), C.byref(passed))
+    assert rc == 0, rc
+    return cnt, si, sf, passed.value
+
+
+# --- reference simdcomp pins ---------------------------------------------
+
+def ref_pack_d1(prev, values128, bits):
+    """simdpackwithoutmaskd1 from the reference's vendored simdcomp."""
+    r = ref_simdcomp()
+    if r is None:
+        return None
+    vals = _u32arr(values128).copy()
+    assert len(vals) == 128
+    out = np.zeros(16 * bits + 16, dtype=np.uint8)
+    r.simdpackwithoutmaskd1(C.c_uint32(prev), vals.ctypes.data_as(PU32),
+                            out.ctypes.data_as(C.c_void_p),
+                            C.c_uint32(bits))
+    return bytes(out[: 16 * bits])
+
+
+def ref_unpack_d1(prev, payload, bits):
+    r = ref_simdcomp()
+    if r is None:
+        return None
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    r.simdunpackd1(C.c_uint32(prev), buf.ctypes.data_as(C.c_void_p),
+                   out.ctypes.data_as(PU32), C.c_uint32(bits))
+    return out
+
+
+def ref_pack(values128, bits):
+    """simdpackwithoutmask (non-delta; freq blocks)."""
+    r = ref_simdcomp()
+    if r is None:
+        return None
+    vals = _u32arr(values128).copy()
+    out = np.zeros(16 * bits + 16, dtype=np.uint8)
+    r.simdpackwithoutmask(vals.ctypes.data_as(PU32),
+                          out.ctypes.data_as(C.c_void_p), C.c_uint32(bits))
+    return bytes(out[: 16 * bits])
+
+
+def ref_unpack(payload, bits):
+    r = ref_simdcomp()
+    if r is None:
+        return None
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    r.simdunpack(buf.ctypes.data_as(C.c_void_p), out.ctypes.data_as(PU32),
+                 C.c_uint32(bits))
+    return out

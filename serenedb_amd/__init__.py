@@ -1,0 +1,314 @@
+"""serenedb_amd — MI355X-native implementation of SereneDB's search-analytics
+hot path (BM25 postings disjunction/conjunction + top-k, fused columnar
+scan->filter->hash-aggregate), per /root/repo/BASELINE.json north_star.
+
+Python here is plumbing only (ctypes over the C ABI in include/sdb_gpu.h);
+the product is the C++/HIP library. The GPU query path REQUIRES a GPU and
+fails loudly otherwise — there is no CPU fallback in this package.
+"""
+
+import ctypes as C
+import os
+
+_PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+
+
+class SdbScoreDoc(C.Structure):
+    """mirrors irs::ScoreDoc (index/iterators.hpp:93-99)"""
+    _fields_ = [
+        ("score", C.c_float),
+        ("doc", C.c_uint32),
+        ("segment_idx", C.c_uint32),
+    ]
+
+
+class SdbTermRef(C.Structure):
+    _fields_ = [("term_idx", C.c_uint32), ("boost", C.c_float)]
+
+
+def _load(name):
+    path = os.path.join(_PKG_DIR, name)
+    if not os.path.exists(path):
+        raise ImportError(
+            f"{name} not built — run `python -m serenedb_amd.build` "
+            f"(or __graft_entry__.build())"
+        )
+    return C.CDLL(path)
+
+
+# ---------------------------------------------------------------------------
+# host library (index-build side; pure C++; always loadable)
+# ---------------------------------------------------------------------------
+_host = None
+
+
+def host():
+    global _host
+    if _host is None:
+        lib = _load("libsdb_host.so")
+        lib.sdb_host_encode_doc_block.restype = C.c_int
+        lib.sdb_host_decode_doc_block.restype = C.c_int
+        lib.sdb_host_encode_freq_block.restype = C.c_int
+        lib.sdb_host_decode_freq_block.restype = C.c_int
+        lib.sdb_host_build_segment.restype = C.c_int
+        lib.sdb_host_build_synth_segment.restype = C.c_int
+        lib.sdb_host_synth_postings.restype = C.c_int
+        lib.sdb_host_synth_norms.restype = C.c_int
+        lib.sdb_host_blob_free.restype = None
+        lib.sdb_host_bm25_stats.restype = None
+        lib.sdb_host_topk_select.restype = C.c_int
+        _host = lib
+    return _host
+
+
+def encode_doc_block(docs, prev):
+    import numpy as np
+
+    docs = np.ascontiguousarray(docs, dtype=np.uint32)
+    out = np.zeros(len(docs) * 5 + 16, dtype=np.uint8)
+    size = C.c_uint32(0)
+    rc = host().sdb_host_encode_doc_block(
+        docs.ctypes.data_as(C.POINTER(C.c_uint32)), len(docs),
+        C.c_uint32(prev), out.ctypes.data_as(C.POINTER(C.c_uint8)),
+        C.byref(size))
+    assert rc == 0, rc
+    return bytes(out[: size.value])
+
+
+def decode_doc_block(payload, length, prev):
+    import numpy as np
+
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    consumed = C.c_uint32(0)
+    rc = host().sdb_host_decode_doc_block(
+        buf.ctypes.data_as(C.POINTER(C.c_uint8)), length, C.c_uint32(prev),
+        out.ctypes.data_as(C.POINTER(C.c_uint32)), C.byref(consumed))
+    assert rc == 0, rc
+    return out[:length].copy(), consumed.value
+
+
+def encode_freq_block(freqs):
+    import numpy as np
+
+    freqs = np.ascontiguousarray(freqs, dtype=np.uint32)
+    out = np.zeros(len(freqs) * 5 + 16, dtype=np.uint8)
+    size = C.c_uint32(0)
+    rc = host().sdb_host_encode_freq_block(
+        freqs.ctypes.data_as(C.POINTER(C.c_uint32)), len(freqs),
+        out.ctypes.data_as(C.POINTER(C.c_uint8)), C.byref(size))
+    assert rc == 0, rc
+    return bytes(out[: size.value])
+
+
+def decode_freq_block(payload, length):
+    import numpy as np
+
+    buf = np.frombuffer(bytes(payload), dtype=np.uint8)
+    out = np.zeros(128, dtype=np.uint32)
+    consumed = C.c_uint32(0)
+    rc = host().sdb_host_decode_freq_block(
+        buf.ctypes.data_as(C.POINTER(C.c_uint8)), length,
+        out.ctypes.data_as(C.POINTER(C.c_uint32)), C.byref(consumed))
+    assert rc == 0, rc
+    return out[:length].copy(), consumed.value
+
+
+def build_segment(doc_count, postings, norms=None):
+    """postings: list of (docs_u32_array, freqs_u32_array) per term.
+    norms: uint32 array of len doc_count+1 (index 0 unused) or None.
+    Returns the serialized segment blob (bytes)."""
+    import numpy as np
+
+    nterms = len(postings)
+    df = np.array([len(d) for d, _ in postings], dtype=np.uint32)
+    doc_arrs = [np.ascontiguousarray(d, dtype=np.uint32) for d, _ in postings]
+    frq_arrs = [np.ascontiguousarray(f, dtype=np.uint32) for _, f in postings]
+    PU32 = C.POINTER(C.c_uint32)
+    doc_ptrs = (PU32 * nterms)(*[a.ctypes.data_as(PU32) for a in doc_arrs])
+    frq_ptrs = (PU32 * nterms)(*[a.ctypes.data_as(PU32) for a in frq_arrs])
+    norm_ptr = None
+    if norms is not None:
+        norms = np.ascontiguousarray(norms, dtype=np.uint32)
+        assert len(norms) == doc_count + 1
+        norm_ptr = norms.ctypes.data_as(PU32)
+    blob = C.c_void_p(0)
+    size = C.c_uint64(0)
+    rc = host().sdb_host_build_segment(
+        C.c_uint32(doc_count), C.c_uint32(nterms),
+        df.ctypes.data_as(PU32), doc_ptrs, frq_ptrs, norm_ptr,
+        C.byref(blob), C.byref(size))
+    assert rc == 0, rc
+    out = C.string_at(blob, size.value)
+    host().sdb_host_blob_free(blob)
+    return out
+
+
+def build_synth_segment(seed, doc_lo, doc_hi, selectivities):
+    import numpy as np
+
+    sel = np.ascontiguousarray(selectivities, dtype=np.float64)
+    blob = C.c_void_p(0)
+    size = C.c_uint64(0)
+    rc = host().sdb_host_build_synth_segment(
+        C.c_uint64(seed), C.c_uint32(doc_lo), C.c_uint32(doc_hi),
+        C.c_uint32(len(sel)), sel.ctypes.data_as(C.POINTER(C.c_double)),
+        C.byref(blob), C.byref(size))
+    assert rc == 0, rc
+    out = C.string_at(blob, size.value)
+    host().sdb_host_blob_free(blob)
+    return out
+
+
+def synth_postings(seed, doc_count, term, sel):
+    import numpy as np
+
+    PU32 = C.POINTER(C.c_uint32)
+    df = C.c_uint32(0)
+    rc = host().sdb_host_synth_postings(
+        C.c_uint64(seed), C.c_uint32(doc_count), C.c_uint32(term),
+        C.c_double(sel), None, None, C.byref(df))
+    assert rc == 0
+    docs = np.zeros(df.value, dtype=np.uint32)
+    freqs = np.zeros(df.value, dtype=np.uint32)
+    rc = host().sdb_host_synth_postings(
+        C.c_uint64(seed), C.c_uint32(doc_count), C.c_uint32(term),
+        C.c_double(sel), docs.ctypes.data_as(PU32),
+        freqs.ctypes.data_as(PU32), C.byref(df))
+    assert rc == 0
+    return docs, freqs
+
+
+def synth_norms(seed, doc_count):
+    import numpy as np
+
+    norms = np.zeros(doc_count + 1, dtype=np.uint32)
+    rc = host().sdb_host_synth_norms(
+        C.c_uint64(seed), C.c_uint32(doc_count),
+        norms.ctypes.data_as(C.POINTER(C.c_uint32)))
+    assert rc == 0
+    return norms
+
+
+def bm25_stats(docs_with_field, docs_with_term, total_term_freq, k=1.2,
+               b=0.75):
+    idf = C.c_float(0)
+    nc = C.c_float(0)
+    nl = C.c_float(0)
+    host().sdb_host_bm25_stats(
+        C.c_uint64(docs_with_field), C.c_uint64(docs_with_term),
+        C.c_uint64(total_term_freq), C.c_float(k), C.c_float(b),
+        C.byref(idf), C.byref(nc), C.byref(nl))
+    return idf.value, nc.value, nl.value
+
+
+# ---------------------------------------------------------------------------
+# GPU library (the product query path; requires a GPU at call time)
+# ---------------------------------------------------------------------------
+_gpu = None
+
+
+def gpu():
+    global _gpu
+    if _gpu is None:
+        lib = _load("libsdb_gpu.so")
+        lib.sdb_gpu_version.restype = C.c_char_p
+        for f in (
+            "sdb_gpu_ctx_create", "sdb_gpu_ctx_destroy",
+            "sdb_gpu_segment_load", "sdb_gpu_segment_free",
+            "sdb_gpu_execute_topk", "sdb_gpu_decode_term",
+            "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
+        ):
+            getattr(lib, f).restype = C.c_int
+        _gpu = lib
+    return _gpu
+
+
+class GpuContext:
+    """Owns an SdbGpuCtx. Raises RuntimeError (loudly) without a GPU."""
+
+    def __init__(self, device=0):
+        self._lib = gpu()
+        self._ctx = C.c_void_p(0)
+        rc = self._lib.sdb_gpu_ctx_create(device, C.byref(self._ctx))
+        if rc != 0:
+            raise RuntimeError(
+                f"sdb_gpu_ctx_create failed rc={rc} "
+                "(no MI355X visible? the GPU path has no CPU fallback)")
+        self._segments = []
+
+    def close(self):
+        if self._ctx:
+            self._lib.sdb_gpu_ctx_destroy(self._ctx)
+            self._ctx = C.c_void_p(0)
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def load_segment(self, blob):
+        seg = C.c_void_p(0)
+        rc = self._lib.sdb_gpu_segment_load(
+            self._ctx, blob, C.c_size_t(len(blob)), C.byref(seg))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_segment_load rc={rc}")
+        self._segments.append(seg)
+        return seg
+
+    def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
+                     b=0.75, global_stats=None):
+        """global_stats: optional (docs_with_field, total_term_freq,
+        [docs_with_term per term]) for sharded execution."""
+        import numpy as np
+
+        class _Plan(C.Structure):
+            _fields_ = [
+                ("terms", C.POINTER(SdbTermRef)),
+                ("nterms", C.c_uint32),
+                ("min_match", C.c_uint32),
+                ("k1", C.c_float),
+                ("b", C.c_float),
+                ("g_docs_with_field", C.c_uint64),
+                ("g_total_term_freq", C.c_uint64),
+                ("g_docs_with_term", C.POINTER(C.c_uint64)),
+            ]
+
+        terms = (SdbTermRef * len(term_idx))(
+            *[SdbTermRef(t, float(bo)) for t, bo in zip(term_idx, boosts)])
+        plan = _Plan(terms, len(term_idx), min_match, k1, b, 0, 0, None)
+        if global_stats is not None:
+            dwf, ttf, dwt = global_stats
+            dwt_arr = (C.c_uint64 * len(dwt))(*[int(x) for x in dwt])
+            plan.g_docs_with_field = int(dwf)
+            plan.g_total_term_freq = int(ttf)
+            plan.g_docs_with_term = dwt_arr
+        seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value) for s in segs])
+        hits = (SdbScoreDoc * k)()
+        out_count = C.c_uint32(0)
+        total = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_execute_topk(
+            self._ctx, seg_arr, len(segs), C.byref(plan), C.c_uint32(k),
+            hits, C.byref(out_count), C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_topk rc={rc}")
+        n = out_count.value
+        res = np.zeros(n, dtype=[("score", "f4"), ("doc", "u4"),
+                                 ("segment", "u4")])
+        for i in range(n):
+            res[i] = (hits[i].score, hits[i].doc, hits[i].segment_idx)
+        return res, total.value
+
+    def decode_term(self, seg, term_idx, df):
+        import numpy as np
+
+        docs = np.zeros(df, dtype=np.uint32)
+        freqs = np.zeros(df, dtype=np.uint32)
+        PU32 = C.POINTER(C.c_uint32)
+        rc = self._lib.sdb_gpu_decode_term(
+            self._ctx, seg, C.c_uint32(term_idx),
+            docs.ctypes.data_as(PU32), freqs.ctypes.data_as(PU32))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_decode_term rc={rc}")
+        return docs, freqs

@@ -1,0 +1,790 @@
+// sdb_host.cpp — host-side (index-build + plan-prep) part of the
+// MI355X-native SereneDB hot path.
+//
+// Contents:
+//   * FormatTraits128-compatible block codec (encode + scalar decode):
+//     restates libs/iresearch/include/iresearch/formats/posting/
+//     format_block_128.hpp:51-379 (write) and :446-636 (read); delta-bitpack
+//     bit layout per third_party/simdcomp simdpackwithoutmaskd1/simdunpackd1
+//     (see include/sdb_format.h header comment for the layout spec).
+//   * Synthetic segment builder (the index-write side is OUT OF SCOPE per
+//     SURVEY.md §2 — CPU-built synthetic segments feed the GPU; this builder
+//     is the project's replacement for the reference's index_writer at test/
+//     bench time).
+//   * Seeded synthetic corpus generator (SURVEY.md §8d distributions).
+//   * BM25 stats preparation (double->f32 exactly as bm25.cpp:288-306).
+//   * Final host-side top-k select (the PrepareEmitBuffer analogue,
+//     server/connector/duckdb_search_full_scan.cpp:1945-2000).
+//
+// This library is pure host C++ (no HIP): it must run both in the CPU-only
+// build container and on the GPU box. The QUERY path lives in libsdb_gpu
+// (csrc/host/api.cpp + csrc/hip/*) and fails loudly without a GPU.
+//
+// NOTE: compiled with -ffp-contract=off so fp32 BM25 arithmetic is
+// bit-identical across gcc/hipcc/oracle (DESIGN.md "Determinism").
+
+#include <algorithm>
+#include <cmath>
+#include <cstdint>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include "../../../include/sdb_format.h"
+
+namespace {
+
+constexpr uint32_t kBlock = SDB_BLOCK_SIZE;
+
+// ---------------------------------------------------------------------------
+// bit utils
+// ---------------------------------------------------------------------------
+inline uint32_t bit_width_u32(uint32_t v) {
+  return v == 0 ? 0 : 32u - __builtin_clz(v);
+}
+
+// ByteSize1234: format_block_128.hpp:784-795
+inline uint32_t byte_size_1234(uint32_t v) {
+  if (v < (1u << 8)) return 1;
+  if (v < (1u << 16)) return 2;
+  if (v < (1u << 24)) return 3;
+  return 4;
+}
+// ByteSize0124: format_block_128.hpp:797-808
+inline uint32_t byte_size_0124(uint32_t v) {
+  if (v == 0) return 0;
+  if (v < (1u << 8)) return 1;
+  if (v < (1u << 16)) return 2;
+  return 4;
+}
+
+// ---------------------------------------------------------------------------
+// simdcomp-layout vertical bitpack (scalar restatement)
+// value index i -> SSE lane c = i&3, group g = i>>2; lane c's groups form an
+// LSB-first bitstream packed into 32-bit words; flat word w*4+c is word w of
+// lane c. (simdintegratedbitpacking.c ipackwithoutmaskN / iunpackN;
+// simdbitpacking.c __SIMD_fastpackwithoutmask_N / __SIMD_fastunpack_N)
+// ---------------------------------------------------------------------------
+static void pack_vertical(const uint32_t* vals /*128*/, uint32_t bits,
+                          uint8_t* out /*16*bits bytes*/) {
+  std::memset(out, 0, 16u * bits);
+  auto* w = reinterpret_cast<uint32_t*>(out);
+  for (uint32_t i = 0; i < kBlock; ++i) {
+    const uint32_t c = i & 3u, g = i >> 2;
+    const uint64_t bitpos = uint64_t(g) * bits;
+    const uint32_t word = uint32_t(bitpos >> 5), sh = uint32_t(bitpos & 31);
+    const uint64_t v = uint64_t(vals[i]) << sh;
+    w[word * 4 + c] |= uint32_t(v);
+    if (sh + bits > 32) w[(word + 1) * 4 + c] |= uint32_t(v >> 32);
+  }
+}
+
+static void unpack_vertical(const uint8_t* in, uint32_t bits,
+                            uint32_t* vals /*128*/) {
+  const uint32_t mask = bits == 32 ? 0xFFFFFFFFu : ((1u << bits) - 1u);
+  const uint32_t* w = reinterpret_cast<const uint32_t*>(in);
+  // in may be unaligned in the payload stream: memcpy-safe word reads
+  auto rdw = [&](uint32_t idx) {
+    uint32_t x;
+    std::memcpy(&x, reinterpret_cast<const uint8_t*>(w) + 4u * idx, 4);
+    return x;
+  };
+  for (uint32_t i = 0; i < kBlock; ++i) {
+    const uint32_t c = i & 3u, g = i >> 2;
+    const uint64_t bitpos = uint64_t(g) * bits;
+    const uint32_t word = uint32_t(bitpos >> 5), sh = uint32_t(bitpos & 31);
+    uint64_t v = rdw(word * 4 + c) >> sh;
+    if (sh + bits > 32) v |= uint64_t(rdw((word + 1) * 4 + c)) << (32 - sh);
+    vals[i] = uint32_t(v) & mask;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// streamvbyte 1234 (public Lemire format; format_block_128.hpp:182-199,
+// 507-517 call sites). Control bytes first (2-bit codes, low bits = first
+// value, code = nbytes-1), then little-endian data bytes.
+// ---------------------------------------------------------------------------
+static uint32_t svb_encode(const uint32_t* vals, uint32_t len, uint8_t* out) {
+  const uint32_t groups = (len + 3) / 4;
+  uint8_t* ctrl = out;
+  uint8_t* data = out + groups;
+  std::memset(ctrl, 0, groups);
+  for (uint32_t i = 0; i < len; ++i) {
+    const uint32_t n = byte_size_1234(vals[i]);
+    ctrl[i >> 2] |= uint8_t((n - 1) << ((i & 3) * 2));
+    uint32_t v = vals[i];
+    for (uint32_t b = 0; b < n; ++b) {
+      *data++ = uint8_t(v);
+      v >>= 8;
+    }
+  }
+  return uint32_t(data - out);
+}
+
+static uint32_t svb_decode(const uint8_t* in, uint32_t* vals, uint32_t len) {
+  const uint32_t groups = (len + 3) / 4;
+  const uint8_t* ctrl = in;
+  const uint8_t* data = in + groups;
+  for (uint32_t i = 0; i < len; ++i) {
+    const uint32_t n = ((ctrl[i >> 2] >> ((i & 3) * 2)) & 3u) + 1u;
+    uint32_t v = 0;
+    for (uint32_t b = 0; b < n; ++b) v |= uint32_t(*data++) << (8 * b);
+    vals[i] = v;
+  }
+  return uint32_t(data - in);
+}
+
+// ---------------------------------------------------------------------------
+// doc-block encode — WriteTailDelta (format_block_128.hpp:57-242)
+// docs sorted strictly ascending, prev < docs[0], 1 <= len <= 128.
+// Returns bytes written (tag byte included).
+// ---------------------------------------------------------------------------
+static uint32_t encode_doc_block(const uint32_t* in, uint32_t len,
+                                 uint32_t prev, uint8_t* out) {
+  uint8_t best_enc = SDB_DE_VALUES;
+  uint32_t best_size = len * 4;
+
+  bool all_same = true;
+  const uint32_t max = in[len - 1];
+  const uint32_t for_base = prev;
+  const uint32_t for_max = max - for_base;
+
+  uint32_t delta_prev = prev;
+  uint32_t delta_max = in[0] - delta_prev;
+
+  const uint32_t groups = (len + 3) / 4;
+  uint32_t size_svb = 2 + groups;
+  uint32_t size_delta_svb = 2 + groups;
+
+  uint32_t deltas[kBlock];
+  for (uint32_t i = 0; i < len; ++i) {
+    const uint32_t value = in[i];
+    const uint32_t delta_value = value - delta_prev;
+    delta_prev = value;
+    deltas[i] = delta_value;
+    all_same &= (delta_max == delta_value);
+    delta_max = std::max(delta_max, delta_value);
+    size_svb += byte_size_1234(value);
+    size_delta_svb += byte_size_1234(delta_value);
+  }
+
+  bool decided = false;
+  if (all_same) {
+    switch (byte_size_0124(delta_max)) {
+      case 1: best_enc = SDB_DE_DELTA_ALL_SAME_08; best_size = 1; break;
+      case 2: best_enc = SDB_DE_DELTA_ALL_SAME_16; best_size = 2; break;
+      default: best_enc = SDB_DE_DELTA_ALL_SAME_32; best_size = 4; break;
+    }
+    decided = true;
+  }
+  if (!decided) {
+    if (len == kBlock) {  // SupportIfBlock
+      const uint32_t bits = bit_width_u32(delta_max);  // >= 2 here
+      const uint32_t size = (kBlock * bits + 7) / 8;
+      if (size < best_size && bits <= 31) {
+        best_enc = uint8_t(SDB_DE_DELTA_BITPACK(bits));
+        best_size = size;
+      }
+    }
+    if (len != kBlock && size_svb < best_size) {  // SupportIfTail
+      best_enc = SDB_DE_STREAMVBYTE1234;
+      best_size = size_svb;
+    }
+    if (len != kBlock && size_delta_svb < best_size) {
+      best_enc = SDB_DE_DELTA_STREAMVBYTE1234;
+      best_size = size_delta_svb;
+    }
+    {
+      const uint32_t words = (for_max + 1 + 63) / 64;
+      const uint32_t size = 1 + words * 8;
+      if (size - 2 < best_size) {
+        best_enc = SDB_DE_FOR_BITSET;
+        best_size = size;
+      }
+    }
+  }
+
+  uint8_t* p = out;
+  *p++ = best_enc;
+  switch (best_enc) {
+    case SDB_DE_VALUES:
+      std::memcpy(p, in, size_t(len) * 4);
+      p += size_t(len) * 4;
+      break;
+    case SDB_DE_DELTA_ALL_SAME_08:
+      *p++ = uint8_t(delta_max);
+      break;
+    case SDB_DE_DELTA_ALL_SAME_16:
+      std::memcpy(p, &delta_max, 2);
+      p += 2;
+      break;
+    case SDB_DE_DELTA_ALL_SAME_32:
+      std::memcpy(p, &delta_max, 4);
+      p += 4;
+      break;
+    case SDB_DE_FOR_BITSET: {
+      // WriteBitset (format_block_128.hpp:815-839)
+      const uint32_t bytes = best_size - 1;
+      const uint32_t words = bytes / 8;
+      uint64_t bitset[kBlock * 4 / 8];  // max 512 bytes = 64 words
+      std::memset(bitset, 0, bytes);
+      for (uint32_t i = 0; i < len; ++i) {
+        const uint32_t v = in[i] - prev;
+        bitset[v >> 6] |= 1ull << (v & 63);
+      }
+      *p++ = uint8_t(words);
+      std::memcpy(p, bitset, bytes);
+      p += bytes;
+      break;
+    }
+    case SDB_DE_STREAMVBYTE1234: {
+      uint8_t buf[kBlock * 5 + 8];
+      const uint32_t size = svb_encode(in, len, buf);
+      const uint16_t s16 = uint16_t(size);
+      std::memcpy(p, &s16, 2);
+      p += 2;
+      std::memcpy(p, buf, size);
+      p += size;
+      break;
+    }
+    case SDB_DE_DELTA_STREAMVBYTE1234: {
+      uint8_t buf[kBlock * 5 + 8];
+      const uint32_t size = svb_encode(deltas, len, buf);
+      const uint16_t s16 = uint16_t(size);
+      std::memcpy(p, &s16, 2);
+      p += 2;
+      std::memcpy(p, buf, size);
+      p += size;
+      break;
+    }
+    default: {  // delta bitpack
+      const uint32_t bits = uint32_t(best_enc - SDB_DE_DELTA_BITPACK_02) + 2;
+      pack_vertical(deltas, bits, p);
+      p += best_size;
+      break;
+    }
+  }
+  return uint32_t(p - out);
+}
+
+// doc-block decode — ReadTailDelta (format_block_128.hpp:466-559).
+// Writes exactly `len` doc ids to out. Returns bytes consumed.
+static uint32_t decode_doc_block(const uint8_t* in, uint32_t len,
+                                 uint32_t prev, uint32_t* out) {
+  const uint8_t* p = in;
+  const uint8_t type = *p++;
+  switch (type) {
+    case SDB_DE_VALUES:
+      std::memcpy(out, p, size_t(len) * 4);
+      p += size_t(len) * 4;
+      break;
+    case SDB_DE_DELTA_ALL_SAME_08: {
+      const uint32_t v = *p++;
+      for (uint32_t i = 0; i < len; ++i) out[i] = prev + v + v * i;
+      break;
+    }
+    case SDB_DE_DELTA_ALL_SAME_16: {
+      uint16_t v16;
+      std::memcpy(&v16, p, 2);
+      p += 2;
+      const uint32_t v = v16;
+      for (uint32_t i = 0; i < len; ++i) out[i] = prev + v + v * i;
+      break;
+    }
+    case SDB_DE_DELTA_ALL_SAME_32: {
+      uint32_t v;
+      std::memcpy(&v, p, 4);
+      p += 4;
+      for (uint32_t i = 0; i < len; ++i) out[i] = prev + v + v * i;
+      break;
+    }
+    case SDB_DE_FOR_BITSET: {
+      const uint32_t words = *p++;
+      uint32_t n = 0;
+      for (uint32_t i = 0; i < words; ++i) {
+        uint64_t word;
+        std::memcpy(&word, p + 8u * i, 8);
+        const uint32_t off = prev + i * 64;
+        while (word) {
+          out[n++] = off + uint32_t(__builtin_ctzll(word));
+          word &= word - 1;
+        }
+      }
+      p += 8u * words;
+      break;
+    }
+    case SDB_DE_STREAMVBYTE1234: {
+      uint16_t size;
+      std::memcpy(&size, p, 2);
+      p += 2;
+      svb_decode(p, out, len);
+      p += size;
+      break;
+    }
+    case SDB_DE_DELTA_STREAMVBYTE1234: {
+      uint16_t size;
+      std::memcpy(&size, p, 2);
+      p += 2;
+      svb_decode(p, out, len);
+      p += size;
+      uint32_t acc = prev;
+      for (uint32_t i = 0; i < len; ++i) {
+        acc += out[i];
+        out[i] = acc;
+      }
+      break;
+    }
+    default: {  // delta bitpack (full blocks only)
+      const uint32_t bits = uint32_t(type - SDB_DE_DELTA_BITPACK_02) + 2;
+      uint32_t deltas[kBlock];
+      unpack_vertical(p, bits, deltas);
+      uint32_t acc = prev;
+      for (uint32_t i = 0; i < kBlock; ++i) {
+        acc += deltas[i];
+        out[i] = acc;
+      }
+      p += 16u * bits;
+      break;
+    }
+  }
+  return uint32_t(p - in);
+}
+
+// freq-block encode — WriteTail (format_block_128.hpp:249-379)
+static uint32_t encode_freq_block(const uint32_t* in, uint32_t len,
+                                  uint8_t* out) {
+  uint8_t best_enc = SDB_E_VALUES;
+  uint32_t best_size = len * 4;
+
+  bool all_same = true;
+  uint32_t max = in[0];
+  const uint32_t groups = (len + 3) / 4;
+  uint32_t size_svb = 2 + groups;
+  for (uint32_t i = 0; i < len; ++i) {
+    const uint32_t value = in[i];
+    all_same &= (max == value);
+    max = std::max(max, value);
+    size_svb += byte_size_1234(value);
+  }
+
+  bool decided = false;
+  if (all_same) {
+    switch (byte_size_0124(max)) {
+      case 0:
+      case 1: best_enc = SDB_E_ALL_SAME_08; best_size = 1; break;
+      case 2: best_enc = SDB_E_ALL_SAME_16; best_size = 2; break;
+      default: best_enc = SDB_E_ALL_SAME_32; best_size = 4; break;
+    }
+    decided = true;
+  }
+  if (!decided) {
+    if (len == kBlock) {
+      const uint32_t bits = bit_width_u32(max);  // >= 1 here
+      const uint32_t size = (kBlock * bits + 7) / 8;
+      if (size < best_size && bits <= 31) {
+        best_enc = uint8_t(SDB_E_BITPACK(bits));
+        best_size = size;
+      }
+    }
+    if (len != kBlock && size_svb < best_size) {
+      best_enc = SDB_E_STREAMVBYTE1234;
+      best_size = size_svb;
+    }
+  }
+
+  uint8_t* p = out;
+  *p++ = best_enc;
+  switch (best_enc) {
+    case SDB_E_VALUES:
+      std::memcpy(p, in, size_t(len) * 4);
+      p += size_t(len) * 4;
+      break;
+    case SDB_E_ALL_SAME_08:
+      *p++ = uint8_t(max);
+      break;
+    case SDB_E_ALL_SAME_16:
+      std::memcpy(p, &max, 2);
+      p += 2;
+      break;
+    case SDB_E_ALL_SAME_32:
+      std::memcpy(p, &max, 4);
+      p += 4;
+      break;
+    case SDB_E_STREAMVBYTE1234: {
+      uint8_t buf[kBlock * 5 + 8];
+      const uint32_t size = svb_encode(in, len, buf);
+      const uint16_t s16 = uint16_t(size);
+      std::memcpy(p, &s16, 2);
+      p += 2;
+      std::memcpy(p, buf, size);
+      p += size;
+      break;
+    }
+    default: {  // bitpack
+      const uint32_t bits = uint32_t(best_enc - SDB_E_BITPACK_01) + 1;
+      pack_vertical(in, bits, p);
+      p += best_size;
+      break;
+    }
+  }
+  return uint32_t(p - out);
+}
+
+// freq-block decode — ReadTail (format_block_128.hpp:568-636)
+static uint32_t decode_freq_block(const uint8_t* in, uint32_t len,
+                                  uint32_t* out) {
+  const uint8_t* p = in;
+  const uint8_t type = *p++;
+  switch (type) {
+    case SDB_E_VALUES:
+      std::memcpy(out, p, size_t(len) * 4);
+      p += size_t(len) * 4;
+      break;
+    case SDB_E_ALL_SAME_08: {
+      const uint32_t v = *p++;
+      for (uint32_t i = 0; i < len; ++i) out[i] = v;
+      break;
+    }
+    case SDB_E_ALL_SAME_16: {
+      uint16_t v;
+      std::memcpy(&v, p, 2);
+      p += 2;
+      for (uint32_t i = 0; i < len; ++i) out[i] = v;
+      break;
+    }
+    case SDB_E_ALL_SAME_32: {
+      uint32_t v;
+      std::memcpy(&v, p, 4);
+      p += 4;
+      for (uint32_t i = 0; i < len; ++i) out[i] = v;
+      break;
+    }
+    case SDB_E_STREAMVBYTE1234: {
+      uint16_t size;
+      std::memcpy(&size, p, 2);
+      p += 2;
+      svb_decode(p, out, len);
+      p += size;
+      break;
+    }
+    default: {  // bitpack (full blocks only)
+      const uint32_t bits = uint32_t(type - SDB_E_BITPACK_01) + 1;
+      unpack_vertical(p, bits, out);
+      p += 16u * bits;
+      break;
+    }
+  }
+  return uint32_t(p - in);
+}
+
+// ---------------------------------------------------------------------------
+// seeded synthetic corpus: stateless splitmix64 hashing so
+// any (term, doc) draw is order-independent and parallelizable.
+// ---------------------------------------------------------------------------
+inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+inline uint64_t hash3(uint64_t seed, uint64_t a, uint64_t b) {
+  return splitmix64(seed ^ splitmix64(a * 0x9E3779B97F4A7C15ull ^
+                                      splitmix64(b + 0xD1B54A32D192ED03ull)));
+}
+inline double u01(uint64_t h) {  // (0,1)
+  return (double(h >> 11) + 0.5) * (1.0 / 9007199254740992.0);
+}
+
+// freq ~ shifted geometric(p=0.6) capped at 255: P(n) = p(1-p)^(n-1), n>=1
+static uint32_t synth_freq(uint64_t seed, uint32_t term, uint32_t doc) {
+  const double u = u01(hash3(seed, 0x66726571u ^ term, doc));
+  const int n = 1 + int(std::log(u) / std::log(0.4));
+  return uint32_t(std::min(255, std::max(1, n)));
+}
+
+// norm ~ round(lognormal(mu=ln 120, sigma=0.5)), >= 1 (Box–Muller)
+static uint32_t synth_norm(uint64_t seed, uint32_t doc) {
+  const double u1 = u01(hash3(seed, 0x6E6F726Du, doc));
+  const double u2 = u01(hash3(seed, 0x6E6F726Eu, doc));
+  const double z =
+    std::sqrt(-2.0 * std::log(u1)) * std::cos(2.0 * M_PI * u2);
+  const double v = std::exp(std::log(120.0) + 0.5 * z);
+  return uint32_t(std::max(1.0, std::min(4.0e9, std::round(v))));
+}
+
+static bool synth_member(uint64_t seed, uint32_t term, uint32_t doc,
+                         double sel) {
+  return u01(hash3(seed, 0x6D656D62u ^ (uint64_t(term) << 32), doc)) < sel;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// exported C ABI (host side)
+// ---------------------------------------------------------------------------
+extern "C" {
+
+int sdb_host_encode_doc_block(const uint32_t* docs, uint32_t len,
+                              uint32_t prev, uint8_t* out,
+                              uint32_t* out_size) {
+  if (!docs || !out || !out_size || len < 1 || len > kBlock) return -1;
+  *out_size = encode_doc_block(docs, len, prev, out);
+  return 0;
+}
+int sdb_host_decode_doc_block(const uint8_t* in, uint32_t len, uint32_t prev,
+                              uint32_t* out, uint32_t* consumed) {
+  if (!in || !out || len < 1 || len > kBlock) return -1;
+  const uint32_t c = decode_doc_block(in, len, prev, out);
+  if (consumed) *consumed = c;
+  return 0;
+}
+int sdb_host_encode_freq_block(const uint32_t* freqs, uint32_t len,
+                               uint8_t* out, uint32_t* out_size) {
+  if (!freqs || !out || !out_size || len < 1 || len > kBlock) return -1;
+  *out_size = encode_freq_block(freqs, len, out);
+  return 0;
+}
+int sdb_host_decode_freq_block(const uint8_t* in, uint32_t len, uint32_t* out,
+                               uint32_t* consumed) {
+  if (!in || !out || len < 1 || len > kBlock) return -1;
+  const uint32_t c = decode_freq_block(in, len, out);
+  if (consumed) *consumed = c;
+  return 0;
+}
+
+// Synthetic corpus: per-term postings of term `t` under (seed, sel).
+// If docs==nullptr, only counts df. docs/freqs sized >= df.
+int sdb_host_synth_postings(uint64_t seed, uint32_t doc_count, uint32_t term,
+                            double sel, uint32_t* docs, uint32_t* freqs,
+                            uint32_t* df_out) {
+  uint32_t n = 0;
+  for (uint32_t d = 1; d <= doc_count; ++d) {
+    if (synth_member(seed, term, d, sel)) {
+      if (docs) {
+        docs[n] = d;
+        freqs[n] = synth_freq(seed, term, d);
+      }
+      ++n;
+    }
+  }
+  if (df_out) *df_out = n;
+  return 0;
+}
+
+int sdb_host_synth_norms(uint64_t seed, uint32_t doc_count, uint32_t* norms) {
+  for (uint32_t d = 1; d <= doc_count; ++d) norms[d] = synth_norm(seed, d);
+  norms[0] = 0;
+  return 0;
+}
+
+// Build a serialized segment blob from explicit postings. Arrays:
+//   df[t], docs[t][0..df[t]), freqs[t][...], norms[1..doc_count].
+// Caller frees *blob with sdb_host_blob_free.
+int sdb_host_build_segment(uint32_t doc_count, uint32_t nterms,
+                           const uint32_t* df, const uint32_t* const* docs,
+                           const uint32_t* const* freqs, const uint32_t* norms,
+                           void** blob_out, uint64_t* blob_size) {
+  if (!df || !blob_out || !blob_size) return -1;
+
+  std::vector<SdbTermEntry> terms(nterms);
+  std::vector<SdbBlockDesc> desc;
+  std::vector<uint8_t> payload;
+  payload.reserve(1 << 20);
+
+  uint64_t total_tf = 0;
+  for (uint32_t t = 0; t < nterms; ++t) {
+    auto& te = terms[t];
+    te.desc_begin = desc.size();
+    te.payload_begin = payload.size();
+    te.df = df[t];
+    te.max_freq = 0;
+    te.total_freq = 0;
+    const uint32_t* td = df[t] ? docs[t] : nullptr;
+    const uint32_t* tf = df[t] ? freqs[t] : nullptr;
+    uint32_t prev = 0;  // doc ids start at 1; initial delta base is 0
+    for (uint32_t pos = 0; pos < df[t]; pos += kBlock) {
+      const uint32_t len = std::min(kBlock, df[t] - pos);
+      SdbBlockDesc bd{};
+      bd.prev_doc = prev;
+      bd.last_doc = td[pos + len - 1];
+      bd.len = uint16_t(len);
+      bd.doc_off = uint32_t(payload.size() - te.payload_begin);
+      uint8_t buf[kBlock * 5 + 16];
+      uint32_t sz = encode_doc_block(td + pos, len, prev, buf);
+      payload.insert(payload.end(), buf, buf + sz);
+      bd.freq_off = uint32_t(payload.size() - te.payload_begin);
+      sz = encode_freq_block(tf + pos, len, buf);
+      payload.insert(payload.end(), buf, buf + sz);
+      uint32_t mf = 0, mn = 0xFFFFFFFFu;
+      for (uint32_t i = 0; i < len; ++i) {
+        mf = std::max(mf, tf[pos + i]);
+        te.total_freq += tf[pos + i];
+        if (norms) mn = std::min(mn, norms[td[pos + i]]);
+      }
+      bd.max_freq = mf;
+      bd.min_norm = mn;
+      te.max_freq = std::max(te.max_freq, mf);
+      desc.push_back(bd);
+      prev = bd.last_doc;
+    }
+    te.desc_end = desc.size();
+    te.payload_end = payload.size();
+  }
+  for (uint32_t d = 1; d <= doc_count; ++d) total_tf += norms ? norms[d] : 1;
+
+  auto align64 = [](uint64_t x) { return (x + 63) & ~63ull; };
+  SdbSegHeader hdr{};
+  hdr.magic = SDB_SEG_MAGIC;
+  hdr.version = 1;
+  hdr.nterms = nterms;
+  hdr.doc_count = doc_count;
+  hdr.docs_with_field = doc_count;
+  hdr.total_term_freq = total_tf;
+  hdr.total_blocks = desc.size();
+  uint64_t off = align64(sizeof(SdbSegHeader));
+  hdr.off_terms = off;
+  off = align64(off + sizeof(SdbTermEntry) * nterms);
+  hdr.off_desc = off;
+  off = align64(off + sizeof(SdbBlockDesc) * desc.size());
+  hdr.off_norms = off;
+  off = align64(off + sizeof(uint32_t) * (uint64_t(doc_count) + 1));
+  hdr.off_payload = off;
+  hdr.payload_size = payload.size();
+  hdr.blob_size = align64(off + payload.size());
+
+  uint8_t* blob = static_cast<uint8_t*>(std::calloc(1, hdr.blob_size));
+  if (!blob) return -4;
+  std::memcpy(blob, &hdr, sizeof(hdr));
+  std::memcpy(blob + hdr.off_terms, terms.data(),
+              sizeof(SdbTermEntry) * nterms);
+  if (!desc.empty())
+    std::memcpy(blob + hdr.off_desc, desc.data(),
+                sizeof(SdbBlockDesc) * desc.size());
+  auto* nb = reinterpret_cast<uint32_t*>(blob + hdr.off_norms);
+  if (norms)
+    std::memcpy(nb, norms, sizeof(uint32_t) * (uint64_t(doc_count) + 1));
+  else
+    for (uint32_t d = 0; d <= doc_count; ++d) nb[d] = 1;
+  if (!payload.empty())
+    std::memcpy(blob + hdr.off_payload, payload.data(), payload.size());
+
+  *blob_out = blob;
+  *blob_size = hdr.blob_size;
+  return 0;
+}
+
+// Convenience: build the whole synthetic segment for docs [doc_lo, doc_hi]
+// of a (seed, selectivities) corpus — doc ids are kept GLOBAL so sharded
+// segments on N ranks score identically to one big segment.
+// NOTE: field stats written into the blob cover ONLY this shard; global BM25
+// stats for sharded execution are supplied via sdb_host_bm25_stats inputs.
+int sdb_host_build_synth_segment(uint64_t seed, uint32_t doc_lo,
+                                 uint32_t doc_hi, uint32_t nterms,
+                                 const double* sel, void** blob_out,
+                                 uint64_t* blob_size) {
+  if (doc_lo < 1 || doc_hi < doc_lo) return -1;
+  const uint32_t n_docs = doc_hi - doc_lo + 1;
+  std::vector<std::vector<uint32_t>> docs(nterms), freqs(nterms);
+  for (uint32_t t = 0; t < nterms; ++t) {
+    for (uint32_t d = doc_lo; d <= doc_hi; ++d) {
+      if (synth_member(seed, t, d, sel[t])) {
+        docs[t].push_back(d);
+        freqs[t].push_back(synth_freq(seed, t, d));
+      }
+    }
+  }
+  // norms indexed by LOCAL offset +1? No: global doc ids -> segment stores
+  // norms for [doc_lo-1 .. doc_hi] at positions [doc_lo-1 .. doc_hi]; to keep
+  // the blob dense we build with doc ids SHIFTED to 1..n_docs and record the
+  // shard base separately at query time. Shift here:
+  std::vector<uint32_t> norms(size_t(n_docs) + 1);
+  norms[0] = 0;
+  for (uint32_t d = 0; d < n_docs; ++d)
+    norms[d + 1] = synth_norm(seed, doc_lo + d);
+  std::vector<const uint32_t*> dp(nterms), fp(nterms);
+  std::vector<uint32_t> df(nterms);
+  std::vector<std::vector<uint32_t>> sdocs(nterms);
+  for (uint32_t t = 0; t < nterms; ++t) {
+    df[t] = uint32_t(docs[t].size());
+    sdocs[t].resize(docs[t].size());
+    for (size_t i = 0; i < docs[t].size(); ++i)
+      sdocs[t][i] = docs[t][i] - (doc_lo - 1);
+    dp[t] = sdocs[t].data();
+    fp[t] = freqs[t].data();
+  }
+  return sdb_host_build_segment(n_docs, nterms, df.data(), dp.data(),
+                                fp.data(), norms.data(), blob_out, blob_size);
+}
+
+void sdb_host_blob_free(void* blob) { std::free(blob); }
+
+int sdb_host_segment_parse(const void* blob, uint64_t size,
+                           SdbSegmentView* out) {
+  if (!blob || size < sizeof(SdbSegHeader)) return -5;
+  const auto* hdr = static_cast<const SdbSegHeader*>(blob);
+  if (hdr->magic != SDB_SEG_MAGIC || hdr->version != 1 ||
+      hdr->blob_size > size)
+    return -5;
+  const auto* base = static_cast<const uint8_t*>(blob);
+  out->hdr = hdr;
+  out->terms = reinterpret_cast<const SdbTermEntry*>(base + hdr->off_terms);
+  out->desc = reinterpret_cast<const SdbBlockDesc*>(base + hdr->off_desc);
+  out->norms = reinterpret_cast<const uint32_t*>(base + hdr->off_norms);
+  out->payload = base + hdr->off_payload;
+  return 0;
+}
+
+// BM25 stats — BM25::collect (search/bm25.cpp:279-306): idf accumulated via
+// double log1p narrowed to f32; norm_const = k(1-b); norm_length = k*b/avgDL
+// with avgDL = (f32)total_term_freq / (f32)docs_with_field.
+void sdb_host_bm25_stats(uint64_t docs_with_field, uint64_t docs_with_term,
+                         uint64_t total_term_freq, float k, float b,
+                         float* idf, float* norm_const, float* norm_length) {
+  *idf = float(std::log1p(
+    (double(docs_with_field - docs_with_term) + 0.5) /
+    (double(docs_with_term) + 0.5)));
+  const float kb = k * b;
+  if (b == 0.0f) {  // BM15: stats->norm_const = k (bm25.cpp:296-299)
+    *norm_const = k;
+    *norm_length = 0.0f;
+    return;
+  }
+  *norm_const = k - kb;
+  if (total_term_freq && docs_with_field) {
+    const float avg_dl = float(total_term_freq) / float(docs_with_field);
+    *norm_length = kb / avg_dl;
+  } else {
+    *norm_length = kb;
+  }
+}
+
+// Final host-side top-k select over candidate (score,doc,segment) triples —
+// the PrepareEmitBuffer analogue (duckdb_search_full_scan.cpp:1945-2000) with
+// the deterministic tie order of DESIGN.md: (score desc, segment asc, doc
+// asc). Acceptance threshold mirrors doc_collector.hpp:58: score > FLT_MIN.
+typedef struct SdbScoreDocC {
+  float score;
+  uint32_t doc;
+  uint32_t segment_idx;
+} SdbScoreDocC;
+
+int sdb_host_topk_select(const SdbScoreDocC* cands, uint64_t n, uint32_t k,
+                         SdbScoreDocC* out, uint32_t* out_count) {
+  std::vector<SdbScoreDocC> v;
+  v.reserve(n);
+  const float kFltMin = 1.17549435e-38f;  // std::numeric_limits<float>::min()
+  for (uint64_t i = 0; i < n; ++i)
+    if (cands[i].score > kFltMin) v.push_back(cands[i]);
+  auto cmp = [](const SdbScoreDocC& a, const SdbScoreDocC& b) {
+    if (a.score != b.score) return a.score > b.score;
+    if (a.segment_idx != b.segment_idx) return a.segment_idx < b.segment_idx;
+    return a.doc < b.doc;
+  };
+  const size_t kk = std::min<size_t>(k, v.size());
+  std::partial_sort(v.begin(), v.begin() + kk, v.end(), cmp);
+  std::copy(v.begin(), v.begin() + kk, out);
+  *out_count = uint32_t(kk);
+  return 0;
+}
+
+}  // extern "C"
