@@ -1,0 +1,242 @@
+"""Segment + BM25 + top-k oracle tests (CPU).
+
+Pins:
+  - the transcribed bm25_test golden corpus rank orders (tests/golden/)
+  - oracle exact vs mechanics vs multithreaded paths agree
+  - oracle scores vs an independent numpy fp32/fp64 scorer
+  - sharded execution with injected global stats == single-segment result
+  - full-term decode round trip on synthetic corpora
+"""
+
+import json
+import os
+
+import numpy as np
+import pytest
+
+import serenedb_amd as sa
+from oracle import pyoracle as po
+
+GOLDEN = os.path.join(os.path.dirname(__file__), "golden")
+
+
+def build_term_corpus(docs_fields):
+    """docs_fields: list (per doc, 1-based ids implied) of token lists.
+    Returns (blob, vocab dict term->idx, norms)."""
+    vocab = {}
+    for f in docs_fields:
+        for t in f:
+            vocab.setdefault(t, len(vocab))
+    postings = [[] for _ in vocab]
+    for d, f in enumerate(docs_fields, start=1):
+        for t in sorted(set(f), key=lambda x: vocab[x]):
+            postings[vocab[t]].append((d, f.count(t)))
+    plist = []
+    for pl in postings:
+        docs = np.array([d for d, _ in pl], dtype=np.uint32)
+        freqs = np.array([c for _, c in pl], dtype=np.uint32)
+        plist.append((docs, freqs))
+    norms = np.zeros(len(docs_fields) + 1, dtype=np.uint32)
+    for d, f in enumerate(docs_fields, start=1):
+        norms[d] = len(f)
+    blob = sa.build_segment(len(docs_fields), plist, norms)
+    return blob, vocab, norms
+
+
+def numpy_bm25_topk(docs_fields, terms, k1=1.2, b=0.75, fp64=False):
+    """independent reference scorer (term-major fp32, or fp64)."""
+    N = len(docs_fields)
+    lens = np.array([len(f) for f in docs_fields], dtype=np.uint32)
+    ttf = int(lens.sum())
+    dt = np.float64 if fp64 else np.float32
+    scores = np.zeros(N, dtype=dt)
+    matched = np.zeros(N, dtype=bool)
+    for t in terms:
+        df = sum(1 for f in docs_fields if t in f)
+        if df == 0:
+            continue
+        idf = dt(np.log1p((np.float64(N - df) + 0.5) / (np.float64(df) + 0.5)))
+        nc = dt(np.float32(k1) - np.float32(k1) * np.float32(b))
+        avg = dt(np.float32(ttf) / np.float32(N))
+        nl = dt(np.float32(np.float32(k1) * np.float32(b)) / avg)
+        num = dt(np.float32(1.0) * np.float32(k1 + 1)) * idf
+        for d, f in enumerate(docs_fields):
+            freq = f.count(t)
+            if freq:
+                matched[d] = True
+                c1 = nc + nl * dt(lens[d])
+                scores[d] += num - num * c1 / (c1 + dt(freq))
+    order = sorted(np.nonzero(matched)[0],
+                   key=lambda d: (-float(scores[d]), d))
+    return order, scores
+
+
+@pytest.mark.parametrize("case_idx", [0, 1, 2])
+def test_bm25_golden_rank_order(case_idx):
+    g = json.load(open(os.path.join(GOLDEN,
+                                    "bm25_simple_sequential_order.json")))
+    docs_fields = g["docs"]
+    case = g["cases"][case_idx]
+    blob, vocab, _ = build_term_corpus(docs_fields)
+    term_idx = [vocab[t] for t in case["terms"]]
+    boosts = [1.0] * len(term_idx)
+    hits, total = po.execute_topk([blob], term_idx, boosts, k=8)
+    got_seq = [int(h["doc"]) - 1 for h in hits]  # seq = doc-1 in this corpus
+    assert got_seq == case["expected_seq_order"], case["cite"]
+    # mechanics path agrees
+    mhits, mtotal = po.execute_topk_mech([blob], term_idx, boosts, k=8)
+    assert total == mtotal
+    assert [int(h["doc"]) for h in mhits[:len(got_seq)]] == \
+        [g + 1 for g in got_seq]
+    # independent numpy scorer agrees bit-for-bit on scores
+    order, scores = numpy_bm25_topk(docs_fields, case["terms"])
+    for h in hits:
+        assert h["score"] == np.float32(scores[int(h["doc"]) - 1])
+
+
+def synth_corpus(seed, doc_count, sels):
+    postings = [sa.synth_postings(seed, doc_count, t, s)
+                for t, s in enumerate(sels)]
+    norms = sa.synth_norms(seed, doc_count)
+    blob = sa.build_segment(doc_count, postings, norms)
+    return blob, postings, norms
+
+
+def test_segment_decode_roundtrip():
+    blob, postings, _ = synth_corpus(42, 50_000, [0.05, 0.02, 0.5, 0.001])
+    for t, (docs, freqs) in enumerate(postings):
+        ddocs, dfreqs = po.decode_term(blob, t, len(docs))
+        np.testing.assert_array_equal(ddocs, docs)
+        np.testing.assert_array_equal(dfreqs, freqs)
+
+
+def brute_topk(postings, norms, doc_count, sels_used, k, min_match=1,
+               k1=1.2, b=0.75):
+    """independent dense fp32 scorer over raw postings"""
+    ttf = int(norms[1:].sum())
+    scores = np.zeros(doc_count + 1, dtype=np.float32)
+    cnt = np.zeros(doc_count + 1, dtype=np.int32)
+    for (docs, freqs) in postings:
+        df = len(docs)
+        if df == 0:
+            continue
+        idf = np.float32(np.log1p(
+            (np.float64(doc_count - df) + 0.5) / (np.float64(df) + 0.5)))
+        nc = np.float32(np.float32(k1) - np.float32(k1) * np.float32(b))
+        avg = np.float32(np.float32(ttf) / np.float32(doc_count))
+        nl = np.float32(np.float32(np.float32(k1) * np.float32(b)) / avg)
+        num = np.float32(np.float32(1.0) * np.float32(k1 + 1)) * idf
+        c1 = nc + nl * norms[docs].astype(np.float32)
+        contrib = num - num * c1 / (c1 + freqs.astype(np.float32))
+        scores[docs] += contrib
+        cnt[docs] += 1
+    match_docs = np.nonzero(cnt >= min_match)[0]
+    flt_min = np.float32(1.17549435e-38)
+    acc = match_docs[scores[match_docs] > flt_min]
+    order = sorted(acc, key=lambda d: (-float(scores[d]), d))[:k]
+    return order, scores, len(match_docs)
+
+
+@pytest.mark.parametrize("seed,mm", [(42, 1), (43, 1), (44, 2), (45, 4)])
+def test_topk_exact_vs_brute(seed, mm):
+    doc_count = 30_000
+    sels = [0.1, 0.05, 0.02, 0.01]
+    blob, postings, norms = synth_corpus(seed, doc_count, sels)
+    k = 100
+    term_idx = list(range(4))
+    boosts = [1.0] * 4
+    hits, total = po.execute_topk([blob], term_idx, boosts, k, min_match=mm)
+    order, scores, nmatch = brute_topk(postings, norms, doc_count, sels, k,
+                                       min_match=mm)
+    assert total == nmatch
+    assert [int(h["doc"]) for h in hits] == [int(d) for d in order]
+    for h in hits:  # fp32 bit-exact (term-major order both sides)
+        assert h["score"] == scores[int(h["doc"])], int(h["doc"])
+
+
+def test_mech_equals_exact():
+    doc_count = 30_000
+    blob, postings, norms = synth_corpus(7, doc_count, [0.08, 0.03, 0.01])
+    for k in (1, 10, 100, 1000):
+        hits, total = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, k)
+        mhits, mtotal = po.execute_topk_mech([blob], [0, 1, 2], [1.0] * 3, k)
+        assert total == mtotal
+        n = len(hits)
+        np.testing.assert_array_equal(hits["doc"], mhits["doc"][:n])
+        np.testing.assert_array_equal(hits["score"], mhits["score"][:n])
+
+
+def test_mt_equals_exact():
+    doc_count = 100_000
+    blob, postings, norms = synth_corpus(9, doc_count, [0.05, 0.02])
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 200)
+    for nthreads in (1, 4, 8):
+        th, tt = po.execute_topk_mt(blob, [0, 1], [1.0, 1.0], 200,
+                                    nthreads=nthreads)
+        assert tt == total
+        np.testing.assert_array_equal(th["doc"], hits["doc"])
+        np.testing.assert_array_equal(th["score"], hits["score"])
+
+
+def test_sharded_with_global_stats():
+    """Two shards + injected global stats == one segment (multi-GPU path)."""
+    doc_count = 40_000
+    sels = [0.06, 0.02, 0.01]
+    seed = 11
+    full_blob, postings, norms = synth_corpus(seed, doc_count, sels)
+    hits, total = po.execute_topk([full_blob], [0, 1, 2], [1.0] * 3, 150)
+
+    half = doc_count // 2
+    shard_blobs = []
+    for lo, hi in ((1, half), (half + 1, doc_count)):
+        shard_blobs.append(sa.build_synth_segment(seed, lo, hi, sels))
+    dwt = [len(d) for d, _ in postings]
+    gstats = (doc_count, int(norms[1:].sum()), dwt)
+    all_cands = []
+    totals = 0
+    for si, (blob, base) in enumerate(zip(shard_blobs, (0, half))):
+        h, t = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 150,
+                               global_stats=gstats)
+        totals += t
+        for x in h:
+            all_cands.append((float(x["score"]), int(x["doc"]) + base))
+    # merge (allgather + host nth_element analogue)
+    all_cands.sort(key=lambda sd: (-sd[0], sd[1]))
+    got = all_cands[:150]
+    assert totals == total
+    assert [d for _, d in got] == [int(d) for d in hits["doc"]]
+    assert [s for s, _ in got] == [float(s) for s in hits["score"]]
+
+
+def test_empty_and_edge_cases():
+    # term with no postings; k > matches; single-doc segment
+    doc_count = 1000
+    docs = np.array([500], dtype=np.uint32)
+    freqs = np.array([3], dtype=np.uint32)
+    norms = np.ones(doc_count + 1, dtype=np.uint32)
+    blob = sa.build_segment(doc_count, [(docs, freqs),
+                                        (np.array([], dtype=np.uint32),
+                                         np.array([], dtype=np.uint32))],
+                            norms)
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 10)
+    assert total == 1
+    assert len(hits) == 1 and hits[0]["doc"] == 500
+    # empty term alone
+    hits, total = po.execute_topk([blob], [1], [1.0], 10)
+    assert total == 0 and len(hits) == 0
+    # conjunction with empty term -> empty
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 10,
+                                  min_match=2)
+    assert total == 0 and len(hits) == 0
+
+
+def test_boost_zero_scores_rejected():
+    """boost 0 -> score 0 -> not accepted (threshold FLT_MIN,
+    doc_collector.hpp:58) but still counted as a match (TotalMatches)."""
+    doc_count = 100
+    docs = np.arange(1, 51, dtype=np.uint32)
+    freqs = np.ones(50, dtype=np.uint32)
+    blob = sa.build_segment(doc_count, [(docs, freqs)], None)
+    hits, total = po.execute_topk([blob], [0], [0.0], 10)
+    assert total == 50
+    assert len(hits) == 0
