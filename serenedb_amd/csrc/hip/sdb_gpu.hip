@@ -1,0 +1,783 @@
+// sdb_gpu.hip — MI355X-native (gfx950/CDNA4) query path of the SereneDB
+// search hot loop, behind the C ABI of include/sdb_gpu.h.
+//
+// Design:
+//   One workgroup owns a WIN-doc window of the doc-id space — the GPU-shaped
+//   descendant of BlockDisjunction's 4096-doc window
+//   (search/block_disjunction.hpp:122-732): an LDS fp32 score window plus a
+//   u8 match-count window. Terms are processed in fixed order with a barrier
+//   between them (term-major fp32 merge order -> scores bit-identical to the
+//   CPU oracle; SURVEY.md §7 "hard parts"). Within a term phase, each of the
+//   8 waves decodes whole 128-doc postings blocks (the FormatTraits128
+//   families, format_block_128.hpp:446-636) straight out of HBM and
+//   accumulates BM25 partial scores (search/bm25.cpp:89-109) into the LDS
+//   window — no atomics needed: a term's doc ids are unique.
+//
+//   Top-k selection mirrors the reference's shared kth-score threshold
+//   (duckdb_search_full_scan.cpp:1884-1921): each window builds a 256-bin
+//   LDS histogram of its match scores, derives a provable lower bound on the
+//   global k-th score (if a window holds >= k scores >= tau, the global k-th
+//   is >= tau), publishes it via a device-wide atomicMax (float-as-uint,
+//   scores are non-negative), and appends only candidates >= the current
+//   global bound. The host then runs the exact final select
+//   (PrepareEmitBuffer analogue) over the small candidate set. Every drop is
+//   justified by a bound <= the true k-th score, so the final top-k is exact
+//   and deterministic regardless of workgroup scheduling.
+//
+// No CPU fallback exists here: every entry point returns SDB_ERR_NO_GPU
+// when no device is present.
+//
+// Compile: hipcc --offload-arch=gfx950 -O3 -ffp-contract=off (fp32 BM25
+// bit-parity with the oracle requires no contraction).
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cfloat>
+#include <cmath>
+#include <cstdint>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include "../../../include/sdb_gpu.h"
+#include "sdb_internal.h"
+
+// ---------------------------------------------------------------------------
+// tunables
+// ---------------------------------------------------------------------------
+#ifndef SDB_WIN_DOCS
+#define SDB_WIN_DOCS 24576u  // docs per workgroup window (96 KB f32 + 24 KB u8)
+#endif
+#define SDB_NTHREADS 512u    // 8 waves
+#define SDB_NWAVES (SDB_NTHREADS / 64u)
+#define SDB_MAX_TERMS 32u
+#define SDB_HIST_BINS 256u
+#define SDB_CAND_CAP (64u * 1024u * 1024u)  // 64M candidates (768 MB)
+
+#define HIP_CHECK(x)                        \
+  do {                                      \
+    hipError_t _e = (x);                    \
+    if (_e == hipErrorNoDevice) return SDB_ERR_NO_GPU; \
+    if (_e != hipSuccess) return SDB_ERR_HIP;          \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// device-side structures
+// ---------------------------------------------------------------------------
+struct SdbGpuSegment {
+  SdbBlockDesc* desc;   // device
+  uint8_t* payload;     // device
+  uint32_t* norms;      // device, doc_count+1
+  SdbTermEntry* terms_host;  // host copy of term table
+  SdbSegHeader hdr;     // host copy
+};
+
+// ---------------------------------------------------------------------------
+// kernel helpers
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint32_t ld_u32_un(const uint8_t* p) {
+  // aligned-pair unaligned read (payload blocks start at arbitrary bytes;
+  // blob has a 64 B tail pad so the +4 word is always in bounds)
+  const uintptr_t a = (uintptr_t)p;
+  const uint32_t* w = (const uint32_t*)(a & ~(uintptr_t)3);
+  const uint32_t sh = ((uint32_t)a & 3u) * 8u;
+  const uint32_t lo = w[0];
+  if (sh == 0) return lo;
+  const uint32_t hi = w[1];
+  return (lo >> sh) | (hi << (32 - sh));
+}
+
+__device__ __forceinline__ uint64_t ld_u64_un(const uint8_t* p) {
+  return (uint64_t)ld_u32_un(p) | ((uint64_t)ld_u32_un(p + 4) << 32);
+}
+
+// inclusive wave-scan (64 lanes) of a uint32
+__device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v, int lane) {
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) {
+    const uint32_t n = __shfl_up(v, off, 64);
+    if (lane >= off) v += n;
+  }
+  return v;
+}
+
+// vertical-layout delta extract: value i of a 128-block, b bits
+// (simdcomp d1 layout — see include/sdb_format.h)
+__device__ __forceinline__ uint32_t extract_packed(const uint8_t* base,
+                                                   uint32_t bits,
+                                                   uint32_t i) {
+  const uint32_t mask = bits >= 32 ? 0xFFFFFFFFu : (1u << bits) - 1u;
+  const uint32_t c = i & 3u, g = i >> 2;
+  const uint32_t bit = g * bits;
+  const uint32_t w = bit >> 5, sh = bit & 31u;
+  uint64_t v = (uint64_t)ld_u32_un(base + 4u * (w * 4 + c)) >> sh;
+  if (sh + bits > 32)
+    v |= (uint64_t)ld_u32_un(base + 4u * ((w + 1) * 4 + c)) << (32 - sh);
+  return (uint32_t)v & mask;
+}
+
+// Decode one doc block (ReadTailDelta, format_block_128.hpp:476-559) with
+// one wave into LDS scratch. Returns nothing; scratch[0..len) = doc ids.
+__device__ void decode_doc_block_wave(const uint8_t* p, uint32_t len,
+                                      uint32_t prev, int lane,
+                                      uint32_t* scratch) {
+  const uint32_t tag = p[0];
+  if (tag >= SDB_DE_DELTA_BITPACK_02) {  // delta bitpack, full blocks
+    const uint32_t bits = tag - SDB_DE_DELTA_BITPACK_02 + 2;
+    const uint8_t* base = p + 1;
+    const uint32_t i0 = 2u * lane;
+    const uint32_t d0 = extract_packed(base, bits, i0);
+    const uint32_t d1 = extract_packed(base, bits, i0 + 1);
+    const uint32_t pair = d0 + d1;
+    const uint32_t incl = wave_incl_scan(pair, lane);
+    const uint32_t excl = incl - pair;
+    scratch[i0] = prev + excl + d0;
+    scratch[i0 + 1] = prev + excl + pair;
+    return;
+  }
+  switch (tag) {
+    case SDB_DE_VALUES: {
+      for (uint32_t i = lane; i < len; i += 64)
+        scratch[i] = ld_u32_un(p + 1 + 4u * i);
+      break;
+    }
+    case SDB_DE_DELTA_ALL_SAME_08:
+    case SDB_DE_DELTA_ALL_SAME_16:
+    case SDB_DE_DELTA_ALL_SAME_32: {
+      const uint32_t nb = tag == SDB_DE_DELTA_ALL_SAME_08
+                            ? 1u
+                            : (tag == SDB_DE_DELTA_ALL_SAME_16 ? 2u : 4u);
+      uint32_t v = ld_u32_un(p + 1);
+      if (nb < 4) v &= (1u << (8 * nb)) - 1u;
+      // FillSameDelta: out[i] = prev + v + v*i
+      for (uint32_t i = lane; i < len; i += 64)
+        scratch[i] = prev + v + v * i;
+      break;
+    }
+    case SDB_DE_FOR_BITSET: {
+      const uint32_t words = p[1];
+      uint64_t word = 0;
+      uint32_t cnt = 0;
+      if ((uint32_t)lane < words) {
+        word = ld_u64_un(p + 2 + 8u * lane);
+        cnt = (uint32_t)__popcll(word);
+      }
+      const uint32_t incl = wave_incl_scan(cnt, lane);
+      uint32_t idx = incl - cnt;
+      const uint32_t off = prev + (uint32_t)lane * 64u;
+      while (word) {
+        scratch[idx++] = off + (uint32_t)__ffsll((long long)word) - 1;
+        word &= word - 1;
+      }
+      break;
+    }
+    case SDB_DE_STREAMVBYTE1234:
+    case SDB_DE_DELTA_STREAMVBYTE1234: {
+      // tails only (len < 128). ctrl at p+3, data after (len+3)/4 ctrl bytes
+      const uint8_t* ctrl = p + 3;
+      const uint8_t* data = ctrl + (len + 3) / 4;
+      const uint32_t i0 = 2u * lane, i1 = i0 + 1;
+      auto vlen = [&](uint32_t i) -> uint32_t {
+        return i < len ? ((ctrl[i >> 2] >> ((i & 3) * 2)) & 3u) + 1u : 0u;
+      };
+      const uint32_t l0 = vlen(i0), l1 = vlen(i1);
+      const uint32_t pair = l0 + l1;
+      const uint32_t incl = wave_incl_scan(pair, lane);
+      const uint32_t excl = incl - pair;
+      auto readv = [&](uint32_t off, uint32_t n) -> uint32_t {
+        uint32_t v = 0;
+        for (uint32_t b = 0; b < n; ++b)
+          v |= (uint32_t)data[off + b] << (8 * b);
+        return v;
+      };
+      uint32_t v0 = i0 < len ? readv(excl, l0) : 0;
+      uint32_t v1 = i1 < len ? readv(excl + l0, l1) : 0;
+      if (tag == SDB_DE_DELTA_STREAMVBYTE1234) {
+        const uint32_t dpair = v0 + v1;
+        const uint32_t dincl = wave_incl_scan(dpair, lane);
+        const uint32_t dexcl = dincl - dpair;
+        if (i0 < len) scratch[i0] = prev + dexcl + v0;
+        if (i1 < len) scratch[i1] = prev + dexcl + dpair;
+      } else {
+        if (i0 < len) scratch[i0] = v0;
+        if (i1 < len) scratch[i1] = v1;
+      }
+      break;
+    }
+    default:
+      break;
+  }
+}
+
+// Decode one freq block (ReadTail, format_block_128.hpp:568-636), one wave.
+__device__ void decode_freq_block_wave(const uint8_t* p, uint32_t len,
+                                       int lane, uint32_t* scratch) {
+  const uint32_t tag = p[0];
+  if (tag >= SDB_E_BITPACK_01) {
+    const uint32_t bits = tag - SDB_E_BITPACK_01 + 1;
+    const uint8_t* base = p + 1;
+    const uint32_t i0 = 2u * lane;
+    scratch[i0] = extract_packed(base, bits, i0);
+    scratch[i0 + 1] = extract_packed(base, bits, i0 + 1);
+    return;
+  }
+  switch (tag) {
+    case SDB_E_VALUES: {
+      for (uint32_t i = lane; i < len; i += 64)
+        scratch[i] = ld_u32_un(p + 1 + 4u * i);
+      break;
+    }
+    case SDB_E_ALL_SAME_08:
+    case SDB_E_ALL_SAME_16:
+    case SDB_E_ALL_SAME_32: {
+      const uint32_t nb =
+        tag == SDB_E_ALL_SAME_08 ? 1u : (tag == SDB_E_ALL_SAME_16 ? 2u : 4u);
+      uint32_t v = ld_u32_un(p + 1);
+      if (nb < 4) v &= (1u << (8 * nb)) - 1u;
+      for (uint32_t i = lane; i < len; i += 64) scratch[i] = v;
+      break;
+    }
+    case SDB_E_STREAMVBYTE1234: {
+      const uint8_t* ctrl = p + 3;
+      const uint8_t* data = ctrl + (len + 3) / 4;
+      const uint32_t i0 = 2u * lane, i1 = i0 + 1;
+      auto vlen = [&](uint32_t i) -> uint32_t {
+        return i < len ? ((ctrl[i >> 2] >> ((i & 3) * 2)) & 3u) + 1u : 0u;
+      };
+      const uint32_t l0 = vlen(i0), l1 = vlen(i1);
+      const uint32_t pair = l0 + l1;
+      const uint32_t incl = wave_incl_scan(pair, lane);
+      const uint32_t excl = incl - pair;
+      auto readv = [&](uint32_t off, uint32_t n) -> uint32_t {
+        uint32_t v = 0;
+        for (uint32_t b = 0; b < n; ++b)
+          v |= (uint32_t)data[off + b] << (8 * b);
+        return v;
+      };
+      if (i0 < len) scratch[i0] = readv(excl, l0);
+      if (i1 < len) scratch[i1] = readv(excl + l0, l1);
+      break;
+    }
+    default:
+      break;
+  }
+}
+
+// binary searches over the descriptor span of one term:
+// first block with last_doc >= lo  /  first block with prev_doc >= hi
+__device__ __forceinline__ uint64_t lower_bound_last_doc(
+  const SdbBlockDesc* d, uint64_t b, uint64_t e, uint32_t lo) {
+  while (b < e) {
+    const uint64_t m = (b + e) >> 1;
+    if (d[m].last_doc < lo)
+      b = m + 1;
+    else
+      e = m;
+  }
+  return b;
+}
+__device__ __forceinline__ uint64_t lower_bound_prev_doc(
+  const SdbBlockDesc* d, uint64_t b, uint64_t e, uint32_t hi) {
+  while (b < e) {
+    const uint64_t m = (b + e) >> 1;
+    if (d[m].prev_doc < hi)
+      b = m + 1;
+    else
+      e = m;
+  }
+  return b;
+}
+
+// ---------------------------------------------------------------------------
+// the window kernel
+// ---------------------------------------------------------------------------
+struct WindowArgs {
+  const SdbBlockDesc* desc;
+  const uint8_t* payload;
+  const uint32_t* norms;
+  uint32_t doc_count;  // docs 1..doc_count (local ids within this segment)
+  uint32_t nterms;
+  uint32_t min_match;
+  uint32_t k;
+  float smax;          // score upper bound (sum of term num)
+  uint32_t seg_idx;
+  uint32_t* gthresh;   // float bits, monotone under atomicMax
+  SdbScoreDoc* cands;
+  uint32_t* cand_count;
+  uint32_t cand_cap;
+  unsigned long long* total_matches;
+  uint32_t* overflow;
+};
+
+__launch_bounds__(SDB_NTHREADS, 1) __global__
+void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
+  // one extern LDS region, 16B-aligned carves (guide §6 G17)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* swin = (float*)smem;                       // SDB_WIN_DOCS * 4
+  uint8_t* cwin = (uint8_t*)(swin + SDB_WIN_DOCS);  // SDB_WIN_DOCS
+  uint32_t* scratch = (uint32_t*)(cwin + SDB_WIN_DOCS);  // 8 waves * 256
+  uint32_t* hist = scratch + SDB_NWAVES * 256;            // 256
+  uint32_t* shared_misc = hist + SDB_HIST_BINS;           // ranges + bcast
+  // shared_misc layout: [2*SDB_MAX_TERMS] block ranges (lo32/hi32 pairs as
+  // u32 relative counts fit 32 bit? desc indices are u64 — store as u32
+  // offsets relative to term desc_begin; per-term blocks < 2^32) + [2] misc
+
+  const uint32_t tid = threadIdx.x;
+  const int lane = tid & 63;
+  const uint32_t wave = tid >> 6;
+
+  const uint32_t lo = 1u + (uint32_t)blockIdx.x * SDB_WIN_DOCS;
+  if (lo > a.doc_count) return;
+  const uint32_t hi = min(lo + SDB_WIN_DOCS - 1u, a.doc_count);
+  const uint32_t wlen = hi - lo + 1u;
+
+  // zero windows + histogram
+  for (uint32_t i = tid; i < SDB_WIN_DOCS; i += SDB_NTHREADS) swin[i] = 0.0f;
+  for (uint32_t i = tid; i < SDB_WIN_DOCS / 4; i += SDB_NTHREADS)
+    ((uint32_t*)cwin)[i] = 0;
+  for (uint32_t i = tid; i < SDB_HIST_BINS; i += SDB_NTHREADS) hist[i] = 0;
+
+  // block ranges per term (parallel over 2*nterms threads)
+  if (tid < 2 * a.nterms) {
+    const uint32_t t = tid >> 1;
+    const TermDev te = terms[t];
+    uint64_t v;
+    if ((tid & 1) == 0)
+      v = lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, lo);
+    else
+      v = lower_bound_prev_doc(a.desc, te.desc_begin, te.desc_end, hi);
+    shared_misc[tid] = (uint32_t)(v - te.desc_begin);
+  }
+  __syncthreads();
+
+  // term-major phases
+  for (uint32_t t = 0; t < a.nterms; ++t) {
+    const TermDev te = terms[t];
+    const uint32_t b_lo = shared_misc[2 * t];
+    const uint32_t b_hi = shared_misc[2 * t + 1];  // first block fully after
+    const uint8_t* pl = a.payload + te.payload_begin;
+    uint32_t* dbuf = scratch + wave * 256;
+    uint32_t* fbuf = dbuf + 128;
+    const float num = te.num, nc = te.nc, nl = te.nl;
+    for (uint64_t b = te.desc_begin + b_lo + wave;
+         b < te.desc_begin + b_hi; b += SDB_NWAVES) {
+      const SdbBlockDesc d = a.desc[b];
+      decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
+      decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+      // wave-local scratch: writes visible to the same wave after lgkm wait
+      // (compiler inserts); score docs inside the window
+      for (uint32_t j = lane; j < d.len; j += 64) {
+        const uint32_t doc = dbuf[j];
+        if (doc < lo || doc > hi) continue;
+        const uint32_t freq = fbuf[j];
+        const uint32_t norm = a.norms[doc];
+        const float c1 = nc + nl * (float)norm;
+        const float s = num - num * c1 / (c1 + (float)freq);
+        const uint32_t off = doc - lo;
+        swin[off] += s;        // unique doc within the term: no atomics
+        cwin[off] = (uint8_t)(cwin[off] + 1u);
+      }
+    }
+    __syncthreads();  // term-major merge order (bit-exact vs oracle)
+  }
+
+  // histogram of matching scores + local match count
+  const uint32_t mm = a.min_match ? a.min_match : 1u;
+  const float inv_smax = (float)SDB_HIST_BINS / a.smax;
+  uint32_t my_matches = 0;
+  for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
+    if (cwin[off] >= mm) {
+      ++my_matches;
+      uint32_t bin = (uint32_t)(swin[off] * inv_smax);
+      if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
+      atomicAdd(&hist[bin], 1u);
+    }
+  }
+  // reduce match count: wave reduce then one atomic per wave
+  uint32_t wm = my_matches;
+#pragma unroll
+  for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
+  __syncthreads();  // hist complete
+  if (lane == 0 && wm) atomicAdd(a.total_matches, (unsigned long long)wm);
+
+  // local k-th lower bound from the histogram (thread 0)
+  if (tid == 0) {
+    uint32_t cum = 0;
+    uint32_t binfloor = 0;
+    for (int b = SDB_HIST_BINS - 1; b >= 0; --b) {
+      cum += hist[b];
+      if (cum >= a.k) {
+        binfloor = (uint32_t)b;
+        // tau_w = binfloor * smax/256: >= k window scores are >= tau_w,
+        // hence the GLOBAL k-th score >= tau_w -> valid global lower bound
+        const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
+        if (tau > 0.0f) {
+          uint32_t bits;
+          __builtin_memcpy(&bits, &tau, 4);
+          atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
+        }
+        break;
+      }
+    }
+    // broadcast current global threshold
+    shared_misc[2 * SDB_MAX_TERMS] =
+      atomicOr(a.gthresh, 0u);  // atomic read
+  }
+  __syncthreads();
+  float gtau;
+  {
+    const uint32_t bits = shared_misc[2 * SDB_MAX_TERMS];
+    __builtin_memcpy(&gtau, &bits, 4);
+  }
+
+  // append candidates with score >= gtau (ties at the k-th kept)
+  for (uint32_t base = 0; base < wlen; base += SDB_NTHREADS) {
+    const uint32_t off = base + tid;
+    bool acc = false;
+    float s = 0.0f;
+    if (off < wlen && cwin[off] >= mm) {
+      s = swin[off];
+      acc = s >= gtau;
+    }
+    const unsigned long long ball = __ballot(acc);
+    const uint32_t n = (uint32_t)__popcll(ball);
+    uint32_t wbase = 0;
+    if (n) {
+      if (lane == 0) wbase = atomicAdd(a.cand_count, n);
+      wbase = __shfl(wbase, 0, 64);
+      if (wbase + n > a.cand_cap) {
+        if (lane == 0) atomicExch(a.overflow, 1u);
+        continue;
+      }
+      if (acc) {
+        const uint32_t pos =
+          wbase + (uint32_t)__popcll(ball & ((1ull << lane) - 1ull));
+        a.cands[pos].score = s;
+        a.cands[pos].doc = lo + off;
+        a.cands[pos].segment_idx = a.seg_idx;
+      }
+    }
+  }
+}
+
+// full-term decode kernel (parity entry): one wave per 128-doc block
+__global__ void decode_term_kernel(const SdbBlockDesc* desc, uint64_t b0,
+                                   uint64_t nblocks, const uint8_t* payload,
+                                   uint64_t payload_begin, uint32_t* docs,
+                                   uint32_t* freqs) {
+  __shared__ uint32_t dbuf[128];
+  __shared__ uint32_t fbuf[128];
+  const uint64_t b = b0 + blockIdx.x;
+  if (blockIdx.x >= nblocks) return;
+  const int lane = threadIdx.x & 63;
+  const SdbBlockDesc d = desc[b];
+  const uint8_t* pl = payload + payload_begin;
+  decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
+  decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+  __syncthreads();
+  const uint64_t out0 = (uint64_t)blockIdx.x * 128u;
+  for (uint32_t i = lane; i < d.len; i += 64) {
+    docs[out0 + i] = dbuf[i];
+    freqs[out0 + i] = fbuf[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host side (C ABI)
+// ---------------------------------------------------------------------------
+namespace {
+
+int check_gpu() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess || n == 0) return SDB_ERR_NO_GPU;
+  return SDB_OK;
+}
+
+int parse_blob(const void* blob, size_t size, SdbSegHeader* hdr_out) {
+  if (!blob || size < sizeof(SdbSegHeader)) return SDB_ERR_BAD_SEGMENT;
+  SdbSegHeader hdr;
+  std::memcpy(&hdr, blob, sizeof(hdr));
+  if (hdr.magic != SDB_SEG_MAGIC || hdr.version != 1 || hdr.blob_size > size)
+    return SDB_ERR_BAD_SEGMENT;
+  *hdr_out = hdr;
+  return SDB_OK;
+}
+
+}  // namespace
+
+extern "C" {
+
+const char* sdb_gpu_version(void) { return "sdb_gpu 0.1 gfx950"; }
+
+int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
+  if (!out) return SDB_ERR_INVALID;
+  int rc = check_gpu();
+  if (rc) return rc;
+  HIP_CHECK(hipSetDevice(device));
+  auto* ctx = new SdbGpuCtx{};
+  ctx->device = device;
+  HIP_CHECK(hipStreamCreate(&ctx->stream));
+  HIP_CHECK(hipMalloc(&ctx->d_cands, sizeof(SdbScoreDoc) * (size_t)SDB_CAND_CAP));
+  HIP_CHECK(hipMalloc(&ctx->d_cand_count, 4));
+  HIP_CHECK(hipMalloc(&ctx->d_total_matches, 8));
+  HIP_CHECK(hipMalloc(&ctx->d_gthresh, 4));
+  HIP_CHECK(hipMalloc(&ctx->d_overflow, 4));
+  HIP_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
+  HIP_CHECK(hipHostMalloc(&ctx->h_counts, 8));
+  HIP_CHECK(hipHostMalloc(&ctx->h_matches, 8));
+  HIP_CHECK(hipEventCreate(&ctx->ev_a));
+  HIP_CHECK(hipEventCreate(&ctx->ev_b));
+  *out = ctx;
+  return SDB_OK;
+}
+
+int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
+  if (!ctx) return SDB_ERR_INVALID;
+  hipFree(ctx->d_cands);
+  hipFree(ctx->d_cand_count);
+  hipFree(ctx->d_total_matches);
+  hipFree(ctx->d_gthresh);
+  hipFree(ctx->d_overflow);
+  hipFree(ctx->d_terms);
+  hipHostFree(ctx->h_counts);
+  hipHostFree(ctx->h_matches);
+  hipEventDestroy(ctx->ev_a);
+  hipEventDestroy(ctx->ev_b);
+  hipStreamDestroy(ctx->stream);
+  delete ctx;
+  return SDB_OK;
+}
+
+int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
+                         SdbGpuSegment** out) {
+  if (!ctx || !out) return SDB_ERR_INVALID;
+  SdbSegHeader hdr;
+  int rc = parse_blob(blob, blob_size, &hdr);
+  if (rc) return rc;
+  auto* seg = new SdbGpuSegment{};
+  seg->hdr = hdr;
+  const uint8_t* base = (const uint8_t*)blob;
+  seg->terms_host = (SdbTermEntry*)std::malloc(sizeof(SdbTermEntry) * hdr.nterms);
+  std::memcpy(seg->terms_host, base + hdr.off_terms,
+              sizeof(SdbTermEntry) * hdr.nterms);
+  HIP_CHECK(hipMalloc(&seg->desc, sizeof(SdbBlockDesc) * hdr.total_blocks + 16));
+  HIP_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 64));
+  HIP_CHECK(
+    hipMalloc(&seg->norms, sizeof(uint32_t) * ((size_t)hdr.doc_count + 1)));
+  HIP_CHECK(hipMemcpy(seg->desc, base + hdr.off_desc,
+                      sizeof(SdbBlockDesc) * hdr.total_blocks,
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(seg->payload, base + hdr.off_payload, hdr.payload_size,
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(seg->norms, base + hdr.off_norms,
+                      sizeof(uint32_t) * ((size_t)hdr.doc_count + 1),
+                      hipMemcpyHostToDevice));
+  *out = seg;
+  return SDB_OK;
+}
+
+int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
+  if (!ctx || !seg) return SDB_ERR_INVALID;
+  hipFree(seg->desc);
+  hipFree(seg->payload);
+  hipFree(seg->norms);
+  std::free(seg->terms_host);
+  delete seg;
+  return SDB_OK;
+}
+
+int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                         uint32_t nsegs, const SdbQueryPlan* plan, uint32_t k,
+                         SdbScoreDoc* hits, uint32_t* out_count,
+                         uint64_t* total_matches) {
+  if (!ctx || !segs || !plan || !hits || !out_count || !total_matches ||
+      plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS || k == 0)
+    return SDB_ERR_INVALID;
+
+  // ---- PreparePhase analogue: global stats (double -> f32, bm25.cpp) ----
+  uint64_t g_dwf = plan->g_docs_with_field;
+  uint64_t g_ttf = plan->g_total_term_freq;
+  std::vector<uint64_t> g_dwt(plan->nterms, 0);
+  if (g_dwf == 0) {
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      g_dwf += segs[s]->hdr.docs_with_field;
+      g_ttf += segs[s]->hdr.total_term_freq;
+      for (uint32_t t = 0; t < plan->nterms; ++t) {
+        const uint32_t ti = plan->terms[t].term_idx;
+        if (ti >= segs[s]->hdr.nterms) return SDB_ERR_INVALID;
+        g_dwt[t] += segs[s]->terms_host[ti].df;
+      }
+    }
+  } else {
+    if (!plan->g_docs_with_term) return SDB_ERR_INVALID;
+    for (uint32_t t = 0; t < plan->nterms; ++t)
+      g_dwt[t] = plan->g_docs_with_term[t];
+  }
+  const float k1 = plan->k1, b = plan->b;
+  std::vector<float> idf(plan->nterms), num(plan->nterms);
+  float nc, nl;
+  float smax = 0.0f;
+  for (uint32_t t = 0; t < plan->nterms; ++t) {
+    if (g_dwt[t] == 0) {
+      idf[t] = 0.0f;
+      num[t] = 0.0f;
+      continue;
+    }
+    // BM25::collect (bm25.cpp:288-306)
+    idf[t] = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
+                          ((double)g_dwt[t] + 0.5));
+    num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
+    smax += num[t] > 0 ? num[t] : 0.0f;
+  }
+  {
+    const float kb = k1 * b;
+    if (b == 0.0f) {
+      nc = k1;
+      nl = 0.0f;
+    } else {
+      nc = k1 - kb;
+      nl = (g_ttf && g_dwf) ? kb / ((float)g_ttf / (float)g_dwf) : kb;
+    }
+  }
+  if (smax <= 0.0f) smax = FLT_MIN;
+
+  // ---- reset device state ----
+  HIP_CHECK(hipMemsetAsync(ctx->d_cand_count, 0, 4, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_total_matches, 0, 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_gthresh, 0, 4, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
+
+  const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
+                           SDB_NWAVES * 256 * 4 + SDB_HIST_BINS * 4 +
+                           (2 * SDB_MAX_TERMS + 2) * 4;
+
+  HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+  for (uint32_t s = 0; s < nsegs; ++s) {
+    SdbGpuSegment* seg = segs[s];
+    // per-segment term table (desc ranges differ per segment)
+    TermDev tdev[SDB_MAX_TERMS];
+    for (uint32_t t = 0; t < plan->nterms; ++t) {
+      const SdbTermEntry& te = seg->terms_host[plan->terms[t].term_idx];
+      tdev[t].desc_begin = te.desc_begin;
+      tdev[t].desc_end = te.desc_end;
+      tdev[t].payload_begin = te.payload_begin;
+      tdev[t].num = num[t];
+      tdev[t].nc = nc;
+      tdev[t].nl = nl;
+    }
+    HIP_CHECK(hipMemcpyAsync(ctx->d_terms, tdev,
+                             sizeof(TermDev) * plan->nterms,
+                             hipMemcpyHostToDevice, ctx->stream));
+    WindowArgs a{};
+    a.desc = seg->desc;
+    a.payload = seg->payload;
+    a.norms = seg->norms;
+    a.doc_count = seg->hdr.doc_count;
+    a.nterms = plan->nterms;
+    a.min_match = plan->min_match ? plan->min_match : 1;
+    a.k = k;
+    a.smax = smax;
+    a.seg_idx = s;
+    a.gthresh = ctx->d_gthresh;
+    a.cands = ctx->d_cands;
+    a.cand_count = ctx->d_cand_count;
+    a.cand_cap = SDB_CAND_CAP;
+    a.total_matches = ctx->d_total_matches;
+    a.overflow = ctx->d_overflow;
+    const uint32_t nwin =
+      (seg->hdr.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
+    hipLaunchKernelGGL(topk_window_kernel, dim3(nwin), dim3(SDB_NTHREADS),
+                       lds_bytes, ctx->stream, a, ctx->d_terms);
+    HIP_CHECK(hipGetLastError());
+    if (nsegs > 1) HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    // (sync between segments: d_terms is reused; single-segment path keeps
+    //  everything async until readback)
+  }
+
+  HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+
+  // ---- readback candidates; exact final select on host ----
+  HIP_CHECK(hipMemcpyAsync(ctx->h_counts, ctx->d_cand_count, 4,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(ctx->h_counts + 1, ctx->d_overflow, 4,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(ctx->h_matches, ctx->d_total_matches, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  {
+    float ms = 0.0f;
+    HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_a, ctx->ev_b));
+    ctx->last_kernel_ms = (double)ms;
+  }
+  if (ctx->h_counts[1]) return SDB_ERR_OOM;  // candidate overflow
+  const uint32_t ncand = ctx->h_counts[0];
+  std::vector<SdbScoreDoc> cands(ncand);
+  if (ncand) {
+    HIP_CHECK(hipMemcpy(cands.data(), ctx->d_cands,
+                        sizeof(SdbScoreDoc) * ncand, hipMemcpyDeviceToHost));
+  }
+  // PrepareEmitBuffer analogue: filter (score > FLT_MIN,
+  // doc_collector.hpp:58), exact select under (score desc, seg, doc)
+  size_t n = 0;
+  for (size_t i = 0; i < cands.size(); ++i)
+    if (cands[i].score > FLT_MIN) cands[n++] = cands[i];
+  cands.resize(n);
+  auto cmp = [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
+    if (x.score != y.score) return x.score > y.score;
+    if (x.segment_idx != y.segment_idx) return x.segment_idx < y.segment_idx;
+    return x.doc < y.doc;
+  };
+  const size_t kk = std::min<size_t>(k, cands.size());
+  std::partial_sort(cands.begin(), cands.begin() + kk, cands.end(), cmp);
+  std::copy(cands.begin(), cands.begin() + kk, hits);
+  *out_count = (uint32_t)kk;
+  *total_matches = *ctx->h_matches;
+  return SDB_OK;
+}
+
+int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
+                        uint32_t* docs, uint32_t* freqs) {
+  if (!ctx || !seg || !docs || !freqs || term_idx >= seg->hdr.nterms)
+    return SDB_ERR_INVALID;
+  const SdbTermEntry& te = seg->terms_host[term_idx];
+  const uint64_t nblocks = te.desc_end - te.desc_begin;
+  if (!nblocks) return SDB_OK;
+  uint32_t *d_docs, *d_freqs;
+  const uint64_t cap = nblocks * 128;
+  HIP_CHECK(hipMalloc(&d_docs, cap * 4));
+  HIP_CHECK(hipMalloc(&d_freqs, cap * 4));
+  hipLaunchKernelGGL(decode_term_kernel, dim3((uint32_t)nblocks), dim3(64), 0,
+                     ctx->stream, seg->desc, te.desc_begin, nblocks,
+                     seg->payload, te.payload_begin, d_docs, d_freqs);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  // compact (tail block writes d.len < 128 at its slot)
+  std::vector<uint32_t> tmp_d(cap), tmp_f(cap);
+  HIP_CHECK(hipMemcpy(tmp_d.data(), d_docs, cap * 4, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(tmp_f.data(), d_freqs, cap * 4, hipMemcpyDeviceToHost));
+  hipFree(d_docs);
+  hipFree(d_freqs);
+  uint32_t n = 0;
+  for (uint64_t b = 0; b < nblocks; ++b) {
+    // need lens: all full except possibly last
+    const uint32_t len =
+      (b + 1 == nblocks) ? (te.df - (uint32_t)(nblocks - 1) * 128u) : 128u;
+    std::memcpy(docs + n, tmp_d.data() + b * 128, len * 4);
+    std::memcpy(freqs + n, tmp_f.data() + b * 128, len * 4);
+    n += len;
+  }
+  return SDB_OK;
+}
+
+// window-kernel time (ms) of the last sdb_gpu_execute_topk on this context,
+// measured with HIP events on the library's own stream (roofline numerator
+// denominator for bench.py; a torch event on another stream cannot see it)
+int sdb_gpu_last_kernel_ms(SdbGpuCtx* ctx, double* ms) {
+  if (!ctx || !ms) return SDB_ERR_INVALID;
+  *ms = ctx->last_kernel_ms;
+  return SDB_OK;
+}
+
+}  // extern "C"

@@ -1,0 +1,29 @@
+// sdb_internal.h — shared internal context layout between sdb_gpu.hip and
+// sdb_scan.hip (not part of the public C ABI).
+#pragma once
+#include <hip/hip_runtime.h>
+#include "../../../include/sdb_gpu.h"
+
+struct TermDev {
+  uint64_t desc_begin;
+  uint64_t desc_end;
+  uint64_t payload_begin;
+  float num;  // boost*(k+1)*idf  (Bm25Score::num, bm25.cpp:225)
+  float nc;   // norm_const = k(1-b)
+  float nl;   // norm_length = k*b/avgDL
+};
+
+struct SdbGpuCtx {
+  int device;
+  hipStream_t stream;
+  SdbScoreDoc* d_cands;
+  uint32_t* d_cand_count;
+  unsigned long long* d_total_matches;
+  uint32_t* d_gthresh;  // float bits
+  TermDev* d_terms;
+  uint32_t* d_overflow;
+  uint32_t* h_counts;  // pinned: [cand_count, overflow]
+  unsigned long long* h_matches;
+  hipEvent_t ev_a, ev_b;   // bracket the window kernels of one execute call
+  double last_kernel_ms;   // read back via sdb_gpu_last_kernel_ms
+};

@@ -1,0 +1,239 @@
+// sdb_scan.hip — columnar scan -> predicate -> hash-group-by on MI355X.
+//
+// Replaces the reference's FullScanner::Scan + ColFilterChain predicate
+// narrowing (server/connector/full_scanner.h:40-90,
+// index/table_filter_iterator.hpp:104-227) fused with the consumer the
+// reference delegates to external DuckDB (PhysicalHashAggregate, un-vendored;
+// result-level parity per SURVEY.md §8c).
+//
+// MI355X design: this is pure HBM-bandwidth work (no MFMA). Columns are
+// device-resident dense arrays (FoR/bitpack codecs: planned next row,
+// SURVEY.md §8f). A grid-stride kernel reads rows with coalesced wide loads;
+// each workgroup accumulates into LDS per-group slots (group keys are dense
+// [0, ngroups), the "LDS-staged open-addressed buckets" of north_star with a
+// perfect hash), then flushes once per workgroup with device atomics.
+// COUNT/SUM(i64) are exact (wrap-around two's complement); SUM over an f32
+// column accumulates in f64 (atomic order nondeterministic; parity vs the
+// oracle's sequential f64 sum is within ~1e-12 relative at 1e9 rows,
+// asserted at 1e-7 in tests — mirrors the reference's own thread-order-
+// dependent fp aggregation).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstring>
+#include <vector>
+
+#include "../../../include/sdb_gpu.h"
+#include "sdb_internal.h"
+
+#define SCAN_NTHREADS 256u
+#define SCAN_MAX_GROUPS 2048u
+#define SCAN_MAX_AGGS 8u
+#define SCAN_MAX_PREDS 4u
+
+#define HIP_CHECK(x)                                   \
+  do {                                                 \
+    hipError_t _e = (x);                               \
+    if (_e == hipErrorNoDevice) return SDB_ERR_NO_GPU; \
+    if (_e != hipSuccess) return SDB_ERR_HIP;          \
+  } while (0)
+
+struct SdbGpuTable {
+  void* cols[16];
+  SdbColType types[16];
+  uint32_t ncols;
+  uint64_t rows;
+};
+
+struct ScanArgs {
+  const int64_t* keys;
+  uint64_t rows;
+  uint32_t ngroups;
+  uint32_t naggs;
+  uint32_t npreds;
+  // preds (on i64 columns; f32 preds can be added when a config needs them)
+  const int64_t* pred_col[SCAN_MAX_PREDS];
+  int pred_op[SCAN_MAX_PREDS];
+  int64_t pred_lo[SCAN_MAX_PREDS];
+  int64_t pred_hi[SCAN_MAX_PREDS];
+  // aggs
+  const void* agg_col[SCAN_MAX_AGGS];
+  int agg_op[SCAN_MAX_AGGS];
+  // outputs: [group * naggs + agg] as u64 (COUNT/SUM_I64) or f64 (SUM_F64)
+  unsigned long long* out;
+  unsigned long long* rows_passed;
+};
+
+__launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned long long* acc = (unsigned long long*)smem;  // ngroups*naggs
+  const uint32_t nslots = a.ngroups * a.naggs;
+  for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) acc[i] = 0;
+  __syncthreads();
+
+  const uint64_t stride = (uint64_t)gridDim.x * SCAN_NTHREADS;
+  uint64_t my_passed = 0;
+  for (uint64_t r = (uint64_t)blockIdx.x * SCAN_NTHREADS + threadIdx.x;
+       r < a.rows; r += stride) {
+    bool ok = true;
+    for (uint32_t p = 0; p < a.npreds; ++p) {
+      const int64_t x = a.pred_col[p][r];
+      switch (a.pred_op[p]) {
+        case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
+        case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
+        case SDB_PRED_BETWEEN:
+          ok &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+          break;
+        default: break;
+      }
+    }
+    if (!ok) continue;
+    ++my_passed;
+    const uint32_t g = (uint32_t)a.keys[r];
+    for (uint32_t q = 0; q < a.naggs; ++q) {
+      unsigned long long* slot = &acc[g * a.naggs + q];
+      switch (a.agg_op[q]) {
+        case SDB_AGG_COUNT:
+          atomicAdd(slot, 1ull);
+          break;
+        case SDB_AGG_SUM_I64:
+          atomicAdd(slot, (unsigned long long)((const int64_t*)a.agg_col[q])[r]);
+          break;
+        case SDB_AGG_SUM_F64:
+          atomicAdd((double*)slot, (double)((const float*)a.agg_col[q])[r]);
+          break;
+      }
+    }
+  }
+  // rows_passed: wave-reduce then one atomic per wave
+  unsigned long long wp = my_passed;
+#pragma unroll
+  for (int off = 32; off; off >>= 1) wp += __shfl_down(wp, off, 64);
+  if ((threadIdx.x & 63) == 0 && wp) atomicAdd(a.rows_passed, wp);
+  __syncthreads();
+  // flush LDS accumulators
+  for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) {
+    const uint32_t q = i % a.naggs;
+    if (a.agg_op[q] == SDB_AGG_SUM_F64) {
+      double v;
+      __builtin_memcpy(&v, &acc[i], 8);
+      if (v != 0.0) atomicAdd((double*)&a.out[i], v);
+    } else if (acc[i]) {
+      atomicAdd(&a.out[i], acc[i]);
+    }
+  }
+}
+
+extern "C" {
+
+int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
+                       uint32_t ncols, uint64_t rows, SdbGpuTable** out) {
+  if (!ctx || !cols || !out || ncols == 0 || ncols > 16)
+    return SDB_ERR_INVALID;
+  auto* tab = new SdbGpuTable{};
+  tab->ncols = ncols;
+  tab->rows = rows;
+  for (uint32_t c = 0; c < ncols; ++c) {
+    const size_t esz = cols[c].type == SDB_COL_I64 ? 8 : 4;
+    tab->types[c] = cols[c].type;
+    HIP_CHECK(hipMalloc(&tab->cols[c], esz * rows));
+    HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, esz * rows,
+                        hipMemcpyHostToDevice));
+  }
+  *out = tab;
+  return SDB_OK;
+}
+
+int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab) {
+  if (!ctx || !tab) return SDB_ERR_INVALID;
+  for (uint32_t c = 0; c < tab->ncols; ++c) hipFree(tab->cols[c]);
+  delete tab;
+  return SDB_OK;
+}
+
+int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
+                     uint32_t ngroups, const SdbPredSpec* preds,
+                     uint32_t npreds, const SdbAggSpec* aggs, uint32_t naggs,
+                     SdbAggResult* out, uint64_t* rows_passed) {
+  if (!ctx || !tab || !out || !rows_passed || group_col >= tab->ncols ||
+      ngroups == 0 || ngroups > SCAN_MAX_GROUPS || naggs == 0 ||
+      naggs > SCAN_MAX_AGGS || npreds > SCAN_MAX_PREDS)
+    return SDB_ERR_INVALID;
+  if (tab->types[group_col] != SDB_COL_I64) return SDB_ERR_INVALID;
+
+  hipStream_t stream = ctx->stream;
+
+  ScanArgs a{};
+  a.keys = (const int64_t*)tab->cols[group_col];
+  a.rows = tab->rows;
+  a.ngroups = ngroups;
+  a.naggs = naggs;
+  a.npreds = npreds;
+  for (uint32_t p = 0; p < npreds; ++p) {
+    if (preds[p].col >= tab->ncols ||
+        tab->types[preds[p].col] != SDB_COL_I64)
+      return SDB_ERR_INVALID;
+    a.pred_col[p] = (const int64_t*)tab->cols[preds[p].col];
+    a.pred_op[p] = preds[p].op;
+    a.pred_lo[p] = preds[p].ilo;
+    a.pred_hi[p] = preds[p].ihi;
+  }
+  for (uint32_t q = 0; q < naggs; ++q) {
+    a.agg_op[q] = aggs[q].op;
+    a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT
+                     ? nullptr
+                     : tab->cols[aggs[q].col];
+    if (aggs[q].op == SDB_AGG_SUM_I64 &&
+        tab->types[aggs[q].col] != SDB_COL_I64)
+      return SDB_ERR_INVALID;
+    if (aggs[q].op == SDB_AGG_SUM_F64 &&
+        tab->types[aggs[q].col] != SDB_COL_F32)
+      return SDB_ERR_INVALID;
+  }
+  const uint32_t nslots = ngroups * naggs;
+  unsigned long long* d_out;
+  unsigned long long* d_passed;
+  HIP_CHECK(hipMalloc(&d_out, 8ull * nslots));
+  HIP_CHECK(hipMalloc(&d_passed, 8));
+  HIP_CHECK(hipMemsetAsync(d_out, 0, 8ull * nslots, stream));
+  HIP_CHECK(hipMemsetAsync(d_passed, 0, 8, stream));
+  a.out = d_out;
+  a.rows_passed = d_passed;
+
+  // memory-bound grid sizing (guide §6 G11): cap ~8 blocks/CU, grid-stride
+  uint32_t nblocks = (uint32_t)((tab->rows + SCAN_NTHREADS - 1) / SCAN_NTHREADS);
+  if (nblocks > 2048) nblocks = 2048;
+  const size_t lds = 8ull * nslots;
+  hipLaunchKernelGGL(scan_agg_kernel, dim3(nblocks), dim3(SCAN_NTHREADS), lds,
+                     stream, a);
+  HIP_CHECK(hipGetLastError());
+  std::vector<unsigned long long> h_out(nslots);
+  unsigned long long h_passed = 0;
+  HIP_CHECK(hipMemcpyAsync(h_out.data(), d_out, 8ull * nslots,
+                           hipMemcpyDeviceToHost, stream));
+  HIP_CHECK(hipMemcpyAsync(&h_passed, d_passed, 8, hipMemcpyDeviceToHost,
+                           stream));
+  HIP_CHECK(hipStreamSynchronize(stream));
+  hipFree(d_out);
+  hipFree(d_passed);
+  for (uint32_t g = 0; g < ngroups; ++g) {
+    for (uint32_t q = 0; q < naggs; ++q) {
+      const unsigned long long raw = h_out[g * naggs + q];
+      SdbAggResult* r = &out[g * naggs + q];
+      if (aggs[q].op == SDB_AGG_SUM_F64) {
+        double v;
+        std::memcpy(&v, &raw, 8);
+        r->f64 = v;
+        r->i64 = 0;
+      } else {
+        r->i64 = (int64_t)raw;
+        r->f64 = 0;
+      }
+    }
+  }
+  *rows_passed = h_passed;
+  return SDB_OK;
+}
+
+}  // extern "C"

@@ -650,7 +650,9 @@ int sdb_host_build_segment(uint32_t doc_count, uint32_t nterms,
   off = align64(off + sizeof(uint32_t) * (uint64_t(doc_count) + 1));
   hdr.off_payload = off;
   hdr.payload_size = payload.size();
-  hdr.blob_size = align64(off + payload.size());
+  /* +64 tail pad: device decode reads payload words via aligned-pair loads
+   * that may touch up to 7 bytes past the last used byte */
+  hdr.blob_size = align64(off + payload.size() + 64);
 
   uint8_t* blob = static_cast<uint8_t*>(std::calloc(1, hdr.blob_size));
   if (!blob) return -4;

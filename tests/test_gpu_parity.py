@@ -1,0 +1,205 @@
+"""GPU parity tests (run on a real MI355X via gpurun / the round-end driver).
+
+Every test compares the product GPU path (libsdb_gpu, HIP kernels) against
+the CPU oracle on identical inputs. Bar (BASELINE.json north_star): doc ids,
+hit sets and integer counts bit-exact; BM25 scores bit-exact in practice
+(term-major fp32 order both sides; tolerance budget 1e-5 relative).
+"""
+
+import numpy as np
+import pytest
+
+import serenedb_amd as sa
+from oracle import pyoracle as po
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    c = sa.GpuContext(0)
+    yield c
+    c.close()
+
+
+def make_corpus(seed, doc_count, sels):
+    postings = [sa.synth_postings(seed, doc_count, t, s)
+                for t, s in enumerate(sels)]
+    norms = sa.synth_norms(seed, doc_count)
+    blob = sa.build_segment(doc_count, postings, norms)
+    return blob, postings, norms
+
+
+def check_parity(ctx, blob, term_idx, boosts, k, min_match=1,
+                 global_stats=None, segs=None):
+    if segs is None:
+        segs = [ctx.load_segment(blob)]
+        blobs = [blob]
+    else:
+        blobs = blob
+    hits, total = ctx.execute_topk(segs, term_idx, boosts, k,
+                                   min_match=min_match,
+                                   global_stats=global_stats)
+    ohits, ototal = po.execute_topk(blobs, term_idx, boosts, k,
+                                    min_match=min_match,
+                                    global_stats=global_stats)
+    assert total == ototal, f"matches {total} != {ototal}"
+    assert len(hits) == len(ohits)
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+    np.testing.assert_array_equal(hits["segment"], ohits["segment"])
+    np.testing.assert_array_equal(
+        hits["score"].view(np.uint32), ohits["score"].view(np.uint32),
+        err_msg="fp32 scores must be bit-exact (term-major order)")
+    return hits, total
+
+
+def test_decode_term_parity(ctx):
+    blob, postings, _ = make_corpus(42, 300_000, [0.05, 0.02, 0.5, 0.0007])
+    seg = ctx.load_segment(blob)
+    for t, (docs, freqs) in enumerate(postings):
+        if len(docs) == 0:
+            continue
+        ddocs, dfreqs = ctx.decode_term(seg, t, len(docs))
+        np.testing.assert_array_equal(ddocs, docs)
+        np.testing.assert_array_equal(dfreqs, freqs)
+
+
+def test_topk_or_parity(ctx):
+    blob, _, _ = make_corpus(43, 1_000_000, [0.10, 0.05, 0.02, 0.01])
+    for k in (10, 1000):
+        check_parity(ctx, blob, [0, 1, 2, 3], [1.0] * 4, k)
+
+
+def test_topk_and_minmatch_parity(ctx):
+    blob, _, _ = make_corpus(44, 500_000, [0.2, 0.1, 0.05])
+    check_parity(ctx, blob, [0, 1, 2], [1.0] * 3, 100, min_match=3)  # AND
+    check_parity(ctx, blob, [0, 1, 2], [1.0] * 3, 100, min_match=2)
+
+
+def test_topk_boosts_and_k1b(ctx):
+    blob, _, _ = make_corpus(45, 400_000, [0.05, 0.02])
+    seg = ctx.load_segment(blob)
+    hits, total = ctx.execute_topk([seg], [0, 1], [2.0, 0.5], 50,
+                                   k1=0.9, b=0.4)
+    ohits, ototal = po.execute_topk([blob], [0, 1], [2.0, 0.5], 50,
+                                    k1=0.9, b=0.4)
+    assert total == ototal
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+    np.testing.assert_array_equal(hits["score"], ohits["score"])
+
+
+def test_topk_edge_cases(ctx):
+    # single posting, empty term, k > matches, dense term (bitset blocks),
+    # tails of every family
+    doc_count = 10_000
+    rng = np.random.default_rng(5)
+    dense = np.sort(rng.choice(np.arange(1, 2000, dtype=np.uint32), 1500,
+                               replace=False))
+    postings = [
+        (np.array([doc_count], dtype=np.uint32),
+         np.array([7], dtype=np.uint32)),            # last-doc posting
+        (np.array([], dtype=np.uint32), np.array([], dtype=np.uint32)),
+        (dense, rng.integers(1, 4, len(dense)).astype(np.uint32)),
+        (np.arange(1, 40, dtype=np.uint32) * 250,
+         np.full(39, 2, dtype=np.uint32)),           # all-same tails
+    ]
+    norms = sa.synth_norms(3, doc_count)
+    blob = sa.build_segment(doc_count, postings, norms)
+    check_parity(ctx, blob, [0, 1, 2, 3], [1.0] * 4, 100)
+    check_parity(ctx, blob, [1], [1.0], 10)           # empty result
+    check_parity(ctx, blob, [0, 2], [1.0, 1.0], 5000)  # k > matches
+
+
+def test_multi_segment(ctx):
+    b1, _, _ = make_corpus(46, 120_000, [0.05, 0.02])
+    b2, _, _ = make_corpus(47, 80_000, [0.03, 0.04])
+    s1 = ctx.load_segment(b1)
+    s2 = ctx.load_segment(b2)
+    check_parity(ctx, [b1, b2], [0, 1], [1.0, 1.0], 200, segs=[s1, s2])
+
+
+def test_sharded_global_stats(ctx):
+    """one GPU, shard segments with injected global stats == full corpus"""
+    seed, doc_count = 48, 200_000
+    sels = [0.05, 0.02, 0.01]
+    full, postings, norms = make_corpus(seed, doc_count, sels)
+    half = doc_count // 2
+    shards = [sa.build_synth_segment(seed, 1, half, sels),
+              sa.build_synth_segment(seed, half + 1, doc_count, sels)]
+    dwt = [len(d) for d, _ in postings]
+    gstats = (doc_count, int(norms[1:].sum()), dwt)
+    fhits, ftotal = po.execute_topk([full], [0, 1, 2], [1.0] * 3, 150)
+    cands = []
+    totals = 0
+    for base, blob in zip((0, half), shards):
+        seg = ctx.load_segment(blob)
+        h, t = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 150,
+                                global_stats=gstats)
+        totals += t
+        for x in h:
+            cands.append((float(x["score"]), int(x["doc"]) + base))
+    cands.sort(key=lambda sd: (-sd[0], sd[1]))
+    assert totals == ftotal
+    assert [d for _, d in cands[:150]] == [int(d) for d in fhits["doc"]]
+
+
+def test_scan_agg_parity(ctx):
+    import ctypes as CT
+
+    rows = 2_000_000
+    ngroups = 1024
+    rng = np.random.default_rng(44)
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(0, 1 << 20, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+
+    lib = sa.gpu()
+
+    class ColView(CT.Structure):
+        _fields_ = [("data", CT.c_void_p), ("rows", CT.c_uint64),
+                    ("type", CT.c_int)]
+
+    class PredSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int),
+                    ("ilo", CT.c_int64), ("ihi", CT.c_int64),
+                    ("flo", CT.c_float), ("fhi", CT.c_float)]
+
+    class AggSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int)]
+
+    class AggResult(CT.Structure):
+        _fields_ = [("i64", CT.c_int64), ("f64", CT.c_double)]
+
+    cols = (ColView * 3)(
+        ColView(keys.ctypes.data_as(CT.c_void_p).value, rows, 0),
+        ColView(v1.ctypes.data_as(CT.c_void_p).value, rows, 0),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, rows, 1))
+    tab = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols, 3, CT.c_uint64(rows),
+                                CT.byref(tab))
+    assert rc == 0, rc
+    # predicate: v1 < c at ~10% selectivity
+    c = int((1 << 20) * 0.1)
+    preds = (PredSpec * 1)(PredSpec(1, 1, c, 0, 0, 0))
+    aggs = (AggSpec * 3)(AggSpec(0, 0), AggSpec(1, 1), AggSpec(2, 2))
+    out = (AggResult * (ngroups * 3))()
+    passed = CT.c_uint64(0)
+    rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab, 0, ngroups, preds, 1, aggs, 3,
+                              out, CT.byref(passed))
+    assert rc == 0, rc
+    ocnt, osi, osf, opassed = po.scan_agg(keys, v1, v2, ngroups, pred_op=1,
+                                          lo=c)
+    assert passed.value == opassed
+    gcnt = np.array([out[g * 3 + 0].i64 for g in range(ngroups)])
+    gsi = np.array([out[g * 3 + 1].i64 for g in range(ngroups)])
+    gsf = np.array([out[g * 3 + 2].f64 for g in range(ngroups)])
+    np.testing.assert_array_equal(gcnt, ocnt)
+    np.testing.assert_array_equal(gsi, osi)
+    np.testing.assert_allclose(gsf, osf, rtol=1e-7)
+    lib.sdb_gpu_table_free(ctx._ctx, tab)
+
+
+def test_smoke_entry():
+    import __graft_entry__ as ge
+
+    ge.smoke()
