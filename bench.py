@@ -1,0 +1,453 @@
+#!/usr/bin/env python3
+"""bench.py — headline benchmark of the MI355X-native SereneDB hot path.
+
+Default workload = BASELINE.json configs[1]: BM25 top-1000, 4-term
+disjunction, 100M synthetic docs (seed 43, SURVEY.md §8d distributions),
+1 GPU. A "step" is one full query execution (decode -> score -> top-k ->
+merge). `--workload scan_agg` runs configs[2] (1B-row scan->filter->group-by,
+seed 44).
+
+  python bench.py --gpus N --steps K --warmup W [--workload bm25_topk]
+
+For N>1 the driver launches this under torch.distributed.run (one rank per
+GPU over RCCL); ranks shard the SAME corpus by doc range (strong scaling,
+BASELINE configs[4]) and merge per-rank top-k candidates with an allgather
+(24 KB — SURVEY.md §8e) plus an allreduce of total-match counts.
+
+Outputs ONE JSON line from rank 0 (driver contract).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_GBS = 8000.0  # 8 TB/s spec (MI355X_MICROARCH.md)
+
+
+def env_rank():
+    return (int(os.environ.get("RANK", "0")),
+            int(os.environ.get("WORLD_SIZE", "1")))
+
+
+def pack_hits(hits, base, k):
+    """(score,doc) -> int64 tensor [k]: score_bits<<32 | global_doc.
+    Scores >= 0 so bit order == float order; padded with -1 (never selected
+    because score_bits>=0 means packed >= 0)."""
+    out = np.full(k, -1, dtype=np.int64)
+    n = len(hits)
+    sb = hits["score"].view(np.uint32).astype(np.int64)
+    gd = hits["doc"].astype(np.int64) + base
+    out[:n] = (sb << 32) | gd
+    return out
+
+
+def unpack_hits(packed):
+    packed = packed[packed >= 0]
+    sb = (packed >> 32).astype(np.uint32)
+    docs = (packed & 0xFFFFFFFF).astype(np.uint64)
+    scores = sb.view(np.float32)
+    return scores, docs
+
+
+def bench_bm25(args):
+    import serenedb_amd as sa
+
+    rank, world = env_rank()
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+
+        tdist.init_process_group(backend="nccl")
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        dist = tdist
+
+    seed = 43
+    doc_count = args.docs
+    sels = [0.10, 0.05, 0.02, 0.01]
+    k = 1000
+    nterms = len(sels)
+
+    # ---- shard build (untimed; index-build side is CPU by design) ----
+    per = doc_count // world
+    lo = rank * per + 1
+    hi = doc_count if rank == world - 1 else (rank + 1) * per
+    t0 = time.time()
+    blob = sa.build_synth_segment(seed, lo, hi, sels)
+    build_s = time.time() - t0
+
+    device = int(os.environ.get("LOCAL_RANK", 0))
+    ctx = sa.GpuContext(device)
+    seg = ctx.load_segment(blob)
+
+    # ---- global stats (PreparePhase analogue over RCCL) ----
+    import ctypes as CT
+    v = np.frombuffer(blob, dtype=np.uint8)
+    host = sa.host()
+
+    class _View(CT.Structure):
+        _fields_ = [("hdr", CT.c_void_p), ("terms", CT.c_void_p),
+                    ("desc", CT.c_void_p), ("norms", CT.c_void_p),
+                    ("payload", CT.c_void_p)]
+
+    class _Hdr(CT.Structure):
+        _fields_ = [("magic", CT.c_uint64), ("version", CT.c_uint32),
+                    ("nterms", CT.c_uint32), ("doc_count", CT.c_uint32),
+                    ("docs_with_field", CT.c_uint32),
+                    ("total_term_freq", CT.c_uint64),
+                    ("total_blocks", CT.c_uint64),
+                    ("off_terms", CT.c_uint64), ("off_desc", CT.c_uint64),
+                    ("off_norms", CT.c_uint64), ("off_payload", CT.c_uint64),
+                    ("payload_size", CT.c_uint64),
+                    ("blob_size", CT.c_uint64)]
+
+    class _Term(CT.Structure):
+        _fields_ = [("desc_begin", CT.c_uint64), ("desc_end", CT.c_uint64),
+                    ("payload_begin", CT.c_uint64),
+                    ("payload_end", CT.c_uint64), ("df", CT.c_uint32),
+                    ("max_freq", CT.c_uint32), ("total_freq", CT.c_uint64)]
+
+    vw = _View()
+    rc = host.sdb_host_segment_parse(
+        v.ctypes.data_as(CT.c_void_p), CT.c_uint64(len(v)), CT.byref(vw))
+    assert rc == 0
+    h = CT.cast(vw.hdr, CT.POINTER(_Hdr)).contents
+    terms = CT.cast(vw.terms, CT.POINTER(_Term * h.nterms)).contents
+    local_stats = np.array(
+        [h.docs_with_field, h.total_term_freq] +
+        [terms[t].df for t in range(nterms)], dtype=np.int64)
+    local_payload_bytes = sum(
+        terms[t].payload_end - terms[t].payload_begin for t in range(nterms))
+    local_desc_blocks = sum(
+        terms[t].desc_end - terms[t].desc_begin for t in range(nterms))
+    if dist:
+        import torch
+        ts = torch.tensor(local_stats, device="cuda")
+        dist.all_reduce(ts)  # PreparePhase stats merge over RCCL
+        gstats_arr = ts.cpu().numpy()
+    else:
+        gstats_arr = local_stats
+    g_dwf, g_ttf = int(gstats_arr[0]), int(gstats_arr[1])
+    g_dwt = [int(x) for x in gstats_arr[2:2 + nterms]]
+    gstats = (g_dwf, g_ttf, g_dwt)
+    total_postings_global = sum(g_dwt)
+    term_idx = list(range(nterms))
+    boosts = [1.0] * nterms
+
+    lib = sa.gpu()
+    lib.sdb_gpu_last_kernel_ms.restype = CT.c_int
+
+    def step():
+        hits, total = ctx.execute_topk([seg], term_idx, boosts, k,
+                                       global_stats=gstats)
+        if dist:
+            import torch
+            packed = torch.from_numpy(pack_hits(hits, lo - 1, k)).cuda()
+            gathered = [torch.empty_like(packed) for _ in range(world)]
+            dist.all_gather(gathered, packed)
+            tm = torch.tensor([total], dtype=torch.int64, device="cuda")
+            dist.all_reduce(tm)
+            total = int(tm.item())
+            allp = torch.cat(gathered).cpu().numpy()
+            allp.sort()
+            topk = allp[::-1][:k]  # score desc; doc tiebreak implicit in bits
+            scores, docs = unpack_hits(topk)
+            return scores, docs, total
+        return hits["score"], hits["doc"], total
+
+    def sync():
+        if dist:
+            import torch
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        step()
+    sync()
+
+    # ---- timed ----
+    kernel_ms_acc = 0.0
+    sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        scores, docs, total = step()
+        ms = CT.c_double(0)
+        lib.sdb_gpu_last_kernel_ms(ctx._ctx, CT.byref(ms))
+        kernel_ms_acc += ms.value
+    sync()
+    elapsed = time.time() - t0
+    if dist:
+        import torch
+        te = torch.tensor([elapsed], device="cuda")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    value = total_postings_global * args.steps / elapsed  # postings scored/s
+
+    # ---- roofline (rank 0's shard kernel): algorithmic bytes per launch ----
+    # postings payload (compressed docs+freqs incl. tag bytes) + 4 B norm per
+    # posting + 32 B descriptor per touched block
+    local_postings = sum(terms[t].df for t in range(nterms))
+    algo_bytes = (local_payload_bytes + 4 * local_postings +
+                  32 * local_desc_blocks)
+    kernel_s = kernel_ms_acc / 1000.0
+    achieved_gbs = (algo_bytes * args.steps / kernel_s / 1e9) if kernel_s else 0
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved_gbs, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+        "traffic": None,
+        "note": "achieved = algorithmic bytes (postings payload + 4B norm/posting + 32B/block desc) / window-kernel time (HIP events); see profiles/ for rocprofv3 evidence",
+    }
+
+    # ---- CPU baseline (rank 0, N=1 only): the oracle's multithreaded
+    # mechanics path (RunTopKScan restatement) on a bounded sample ----
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        from oracle import pyoracle as po
+
+        sample_docs = min(doc_count, 12_500_000)
+        sblob = (blob if (lo == 1 and hi == sample_docs) else
+                 sa.build_synth_segment(seed, 1, sample_docs, sels))
+        # postings in the sample
+        sv = _View()
+        sb = np.frombuffer(sblob, dtype=np.uint8)
+        host.sdb_host_segment_parse(sb.ctypes.data_as(CT.c_void_p),
+                                    CT.c_uint64(len(sb)), CT.byref(sv))
+        sh = CT.cast(sv.hdr, CT.POINTER(_Hdr)).contents
+        st = CT.cast(sv.terms, CT.POINTER(_Term * sh.nterms)).contents
+        sample_postings = sum(st[t].df for t in range(nterms))
+        ncores = os.cpu_count() or 1
+        iters = 0
+        tcpu = time.time()
+        while time.time() - tcpu < args.cpu_seconds and iters < 200:
+            po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
+                               global_stats=gstats)
+            iters += 1
+        tcpu = time.time() - tcpu
+        cpu_baseline = {
+            "value": round(sample_postings * iters / tcpu, 1),
+            "unit": "postings scored/s",
+            "cores": ncores,
+            "kind": "port",
+            "sample": f"{sample_docs/1e6:.1f}M-doc shard of the same corpus "
+                      f"({sample_postings} postings/query, {iters} iters, "
+                      f"{tcpu:.1f}s; oracle RunTopKScan restatement)",
+        }
+
+    result = {
+        "metric": "docs scored/sec BM25 top-1000",
+        "value": round(value, 1),
+        "unit": "postings/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "f32",
+        "data": "synthetic",
+        "config": {
+            "workload": "bm25_top1000_4term_or_100M" if doc_count == 100_000_000
+                        else f"bm25_top1000_4term_or_{doc_count}",
+            "doc_count": doc_count,
+            "selectivities": sels,
+            "k": k,
+            "seed": seed,
+            "postings_per_query": int(total_postings_global),
+            "total_matches": int(total),
+            "parallelism": f"doc-range shards x{world}, RCCL allgather merge",
+            "segment_build_s": round(build_s, 1),
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+    if dist:
+        dist.destroy_process_group()
+
+
+def bench_scan(args):
+    import ctypes as CT
+
+    import serenedb_amd as sa
+
+    rank, world = env_rank()
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+
+        tdist.init_process_group(backend="nccl")
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        dist = tdist
+
+    rows_total = args.rows
+    ngroups = 1024
+    seed = 44
+    per = rows_total // world
+    my_rows = per if rank != world - 1 else rows_total - per * (world - 1)
+    rng = np.random.default_rng(seed + rank)
+    keys = rng.integers(0, ngroups, my_rows).astype(np.int64)
+    v1 = rng.integers(0, 1 << 20, my_rows).astype(np.int64)
+    v2 = rng.normal(0, 1, my_rows).astype(np.float32)
+
+    device = int(os.environ.get("LOCAL_RANK", 0))
+    ctx = sa.GpuContext(device)
+    lib = sa.gpu()
+
+    class ColView(CT.Structure):
+        _fields_ = [("data", CT.c_void_p), ("rows", CT.c_uint64),
+                    ("type", CT.c_int)]
+
+    class PredSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int),
+                    ("ilo", CT.c_int64), ("ihi", CT.c_int64),
+                    ("flo", CT.c_float), ("fhi", CT.c_float)]
+
+    class AggSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int)]
+
+    class AggResult(CT.Structure):
+        _fields_ = [("i64", CT.c_int64), ("f64", CT.c_double)]
+
+    cols = (ColView * 3)(
+        ColView(keys.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
+        ColView(v1.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, my_rows, 1))
+    tab = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols, 3, CT.c_uint64(my_rows),
+                                CT.byref(tab))
+    assert rc == 0, rc
+    c = int((1 << 20) * 0.1)  # 10% selectivity predicate
+    preds = (PredSpec * 1)(PredSpec(1, 1, c, 0, 0, 0))
+    aggs = (AggSpec * 3)(AggSpec(0, 0), AggSpec(1, 1), AggSpec(2, 2))
+    out = (AggResult * (ngroups * 3))()
+    passed = CT.c_uint64(0)
+
+    def step():
+        rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab, 0, ngroups, preds, 1, aggs,
+                                  3, out, CT.byref(passed))
+        assert rc == 0, rc
+        if dist:
+            import torch
+            arr = np.array([out[i].i64 for i in range(ngroups * 3)],
+                           dtype=np.int64)
+            t = torch.from_numpy(arr).cuda()
+            dist.all_reduce(t)  # 24 KB partial-agg merge
+            _ = t.cpu().numpy()
+
+    def sync():
+        if dist:
+            import torch
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    sync()
+    elapsed = time.time() - t0
+    if dist:
+        import torch
+        te = torch.tensor([elapsed], device="cuda")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+    ms_per_step = elapsed * 1000 / args.steps
+    value = rows_total * args.steps / elapsed
+
+    algo_bytes = my_rows * 20  # 8B key + 8B v1 + 4B v2, no zonemap skips
+    achieved_gbs = algo_bytes * args.steps / elapsed / 1e9  # whole step ~ kernel
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        from oracle import pyoracle as po
+
+        n = min(my_rows, 50_000_000)
+        iters = 0
+        t1 = time.time()
+        while time.time() - t1 < args.cpu_seconds and iters < 50:
+            po.scan_agg(keys[:n], v1[:n], v2[:n], ngroups, pred_op=1, lo=c)
+            iters += 1
+        t1 = time.time() - t1
+        cpu_baseline = {
+            "value": round(n * iters / t1, 1),
+            "unit": "rows/s",
+            "cores": 1,
+            "kind": "port",
+            "sample": f"{n/1e6:.0f}M-row prefix, {iters} iters, {t1:.1f}s "
+                      "(oracle scan_agg, single-thread)",
+        }
+    result = {
+        "metric": "rows/sec filter-agg",
+        "value": round(value, 1),
+        "unit": "rows/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": f"scan_filter_groupby_{rows_total//10**9}B" if rows_total >= 10**9 else f"scan_filter_groupby_{rows_total}",
+            "rows": rows_total,
+            "ngroups": ngroups,
+            "predicate": "v1 < 10% quantile",
+            "aggs": "COUNT, SUM(i64), SUM(f32->f64)",
+            "seed": seed,
+            "parallelism": f"row shards x{world}, RCCL allreduce merge",
+        },
+        "roofline": {
+            "bound": "hbm",
+            "achieved": round(achieved_gbs, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+            "traffic": None,
+        },
+        "cpu_baseline": cpu_baseline,
+    }
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+    lib.sdb_gpu_table_free(ctx._ctx, tab)
+    if dist:
+        dist.destroy_process_group()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--workload", default="bm25_topk",
+                    choices=["bm25_topk", "scan_agg"])
+    ap.add_argument("--docs", type=int, default=100_000_000)
+    ap.add_argument("--rows", type=int, default=1_000_000_000)
+    ap.add_argument("--cpu-seconds", type=float, default=10.0)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+    if args.workload == "bm25_topk":
+        bench_bm25(args)
+    else:
+        bench_scan(args)
+
+
+if __name__ == "__main__":
+    main()
