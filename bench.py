@@ -36,21 +36,22 @@ def env_rank():
 
 
 def pack_hits(hits, base, k):
-    """(score,doc) -> int64 tensor [k]: score_bits<<32 | global_doc.
-    Scores >= 0 so bit order == float order; padded with -1 (never selected
-    because score_bits>=0 means packed >= 0)."""
+    """(score,doc) -> int64 [k]: score_bits<<32 | (2^32-1 - global_doc), so a
+    DESCENDING sort of packed values orders by (score desc, doc asc) — the
+    deterministic tie order of DESIGN.md. Scores >= 0 so float order ==
+    bit order; padding is -1 (negative, filtered out)."""
     out = np.full(k, -1, dtype=np.int64)
     n = len(hits)
     sb = hits["score"].view(np.uint32).astype(np.int64)
     gd = hits["doc"].astype(np.int64) + base
-    out[:n] = (sb << 32) | gd
+    out[:n] = (sb << 32) | (0xFFFFFFFF - gd)
     return out
 
 
 def unpack_hits(packed):
     packed = packed[packed >= 0]
     sb = (packed >> 32).astype(np.uint32)
-    docs = (packed & 0xFFFFFFFF).astype(np.uint64)
+    docs = (0xFFFFFFFF - (packed & 0xFFFFFFFF)).astype(np.uint64)
     scores = sb.view(np.float32)
     return scores, docs
 

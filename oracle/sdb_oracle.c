@@ -638,6 +638,32 @@ static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
   return 0;
 }
 
+/* Cross-segment stats merge — the reference merges Field/TermCollector
+ * counters over ALL segments before BM25::collect runs (PrepareCollector
+ * Finish, search/collectors.cpp; PreparePhase barrier,
+ * duckdb_search_full_scan.cpp:1359-1384): idf/avgDL are GLOBAL. When the
+ * caller passes no explicit global stats, derive them by summing the
+ * provided segments. Returns 0 and fills dwf/ttf/dwt[nterms]. */
+static int o_global_stats(const OSegBlob* segs, uint32_t nsegs,
+                          const uint32_t* term_idx, uint32_t nterms,
+                          uint64_t* dwf, uint64_t* ttf, uint64_t* dwt) {
+  *dwf = 0;
+  *ttf = 0;
+  memset(dwt, 0, sizeof(uint64_t) * nterms);
+  for (uint32_t s = 0; s < nsegs; ++s) {
+    SdbSegmentView v;
+    int rc = o_segment_parse(segs[s].blob, segs[s].size, &v);
+    if (rc) return rc;
+    *dwf += v.hdr->docs_with_field;
+    *ttf += v.hdr->total_term_freq;
+    for (uint32_t t = 0; t < nterms; ++t) {
+      if (term_idx[t] >= v.hdr->nterms) return -1;
+      dwt[t] += v.terms[term_idx[t]].df;
+    }
+  }
+  return 0;
+}
+
 /* EXACT top-k (the parity oracle): gathers every matching (doc,score), then
  * selects under the deterministic total order. Mirrors
  * ExecuteTopKWithCount's results up to nth_element tie refinement. */
@@ -649,18 +675,29 @@ int o_execute_topk(const OSegBlob* segs, uint32_t nsegs,
                    uint64_t* total_matches) {
   OCandVec cands = {0, 0, 0};
   uint64_t matches = 0;
+  uint64_t auto_dwf = 0, auto_ttf = 0;
+  uint64_t* auto_dwt = (uint64_t*)malloc(sizeof(uint64_t) * nterms);
+  if (g_dwf == 0) {
+    int rc = o_global_stats(segs, nsegs, term_idx, nterms, &auto_dwf,
+                            &auto_ttf, auto_dwt);
+    if (rc) { free(auto_dwt); return rc; }
+    g_dwf = auto_dwf;
+    g_ttf = auto_ttf;
+    g_dwt = auto_dwt;
+  }
   OCursor* cur = (OCursor*)malloc(sizeof(OCursor) * nterms);
   for (uint32_t s = 0; s < nsegs; ++s) {
     SdbSegmentView v;
     int rc = o_segment_parse(segs[s].blob, segs[s].size, &v);
-    if (rc) { free(cur); free(cands.v); return rc; }
+    if (rc) { free(cur); free(cands.v); free(auto_dwt); return rc; }
     rc = o_prep_cursors(&v, term_idx, boosts, nterms, k1, b, g_dwf, g_dwt,
                         g_ttf, cur);
-    if (rc) { free(cur); free(cands.v); return rc; }
+    if (rc) { free(cur); free(cands.v); free(auto_dwt); return rc; }
     matches += o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1,
                             1, v.hdr->doc_count, NULL, &cands, s);
   }
   free(cur);
+  free(auto_dwt);
   /* threshold semantics: accept score > FLT_MIN (doc_collector.hpp:58) */
   uint64_t n = 0;
   const float kFltMin = 1.17549435e-38f;
@@ -683,6 +720,16 @@ int o_execute_topk_mech(const OSegBlob* segs, uint32_t nsegs,
                         uint64_t g_dwf, const uint64_t* g_dwt, uint64_t g_ttf,
                         uint32_t k, OScoreDoc* hits2k /* 2k slots */,
                         uint32_t* out_count, uint64_t* total_matches) {
+  uint64_t auto_dwf = 0, auto_ttf = 0;
+  uint64_t* auto_dwt = (uint64_t*)malloc(sizeof(uint64_t) * nterms);
+  if (g_dwf == 0) {
+    int rc = o_global_stats(segs, nsegs, term_idx, nterms, &auto_dwf,
+                            &auto_ttf, auto_dwt);
+    if (rc) { free(auto_dwt); return rc; }
+    g_dwf = auto_dwf;
+    g_ttf = auto_ttf;
+    g_dwt = auto_dwt;
+  }
   float threshold = 1.17549435e-38f; /* FLT_MIN, doc_collector.hpp:58 */
   OCollector coll;
   coll.hits = hits2k;
@@ -694,15 +741,16 @@ int o_execute_topk_mech(const OSegBlob* segs, uint32_t nsegs,
   for (uint32_t s = 0; s < nsegs; ++s) {
     SdbSegmentView v;
     int rc = o_segment_parse(segs[s].blob, segs[s].size, &v);
-    if (rc) { free(cur); return rc; }
+    if (rc) { free(cur); free(auto_dwt); return rc; }
     rc = o_prep_cursors(&v, term_idx, boosts, nterms, k1, b, g_dwf, g_dwt,
                         g_ttf, cur);
-    if (rc) { free(cur); return rc; }
+    if (rc) { free(cur); free(auto_dwt); return rc; }
     coll.seg = s;
     o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1, 1,
                  v.hdr->doc_count, &coll, NULL, s);
   }
   free(cur);
+  free(auto_dwt);
   qsort(hits2k, coll.it, sizeof(OScoreDoc), o_sd_cmp); /* :82-85 final sort */
   *out_count = coll.it;
   *total_matches = coll.count;

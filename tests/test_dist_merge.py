@@ -1,0 +1,92 @@
+"""Multi-process (gloo, world_size=2) test of the distributed merge path on
+CPU: per-rank shard execution (oracle standing in for the per-GPU kernel) +
+torch.distributed stats-allreduce + top-k allgather + host merge must equal
+the single-segment result. This covers bench.py's N>1 logic (pack/unpack,
+stats merge, candidate merge) without a GPU.
+"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+DOCS = 60_000
+SELS = [0.05, 0.02, 0.01]
+K = 120
+SEED = 13
+
+
+def _worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    import serenedb_amd as sa
+    from oracle import pyoracle as po
+    import bench
+
+    per = DOCS // world
+    lo = rank * per + 1
+    hi = DOCS if rank == world - 1 else (rank + 1) * per
+    blob = sa.build_synth_segment(SEED, lo, hi, SELS)
+
+    # stats allreduce (PreparePhase analogue)
+    postings = [sa.synth_postings(SEED, DOCS, t, s)[0][
+        (sa.synth_postings(SEED, DOCS, t, s)[0] >= lo) &
+        (sa.synth_postings(SEED, DOCS, t, s)[0] <= hi)]
+        for t, s in enumerate(SELS)]
+    norms = sa.synth_norms(SEED, DOCS)
+    local = torch.tensor(
+        [hi - lo + 1, int(norms[lo:hi + 1].sum())] +
+        [len(p) for p in postings], dtype=torch.int64)
+    dist.all_reduce(local)
+    g_dwf, g_ttf = int(local[0]), int(local[1])
+    g_dwt = [int(x) for x in local[2:]]
+
+    hits, total = po.execute_topk([blob], list(range(len(SELS))),
+                                  [1.0] * len(SELS), K,
+                                  global_stats=(g_dwf, g_ttf, g_dwt))
+    packed = torch.from_numpy(bench.pack_hits(hits, lo - 1, K))
+    gathered = [torch.empty_like(packed) for _ in range(world)]
+    dist.all_gather(gathered, packed)
+    tm = torch.tensor([total], dtype=torch.int64)
+    dist.all_reduce(tm)
+
+    if rank == 0:
+        allp = torch.cat(gathered).numpy()
+        allp = np.sort(allp)[::-1][:K]
+        scores, docs = bench.unpack_hits(allp)
+        q.put((scores.copy(), docs.copy(), int(tm.item())))
+    dist.destroy_process_group()
+
+
+def test_gloo_sharded_merge_equals_full():
+    import serenedb_amd as sa
+    from oracle import pyoracle as po
+
+    full = sa.build_synth_segment(SEED, 1, DOCS, SELS)
+    fhits, ftotal = po.execute_topk([full], list(range(len(SELS))),
+                                    [1.0] * len(SELS), K)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        scores, docs, total = q.get(timeout=300)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    assert total == ftotal
+    np.testing.assert_array_equal(docs.astype(np.uint32), fhits["doc"])
+    np.testing.assert_array_equal(scores, fhits["score"])

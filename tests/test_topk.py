@@ -240,3 +240,21 @@ def test_boost_zero_scores_rejected():
     hits, total = po.execute_topk([blob], [0], [0.0], 10)
     assert total == 50
     assert len(hits) == 0
+
+
+def test_multi_segment_stats_merge():
+    """Cross-segment BM25 stats must be merged BEFORE scoring (PrepareCollector
+    Finish / PreparePhase, duckdb_search_full_scan.cpp:1359-1384): two
+    segments == one concatenated segment with the same global stats."""
+    seed, n1, n2 = 21, 20_000, 30_000
+    sels = [0.05, 0.02]
+    b1 = sa.build_synth_segment(seed, 1, n1, sels)
+    b2 = sa.build_synth_segment(seed, n1 + 1, n1 + n2, sels)
+    full = sa.build_synth_segment(seed, 1, n1 + n2, sels)
+    hits, total = po.execute_topk([b1, b2], [0, 1], [1.0, 1.0], 100)
+    fhits, ftotal = po.execute_topk([full], [0, 1], [1.0, 1.0], 100)
+    assert total == ftotal
+    # map (seg, local doc) -> global doc
+    got = [int(h["doc"]) + (n1 if h["segment"] == 1 else 0) for h in hits]
+    np.testing.assert_array_equal(got, fhits["doc"])
+    np.testing.assert_array_equal(hits["score"], fhits["score"])
