@@ -294,10 +294,9 @@ class GpuContext:
         if rc != 0:
             raise RuntimeError(f"sdb_gpu_execute_topk rc={rc}")
         n = out_count.value
-        res = np.zeros(n, dtype=[("score", "f4"), ("doc", "u4"),
-                                 ("segment", "u4")])
-        for i in range(n):
-            res[i] = (hits[i].score, hits[i].doc, hits[i].segment_idx)
+        dt = np.dtype([("score", "f4"), ("doc", "u4"), ("segment", "u4")])
+        res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
+                            dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value
 
     def decode_term(self, seg, term_idx, df):

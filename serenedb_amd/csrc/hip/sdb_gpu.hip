@@ -401,32 +401,42 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   if (lane == 0) shared_misc[2 * SDB_MAX_TERMS + 2 + wave] = wm;
   __syncthreads();  // hist + per-wave match counts complete
 
-  // local k-th lower bound from the histogram (thread 0)
-  if (tid == 0) {
-    uint32_t total_m = 0;
-    for (uint32_t w = 0; w < SDB_NWAVES; ++w)
-      total_m += shared_misc[2 * SDB_MAX_TERMS + 2 + w];
-    if (total_m) atomicAdd(a.total_matches, (unsigned long long)total_m);
-    uint32_t cum = 0;
-    uint32_t binfloor = 0;
-    for (int b = SDB_HIST_BINS - 1; b >= 0; --b) {
-      cum += hist[b];
-      if (cum >= a.k) {
-        binfloor = (uint32_t)b;
-        // tau_w = binfloor * smax/256: >= k window scores are >= tau_w,
-        // hence the GLOBAL k-th score >= tau_w -> valid global lower bound
-        const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
-        if (tau > 0.0f) {
-          uint32_t bits;
-          __builtin_memcpy(&bits, &tau, 4);
-          atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
+  // local k-th lower bound from the histogram: wave 0 does a suffix scan
+  // (lane l owns the 4 bins [252-4l .. 255-4l]); the first lane whose
+  // suffix count reaches k refines the exact bin and publishes tau_w
+  if (wave == 0) {
+    uint32_t part = 0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      part += hist[SDB_HIST_BINS - 4 * lane - 4 + j];
+    const uint32_t suff_incl = wave_incl_scan(part, lane);  // bins >= 252-4l
+    const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
+    const bool winner =
+      suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
+    if (winner) {
+      uint32_t cum = suff_incl - part;  // count in bins above my range
+      uint32_t binfloor = 0;
+      for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
+        cum += hist[b];
+        if (cum >= a.k) {
+          binfloor = (uint32_t)b;
+          break;
         }
-        break;
+      }
+      const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
+      if (tau > 0.0f) {
+        uint32_t bits;
+        __builtin_memcpy(&bits, &tau, 4);
+        atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
       }
     }
-    // broadcast current global threshold
-    shared_misc[2 * SDB_MAX_TERMS] =
-      atomicOr(a.gthresh, 0u);  // atomic read
+    if (lane == 0) {
+      uint32_t total_m = 0;
+      for (uint32_t w = 0; w < SDB_NWAVES; ++w)
+        total_m += shared_misc[2 * SDB_MAX_TERMS + 2 + w];
+      if (total_m) atomicAdd(a.total_matches, (unsigned long long)total_m);
+      shared_misc[2 * SDB_MAX_TERMS] = atomicOr(a.gthresh, 0u);
+    }
   }
   __syncthreads();
   float gtau;
@@ -736,10 +746,21 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                         sizeof(SdbScoreDoc) * ncand, hipMemcpyDeviceToHost));
   }
   // PrepareEmitBuffer analogue: filter (score > FLT_MIN,
-  // doc_collector.hpp:58), exact select under (score desc, seg, doc)
+  // doc_collector.hpp:58) AND by the final global threshold (early windows
+  // appended against a weaker bound; the final bound is still <= the true
+  // k-th score, so this drops no top-k member), then exact select under
+  // (score desc, seg, doc).
+  float gtau_final = 0.0f;
+  {
+    uint32_t bits = 0;
+    HIP_CHECK(hipMemcpy(&bits, ctx->d_gthresh, 4, hipMemcpyDeviceToHost));
+    std::memcpy(&gtau_final, &bits, 4);
+  }
+  const float floor_score = gtau_final > FLT_MIN ? gtau_final : FLT_MIN;
   size_t n = 0;
   for (size_t i = 0; i < cands.size(); ++i)
-    if (cands[i].score > FLT_MIN) cands[n++] = cands[i];
+    if (cands[i].score >= floor_score && cands[i].score > FLT_MIN)
+      cands[n++] = cands[i];
   cands.resize(n);
   auto cmp = [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
     if (x.score != y.score) return x.score > y.score;
@@ -747,7 +768,9 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     return x.doc < y.doc;
   };
   const size_t kk = std::min<size_t>(k, cands.size());
-  std::partial_sort(cands.begin(), cands.begin() + kk, cands.end(), cmp);
+  if (kk < cands.size())
+    std::nth_element(cands.begin(), cands.begin() + kk, cands.end(), cmp);
+  std::sort(cands.begin(), cands.begin() + kk, cmp);
   std::copy(cands.begin(), cands.begin() + kk, hits);
   *out_count = (uint32_t)kk;
   *total_matches = *ctx->h_matches;
