@@ -39,6 +39,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <vector>
+#include <chrono>
 
 #include "../../../include/sdb_gpu.h"
 #include "sdb_internal.h"
@@ -740,11 +741,18 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   }
   if (ctx->h_counts[1]) return SDB_ERR_OOM;  // candidate overflow
   const uint32_t ncand = ctx->h_counts[0];
+  ctx->last_ncand = ncand;
+  const auto t_rb0 = std::chrono::steady_clock::now();
   std::vector<SdbScoreDoc> cands(ncand);
   if (ncand) {
     HIP_CHECK(hipMemcpy(cands.data(), ctx->d_cands,
                         sizeof(SdbScoreDoc) * ncand, hipMemcpyDeviceToHost));
   }
+  ctx->last_readback_ms =
+    std::chrono::duration<double, std::milli>(
+      std::chrono::steady_clock::now() - t_rb0)
+      .count();
+  const auto t_sel0 = std::chrono::steady_clock::now();
   // PrepareEmitBuffer analogue: filter (score > FLT_MIN,
   // doc_collector.hpp:58) AND by the final global threshold (early windows
   // appended against a weaker bound; the final bound is still <= the true
@@ -774,6 +782,11 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   std::copy(cands.begin(), cands.begin() + kk, hits);
   *out_count = (uint32_t)kk;
   *total_matches = *ctx->h_matches;
+  ctx->last_gtau = gtau_final;
+  ctx->last_select_ms =
+    std::chrono::duration<double, std::milli>(
+      std::chrono::steady_clock::now() - t_sel0)
+      .count();
   return SDB_OK;
 }
 
@@ -817,6 +830,18 @@ int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
 int sdb_gpu_last_kernel_ms(SdbGpuCtx* ctx, double* ms) {
   if (!ctx || !ms) return SDB_ERR_INVALID;
   *ms = ctx->last_kernel_ms;
+  return SDB_OK;
+}
+
+// breakdown of the last execute_topk (diagnostics for bench --debug)
+int sdb_gpu_last_stats(SdbGpuCtx* ctx, double* kernel_ms, unsigned* ncand,
+                       float* gtau, double* readback_ms, double* select_ms) {
+  if (!ctx) return SDB_ERR_INVALID;
+  if (kernel_ms) *kernel_ms = ctx->last_kernel_ms;
+  if (ncand) *ncand = ctx->last_ncand;
+  if (gtau) *gtau = ctx->last_gtau;
+  if (readback_ms) *readback_ms = ctx->last_readback_ms;
+  if (select_ms) *select_ms = ctx->last_select_ms;
   return SDB_OK;
 }
 

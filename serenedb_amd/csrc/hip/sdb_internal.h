@@ -26,4 +26,9 @@ struct SdbGpuCtx {
   unsigned long long* h_matches;
   hipEvent_t ev_a, ev_b;   // bracket the window kernels of one execute call
   double last_kernel_ms;   // read back via sdb_gpu_last_kernel_ms
+  // breakdown of the last execute_topk (sdb_gpu_last_stats)
+  unsigned last_ncand;
+  float last_gtau;
+  double last_readback_ms;
+  double last_select_ms;
 };
