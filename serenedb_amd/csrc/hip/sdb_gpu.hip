@@ -304,6 +304,7 @@ struct WindowArgs {
   float smax;          // score upper bound (sum of term num)
   uint32_t seg_idx;
   uint32_t* gthresh;   // float bits, monotone under atomicMax
+  uint32_t* ghist;     // global 256-bin histogram (monotone counts)
   SdbScoreDoc* cands;
   uint32_t* cand_count;
   uint32_t cand_cap;
@@ -402,14 +403,21 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   if (lane == 0) shared_misc[2 * SDB_MAX_TERMS + 2 + wave] = wm;
   __syncthreads();  // hist + per-wave match counts complete
 
-  // local k-th lower bound from the histogram: wave 0 does a suffix scan
-  // (lane l owns the 4 bins [252-4l .. 255-4l]); the first lane whose
-  // suffix count reaches k refines the exact bin and publishes tau_w
+  // merge this window's histogram into the GLOBAL histogram, then derive
+  // the k-th lower bound from the GLOBAL suffix counts. A per-window bound
+  // is weak when k ~ window matches (measured: 3.85M candidates at k=1000);
+  // the global histogram converges to the true k-th's bin after a few
+  // hundred windows. Counts are monotone, so any torn read of bins still
+  // certifies ">= k real scores >= bin floor" — a valid lower bound.
+  for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
+    if (hist[b]) atomicAdd(&a.ghist[b], hist[b]);
+  __syncthreads();
   if (wave == 0) {
     uint32_t part = 0;
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      part += hist[SDB_HIST_BINS - 4 * lane - 4 + j];
+      part += __hip_atomic_load(&a.ghist[SDB_HIST_BINS - 4 * lane - 4 + j],
+                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     const uint32_t suff_incl = wave_incl_scan(part, lane);  // bins >= 252-4l
     const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
     const bool winner =
@@ -418,7 +426,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       uint32_t cum = suff_incl - part;  // count in bins above my range
       uint32_t binfloor = 0;
       for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
-        cum += hist[b];
+        cum += __hip_atomic_load(&a.ghist[b], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT);
         if (cum >= a.k) {
           binfloor = (uint32_t)b;
           break;
@@ -551,6 +560,7 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   HIP_CHECK(hipMalloc(&ctx->d_cand_count, 4));
   HIP_CHECK(hipMalloc(&ctx->d_total_matches, 8));
   HIP_CHECK(hipMalloc(&ctx->d_gthresh, 4));
+  HIP_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS));
   HIP_CHECK(hipMalloc(&ctx->d_overflow, 4));
   HIP_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
   HIP_CHECK(hipHostMalloc(&ctx->h_counts, 8));
@@ -567,6 +577,7 @@ int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   hipFree(ctx->d_cand_count);
   hipFree(ctx->d_total_matches);
   hipFree(ctx->d_gthresh);
+  hipFree(ctx->d_ghist);
   hipFree(ctx->d_overflow);
   hipFree(ctx->d_terms);
   hipHostFree(ctx->h_counts);
@@ -675,6 +686,7 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_cand_count, 0, 4, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_total_matches, 0, 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_gthresh, 0, 4, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_ghist, 0, 4 * SDB_HIST_BINS, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
@@ -709,6 +721,7 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.smax = smax;
     a.seg_idx = s;
     a.gthresh = ctx->d_gthresh;
+    a.ghist = ctx->d_ghist;
     a.cands = ctx->d_cands;
     a.cand_count = ctx->d_cand_count;
     a.cand_cap = SDB_CAND_CAP;

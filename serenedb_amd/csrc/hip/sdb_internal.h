@@ -20,6 +20,7 @@ struct SdbGpuCtx {
   uint32_t* d_cand_count;
   unsigned long long* d_total_matches;
   uint32_t* d_gthresh;  // float bits
+  uint32_t* d_ghist;    // global 256-bin score histogram (threshold tightening)
   TermDev* d_terms;
   uint32_t* d_overflow;
   uint32_t* h_counts;  // pinned: [cand_count, overflow]
