@@ -194,11 +194,11 @@ def bench_bm25(args):
     value = total_postings_global * args.steps / elapsed  # postings scored/s
 
     # ---- roofline (rank 0's shard kernel): algorithmic bytes per launch ----
-    # postings payload (compressed docs+freqs incl. tag bytes) + 4 B norm per
-    # posting + 32 B descriptor per touched block
+    # postings payload (compressed docs+freqs+embedded norm blocks, tag
+    # bytes included — format v2 materializes the norm-column gather into
+    # the payload stream) + 28 B descriptor per touched block
     local_postings = sum(terms[t].df for t in range(nterms))
-    algo_bytes = (local_payload_bytes + 4 * local_postings +
-                  32 * local_desc_blocks)
+    algo_bytes = local_payload_bytes + 28 * local_desc_blocks
     kernel_s = kernel_ms_acc / 1000.0
     achieved_gbs = (algo_bytes * args.steps / kernel_s / 1e9) if kernel_s else 0
     roofline = {

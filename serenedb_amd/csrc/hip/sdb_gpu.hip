@@ -298,6 +298,7 @@ struct WindowArgs {
   const uint8_t* payload;
   const uint32_t* norms;
   uint32_t doc_count;  // docs 1..doc_count (local ids within this segment)
+  uint32_t norm_stream;  // v2 segments: per-block norm blocks (flags=freq sz)
   uint32_t nterms;
   uint32_t min_match;
   uint32_t k;
@@ -318,8 +319,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* swin = (float*)smem;                       // SDB_WIN_DOCS * 4
   uint8_t* cwin = (uint8_t*)(swin + SDB_WIN_DOCS);  // SDB_WIN_DOCS
-  uint32_t* scratch = (uint32_t*)(cwin + SDB_WIN_DOCS);  // 8 waves * 256
-  uint32_t* hist = scratch + SDB_NWAVES * 256;            // 256
+  uint32_t* scratch = (uint32_t*)(cwin + SDB_WIN_DOCS);  // 8 waves * 384
+  uint32_t* hist = scratch + SDB_NWAVES * 384;            // 256
   uint32_t* shared_misc = hist + SDB_HIST_BINS;           // ranges + bcast
   // shared_misc layout: [2*SDB_MAX_TERMS] block ranges (lo32/hi32 pairs as
   // u32 relative counts fit 32 bit? desc indices are u64 — store as u32
@@ -359,21 +360,24 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     const uint32_t b_lo = shared_misc[2 * t];
     const uint32_t b_hi = shared_misc[2 * t + 1];  // first block fully after
     const uint8_t* pl = a.payload + te.payload_begin;
-    uint32_t* dbuf = scratch + wave * 256;
+    uint32_t* dbuf = scratch + wave * 384;
     uint32_t* fbuf = dbuf + 128;
+    uint32_t* nbuf = fbuf + 128;
     const float num = te.num, nc = te.nc, nl = te.nl;
     for (uint64_t b = te.desc_begin + b_lo + wave;
          b < te.desc_begin + b_hi; b += SDB_NWAVES) {
       const SdbBlockDesc d = a.desc[b];
       decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
       decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+      if (a.norm_stream)  // v2: norm block follows the freq block
+        decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane, nbuf);
       // wave-local scratch: writes visible to the same wave after lgkm wait
       // (compiler inserts); score docs inside the window
       for (uint32_t j = lane; j < d.len; j += 64) {
         const uint32_t doc = dbuf[j];
         if (doc < lo || doc > hi) continue;
         const uint32_t freq = fbuf[j];
-        const uint32_t norm = a.norms[doc];
+        const uint32_t norm = a.norm_stream ? nbuf[j] : a.norms[doc];
         const float c1 = nc + nl * (float)norm;
         const float s = num - num * c1 / (c1 + (float)freq);
         const uint32_t off = doc - lo;
@@ -536,7 +540,8 @@ int parse_blob(const void* blob, size_t size, SdbSegHeader* hdr_out) {
   if (!blob || size < sizeof(SdbSegHeader)) return SDB_ERR_BAD_SEGMENT;
   SdbSegHeader hdr;
   std::memcpy(&hdr, blob, sizeof(hdr));
-  if (hdr.magic != SDB_SEG_MAGIC || hdr.version != 1 || hdr.blob_size > size)
+  if (hdr.magic != SDB_SEG_MAGIC || hdr.version < 1 || hdr.version > 2 ||
+      hdr.blob_size > size)
     return SDB_ERR_BAD_SEGMENT;
   *hdr_out = hdr;
   return SDB_OK;
@@ -690,7 +695,7 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
-                           SDB_NWAVES * 256 * 4 + SDB_HIST_BINS * 4 +
+                           SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
                            (2 * SDB_MAX_TERMS + 2 + SDB_NWAVES) * 4;
 
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
@@ -715,6 +720,7 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.payload = seg->payload;
     a.norms = seg->norms;
     a.doc_count = seg->hdr.doc_count;
+    a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
     a.nterms = plan->nterms;
     a.min_match = plan->min_match ? plan->min_match : 1;
     a.k = k;
