@@ -367,10 +367,19 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (uint64_t b = te.desc_begin + b_lo + wave;
          b < te.desc_begin + b_hi; b += SDB_NWAVES) {
       const SdbBlockDesc d = a.desc[b];
+#ifdef SDB_ABLATE_DECODE
+      // ablation: skip decode, fabricate in-window docs (measures the rest)
+      for (uint32_t j = lane; j < d.len; j += 64) {
+        dbuf[j] = lo + ((uint32_t)(b * 131u) + j * 7u) % (hi - lo + 1u);
+        fbuf[j] = 1u + (j & 7u);
+        nbuf[j] = 100u + j;
+      }
+#else
       decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
       decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
       if (a.norm_stream)  // v2: norm block follows the freq block
         decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane, nbuf);
+#endif
       // wave-local scratch: writes visible to the same wave after lgkm wait
       // (compiler inserts); score docs inside the window
       for (uint32_t j = lane; j < d.len; j += 64) {
@@ -378,16 +387,30 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         if (doc < lo || doc > hi) continue;
         const uint32_t freq = fbuf[j];
         const uint32_t norm = a.norm_stream ? nbuf[j] : a.norms[doc];
+#ifdef SDB_ABLATE_SCORE
+        // ablation: keep decoded values alive without scoring/window writes
+        asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
+#else
         const float c1 = nc + nl * (float)norm;
         const float s = num - num * c1 / (c1 + (float)freq);
         const uint32_t off = doc - lo;
         swin[off] += s;        // unique doc within the term: no atomics
         cwin[off] = (uint8_t)(cwin[off] + 1u);
+#endif
       }
     }
     __syncthreads();  // term-major merge order (bit-exact vs oracle)
   }
 
+#ifdef SDB_ABLATE_TAIL
+  // ablation: keep the windows alive, skip histogram/threshold/append
+  if (tid == 0) {
+    uint32_t x = (uint32_t)swin[0] + cwin[0];
+    asm volatile("" ::"v"(x));
+    atomicAdd(a.total_matches, 0ull);
+  }
+  return;
+#endif
   // histogram of matching scores + local match count
   const uint32_t mm = a.min_match ? a.min_match : 1u;
   const float inv_smax = (float)SDB_HIST_BINS / a.smax;
