@@ -313,236 +313,252 @@ struct WindowArgs {
   uint32_t* overflow;
 };
 
+// Persistent-range window kernel: each workgroup owns a CONTIGUOUS range of
+// doc windows and walks them with per-term block cursors carried in LDS —
+// the GPU analogue of the reference's streaming PostingIterator cursors
+// (formats/posting/iterator_doc.hpp:36-344) over BlockDisjunction windows.
+// One binary search per term per WORKGROUP (not per window); block
+// descriptors advance monotonically. Ablation history: the
+// per-window-grid version spent 0.18 ms in per-window init (dominated by
+// 2*nterms serial-latency binary searches) and ran 16 dispatch rounds.
 __launch_bounds__(SDB_NTHREADS, 1) __global__
 void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
-  // one extern LDS region, 16B-aligned carves (guide §6 G17)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* swin = (float*)smem;                       // SDB_WIN_DOCS * 4
   uint8_t* cwin = (uint8_t*)(swin + SDB_WIN_DOCS);  // SDB_WIN_DOCS
   uint32_t* scratch = (uint32_t*)(cwin + SDB_WIN_DOCS);  // 8 waves * 384
   uint32_t* hist = scratch + SDB_NWAVES * 384;            // 256
-  uint32_t* shared_misc = hist + SDB_HIST_BINS;           // ranges + bcast
-  // shared_misc layout: [2*SDB_MAX_TERMS] block ranges (lo32/hi32 pairs as
-  // u32 relative counts fit 32 bit? desc indices are u64 — store as u32
-  // offsets relative to term desc_begin; per-term blocks < 2^32) + [2] misc
+  uint32_t* shared_misc = hist + SDB_HIST_BINS;     // bcast + wave counts
+  uint32_t* cursors = shared_misc + 2 + SDB_NWAVES; // per-term block cursor
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
   const uint32_t wave = tid >> 6;
-
-  const uint32_t lo = 1u + (uint32_t)blockIdx.x * SDB_WIN_DOCS;
-  if (lo > a.doc_count) return;
-  const uint32_t hi = min(lo + SDB_WIN_DOCS - 1u, a.doc_count);
-  const uint32_t wlen = hi - lo + 1u;
-
-  // zero windows + histogram
-  for (uint32_t i = tid; i < SDB_WIN_DOCS; i += SDB_NTHREADS) swin[i] = 0.0f;
-  for (uint32_t i = tid; i < SDB_WIN_DOCS / 4; i += SDB_NTHREADS)
-    ((uint32_t*)cwin)[i] = 0;
-  for (uint32_t i = tid; i < SDB_HIST_BINS; i += SDB_NTHREADS) hist[i] = 0;
-
-  // block ranges per term (parallel over 2*nterms threads)
-  if (tid < 2 * a.nterms) {
-    const uint32_t t = tid >> 1;
-    const TermDev te = terms[t];
-    uint64_t v;
-    if ((tid & 1) == 0)
-      v = lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, lo);
-    else
-      v = lower_bound_prev_doc(a.desc, te.desc_begin, te.desc_end, hi);
-    shared_misc[tid] = (uint32_t)(v - te.desc_begin);
-  }
-  __syncthreads();
-
-  // term-major phases
-  for (uint32_t t = 0; t < a.nterms; ++t) {
-    const TermDev te = terms[t];
-    const uint32_t b_lo = shared_misc[2 * t];
-    const uint32_t b_hi = shared_misc[2 * t + 1];  // first block fully after
-    const uint8_t* pl = a.payload + te.payload_begin;
-    uint32_t* dbuf = scratch + wave * 384;
-    uint32_t* fbuf = dbuf + 128;
-    uint32_t* nbuf = fbuf + 128;
-    const float num = te.num, nc = te.nc, nl = te.nl;
-    for (uint64_t b = te.desc_begin + b_lo + wave;
-         b < te.desc_begin + b_hi; b += SDB_NWAVES) {
-      const SdbBlockDesc d = a.desc[b];
-#ifdef SDB_ABLATE_DECODE
-      // ablation: skip decode, fabricate in-window docs (measures the rest)
-      for (uint32_t j = lane; j < d.len; j += 64) {
-        dbuf[j] = lo + ((uint32_t)(b * 131u) + j * 7u) % (hi - lo + 1u);
-        fbuf[j] = 1u + (j & 7u);
-        nbuf[j] = 100u + j;
-      }
-#else
-      decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
-      decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
-      if (a.norm_stream)  // v2: norm block follows the freq block
-        decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane, nbuf);
-#endif
-      // wave-local scratch: writes visible to the same wave after lgkm wait
-      // (compiler inserts); score docs inside the window
-      for (uint32_t j = lane; j < d.len; j += 64) {
-        const uint32_t doc = dbuf[j];
-        if (doc < lo || doc > hi) continue;
-        const uint32_t freq = fbuf[j];
-        const uint32_t norm = a.norm_stream ? nbuf[j] : a.norms[doc];
-#ifdef SDB_ABLATE_SCORE
-        // ablation: keep decoded values alive without scoring/window writes
-        asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
-#else
-        const float c1 = nc + nl * (float)norm;
-        const float s = num - num * c1 / (c1 + (float)freq);
-        const uint32_t off = doc - lo;
-        swin[off] += s;        // unique doc within the term: no atomics
-        cwin[off] = (uint8_t)(cwin[off] + 1u);
-#endif
-      }
-    }
-    __syncthreads();  // term-major merge order (bit-exact vs oracle)
-  }
-
-#ifdef SDB_ABLATE_TAIL
-  // ablation: keep the windows alive, skip histogram/threshold/append
-  if (tid == 0) {
-    uint32_t x = (uint32_t)swin[0] + cwin[0];
-    asm volatile("" ::"v"(x));
-    atomicAdd(a.total_matches, 0ull);
-  }
-  return;
-#endif
-  // histogram of matching scores + local match count
   const uint32_t mm = a.min_match ? a.min_match : 1u;
   const float inv_smax = (float)SDB_HIST_BINS / a.smax;
-  uint32_t my_matches = 0;
-  for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
-    if (cwin[off] >= mm) {
-      ++my_matches;
-      uint32_t bin = (uint32_t)(swin[off] * inv_smax);
-      if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
-      atomicAdd(&hist[bin], 1u);
-    }
-  }
-  // reduce match count: wave reduce, then ONE atomic per workgroup
-  uint32_t wm = my_matches;
-#pragma unroll
-  for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
-  if (lane == 0) shared_misc[2 * SDB_MAX_TERMS + 2 + wave] = wm;
-  __syncthreads();  // hist + per-wave match counts complete
-
-  // merge this window's histogram into the GLOBAL histogram, then derive
-  // the k-th lower bound from the GLOBAL suffix counts. A per-window bound
-  // is weak when k ~ window matches (measured: 3.85M candidates at k=1000);
-  // the global histogram converges to the true k-th's bin after a few
-  // hundred windows. Counts are monotone, so any torn read of bins still
-  // certifies ">= k real scores >= bin floor" — a valid lower bound.
-  //
-  // The histogram is sharded 8 ways by XCD (ablation showed 0.43 ms in this
-  // tail, dominated by ~1M atomics onto 4 cachelines in one L2 home), and
-  // bins strictly below the already-published global threshold's bin are
-  // skipped — they can never change a suffix count at or above the k-th bin.
   uint32_t* gh = a.ghist + (blockIdx.x & 7u) * SDB_HIST_BINS;
-  uint32_t known_bin = 0;
-  {
-    float gt_now;
-    const uint32_t bits =
-      __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    __builtin_memcpy(&gt_now, &bits, 4);
-    known_bin = (uint32_t)(gt_now * inv_smax);  // tau is an exact bin floor
-    if (known_bin >= SDB_HIST_BINS) known_bin = SDB_HIST_BINS - 1;
+
+  // contiguous window range of this workgroup
+  const uint32_t nwin = (a.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
+  const uint32_t per = (nwin + gridDim.x - 1) / gridDim.x;
+  const uint32_t w_lo = blockIdx.x * per;
+  const uint32_t w_hi = min(nwin, w_lo + per);
+  if (w_lo >= w_hi) return;
+
+  // one binary search per term: first block with last_doc >= first window lo
+  if (tid < a.nterms) {
+    const TermDev te = terms[tid];
+    const uint32_t first_lo = 1u + w_lo * SDB_WIN_DOCS;
+    cursors[tid] = (uint32_t)(
+      lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, first_lo) -
+      te.desc_begin);
   }
-  for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
-    if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
   __syncthreads();
-  if (wave == 0) {
-    uint32_t part = 0;
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
-#pragma unroll
-      for (int s = 0; s < 8; ++s)
-        part += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
-                                  __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    }
-    const uint32_t suff_incl = wave_incl_scan(part, lane);  // bins >= 252-4l
-    const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
-    const bool winner =
-      suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
-    if (winner) {
-      uint32_t cum = suff_incl - part;  // count in bins above my range
-      uint32_t binfloor = 0;
-      for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
-#pragma unroll
-        for (int s = 0; s < 8; ++s)
-          cum += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        if (cum >= a.k) {
-          binfloor = (uint32_t)b;
-          break;
+
+  for (uint32_t w = w_lo; w < w_hi; ++w) {
+    const uint32_t lo = 1u + w * SDB_WIN_DOCS;
+    if (lo > a.doc_count) break;
+    const uint32_t hi = min(lo + SDB_WIN_DOCS - 1u, a.doc_count);
+    const uint32_t wlen = hi - lo + 1u;
+
+    // zero windows + histogram
+    for (uint32_t i = tid; i < SDB_WIN_DOCS; i += SDB_NTHREADS)
+      swin[i] = 0.0f;
+    for (uint32_t i = tid; i < SDB_WIN_DOCS / 4; i += SDB_NTHREADS)
+      ((uint32_t*)cwin)[i] = 0;
+    for (uint32_t i = tid; i < SDB_HIST_BINS; i += SDB_NTHREADS) hist[i] = 0;
+    __syncthreads();
+
+    // term-major phases (fixed fp32 merge order -> bit-exact vs the oracle)
+    for (uint32_t t = 0; t < a.nterms; ++t) {
+      const TermDev te = terms[t];
+      const uint8_t* pl = a.payload + te.payload_begin;
+      const uint64_t dend = te.desc_end;
+      uint32_t* dbuf = scratch + wave * 384;
+      uint32_t* fbuf = dbuf + 128;
+      uint32_t* nbuf = fbuf + 128;
+      const float num = te.num, nc = te.nc, nl = te.nl;
+      // waves walk blocks from the shared cursor; every block from the
+      // cursor has last_doc >= lo (maintained below); stop at first block
+      // whose first doc (> prev_doc) lies beyond the window
+      for (uint64_t b = te.desc_begin + cursors[t] + wave; b < dend;
+           b += SDB_NWAVES) {
+        const SdbBlockDesc d = a.desc[b];
+        if (d.prev_doc >= hi) break;  // first doc > hi
+#ifdef SDB_ABLATE_DECODE
+        for (uint32_t j = lane; j < d.len; j += 64) {
+          dbuf[j] = lo + ((uint32_t)(b * 131u) + j * 7u) % (hi - lo + 1u);
+          fbuf[j] = 1u + (j & 7u);
+          nbuf[j] = 100u + j;
+        }
+#else
+        decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
+        decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+        if (a.norm_stream)  // v2: norm block follows the freq block
+          decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane,
+                                 nbuf);
+#endif
+        for (uint32_t j = lane; j < d.len; j += 64) {
+          const uint32_t doc = dbuf[j];
+          if (doc < lo || doc > hi) continue;
+          const uint32_t freq = fbuf[j];
+          const uint32_t norm = a.norm_stream ? nbuf[j] : a.norms[doc];
+#ifdef SDB_ABLATE_SCORE
+          asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
+#else
+          const float c1 = nc + nl * (float)norm;
+          const float s = num - num * c1 / (c1 + (float)freq);
+          const uint32_t off = doc - lo;
+          swin[off] += s;      // unique doc within the term: no atomics
+          cwin[off] = (uint8_t)(cwin[off] + 1u);
+#endif
         }
       }
-      const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
-      if (tau > 0.0f) {
-        uint32_t bits;
-        __builtin_memcpy(&bits, &tau, 4);
-        atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
+      __syncthreads();
+      // advance the cursor past blocks fully consumed by this window
+      // (desc entries just touched; L1-warm)
+      if (tid == t) {
+        uint32_t cur = cursors[t];
+        while (te.desc_begin + cur < dend &&
+               a.desc[te.desc_begin + cur].last_doc <= hi)
+          ++cur;
+        cursors[t] = cur;
       }
+      __syncthreads();
     }
-    if (lane == 0) {
-      uint32_t total_m = 0;
-      for (uint32_t w = 0; w < SDB_NWAVES; ++w)
-        total_m += shared_misc[2 * SDB_MAX_TERMS + 2 + w];
-      if (total_m) atomicAdd(a.total_matches, (unsigned long long)total_m);
-      shared_misc[2 * SDB_MAX_TERMS] = atomicOr(a.gthresh, 0u);
-    }
-  }
-  __syncthreads();
-  float gtau;
-  {
-    const uint32_t bits = shared_misc[2 * SDB_MAX_TERMS];
-    __builtin_memcpy(&gtau, &bits, 4);
-  }
 
-  // append candidates with score >= gtau (ties at the k-th kept).
-  // ONE global atomicAdd per WORKGROUP: a per-word cursor would serialize
-  // ~1.5M atomics across the grid (measured 16 ms at 100M docs — the
-  // microarch 'dequeue' row: one word sustains ~88 ops/us). Per-thread
-  // accept counts -> block exclusive scan (in the now-free decode scratch)
-  // -> one cursor bump -> scatter.
-  uint32_t my_cnt = 0;
-  for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS)
-    if (cwin[off] >= mm && swin[off] >= gtau) ++my_cnt;
-  // block exclusive scan over 512 per-thread counts (scratch is free here)
-  uint32_t* cnts = scratch;  // [SDB_NTHREADS] + [SDB_NWAVES] wave totals
-  const uint32_t incl = wave_incl_scan(my_cnt, lane);
-  if (lane == 63) cnts[SDB_NTHREADS + wave] = incl;
-  __syncthreads();
-  uint32_t wave_base = 0;
-  for (uint32_t w = 0; w < wave; ++w) wave_base += cnts[SDB_NTHREADS + w];
-  const uint32_t my_excl = wave_base + incl - my_cnt;
-  uint32_t block_total = 0;
-  for (uint32_t w = 0; w < SDB_NWAVES; ++w)
-    block_total += cnts[SDB_NTHREADS + w];
-  if (tid == 0)
-    shared_misc[2 * SDB_MAX_TERMS + 1] =
-      block_total ? atomicAdd(a.cand_count, block_total) : 0u;
-  __syncthreads();
-  const uint32_t base = shared_misc[2 * SDB_MAX_TERMS + 1];
-  if (base + block_total > a.cand_cap) {
-    if (tid == 0) atomicExch(a.overflow, 1u);
-    return;
-  }
-  uint32_t pos = base + my_excl;
-  for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
-    if (cwin[off] >= mm) {
-      const float s = swin[off];
-      if (s >= gtau) {
-        a.cands[pos].score = s;
-        a.cands[pos].doc = lo + off;
-        a.cands[pos].segment_idx = a.seg_idx;
-        ++pos;
+#ifdef SDB_ABLATE_TAIL
+    if (tid == 0) {
+      uint32_t x = (uint32_t)swin[0] + cwin[0];
+      asm volatile("" ::"v"(x));
+      atomicAdd(a.total_matches, 0ull);
+    }
+    __syncthreads();
+    continue;
+#else
+    // histogram of matching scores + local match count
+    uint32_t my_matches = 0;
+    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
+      if (cwin[off] >= mm) {
+        ++my_matches;
+        uint32_t bin = (uint32_t)(swin[off] * inv_smax);
+        if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
+        atomicAdd(&hist[bin], 1u);
       }
     }
+    uint32_t wm = my_matches;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
+    if (lane == 0) shared_misc[2 + wave] = wm;
+    __syncthreads();  // hist + per-wave match counts complete
+
+    // merge window histogram into the per-XCD global shard (skip bins below
+    // the published threshold bin: they cannot change any suffix count at or
+    // above the k-th bin), then derive tau from the global suffix counts
+    uint32_t known_bin = 0;
+    {
+      float gt_now;
+      const uint32_t bits = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                              __HIP_MEMORY_SCOPE_AGENT);
+      __builtin_memcpy(&gt_now, &bits, 4);
+      known_bin = (uint32_t)(gt_now * inv_smax);
+      if (known_bin >= SDB_HIST_BINS) known_bin = SDB_HIST_BINS - 1;
+    }
+    for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
+      if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
+    __syncthreads();
+    if (wave == 0) {
+      uint32_t part = 0;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
+#pragma unroll
+        for (int s = 0; s < 8; ++s)
+          part += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
+                                    __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+      }
+      const uint32_t suff_incl = wave_incl_scan(part, lane);
+      const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
+      const bool winner =
+        suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
+      if (winner) {
+        uint32_t cum = suff_incl - part;
+        uint32_t binfloor = 0;
+        for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
+          uint32_t add = 0;
+#pragma unroll
+          for (int s = 0; s < 8; ++s)
+            add += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+          cum += add;
+          if (cum >= a.k) {
+            binfloor = (uint32_t)b;
+            break;
+          }
+        }
+        const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
+        if (tau > 0.0f) {
+          uint32_t bits;
+          __builtin_memcpy(&bits, &tau, 4);
+          atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
+        }
+      }
+      if (lane == 0) {
+        uint32_t total_m = 0;
+        for (uint32_t v = 0; v < SDB_NWAVES; ++v)
+          total_m += shared_misc[2 + v];
+        if (total_m) atomicAdd(a.total_matches, (unsigned long long)total_m);
+        shared_misc[0] = atomicOr(a.gthresh, 0u);
+      }
+    }
+    __syncthreads();
+    float gtau;
+    {
+      const uint32_t bits = shared_misc[0];
+      __builtin_memcpy(&gtau, &bits, 4);
+    }
+
+    // append candidates with score >= gtau: per-thread counts, block scan,
+    // ONE cursor atomicAdd per workgroup per window (the per-wave version
+    // serialized ~1.5M atomics on one word — microarch 'dequeue' row)
+    uint32_t my_cnt = 0;
+    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS)
+      if (cwin[off] >= mm && swin[off] >= gtau) ++my_cnt;
+    uint32_t* cnts = scratch;  // decode scratch is free here
+    const uint32_t incl = wave_incl_scan(my_cnt, lane);
+    if (lane == 63) cnts[SDB_NTHREADS + wave] = incl;
+    __syncthreads();
+    uint32_t wave_base = 0;
+    for (uint32_t v = 0; v < wave; ++v) wave_base += cnts[SDB_NTHREADS + v];
+    const uint32_t my_excl = wave_base + incl - my_cnt;
+    uint32_t block_total = 0;
+    for (uint32_t v = 0; v < SDB_NWAVES; ++v)
+      block_total += cnts[SDB_NTHREADS + v];
+    if (tid == 0)
+      shared_misc[1] =
+        block_total ? atomicAdd(a.cand_count, block_total) : 0u;
+    __syncthreads();
+    const uint32_t base = shared_misc[1];
+    if (base + block_total > a.cand_cap) {
+      if (tid == 0) atomicExch(a.overflow, 1u);
+      return;
+    }
+    uint32_t pos = base + my_excl;
+    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
+      if (cwin[off] >= mm) {
+        const float s = swin[off];
+        if (s >= gtau) {
+          a.cands[pos].score = s;
+          a.cands[pos].doc = lo + off;
+          a.cands[pos].segment_idx = a.seg_idx;
+          ++pos;
+        }
+      }
+    }
+    __syncthreads();  // window state reused next iteration
+#endif
   }
 }
 
@@ -740,7 +756,7 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
-                           (2 * SDB_MAX_TERMS + 2 + SDB_NWAVES) * 4;
+                           (2 + SDB_NWAVES + SDB_MAX_TERMS) * 4;
 
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   for (uint32_t s = 0; s < nsegs; ++s) {
@@ -779,7 +795,8 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.overflow = ctx->d_overflow;
     const uint32_t nwin =
       (seg->hdr.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
-    hipLaunchKernelGGL(topk_window_kernel, dim3(nwin), dim3(SDB_NTHREADS),
+    const uint32_t ngrid = nwin < 256u ? nwin : 256u;  // 1 WG/CU persistent
+    hipLaunchKernelGGL(topk_window_kernel, dim3(ngrid), dim3(SDB_NTHREADS),
                        lds_bytes, ctx->stream, a, ctx->d_terms);
     HIP_CHECK(hipGetLastError());
     if (nsegs > 1) HIP_CHECK(hipStreamSynchronize(ctx->stream));
