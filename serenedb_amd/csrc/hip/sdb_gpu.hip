@@ -54,6 +54,7 @@
 #define SDB_NWAVES (SDB_NTHREADS / 64u)
 #define SDB_MAX_TERMS 32u
 #define SDB_HIST_BINS 256u
+#define SDB_DESC_CACHE 32u  // staged descriptors per term per window
 #define SDB_CAND_CAP (64u * 1024u * 1024u)  // 64M candidates (768 MB)
 
 #define HIP_CHECK(x)                        \
@@ -330,6 +331,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* hist = scratch + SDB_NWAVES * 384;            // 256
   uint32_t* shared_misc = hist + SDB_HIST_BINS;     // bcast + wave counts
   uint32_t* cursors = shared_misc + 2 + SDB_NWAVES; // per-term block cursor
+  // staged descriptors: SDB_DESC_CACHE per term, one coalesced load per
+  // window removes a ~900-cycle dependent desc load from every block chain
+  SdbBlockDesc* dcache = (SdbBlockDesc*)(cursors + SDB_MAX_TERMS);
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
@@ -367,6 +371,25 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (uint32_t i = tid; i < SDB_WIN_DOCS / 4; i += SDB_NTHREADS)
       ((uint32_t*)cwin)[i] = 0;
     for (uint32_t i = tid; i < SDB_HIST_BINS; i += SDB_NTHREADS) hist[i] = 0;
+    // stage this window's descriptors: term t's next SDB_DESC_CACHE descs
+    // from its cursor, as coalesced u32 reads (7 words per desc)
+    {
+      const uint32_t words_per_term = SDB_DESC_CACHE * 7u;
+      for (uint32_t i = tid; i < a.nterms * words_per_term;
+           i += SDB_NTHREADS) {
+        const uint32_t t = i / words_per_term;
+        const uint32_t wrd = i % words_per_term;
+        const TermDev te = terms[t];
+        const uint64_t b0 = te.desc_begin + cursors[t];
+        const uint32_t avail = (uint32_t)(te.desc_end > b0
+                                            ? te.desc_end - b0
+                                            : 0);
+        if (wrd < avail * 7u) {
+          ((uint32_t*)&dcache[t * SDB_DESC_CACHE])[wrd] =
+            ((const uint32_t*)&a.desc[b0])[wrd];
+        }
+      }
+    }
     __syncthreads();
 
     // term-major phases (fixed fp32 merge order -> bit-exact vs the oracle)
@@ -381,9 +404,13 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       // waves walk blocks from the shared cursor; every block from the
       // cursor has last_doc >= lo (maintained below); stop at first block
       // whose first doc (> prev_doc) lies beyond the window
-      for (uint64_t b = te.desc_begin + cursors[t] + wave; b < dend;
+      const uint32_t cur0 = cursors[t];
+      for (uint64_t b = te.desc_begin + cur0 + wave; b < dend;
            b += SDB_NWAVES) {
-        const SdbBlockDesc d = a.desc[b];
+        const uint32_t rel = (uint32_t)(b - te.desc_begin) - cur0;
+        const SdbBlockDesc d = rel < SDB_DESC_CACHE
+                                 ? dcache[t * SDB_DESC_CACHE + rel]
+                                 : a.desc[b];
         if (d.prev_doc >= hi) break;  // first doc > hi
 #ifdef SDB_ABLATE_DECODE
         for (uint32_t j = lane; j < d.len; j += 64) {
@@ -419,9 +446,15 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       // (desc entries just touched; L1-warm)
       if (tid == t) {
         uint32_t cur = cursors[t];
-        while (te.desc_begin + cur < dend &&
-               a.desc[te.desc_begin + cur].last_doc <= hi)
+        const uint32_t cur0 = cursors[t];
+        while (te.desc_begin + cur < dend) {
+          const uint32_t rel = cur - cur0;
+          const uint32_t last = rel < SDB_DESC_CACHE
+                                  ? dcache[t * SDB_DESC_CACHE + rel].last_doc
+                                  : a.desc[te.desc_begin + cur].last_doc;
+          if (last > hi) break;
           ++cur;
+        }
         cursors[t] = cur;
       }
       __syncthreads();
@@ -756,7 +789,9 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
-                           (2 + SDB_NWAVES + SDB_MAX_TERMS) * 4;
+                           (2 + SDB_NWAVES + SDB_MAX_TERMS) * 4 +
+                           sizeof(SdbBlockDesc) * SDB_DESC_CACHE *
+                             plan->nterms;
 
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   for (uint32_t s = 0; s < nsegs; ++s) {
@@ -795,7 +830,11 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.overflow = ctx->d_overflow;
     const uint32_t nwin =
       (seg->hdr.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
-    const uint32_t ngrid = nwin < 256u ? nwin : 256u;  // 1 WG/CU persistent
+    uint32_t wgs_per_cu = (uint32_t)(163840 / lds_bytes);
+    if (wgs_per_cu < 1) wgs_per_cu = 1;
+    if (wgs_per_cu > 4) wgs_per_cu = 4;
+    uint32_t ngrid = 256u * wgs_per_cu;
+    if (ngrid > nwin) ngrid = nwin;
     hipLaunchKernelGGL(topk_window_kernel, dim3(ngrid), dim3(SDB_NTHREADS),
                        lds_bytes, ctx->stream, a, ctx->d_terms);
     HIP_CHECK(hipGetLastError());
