@@ -441,23 +441,24 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #endif
         }
       }
-      __syncthreads();
-      // advance the cursor past blocks fully consumed by this window
-      // (desc entries just touched; L1-warm)
-      if (tid == t) {
-        uint32_t cur = cursors[t];
-        const uint32_t cur0 = cursors[t];
-        while (te.desc_begin + cur < dend) {
-          const uint32_t rel = cur - cur0;
-          const uint32_t last = rel < SDB_DESC_CACHE
-                                  ? dcache[t * SDB_DESC_CACHE + rel].last_doc
-                                  : a.desc[te.desc_begin + cur].last_doc;
-          if (last > hi) break;
-          ++cur;
-        }
-        cursors[t] = cur;
+      __syncthreads();  // term-major merge order (bit-exact vs oracle)
+    }
+    // advance every term's cursor once per window (cursors are only read
+    // at the NEXT window's staging/phases, after the barrier below)
+    if (tid < a.nterms) {
+      const uint32_t t = tid;
+      const TermDev te = terms[t];
+      uint32_t cur = cursors[t];
+      const uint32_t cur0 = cur;
+      while (te.desc_begin + cur < te.desc_end) {
+        const uint32_t rel = cur - cur0;
+        const uint32_t last = rel < SDB_DESC_CACHE
+                                ? dcache[t * SDB_DESC_CACHE + rel].last_doc
+                                : a.desc[te.desc_begin + cur].last_doc;
+        if (last > hi) break;
+        ++cur;
       }
-      __syncthreads();
+      cursors[t] = cur;
     }
 
 #ifdef SDB_ABLATE_TAIL
@@ -466,7 +467,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       asm volatile("" ::"v"(x));
       atomicAdd(a.total_matches, 0ull);
     }
-    __syncthreads();
+    __syncthreads();  // cursor updates visible before next window stages
     continue;
 #else
     // histogram of matching scores + local match count
