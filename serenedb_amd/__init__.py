@@ -26,6 +26,17 @@ class SdbTermRef(C.Structure):
     _fields_ = [("term_idx", C.c_uint32), ("boost", C.c_float)]
 
 
+def _copy_blob(ptr, size):
+    """Copy `size` bytes from a C pointer. ctypes.string_at truncates its
+    size argument to a C int (a >4 GB segment blob came back mod 2^32);
+    ctypes.memmove takes size_t."""
+    import numpy as np
+
+    out = np.empty(size, dtype=np.uint8)
+    C.memmove(out.ctypes.data, ptr, size)
+    return out.tobytes()
+
+
 def _load(name):
     path = os.path.join(_PKG_DIR, name)
     if not os.path.exists(path):
@@ -139,7 +150,7 @@ def build_segment(doc_count, postings, norms=None):
         df.ctypes.data_as(PU32), doc_ptrs, frq_ptrs, norm_ptr,
         C.byref(blob), C.byref(size))
     assert rc == 0, rc
-    out = C.string_at(blob, size.value)
+    out = _copy_blob(blob, size.value)
     host().sdb_host_blob_free(blob)
     return out
 
@@ -155,7 +166,7 @@ def build_synth_segment(seed, doc_lo, doc_hi, selectivities):
         C.c_uint32(len(sel)), sel.ctypes.data_as(C.POINTER(C.c_double)),
         C.byref(blob), C.byref(size))
     assert rc == 0, rc
-    out = C.string_at(blob, size.value)
+    out = _copy_blob(blob, size.value)
     host().sdb_host_blob_free(blob)
     return out
 

@@ -631,12 +631,12 @@ int check_gpu() {
 }
 
 int parse_blob(const void* blob, size_t size, SdbSegHeader* hdr_out) {
-  if (!blob || size < sizeof(SdbSegHeader)) return SDB_ERR_BAD_SEGMENT;
+  if (!blob || size < sizeof(SdbSegHeader)) return -51;
   SdbSegHeader hdr;
   std::memcpy(&hdr, blob, sizeof(hdr));
-  if (hdr.magic != SDB_SEG_MAGIC || hdr.version < 1 || hdr.version > 2 ||
-      hdr.blob_size > size)
-    return SDB_ERR_BAD_SEGMENT;
+  if (hdr.magic != SDB_SEG_MAGIC) return -52;
+  if (hdr.version < 1 || hdr.version > 2) return -53;
+  if (hdr.blob_size > size) return -54;
   *hdr_out = hdr;
   return SDB_OK;
 }
