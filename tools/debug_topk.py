@@ -11,8 +11,15 @@ import serenedb_amd as sa
 docs = int(sys.argv[1]) if len(sys.argv) > 1 else 100_000_000
 sels = [0.10, 0.05, 0.02, 0.01]
 t0 = time.time()
-blob = sa.build_synth_segment(43, 1, docs, sels)
-print(f"build {time.time()-t0:.1f}s blob {len(blob)/1e6:.1f}MB")
+cache = os.environ.get("SDB_BLOB_CACHE")
+if cache and os.path.exists(cache):
+    blob = open(cache, "rb").read()
+    print(f"loaded cached blob {len(blob)/1e6:.1f}MB")
+else:
+    blob = sa.build_synth_segment(43, 1, docs, sels)
+    print(f"build {time.time()-t0:.1f}s blob {len(blob)/1e6:.1f}MB")
+    if cache:
+        open(cache, "wb").write(blob)
 ctx = sa.GpuContext(0)
 seg = ctx.load_segment(blob)
 lib = sa.gpu()
