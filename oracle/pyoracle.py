@@ -290,3 +290,36 @@ def ref_unpack(payload, bits):
     r.simdunpack(buf.ctypes.data_as(C.c_void_p), out.ctypes.data_as(PU32),
                  C.c_uint32(bits))
     return out
+
+
+def execute_topk_hybrid(blob, term_idx, boosts, k, col, flo, fhi, nbuckets,
+                        min_match=1, k1=1.2, b=0.75, global_stats=None):
+    """Hybrid exact top-k: BM25 match AND col BETWEEN [flo,fhi], plus
+    per-bucket COUNT/SUM over surviving matches."""
+    buf = np.frombuffer(blob, dtype=np.uint8)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    col = np.ascontiguousarray(col, dtype=np.int64)
+    g_dwf, g_ttf, g_dwt_ptr, keep2 = 0, 0, None, None
+    if global_stats is not None:
+        g_dwf, g_ttf, dwt = global_stats
+        keep2 = np.ascontiguousarray(dwt, dtype=np.uint64)
+        g_dwt_ptr = keep2.ctypes.data_as(PU64)
+    bcnt = np.zeros(nbuckets, dtype=np.int64)
+    bsum = np.zeros(nbuckets, dtype=np.int64)
+    hits = (OScoreDoc * k)()
+    out_count = C.c_uint32(0)
+    total = C.c_uint64(0)
+    PI64 = C.POINTER(C.c_int64)
+    rc = lib().o_execute_topk_hybrid(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)),
+        ti.ctypes.data_as(PU32), bo.ctypes.data_as(C.POINTER(C.c_float)),
+        C.c_uint32(len(ti)), C.c_uint32(min_match), C.c_float(k1),
+        C.c_float(b), C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf),
+        C.c_uint32(k), col.ctypes.data_as(PI64), C.c_int64(flo),
+        C.c_int64(fhi), C.c_uint32(nbuckets),
+        bcnt.ctypes.data_as(PI64), bsum.ctypes.data_as(PI64),
+        hits, C.byref(out_count), C.byref(total))
+    assert rc == 0, rc
+    del keep2
+    return _hits_to_np(hits, out_count.value), total.value, bcnt, bsum

@@ -528,10 +528,22 @@ static void o_cand_push(OCandVec* cv, float score, uint32_t doc,
  * local to the segment). Cursors must be positioned before range_lo.
  * Exactly one of coll / cands is non-NULL.
  * Returns the number of matching docs in the range. */
+/* optional hybrid context (TableFilterDocIterator semantics,
+ * index/table_filter_iterator.hpp:104-312: survivors of the pushed column
+ * predicate; plus the analytics consumer's COUNT/SUM per bucket) */
+typedef struct {
+  const int64_t* col; /* indexed by doc (1-based), NULL = no filter */
+  int64_t lo, hi;     /* BETWEEN, inclusive */
+  uint32_t nbuckets;  /* bucket = (col-lo)*nbuckets/(hi-lo+1) */
+  int64_t* bucket_count;
+  int64_t* bucket_sum;
+} OHybrid;
+
 static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
                              const uint32_t* norms, uint32_t min_match,
                              uint32_t range_lo, uint32_t range_hi,
-                             OCollector* coll, OCandVec* cands, uint32_t seg) {
+                             OCollector* coll, OCandVec* cands, uint32_t seg,
+                             const OHybrid* hy) {
   float score_win[O_WINDOW];
   uint8_t cnt_win[O_WINDOW];
   uint64_t mask[O_WINDOW / 64];
@@ -578,16 +590,30 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
       uint64_t word = mask[w];
       if (!word) continue;
       matches += (uint64_t)__builtin_popcountll(word);
-      const uint32_t base = lo + w * 64;
+      const uint32_t base = lo + w * 64;  /* hy filter decrements below */
       while (word) {
         const uint32_t bit = (uint32_t)__builtin_ctzll(word);
         word &= word - 1;
+        const uint32_t doc = base + bit;
+        if (hy && hy->col) {
+          const int64_t v = hy->col[doc];
+          if (v < hy->lo || v > hy->hi) {
+            --matches; /* popcount above pre-counted this doc */
+            continue;
+          }
+          const uint64_t span = (uint64_t)(hy->hi - hy->lo) + 1;
+          uint32_t bkt =
+            (uint32_t)(((uint64_t)(v - hy->lo) * hy->nbuckets) / span);
+          if (bkt >= hy->nbuckets) bkt = hy->nbuckets - 1;
+          hy->bucket_count[bkt] += 1;
+          hy->bucket_sum[bkt] += v;
+        }
         const float s = score_win[w * 64 + bit];
         if (coll) {
           coll->count++;
-          o_coll_try(coll, s, base + bit);
+          o_coll_try(coll, s, doc);
         } else {
-          o_cand_push(cands, s, base + bit, seg);
+          o_cand_push(cands, s, doc, seg);
         }
       }
     }
@@ -694,11 +720,66 @@ int o_execute_topk(const OSegBlob* segs, uint32_t nsegs,
                         g_ttf, cur);
     if (rc) { free(cur); free(cands.v); free(auto_dwt); return rc; }
     matches += o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1,
-                            1, v.hdr->doc_count, NULL, &cands, s);
+                            1, v.hdr->doc_count, NULL, &cands, s, NULL);
   }
   free(cur);
   free(auto_dwt);
   /* threshold semantics: accept score > FLT_MIN (doc_collector.hpp:58) */
+  uint64_t n = 0;
+  const float kFltMin = 1.17549435e-38f;
+  for (uint64_t i = 0; i < cands.n; ++i)
+    if (cands.v[i].score > kFltMin) cands.v[n++] = cands.v[i];
+  qsort(cands.v, n, sizeof(OScoreDoc), o_sd_cmp);
+  const uint32_t kk = n < k ? (uint32_t)n : k;
+  memcpy(hits, cands.v, sizeof(OScoreDoc) * kk);
+  *out_count = kk;
+  *total_matches = matches;
+  free(cands.v);
+  return 0;
+}
+
+/* HYBRID exact top-k: BM25 match AND col BETWEEN [lo,hi], plus per-bucket
+ * COUNT/SUM over the surviving matches (BASELINE configs[3]; reference
+ * semantics: MaybeWrapColFilter + TableFilterDocIterator::Collect,
+ * duckdb_search_full_scan.cpp:1900-1912, table_filter_iterator.hpp:229-312,
+ * aggregate consumer external). col indexed by LOCAL doc id of the single
+ * provided segment. */
+int o_execute_topk_hybrid(const void* blob, uint64_t size,
+                          const uint32_t* term_idx, const float* boosts,
+                          uint32_t nterms, uint32_t min_match, float k1,
+                          float b, uint64_t g_dwf, const uint64_t* g_dwt,
+                          uint64_t g_ttf, uint32_t k, const int64_t* col,
+                          int64_t flo, int64_t fhi, uint32_t nbuckets,
+                          int64_t* bucket_count, int64_t* bucket_sum,
+                          OScoreDoc* hits, uint32_t* out_count,
+                          uint64_t* total_matches) {
+  SdbSegmentView v;
+  int rc = o_segment_parse(blob, size, &v);
+  if (rc) return rc;
+  OSegBlob sb = {blob, size};
+  uint64_t auto_dwf = 0, auto_ttf = 0;
+  uint64_t* auto_dwt = (uint64_t*)malloc(sizeof(uint64_t) * nterms);
+  if (g_dwf == 0) {
+    rc = o_global_stats(&sb, 1, term_idx, nterms, &auto_dwf, &auto_ttf,
+                        auto_dwt);
+    if (rc) { free(auto_dwt); return rc; }
+    g_dwf = auto_dwf;
+    g_ttf = auto_ttf;
+    g_dwt = auto_dwt;
+  }
+  OCursor* cur = (OCursor*)malloc(sizeof(OCursor) * nterms);
+  rc = o_prep_cursors(&v, term_idx, boosts, nterms, k1, b, g_dwf, g_dwt,
+                      g_ttf, cur);
+  if (rc) { free(cur); free(auto_dwt); return rc; }
+  memset(bucket_count, 0, sizeof(int64_t) * nbuckets);
+  memset(bucket_sum, 0, sizeof(int64_t) * nbuckets);
+  OHybrid hy = {col, flo, fhi, nbuckets, bucket_count, bucket_sum};
+  OCandVec cands = {0, 0, 0};
+  uint64_t matches =
+    o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1, 1,
+                 v.hdr->doc_count, NULL, &cands, 0, &hy);
+  free(cur);
+  free(auto_dwt);
   uint64_t n = 0;
   const float kFltMin = 1.17549435e-38f;
   for (uint64_t i = 0; i < cands.n; ++i)
@@ -747,7 +828,7 @@ int o_execute_topk_mech(const OSegBlob* segs, uint32_t nsegs,
     if (rc) { free(cur); free(auto_dwt); return rc; }
     coll.seg = s;
     o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1, 1,
-                 v.hdr->doc_count, &coll, NULL, s);
+                 v.hdr->doc_count, &coll, NULL, s, NULL);
   }
   free(cur);
   free(auto_dwt);
@@ -828,7 +909,7 @@ static void* o_mt_worker(void* argp) {
     if (g > threshold) threshold = g;
     matches += o_exec_range(cur, a->nterms, a->v->norms,
                             a->min_match ? a->min_match : 1, lo, hi, &coll,
-                            NULL, 0);
+                            NULL, 0, NULL);
     o_atomic_thresh_max(a->g_thresh_bits, threshold);
   }
   free(cur);

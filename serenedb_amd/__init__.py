@@ -229,6 +229,7 @@ def gpu():
             "sdb_gpu_segment_load", "sdb_gpu_segment_free",
             "sdb_gpu_execute_topk", "sdb_gpu_decode_term",
             "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
+            "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
         ):
             getattr(lib, f).restype = C.c_int
         _gpu = lib
@@ -268,12 +269,7 @@ class GpuContext:
         self._segments.append(seg)
         return seg
 
-    def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
-                     b=0.75, global_stats=None):
-        """global_stats: optional (docs_with_field, total_term_freq,
-        [docs_with_term per term]) for sharded execution."""
-        import numpy as np
-
+    def _make_plan(self, term_idx, boosts, min_match, k1, b, global_stats):
         class _Plan(C.Structure):
             _fields_ = [
                 ("terms", C.POINTER(SdbTermRef)),
@@ -289,12 +285,24 @@ class GpuContext:
         terms = (SdbTermRef * len(term_idx))(
             *[SdbTermRef(t, float(bo)) for t, bo in zip(term_idx, boosts)])
         plan = _Plan(terms, len(term_idx), min_match, k1, b, 0, 0, None)
+        plan._keep = terms
         if global_stats is not None:
             dwf, ttf, dwt = global_stats
             dwt_arr = (C.c_uint64 * len(dwt))(*[int(x) for x in dwt])
             plan.g_docs_with_field = int(dwf)
             plan.g_total_term_freq = int(ttf)
             plan.g_docs_with_term = dwt_arr
+            plan._keep2 = dwt_arr
+        return plan
+
+    def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
+                     b=0.75, global_stats=None):
+        """global_stats: optional (docs_with_field, total_term_freq,
+        [docs_with_term per term]) for sharded execution."""
+        import numpy as np
+
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b,
+                               global_stats)
         seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value) for s in segs])
         hits = (SdbScoreDoc * k)()
         out_count = C.c_uint32(0)
@@ -309,6 +317,43 @@ class GpuContext:
         res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
                             dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value
+
+    def attach_column(self, seg, col):
+        import numpy as np
+
+        col = np.ascontiguousarray(col, dtype=np.int64)
+        rc = self._lib.sdb_gpu_segment_attach_column(
+            self._ctx, seg, col.ctypes.data_as(C.POINTER(C.c_int64)))
+        if rc != 0:
+            raise RuntimeError(f"attach_column rc={rc}")
+
+    def execute_topk_hybrid(self, segs, term_idx, boosts, k, flo, fhi,
+                            nbuckets, min_match=1, k1=1.2, b=0.75,
+                            global_stats=None):
+        import numpy as np
+
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b,
+                               global_stats)
+        seg_arr = (C.c_void_p * len(segs))(
+            *[C.c_void_p(s.value) for s in segs])
+        hits = (SdbScoreDoc * k)()
+        out_count = C.c_uint32(0)
+        total = C.c_uint64(0)
+        bcnt = np.zeros(nbuckets, dtype=np.int64)
+        bsum = np.zeros(nbuckets, dtype=np.int64)
+        PI64 = C.POINTER(C.c_int64)
+        rc = self._lib.sdb_gpu_execute_topk_hybrid(
+            self._ctx, seg_arr, len(segs), C.byref(plan), C.c_uint32(k),
+            C.c_int64(flo), C.c_int64(fhi), C.c_uint32(nbuckets),
+            bcnt.ctypes.data_as(PI64), bsum.ctypes.data_as(PI64),
+            hits, C.byref(out_count), C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_topk_hybrid rc={rc}")
+        n = out_count.value
+        dt = np.dtype([("score", "f4"), ("doc", "u4"), ("segment", "u4")])
+        res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
+                            dtype=dt).copy() if n else np.zeros(0, dtype=dt)
+        return res, total.value, bcnt, bsum
 
     def decode_term(self, seg, term_idx, df):
         import numpy as np

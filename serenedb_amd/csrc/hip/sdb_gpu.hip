@@ -55,6 +55,7 @@
 #define SDB_MAX_TERMS 32u
 #define SDB_HIST_BINS 256u
 #define SDB_DESC_CACHE 32u  // staged descriptors per term per window
+#define SDB_MAX_BUCKETS 128u  // hybrid group-by buckets
 #define SDB_CAND_CAP (64u * 1024u * 1024u)  // 64M candidates (768 MB)
 
 #define HIP_CHECK(x)                        \
@@ -71,6 +72,7 @@ struct SdbGpuSegment {
   SdbBlockDesc* desc;   // device
   uint8_t* payload;     // device
   uint32_t* norms;      // device, doc_count+1
+  long long* fcol;      // device, doc_count+1 (hybrid filter column) or null
   SdbTermEntry* terms_host;  // host copy of term table
   SdbSegHeader hdr;     // host copy
 };
@@ -312,6 +314,13 @@ struct WindowArgs {
   uint32_t cand_cap;
   unsigned long long* total_matches;
   uint32_t* overflow;
+  // hybrid (BASELINE configs[3]): column BETWEEN filter over matches +
+  // per-bucket COUNT/SUM (TableFilterDocIterator semantics,
+  // index/table_filter_iterator.hpp:104-312)
+  const long long* fcol;  // device, indexed by doc; NULL = no filter
+  long long flo, fhi;     // inclusive
+  uint32_t nbuckets;
+  unsigned long long* bucket_out;  // [2*nbuckets]: count, sum (i64 bits)
 };
 
 // Persistent-range window kernel: each workgroup owns a CONTIGUOUS range of
@@ -333,7 +342,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* cursors = shared_misc + 2 + SDB_NWAVES; // per-term block cursor
   // staged descriptors: SDB_DESC_CACHE per term, one coalesced load per
   // window removes a ~900-cycle dependent desc load from every block chain
-  SdbBlockDesc* dcache = (SdbBlockDesc*)(cursors + SDB_MAX_TERMS);
+  unsigned long long* lbuck =
+    (unsigned long long*)(cursors + SDB_MAX_TERMS);  // 2*SDB_MAX_BUCKETS
+  SdbBlockDesc* dcache = (SdbBlockDesc*)(lbuck + 2 * SDB_MAX_BUCKETS);
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
@@ -371,6 +382,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (uint32_t i = tid; i < SDB_WIN_DOCS / 4; i += SDB_NTHREADS)
       ((uint32_t*)cwin)[i] = 0;
     for (uint32_t i = tid; i < SDB_HIST_BINS; i += SDB_NTHREADS) hist[i] = 0;
+    if (a.fcol)
+      for (uint32_t i = tid; i < 2 * a.nbuckets; i += SDB_NTHREADS)
+        lbuck[i] = 0;
     // stage this window's descriptors: term t's next SDB_DESC_CACHE descs
     // from its cursor, as coalesced u32 reads (7 words per desc)
     {
@@ -470,10 +484,26 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     __syncthreads();  // cursor updates visible before next window stages
     continue;
 #else
-    // histogram of matching scores + local match count
+    // histogram of matching scores + local match count (hybrid: apply the
+    // column BETWEEN filter here; filtered docs are unmarked so the append
+    // pass skips them)
     uint32_t my_matches = 0;
     for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
       if (cwin[off] >= mm) {
+        if (a.fcol) {
+          const long long vv = a.fcol[lo + off];
+          if (vv < a.flo || vv > a.fhi) {
+            cwin[off] = 0;
+            continue;
+          }
+          const unsigned long long span =
+            (unsigned long long)(a.fhi - a.flo) + 1ull;
+          uint32_t bkt = (uint32_t)(
+            ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
+          if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
+          atomicAdd(&lbuck[2 * bkt], 1ull);
+          atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
+        }
         ++my_matches;
         uint32_t bin = (uint32_t)(swin[off] * inv_smax);
         if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
@@ -485,6 +515,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
     if (lane == 0) shared_misc[2 + wave] = wm;
     __syncthreads();  // hist + per-wave match counts complete
+    if (a.fcol)
+      for (uint32_t i = tid; i < 2 * a.nbuckets; i += SDB_NTHREADS)
+        if (lbuck[i]) atomicAdd(&a.bucket_out[i], lbuck[i]);
 
     // merge window histogram into the per-XCD global shard (skip bins below
     // the published threshold bin: they cannot change any suffix count at or
@@ -660,6 +693,7 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   HIP_CHECK(hipMalloc(&ctx->d_total_matches, 8));
   HIP_CHECK(hipMalloc(&ctx->d_gthresh, 4));
   HIP_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 8));
+  HIP_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
   HIP_CHECK(hipMalloc(&ctx->d_overflow, 4));
   HIP_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
   HIP_CHECK(hipHostMalloc(&ctx->h_counts, 8));
@@ -677,6 +711,7 @@ int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   hipFree(ctx->d_total_matches);
   hipFree(ctx->d_gthresh);
   hipFree(ctx->d_ghist);
+  hipFree(ctx->d_buckets);
   hipFree(ctx->d_overflow);
   hipFree(ctx->d_terms);
   hipHostFree(ctx->h_counts);
@@ -721,18 +756,25 @@ int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
   hipFree(seg->desc);
   hipFree(seg->payload);
   hipFree(seg->norms);
+  if (seg->fcol) hipFree(seg->fcol);
   std::free(seg->terms_host);
   delete seg;
   return SDB_OK;
 }
 
-int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
-                         uint32_t nsegs, const SdbQueryPlan* plan, uint32_t k,
-                         SdbScoreDoc* hits, uint32_t* out_count,
-                         uint64_t* total_matches) {
+static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                          uint32_t nsegs, const SdbQueryPlan* plan,
+                          uint32_t k, int hybrid, long long h_flo,
+                          long long h_fhi, uint32_t h_nbuckets,
+                          int64_t* bucket_count, int64_t* bucket_sum,
+                          SdbScoreDoc* hits, uint32_t* out_count,
+                          uint64_t* total_matches) {
   if (!ctx || !segs || !plan || !hits || !out_count || !total_matches ||
       plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS || k == 0)
     return SDB_ERR_INVALID;
+  if (hybrid)
+    for (uint32_t s = 0; s < nsegs; ++s)
+      if (!segs[s]->fcol) return SDB_ERR_INVALID;
 
   // ---- PreparePhase analogue: global stats (double -> f32, bm25.cpp) ----
   uint64_t g_dwf = plan->g_docs_with_field;
@@ -787,10 +829,14 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_gthresh, 0, 4, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_ghist, 0, 4 * SDB_HIST_BINS * 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
+  if (hybrid)
+    HIP_CHECK(hipMemsetAsync(ctx->d_buckets, 0, 8 * 2 * h_nbuckets,
+                             ctx->stream));
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
                            (2 + SDB_NWAVES + SDB_MAX_TERMS) * 4 +
+                           8 * 2 * SDB_MAX_BUCKETS +
                            sizeof(SdbBlockDesc) * SDB_DESC_CACHE *
                              plan->nterms;
 
@@ -812,6 +858,11 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                              sizeof(TermDev) * plan->nterms,
                              hipMemcpyHostToDevice, ctx->stream));
     WindowArgs a{};
+    a.fcol = hybrid ? seg->fcol : nullptr;
+    a.flo = h_flo;
+    a.fhi = h_fhi;
+    a.nbuckets = h_nbuckets;
+    a.bucket_out = ctx->d_buckets;
     a.desc = seg->desc;
     a.payload = seg->payload;
     a.norms = seg->norms;
@@ -902,12 +953,59 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   std::copy(cands.begin(), cands.begin() + kk, hits);
   *out_count = (uint32_t)kk;
   *total_matches = *ctx->h_matches;
+  if (hybrid && bucket_count && bucket_sum) {
+    std::vector<unsigned long long> hb(2 * h_nbuckets);
+    HIP_CHECK(hipMemcpy(hb.data(), ctx->d_buckets, 8ull * 2 * h_nbuckets,
+                        hipMemcpyDeviceToHost));
+    for (uint32_t i = 0; i < h_nbuckets; ++i) {
+      bucket_count[i] = (int64_t)hb[2 * i];
+      bucket_sum[i] = (int64_t)hb[2 * i + 1];
+    }
+  }
   ctx->last_gtau = gtau_final;
   ctx->last_select_ms =
     std::chrono::duration<double, std::milli>(
       std::chrono::steady_clock::now() - t_sel0)
       .count();
   return SDB_OK;
+}
+
+int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                         uint32_t nsegs, const SdbQueryPlan* plan, uint32_t k,
+                         SdbScoreDoc* hits, uint32_t* out_count,
+                         uint64_t* total_matches) {
+  return exec_topk_impl(ctx, segs, nsegs, plan, k, 0, 0, 0, 0, nullptr,
+                        nullptr, hits, out_count, total_matches);
+}
+
+// Attach the hybrid filter column (i64[doc_count+1], index 0 unused) to a
+// resident segment — the analytics column the reference's MaybeWrapColFilter
+// pushes into the scan (duckdb_search_full_scan.cpp:1900-1912).
+int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                  const int64_t* data) {
+  if (!ctx || !seg || !data) return SDB_ERR_INVALID;
+  if (!seg->fcol)
+    HIP_CHECK(
+      hipMalloc(&seg->fcol, 8ull * ((uint64_t)seg->hdr.doc_count + 1)));
+  HIP_CHECK(hipMemcpy(seg->fcol, data,
+                      8ull * ((uint64_t)seg->hdr.doc_count + 1),
+                      hipMemcpyHostToDevice));
+  return SDB_OK;
+}
+
+// Hybrid: BM25 top-k AND col BETWEEN [flo,fhi] + per-bucket COUNT/SUM over
+// the surviving matches (BASELINE configs[3]).
+int sdb_gpu_execute_topk_hybrid(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                                uint32_t nsegs, const SdbQueryPlan* plan,
+                                uint32_t k, int64_t flo, int64_t fhi,
+                                uint32_t nbuckets, int64_t* bucket_count,
+                                int64_t* bucket_sum, SdbScoreDoc* hits,
+                                uint32_t* out_count,
+                                uint64_t* total_matches) {
+  if (nbuckets == 0 || nbuckets > SDB_MAX_BUCKETS) return SDB_ERR_INVALID;
+  return exec_topk_impl(ctx, segs, nsegs, plan, k, 1, flo, fhi, nbuckets,
+                        bucket_count, bucket_sum, hits, out_count,
+                        total_matches);
 }
 
 int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,

@@ -21,6 +21,7 @@ struct SdbGpuCtx {
   unsigned long long* d_total_matches;
   uint32_t* d_gthresh;  // float bits
   uint32_t* d_ghist;    // global 256-bin score histogram (threshold tightening)
+  unsigned long long* d_buckets;  // hybrid bucket aggregates [2*128]
   TermDev* d_terms;
   uint32_t* d_overflow;
   uint32_t* h_counts;  // pinned: [cand_count, overflow]

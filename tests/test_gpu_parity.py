@@ -203,3 +203,29 @@ def test_smoke_entry():
     import __graft_entry__ as ge
 
     ge.smoke()
+
+
+def test_hybrid_parity(ctx):
+    """BM25 top-k AND column BETWEEN + bucket aggregates vs oracle
+    (BASELINE configs[3] semantics)."""
+    doc_count = 400_000
+    sels = [0.1, 0.05, 0.02, 0.01]
+    seed = 52
+    blob, _, _ = make_corpus(seed, doc_count, sels)
+    rng = np.random.default_rng(45)
+    col = rng.integers(0, 1 << 31, doc_count + 1).astype(np.int64)
+    span = 1 << 31
+    flo, fhi = int(span * 0.4), int(span * 0.6) - 1
+    nbuckets = 64
+    seg = ctx.load_segment(blob)
+    ctx.attach_column(seg, col)
+    hits, total, bcnt, bsum = ctx.execute_topk_hybrid(
+        [seg], [0, 1, 2, 3], [1.0] * 4, 1000, flo, fhi, nbuckets)
+    ohits, ototal, obcnt, obsum = po.execute_topk_hybrid(
+        blob, [0, 1, 2, 3], [1.0] * 4, 1000, col, flo, fhi, nbuckets)
+    assert total == ototal
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+    np.testing.assert_array_equal(
+        hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
+    np.testing.assert_array_equal(bcnt, obcnt)
+    np.testing.assert_array_equal(bsum, obsum)

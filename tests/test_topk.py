@@ -258,3 +258,39 @@ def test_multi_segment_stats_merge():
     got = [int(h["doc"]) + (n1 if h["segment"] == 1 else 0) for h in hits]
     np.testing.assert_array_equal(got, fhits["doc"])
     np.testing.assert_array_equal(hits["score"], fhits["score"])
+
+
+def test_hybrid_vs_brute():
+    """BM25 top-k AND col BETWEEN + per-bucket COUNT/SUM (configs[3]
+    semantics: TableFilterDocIterator + aggregate consumer)."""
+    doc_count = 30_000
+    sels = [0.1, 0.05, 0.02]
+    seed = 51
+    blob, postings, norms = synth_corpus(seed, doc_count, sels)
+    rng = np.random.default_rng(45)
+    col = rng.integers(0, 1 << 31, doc_count + 1).astype(np.int64)
+    span = 1 << 31
+    flo, fhi = int(span * 0.4), int(span * 0.6) - 1  # 20% selectivity
+    nbuckets = 64
+    k = 100
+    hits, total, bcnt, bsum = po.execute_topk_hybrid(
+        blob, [0, 1, 2], [1.0] * 3, k, col, flo, fhi, nbuckets)
+    # brute force
+    order, scores, _ = brute_topk(postings, norms, doc_count, sels, 10**9)
+    match = np.zeros(doc_count + 1, dtype=bool)
+    for docs, _f in postings:
+        match[docs] = True
+    docs_all = np.nonzero(match)[0]
+    passing = docs_all[(col[docs_all] >= flo) & (col[docs_all] <= fhi)]
+    assert total == len(passing)
+    exp_order = sorted(passing, key=lambda d: (-float(scores[d]), d))[:k]
+    assert [int(h["doc"]) for h in hits] == [int(d) for d in exp_order]
+    # buckets
+    w = (fhi - flo + 1)
+    bks = ((col[passing] - flo) * nbuckets // w).astype(np.int64)
+    bks = np.minimum(bks, nbuckets - 1)
+    exp_cnt = np.bincount(bks, minlength=nbuckets)
+    exp_sum = np.bincount(bks, weights=col[passing].astype(np.float64),
+                          minlength=nbuckets).astype(np.int64)
+    np.testing.assert_array_equal(bcnt, exp_cnt)
+    np.testing.assert_array_equal(bsum, exp_sum)
