@@ -56,7 +56,7 @@ def unpack_hits(packed):
     return scores, docs
 
 
-def bench_bm25(args):
+def bench_bm25(args, hybrid=False):
     import serenedb_amd as sa
 
     rank, world = env_rank()
@@ -86,6 +86,16 @@ def bench_bm25(args):
     device = int(os.environ.get("LOCAL_RANK", 0))
     ctx = sa.GpuContext(device)
     seg = ctx.load_segment(blob)
+
+    # hybrid (configs[3]): i64 filter column over this shard's docs,
+    # BETWEEN at 20% selectivity, 64-bucket COUNT/SUM group-by (seed 45)
+    nbuckets = 64
+    span = 1 << 31
+    flo, fhi = int(span * 0.4), int(span * 0.6) - 1
+    if hybrid:
+        crng = np.random.default_rng(45 + rank)
+        col = crng.integers(0, span, (hi - lo + 1) + 1).astype(np.int64)
+        ctx.attach_column(seg, col)
 
     # ---- global stats (PreparePhase analogue over RCCL) ----
     import ctypes as CT
@@ -145,8 +155,18 @@ def bench_bm25(args):
     lib.sdb_gpu_last_kernel_ms.restype = CT.c_int
 
     def step():
-        hits, total = ctx.execute_topk([seg], term_idx, boosts, k,
-                                       global_stats=gstats)
+        if hybrid:
+            hits, total, bcnt, bsum = ctx.execute_topk_hybrid(
+                [seg], term_idx, boosts, k, flo, fhi, nbuckets,
+                global_stats=gstats)
+            if dist:
+                import torch
+                bb = torch.from_numpy(np.concatenate([bcnt, bsum])).cuda()
+                dist.all_reduce(bb)  # partial-agg merge
+                _ = bb.cpu().numpy()
+        else:
+            hits, total = ctx.execute_topk([seg], term_idx, boosts, k,
+                                           global_stats=gstats)
         if dist:
             import torch
             packed = torch.from_numpy(pack_hits(hits, lo - 1, k)).cuda()
@@ -230,24 +250,37 @@ def bench_bm25(args):
         sample_postings = sum(st[t].df for t in range(nterms))
         ncores = os.cpu_count() or 1
         iters = 0
+        scol = None
+        if hybrid:
+            crng = np.random.default_rng(45)
+            scol = crng.integers(0, span, sample_docs + 1).astype(np.int64)
         tcpu = time.time()
         while time.time() - tcpu < args.cpu_seconds and iters < 200:
-            po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
-                               global_stats=gstats)
+            if hybrid:
+                po.execute_topk_hybrid(sblob, term_idx, boosts, k, scol,
+                                       flo, fhi, nbuckets,
+                                       global_stats=gstats)
+            else:
+                po.execute_topk_mt(sblob, term_idx, boosts, k,
+                                   nthreads=ncores, global_stats=gstats)
             iters += 1
         tcpu = time.time() - tcpu
         cpu_baseline = {
             "value": round(sample_postings * iters / tcpu, 1),
             "unit": "postings scored/s",
-            "cores": ncores,
+            "cores": 1 if hybrid else ncores,
             "kind": "port",
             "sample": f"{sample_docs/1e6:.1f}M-doc shard of the same corpus "
                       f"({sample_postings} postings/query, {iters} iters, "
-                      f"{tcpu:.1f}s; oracle RunTopKScan restatement)",
+                      f"{tcpu:.1f}s; oracle "
+                      + ("hybrid exact path, single-thread"
+                         if hybrid else "RunTopKScan restatement")
+                      + ")",
         }
 
     result = {
-        "metric": "docs scored/sec BM25 top-1000",
+        "metric": ("docs scored/sec BM25 top-1000 + range filter-agg"
+                   if hybrid else "docs scored/sec BM25 top-1000"),
         "value": round(value, 1),
         "unit": "postings/s",
         "n_gpus": world,
@@ -260,8 +293,10 @@ def bench_bm25(args):
         "dtype": "f32",
         "data": "synthetic",
         "config": {
-            "workload": "bm25_top1000_4term_or_100M" if doc_count == 100_000_000
-                        else f"bm25_top1000_4term_or_{doc_count}",
+            "workload": (("hybrid_" if hybrid else "") +
+                         ("bm25_top1000_4term_or_100M"
+                          if doc_count == 100_000_000
+                          else f"bm25_top1000_4term_or_{doc_count}")),
             "doc_count": doc_count,
             "selectivities": sels,
             "k": k,
@@ -438,14 +473,16 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--workload", default="bm25_topk",
-                    choices=["bm25_topk", "scan_agg"])
+                    choices=["bm25_topk", "scan_agg", "hybrid"])
     ap.add_argument("--docs", type=int, default=100_000_000)
     ap.add_argument("--rows", type=int, default=1_000_000_000)
     ap.add_argument("--cpu-seconds", type=float, default=10.0)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
     if args.workload == "bm25_topk":
-        bench_bm25(args)
+        bench_bm25(args, hybrid=False)
+    elif args.workload == "hybrid":
+        bench_bm25(args, hybrid=True)
     else:
         bench_scan(args)
 
