@@ -426,6 +426,24 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                  ? dcache[t * SDB_DESC_CACHE + rel]
                                  : a.desc[b];
         if (d.prev_doc >= hi) break;  // first doc > hi
+        // prefetch the NEXT block's payload (256 B per wave) so its decode
+        // loads hit L1: the per-block chain was serial cold loads
+        // (desc -> docs -> freqs -> norms, ~900 cy each)
+        {
+          const uint64_t nb = b + SDB_NWAVES;
+          if (nb < dend) {
+            const uint32_t nrel = (uint32_t)(nb - te.desc_begin) - cur0;
+            const SdbBlockDesc dn = nrel < SDB_DESC_CACHE
+                                      ? dcache[t * SDB_DESC_CACHE + nrel]
+                                      : a.desc[nb];
+            if (dn.prev_doc < hi) {
+              const uint32_t* pfp = (const uint32_t*)(
+                (uintptr_t)(pl + dn.doc_off) & ~(uintptr_t)3);
+              const uint32_t pf = pfp[lane];  // 256 B line-touch
+              asm volatile("" ::"v"(pf));
+            }
+          }
+        }
 #ifdef SDB_ABLATE_DECODE
         for (uint32_t j = lane; j < d.len; j += 64) {
           dbuf[j] = lo + ((uint32_t)(b * 131u) + j * 7u) % (hi - lo + 1u);
@@ -736,7 +754,7 @@ int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
   std::memcpy(seg->terms_host, base + hdr.off_terms,
               sizeof(SdbTermEntry) * hdr.nterms);
   HIP_CHECK(hipMalloc(&seg->desc, sizeof(SdbBlockDesc) * hdr.total_blocks + 16));
-  HIP_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 64));
+  HIP_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 512));
   HIP_CHECK(
     hipMalloc(&seg->norms, sizeof(uint32_t) * ((size_t)hdr.doc_count + 1)));
   HIP_CHECK(hipMemcpy(seg->desc, base + hdr.off_desc,
