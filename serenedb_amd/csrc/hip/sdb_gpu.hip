@@ -268,6 +268,64 @@ __device__ void decode_freq_block_wave(const uint8_t* p, uint32_t len,
   }
 }
 
+// Fused fast path for the dominant block shape (delta-bitpack docs +
+// bitpack freqs + bitpack norms, always full 128-doc blocks): issue every
+// stream's packed-word loads up front (one memory round trip instead of
+// three serialized ones), keep the two values per lane in registers, score
+// directly — no LDS scratch round trip. Returns false for any other family
+// combination (caller falls back to the generic per-stream decode).
+__device__ __forceinline__ bool try_block_fused(
+  const uint8_t* pl, const SdbBlockDesc& d, int lane, uint32_t norm_stream,
+  uint32_t lo, uint32_t hi, float num, float nc, float nl,
+  const uint32_t* norms_col, float* swin, uint8_t* cwin) {
+  const uint8_t* db = pl + d.doc_off;
+  const uint8_t* fb = pl + d.freq_off;
+  const uint8_t* nb = fb + d.flags;
+  const uint32_t dtag = db[0], ftag = fb[0];
+  const uint32_t ntag = norm_stream ? nb[0] : SDB_E_BITPACK_01;
+  if (dtag < SDB_DE_DELTA_BITPACK_02 || ftag < SDB_E_BITPACK_01 ||
+      ntag < SDB_E_BITPACK_01)
+    return false;
+  const uint32_t dbits = dtag - SDB_DE_DELTA_BITPACK_02 + 2;
+  const uint32_t fbits = ftag - SDB_E_BITPACK_01 + 1;
+  const uint32_t nbits = ntag - SDB_E_BITPACK_01 + 1;
+  const uint32_t i0 = 2u * lane, i1 = i0 + 1;
+  // all loads issue here, before any cross-lane dependency
+  const uint32_t dd0 = extract_packed(db + 1, dbits, i0);
+  const uint32_t dd1 = extract_packed(db + 1, dbits, i1);
+  const uint32_t f0 = extract_packed(fb + 1, fbits, i0);
+  const uint32_t f1 = extract_packed(fb + 1, fbits, i1);
+  uint32_t n0 = 0, n1 = 0;
+  if (norm_stream) {
+    n0 = extract_packed(nb + 1, nbits, i0);
+    n1 = extract_packed(nb + 1, nbits, i1);
+  }
+  const uint32_t pair = dd0 + dd1;
+  const uint32_t incl = wave_incl_scan(pair, lane);
+  const uint32_t excl = incl - pair;
+  const uint32_t doc0 = d.prev_doc + excl + dd0;
+  const uint32_t doc1 = d.prev_doc + excl + pair;
+  if (!norm_stream) {
+    n0 = doc0 >= lo && doc0 <= hi ? norms_col[doc0] : 1u;
+    n1 = doc1 >= lo && doc1 <= hi ? norms_col[doc1] : 1u;
+  }
+  if (doc0 >= lo && doc0 <= hi) {
+    const float c1 = nc + nl * (float)n0;
+    const float s = num - num * c1 / (c1 + (float)f0);
+    const uint32_t off = doc0 - lo;
+    swin[off] += s;
+    cwin[off] = (uint8_t)(cwin[off] + 1u);
+  }
+  if (doc1 >= lo && doc1 <= hi) {
+    const float c1 = nc + nl * (float)n1;
+    const float s = num - num * c1 / (c1 + (float)f1);
+    const uint32_t off = doc1 - lo;
+    swin[off] += s;
+    cwin[off] = (uint8_t)(cwin[off] + 1u);
+  }
+  return true;
+}
+
 // binary searches over the descriptor span of one term:
 // first block with last_doc >= lo  /  first block with prev_doc >= hi
 __device__ __forceinline__ uint64_t lower_bound_last_doc(
@@ -451,6 +509,11 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           nbuf[j] = 100u + j;
         }
 #else
+#ifndef SDB_ABLATE_SCORE
+        if (try_block_fused(pl, d, lane, a.norm_stream, lo, hi, num, nc, nl,
+                            a.norms, swin, cwin))
+          continue;
+#endif
         decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
         decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
         if (a.norm_stream)  // v2: norm block follows the freq block
