@@ -411,6 +411,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   const float inv_smax = (float)SDB_HIST_BINS / a.smax;
   uint32_t* gh = a.ghist + (blockIdx.x & 7u) * SDB_HIST_BINS;
 
+  unsigned long long wg_matches = 0;  // summed on tid 0, flushed once
+
   // contiguous window range of this workgroup
   const uint32_t nwin = (a.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
   const uint32_t per = (nwin + gridDim.x - 1) / gridDim.x;
@@ -617,14 +619,16 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     __syncthreads();
     if (wave == 0) {
       uint32_t part = 0;
+      if (SDB_HIST_BINS - 1 - 4 * (uint32_t)lane + 3 >= known_bin) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
+        for (int j = 0; j < 4; ++j) {
+          const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
 #pragma unroll
-        for (int s = 0; s < 8; ++s)
-          part += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
-                                    __ATOMIC_RELAXED,
-                                    __HIP_MEMORY_SCOPE_AGENT);
+          for (int s = 0; s < 8; ++s)
+            part += __hip_atomic_load(&a.ghist[s * SDB_HIST_BINS + b],
+                                      __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+        }
       }
       const uint32_t suff_incl = wave_incl_scan(part, lane);
       const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
@@ -647,7 +651,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           }
         }
         const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
-        if (tau > 0.0f) {
+        if (tau > 0.0f && binfloor > known_bin) {
           uint32_t bits;
           __builtin_memcpy(&bits, &tau, 4);
           atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
@@ -657,8 +661,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         uint32_t total_m = 0;
         for (uint32_t v = 0; v < SDB_NWAVES; ++v)
           total_m += shared_misc[2 + v];
-        if (total_m) atomicAdd(a.total_matches, (unsigned long long)total_m);
-        shared_misc[0] = atomicOr(a.gthresh, 0u);
+        wg_matches += total_m;  // ONE global atomic at kernel end
+        shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
       }
     }
     __syncthreads();
@@ -708,6 +713,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     __syncthreads();  // window state reused next iteration
 #endif
   }
+  if (tid == 0 && wg_matches) atomicAdd(a.total_matches, wg_matches);
 }
 
 // full-term decode kernel (parity entry): one wave per 128-doc block
