@@ -40,6 +40,7 @@
 #include <cstring>
 #include <vector>
 #include <chrono>
+#include <cstdio>
 
 #include "../../../include/sdb_gpu.h"
 #include "sdb_internal.h"
@@ -412,6 +413,19 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* gh = a.ghist + (blockIdx.x & 7u) * SDB_HIST_BINS;
 
   unsigned long long wg_matches = 0;  // summed on tid 0, flushed once
+#ifdef SDB_TIMING
+  // per-phase cycle accounting (tid 0; flushed to bucket_out[0..5])
+  unsigned long long t_acc[6] = {0, 0, 0, 0, 0, 0};
+  long long t_mark = clock64();
+#define SDB_T(idx)                                         \
+  if (tid == 0) {                                          \
+    const long long now_ = clock64();                      \
+    t_acc[idx] += (unsigned long long)(now_ - t_mark);     \
+    t_mark = now_;                                         \
+  }
+#else
+#define SDB_T(idx)
+#endif
 
   // contiguous window range of this workgroup
   const uint32_t nwin = (a.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
@@ -465,6 +479,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
     }
     __syncthreads();
+    SDB_T(0)
 
     // term-major phases (fixed fp32 merge order -> bit-exact vs the oracle)
     for (uint32_t t = 0; t < a.nterms; ++t) {
@@ -540,6 +555,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
       __syncthreads();  // term-major merge order (bit-exact vs oracle)
     }
+    SDB_T(1)
     // advance every term's cursor once per window (cursors are only read
     // at the NEXT window's staging/phases, after the barrier below)
     if (tid < a.nterms) {
@@ -598,6 +614,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
     if (lane == 0) shared_misc[2 + wave] = wm;
     __syncthreads();  // hist + per-wave match counts complete
+    SDB_T(2)
     if (a.fcol)
       for (uint32_t i = tid; i < 2 * a.nbuckets; i += SDB_NTHREADS)
         if (lbuck[i]) atomicAdd(&a.bucket_out[i], lbuck[i]);
@@ -667,6 +684,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
     }
     __syncthreads();
+    SDB_T(3)
     float gtau;
     {
       const uint32_t bits = shared_misc[0];
@@ -711,9 +729,15 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
     }
     __syncthreads();  // window state reused next iteration
+    SDB_T(4)
 #endif
   }
   if (tid == 0 && wg_matches) atomicAdd(a.total_matches, wg_matches);
+#ifdef SDB_TIMING
+  if (tid == 0)
+    for (int i = 0; i < 6; ++i)
+      atomicAdd(&a.bucket_out[i], t_acc[i]);
+#endif
 }
 
 // full-term decode kernel (parity entry): one wave per 128-doc block
@@ -916,9 +940,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_gthresh, 0, 4, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_ghist, 0, 4 * SDB_HIST_BINS * 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
-  if (hybrid)
-    HIP_CHECK(hipMemsetAsync(ctx->d_buckets, 0, 8 * 2 * h_nbuckets,
-                             ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_buckets, 0, 8 * 2 * SDB_MAX_BUCKETS,
+                            ctx->stream));
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
@@ -1040,6 +1063,17 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   std::copy(cands.begin(), cands.begin() + kk, hits);
   *out_count = (uint32_t)kk;
   *total_matches = *ctx->h_matches;
+#ifdef SDB_TIMING
+  {
+    unsigned long long tdbg[6];
+    hipMemcpy(tdbg, ctx->d_buckets, 48, hipMemcpyDeviceToHost);
+    fprintf(stderr,
+            "[timing cyc/WG avg] zero+stage=%llu phases=%llu hist=%llu "
+            "tau=%llu append=%llu\n",
+            tdbg[0] / 256, tdbg[1] / 256, tdbg[2] / 256, tdbg[3] / 256,
+            tdbg[4] / 256);
+  }
+#endif
   if (hybrid && bucket_count && bucket_sum) {
     std::vector<unsigned long long> hb(2 * h_nbuckets);
     HIP_CHECK(hipMemcpy(hb.data(), ctx->d_buckets, 8ull * 2 * h_nbuckets,
