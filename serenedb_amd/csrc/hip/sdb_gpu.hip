@@ -587,8 +587,16 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // column BETWEEN filter here; filtered docs are unmarked so the append
     // pass skips them)
     uint32_t my_matches = 0;
-    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
-      if (cwin[off] >= mm) {
+    for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
+      // 4 match-counts per u32 read; half the words are all-zero at the
+      // bench densities (the byte-per-iteration walk measured ~500 cy/doc)
+      const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+      if (cw == 0) continue;
+#pragma unroll
+      for (uint32_t e = 0; e < 4; ++e) {
+        const uint32_t off = base + e;
+        if (off >= wlen) break;
+        if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
         if (a.fcol) {
           const long long vv = a.fcol[lo + off];
           if (vv < a.flo || vv > a.fhi) {
@@ -631,10 +639,12 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       known_bin = (uint32_t)(gt_now * inv_smax);
       if (known_bin >= SDB_HIST_BINS) known_bin = SDB_HIST_BINS - 1;
     }
-    for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
-      if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
+    const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
+    if (derive)
+      for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
+        if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
     __syncthreads();
-    if (wave == 0) {
+    if (derive && wave == 0) {
       uint32_t part = 0;
       if (SDB_HIST_BINS - 1 - 4 * (uint32_t)lane + 3 >= known_bin) {
 #pragma unroll
@@ -679,10 +689,17 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         for (uint32_t v = 0; v < SDB_NWAVES; ++v)
           total_m += shared_misc[2 + v];
         wg_matches += total_m;  // ONE global atomic at kernel end
-        shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
-                                           __HIP_MEMORY_SCOPE_AGENT);
       }
     }
+    if (!derive && wave == 0 && lane == 0) {
+      uint32_t total_m = 0;
+      for (uint32_t v = 0; v < SDB_NWAVES; ++v)
+        total_m += shared_misc[2 + v];
+      wg_matches += total_m;
+    }
+    if (wave == 0 && lane == 0)
+      shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
     __syncthreads();
     SDB_T(3)
     float gtau;
@@ -695,8 +712,16 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // ONE cursor atomicAdd per workgroup per window (the per-wave version
     // serialized ~1.5M atomics on one word — microarch 'dequeue' row)
     uint32_t my_cnt = 0;
-    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS)
-      if (cwin[off] >= mm && swin[off] >= gtau) ++my_cnt;
+    for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
+      const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+      if (cw == 0) continue;
+#pragma unroll
+      for (uint32_t e = 0; e < 4; ++e) {
+        const uint32_t off = base + e;
+        if (off >= wlen) break;
+        if (((cw >> (8 * e)) & 0xFFu) >= mm && swin[off] >= gtau) ++my_cnt;
+      }
+    }
     uint32_t* cnts = scratch;  // decode scratch is free here
     const uint32_t incl = wave_incl_scan(my_cnt, lane);
     if (lane == 63) cnts[SDB_NTHREADS + wave] = incl;
@@ -717,8 +742,14 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       return;
     }
     uint32_t pos = base + my_excl;
-    for (uint32_t off = tid; off < wlen; off += SDB_NTHREADS) {
-      if (cwin[off] >= mm) {
+    for (uint32_t wb = 4 * tid; wb < wlen; wb += 4 * SDB_NTHREADS) {
+      const uint32_t cw = ((const uint32_t*)cwin)[wb >> 2];
+      if (cw == 0) continue;
+#pragma unroll
+      for (uint32_t e = 0; e < 4; ++e) {
+        const uint32_t off = wb + e;
+        if (off >= wlen) break;
+        if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
         const float s = swin[off];
         if (s >= gtau) {
           a.cands[pos].score = s;
