@@ -586,35 +586,52 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // histogram of matching scores + local match count (hybrid: apply the
     // column BETWEEN filter here; filtered docs are unmarked so the append
     // pass skips them)
+    // the histogram feeds only the (sampled) tau derivation; non-derive
+    // windows take the branchless nonzero-byte count (mm==1, no filter)
+    const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
     uint32_t my_matches = 0;
-    for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
-      // 4 match-counts per u32 read; half the words are all-zero at the
-      // bench densities (the byte-per-iteration walk measured ~500 cy/doc)
-      const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
-      if (cw == 0) continue;
+    if (!derive && mm == 1 && !a.fcol) {
+      for (uint32_t base = 4 * tid; base < wlen;
+           base += 4 * SDB_NTHREADS) {
+        const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+        if (!cw) continue;
+        // 0x80 per nonzero byte -> popcount (tail bytes beyond wlen are
+        // zeroed LDS, never counted)
+        const uint32_t nz =
+          (cw | (0x7F7F7F7Fu + (cw & 0x7F7F7F7Fu))) & 0x80808080u;
+        my_matches += (uint32_t)__popc(nz);
+      }
+    } else {
+      for (uint32_t base = 4 * tid; base < wlen;
+           base += 4 * SDB_NTHREADS) {
+        const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+        if (cw == 0) continue;
 #pragma unroll
-      for (uint32_t e = 0; e < 4; ++e) {
-        const uint32_t off = base + e;
-        if (off >= wlen) break;
-        if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
-        if (a.fcol) {
-          const long long vv = a.fcol[lo + off];
-          if (vv < a.flo || vv > a.fhi) {
-            cwin[off] = 0;
-            continue;
+        for (uint32_t e = 0; e < 4; ++e) {
+          const uint32_t off = base + e;
+          if (off >= wlen) break;
+          if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
+          if (a.fcol) {
+            const long long vv = a.fcol[lo + off];
+            if (vv < a.flo || vv > a.fhi) {
+              cwin[off] = 0;
+              continue;
+            }
+            const unsigned long long span =
+              (unsigned long long)(a.fhi - a.flo) + 1ull;
+            uint32_t bkt = (uint32_t)(
+              ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
+            if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
+            atomicAdd(&lbuck[2 * bkt], 1ull);
+            atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
           }
-          const unsigned long long span =
-            (unsigned long long)(a.fhi - a.flo) + 1ull;
-          uint32_t bkt = (uint32_t)(
-            ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
-          if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
-          atomicAdd(&lbuck[2 * bkt], 1ull);
-          atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
+          ++my_matches;
+          if (derive) {
+            uint32_t bin = (uint32_t)(swin[off] * inv_smax);
+            if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
+            atomicAdd(&hist[bin], 1u);
+          }
         }
-        ++my_matches;
-        uint32_t bin = (uint32_t)(swin[off] * inv_smax);
-        if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
-        atomicAdd(&hist[bin], 1u);
       }
     }
     uint32_t wm = my_matches;
@@ -639,7 +656,6 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       known_bin = (uint32_t)(gt_now * inv_smax);
       if (known_bin >= SDB_HIST_BINS) known_bin = SDB_HIST_BINS - 1;
     }
-    const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
     if (derive)
       for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
         if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
