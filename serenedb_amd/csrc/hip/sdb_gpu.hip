@@ -449,6 +449,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     if (lo > a.doc_count) break;
     const uint32_t hi = min(lo + SDB_WIN_DOCS - 1u, a.doc_count);
     const uint32_t wlen = hi - lo + 1u;
+    uint32_t my_excl_snap = 0;
 
     // zero windows + histogram
     for (uint32_t i = tid; i < SDB_WIN_DOCS; i += SDB_NTHREADS)
@@ -478,6 +479,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         }
       }
     }
+    if (tid == 0)
+      shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
     __syncthreads();
     SDB_T(0)
 
@@ -586,53 +590,55 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // histogram of matching scores + local match count (hybrid: apply the
     // column BETWEEN filter here; filtered docs are unmarked so the append
     // pass skips them)
-    // the histogram feeds only the (sampled) tau derivation; non-derive
-    // windows take the branchless nonzero-byte count (mm==1, no filter)
+    // ONE fused pass: match count, (sampled) histogram, and the candidate
+    // count against the window's threshold snapshot (read during staging;
+    // one window stale = smaller = conservative: a few extra candidates,
+    // never a dropped top-k member)
     const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
+    float gtau_w;
+    {
+      const uint32_t bits = shared_misc[0];
+      __builtin_memcpy(&gtau_w, &bits, 4);
+    }
     uint32_t my_matches = 0;
-    if (!derive && mm == 1 && !a.fcol) {
-      for (uint32_t base = 4 * tid; base < wlen;
-           base += 4 * SDB_NTHREADS) {
-        const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
-        if (!cw) continue;
-        // 0x80 per nonzero byte -> popcount (tail bytes beyond wlen are
-        // zeroed LDS, never counted)
-        const uint32_t nz =
-          (cw | (0x7F7F7F7Fu + (cw & 0x7F7F7F7Fu))) & 0x80808080u;
-        my_matches += (uint32_t)__popc(nz);
-      }
-    } else {
-      for (uint32_t base = 4 * tid; base < wlen;
-           base += 4 * SDB_NTHREADS) {
-        const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
-        if (cw == 0) continue;
+    uint32_t my_cnt = 0;
+    for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
+      const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+      if (cw == 0) continue;
 #pragma unroll
-        for (uint32_t e = 0; e < 4; ++e) {
-          const uint32_t off = base + e;
-          if (off >= wlen) break;
-          if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
-          if (a.fcol) {
-            const long long vv = a.fcol[lo + off];
-            if (vv < a.flo || vv > a.fhi) {
-              cwin[off] = 0;
-              continue;
-            }
-            const unsigned long long span =
-              (unsigned long long)(a.fhi - a.flo) + 1ull;
-            uint32_t bkt = (uint32_t)(
-              ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
-            if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
-            atomicAdd(&lbuck[2 * bkt], 1ull);
-            atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
+      for (uint32_t e = 0; e < 4; ++e) {
+        const uint32_t off = base + e;
+        if (off >= wlen) break;
+        if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
+        if (a.fcol) {
+          const long long vv = a.fcol[lo + off];
+          if (vv < a.flo || vv > a.fhi) {
+            cwin[off] = 0;
+            continue;
           }
-          ++my_matches;
-          if (derive) {
-            uint32_t bin = (uint32_t)(swin[off] * inv_smax);
-            if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
-            atomicAdd(&hist[bin], 1u);
-          }
+          const unsigned long long span =
+            (unsigned long long)(a.fhi - a.flo) + 1ull;
+          uint32_t bkt = (uint32_t)(
+            ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
+          if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
+          atomicAdd(&lbuck[2 * bkt], 1ull);
+          atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
+        }
+        ++my_matches;
+        const float s = swin[off];
+        if (s >= gtau_w) ++my_cnt;
+        if (derive) {
+          uint32_t bin = (uint32_t)(s * inv_smax);
+          if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
+          atomicAdd(&hist[bin], 1u);
         }
       }
+    }
+    // per-wave candidate-count scan inputs (consumed after the barrier)
+    {
+      const uint32_t incl0 = wave_incl_scan(my_cnt, lane);
+      if (lane == 63) scratch[SDB_NTHREADS + wave] = incl0;
+      my_excl_snap = incl0 - my_cnt;
     }
     uint32_t wm = my_matches;
 #pragma unroll
@@ -659,7 +665,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     if (derive)
       for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
         if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
-    __syncthreads();
+    // (no barrier: wave 0 may miss this WG's freshest counts in the global
+    //  histogram — the derived bound is then merely looser, never invalid)
     if (derive && wave == 0) {
       uint32_t part = 0;
       if (SDB_HIST_BINS - 1 - 4 * (uint32_t)lane + 3 >= known_bin) {
@@ -713,41 +720,49 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         total_m += shared_misc[2 + v];
       wg_matches += total_m;
     }
-    if (wave == 0 && lane == 0)
-      shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
-                                         __HIP_MEMORY_SCOPE_AGENT);
-    __syncthreads();
     SDB_T(3)
-    float gtau;
-    {
-      const uint32_t bits = shared_misc[0];
-      __builtin_memcpy(&gtau, &bits, 4);
+    // First window of this workgroup: the staged threshold snapshot was 0
+    // (or near-0), which would append every match (~1M candidates across
+    // the grid). Refresh from the just-derived global threshold and
+    // recount. w_lo is always a derive window.
+    if (w == w_lo) {
+      if (tid == 0)
+        shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+      __syncthreads();
+      {
+        const uint32_t bits = shared_misc[0];
+        __builtin_memcpy(&gtau_w, &bits, 4);
+      }
+      uint32_t cnt2 = 0;
+      for (uint32_t base = 4 * tid; base < wlen;
+           base += 4 * SDB_NTHREADS) {
+        const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
+        if (cw == 0) continue;
+#pragma unroll
+        for (uint32_t e = 0; e < 4; ++e) {
+          const uint32_t off = base + e;
+          if (off >= wlen) break;
+          if (((cw >> (8 * e)) & 0xFFu) >= mm && swin[off] >= gtau_w)
+            ++cnt2;
+        }
+      }
+      const uint32_t incl2 = wave_incl_scan(cnt2, lane);
+      if (lane == 63) scratch[SDB_NTHREADS + wave] = incl2;
+      my_excl_snap = incl2 - cnt2;
+      __syncthreads();
     }
 
-    // append candidates with score >= gtau: per-thread counts, block scan,
-    // ONE cursor atomicAdd per workgroup per window (the per-wave version
-    // serialized ~1.5M atomics on one word — microarch 'dequeue' row)
-    uint32_t my_cnt = 0;
-    for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
-      const uint32_t cw = ((const uint32_t*)cwin)[base >> 2];
-      if (cw == 0) continue;
-#pragma unroll
-      for (uint32_t e = 0; e < 4; ++e) {
-        const uint32_t off = base + e;
-        if (off >= wlen) break;
-        if (((cw >> (8 * e)) & 0xFFu) >= mm && swin[off] >= gtau) ++my_cnt;
-      }
-    }
-    uint32_t* cnts = scratch;  // decode scratch is free here
-    const uint32_t incl = wave_incl_scan(my_cnt, lane);
-    if (lane == 63) cnts[SDB_NTHREADS + wave] = incl;
-    __syncthreads();
+    // append candidates with score >= the window's threshold snapshot:
+    // counts and per-wave scans were computed in the fused pass; ONE cursor
+    // atomicAdd per workgroup per window
     uint32_t wave_base = 0;
-    for (uint32_t v = 0; v < wave; ++v) wave_base += cnts[SDB_NTHREADS + v];
-    const uint32_t my_excl = wave_base + incl - my_cnt;
+    for (uint32_t v = 0; v < wave; ++v)
+      wave_base += scratch[SDB_NTHREADS + v];
+    const uint32_t my_excl = wave_base + my_excl_snap;
     uint32_t block_total = 0;
     for (uint32_t v = 0; v < SDB_NWAVES; ++v)
-      block_total += cnts[SDB_NTHREADS + v];
+      block_total += scratch[SDB_NTHREADS + v];
     if (tid == 0)
       shared_misc[1] =
         block_total ? atomicAdd(a.cand_count, block_total) : 0u;
@@ -767,7 +782,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         if (off >= wlen) break;
         if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
         const float s = swin[off];
-        if (s >= gtau) {
+        if (s >= gtau_w) {
           a.cands[pos].score = s;
           a.cands[pos].doc = lo + off;
           a.cands[pos].segment_idx = a.seg_idx;
