@@ -633,6 +633,22 @@ typedef struct {
 /* Prepare cursors + scorer constants for one segment.
  * Global stats (sharded execution) may be injected; zeros mean "derive from
  * the provided segments" (ExecuteTopKWithCount single-node semantics). */
+/* binary search: first block with last_doc >= lo (the skip list's
+ * SeekToBlock analogue, skip_list.hpp) — the multithreaded baseline seeks
+ * each work chunk in O(log blocks); a linear walk from the term start made
+ * per-chunk positioning quadratic and understated the CPU baseline ~50x. */
+static const SdbBlockDesc* o_seek_block(const SdbBlockDesc* b,
+                                        const SdbBlockDesc* e, uint32_t lo) {
+  while (b < e) {
+    const SdbBlockDesc* m = b + (e - b) / 2;
+    if (m->last_doc < lo)
+      b = m + 1;
+    else
+      e = m;
+  }
+  return b;
+}
+
 static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
                           const float* boosts, uint32_t nterms, float k1,
                           float b, uint64_t g_dwf, const uint64_t* g_dwt,
@@ -903,6 +919,8 @@ static void* o_mt_worker(void* argp) {
     if (hi > a->v->hdr->doc_count) hi = a->v->hdr->doc_count;
     o_prep_cursors(a->v, a->term_idx, a->boosts, a->nterms, a->k1, a->b,
                    a->g_dwf, a->g_dwt, a->g_ttf, cur);
+    for (uint32_t t = 0; t < a->nterms; ++t)
+      cur[t].d = o_seek_block(cur[t].d, cur[t].dend, lo);
     /* pull the shared threshold before the chunk; publish after
      * (duckdb_search_full_scan.cpp:1918-1921 CAS-max mirror) */
     const float g = o_atomic_thresh_get(a->g_thresh_bits);
