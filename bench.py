@@ -228,7 +228,7 @@ def bench_bm25(args, hybrid=False):
         "unit": "GB/s",
         "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
         "traffic": None,
-        "note": "achieved = algorithmic bytes (postings payload + 4B norm/posting + 32B/block desc) / window-kernel time (HIP events); see profiles/ for rocprofv3 evidence",
+        "note": "achieved = algorithmic bytes (compressed postings payload incl. per-block doc/freq/norm streams + 28B descriptor per block) / window-kernel time (HIP events on the library stream); rocprofv3 evidence under profiles/",
     }
 
     # ---- CPU baseline (rank 0, N=1 only): the oracle's multithreaded
