@@ -254,16 +254,24 @@ def bench_bm25(args, hybrid=False):
         if hybrid:
             crng = np.random.default_rng(45)
             scol = crng.integers(0, span, sample_docs + 1).astype(np.int64)
-        tcpu = time.time()
-        while time.time() - tcpu < args.cpu_seconds and iters < 200:
-            if hybrid:
+        if hybrid:
+            tcpu = time.time()
+            while time.time() - tcpu < args.cpu_seconds and iters < 200:
                 po.execute_topk_hybrid(sblob, term_idx, boosts, k, scol,
                                        flo, fhi, nbuckets,
                                        global_stats=gstats)
-            else:
-                po.execute_topk_mt(sblob, term_idx, boosts, k,
-                                   nthreads=ncores, global_stats=gstats)
-            iters += 1
+                iters += 1
+        else:
+            # estimate per-query time, then run the whole workload inside
+            # the thread pool (amortizes 256 pthread spawns per query)
+            t1 = time.time()
+            po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
+                               global_stats=gstats, iters=4)
+            est = (time.time() - t1) / 4
+            iters = max(1, min(2000, int(args.cpu_seconds / max(est, 1e-4))))
+            tcpu = time.time()
+            po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
+                               global_stats=gstats, iters=iters)
         tcpu = time.time() - tcpu
         cpu_baseline = {
             "value": round(sample_postings * iters / tcpu, 1),

@@ -196,8 +196,10 @@ def execute_topk_mech(blobs, term_idx, boosts, k, min_match=1, k1=1.2,
 
 
 def execute_topk_mt(blob, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
-                    nthreads=0, global_stats=None):
-    """Multithreaded mechanics baseline over one segment (timed CPU leg)."""
+                    nthreads=0, global_stats=None, iters=1):
+    """Multithreaded mechanics baseline over one segment (timed CPU leg).
+    iters > 1 repeats the whole query inside the thread pool (thread
+    creation would otherwise dominate sub-ms queries)."""
     if nthreads == 0:
         nthreads = os.cpu_count() or 1
     buf = np.frombuffer(blob, dtype=np.uint8)
@@ -211,13 +213,13 @@ def execute_topk_mt(blob, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
     hits = (OScoreDoc * k)()
     out_count = C.c_uint32(0)
     total = C.c_uint64(0)
-    rc = lib().o_execute_topk_mt(
+    rc = lib().o_execute_topk_mt_iters(
         buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(blob)),
         ti.ctypes.data_as(PU32), bo.ctypes.data_as(C.POINTER(C.c_float)),
         C.c_uint32(len(ti)), C.c_uint32(min_match), C.c_float(k1),
         C.c_float(b), C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf),
-        C.c_uint32(k), C.c_uint32(nthreads), hits, C.byref(out_count),
-        C.byref(total))
+        C.c_uint32(k), C.c_uint32(nthreads), C.c_uint32(iters), hits,
+        C.byref(out_count), C.byref(total))
     assert rc == 0, rc
     del keep2
     return _hits_to_np(hits, out_count.value), total.value
