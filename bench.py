@@ -272,7 +272,7 @@ def bench_bm25(args, hybrid=False):
             po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
                                global_stats=gstats, iters=4)
             est = (time.time() - t1) / 4
-            iters = max(1, min(2000, int(args.cpu_seconds / max(est, 1e-4))))
+            iters = max(1, min(600, int(args.cpu_seconds / max(est, 1e-4))))
             tcpu = time.time()
             po.execute_topk_mt(sblob, term_idx, boosts, k, nthreads=ncores,
                                global_stats=gstats, iters=iters)
