@@ -229,3 +229,16 @@ def test_hybrid_parity(ctx):
         hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
     np.testing.assert_array_equal(bcnt, obcnt)
     np.testing.assert_array_equal(bsum, obsum)
+
+
+def test_bm15_parity(ctx):
+    """BM15 (b=0): norm_const = k, no length normalization
+    (bm25.cpp:296-299 stats branch)."""
+    blob, _, _ = make_corpus(53, 200_000, [0.05, 0.02])
+    seg = ctx.load_segment(blob)
+    hits, total = ctx.execute_topk([seg], [0, 1], [1.0, 1.0], 100, b=0.0)
+    ohits, ototal = po.execute_topk([blob], [0, 1], [1.0, 1.0], 100, b=0.0)
+    assert total == ototal
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+    np.testing.assert_array_equal(
+        hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
