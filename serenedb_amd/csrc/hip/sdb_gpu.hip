@@ -327,6 +327,76 @@ __device__ __forceinline__ bool try_block_fused(
   return true;
 }
 
+// Dual-block fused path: both blocks' packed-word loads are issued before
+// either scan, so the two dependent-load chains overlap (the decode phase
+// is latency-bound — SDB_TIMING showed it at 43% of the kernel).
+__device__ __forceinline__ bool try_block_fused2(
+  const uint8_t* pl, const SdbBlockDesc& da, const SdbBlockDesc& db_,
+  int lane, uint32_t norm_stream, uint32_t lo, uint32_t hi, float num,
+  float nc, float nl, const uint32_t* norms_col, float* swin,
+  uint8_t* cwin) {
+  const uint8_t* adoc = pl + da.doc_off;
+  const uint8_t* afrq = pl + da.freq_off;
+  const uint8_t* anrm = afrq + da.flags;
+  const uint8_t* bdoc = pl + db_.doc_off;
+  const uint8_t* bfrq = pl + db_.freq_off;
+  const uint8_t* bnrm = bfrq + db_.flags;
+  const uint32_t adt = adoc[0], aft = afrq[0];
+  const uint32_t bdt = bdoc[0], bft = bfrq[0];
+  const uint32_t ant = norm_stream ? anrm[0] : SDB_E_BITPACK_01;
+  const uint32_t bnt = norm_stream ? bnrm[0] : SDB_E_BITPACK_01;
+  if (adt < SDB_DE_DELTA_BITPACK_02 || aft < SDB_E_BITPACK_01 ||
+      ant < SDB_E_BITPACK_01 || bdt < SDB_DE_DELTA_BITPACK_02 ||
+      bft < SDB_E_BITPACK_01 || bnt < SDB_E_BITPACK_01 || !norm_stream)
+    return false;
+  const uint32_t i0 = 2u * lane, i1 = i0 + 1;
+  // every load issues here
+  const uint32_t a_d0 = extract_packed(adoc + 1, adt - SDB_DE_DELTA_BITPACK_02 + 2, i0);
+  const uint32_t a_d1 = extract_packed(adoc + 1, adt - SDB_DE_DELTA_BITPACK_02 + 2, i1);
+  const uint32_t b_d0 = extract_packed(bdoc + 1, bdt - SDB_DE_DELTA_BITPACK_02 + 2, i0);
+  const uint32_t b_d1 = extract_packed(bdoc + 1, bdt - SDB_DE_DELTA_BITPACK_02 + 2, i1);
+  const uint32_t a_f0 = extract_packed(afrq + 1, aft - SDB_E_BITPACK_01 + 1, i0);
+  const uint32_t a_f1 = extract_packed(afrq + 1, aft - SDB_E_BITPACK_01 + 1, i1);
+  const uint32_t b_f0 = extract_packed(bfrq + 1, bft - SDB_E_BITPACK_01 + 1, i0);
+  const uint32_t b_f1 = extract_packed(bfrq + 1, bft - SDB_E_BITPACK_01 + 1, i1);
+  const uint32_t a_n0 = extract_packed(anrm + 1, ant - SDB_E_BITPACK_01 + 1, i0);
+  const uint32_t a_n1 = extract_packed(anrm + 1, ant - SDB_E_BITPACK_01 + 1, i1);
+  const uint32_t b_n0 = extract_packed(bnrm + 1, bnt - SDB_E_BITPACK_01 + 1, i0);
+  const uint32_t b_n1 = extract_packed(bnrm + 1, bnt - SDB_E_BITPACK_01 + 1, i1);
+  // two independent wave scans (shfl chains interleave)
+  const uint32_t a_pair = a_d0 + a_d1;
+  const uint32_t b_pair = b_d0 + b_d1;
+  uint32_t a_incl = a_pair, b_incl = b_pair;
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) {
+    const uint32_t an = __shfl_up(a_incl, off, 64);
+    const uint32_t bn = __shfl_up(b_incl, off, 64);
+    if (lane >= off) {
+      a_incl += an;
+      b_incl += bn;
+    }
+  }
+  const uint32_t a_excl = a_incl - a_pair;
+  const uint32_t b_excl = b_incl - b_pair;
+  const uint32_t docs[4] = {da.prev_doc + a_excl + a_d0,
+                            da.prev_doc + a_excl + a_pair,
+                            db_.prev_doc + b_excl + b_d0,
+                            db_.prev_doc + b_excl + b_pair};
+  const uint32_t frqs[4] = {a_f0, a_f1, b_f0, b_f1};
+  const uint32_t nrms[4] = {a_n0, a_n1, b_n0, b_n1};
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const uint32_t doc = docs[e];
+    if (doc < lo || doc > hi) continue;
+    const float c1 = nc + nl * (float)nrms[e];
+    const float s = num - num * c1 / (c1 + (float)frqs[e]);
+    const uint32_t off = doc - lo;
+    swin[off] += s;
+    cwin[off] = (uint8_t)(cwin[off] + 1u);
+  }
+  return true;
+}
+
 // binary searches over the descriptor span of one term:
 // first block with last_doc >= lo  /  first block with prev_doc >= hi
 __device__ __forceinline__ uint64_t lower_bound_last_doc(
@@ -498,13 +568,32 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       // cursor has last_doc >= lo (maintained below); stop at first block
       // whose first doc (> prev_doc) lies beyond the window
       const uint32_t cur0 = cursors[t];
-      for (uint64_t b = te.desc_begin + cur0 + wave; b < dend;
-           b += SDB_NWAVES) {
+      uint64_t b = te.desc_begin + cur0 + wave;
+      while (b < dend) {
         const uint32_t rel = (uint32_t)(b - te.desc_begin) - cur0;
         const SdbBlockDesc d = rel < SDB_DESC_CACHE
                                  ? dcache[t * SDB_DESC_CACHE + rel]
                                  : a.desc[b];
         if (d.prev_doc >= hi) break;  // first doc > hi
+#if !defined(SDB_ABLATE_DECODE) && !defined(SDB_ABLATE_SCORE)
+        // pair this block with the wave's next one when both are the
+        // common fused shape: both chains' loads fly together
+        {
+          const uint64_t b2 = b + SDB_NWAVES;
+          if (b2 < dend) {
+            const uint32_t rel2 = (uint32_t)(b2 - te.desc_begin) - cur0;
+            const SdbBlockDesc d2 = rel2 < SDB_DESC_CACHE
+                                      ? dcache[t * SDB_DESC_CACHE + rel2]
+                                      : a.desc[b2];
+            if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
+                try_block_fused2(pl, d, d2, lane, a.norm_stream, lo, hi,
+                                 num, nc, nl, a.norms, swin, cwin)) {
+              b += 2 * SDB_NWAVES;
+              continue;
+            }
+          }
+        }
+#endif
         // prefetch the NEXT block's payload (256 B per wave) so its decode
         // loads hit L1: the per-block chain was serial cold loads
         // (desc -> docs -> freqs -> norms, ~900 cy each)
@@ -532,8 +621,10 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #else
 #ifndef SDB_ABLATE_SCORE
         if (try_block_fused(pl, d, lane, a.norm_stream, lo, hi, num, nc, nl,
-                            a.norms, swin, cwin))
+                            a.norms, swin, cwin)) {
+          b += SDB_NWAVES;  // while-loop: explicit advance before continue
           continue;
+        }
 #endif
         decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
         decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
@@ -556,6 +647,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           cwin[off] = (uint8_t)(cwin[off] + 1u);
 #endif
         }
+        b += SDB_NWAVES;
       }
       __syncthreads();  // term-major merge order (bit-exact vs oracle)
     }
