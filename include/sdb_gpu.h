@@ -130,6 +130,16 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                          SdbScoreDoc* hits, uint32_t* out_count,
                          uint64_t* total_matches);
 
+/* Streaming scan — the RunStreamingScan / HitBatcher analogue
+ * (duckdb_search_full_scan.cpp:2370, index/hit_batcher.hpp:39-190): emit
+ * every matching doc id ascending into docs_out (up to cap) and, when
+ * col_out != NULL and a column is attached, the gathered i64 value per hit.
+ * *total_matches = full match count. Single segment. */
+int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                               const SdbQueryPlan* plan, uint32_t* docs_out,
+                               int64_t* col_out, uint64_t cap,
+                               uint64_t* out_count, uint64_t* total_matches);
+
 /* Raw postings-block decode of one term into caller buffers (docs+freqs,
  * df entries each). Parity/diagnostic entry (mirrors
  * FormatTraits128::ReadBlockDelta/ReadBlock, format_block_128.hpp:446-636);

@@ -325,3 +325,31 @@ def execute_topk_hybrid(blob, term_idx, boosts, k, col, flo, fhi, nbuckets,
     assert rc == 0, rc
     del keep2
     return _hits_to_np(hits, out_count.value), total.value, bcnt, bsum
+
+
+def execute_match_docs(blob, term_idx, boosts, cap, col=None, min_match=1,
+                       k1=1.2, b=0.75):
+    """Streaming match emission (doc-ascending) + optional column gather."""
+    buf = np.frombuffer(blob, dtype=np.uint8)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    docs = np.zeros(cap, dtype=np.uint32)
+    PI64 = C.POINTER(C.c_int64)
+    col_ptr, col_out, col_out_ptr = None, None, None
+    if col is not None:
+        col = np.ascontiguousarray(col, dtype=np.int64)
+        col_ptr = col.ctypes.data_as(PI64)
+        col_out = np.zeros(cap, dtype=np.int64)
+        col_out_ptr = col_out.ctypes.data_as(PI64)
+    out_n = C.c_uint64(0)
+    total = C.c_uint64(0)
+    rc = lib().o_execute_match_docs(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)),
+        ti.ctypes.data_as(PU32), bo.ctypes.data_as(C.POINTER(C.c_float)),
+        C.c_uint32(len(ti)), C.c_uint32(min_match), C.c_float(k1),
+        C.c_float(b), col_ptr, col_out_ptr,
+        docs.ctypes.data_as(PU32), C.c_uint64(cap), C.byref(out_n),
+        C.byref(total))
+    assert rc == 0, rc
+    n = out_n.value
+    return docs[:n], (col_out[:n] if col is not None else None), total.value

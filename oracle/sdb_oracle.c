@@ -809,6 +809,48 @@ int o_execute_topk_hybrid(const void* blob, uint64_t size,
   return 0;
 }
 
+/* STREAMING scan: emit every matching doc id ascending, plus (optionally)
+ * the gathered values of an attached i64 column — the RunStreamingScan /
+ * HitBatcher analogue (duckdb_search_full_scan.cpp:2370,
+ * index/hit_batcher.hpp:39-190: windows of hit doc-ids -> column gather ->
+ * DataChunk). Single segment. Returns up to cap docs; *total_matches is the
+ * full count (emission stops at cap but counting continues). */
+int o_execute_match_docs(const void* blob, uint64_t size,
+                         const uint32_t* term_idx, const float* boosts,
+                         uint32_t nterms, uint32_t min_match, float k1,
+                         float b, const int64_t* col, int64_t* col_out,
+                         uint32_t* docs_out, uint64_t cap,
+                         uint64_t* out_count, uint64_t* total_matches) {
+  SdbSegmentView v;
+  int rc = o_segment_parse(blob, size, &v);
+  if (rc) return rc;
+  OSegBlob sb = {blob, size};
+  uint64_t dwf, ttf;
+  uint64_t* dwt = (uint64_t*)malloc(sizeof(uint64_t) * nterms);
+  rc = o_global_stats(&sb, 1, term_idx, nterms, &dwf, &ttf, dwt);
+  if (rc) { free(dwt); return rc; }
+  OCursor* cur = (OCursor*)malloc(sizeof(OCursor) * nterms);
+  rc = o_prep_cursors(&v, term_idx, boosts, nterms, k1, b, dwf, dwt, ttf,
+                      cur);
+  if (rc) { free(cur); free(dwt); return rc; }
+  OCandVec cands = {0, 0, 0};
+  uint64_t matches =
+    o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1, 1,
+                 v.hdr->doc_count, NULL, &cands, 0, NULL);
+  /* cands are window-ordered = doc-ascending already (single segment) */
+  uint64_t n = cands.n < cap ? cands.n : cap;
+  for (uint64_t i = 0; i < n; ++i) {
+    docs_out[i] = cands.v[i].doc;
+    if (col && col_out) col_out[i] = col[cands.v[i].doc];
+  }
+  *out_count = n;
+  *total_matches = matches;
+  free(cands.v);
+  free(cur);
+  free(dwt);
+  return 0;
+}
+
 /* MECHANICS emulation (2k buffer + threshold), single-threaded — validates
  * transcribed reference fixtures and is the timed single-thread CPU path. */
 int o_execute_topk_mech(const OSegBlob* segs, uint32_t nsegs,

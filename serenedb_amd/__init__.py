@@ -230,6 +230,7 @@ def gpu():
             "sdb_gpu_execute_topk", "sdb_gpu_decode_term",
             "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
+            "sdb_gpu_execute_match_docs",
         ):
             getattr(lib, f).restype = C.c_int
         _gpu = lib
@@ -354,6 +355,25 @@ class GpuContext:
         res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
                             dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value, bcnt, bsum
+
+    def execute_match_docs(self, seg, term_idx, boosts, cap, min_match=1,
+                           k1=1.2, b=0.75, with_col=False):
+        import numpy as np
+
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b, None)
+        docs = np.zeros(cap, dtype=np.uint32)
+        cols = np.zeros(cap, dtype=np.int64) if with_col else None
+        out_n = C.c_uint64(0)
+        total = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_execute_match_docs(
+            self._ctx, seg, C.byref(plan),
+            docs.ctypes.data_as(C.POINTER(C.c_uint32)),
+            cols.ctypes.data_as(C.POINTER(C.c_int64)) if with_col else None,
+            C.c_uint64(cap), C.byref(out_n), C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_match_docs rc={rc}")
+        n = out_n.value
+        return docs[:n], (cols[:n] if with_col else None), total.value
 
     def decode_term(self, seg, term_idx, df):
         import numpy as np

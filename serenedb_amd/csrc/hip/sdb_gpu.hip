@@ -894,6 +894,14 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #endif
 }
 
+// column gather for the streaming scan: out[i] = col[docs[i]]
+__global__ void gather_col_kernel(const uint32_t* __restrict__ docs,
+                                  const long long* __restrict__ col,
+                                  long long* __restrict__ out, uint64_t n) {
+  const uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = col[docs[i]];
+}
+
 // full-term decode kernel (parity entry): one wave per 128-doc block
 __global__ void decode_term_kernel(const SdbBlockDesc* desc, uint64_t b0,
                                    uint64_t nblocks, const uint8_t* payload,
@@ -1034,9 +1042,11 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                           int64_t* bucket_count, int64_t* bucket_sum,
                           SdbScoreDoc* hits, uint32_t* out_count,
                           uint64_t* total_matches) {
-  if (!ctx || !segs || !plan || !hits || !out_count || !total_matches ||
+  if (!ctx || !segs || !plan || !out_count || !total_matches ||
       plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS || k == 0)
     return SDB_ERR_INVALID;
+  /* hits == NULL: candidate-only mode (streaming match emission reads the
+   * device candidate buffer itself; no host select) */
   if (hybrid)
     for (uint32_t s = 0; s < nsegs; ++s)
       if (!segs[s]->fcol) return SDB_ERR_INVALID;
@@ -1177,6 +1187,12 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   if (ctx->h_counts[1]) return SDB_ERR_OOM;  // candidate overflow
   const uint32_t ncand = ctx->h_counts[0];
   ctx->last_ncand = ncand;
+  if (!hits) {
+    *out_count = 0;
+    *total_matches = *ctx->h_matches;
+    ctx->last_gtau = 0.0f;
+    return SDB_OK;
+  }
   const auto t_rb0 = std::chrono::steady_clock::now();
   std::vector<SdbScoreDoc> cands(ncand);
   if (ncand) {
@@ -1281,6 +1297,59 @@ int sdb_gpu_execute_topk_hybrid(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   return exec_topk_impl(ctx, segs, nsegs, plan, k, 1, flo, fhi, nbuckets,
                         bucket_count, bucket_sum, hits, out_count,
                         total_matches);
+}
+
+// STREAMING scan (RunStreamingScan / HitBatcher analogue,
+// duckdb_search_full_scan.cpp:2370, index/hit_batcher.hpp:39-190): emit
+// every matching doc ascending plus (optionally) the gathered values of
+// the attached filter column. Implemented over the window kernel with
+// k = UINT32_MAX (the k-th bound never fires, every match is appended);
+// the host orders by doc and gathers. Single segment.
+int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                               const SdbQueryPlan* plan, uint32_t* docs_out,
+                               int64_t* col_out, uint64_t cap,
+                               uint64_t* out_count, uint64_t* total_matches) {
+  if (!ctx || !seg || !plan || !docs_out || !out_count || !total_matches)
+    return SDB_ERR_INVALID;
+  // run with a k no window can reach: threshold stays 0, all matches append
+  uint32_t dummy_n = 0;
+  SdbGpuSegment* segs1[1] = {seg};
+  int rc = exec_topk_impl(ctx, segs1, 1, plan, 0xFFFFFFFFu, 0, 0, 0, 0,
+                          nullptr, nullptr, /*hits=*/nullptr, &dummy_n,
+                          total_matches);
+  if (rc) return rc;
+  // all candidates are still on the device; re-read and order by doc
+  const uint32_t ncand = ctx->last_ncand;
+  std::vector<SdbScoreDoc> cands(ncand);
+  if (ncand)
+    HIP_CHECK(hipMemcpy(cands.data(), ctx->d_cands,
+                        sizeof(SdbScoreDoc) * ncand, hipMemcpyDeviceToHost));
+  std::sort(cands.begin(), cands.end(),
+            [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
+              return x.doc < y.doc;
+            });
+  const uint64_t n = std::min<uint64_t>(cap, cands.size());
+  for (uint64_t i = 0; i < n; ++i) docs_out[i] = cands[i].doc;
+  if (col_out && seg->fcol && n) {
+    // device gather: upload the doc-ordered hit ids, one gather kernel,
+    // one D2H of the values (HitBatcher's dense/scatter gather analogue)
+    uint32_t* d_docs;
+    long long* d_vals;
+    HIP_CHECK(hipMalloc(&d_docs, 4 * n));
+    HIP_CHECK(hipMalloc(&d_vals, 8 * n));
+    HIP_CHECK(hipMemcpy(d_docs, docs_out, 4 * n, hipMemcpyHostToDevice));
+    const uint32_t nb = (uint32_t)((n + 255) / 256);
+    hipLaunchKernelGGL(gather_col_kernel, dim3(nb), dim3(256), 0,
+                       ctx->stream, d_docs, seg->fcol, d_vals, n);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipMemcpyAsync(col_out, d_vals, 8 * n, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    hipFree(d_docs);
+    hipFree(d_vals);
+  }
+  *out_count = n;
+  return SDB_OK;
 }
 
 int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,

@@ -242,3 +242,20 @@ def test_bm15_parity(ctx):
     np.testing.assert_array_equal(hits["doc"], ohits["doc"])
     np.testing.assert_array_equal(
         hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
+
+
+def test_match_docs_parity(ctx):
+    """streaming scan: GPU match emission + device column gather == oracle"""
+    doc_count = 300_000
+    blob, _, _ = make_corpus(62, doc_count, [0.05, 0.02, 0.01])
+    col = np.random.default_rng(9).integers(0, 1 << 40, doc_count + 1
+                                            ).astype(np.int64)
+    seg = ctx.load_segment(blob)
+    ctx.attach_column(seg, col)
+    docs, vals, total = ctx.execute_match_docs(seg, [0, 1, 2], [1.0] * 3,
+                                               doc_count, with_col=True)
+    odocs, ovals, ototal = po.execute_match_docs(blob, [0, 1, 2], [1.0] * 3,
+                                                 doc_count, col=col)
+    assert total == ototal
+    np.testing.assert_array_equal(docs, odocs)
+    np.testing.assert_array_equal(vals, ovals)

@@ -294,3 +294,18 @@ def test_hybrid_vs_brute():
                           minlength=nbuckets).astype(np.int64)
     np.testing.assert_array_equal(bcnt, exp_cnt)
     np.testing.assert_array_equal(bsum, exp_sum)
+
+
+def test_match_docs_streaming():
+    """Streaming emission (RunStreamingScan/HitBatcher analogue): every
+    matching doc ascending + gathered column values."""
+    doc_count = 25_000
+    blob, postings, _ = synth_corpus(61, doc_count, [0.05, 0.02])
+    col = np.random.default_rng(3).integers(0, 1000, doc_count + 1
+                                            ).astype(np.int64)
+    docs, vals, total = po.execute_match_docs(blob, [0, 1], [1.0, 1.0],
+                                              doc_count, col=col)
+    exp = np.union1d(postings[0][0], postings[1][0])
+    assert total == len(exp)
+    np.testing.assert_array_equal(docs, exp)
+    np.testing.assert_array_equal(vals, col[exp])
