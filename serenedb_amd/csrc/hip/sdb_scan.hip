@@ -27,7 +27,12 @@
 #include "../../../include/sdb_gpu.h"
 #include "sdb_internal.h"
 
+#ifndef SCAN_NTHREADS
 #define SCAN_NTHREADS 256u
+#endif
+#ifndef SCAN_MAXB
+#define SCAN_MAXB 4096u
+#endif
 #define SCAN_MAX_GROUPS 2048u
 #define SCAN_MAX_AGGS 8u
 #define SCAN_MAX_PREDS 4u
@@ -256,7 +261,7 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   // memory-bound grid sizing (guide §6 G11): cap ~8 blocks/CU, grid-stride
   uint32_t nblocks =
     (uint32_t)((tab->rows / 2 + SCAN_NTHREADS - 1) / SCAN_NTHREADS);
-  if (nblocks > 4096) nblocks = 4096;
+  if (nblocks > SCAN_MAXB) nblocks = SCAN_MAXB;
   if (nblocks < 1) nblocks = 1;
   const size_t lds = 8ull * nslots;
   hipLaunchKernelGGL(scan_agg_kernel, dim3(nblocks), dim3(SCAN_NTHREADS), lds,
