@@ -149,11 +149,14 @@ int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
 
 /* ---- columnar scan -> filter -> hash aggregate ---- */
 
-/* Column layout: dense device-resident columns (round 1: uncompressed i64 /
- * f32; FoR/bitpack codecs are a planned next row). */
+/* Column layout: dense device-resident columns, or FoR/bitpack row groups
+ * with min/max zonemaps (this repo's own codec — the reference's column
+ * codecs live in the un-vendored DuckDB fork). */
 typedef enum SdbColType {
-  SDB_COL_I64 = 0,
-  SDB_COL_F32 = 1,
+  SDB_COL_I64 = 0,     /* dense i64 array */
+  SDB_COL_F32 = 1,     /* dense f32 array */
+  SDB_COL_I64_FOR = 2, /* FoR/bitpack row groups + zonemaps: the blob from
+                          sdb_host_encode_col_i64 (data = blob pointer) */
 } SdbColType;
 
 typedef struct SdbColumnView {

@@ -201,6 +201,34 @@ def synth_norms(seed, doc_count):
     return norms
 
 
+def encode_col_i64(vals, group_rows=65536):
+    """FoR/bitpack row-group encoding of an i64 column (+zonemaps)."""
+    import numpy as np
+
+    vals = np.ascontiguousarray(vals, dtype=np.int64)
+    blob = C.c_void_p(0)
+    size = C.c_uint64(0)
+    rc = host().sdb_host_encode_col_i64(
+        vals.ctypes.data_as(C.POINTER(C.c_int64)), C.c_uint64(len(vals)),
+        C.c_uint32(group_rows), C.byref(blob), C.byref(size))
+    assert rc == 0, rc
+    out = _copy_blob(blob, size.value)
+    host().sdb_host_blob_free(blob)
+    return out
+
+
+def decode_col_i64(blob, rows):
+    import numpy as np
+
+    buf = np.frombuffer(blob, dtype=np.uint8)
+    out = np.zeros(rows, dtype=np.int64)
+    rc = host().sdb_host_decode_col_i64(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)),
+        out.ctypes.data_as(C.POINTER(C.c_int64)), C.c_uint64(rows))
+    assert rc == 0, rc
+    return out
+
+
 def bm25_stats(docs_with_field, docs_with_term, total_term_freq, k=1.2,
                b=0.75):
     idf = C.c_float(0)

@@ -6,17 +6,19 @@
 // reference delegates to external DuckDB (PhysicalHashAggregate, un-vendored;
 // result-level parity per SURVEY.md §8c).
 //
-// MI355X design: this is pure HBM-bandwidth work (no MFMA). Columns are
-// device-resident dense arrays (FoR/bitpack codecs: planned next row,
-// SURVEY.md §8f). A grid-stride kernel reads rows with coalesced wide loads;
-// each workgroup accumulates into LDS per-group slots (group keys are dense
-// [0, ngroups), the "LDS-staged open-addressed buckets" of north_star with a
-// perfect hash), then flushes once per workgroup with device atomics.
-// COUNT/SUM(i64) are exact (wrap-around two's complement); SUM over an f32
-// column accumulates in f64 (atomic order nondeterministic; parity vs the
-// oracle's sequential f64 sum is within ~1e-12 relative at 1e9 rows,
-// asserted at 1e-7 in tests — mirrors the reference's own thread-order-
-// dependent fp aggregation).
+// MI355X design: pure HBM-bandwidth work (no MFMA). Columns are either
+// dense device-resident arrays or FoR/bitpack row groups (this repo's own
+// codec, sdb_host.cpp: per-group frame-of-reference base + horizontal
+// bitpack + min/max zonemap — the reference's block codecs live in the
+// un-vendored DuckDB fork, so parity is at result level). The kernel walks
+// row GROUPS grid-stride: one zonemap check per group can skip the whole
+// group (DeadUntil analogue, full_scanner.h:61-71); within a group,
+// threads decode packed values in registers (coalesced u32 payload reads),
+// evaluate predicates, and accumulate into LDS per-group-key slots (dense
+// keys = perfect hash); one device-atomic flush per block at kernel end.
+// COUNT/SUM(i64) exact (wrap-around); SUM(f32) in f64 (atomic order
+// nondeterministic, like the reference's thread-order-dependent fp
+// aggregation; parity vs the oracle's sequential sum at ~1e-12 relative).
 
 #include <hip/hip_runtime.h>
 
@@ -28,10 +30,10 @@
 #include "sdb_internal.h"
 
 #ifndef SCAN_NTHREADS
-#define SCAN_NTHREADS 256u
-#endif
+#define SCAN_NTHREADS 1024u  // swept on-box: 1024t/2048b = 4.55 TB/s vs
+#endif                       // 256t/4096b = 3.64 TB/s (profiles/)
 #ifndef SCAN_MAXB
-#define SCAN_MAXB 4096u
+#define SCAN_MAXB 2048u
 #endif
 #define SCAN_MAX_GROUPS 2048u
 #define SCAN_MAX_AGGS 8u
@@ -44,31 +46,71 @@
     if (_e != hipSuccess) return SDB_ERR_HIP;          \
   } while (0)
 
+// mirrors the host codec (sdb_host.cpp sdb_host_encode_col_i64)
+struct SdbColHeaderDev {
+  uint64_t magic;
+  uint64_t rows;
+  uint32_t group_rows;
+  uint32_t ngroups;
+  uint64_t off_desc;
+  uint64_t off_payload;
+  uint64_t size;
+};
+struct SdbColGroupDescDev {
+  int64_t base;
+  int64_t vmin;
+  int64_t vmax;
+  uint64_t word_off;
+  uint16_t width;
+  uint16_t pad[3];
+};
+#define SDB_COL_MAGIC_DEV 0x31304C4F43424453ull
+
+struct ColRef {
+  const void* data;                  // raw array, or FoR payload (u32*)
+  const SdbColGroupDescDev* desc;    // FoR group table (null = raw)
+};
+
 struct SdbGpuTable {
-  void* cols[16];
+  void* cols[16];        // raw device array or whole FoR blob
   SdbColType types[16];
+  ColRef refs[16];       // device pointers into cols[] allocations
   uint32_t ncols;
   uint64_t rows;
+  uint32_t group_rows;   // shared by every FoR column (0 if none)
 };
 
 struct ScanArgs {
-  const int64_t* keys;
+  ColRef keys;
   uint64_t rows;
+  uint32_t group_rows;  // row-group tiling (also used for raw-only tables)
   uint32_t ngroups;
   uint32_t naggs;
   uint32_t npreds;
-  // preds (on i64 columns; f32 preds can be added when a config needs them)
-  const int64_t* pred_col[SCAN_MAX_PREDS];
+  ColRef pred_col[SCAN_MAX_PREDS];
   int pred_op[SCAN_MAX_PREDS];
   int64_t pred_lo[SCAN_MAX_PREDS];
   int64_t pred_hi[SCAN_MAX_PREDS];
-  // aggs
-  const void* agg_col[SCAN_MAX_AGGS];
+  ColRef agg_col[SCAN_MAX_AGGS];
   int agg_op[SCAN_MAX_AGGS];
-  // outputs: [group * naggs + agg] as u64 (COUNT/SUM_I64) or f64 (SUM_F64)
   unsigned long long* out;
   unsigned long long* rows_passed;
 };
+
+__device__ __forceinline__ int64_t col_read(const ColRef& c, uint64_t r0,
+                                            uint64_t r,
+                                            const SdbColGroupDescDev& d) {
+  if (!c.desc) return ((const long long*)c.data)[r];
+  if (d.width == 0) return d.base;
+  const uint32_t* w = (const uint32_t*)c.data + d.word_off;
+  const uint64_t bit = (r - r0) * d.width;
+  uint64_t v = w[bit >> 5] >> (bit & 31);
+  if ((bit & 31) + d.width > 32)
+    v |= (uint64_t)w[(bit >> 5) + 1] << (32 - (bit & 31));
+  const uint64_t mask =
+    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
+  return d.base + (int64_t)(v & mask);
+}
 
 __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -77,99 +119,75 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) acc[i] = 0;
   __syncthreads();
 
-  // 2 rows per thread with 16-byte loads (guide G13: vectorize ALWAYS);
-  // consecutive lanes read consecutive longlong2 -> 1 KiB per wave per
-  // instruction on the i64 columns
-  const uint64_t pair_stride = (uint64_t)gridDim.x * SCAN_NTHREADS * 2u;
-  uint64_t my_passed = 0;
-  const uint64_t rows2 = a.rows & ~1ull;
-  for (uint64_t r = ((uint64_t)blockIdx.x * SCAN_NTHREADS + threadIdx.x) * 2u;
-       r < rows2; r += pair_stride) {
-    bool okv[2] = {true, true};
+  const uint32_t n_rowgroups =
+    (uint32_t)((a.rows + a.group_rows - 1) / a.group_rows);
+  unsigned long long my_passed = 0;
+  SdbColGroupDescDev kd{}, pd[SCAN_MAX_PREDS], ad[SCAN_MAX_AGGS];
+
+  for (uint32_t rg = blockIdx.x; rg < n_rowgroups; rg += gridDim.x) {
+    const uint64_t r0 = (uint64_t)rg * a.group_rows;
+    const uint64_t r1 = min(a.rows, r0 + a.group_rows);
+    // group descriptors + zonemap skip (DeadUntil analogue)
+    bool dead = false;
     for (uint32_t p = 0; p < a.npreds; ++p) {
-      longlong2 x;
-      __builtin_memcpy(&x, &a.pred_col[p][r], 16);
-      const int64_t xs[2] = {x.x, x.y};
-#pragma unroll
-      for (int e = 0; e < 2; ++e) {
+      if (a.pred_col[p].desc) {
+        pd[p] = a.pred_col[p].desc[rg];
         switch (a.pred_op[p]) {
-          case SDB_PRED_LT: okv[e] &= xs[e] < a.pred_lo[p]; break;
-          case SDB_PRED_GE: okv[e] &= xs[e] >= a.pred_lo[p]; break;
+          case SDB_PRED_LT: dead |= pd[p].vmin >= a.pred_lo[p]; break;
+          case SDB_PRED_GE: dead |= pd[p].vmax < a.pred_lo[p]; break;
           case SDB_PRED_BETWEEN:
-            okv[e] &= (xs[e] >= a.pred_lo[p]) & (xs[e] <= a.pred_hi[p]);
+            dead |= (pd[p].vmax < a.pred_lo[p]) |
+                    (pd[p].vmin > a.pred_hi[p]);
             break;
           default: break;
         }
       }
     }
-    if (!okv[0] && !okv[1]) continue;
-    longlong2 kk;
-    __builtin_memcpy(&kk, &a.keys[r], 16);
-    const int64_t ks[2] = {kk.x, kk.y};
-#pragma unroll
-    for (int e = 0; e < 2; ++e) {
-      if (!okv[e]) continue;
+    if (dead) continue;
+    if (a.keys.desc) kd = a.keys.desc[rg];
+    for (uint32_t q = 0; q < a.naggs; ++q)
+      if (a.agg_col[q].desc) ad[q] = a.agg_col[q].desc[rg];
+
+    for (uint64_t r = r0 + threadIdx.x; r < r1; r += SCAN_NTHREADS) {
+      bool ok = true;
+      for (uint32_t p = 0; p < a.npreds; ++p) {
+        const int64_t x = col_read(a.pred_col[p], r0, r, pd[p]);
+        switch (a.pred_op[p]) {
+          case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
+          case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
+          case SDB_PRED_BETWEEN:
+            ok &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+            break;
+          default: break;
+        }
+      }
+      if (!ok) continue;
       ++my_passed;
-      const uint32_t g = (uint32_t)ks[e];
+      const uint32_t grp = (uint32_t)col_read(a.keys, r0, r, kd);
       for (uint32_t q = 0; q < a.naggs; ++q) {
-        unsigned long long* slot = &acc[g * a.naggs + q];
+        unsigned long long* slot = &acc[grp * a.naggs + q];
         switch (a.agg_op[q]) {
           case SDB_AGG_COUNT:
             atomicAdd(slot, 1ull);
             break;
           case SDB_AGG_SUM_I64:
-            atomicAdd(slot, (unsigned long long)((const int64_t*)
-                                                   a.agg_col[q])[r + e]);
+            atomicAdd(slot, (unsigned long long)col_read(a.agg_col[q], r0,
+                                                         r, ad[q]));
             break;
           case SDB_AGG_SUM_F64:
             atomicAdd((double*)slot,
-                      (double)((const float*)a.agg_col[q])[r + e]);
+                      (double)((const float*)a.agg_col[q].data)[r]);
             break;
         }
       }
     }
   }
-  // odd tail row
-  if (blockIdx.x == 0 && threadIdx.x == 0 && (a.rows & 1ull)) {
-    const uint64_t r = a.rows - 1;
-    bool ok = true;
-    for (uint32_t p = 0; p < a.npreds; ++p) {
-      const int64_t x = a.pred_col[p][r];
-      switch (a.pred_op[p]) {
-        case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
-        case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
-        case SDB_PRED_BETWEEN:
-          ok &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
-          break;
-        default: break;
-      }
-    }
-    if (ok) {
-      ++my_passed;
-      const uint32_t g = (uint32_t)a.keys[r];
-      for (uint32_t q = 0; q < a.naggs; ++q) {
-        unsigned long long* slot = &acc[g * a.naggs + q];
-        switch (a.agg_op[q]) {
-          case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
-          case SDB_AGG_SUM_I64:
-            atomicAdd(slot,
-                      (unsigned long long)((const int64_t*)a.agg_col[q])[r]);
-            break;
-          case SDB_AGG_SUM_F64:
-            atomicAdd((double*)slot,
-                      (double)((const float*)a.agg_col[q])[r]);
-            break;
-        }
-      }
-    }
-  }
-  // rows_passed: wave-reduce then one atomic per wave
+
   unsigned long long wp = my_passed;
 #pragma unroll
   for (int off = 32; off; off >>= 1) wp += __shfl_down(wp, off, 64);
   if ((threadIdx.x & 63) == 0 && wp) atomicAdd(a.rows_passed, wp);
   __syncthreads();
-  // flush LDS accumulators
   for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) {
     const uint32_t q = i % a.naggs;
     if (a.agg_op[q] == SDB_AGG_SUM_F64) {
@@ -191,13 +209,34 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
   auto* tab = new SdbGpuTable{};
   tab->ncols = ncols;
   tab->rows = rows;
+  tab->group_rows = 0;
   for (uint32_t c = 0; c < ncols; ++c) {
-    const size_t esz = cols[c].type == SDB_COL_I64 ? 8 : 4;
     tab->types[c] = cols[c].type;
-    HIP_CHECK(hipMalloc(&tab->cols[c], esz * rows));
-    HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, esz * rows,
-                        hipMemcpyHostToDevice));
+    if (cols[c].type == SDB_COL_I64_FOR) {
+      // encoded blob: upload whole blob, point refs into it
+      SdbColHeaderDev hdr;
+      std::memcpy(&hdr, cols[c].data, sizeof(hdr));
+      if (hdr.magic != SDB_COL_MAGIC_DEV || hdr.rows != rows)
+        return SDB_ERR_INVALID;
+      if (tab->group_rows && tab->group_rows != hdr.group_rows)
+        return SDB_ERR_INVALID;  // FoR columns must share the group tiling
+      tab->group_rows = hdr.group_rows;
+      HIP_CHECK(hipMalloc(&tab->cols[c], hdr.size));
+      HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, hdr.size,
+                          hipMemcpyHostToDevice));
+      tab->refs[c].data = (const uint8_t*)tab->cols[c] + hdr.off_payload;
+      tab->refs[c].desc = (const SdbColGroupDescDev*)((const uint8_t*)
+                            tab->cols[c] + hdr.off_desc);
+    } else {
+      const size_t esz = cols[c].type == SDB_COL_I64 ? 8 : 4;
+      HIP_CHECK(hipMalloc(&tab->cols[c], esz * rows));
+      HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, esz * rows,
+                          hipMemcpyHostToDevice));
+      tab->refs[c].data = tab->cols[c];
+      tab->refs[c].desc = nullptr;
+    }
   }
+  if (tab->group_rows == 0) tab->group_rows = 65536;  // raw-only tiling
   *out = tab;
   return SDB_OK;
 }
@@ -217,32 +256,32 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
       ngroups == 0 || ngroups > SCAN_MAX_GROUPS || naggs == 0 ||
       naggs > SCAN_MAX_AGGS || npreds > SCAN_MAX_PREDS)
     return SDB_ERR_INVALID;
-  if (tab->types[group_col] != SDB_COL_I64) return SDB_ERR_INVALID;
+  if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
 
   hipStream_t stream = ctx->stream;
 
   ScanArgs a{};
-  a.keys = (const int64_t*)tab->cols[group_col];
+  a.keys = tab->refs[group_col];
   a.rows = tab->rows;
+  a.group_rows = tab->group_rows;
   a.ngroups = ngroups;
   a.naggs = naggs;
   a.npreds = npreds;
   for (uint32_t p = 0; p < npreds; ++p) {
     if (preds[p].col >= tab->ncols ||
-        tab->types[preds[p].col] != SDB_COL_I64)
+        tab->types[preds[p].col] == SDB_COL_F32)
       return SDB_ERR_INVALID;
-    a.pred_col[p] = (const int64_t*)tab->cols[preds[p].col];
+    a.pred_col[p] = tab->refs[preds[p].col];
     a.pred_op[p] = preds[p].op;
     a.pred_lo[p] = preds[p].ilo;
     a.pred_hi[p] = preds[p].ihi;
   }
   for (uint32_t q = 0; q < naggs; ++q) {
     a.agg_op[q] = aggs[q].op;
-    a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT
-                     ? nullptr
-                     : tab->cols[aggs[q].col];
+    a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT ? ColRef{nullptr, nullptr}
+                                               : tab->refs[aggs[q].col];
     if (aggs[q].op == SDB_AGG_SUM_I64 &&
-        tab->types[aggs[q].col] != SDB_COL_I64)
+        tab->types[aggs[q].col] == SDB_COL_F32)
       return SDB_ERR_INVALID;
     if (aggs[q].op == SDB_AGG_SUM_F64 &&
         tab->types[aggs[q].col] != SDB_COL_F32)
@@ -258,9 +297,8 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   a.out = d_out;
   a.rows_passed = d_passed;
 
-  // memory-bound grid sizing (guide §6 G11): cap ~8 blocks/CU, grid-stride
   uint32_t nblocks =
-    (uint32_t)((tab->rows / 2 + SCAN_NTHREADS - 1) / SCAN_NTHREADS);
+    (uint32_t)((tab->rows + a.group_rows - 1) / a.group_rows);
   if (nblocks > SCAN_MAXB) nblocks = SCAN_MAXB;
   if (nblocks < 1) nblocks = 1;
   const size_t lds = 8ull * nslots;

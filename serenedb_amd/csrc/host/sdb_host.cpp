@@ -478,6 +478,138 @@ static uint32_t decode_freq_block(const uint8_t* in, uint32_t len,
 }
 
 // ---------------------------------------------------------------------------
+// FoR/bitpack i64 column codec — this repo's own
+// layout (the reference's column codecs live in the un-vendored DuckDB
+// fork; parity is at result level). Per row group:
+// frame-of-reference base (min) + horizontal bitpack of (v - base) at the
+// group's required width, plus a min/max zonemap (BaseStatistics analogue,
+// column_reader.hpp:201) for whole-group predicate skips (DeadUntil,
+// full_scanner.h:61-71).
+//
+// Blob layout (64-bit aligned):
+//   SdbColHeader { u64 magic='SDBCOL01'; u64 rows; u32 group_rows;
+//                  u32 ngroups; u64 off_desc; u64 off_payload; u64 size; }
+//   SdbColGroupDesc[ngroups] { i64 base; i64 vmin; i64 vmax; u64 bit_off;
+//                              u16 width; u16 pad[3]; }
+//   payload: per group, ceil(group_rows*width/32) u32 words, packed
+//   little-endian, value i at bits [i*width, (i+1)*width).
+// ---------------------------------------------------------------------------
+extern "C" {
+typedef struct SdbColHeader {
+  uint64_t magic;
+  uint64_t rows;
+  uint32_t group_rows;
+  uint32_t ngroups;
+  uint64_t off_desc;
+  uint64_t off_payload;
+  uint64_t size;
+} SdbColHeader;
+typedef struct SdbColGroupDesc {
+  int64_t base;
+  int64_t vmin;
+  int64_t vmax;
+  uint64_t word_off; /* u32-word offset into payload */
+  uint16_t width;    /* bits per value, 0..32 (delta range must fit u32) */
+  uint16_t pad[3];
+} SdbColGroupDesc;
+}
+#define SDB_COL_MAGIC 0x31304C4F43424453ull
+
+extern "C" int sdb_host_encode_col_i64(const int64_t* vals, uint64_t rows,
+                                       uint32_t group_rows, void** blob_out,
+                                       uint64_t* blob_size) {
+  if (!vals || !blob_out || !blob_size || rows == 0 || group_rows == 0)
+    return -1;
+  const uint32_t ngroups = uint32_t((rows + group_rows - 1) / group_rows);
+  std::vector<SdbColGroupDesc> desc(ngroups);
+  std::vector<uint32_t> payload;
+  payload.reserve(rows / 2);
+  for (uint32_t g = 0; g < ngroups; ++g) {
+    const uint64_t r0 = uint64_t(g) * group_rows;
+    const uint64_t r1 = std::min<uint64_t>(rows, r0 + group_rows);
+    int64_t mn = vals[r0], mx = vals[r0];
+    for (uint64_t r = r0; r < r1; ++r) {
+      mn = std::min(mn, vals[r]);
+      mx = std::max(mx, vals[r]);
+    }
+    const uint64_t range = uint64_t(mx - mn);
+    if (range > 0xFFFFFFFFull) return -2; /* delta must fit u32 (round 1) */
+    uint32_t width = 0;
+    while (width < 33 && (width == 64 ? 0 : (range >> width)))
+      ++width; /* bit_width(range) */
+    SdbColGroupDesc d{};
+    d.base = mn;
+    d.vmin = mn;
+    d.vmax = mx;
+    d.width = uint16_t(width);
+    d.word_off = payload.size();
+    const uint64_t nw = (uint64_t(r1 - r0) * width + 31) / 32;
+    const size_t p0 = payload.size();
+    payload.resize(p0 + nw, 0u);
+    for (uint64_t r = r0; r < r1; ++r) {
+      const uint64_t v = uint64_t(vals[r] - mn);
+      const uint64_t bit = (r - r0) * width;
+      const uint64_t w = bit >> 5;
+      const uint32_t sh = uint32_t(bit & 31);
+      payload[p0 + w] |= uint32_t(v << sh);
+      if (sh + width > 32) payload[p0 + w + 1] |= uint32_t(v >> (32 - sh));
+    }
+    desc[g] = d;
+  }
+  auto align64 = [](uint64_t x) { return (x + 63) & ~63ull; };
+  SdbColHeader hdr{};
+  hdr.magic = SDB_COL_MAGIC;
+  hdr.rows = rows;
+  hdr.group_rows = group_rows;
+  hdr.ngroups = ngroups;
+  hdr.off_desc = align64(sizeof(SdbColHeader));
+  hdr.off_payload = align64(hdr.off_desc + sizeof(SdbColGroupDesc) * ngroups);
+  hdr.size = align64(hdr.off_payload + payload.size() * 4 + 64);
+  uint8_t* blob = static_cast<uint8_t*>(std::calloc(1, hdr.size));
+  if (!blob) return -4;
+  std::memcpy(blob, &hdr, sizeof(hdr));
+  std::memcpy(blob + hdr.off_desc, desc.data(),
+              sizeof(SdbColGroupDesc) * ngroups);
+  if (!payload.empty())
+    std::memcpy(blob + hdr.off_payload, payload.data(), payload.size() * 4);
+  *blob_out = blob;
+  *blob_size = hdr.size;
+  return 0;
+}
+
+// scalar decode (builder verification + oracle cross-check)
+extern "C" int sdb_host_decode_col_i64(const void* blob, uint64_t size,
+                                       int64_t* out, uint64_t rows) {
+  const SdbColHeader* hdr = (const SdbColHeader*)blob;
+  if (!hdr || hdr->magic != SDB_COL_MAGIC || hdr->rows != rows ||
+      hdr->size > size)
+    return -5;
+  const auto* desc =
+    (const SdbColGroupDesc*)((const uint8_t*)blob + hdr->off_desc);
+  const auto* pl = (const uint32_t*)((const uint8_t*)blob + hdr->off_payload);
+  for (uint32_t g = 0; g < hdr->ngroups; ++g) {
+    const uint64_t r0 = uint64_t(g) * hdr->group_rows;
+    const uint64_t r1 = std::min<uint64_t>(rows, r0 + hdr->group_rows);
+    const SdbColGroupDesc d = desc[g];
+    const uint32_t* w = pl + d.word_off;
+    const uint64_t mask =
+      d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
+    for (uint64_t r = r0; r < r1; ++r) {
+      if (d.width == 0) {
+        out[r] = d.base;
+        continue;
+      }
+      const uint64_t bit = (r - r0) * d.width;
+      uint64_t v = w[bit >> 5] >> (bit & 31);
+      if ((bit & 31) + d.width > 32)
+        v |= uint64_t(w[(bit >> 5) + 1]) << (32 - (bit & 31));
+      out[r] = d.base + int64_t(v & mask);
+    }
+  }
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
 // seeded synthetic corpus: stateless splitmix64 hashing so
 // any (term, doc) draw is order-independent and parallelizable.
 // ---------------------------------------------------------------------------

@@ -259,3 +259,66 @@ def test_match_docs_parity(ctx):
     assert total == ototal
     np.testing.assert_array_equal(docs, odocs)
     np.testing.assert_array_equal(vals, ovals)
+
+
+def test_scan_agg_for_codec_parity(ctx):
+    """FoR/bitpack columns + zonemap skips == oracle over the raw values.
+    Keys clustered so zonemaps actually kill row groups under a BETWEEN."""
+    import ctypes as CT
+
+    rows = 3_000_000
+    ngroups = 1024
+    rng = np.random.default_rng(46)
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = np.sort(rng.integers(0, 1 << 20, rows)).astype(np.int64)  # clustered
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    keys_blob = sa.encode_col_i64(keys)
+    v1_blob = sa.encode_col_i64(v1)
+
+    lib = sa.gpu()
+
+    class ColView(CT.Structure):
+        _fields_ = [("data", CT.c_void_p), ("rows", CT.c_uint64),
+                    ("type", CT.c_int)]
+
+    class PredSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int),
+                    ("ilo", CT.c_int64), ("ihi", CT.c_int64),
+                    ("flo", CT.c_float), ("fhi", CT.c_float)]
+
+    class AggSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int)]
+
+    class AggResult(CT.Structure):
+        _fields_ = [("i64", CT.c_int64), ("f64", CT.c_double)]
+
+    kb = np.frombuffer(keys_blob, dtype=np.uint8)
+    vb = np.frombuffer(v1_blob, dtype=np.uint8)
+    cols = (ColView * 3)(
+        ColView(kb.ctypes.data_as(CT.c_void_p).value, rows, 2),
+        ColView(vb.ctypes.data_as(CT.c_void_p).value, rows, 2),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, rows, 1))
+    tab = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols, 3, CT.c_uint64(rows),
+                                CT.byref(tab))
+    assert rc == 0, rc
+    # clustered v1 + BETWEEN in the middle: most groups zonemap-dead
+    lo_v = int((1 << 20) * 0.45)
+    hi_v = int((1 << 20) * 0.55)
+    preds = (PredSpec * 1)(PredSpec(1, 3, lo_v, hi_v, 0, 0))
+    aggs = (AggSpec * 3)(AggSpec(0, 0), AggSpec(1, 1), AggSpec(2, 2))
+    out = (AggResult * (ngroups * 3))()
+    passed = CT.c_uint64(0)
+    rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab, 0, ngroups, preds, 1, aggs, 3,
+                              out, CT.byref(passed))
+    assert rc == 0, rc
+    ocnt, osi, osf, opassed = po.scan_agg(keys, v1, v2, ngroups, pred_op=3,
+                                          lo=lo_v, hi=hi_v)
+    assert passed.value == opassed
+    gcnt = np.array([out[g * 3 + 0].i64 for g in range(ngroups)])
+    gsi = np.array([out[g * 3 + 1].i64 for g in range(ngroups)])
+    gsf = np.array([out[g * 3 + 2].f64 for g in range(ngroups)])
+    np.testing.assert_array_equal(gcnt, ocnt)
+    np.testing.assert_array_equal(gsi, osi)
+    np.testing.assert_allclose(gsf, osf, rtol=1e-7)
+    lib.sdb_gpu_table_free(ctx._ctx, tab)
