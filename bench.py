@@ -351,6 +351,15 @@ def bench_scan(args):
     keys = rng.integers(0, ngroups, my_rows).astype(np.int64)
     v1 = rng.integers(0, 1 << 20, my_rows).astype(np.int64)
     v2 = rng.normal(0, 1, my_rows).astype(np.float32)
+    use_for = args.scan_codec == "for"
+    if use_for:
+        keys_blob = sa.encode_col_i64(keys)
+        v1_blob = sa.encode_col_i64(v1)
+        kb = np.frombuffer(keys_blob, dtype=np.uint8)
+        vb = np.frombuffer(v1_blob, dtype=np.uint8)
+        scan_bytes = len(keys_blob) + len(v1_blob) + 4 * my_rows
+    else:
+        scan_bytes = 20 * my_rows
 
     device = int(os.environ.get("LOCAL_RANK", 0))
     ctx = sa.GpuContext(device)
@@ -371,10 +380,16 @@ def bench_scan(args):
     class AggResult(CT.Structure):
         _fields_ = [("i64", CT.c_int64), ("f64", CT.c_double)]
 
-    cols = (ColView * 3)(
-        ColView(keys.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
-        ColView(v1.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
-        ColView(v2.ctypes.data_as(CT.c_void_p).value, my_rows, 1))
+    if use_for:
+        cols = (ColView * 3)(
+            ColView(kb.ctypes.data_as(CT.c_void_p).value, my_rows, 2),
+            ColView(vb.ctypes.data_as(CT.c_void_p).value, my_rows, 2),
+            ColView(v2.ctypes.data_as(CT.c_void_p).value, my_rows, 1))
+    else:
+        cols = (ColView * 3)(
+            ColView(keys.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
+            ColView(v1.ctypes.data_as(CT.c_void_p).value, my_rows, 0),
+            ColView(v2.ctypes.data_as(CT.c_void_p).value, my_rows, 1))
     tab = CT.c_void_p(0)
     rc = lib.sdb_gpu_table_load(ctx._ctx, cols, 3, CT.c_uint64(my_rows),
                                 CT.byref(tab))
@@ -419,7 +434,7 @@ def bench_scan(args):
     ms_per_step = elapsed * 1000 / args.steps
     value = rows_total * args.steps / elapsed
 
-    algo_bytes = my_rows * 20  # 8B key + 8B v1 + 4B v2, no zonemap skips
+    algo_bytes = scan_bytes  # compressed (FoR) or raw column bytes scanned
     achieved_gbs = algo_bytes * args.steps / elapsed / 1e9  # whole step ~ kernel
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
@@ -459,6 +474,8 @@ def bench_scan(args):
             "ngroups": ngroups,
             "predicate": "v1 < 10% quantile",
             "aggs": "COUNT, SUM(i64), SUM(f32->f64)",
+            "codec": args.scan_codec,
+            "bytes_per_row": round(scan_bytes / my_rows, 2),
             "seed": seed,
             "parallelism": f"row shards x{world}, RCCL allreduce merge",
         },
@@ -488,6 +505,7 @@ def main():
                     choices=["bm25_topk", "scan_agg", "hybrid"])
     ap.add_argument("--docs", type=int, default=100_000_000)
     ap.add_argument("--rows", type=int, default=1_000_000_000)
+    ap.add_argument("--scan-codec", default="for", choices=["for", "raw"])
     ap.add_argument("--cpu-seconds", type=float, default=10.0)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
