@@ -93,14 +93,19 @@ struct ScanArgs {
   int64_t pred_hi[SCAN_MAX_PREDS];
   ColRef agg_col[SCAN_MAX_AGGS];
   int agg_op[SCAN_MAX_AGGS];
+  int agg_src[SCAN_MAX_AGGS];  // 0=own col_read, 1+p=pred p's value, 9=key
   unsigned long long* out;
   unsigned long long* rows_passed;
 };
 
-__device__ __forceinline__ int64_t col_read(const ColRef& c, uint64_t r0,
-                                            uint64_t r,
-                                            const SdbColGroupDescDev& d) {
+// desc fields are read per call from desc[rg]: the address is wave-uniform
+// so the loads scalarize/broadcast from cache (a register-cached desc array
+// indexed by a runtime agg index would spill to scratch — guide rule 20:
+// that variant measured 2x slower than the raw path)
+__device__ __forceinline__ int64_t col_read(const ColRef& c, uint32_t rg,
+                                            uint64_t r0, uint64_t r) {
   if (!c.desc) return ((const long long*)c.data)[r];
+  const SdbColGroupDescDev& d = c.desc[rg];
   if (d.width == 0) return d.base;
   const uint32_t* w = (const uint32_t*)c.data + d.word_off;
   const uint64_t bit = (r - r0) * d.width;
@@ -122,36 +127,38 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   const uint32_t n_rowgroups =
     (uint32_t)((a.rows + a.group_rows - 1) / a.group_rows);
   unsigned long long my_passed = 0;
-  SdbColGroupDescDev kd{}, pd[SCAN_MAX_PREDS], ad[SCAN_MAX_AGGS];
 
   for (uint32_t rg = blockIdx.x; rg < n_rowgroups; rg += gridDim.x) {
     const uint64_t r0 = (uint64_t)rg * a.group_rows;
     const uint64_t r1 = min(a.rows, r0 + a.group_rows);
-    // group descriptors + zonemap skip (DeadUntil analogue)
+    // zonemap skip (DeadUntil analogue)
     bool dead = false;
     for (uint32_t p = 0; p < a.npreds; ++p) {
       if (a.pred_col[p].desc) {
-        pd[p] = a.pred_col[p].desc[rg];
+        const SdbColGroupDescDev& d = a.pred_col[p].desc[rg];
         switch (a.pred_op[p]) {
-          case SDB_PRED_LT: dead |= pd[p].vmin >= a.pred_lo[p]; break;
-          case SDB_PRED_GE: dead |= pd[p].vmax < a.pred_lo[p]; break;
+          case SDB_PRED_LT: dead |= d.vmin >= a.pred_lo[p]; break;
+          case SDB_PRED_GE: dead |= d.vmax < a.pred_lo[p]; break;
           case SDB_PRED_BETWEEN:
-            dead |= (pd[p].vmax < a.pred_lo[p]) |
-                    (pd[p].vmin > a.pred_hi[p]);
+            dead |= (d.vmax < a.pred_lo[p]) | (d.vmin > a.pred_hi[p]);
             break;
           default: break;
         }
       }
     }
     if (dead) continue;
-    if (a.keys.desc) kd = a.keys.desc[rg];
-    for (uint32_t q = 0; q < a.naggs; ++q)
-      if (a.agg_col[q].desc) ad[q] = a.agg_col[q].desc[rg];
 
     for (uint64_t r = r0 + threadIdx.x; r < r1; r += SCAN_NTHREADS) {
       bool ok = true;
-      for (uint32_t p = 0; p < a.npreds; ++p) {
-        const int64_t x = col_read(a.pred_col[p], r0, r, pd[p]);
+      int64_t pv0 = 0, pv1 = 0, pv2 = 0, pv3 = 0;
+#pragma unroll
+      for (uint32_t p = 0; p < SCAN_MAX_PREDS; ++p) {
+        if (p >= a.npreds) break;
+        const int64_t x = col_read(a.pred_col[p], rg, r0, r);
+        if (p == 0) pv0 = x;
+        else if (p == 1) pv1 = x;
+        else if (p == 2) pv2 = x;
+        else pv3 = x;
         switch (a.pred_op[p]) {
           case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
           case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
@@ -163,17 +170,26 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
       }
       if (!ok) continue;
       ++my_passed;
-      const uint32_t grp = (uint32_t)col_read(a.keys, r0, r, kd);
+      const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
       for (uint32_t q = 0; q < a.naggs; ++q) {
         unsigned long long* slot = &acc[grp * a.naggs + q];
         switch (a.agg_op[q]) {
           case SDB_AGG_COUNT:
             atomicAdd(slot, 1ull);
             break;
-          case SDB_AGG_SUM_I64:
-            atomicAdd(slot, (unsigned long long)col_read(a.agg_col[q], r0,
-                                                         r, ad[q]));
+          case SDB_AGG_SUM_I64: {
+            int64_t x;
+            switch (a.agg_src[q]) {  // decode-once dedup vs preds/key
+              case 1: x = pv0; break;
+              case 2: x = pv1; break;
+              case 3: x = pv2; break;
+              case 4: x = pv3; break;
+              case 9: x = (int64_t)grp; break;
+              default: x = col_read(a.agg_col[q], rg, r0, r); break;
+            }
+            atomicAdd(slot, (unsigned long long)x);
             break;
+          }
           case SDB_AGG_SUM_F64:
             atomicAdd((double*)slot,
                       (double)((const float*)a.agg_col[q].data)[r]);
@@ -280,6 +296,12 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
     a.agg_op[q] = aggs[q].op;
     a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT ? ColRef{nullptr, nullptr}
                                                : tab->refs[aggs[q].col];
+    a.agg_src[q] = 0;
+    if (aggs[q].op == SDB_AGG_SUM_I64) {
+      if (aggs[q].col == group_col) a.agg_src[q] = 9;
+      for (uint32_t p = 0; p < npreds; ++p)
+        if (aggs[q].col == preds[p].col) a.agg_src[q] = 1 + (int)p;
+    }
     if (aggs[q].op == SDB_AGG_SUM_I64 &&
         tab->types[aggs[q].col] == SDB_COL_F32)
       return SDB_ERR_INVALID;
