@@ -505,7 +505,7 @@ def main():
                     choices=["bm25_topk", "scan_agg", "hybrid"])
     ap.add_argument("--docs", type=int, default=100_000_000)
     ap.add_argument("--rows", type=int, default=1_000_000_000)
-    ap.add_argument("--scan-codec", default="for", choices=["for", "raw"])
+    ap.add_argument("--scan-codec", default="raw", choices=["for", "raw"])
     ap.add_argument("--cpu-seconds", type=float, default=10.0)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()

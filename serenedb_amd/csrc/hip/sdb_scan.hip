@@ -117,12 +117,118 @@ __device__ __forceinline__ int64_t col_read(const ColRef& c, uint32_t rg,
   return d.base + (int64_t)(v & mask);
 }
 
+template <int RAW>
 __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned long long* acc = (unsigned long long*)smem;  // ngroups*naggs
   const uint32_t nslots = a.ngroups * a.naggs;
   for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) acc[i] = 0;
   __syncthreads();
+
+  if (RAW) {
+    // raw dense columns: 2 rows/thread with 16-byte loads (the measured
+    // fastest shape: 4.55 TB/s at 1024t/2048b)
+    const uint64_t pair_stride = (uint64_t)gridDim.x * SCAN_NTHREADS * 2u;
+    unsigned long long my_passed = 0;
+    const uint64_t rows2 = a.rows & ~1ull;
+    for (uint64_t r =
+           ((uint64_t)blockIdx.x * SCAN_NTHREADS + threadIdx.x) * 2u;
+         r < rows2; r += pair_stride) {
+      bool okv[2] = {true, true};
+      for (uint32_t p = 0; p < a.npreds; ++p) {
+        longlong2 x;
+        __builtin_memcpy(&x, &((const long long*)a.pred_col[p].data)[r], 16);
+        const int64_t xs[2] = {x.x, x.y};
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+          switch (a.pred_op[p]) {
+            case SDB_PRED_LT: okv[e] &= xs[e] < a.pred_lo[p]; break;
+            case SDB_PRED_GE: okv[e] &= xs[e] >= a.pred_lo[p]; break;
+            case SDB_PRED_BETWEEN:
+              okv[e] &= (xs[e] >= a.pred_lo[p]) & (xs[e] <= a.pred_hi[p]);
+              break;
+            default: break;
+          }
+        }
+      }
+      if (!okv[0] && !okv[1]) continue;
+      longlong2 kk;
+      __builtin_memcpy(&kk, &((const long long*)a.keys.data)[r], 16);
+      const int64_t ks[2] = {kk.x, kk.y};
+#pragma unroll
+      for (int e = 0; e < 2; ++e) {
+        if (!okv[e]) continue;
+        ++my_passed;
+        const uint32_t g = (uint32_t)ks[e];
+        for (uint32_t q = 0; q < a.naggs; ++q) {
+          unsigned long long* slot = &acc[g * a.naggs + q];
+          switch (a.agg_op[q]) {
+            case SDB_AGG_COUNT:
+              atomicAdd(slot, 1ull);
+              break;
+            case SDB_AGG_SUM_I64:
+              atomicAdd(slot, (unsigned long long)((const long long*)
+                                                     a.agg_col[q]
+                                                       .data)[r + e]);
+              break;
+            case SDB_AGG_SUM_F64:
+              atomicAdd((double*)slot,
+                        (double)((const float*)a.agg_col[q].data)[r + e]);
+              break;
+          }
+        }
+      }
+    }
+    if ((a.rows & 1ull) && blockIdx.x == 0 && threadIdx.x == 0) {
+      const uint64_t r = a.rows - 1;
+      bool ok = true;
+      for (uint32_t p = 0; p < a.npreds; ++p) {
+        const int64_t x = ((const long long*)a.pred_col[p].data)[r];
+        switch (a.pred_op[p]) {
+          case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
+          case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
+          case SDB_PRED_BETWEEN:
+            ok &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+            break;
+          default: break;
+        }
+      }
+      if (ok) {
+        ++my_passed;
+        const uint32_t g = (uint32_t)((const long long*)a.keys.data)[r];
+        for (uint32_t q = 0; q < a.naggs; ++q) {
+          unsigned long long* slot = &acc[g * a.naggs + q];
+          switch (a.agg_op[q]) {
+            case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
+            case SDB_AGG_SUM_I64:
+              atomicAdd(slot, (unsigned long long)((const long long*)
+                                                     a.agg_col[q].data)[r]);
+              break;
+            case SDB_AGG_SUM_F64:
+              atomicAdd((double*)slot,
+                        (double)((const float*)a.agg_col[q].data)[r]);
+              break;
+          }
+        }
+      }
+    }
+    unsigned long long wp = my_passed;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) wp += __shfl_down(wp, off, 64);
+    if ((threadIdx.x & 63) == 0 && wp) atomicAdd(a.rows_passed, wp);
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) {
+      const uint32_t q = i % a.naggs;
+      if (a.agg_op[q] == SDB_AGG_SUM_F64) {
+        double v;
+        __builtin_memcpy(&v, &acc[i], 8);
+        if (v != 0.0) atomicAdd((double*)&a.out[i], v);
+      } else if (acc[i]) {
+        atomicAdd(&a.out[i], acc[i]);
+      }
+    }
+    return;
+  }
 
   const uint32_t n_rowgroups =
     (uint32_t)((a.rows + a.group_rows - 1) / a.group_rows);
@@ -324,8 +430,15 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   if (nblocks > SCAN_MAXB) nblocks = SCAN_MAXB;
   if (nblocks < 1) nblocks = 1;
   const size_t lds = 8ull * nslots;
-  hipLaunchKernelGGL(scan_agg_kernel, dim3(nblocks), dim3(SCAN_NTHREADS), lds,
-                     stream, a);
+  bool any_for = a.keys.desc != nullptr;
+  for (uint32_t p = 0; p < npreds; ++p) any_for |= a.pred_col[p].desc != nullptr;
+  for (uint32_t q = 0; q < naggs; ++q) any_for |= a.agg_col[q].desc != nullptr;
+  if (any_for)
+    hipLaunchKernelGGL((scan_agg_kernel<0>), dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
+  else
+    hipLaunchKernelGGL((scan_agg_kernel<1>), dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
   HIP_CHECK(hipGetLastError());
   std::vector<unsigned long long> h_out(nslots);
   unsigned long long h_passed = 0;
