@@ -353,3 +353,27 @@ def execute_match_docs(blob, term_idx, boosts, cap, col=None, min_match=1,
     assert rc == 0, rc
     n = out_n.value
     return docs[:n], (col_out[:n] if col is not None else None), total.value
+
+
+def scan_agg_mt(keys, v1, v2, ngroups, pred_op=0, lo=0, hi=0, nthreads=0,
+                iters=1):
+    """Multithreaded scan baseline (timed CPU leg; deterministic merge)."""
+    if nthreads == 0:
+        nthreads = os.cpu_count() or 1
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    v1 = np.ascontiguousarray(v1, dtype=np.int64)
+    v2 = np.ascontiguousarray(v2, dtype=np.float32)
+    cnt = np.zeros(ngroups, dtype=np.int64)
+    si = np.zeros(ngroups, dtype=np.int64)
+    sf = np.zeros(ngroups, dtype=np.float64)
+    passed = C.c_uint64(0)
+    PI64 = C.POINTER(C.c_int64)
+    rc = lib().o_scan_agg_mt(
+        keys.ctypes.data_as(PI64), v1.ctypes.data_as(PI64),
+        v2.ctypes.data_as(C.POINTER(C.c_float)), C.c_uint64(len(keys)),
+        C.c_uint32(ngroups), C.c_int(pred_op), C.c_int64(lo), C.c_int64(hi),
+        C.c_uint32(nthreads), C.c_uint32(iters),
+        cnt.ctypes.data_as(PI64), si.ctypes.data_as(PI64),
+        sf.ctypes.data_as(C.POINTER(C.c_double)), C.byref(passed))
+    assert rc == 0, rc
+    return cnt, si, sf, passed.value

@@ -1080,6 +1080,103 @@ int o_execute_topk_mt(const void* blob, uint64_t size,
 /*  SURVEY.md §8c). COUNT/SUM(i64) exact; SUM(f32) accumulated in f64  */
 /*  sequentially (deterministic oracle order).                         */
 /* ------------------------------------------------------------------ */
+/* multithreaded scan baseline: row chunks over a thread pool, per-thread
+ * group arrays merged in thread-index order (deterministic i64; f64 merge
+ * order fixed). iters repeats the scan inside the pool (thread-spawn
+ * amortization, as in the top-k baseline). */
+typedef struct {
+  const int64_t* keys;
+  const int64_t* v1;
+  const float* v2;
+  uint64_t rows;
+  uint32_t ngroups;
+  int pred_op;
+  int64_t lo, hi;
+  uint32_t nthreads, thread_id, iters;
+  pthread_barrier_t* barrier;
+  int64_t* cnt; /* per-thread arrays [ngroups] */
+  int64_t* si;
+  double* sf;
+  uint64_t passed;
+} OScanArg;
+
+static void* o_scan_worker(void* argp) {
+  OScanArg* a = (OScanArg*)argp;
+  for (uint32_t it = 0; it < a->iters; ++it) {
+    pthread_barrier_wait(a->barrier);
+    memset(a->cnt, 0, sizeof(int64_t) * a->ngroups);
+    memset(a->si, 0, sizeof(int64_t) * a->ngroups);
+    memset(a->sf, 0, sizeof(double) * a->ngroups);
+    a->passed = 0;
+    const uint64_t per = (a->rows + a->nthreads - 1) / a->nthreads;
+    const uint64_t r0 = (uint64_t)a->thread_id * per;
+    const uint64_t r1 = r0 + per < a->rows ? r0 + per : a->rows;
+    for (uint64_t r = r0; r < r1; ++r) {
+      const int64_t x = a->v1[r];
+      int ok;
+      switch (a->pred_op) {
+        case 1: ok = x < a->lo; break;
+        case 2: ok = x >= a->lo; break;
+        case 3: ok = x >= a->lo && x <= a->hi; break;
+        default: ok = 1; break;
+      }
+      if (!ok) continue;
+      ++a->passed;
+      const uint32_t g = (uint32_t)a->keys[r];
+      a->cnt[g] += 1;
+      a->si[g] += x;
+      a->sf[g] += (double)a->v2[r];
+    }
+    pthread_barrier_wait(a->barrier);
+  }
+  return NULL;
+}
+
+int o_scan_agg_mt(const int64_t* keys, const int64_t* v1, const float* v2,
+                  uint64_t rows, uint32_t ngroups, int pred_op, int64_t lo,
+                  int64_t hi, uint32_t nthreads, uint32_t iters,
+                  int64_t* out_count, int64_t* out_sum_i64,
+                  double* out_sum_f64, uint64_t* rows_passed) {
+  if (nthreads == 0) nthreads = 1;
+  if (iters == 0) iters = 1;
+  OScanArg* args = (OScanArg*)calloc(nthreads, sizeof(OScanArg));
+  pthread_t* th = (pthread_t*)malloc(sizeof(pthread_t) * nthreads);
+  int64_t* cnt = (int64_t*)malloc(8ull * ngroups * nthreads);
+  int64_t* si = (int64_t*)malloc(8ull * ngroups * nthreads);
+  double* sf = (double*)malloc(8ull * ngroups * nthreads);
+  pthread_barrier_t barrier;
+  pthread_barrier_init(&barrier, NULL, nthreads);
+  for (uint32_t t = 0; t < nthreads; ++t) {
+    args[t] = (OScanArg){keys, v1, v2, rows, ngroups, pred_op, lo, hi,
+                         nthreads, t, iters, &barrier,
+                         cnt + (uint64_t)t * ngroups,
+                         si + (uint64_t)t * ngroups,
+                         sf + (uint64_t)t * ngroups, 0};
+    pthread_create(&th[t], NULL, o_scan_worker, &args[t]);
+  }
+  memset(out_count, 0, 8ull * ngroups);
+  memset(out_sum_i64, 0, 8ull * ngroups);
+  memset(out_sum_f64, 0, 8ull * ngroups);
+  uint64_t passed = 0;
+  for (uint32_t t = 0; t < nthreads; ++t) {
+    pthread_join(th[t], NULL);
+    passed += args[t].passed;
+    for (uint32_t g = 0; g < ngroups; ++g) {
+      out_count[g] += args[t].cnt[g];
+      out_sum_i64[g] += args[t].si[g];
+      out_sum_f64[g] += args[t].sf[g];
+    }
+  }
+  *rows_passed = passed;
+  pthread_barrier_destroy(&barrier);
+  free(args);
+  free(th);
+  free(cnt);
+  free(si);
+  free(sf);
+  return 0;
+}
+
 int o_scan_agg(const int64_t* keys, const int64_t* v1, const float* v2,
                uint64_t rows, uint32_t ngroups, int pred_op, int64_t lo,
                int64_t hi, int64_t* out_count, int64_t* out_sum_i64,

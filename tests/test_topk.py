@@ -309,3 +309,19 @@ def test_match_docs_streaming():
     assert total == len(exp)
     np.testing.assert_array_equal(docs, exp)
     np.testing.assert_array_equal(vals, col[exp])
+
+
+def test_scan_agg_mt_equals_single():
+    rng = np.random.default_rng(71)
+    rows, ngroups = 500_000, 64
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(0, 1 << 20, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    c1, s1, f1, p1 = po.scan_agg(keys, v1, v2, ngroups, pred_op=1,
+                                 lo=100000)
+    c2, s2, f2, p2 = po.scan_agg_mt(keys, v1, v2, ngroups, pred_op=1,
+                                    lo=100000, nthreads=4, iters=2)
+    assert p1 == p2
+    np.testing.assert_array_equal(c1, c2)
+    np.testing.assert_array_equal(s1, s2)
+    np.testing.assert_allclose(f1, f2, rtol=1e-12)
