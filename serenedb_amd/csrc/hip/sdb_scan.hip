@@ -117,6 +117,42 @@ __device__ __forceinline__ int64_t col_read(const ColRef& c, uint32_t rg,
   return d.base + (int64_t)(v & mask);
 }
 
+// paired FoR extraction: values r and r+1 from <=3 payload words
+__device__ __forceinline__ void col_read2(const ColRef& c, uint32_t rg,
+                                          uint64_t r0, uint64_t r,
+                                          int64_t& x0, int64_t& x1) {
+  if (!c.desc) {
+    longlong2 x;
+    __builtin_memcpy(&x, &((const long long*)c.data)[r], 16);
+    x0 = x.x;
+    x1 = x.y;
+    return;
+  }
+  const SdbColGroupDescDev& d = c.desc[rg];
+  if (d.width == 0) {
+    x0 = x1 = d.base;
+    return;
+  }
+  const uint32_t* w = (const uint32_t*)c.data + d.word_off;
+  const uint64_t bit0 = (r - r0) * d.width;
+  const uint64_t w0i = bit0 >> 5;
+  const uint32_t w0 = w[w0i], w1 = w[w0i + 1], w2 = w[w0i + 2];
+  const uint64_t mask =
+    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
+  const uint32_t sh0 = (uint32_t)(bit0 & 31);
+  uint64_t v0 = (uint64_t)w0 >> sh0;
+  if (sh0 + d.width > 32) v0 |= (uint64_t)w1 << (32 - sh0);
+  const uint64_t bit1 = bit0 + d.width;
+  const uint32_t rel = (uint32_t)((bit1 >> 5) - w0i);  // 0, 1 or 2
+  const uint32_t wa = rel == 0 ? w0 : (rel == 1 ? w1 : w2);
+  const uint32_t wb = rel == 0 ? w1 : w2;
+  const uint32_t sh1 = (uint32_t)(bit1 & 31);
+  uint64_t v1 = (uint64_t)wa >> sh1;
+  if (sh1 + d.width > 32) v1 |= (uint64_t)wb << (32 - sh1);
+  x0 = d.base + (int64_t)(v0 & mask);
+  x1 = d.base + (int64_t)(v1 & mask);
+}
+
 template <int RAW>
 __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -254,17 +290,89 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
     }
     if (dead) continue;
 
-    for (uint64_t r = r0 + threadIdx.x; r < r1; r += SCAN_NTHREADS) {
-      bool ok = true;
-      int64_t pv0 = 0, pv1 = 0, pv2 = 0, pv3 = 0;
+    // 2 rows per thread, paired word extraction; odd tail row by thread 0
+    const uint64_t glen = r1 - r0;
+    const uint64_t gpairs = glen >> 1;
+    for (uint64_t pr = threadIdx.x; pr < gpairs; pr += SCAN_NTHREADS) {
+      const uint64_t r = r0 + 2 * pr;
+      bool okv[2] = {true, true};
+      int64_t pva[2] = {0, 0}, pvb[2] = {0, 0};
 #pragma unroll
-      for (uint32_t p = 0; p < SCAN_MAX_PREDS; ++p) {
+      for (uint32_t p = 0; p < 2; ++p) {  // fast path: first two preds
         if (p >= a.npreds) break;
+        int64_t x0, x1;
+        col_read2(a.pred_col[p], rg, r0, r, x0, x1);
+        if (p == 0) { pva[0] = x0; pva[1] = x1; }
+        else { pvb[0] = x0; pvb[1] = x1; }
+        const int64_t xs[2] = {x0, x1};
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+          switch (a.pred_op[p]) {
+            case SDB_PRED_LT: okv[e] &= xs[e] < a.pred_lo[p]; break;
+            case SDB_PRED_GE: okv[e] &= xs[e] >= a.pred_lo[p]; break;
+            case SDB_PRED_BETWEEN:
+              okv[e] &= (xs[e] >= a.pred_lo[p]) & (xs[e] <= a.pred_hi[p]);
+              break;
+            default: break;
+          }
+        }
+      }
+      for (uint32_t p = 2; p < a.npreds; ++p) {  // rare: >2 predicates
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+          const int64_t x = col_read(a.pred_col[p], rg, r0, r + e);
+          switch (a.pred_op[p]) {
+            case SDB_PRED_LT: okv[e] &= x < a.pred_lo[p]; break;
+            case SDB_PRED_GE: okv[e] &= x >= a.pred_lo[p]; break;
+            case SDB_PRED_BETWEEN:
+              okv[e] &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+              break;
+            default: break;
+          }
+        }
+      }
+      if (!okv[0] && !okv[1]) continue;
+      int64_t k0, k1;
+      col_read2(a.keys, rg, r0, r, k0, k1);
+      const int64_t ks[2] = {k0, k1};
+#pragma unroll
+      for (int e = 0; e < 2; ++e) {
+        if (!okv[e]) continue;
+        ++my_passed;
+        const uint32_t grp = (uint32_t)ks[e];
+        for (uint32_t q = 0; q < a.naggs; ++q) {
+          unsigned long long* slot = &acc[grp * a.naggs + q];
+          switch (a.agg_op[q]) {
+            case SDB_AGG_COUNT:
+              atomicAdd(slot, 1ull);
+              break;
+            case SDB_AGG_SUM_I64: {
+              int64_t x;
+              switch (a.agg_src[q]) {  // decode-once dedup vs preds/key
+                case 1: x = pva[e]; break;
+                case 2: x = pvb[e]; break;
+                case 9: x = ks[e]; break;
+                default: x = col_read(a.agg_col[q], rg, r0, r + e); break;
+              }
+              atomicAdd(slot, (unsigned long long)x);
+              break;
+            }
+            case SDB_AGG_SUM_F64:
+              atomicAdd((double*)slot,
+                        (double)((const float*)a.agg_col[q].data)[r + e]);
+              break;
+          }
+        }
+      }
+    }
+    if ((glen & 1ull) && threadIdx.x == 0) {
+      const uint64_t r = r1 - 1;
+      bool ok = true;
+      int64_t pv0s = 0, pv1s = 0;
+      for (uint32_t p = 0; p < a.npreds; ++p) {
         const int64_t x = col_read(a.pred_col[p], rg, r0, r);
-        if (p == 0) pv0 = x;
-        else if (p == 1) pv1 = x;
-        else if (p == 2) pv2 = x;
-        else pv3 = x;
+        if (p == 0) pv0s = x;
+        else if (p == 1) pv1s = x;
         switch (a.pred_op[p]) {
           case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
           case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
@@ -274,32 +382,29 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
           default: break;
         }
       }
-      if (!ok) continue;
-      ++my_passed;
-      const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
-      for (uint32_t q = 0; q < a.naggs; ++q) {
-        unsigned long long* slot = &acc[grp * a.naggs + q];
-        switch (a.agg_op[q]) {
-          case SDB_AGG_COUNT:
-            atomicAdd(slot, 1ull);
-            break;
-          case SDB_AGG_SUM_I64: {
-            int64_t x;
-            switch (a.agg_src[q]) {  // decode-once dedup vs preds/key
-              case 1: x = pv0; break;
-              case 2: x = pv1; break;
-              case 3: x = pv2; break;
-              case 4: x = pv3; break;
-              case 9: x = (int64_t)grp; break;
-              default: x = col_read(a.agg_col[q], rg, r0, r); break;
+      if (ok) {
+        ++my_passed;
+        const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
+        for (uint32_t q = 0; q < a.naggs; ++q) {
+          unsigned long long* slot = &acc[grp * a.naggs + q];
+          switch (a.agg_op[q]) {
+            case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
+            case SDB_AGG_SUM_I64: {
+              int64_t x;
+              switch (a.agg_src[q]) {
+                case 1: x = pv0s; break;
+                case 2: x = pv1s; break;
+                case 9: x = (int64_t)grp; break;
+                default: x = col_read(a.agg_col[q], rg, r0, r); break;
+              }
+              atomicAdd(slot, (unsigned long long)x);
+              break;
             }
-            atomicAdd(slot, (unsigned long long)x);
-            break;
+            case SDB_AGG_SUM_F64:
+              atomicAdd((double*)slot,
+                        (double)((const float*)a.agg_col[q].data)[r]);
+              break;
           }
-          case SDB_AGG_SUM_F64:
-            atomicAdd((double*)slot,
-                      (double)((const float*)a.agg_col[q].data)[r]);
-            break;
         }
       }
     }
