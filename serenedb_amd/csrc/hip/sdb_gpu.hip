@@ -51,7 +51,9 @@
 #ifndef SDB_WIN_DOCS
 #define SDB_WIN_DOCS 24576u  // docs per workgroup window (96 KB f32 + 24 KB u8)
 #endif
-#define SDB_NTHREADS 512u    // 8 waves
+#ifndef SDB_NTHREADS
+#define SDB_NTHREADS 512u  // 8 waves (overridable for sweeps)
+#endif
 #define SDB_NWAVES (SDB_NTHREADS / 64u)
 #define SDB_MAX_TERMS 32u
 #define SDB_HIST_BINS 256u
