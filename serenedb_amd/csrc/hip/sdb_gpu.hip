@@ -52,8 +52,8 @@
 #define SDB_WIN_DOCS 24576u  // docs per workgroup window (96 KB f32 + 24 KB u8)
 #endif
 #ifndef SDB_NTHREADS
-#define SDB_NTHREADS 512u  // 8 waves (overridable for sweeps)
-#endif
+#define SDB_NTHREADS 1024u  // 16 waves = 4/SIMD: swept 0.51 ms vs 0.66 ms
+#endif                      // at 512 threads (more TLP per phase)
 #define SDB_NWAVES (SDB_NTHREADS / 64u)
 #define SDB_MAX_TERMS 32u
 #define SDB_HIST_BINS 256u
