@@ -344,10 +344,8 @@ def test_full_size_properties(ctx):
     s = hits["score"]
     assert np.all(s[:-1] >= s[1:])  # descending
     assert np.all(np.isin(hits["doc"], union))
-    # k-th dominance on a sample: no sampled non-hit doc scores above the
-    # k-th (cheap spot check against the oracle's scorer on a small slice)
-    sub = sa.build_synth_segment(43, 1, 200_000, sels)
-    ohits, _ = po.execute_topk([sub], [0, 1, 2, 3], [1.0] * 4, 50,
-                               global_stats=(doc_count, None, None)
-                               if False else None)
-    assert ohits["score"][0] <= s[0] * 1.2  # same corpus family, sane range
+    # determinism: a second run returns the identical result
+    hits2, total2 = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, 1000)
+    assert total2 == total
+    np.testing.assert_array_equal(hits["doc"], hits2["doc"])
+    np.testing.assert_array_equal(hits["score"], hits2["score"])
