@@ -90,12 +90,20 @@ typedef struct SdbTermRef {
   float boost;
 } SdbTermRef;
 
+typedef enum SdbScorerType {
+  SDB_SCORER_BM25 = 0,       /* search/bm25.cpp:89-109 */
+  SDB_SCORER_TFIDF = 1,      /* search/tfidf.cpp:60-62: idf*sqrt(freq) */
+  SDB_SCORER_TFIDF_NORM = 2, /* tfidf.cpp:72-76: idf*sqrt(freq)/sqrt(norm) */
+} SdbScorerType;
+
 typedef struct SdbQueryPlan {
   const SdbTermRef* terms;
   uint32_t nterms;
   uint32_t min_match;
   float k1; /* BM25 k (search/bm25.hpp:62: default 1.2) */
   float b;  /* BM25 b (search/bm25.hpp:64: default 0.75) */
+  uint32_t scorer; /* SdbScorerType; TFIDF idf = log1p((N+1)/(df+1)),
+                      tfidf.cpp:148-151 */
   /* Optional GLOBAL BM25 stats for sharded (multi-rank) execution, where
    * each device holds only its shard: the cross-rank stats merge is the
    * PreparePhase barrier analogue (duckdb_search_full_scan.cpp:1359-1384)

@@ -144,9 +144,13 @@ def _hits_to_np(hits, n):
     return res
 
 
+SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
+
+
 def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
-                 global_stats=None):
+                 global_stats=None, scorer="bm25"):
     """EXACT top-k (parity oracle). blobs: list of segment blob bytes."""
+    lib().o_set_scorer(C.c_uint32(SCORERS[scorer]))
     arr, keep = _mkblobs(blobs)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)
@@ -173,6 +177,7 @@ def execute_topk_mech(blobs, term_idx, boosts, k, min_match=1, k1=1.2,
                       b=0.75, global_stats=None):
     """2k-buffer/nth_element mechanics emulation (reference fixture
     validation; single-thread timed CPU path)."""
+    lib().o_set_scorer(0)  # these entry points are BM25-only
     arr, keep = _mkblobs(blobs)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)
@@ -200,6 +205,7 @@ def execute_topk_mt(blob, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
     """Multithreaded mechanics baseline over one segment (timed CPU leg).
     iters > 1 repeats the whole query inside the thread pool (thread
     creation would otherwise dominate sub-ms queries)."""
+    lib().o_set_scorer(0)  # these entry points are BM25-only
     if nthreads == 0:
         nthreads = os.cpu_count() or 1
     buf = np.frombuffer(blob, dtype=np.uint8)
@@ -298,6 +304,7 @@ def execute_topk_hybrid(blob, term_idx, boosts, k, col, flo, fhi, nbuckets,
                         min_match=1, k1=1.2, b=0.75, global_stats=None):
     """Hybrid exact top-k: BM25 match AND col BETWEEN [flo,fhi], plus
     per-bucket COUNT/SUM over surviving matches."""
+    lib().o_set_scorer(0)  # these entry points are BM25-only
     buf = np.frombuffer(blob, dtype=np.uint8)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)
@@ -330,6 +337,7 @@ def execute_topk_hybrid(blob, term_idx, boosts, k, col, flo, fhi, nbuckets,
 def execute_match_docs(blob, term_idx, boosts, cap, col=None, min_match=1,
                        k1=1.2, b=0.75):
     """Streaming match emission (doc-ascending) + optional column gather."""
+    lib().o_set_scorer(0)  # these entry points are BM25-only
     buf = np.frombuffer(blob, dtype=np.uint8)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)

@@ -411,10 +411,13 @@ void o_bm25_stats(uint64_t docs_with_field, uint64_t docs_with_term,
   }
 }
 
-/* Bm25 score kernel (bm25.cpp:89-109): num precomputed boost*(k+1)*idf */
-static inline float o_bm25_score(float num, float norm_const,
-                                 float norm_length, uint32_t freq,
-                                 uint32_t norm) {
+/* per-(doc,term) score — BM25 (bm25.cpp:89-109, num = boost*(k+1)*idf) or
+ * TFIDF (tfidf.cpp:60-76, num = boost*idf) */
+static inline float o_score_one(uint32_t scorer, float num, float norm_const,
+                                float norm_length, uint32_t freq,
+                                uint32_t norm) {
+  if (scorer == 1) return num * sqrtf((float)freq);
+  if (scorer == 2) return num * sqrtf((float)freq) / sqrtf((float)norm);
   const float c1 = norm_const + norm_length * (float)norm;
   return num - num * c1 / (c1 + (float)freq);
 }
@@ -442,6 +445,8 @@ typedef struct {
   uint32_t buf_freqs[128];
   uint32_t buf_len, buf_pos;
   float num, nc, nl; /* prepared scorer constants */
+  uint32_t scorer;   /* 0=BM25 (bm25.cpp:89-109), 1=TFIDF, 2=TFIDF+norms
+                        (tfidf.cpp:60-76; idf per :148-151) */
 } OCursor;
 
 static void o_cursor_refill(OCursor* c) {
@@ -573,8 +578,8 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
         while (i < c->buf_len && c->buf_docs[i] <= hi) {
           const uint32_t doc = c->buf_docs[i];
           const uint32_t off = doc - lo;
-          score_win[off] += o_bm25_score(c->num, c->nc, c->nl,
-                                         c->buf_freqs[i], norms[doc]);
+          score_win[off] += o_score_one(c->scorer, c->num, c->nc, c->nl,
+                                        c->buf_freqs[i], norms[doc]);
           const uint8_t cc = ++cnt_win[off];
           if (cc == min_match) mask[off >> 6] |= 1ull << (off & 63);
           ++i;
@@ -649,6 +654,9 @@ static const SdbBlockDesc* o_seek_block(const SdbBlockDesc* b,
   return b;
 }
 
+static uint32_t o_scorer_kind = 0; /* set per call by the drivers (the C
+  API keeps single-threaded-per-call semantics; sdb_oracle is test infra) */
+
 static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
                           const float* boosts, uint32_t nterms, float k1,
                           float b, uint64_t g_dwf, const uint64_t* g_dwt,
@@ -665,10 +673,18 @@ static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
     const uint64_t dwf = g_dwf ? g_dwf : v->hdr->docs_with_field;
     const uint64_t dwt = g_dwt ? g_dwt[t] : te->df;
     const uint64_t ttf = g_ttf ? g_ttf : v->hdr->total_term_freq;
+    c->scorer = o_scorer_kind;
     if (dwt == 0) { /* term absent everywhere: no iterator */
       c->d = c->dend;
       c->num = 0.0f;
       c->nc = k1;
+      c->nl = 0.0f;
+      continue;
+    }
+    if (o_scorer_kind != 0) { /* TFIDF::collect (tfidf.cpp:148-151) */
+      idf = (float)log1p(((double)dwf + 1.0) / ((double)dwt + 1.0));
+      c->num = boosts[t] * idf;
+      c->nc = 0.0f;
       c->nl = 0.0f;
       continue;
     }
@@ -679,6 +695,10 @@ static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
   }
   return 0;
 }
+
+/* scorer selection for subsequent o_execute_* calls (0=BM25, 1=TFIDF,
+ * 2=TFIDF with norms) */
+void o_set_scorer(uint32_t kind) { o_scorer_kind = kind; }
 
 /* Cross-segment stats merge — the reference merges Field/TermCollector
  * counters over ALL segments before BM25::collect runs (PrepareCollector

@@ -298,7 +298,10 @@ class GpuContext:
         self._segments.append(seg)
         return seg
 
-    def _make_plan(self, term_idx, boosts, min_match, k1, b, global_stats):
+    SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
+
+    def _make_plan(self, term_idx, boosts, min_match, k1, b, global_stats,
+                   scorer="bm25"):
         class _Plan(C.Structure):
             _fields_ = [
                 ("terms", C.POINTER(SdbTermRef)),
@@ -306,6 +309,7 @@ class GpuContext:
                 ("min_match", C.c_uint32),
                 ("k1", C.c_float),
                 ("b", C.c_float),
+                ("scorer", C.c_uint32),
                 ("g_docs_with_field", C.c_uint64),
                 ("g_total_term_freq", C.c_uint64),
                 ("g_docs_with_term", C.POINTER(C.c_uint64)),
@@ -313,7 +317,8 @@ class GpuContext:
 
         terms = (SdbTermRef * len(term_idx))(
             *[SdbTermRef(t, float(bo)) for t, bo in zip(term_idx, boosts)])
-        plan = _Plan(terms, len(term_idx), min_match, k1, b, 0, 0, None)
+        plan = _Plan(terms, len(term_idx), min_match, k1, b,
+                     self.SCORERS[scorer], 0, 0, None)
         plan._keep = terms
         if global_stats is not None:
             dwf, ttf, dwt = global_stats
@@ -325,13 +330,13 @@ class GpuContext:
         return plan
 
     def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
-                     b=0.75, global_stats=None):
+                     b=0.75, global_stats=None, scorer="bm25"):
         """global_stats: optional (docs_with_field, total_term_freq,
         [docs_with_term per term]) for sharded execution."""
         import numpy as np
 
         plan = self._make_plan(term_idx, boosts, min_match, k1, b,
-                               global_stats)
+                               global_stats, scorer)
         seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value) for s in segs])
         hits = (SdbScoreDoc * k)()
         out_count = C.c_uint32(0)

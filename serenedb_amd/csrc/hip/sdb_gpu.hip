@@ -109,6 +109,19 @@ __device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v, int lane) {
   return v;
 }
 
+// per-(doc,term) score — the reference scorer kernels:
+//   BM25  (bm25.cpp:89-109):  num - num*c1/(c1+freq), c1 = nc + nl*norm
+//   TFIDF (tfidf.cpp:60-76):  num*sqrt(freq) [/ sqrt(norm) with norms]
+// sqrtf/div are IEEE-exact: bit parity with the CPU oracle holds.
+__device__ __forceinline__ float score_one(uint32_t scorer, float num,
+                                           float nc, float nl, uint32_t freq,
+                                           uint32_t norm) {
+  if (scorer == 1) return num * sqrtf((float)freq);
+  if (scorer == 2) return num * sqrtf((float)freq) / sqrtf((float)norm);
+  const float c1 = nc + nl * (float)norm;
+  return num - num * c1 / (c1 + (float)freq);
+}
+
 // vertical-layout delta extract: value i of a 128-block, b bits
 // (simdcomp d1 layout — see include/sdb_format.h)
 __device__ __forceinline__ uint32_t extract_packed(const uint8_t* base,
@@ -279,7 +292,7 @@ __device__ void decode_freq_block_wave(const uint8_t* p, uint32_t len,
 // combination (caller falls back to the generic per-stream decode).
 __device__ __forceinline__ bool try_block_fused(
   const uint8_t* pl, const SdbBlockDesc& d, int lane, uint32_t norm_stream,
-  uint32_t lo, uint32_t hi, float num, float nc, float nl,
+  uint32_t lo, uint32_t hi, float num, float nc, float nl, uint32_t scorer,
   const uint32_t* norms_col, float* swin, uint8_t* cwin) {
   const uint8_t* db = pl + d.doc_off;
   const uint8_t* fb = pl + d.freq_off;
@@ -313,15 +326,13 @@ __device__ __forceinline__ bool try_block_fused(
     n1 = doc1 >= lo && doc1 <= hi ? norms_col[doc1] : 1u;
   }
   if (doc0 >= lo && doc0 <= hi) {
-    const float c1 = nc + nl * (float)n0;
-    const float s = num - num * c1 / (c1 + (float)f0);
+    const float s = score_one(scorer, num, nc, nl, f0, n0);
     const uint32_t off = doc0 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
   }
   if (doc1 >= lo && doc1 <= hi) {
-    const float c1 = nc + nl * (float)n1;
-    const float s = num - num * c1 / (c1 + (float)f1);
+    const float s = score_one(scorer, num, nc, nl, f1, n1);
     const uint32_t off = doc1 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -335,8 +346,8 @@ __device__ __forceinline__ bool try_block_fused(
 __device__ __forceinline__ bool try_block_fused2(
   const uint8_t* pl, const SdbBlockDesc& da, const SdbBlockDesc& db_,
   int lane, uint32_t norm_stream, uint32_t lo, uint32_t hi, float num,
-  float nc, float nl, const uint32_t* norms_col, float* swin,
-  uint8_t* cwin) {
+  float nc, float nl, uint32_t scorer, const uint32_t* norms_col,
+  float* swin, uint8_t* cwin) {
   const uint8_t* adoc = pl + da.doc_off;
   const uint8_t* afrq = pl + da.freq_off;
   const uint8_t* anrm = afrq + da.flags;
@@ -390,8 +401,7 @@ __device__ __forceinline__ bool try_block_fused2(
   for (int e = 0; e < 4; ++e) {
     const uint32_t doc = docs[e];
     if (doc < lo || doc > hi) continue;
-    const float c1 = nc + nl * (float)nrms[e];
-    const float s = num - num * c1 / (c1 + (float)frqs[e]);
+    const float s = score_one(scorer, num, nc, nl, frqs[e], nrms[e]);
     const uint32_t off = doc - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -432,6 +442,7 @@ struct WindowArgs {
   const uint8_t* payload;
   const uint32_t* norms;
   uint32_t doc_count;  // docs 1..doc_count (local ids within this segment)
+  uint32_t scorer;     // SdbScorerType (uniform per plan)
   uint32_t norm_stream;  // v2 segments: per-block norm blocks (flags=freq sz)
   uint32_t nterms;
   uint32_t min_match;
@@ -589,7 +600,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                       : a.desc[b2];
             if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
                 try_block_fused2(pl, d, d2, lane, a.norm_stream, lo, hi,
-                                 num, nc, nl, a.norms, swin, cwin)) {
+                                 num, nc, nl, a.scorer, a.norms, swin,
+                                 cwin)) {
               b += 2 * SDB_NWAVES;
               continue;
             }
@@ -623,7 +635,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #else
 #ifndef SDB_ABLATE_SCORE
         if (try_block_fused(pl, d, lane, a.norm_stream, lo, hi, num, nc, nl,
-                            a.norms, swin, cwin)) {
+                            a.scorer, a.norms, swin, cwin)) {
           b += SDB_NWAVES;  // while-loop: explicit advance before continue
           continue;
         }
@@ -642,8 +654,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #ifdef SDB_ABLATE_SCORE
           asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
 #else
-          const float c1 = nc + nl * (float)norm;
-          const float s = num - num * c1 / (c1 + (float)freq);
+          const float s = score_one(a.scorer, num, nc, nl, freq, norm);
           const uint32_t off = doc - lo;
           swin[off] += s;      // unique doc within the term: no atomics
           cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -1073,8 +1084,9 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       g_dwt[t] = plan->g_docs_with_term[t];
   }
   const float k1 = plan->k1, b = plan->b;
+  const uint32_t scorer = plan->scorer;
   std::vector<float> idf(plan->nterms), num(plan->nterms);
-  float nc, nl;
+  float nc = 0.0f, nl = 0.0f;
   float smax = 0.0f;
   for (uint32_t t = 0; t < plan->nterms; ++t) {
     if (g_dwt[t] == 0) {
@@ -1082,13 +1094,29 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       num[t] = 0.0f;
       continue;
     }
-    // BM25::collect (bm25.cpp:288-306)
-    idf[t] = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
-                          ((double)g_dwt[t] + 0.5));
-    num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
-    smax += num[t] > 0 ? num[t] : 0.0f;
+    uint32_t term_max_freq = 1;
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      const uint32_t ti = plan->terms[t].term_idx;
+      term_max_freq =
+        std::max(term_max_freq, segs[s]->terms_host[ti].max_freq);
+    }
+    if (scorer == SDB_SCORER_BM25) {
+      // BM25::collect (bm25.cpp:288-306)
+      idf[t] = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
+                            ((double)g_dwt[t] + 0.5));
+      num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
+      smax += num[t] > 0 ? num[t] : 0.0f;
+    } else {
+      // TFIDF::collect (tfidf.cpp:148-151)
+      idf[t] = (float)log1p(((double)g_dwf + 1.0) /
+                            ((double)g_dwt[t] + 1.0));
+      num[t] = plan->terms[t].boost * idf[t];
+      // score <= num*sqrt(max_freq) (norm >= 1 only helps the bound)
+      const float ub = num[t] * sqrtf((float)term_max_freq);
+      smax += ub > 0 ? ub : 0.0f;
+    }
   }
-  {
+  if (scorer == SDB_SCORER_BM25) {
     const float kb = k1 * b;
     if (b == 0.0f) {
       nc = k1;
@@ -1143,6 +1171,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.payload = seg->payload;
     a.norms = seg->norms;
     a.doc_count = seg->hdr.doc_count;
+    a.scorer = scorer;
     a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
     a.nterms = plan->nterms;
     a.min_match = plan->min_match ? plan->min_match : 1;

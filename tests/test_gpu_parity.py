@@ -349,3 +349,18 @@ def test_full_size_properties(ctx):
     assert total2 == total
     np.testing.assert_array_equal(hits["doc"], hits2["doc"])
     np.testing.assert_array_equal(hits["score"], hits2["score"])
+
+
+def test_tfidf_parity(ctx):
+    """TFIDF scorers (tfidf.cpp:60-76,148-151) through the same interface."""
+    blob, _, _ = make_corpus(57, 300_000, [0.05, 0.02, 0.01])
+    seg = ctx.load_segment(blob)
+    for scorer in ("tfidf", "tfidf_norm"):
+        hits, total = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 200,
+                                       scorer=scorer)
+        ohits, ototal = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 200,
+                                        scorer=scorer)
+        assert total == ototal
+        np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+        np.testing.assert_array_equal(
+            hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
