@@ -104,6 +104,14 @@ typedef struct SdbQueryPlan {
   float b;  /* BM25 b (search/bm25.hpp:64: default 0.75) */
   uint32_t scorer; /* SdbScorerType; TFIDF idf = log1p((N+1)/(df+1)),
                       tfidf.cpp:148-151 */
+  /* WAND block-max pruning (max_score_iterator.hpp Block-Max MaxScore
+   * analogue over the descriptor max_freq/min_norm bounds): skips postings
+   * blocks that provably cannot reach the current k-th score. EXACT top-k
+   * (every skipped doc's full score < the threshold); only valid for pure
+   * disjunctions (min_match == 1); total_matches then counts only VISITED
+   * matches (the reference's WAND path likewise stops counting,
+   * reader.hpp:517-524 wand gate). 0 = off. */
+  uint32_t wand;
   /* Optional GLOBAL BM25 stats for sharded (multi-rank) execution, where
    * each device holds only its shard: the cross-rank stats merge is the
    * PreparePhase barrier analogue (duckdb_search_full_scan.cpp:1359-1384)

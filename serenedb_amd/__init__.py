@@ -301,7 +301,7 @@ class GpuContext:
     SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
 
     def _make_plan(self, term_idx, boosts, min_match, k1, b, global_stats,
-                   scorer="bm25"):
+                   scorer="bm25", wand=False):
         class _Plan(C.Structure):
             _fields_ = [
                 ("terms", C.POINTER(SdbTermRef)),
@@ -310,6 +310,7 @@ class GpuContext:
                 ("k1", C.c_float),
                 ("b", C.c_float),
                 ("scorer", C.c_uint32),
+                ("wand", C.c_uint32),
                 ("g_docs_with_field", C.c_uint64),
                 ("g_total_term_freq", C.c_uint64),
                 ("g_docs_with_term", C.POINTER(C.c_uint64)),
@@ -318,7 +319,7 @@ class GpuContext:
         terms = (SdbTermRef * len(term_idx))(
             *[SdbTermRef(t, float(bo)) for t, bo in zip(term_idx, boosts)])
         plan = _Plan(terms, len(term_idx), min_match, k1, b,
-                     self.SCORERS[scorer], 0, 0, None)
+                     self.SCORERS[scorer], 1 if wand else 0, 0, 0, None)
         plan._keep = terms
         if global_stats is not None:
             dwf, ttf, dwt = global_stats
@@ -330,13 +331,15 @@ class GpuContext:
         return plan
 
     def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
-                     b=0.75, global_stats=None, scorer="bm25"):
+                     b=0.75, global_stats=None, scorer="bm25", wand=False):
         """global_stats: optional (docs_with_field, total_term_freq,
-        [docs_with_term per term]) for sharded execution."""
+        [docs_with_term per term]) for sharded execution. wand=True enables
+        exact block-max pruning (OR plans only; total_matches then counts
+        visited matches only)."""
         import numpy as np
 
         plan = self._make_plan(term_idx, boosts, min_match, k1, b,
-                               global_stats, scorer)
+                               global_stats, scorer, wand)
         seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value) for s in segs])
         hits = (SdbScoreDoc * k)()
         out_count = C.c_uint32(0)

@@ -443,6 +443,7 @@ struct WindowArgs {
   const uint32_t* norms;
   uint32_t doc_count;  // docs 1..doc_count (local ids within this segment)
   uint32_t scorer;     // SdbScorerType (uniform per plan)
+  uint32_t wand;       // block-max pruning (OR only; see sdb_gpu.h)
   uint32_t norm_stream;  // v2 segments: per-block norm blocks (flags=freq sz)
   uint32_t nterms;
   uint32_t min_match;
@@ -482,10 +483,11 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* hist = scratch + SDB_NWAVES * 384;            // 256
   uint32_t* shared_misc = hist + SDB_HIST_BINS;     // bcast + wave counts
   uint32_t* cursors = shared_misc + 2 + SDB_NWAVES; // per-term block cursor
+  // (cursors + SDB_MAX_TERMS .. +2*SDB_MAX_TERMS) = WAND per-term bounds
   // staged descriptors: SDB_DESC_CACHE per term, one coalesced load per
   // window removes a ~900-cycle dependent desc load from every block chain
   unsigned long long* lbuck =
-    (unsigned long long*)(cursors + SDB_MAX_TERMS);  // 2*SDB_MAX_BUCKETS
+    (unsigned long long*)(cursors + 2 * SDB_MAX_TERMS);  // 2*SDB_MAX_BUCKETS
   SdbBlockDesc* dcache = (SdbBlockDesc*)(lbuck + 2 * SDB_MAX_BUCKETS);
 
   const uint32_t tid = threadIdx.x;
@@ -565,8 +567,30 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     if (tid == 0)
       shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
                                          __HIP_MEMORY_SCOPE_AGENT);
+    if (a.wand) __syncthreads();  // wub scan reads other threads' staging
+    // WAND: per-term window score upper bounds from the staged descriptors
+    // (score is monotone up in freq, down in norm, so (max_freq, min_norm)
+    // bounds every doc in a block)
+    float* wub = (float*)(shared_misc + 2 + SDB_NWAVES + SDB_MAX_TERMS);
+    if (a.wand && tid < a.nterms) {
+      const TermDev te = terms[tid];
+      const uint32_t cur0w = cursors[tid];
+      float ub = 0.0f;
+      for (uint32_t i = 0; i < SDB_DESC_CACHE; ++i) {
+        if (te.desc_begin + cur0w + i >= te.desc_end) break;
+        const SdbBlockDesc d = dcache[tid * SDB_DESC_CACHE + i];
+        if (d.prev_doc >= hi) break;
+        const float u =
+          score_one(a.scorer, te.num, te.nc, te.nl, d.max_freq, d.min_norm);
+        ub = u > ub ? u : ub;
+      }
+      wub[tid] = ub;
+    }
     __syncthreads();
     SDB_T(0)
+    float wand_total_ub = 0.0f;
+    if (a.wand)
+      for (uint32_t t = 0; t < a.nterms; ++t) wand_total_ub += wub[t];
 
     // term-major phases (fixed fp32 merge order -> bit-exact vs the oracle)
     for (uint32_t t = 0; t < a.nterms; ++t) {
@@ -588,6 +612,21 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                  ? dcache[t * SDB_DESC_CACHE + rel]
                                  : a.desc[b];
         if (d.prev_doc >= hi) break;  // first doc > hi
+        if (a.wand) {
+          // skip if even this block's best doc cannot reach the threshold
+          // with every other term's window-best contribution
+          float gtw;
+          {
+            const uint32_t bits = shared_misc[0];
+            __builtin_memcpy(&gtw, &bits, 4);
+          }
+          const float own =
+            score_one(a.scorer, num, nc, nl, d.max_freq, d.min_norm);
+          if (own + (wand_total_ub - wub[t]) < gtw) {
+            b += SDB_NWAVES;
+            continue;
+          }
+        }
 #if !defined(SDB_ABLATE_DECODE) && !defined(SDB_ABLATE_SCORE)
         // pair this block with the wave's next one when both are the
         // common fused shape: both chains' loads fly together
@@ -1139,7 +1178,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
 
   const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
-                           (2 + SDB_NWAVES + SDB_MAX_TERMS) * 4 +
+                           (2 + SDB_NWAVES + 2 * SDB_MAX_TERMS) * 4 +
                            8 * 2 * SDB_MAX_BUCKETS +
                            sizeof(SdbBlockDesc) * SDB_DESC_CACHE *
                              plan->nterms;
@@ -1172,6 +1211,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.norms = seg->norms;
     a.doc_count = seg->hdr.doc_count;
     a.scorer = scorer;
+    a.wand = (plan->wand && (plan->min_match <= 1) && !hybrid) ? 1u : 0u;
     a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
     a.nterms = plan->nterms;
     a.min_match = plan->min_match ? plan->min_match : 1;

@@ -364,3 +364,20 @@ def test_tfidf_parity(ctx):
         np.testing.assert_array_equal(hits["doc"], ohits["doc"])
         np.testing.assert_array_equal(
             hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
+
+
+def test_wand_exactness(ctx):
+    """WAND block-max pruning must return the IDENTICAL top-k (hits and
+    scores) as the unpruned path — the reference's WAND is exact
+    (formats_15_tests.cpp AssertWandPostings pins pruning correctness)."""
+    blob, _, _ = make_corpus(58, 2_000_000, [0.10, 0.05, 0.02, 0.01])
+    seg = ctx.load_segment(blob)
+    for k in (10, 100, 1000):
+        for scorer in ("bm25", "tfidf_norm"):
+            base, _bt = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, k,
+                                         scorer=scorer)
+            wand, _wt = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, k,
+                                         scorer=scorer, wand=True)
+            np.testing.assert_array_equal(base["doc"], wand["doc"])
+            np.testing.assert_array_equal(
+                base["score"].view(np.uint32), wand["score"].view(np.uint32))
