@@ -1182,6 +1182,9 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                            8 * 2 * SDB_MAX_BUCKETS +
                            sizeof(SdbBlockDesc) * SDB_DESC_CACHE *
                              plan->nterms;
+  if (lds_bytes > 160 * 1024)  // LDS/CU cap: ~14 terms at the default
+    return SDB_ERR_INVALID;    // window size (shrink the desc
+                               // cache for wider plans in a later round)
 
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   for (uint32_t s = 0; s < nsegs; ++s) {

@@ -381,3 +381,11 @@ def test_wand_exactness(ctx):
             np.testing.assert_array_equal(base["doc"], wand["doc"])
             np.testing.assert_array_equal(
                 base["score"].view(np.uint32), wand["score"].view(np.uint32))
+
+
+def test_eight_term_disjunction(ctx):
+    sels = [0.08, 0.05, 0.04, 0.03, 0.02, 0.015, 0.01, 0.005]
+    blob, _, _ = make_corpus(59, 400_000, sels)
+    check_parity(ctx, blob, list(range(8)),
+                 [1.0, 2.0, 0.5, 1.0, 3.0, 1.0, 0.25, 1.0], 300)
+    check_parity(ctx, blob, list(range(8)), [1.0] * 8, 100, min_match=3)
