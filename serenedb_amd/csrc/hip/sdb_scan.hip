@@ -172,9 +172,16 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
          r < rows2; r += pair_stride) {
       bool okv[2] = {true, true};
       for (uint32_t p = 0; p < a.npreds; ++p) {
+#ifdef SDB_SCAN_NT
+        // streamed-once columns: nontemporal loads keep L2 for reuse data
+        const long long* pp = &((const long long*)a.pred_col[p].data)[r];
+        const int64_t xs[2] = {__builtin_nontemporal_load(pp),
+                               __builtin_nontemporal_load(pp + 1)};
+#else
         longlong2 x;
         __builtin_memcpy(&x, &((const long long*)a.pred_col[p].data)[r], 16);
         const int64_t xs[2] = {x.x, x.y};
+#endif
 #pragma unroll
         for (int e = 0; e < 2; ++e) {
           switch (a.pred_op[p]) {

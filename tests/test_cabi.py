@@ -45,3 +45,37 @@ def test_gpu_ctx_fails_loudly_without_gpu():
 
     with pytest.raises(RuntimeError, match="sdb_gpu_ctx_create"):
         sa.GpuContext(0)
+
+
+def test_malformed_blob_rejected():
+    """parse must reject garbage/truncated/wrong-version blobs loudly"""
+    import numpy as np
+    import serenedb_amd as sa
+
+    class _View(ctypes.Structure):
+        _fields_ = [("hdr", ctypes.c_void_p), ("terms", ctypes.c_void_p),
+                    ("desc", ctypes.c_void_p), ("norms", ctypes.c_void_p),
+                    ("payload", ctypes.c_void_p)]
+
+    host = sa.host()
+    v = _View()
+    garbage = np.frombuffer(b"\x00" * 256, dtype=np.uint8)
+    rc = host.sdb_host_segment_parse(
+        garbage.ctypes.data_as(ctypes.c_void_p), ctypes.c_uint64(256),
+        ctypes.byref(v))
+    assert rc != 0
+    # valid blob, truncated
+    blob = sa.build_synth_segment(1, 1, 1000, [0.1])
+    cut = np.frombuffer(blob[: len(blob) // 2], dtype=np.uint8)
+    rc = host.sdb_host_segment_parse(
+        cut.ctypes.data_as(ctypes.c_void_p), ctypes.c_uint64(len(cut)),
+        ctypes.byref(v))
+    assert rc != 0
+    # wrong version
+    bad = bytearray(blob[:256])
+    bad[8] = 99
+    arr = np.frombuffer(bytes(bad), dtype=np.uint8)
+    rc = host.sdb_host_segment_parse(
+        arr.ctypes.data_as(ctypes.c_void_p), ctypes.c_uint64(len(arr)),
+        ctypes.byref(v))
+    assert rc != 0
