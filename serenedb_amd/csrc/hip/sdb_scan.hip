@@ -436,6 +436,12 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
 
 extern "C" {
 
+static void table_free_partial(SdbGpuTable* tab) {
+  for (uint32_t c = 0; c < 16; ++c)
+    if (tab->cols[c]) (void)hipFree(tab->cols[c]);
+  delete tab;
+}
+
 int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
                        uint32_t ncols, uint64_t rows, SdbGpuTable** out) {
   if (!ctx || !cols || !out || ncols == 0 || ncols > 16)
@@ -450,10 +456,14 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
       // encoded blob: upload whole blob, point refs into it
       SdbColHeaderDev hdr;
       std::memcpy(&hdr, cols[c].data, sizeof(hdr));
-      if (hdr.magic != SDB_COL_MAGIC_DEV || hdr.rows != rows)
+      if (hdr.magic != SDB_COL_MAGIC_DEV || hdr.rows != rows) {
+        table_free_partial(tab);
         return SDB_ERR_INVALID;
-      if (tab->group_rows && tab->group_rows != hdr.group_rows)
-        return SDB_ERR_INVALID;  // FoR columns must share the group tiling
+      }
+      if (tab->group_rows && tab->group_rows != hdr.group_rows) {
+        table_free_partial(tab);  // FoR columns must share the group tiling
+        return SDB_ERR_INVALID;
+      }
       tab->group_rows = hdr.group_rows;
       HIP_CHECK(hipMalloc(&tab->cols[c], hdr.size));
       HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, hdr.size,

@@ -328,3 +328,23 @@ def test_scan_agg_mt_equals_single():
     np.testing.assert_array_equal(c1, c2)
     np.testing.assert_array_equal(s1, s2)
     np.testing.assert_allclose(f1, f2, rtol=1e-12)
+
+
+def test_config1_cpu_plumbing():
+    """BASELINE configs[0]: BM25 top-10, 2-term OR, 1M-doc synthetic
+    segment, CPU oracle only (seed 42, selectivities 5%/2%, geometric
+    freqs, lognormal norms — SURVEY.md §8d). Pins the oracle end to end at
+    the named shape: exact vs mechanics vs brute force."""
+    doc_count = 1_000_000
+    sels = [0.05, 0.02]
+    seed = 42
+    blob, postings, norms = synth_corpus(seed, doc_count, sels)
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 10)
+    mhits, mtotal = po.execute_topk_mech([blob], [0, 1], [1.0, 1.0], 10)
+    assert total == mtotal
+    np.testing.assert_array_equal(hits["doc"], mhits["doc"][:len(hits)])
+    order, scores, nmatch = brute_topk(postings, norms, doc_count, sels, 10)
+    assert total == nmatch
+    assert [int(h["doc"]) for h in hits] == [int(d) for d in order]
+    for h in hits:
+        assert h["score"] == scores[int(h["doc"])]
