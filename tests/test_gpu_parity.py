@@ -389,3 +389,51 @@ def test_eight_term_disjunction(ctx):
     check_parity(ctx, blob, list(range(8)),
                  [1.0, 2.0, 0.5, 1.0, 3.0, 1.0, 0.25, 1.0], 300)
     check_parity(ctx, blob, list(range(8)), [1.0] * 8, 100, min_match=3)
+
+
+def test_hybrid_multi_segment(ctx):
+    """hybrid across two resident segments: bucket aggregates accumulate
+    and the merged top-k equals per-segment oracle runs merged with global
+    stats (the multi-segment PreparePhase semantics)."""
+    seed = 63
+    sels = [0.08, 0.03]
+    n1, n2 = 150_000, 100_000
+    b1 = sa.build_synth_segment(seed, 1, n1, sels)
+    b2 = sa.build_synth_segment(seed, n1 + 1, n1 + n2, sels)
+    rng = np.random.default_rng(12)
+    span = 1 << 31
+    col1 = rng.integers(0, span, n1 + 1).astype(np.int64)
+    col2 = rng.integers(0, span, n2 + 1).astype(np.int64)
+    flo, fhi = int(span * 0.3), int(span * 0.7) - 1
+    nb = 32
+    s1 = ctx.load_segment(b1)
+    s2 = ctx.load_segment(b2)
+    ctx.attach_column(s1, col1)
+    ctx.attach_column(s2, col2)
+    hits, total, bcnt, bsum = ctx.execute_topk_hybrid(
+        [s1, s2], [0, 1], [1.0, 1.0], 200, flo, fhi, nb)
+    # oracle: per-segment with merged global stats, buckets summed
+    d0 = sa.synth_postings(seed, n1 + n2, 0, sels[0])[0]
+    d1 = sa.synth_postings(seed, n1 + n2, 1, sels[1])[0]
+    norms = sa.synth_norms(seed, n1 + n2)
+    gstats = (n1 + n2, int(norms[1:].sum()), [len(d0), len(d1)])
+    tot = 0
+    cands = []
+    cnt = np.zeros(nb, dtype=np.int64)
+    sm = np.zeros(nb, dtype=np.int64)
+    for si, (blob, col) in enumerate(((b1, col1), (b2, col2))):
+        h, t, bc, bs = po.execute_topk_hybrid(
+            blob, [0, 1], [1.0, 1.0], 200, col, flo, fhi, nb,
+            global_stats=gstats)
+        tot += t
+        cnt += bc
+        sm += bs
+        for x in h:
+            cands.append((float(x["score"]), si, int(x["doc"])))
+    cands.sort(key=lambda v: (-v[0], v[1], v[2]))
+    assert total == tot
+    np.testing.assert_array_equal(bcnt, cnt)
+    np.testing.assert_array_equal(bsum, sm)
+    got = [(float(h["score"]), int(h["segment"]), int(h["doc"]))
+           for h in hits]
+    assert got == cands[:len(got)]
