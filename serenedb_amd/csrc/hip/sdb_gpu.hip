@@ -1005,6 +1005,33 @@ extern "C" {
 
 const char* sdb_gpu_version(void) { return "sdb_gpu 0.1 gfx950"; }
 
+// free whatever a partially-constructed context owns (error paths)
+static void ctx_free_partial(SdbGpuCtx* ctx) {
+  if (ctx->d_cands) (void)hipFree(ctx->d_cands);
+  if (ctx->d_cand_count) (void)hipFree(ctx->d_cand_count);
+  if (ctx->d_total_matches) (void)hipFree(ctx->d_total_matches);
+  if (ctx->d_gthresh) (void)hipFree(ctx->d_gthresh);
+  if (ctx->d_ghist) (void)hipFree(ctx->d_ghist);
+  if (ctx->d_buckets) (void)hipFree(ctx->d_buckets);
+  if (ctx->d_terms) (void)hipFree(ctx->d_terms);
+  if (ctx->d_overflow) (void)hipFree(ctx->d_overflow);
+  if (ctx->h_counts) (void)hipHostFree(ctx->h_counts);
+  if (ctx->h_matches) (void)hipHostFree(ctx->h_matches);
+  if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
+  delete ctx;
+}
+
+#define CTX_CHECK(x)                                   \
+  do {                                                 \
+    hipError_t _e = (x);                               \
+    if (_e != hipSuccess) {                            \
+      ctx_free_partial(ctx);                           \
+      return _e == hipErrorNoDevice ? SDB_ERR_NO_GPU   \
+             : _e == hipErrorOutOfMemory ? SDB_ERR_OOM \
+                                         : SDB_ERR_HIP; \
+    }                                                  \
+  } while (0)
+
 int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   if (!out) return SDB_ERR_INVALID;
   int rc = check_gpu();
@@ -1012,19 +1039,19 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   HIP_CHECK(hipSetDevice(device));
   auto* ctx = new SdbGpuCtx{};
   ctx->device = device;
-  HIP_CHECK(hipStreamCreate(&ctx->stream));
-  HIP_CHECK(hipMalloc(&ctx->d_cands, sizeof(SdbScoreDoc) * (size_t)SDB_CAND_CAP));
-  HIP_CHECK(hipMalloc(&ctx->d_cand_count, 4));
-  HIP_CHECK(hipMalloc(&ctx->d_total_matches, 8));
-  HIP_CHECK(hipMalloc(&ctx->d_gthresh, 4));
-  HIP_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 8));
-  HIP_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
-  HIP_CHECK(hipMalloc(&ctx->d_overflow, 4));
-  HIP_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
-  HIP_CHECK(hipHostMalloc(&ctx->h_counts, 8));
-  HIP_CHECK(hipHostMalloc(&ctx->h_matches, 8));
-  HIP_CHECK(hipEventCreate(&ctx->ev_a));
-  HIP_CHECK(hipEventCreate(&ctx->ev_b));
+  CTX_CHECK(hipStreamCreate(&ctx->stream));
+  CTX_CHECK(hipMalloc(&ctx->d_cands, sizeof(SdbScoreDoc) * (size_t)SDB_CAND_CAP));
+  CTX_CHECK(hipMalloc(&ctx->d_cand_count, 4));
+  CTX_CHECK(hipMalloc(&ctx->d_total_matches, 8));
+  CTX_CHECK(hipMalloc(&ctx->d_gthresh, 4));
+  CTX_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 8));
+  CTX_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
+  CTX_CHECK(hipMalloc(&ctx->d_overflow, 4));
+  CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
+  CTX_CHECK(hipHostMalloc(&ctx->h_counts, 8));
+  CTX_CHECK(hipHostMalloc(&ctx->h_matches, 8));
+  CTX_CHECK(hipEventCreate(&ctx->ev_a));
+  CTX_CHECK(hipEventCreate(&ctx->ev_b));
   *out = ctx;
   return SDB_OK;
 }
@@ -1057,23 +1084,36 @@ int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
   auto* seg = new SdbGpuSegment{};
   seg->hdr = hdr;
   const uint8_t* base = (const uint8_t*)blob;
+#define SEG_CHECK(x)                                                     \
+  do {                                                                   \
+    hipError_t _e = (x);                                                 \
+    if (_e != hipSuccess) {                                              \
+      if (seg->desc) (void)hipFree(seg->desc);                           \
+      if (seg->payload) (void)hipFree(seg->payload);                     \
+      if (seg->norms) (void)hipFree(seg->norms);                         \
+      std::free(seg->terms_host);                                        \
+      delete seg;                                                        \
+      return _e == hipErrorOutOfMemory ? SDB_ERR_OOM : SDB_ERR_HIP;      \
+    }                                                                    \
+  } while (0)
   seg->terms_host = (SdbTermEntry*)std::malloc(sizeof(SdbTermEntry) * hdr.nterms);
   std::memcpy(seg->terms_host, base + hdr.off_terms,
               sizeof(SdbTermEntry) * hdr.nterms);
-  HIP_CHECK(hipMalloc(&seg->desc, sizeof(SdbBlockDesc) * hdr.total_blocks + 16));
-  HIP_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 512));
-  HIP_CHECK(
+  SEG_CHECK(hipMalloc(&seg->desc, sizeof(SdbBlockDesc) * hdr.total_blocks + 16));
+  SEG_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 512));
+  SEG_CHECK(
     hipMalloc(&seg->norms, sizeof(uint32_t) * ((size_t)hdr.doc_count + 1)));
-  HIP_CHECK(hipMemcpy(seg->desc, base + hdr.off_desc,
+  SEG_CHECK(hipMemcpy(seg->desc, base + hdr.off_desc,
                       sizeof(SdbBlockDesc) * hdr.total_blocks,
                       hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(seg->payload, base + hdr.off_payload, hdr.payload_size,
+  SEG_CHECK(hipMemcpy(seg->payload, base + hdr.off_payload, hdr.payload_size,
                       hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(seg->norms, base + hdr.off_norms,
+  SEG_CHECK(hipMemcpy(seg->norms, base + hdr.off_norms,
                       sizeof(uint32_t) * ((size_t)hdr.doc_count + 1),
                       hipMemcpyHostToDevice));
   *out = seg;
   return SDB_OK;
+#undef SEG_CHECK
 }
 
 int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
