@@ -146,6 +146,12 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                          SdbScoreDoc* hits, uint32_t* out_count,
                          uint64_t* total_matches);
 
+/* CountFast — exact match count without scoring (docs-only decode;
+ * DecideScanMode Count/CountFast, duckdb_search_full_scan.cpp:972). */
+int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                          uint32_t nsegs, const SdbQueryPlan* plan,
+                          uint64_t* total_matches);
+
 /* Streaming scan — the RunStreamingScan / HitBatcher analogue
  * (duckdb_search_full_scan.cpp:2370, index/hit_batcher.hpp:39-190): emit
  * every matching doc id ascending into docs_out (up to cap) and, when

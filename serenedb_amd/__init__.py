@@ -258,7 +258,7 @@ def gpu():
             "sdb_gpu_execute_topk", "sdb_gpu_decode_term",
             "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
-            "sdb_gpu_execute_match_docs",
+            "sdb_gpu_execute_match_docs", "sdb_gpu_execute_count",
         ):
             getattr(lib, f).restype = C.c_int
         _gpu = lib
@@ -391,6 +391,18 @@ class GpuContext:
         res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
                             dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value, bcnt, bsum
+
+    def execute_count(self, segs, term_idx, boosts, min_match=1, k1=1.2,
+                      b=0.75):
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b, None)
+        seg_arr = (C.c_void_p * len(segs))(
+            *[C.c_void_p(s.value) for s in segs])
+        total = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_execute_count(
+            self._ctx, seg_arr, len(segs), C.byref(plan), C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_count rc={rc}")
+        return total.value
 
     def execute_match_docs(self, seg, term_idx, boosts, cap, min_match=1,
                            k1=1.2, b=0.75, with_col=False):

@@ -444,6 +444,7 @@ struct WindowArgs {
   uint32_t doc_count;  // docs 1..doc_count (local ids within this segment)
   uint32_t scorer;     // SdbScorerType (uniform per plan)
   uint32_t wand;       // block-max pruning (OR only; see sdb_gpu.h)
+  uint32_t count_only; // CountFast mode: docs-only decode, no scoring
   uint32_t norm_stream;  // v2 segments: per-block norm blocks (flags=freq sz)
   uint32_t nterms;
   uint32_t min_match;
@@ -627,6 +628,19 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             continue;
           }
         }
+        if (a.count_only) {
+          // CountFast (DecideScanMode Count/CountFast analogue,
+          // duckdb_search_full_scan.cpp:972): decode doc ids only
+          decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane,
+                                dbuf);
+          for (uint32_t j = lane; j < d.len; j += 64) {
+            const uint32_t doc = dbuf[j];
+            if (doc < lo || doc > hi) continue;
+            cwin[doc - lo] = (uint8_t)(cwin[doc - lo] + 1u);
+          }
+          b += SDB_NWAVES;
+          continue;
+        }
 #if !defined(SDB_ABLATE_DECODE) && !defined(SDB_ABLATE_SCORE)
         // pair this block with the wave's next one when both are the
         // common fused shape: both chains' loads fly together
@@ -738,7 +752,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // count against the window's threshold snapshot (read during staging;
     // one window stale = smaller = conservative: a few extra candidates,
     // never a dropped top-k member)
-    const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
+    const bool derive =
+      !a.count_only && (((w & 3u) == 0) || (w < w_lo + 2));
     float gtau_w;
     {
       const uint32_t bits = shared_misc[0];
@@ -780,9 +795,10 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     }
     // per-wave candidate-count scan inputs (consumed after the barrier)
     {
-      const uint32_t incl0 = wave_incl_scan(my_cnt, lane);
+      const uint32_t incl0 = wave_incl_scan(a.count_only ? 0u : my_cnt,
+                                            lane);
       if (lane == 63) scratch[SDB_NTHREADS + wave] = incl0;
-      my_excl_snap = incl0 - my_cnt;
+      my_excl_snap = incl0 - (a.count_only ? 0u : my_cnt);
     }
     uint32_t wm = my_matches;
 #pragma unroll
@@ -1133,7 +1149,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                           long long h_fhi, uint32_t h_nbuckets,
                           int64_t* bucket_count, int64_t* bucket_sum,
                           SdbScoreDoc* hits, uint32_t* out_count,
-                          uint64_t* total_matches) {
+                          uint64_t* total_matches, int count_only = 0) {
   if (!ctx || !segs || !plan || !out_count || !total_matches ||
       plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS || k == 0)
     return SDB_ERR_INVALID;
@@ -1255,6 +1271,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.doc_count = seg->hdr.doc_count;
     a.scorer = scorer;
     a.wand = (plan->wand && (plan->min_match <= 1) && !hybrid) ? 1u : 0u;
+    a.count_only = count_only;
     a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
     a.nterms = plan->nterms;
     a.min_match = plan->min_match ? plan->min_match : 1;
@@ -1381,6 +1398,21 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                          uint64_t* total_matches) {
   return exec_topk_impl(ctx, segs, nsegs, plan, k, 0, 0, 0, 0, nullptr,
                         nullptr, hits, out_count, total_matches);
+}
+
+// CountFast: exact match count without scoring (docs-only decode —
+// DecideScanMode Count/CountFast, duckdb_search_full_scan.cpp:972).
+// WAND never applies (counting must visit every match).
+int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                          uint32_t nsegs, const SdbQueryPlan* plan,
+                          uint64_t* total_matches) {
+  if (!total_matches) return SDB_ERR_INVALID;
+  SdbQueryPlan p = *plan;
+  p.wand = 0;
+  uint32_t n = 0;
+  return exec_topk_impl(ctx, segs, nsegs, &p, 1, 0, 0, 0, 0, nullptr,
+                        nullptr, /*hits=*/nullptr, &n, total_matches,
+                        /*count_only=*/1);
 }
 
 // Attach the hybrid filter column (i64[doc_count+1], index 0 unused) to a

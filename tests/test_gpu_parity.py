@@ -437,3 +437,15 @@ def test_hybrid_multi_segment(ctx):
     got = [(float(h["score"]), int(h["segment"]), int(h["doc"]))
            for h in hits]
     assert got == cands[:len(got)]
+
+
+def test_count_fast(ctx):
+    """CountFast (docs-only decode) == full-path total_matches, OR and
+    min-match variants."""
+    blob, _, _ = make_corpus(64, 800_000, [0.08, 0.04, 0.01])
+    seg = ctx.load_segment(blob)
+    for mm in (1, 2, 3):
+        _, full = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 10,
+                                   min_match=mm)
+        fast = ctx.execute_count([seg], [0, 1, 2], [1.0] * 3, min_match=mm)
+        assert fast == full, (mm, fast, full)
