@@ -1158,6 +1158,10 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   if (hybrid)
     for (uint32_t s = 0; s < nsegs; ++s)
       if (!segs[s]->fcol) return SDB_ERR_INVALID;
+  for (uint32_t i = 0; i < plan->nterms; ++i)  // dup terms would double
+    for (uint32_t j = i + 1; j < plan->nterms; ++j)  // count match tallies
+      if (plan->terms[i].term_idx == plan->terms[j].term_idx)
+        return SDB_ERR_INVALID;
 
   // ---- PreparePhase analogue: global stats (double -> f32, bm25.cpp) ----
   uint64_t g_dwf = plan->g_docs_with_field;

@@ -267,3 +267,20 @@ def test_for_column_codec_roundtrip():
     v = rng.integers(0, 1 << 20, 1_000_000).astype(np.int64)
     blob = sa.encode_col_i64(v)
     assert len(blob) < 2.7 * len(v), len(blob) / len(v)
+
+
+def test_segment_blob_format_stability():
+    """Golden hash of a small synthetic segment blob: the serialized format
+    (container layout + reference-exact block encodings + v2 norm streams)
+    must not drift silently across commits. Update CONSCIOUSLY on any
+    deliberate format change (bump SdbSegHeader.version too)."""
+    import hashlib
+
+    blob = sa.build_synth_segment(123, 1, 50_000, [0.07, 0.02])
+    h = hashlib.sha256(blob).hexdigest()
+    assert h == GOLDEN_BLOB_SHA, (
+        f"segment format changed: {h} (deliberate? update GOLDEN_BLOB_SHA "
+        "and bump the format version)")
+
+
+GOLDEN_BLOB_SHA = "f5687e1d73b2af986569ff30951718cbde6b4c6b399b92bb0d151bbbd4bb0d62"

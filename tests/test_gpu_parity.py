@@ -449,3 +449,30 @@ def test_count_fast(ctx):
                                    min_match=mm)
         fast = ctx.execute_count([seg], [0, 1, 2], [1.0] * 3, min_match=mm)
         assert fast == full, (mm, fast, full)
+
+
+def test_hybrid_and_minmatch(ctx):
+    """hybrid with conjunction semantics (min_match = nterms)"""
+    blob, _, _ = make_corpus(66, 300_000, [0.15, 0.1])
+    rng = np.random.default_rng(8)
+    col = rng.integers(0, 1 << 30, 300_001).astype(np.int64)
+    flo, fhi = 0, (1 << 29) - 1  # ~50%
+    seg = ctx.load_segment(blob)
+    ctx.attach_column(seg, col)
+    hits, total, bcnt, bsum = ctx.execute_topk_hybrid(
+        [seg], [0, 1], [1.0, 1.0], 100, flo, fhi, 16, min_match=2)
+    ohits, ototal, obc, obs = po.execute_topk_hybrid(
+        blob, [0, 1], [1.0, 1.0], 100, col, flo, fhi, 16, min_match=2)
+    assert total == ototal
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+    np.testing.assert_array_equal(bcnt, obc)
+    np.testing.assert_array_equal(bsum, obs)
+
+
+def test_duplicate_terms_rejected(ctx):
+    blob, _, _ = make_corpus(67, 10_000, [0.1, 0.05])
+    seg = ctx.load_segment(blob)
+    import pytest as _pt
+
+    with _pt.raises(RuntimeError):
+        ctx.execute_topk([seg], [0, 0], [1.0, 1.0], 10)
