@@ -75,6 +75,9 @@ struct SdbGpuTable {
   void* cols[16];        // raw device array or whole FoR blob
   SdbColType types[16];
   ColRef refs[16];       // device pointers into cols[] allocations
+  uint16_t max_w[16];    // FoR cols: max bit width over all groups
+  uint64_t paywords[16]; // FoR cols: total payload u32 words (for clamped
+                         // slack loads in the staged kernel)
   uint32_t ncols;
   uint64_t rows;
   uint32_t group_rows;   // shared by every FoR column (0 if none)
@@ -96,6 +99,15 @@ struct ScanArgs {
   int agg_src[SCAN_MAX_AGGS];  // 0=own col_read, 1+p=pred p's value, 9=key
   unsigned long long* out;
   unsigned long long* rows_passed;
+  // --- staged-FoR variant only (scan_agg_staged_kernel) ---
+  uint32_t chunk_rows;  // LDS staging chunk (multiple of 32)
+  uint32_t nstage;      // distinct FoR cols staged (<=3: keys, pred0, pred1)
+  int key_st;           // stage slot of the key column, -1 = unstaged
+  int pred_st[2];       // stage slot of pred 0/1's column, -1 = unstaged
+  const SdbColGroupDescDev* st_desc[3];
+  const uint32_t* st_pay[3];
+  uint64_t st_paywords[3];  // clamp for slack loads past the chunk
+  uint32_t st_lds_off[3];   // u32 offset of each stage buffer in smem
 };
 
 // desc fields are read per call from desc[rg]: the address is wave-uniform
@@ -137,6 +149,39 @@ __device__ __forceinline__ void col_read2(const ColRef& c, uint32_t rg,
   const uint64_t bit0 = (r - r0) * d.width;
   const uint64_t w0i = bit0 >> 5;
   const uint32_t w0 = w[w0i], w1 = w[w0i + 1], w2 = w[w0i + 2];
+  const uint64_t mask =
+    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
+  const uint32_t sh0 = (uint32_t)(bit0 & 31);
+  uint64_t v0 = (uint64_t)w0 >> sh0;
+  if (sh0 + d.width > 32) v0 |= (uint64_t)w1 << (32 - sh0);
+  const uint64_t bit1 = bit0 + d.width;
+  const uint32_t rel = (uint32_t)((bit1 >> 5) - w0i);  // 0, 1 or 2
+  const uint32_t wa = rel == 0 ? w0 : (rel == 1 ? w1 : w2);
+  const uint32_t wb = rel == 0 ? w1 : w2;
+  const uint32_t sh1 = (uint32_t)(bit1 & 31);
+  uint64_t v1 = (uint64_t)wa >> sh1;
+  if (sh1 + d.width > 32) v1 |= (uint64_t)wb << (32 - sh1);
+  x0 = d.base + (int64_t)(v0 & mask);
+  x1 = d.base + (int64_t)(v1 & mask);
+}
+
+// staged analogue of col_read2: the chunk's packed words sit in LDS (sw),
+// bit offsets are chunk-relative (lr = row - chunk_start; chunks start on a
+// 32-row multiple so the chunk's first bit is word-aligned). Desc fields are
+// read per call — the address is wave-uniform so the loads scalarize, same
+// rationale as col_read.
+__device__ __forceinline__ void col_read2_lds(const SdbColGroupDescDev* desc,
+                                              const uint32_t* sw, uint32_t rg,
+                                              uint32_t lr, int64_t& x0,
+                                              int64_t& x1) {
+  const SdbColGroupDescDev& d = desc[rg];
+  if (d.width == 0) {
+    x0 = x1 = d.base;
+    return;
+  }
+  const uint64_t bit0 = (uint64_t)lr * d.width;
+  const uint64_t w0i = bit0 >> 5;
+  const uint32_t w0 = sw[w0i], w1 = sw[w0i + 1], w2 = sw[w0i + 2];
   const uint64_t mask =
     d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
   const uint32_t sh0 = (uint32_t)(bit0 & 31);
@@ -434,6 +479,211 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   }
 }
 
+// FoR walker with LDS payload staging: the plain walker (scan_agg_kernel<0>)
+// is latency-bound on its per-pair global extract chains (3 dependent word
+// loads per column per pair; measured 137G rows/s vs the raw path's 233G).
+// Here each chunk's packed words for the key + first two predicate columns
+// are staged cooperatively (one coalesced burst per column), and all pair
+// extraction reads hit LDS. Zonemap skips, predicate fast path, and the
+// agg_src decode-once dedup are identical to the unstaged walker.
+__launch_bounds__(SCAN_NTHREADS) __global__
+void scan_agg_staged_kernel(ScanArgs a) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned long long* acc = (unsigned long long*)smem;  // ngroups*naggs
+  uint32_t* sw = (uint32_t*)smem;  // stage buffers live past acc
+  const uint32_t nslots = a.ngroups * a.naggs;
+  for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) acc[i] = 0;
+  __syncthreads();  // a 1-row group accumulates before any chunk barrier
+
+  const uint32_t n_rowgroups =
+    (uint32_t)((a.rows + a.group_rows - 1) / a.group_rows);
+  unsigned long long my_passed = 0;
+
+  for (uint32_t rg = blockIdx.x; rg < n_rowgroups; rg += gridDim.x) {
+    const uint64_t r0 = (uint64_t)rg * a.group_rows;
+    const uint64_t r1 = min(a.rows, r0 + a.group_rows);
+    // zonemap skip (DeadUntil analogue)
+    bool dead = false;
+    for (uint32_t p = 0; p < a.npreds; ++p) {
+      if (a.pred_col[p].desc) {
+        const SdbColGroupDescDev& d = a.pred_col[p].desc[rg];
+        switch (a.pred_op[p]) {
+          case SDB_PRED_LT: dead |= d.vmin >= a.pred_lo[p]; break;
+          case SDB_PRED_GE: dead |= d.vmax < a.pred_lo[p]; break;
+          case SDB_PRED_BETWEEN:
+            dead |= (d.vmax < a.pred_lo[p]) | (d.vmin > a.pred_hi[p]);
+            break;
+          default: break;
+        }
+      }
+    }
+    if (dead) continue;
+
+    const uint64_t glen = r1 - r0;
+    const uint64_t glen_even = glen & ~1ull;
+    for (uint64_t c0 = 0; c0 < glen_even; c0 += a.chunk_rows) {
+      const uint32_t clen =
+        (uint32_t)min((uint64_t)a.chunk_rows, glen_even - c0);
+      __syncthreads();  // prior chunk's extraction done before overwrite
+      for (uint32_t s = 0; s < a.nstage; ++s) {
+        const SdbColGroupDescDev& d = a.st_desc[s][rg];
+        if (d.width == 0) continue;
+        // chunk starts on a 32-row multiple => word-aligned bit offset
+        const uint64_t wstart = d.word_off + ((c0 * d.width) >> 5);
+        const uint32_t nw =
+          (uint32_t)(((uint64_t)clen * d.width + 31) >> 5) + 2;
+        const uint64_t limit = a.st_paywords[s] - 1;
+        uint32_t* dst = sw + a.st_lds_off[s];
+        const uint32_t* src = a.st_pay[s];
+        for (uint32_t i = threadIdx.x; i < nw; i += SCAN_NTHREADS)
+          dst[i] = src[min(wstart + i, limit)];
+      }
+      __syncthreads();
+      const uint32_t cpairs = clen >> 1;
+      for (uint32_t pr = threadIdx.x; pr < cpairs; pr += SCAN_NTHREADS) {
+        const uint32_t lr = 2 * pr;       // chunk-relative row
+        const uint64_t r = r0 + c0 + lr;  // absolute row
+        bool okv[2] = {true, true};
+        int64_t pva[2] = {0, 0}, pvb[2] = {0, 0};
+#pragma unroll
+        for (uint32_t p = 0; p < 2; ++p) {  // fast path: first two preds
+          if (p >= a.npreds) break;
+          int64_t x0, x1;
+          if (a.pred_st[p] >= 0)
+            col_read2_lds(a.st_desc[a.pred_st[p]],
+                          sw + a.st_lds_off[a.pred_st[p]], rg, lr, x0, x1);
+          else
+            col_read2(a.pred_col[p], rg, r0, r, x0, x1);
+          if (p == 0) { pva[0] = x0; pva[1] = x1; }
+          else { pvb[0] = x0; pvb[1] = x1; }
+          const int64_t xs[2] = {x0, x1};
+#pragma unroll
+          for (int e = 0; e < 2; ++e) {
+            switch (a.pred_op[p]) {
+              case SDB_PRED_LT: okv[e] &= xs[e] < a.pred_lo[p]; break;
+              case SDB_PRED_GE: okv[e] &= xs[e] >= a.pred_lo[p]; break;
+              case SDB_PRED_BETWEEN:
+                okv[e] &= (xs[e] >= a.pred_lo[p]) & (xs[e] <= a.pred_hi[p]);
+                break;
+              default: break;
+            }
+          }
+        }
+        for (uint32_t p = 2; p < a.npreds; ++p) {  // rare: >2 predicates
+#pragma unroll
+          for (int e = 0; e < 2; ++e) {
+            const int64_t x = col_read(a.pred_col[p], rg, r0, r + e);
+            switch (a.pred_op[p]) {
+              case SDB_PRED_LT: okv[e] &= x < a.pred_lo[p]; break;
+              case SDB_PRED_GE: okv[e] &= x >= a.pred_lo[p]; break;
+              case SDB_PRED_BETWEEN:
+                okv[e] &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+                break;
+              default: break;
+            }
+          }
+        }
+        if (!okv[0] && !okv[1]) continue;
+        int64_t k0, k1;
+        if (a.key_st >= 0)
+          col_read2_lds(a.st_desc[a.key_st], sw + a.st_lds_off[a.key_st],
+                        rg, lr, k0, k1);
+        else
+          col_read2(a.keys, rg, r0, r, k0, k1);
+        const int64_t ks[2] = {k0, k1};
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+          if (!okv[e]) continue;
+          ++my_passed;
+          const uint32_t grp = (uint32_t)ks[e];
+          for (uint32_t q = 0; q < a.naggs; ++q) {
+            unsigned long long* slot = &acc[grp * a.naggs + q];
+            switch (a.agg_op[q]) {
+              case SDB_AGG_COUNT:
+                atomicAdd(slot, 1ull);
+                break;
+              case SDB_AGG_SUM_I64: {
+                int64_t x;
+                switch (a.agg_src[q]) {  // decode-once dedup vs preds/key
+                  case 1: x = pva[e]; break;
+                  case 2: x = pvb[e]; break;
+                  case 9: x = ks[e]; break;
+                  default: x = col_read(a.agg_col[q], rg, r0, r + e); break;
+                }
+                atomicAdd(slot, (unsigned long long)x);
+                break;
+              }
+              case SDB_AGG_SUM_F64:
+                atomicAdd((double*)slot,
+                          (double)((const float*)a.agg_col[q].data)[r + e]);
+                break;
+            }
+          }
+        }
+      }
+    }
+    if ((glen & 1ull) && threadIdx.x == 0) {  // odd tail row: global reads
+      const uint64_t r = r1 - 1;
+      bool ok = true;
+      int64_t pv0s = 0, pv1s = 0;
+      for (uint32_t p = 0; p < a.npreds; ++p) {
+        const int64_t x = col_read(a.pred_col[p], rg, r0, r);
+        if (p == 0) pv0s = x;
+        else if (p == 1) pv1s = x;
+        switch (a.pred_op[p]) {
+          case SDB_PRED_LT: ok &= x < a.pred_lo[p]; break;
+          case SDB_PRED_GE: ok &= x >= a.pred_lo[p]; break;
+          case SDB_PRED_BETWEEN:
+            ok &= (x >= a.pred_lo[p]) & (x <= a.pred_hi[p]);
+            break;
+          default: break;
+        }
+      }
+      if (ok) {
+        ++my_passed;
+        const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
+        for (uint32_t q = 0; q < a.naggs; ++q) {
+          unsigned long long* slot = &acc[grp * a.naggs + q];
+          switch (a.agg_op[q]) {
+            case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
+            case SDB_AGG_SUM_I64: {
+              int64_t x;
+              switch (a.agg_src[q]) {
+                case 1: x = pv0s; break;
+                case 2: x = pv1s; break;
+                case 9: x = (int64_t)grp; break;
+                default: x = col_read(a.agg_col[q], rg, r0, r); break;
+              }
+              atomicAdd(slot, (unsigned long long)x);
+              break;
+            }
+            case SDB_AGG_SUM_F64:
+              atomicAdd((double*)slot,
+                        (double)((const float*)a.agg_col[q].data)[r]);
+              break;
+          }
+        }
+      }
+    }
+  }
+
+  unsigned long long wp = my_passed;
+#pragma unroll
+  for (int off = 32; off; off >>= 1) wp += __shfl_down(wp, off, 64);
+  if ((threadIdx.x & 63) == 0 && wp) atomicAdd(a.rows_passed, wp);
+  __syncthreads();
+  for (uint32_t i = threadIdx.x; i < nslots; i += SCAN_NTHREADS) {
+    const uint32_t q = i % a.naggs;
+    if (a.agg_op[q] == SDB_AGG_SUM_F64) {
+      double v;
+      __builtin_memcpy(&v, &acc[i], 8);
+      if (v != 0.0) atomicAdd((double*)&a.out[i], v);
+    } else if (acc[i]) {
+      atomicAdd(&a.out[i], acc[i]);
+    }
+  }
+}
+
 extern "C" {
 
 static void table_free_partial(SdbGpuTable* tab) {
@@ -471,6 +721,16 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
       tab->refs[c].data = (const uint8_t*)tab->cols[c] + hdr.off_payload;
       tab->refs[c].desc = (const SdbColGroupDescDev*)((const uint8_t*)
                             tab->cols[c] + hdr.off_desc);
+      // host-side pass over the group table: max width sizes the staged
+      // kernel's LDS budget; payword count bounds its slack loads
+      const SdbColGroupDescDev* hdesc =
+        (const SdbColGroupDescDev*)((const uint8_t*)cols[c].data +
+                                    hdr.off_desc);
+      uint16_t mw = 0;
+      for (uint32_t g = 0; g < hdr.ngroups; ++g)
+        if (hdesc[g].width > mw) mw = hdesc[g].width;
+      tab->max_w[c] = mw;
+      tab->paywords[c] = (hdr.size - hdr.off_payload) / 4;
     } else {
       const size_t esz = cols[c].type == SDB_COL_I64 ? 8 : 4;
       HIP_CHECK(hipMalloc(&tab->cols[c], esz * rows));
@@ -551,11 +811,59 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
     (uint32_t)((tab->rows + a.group_rows - 1) / a.group_rows);
   if (nblocks > SCAN_MAXB) nblocks = SCAN_MAXB;
   if (nblocks < 1) nblocks = 1;
-  const size_t lds = 8ull * nslots;
+  size_t lds = 8ull * nslots;
   bool any_for = a.keys.desc != nullptr;
   for (uint32_t p = 0; p < npreds; ++p) any_for |= a.pred_col[p].desc != nullptr;
   for (uint32_t q = 0; q < naggs; ++q) any_for |= a.agg_col[q].desc != nullptr;
-  if (any_for)
+
+  // staged-FoR eligibility: distinct FoR columns among {keys, pred0, pred1}
+  // get LDS chunk staging; the chunk is shrunk until acc + stage buffers fit
+  // the 160 KB LDS. SDB_SCAN_NOSTAGE falls back to the unstaged walker.
+  uint32_t st_col[3];
+  a.nstage = 0;
+  a.key_st = -1;
+  a.pred_st[0] = a.pred_st[1] = -1;
+  auto add_stage = [&](uint32_t col) -> int {
+    if (!tab->refs[col].desc) return -1;
+    for (uint32_t s = 0; s < a.nstage; ++s)
+      if (st_col[s] == col) return (int)s;
+    st_col[a.nstage] = col;
+    return (int)a.nstage++;
+  };
+  a.key_st = add_stage(group_col);
+  for (uint32_t p = 0; p < npreds && p < 2; ++p)
+    a.pred_st[p] = add_stage(preds[p].col);
+  bool staged = any_for && a.nstage > 0 && !getenv("SDB_SCAN_NOSTAGE");
+  if (staged) {
+    uint32_t C = 8192;  // rows per chunk; multiple of 32 (word alignment)
+    size_t ldsb = 0;
+    for (;;) {
+      ldsb = (8ull * nslots + 15) & ~15ull;
+      for (uint32_t s = 0; s < a.nstage; ++s) {
+        a.st_lds_off[s] = (uint32_t)(ldsb / 4);
+        const uint64_t capw = (uint64_t)C * tab->max_w[st_col[s]] / 32 + 4;
+        ldsb = (ldsb + 4 * capw + 15) & ~15ull;
+      }
+      if (ldsb <= 160 * 1024 || C <= 1024) break;
+      C >>= 1;
+    }
+    if (ldsb > 160 * 1024) {
+      staged = false;  // huge group table: unstaged walker
+    } else {
+      a.chunk_rows = C;
+      for (uint32_t s = 0; s < a.nstage; ++s) {
+        a.st_desc[s] = tab->refs[st_col[s]].desc;
+        a.st_pay[s] = (const uint32_t*)tab->refs[st_col[s]].data;
+        a.st_paywords[s] = tab->paywords[st_col[s]];
+      }
+      lds = ldsb;
+    }
+  }
+
+  if (staged)
+    hipLaunchKernelGGL(scan_agg_staged_kernel, dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
+  else if (any_for)
     hipLaunchKernelGGL((scan_agg_kernel<0>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   else

@@ -324,6 +324,123 @@ def test_scan_agg_for_codec_parity(ctx):
     lib.sdb_gpu_table_free(ctx._ctx, tab)
 
 
+def test_scan_agg_for_staged_edges(ctx):
+    """Staged-FoR walker edge cases: odd row count with a 1-row tail group,
+    a width-0 (constant) group, raw key + FoR predicate mix, a SUM over an
+    unstaged FoR column (global col_read inside the staged kernel), pred on
+    the key column (shared stage slot), and staged == unstaged
+    (SDB_SCAN_NOSTAGE) on the same inputs."""
+    import ctypes as CT
+    import os
+
+    rows = 65536 * 3 + 1  # 3 full groups + a 1-row tail group
+    ngroups = 512
+    rng = np.random.default_rng(47)
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(0, 1 << 20, rows).astype(np.int64)
+    v1[:65536] = 5000  # constant group -> width 0, staging skipped
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    v3 = rng.integers(-(1 << 30), 1 << 30, rows).astype(np.int64)
+    v1_blob = sa.encode_col_i64(v1)
+    v3_blob = sa.encode_col_i64(v3)
+
+    lib = sa.gpu()
+
+    class ColView(CT.Structure):
+        _fields_ = [("data", CT.c_void_p), ("rows", CT.c_uint64),
+                    ("type", CT.c_int)]
+
+    class PredSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int),
+                    ("ilo", CT.c_int64), ("ihi", CT.c_int64),
+                    ("flo", CT.c_float), ("fhi", CT.c_float)]
+
+    class AggSpec(CT.Structure):
+        _fields_ = [("col", CT.c_uint32), ("op", CT.c_int)]
+
+    class AggResult(CT.Structure):
+        _fields_ = [("i64", CT.c_int64), ("f64", CT.c_double)]
+
+    vb1 = np.frombuffer(v1_blob, dtype=np.uint8)
+    vb3 = np.frombuffer(v3_blob, dtype=np.uint8)
+    cols = (ColView * 4)(
+        ColView(keys.ctypes.data_as(CT.c_void_p).value, rows, 0),  # raw key
+        ColView(vb1.ctypes.data_as(CT.c_void_p).value, rows, 2),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, rows, 1),
+        ColView(vb3.ctypes.data_as(CT.c_void_p).value, rows, 2))
+    tab = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols, 4, CT.c_uint64(rows),
+                                CT.byref(tab))
+    assert rc == 0, rc
+    lo = 4000  # constant region (5000) passes
+    preds = (PredSpec * 1)(PredSpec(1, 2, lo, 0, 0, 0))  # v1 >= lo
+    aggs = (AggSpec * 3)(AggSpec(0, 0), AggSpec(3, 1), AggSpec(2, 2))
+    out = (AggResult * (ngroups * 3))()
+    passed = CT.c_uint64(0)
+
+    def run():
+        rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab, 0, ngroups, preds, 1, aggs,
+                                  3, out, CT.byref(passed))
+        assert rc == 0, rc
+        return (np.array([out[g * 3 + 0].i64 for g in range(ngroups)]),
+                np.array([out[g * 3 + 1].i64 for g in range(ngroups)]),
+                np.array([out[g * 3 + 2].f64 for g in range(ngroups)]),
+                passed.value)
+
+    gcnt, gsi, gsf, gpassed = run()
+    mask = v1 >= lo
+    ecnt = np.bincount(keys[mask], minlength=ngroups)
+    # f64 bincount is exact here: |v3| < 2^31 and |sums| < 2^53
+    esi = np.bincount(keys[mask], weights=v3[mask].astype(np.float64),
+                      minlength=ngroups).astype(np.int64)
+    esf = np.bincount(keys[mask], weights=v2[mask].astype(np.float64),
+                      minlength=ngroups)
+    assert gpassed == int(mask.sum())
+    np.testing.assert_array_equal(gcnt, ecnt)
+    np.testing.assert_array_equal(gsi, esi)
+    np.testing.assert_allclose(gsf, esf, rtol=1e-7, atol=1e-6)
+
+    # unstaged walker on the same table must agree (ints bit-exact; the f64
+    # sums only to atomic-order tolerance)
+    os.environ["SDB_SCAN_NOSTAGE"] = "1"
+    try:
+        ucnt, usi, usf, upassed = run()
+    finally:
+        del os.environ["SDB_SCAN_NOSTAGE"]
+    assert upassed == gpassed
+    np.testing.assert_array_equal(ucnt, gcnt)
+    np.testing.assert_array_equal(usi, gsi)
+    np.testing.assert_allclose(usf, gsf, rtol=1e-7, atol=1e-6)
+    lib.sdb_gpu_table_free(ctx._ctx, tab)
+
+    # FoR key with the predicate on the key column itself: shared stage
+    # slot (add_stage dedup) + agg_src=9 (key-as-value SUM)
+    keys_blob = sa.encode_col_i64(keys)
+    kb = np.frombuffer(keys_blob, dtype=np.uint8)
+    cols2 = (ColView * 2)(
+        ColView(kb.ctypes.data_as(CT.c_void_p).value, rows, 2),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, rows, 1))
+    tab2 = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols2, 2, CT.c_uint64(rows),
+                                CT.byref(tab2))
+    assert rc == 0, rc
+    preds2 = (PredSpec * 1)(PredSpec(0, 3, 100, 300, 0, 0))  # key BETWEEN
+    aggs2 = (AggSpec * 2)(AggSpec(0, 0), AggSpec(0, 1))
+    out2 = (AggResult * (ngroups * 2))()
+    rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab2, 0, ngroups, preds2, 1, aggs2,
+                              2, out2, CT.byref(passed))
+    assert rc == 0, rc
+    kmask = (keys >= 100) & (keys <= 300)
+    ecnt2 = np.bincount(keys[kmask], minlength=ngroups)
+    np.testing.assert_array_equal(
+        np.array([out2[g * 2 + 0].i64 for g in range(ngroups)]), ecnt2)
+    np.testing.assert_array_equal(
+        np.array([out2[g * 2 + 1].i64 for g in range(ngroups)]),
+        ecnt2 * np.arange(ngroups))
+    assert passed.value == int(kmask.sum())
+    lib.sdb_gpu_table_free(ctx._ctx, tab2)
+
+
 def test_full_size_properties(ctx):
     """Size-independent properties at the full headline config (100M docs,
     the oracle is too slow to replay here): exact match count vs the
