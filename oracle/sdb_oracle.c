@@ -692,6 +692,10 @@ static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
     c->num = boosts[t] * (k1 + 1.0f) * idf; /* bm25.cpp Bm25Score::num */
     c->nc = nc;
     c->nl = nl;
+    /* BM1 (k == 0, bm25.cpp:112-140 Bm1Score + :333-336 dispatch): without
+     * a filter boost every score is 0 (memset), so nothing beats the
+     * collector's FLT_MIN threshold; num = 0 reproduces that exactly */
+    if (k1 == 0.0f) c->num = 0.0f;
   }
   return 0;
 }

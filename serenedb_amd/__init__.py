@@ -423,6 +423,36 @@ class GpuContext:
         n = out_n.value
         return docs[:n], (cols[:n] if with_col else None), total.value
 
+    def execute_match_docs_multi(self, segs, term_idx, boosts, cap,
+                                 min_match=1, k1=1.2, b=0.75,
+                                 with_col=False):
+        """Streaming scan over several resident segments: the reference's
+        RunStreamingScan worker loop claims segments one at a time
+        (duckdb_search_full_scan.cpp:2370 via g.next_segment), so the
+        multi-segment form IS a per-segment loop. Emits (segment_idx, doc)
+        pairs segment-ascending then doc-ascending, column values gathered
+        per segment when attached."""
+        import numpy as np
+
+        seg_out, doc_out, col_out = [], [], []
+        total = 0
+        left = cap
+        for si, seg in enumerate(segs):
+            docs, cols, t = self.execute_match_docs(
+                seg, term_idx, boosts, left, min_match=min_match, k1=k1,
+                b=b, with_col=with_col)
+            total += t
+            seg_out.append(np.full(len(docs), si, dtype=np.uint32))
+            doc_out.append(docs)
+            if with_col:
+                col_out.append(cols)
+            left = max(left - len(docs), 0)  # keep counting total_matches
+                                             # in later segments past cap
+        segcat = np.concatenate(seg_out) if seg_out else np.zeros(0, "u4")
+        doccat = np.concatenate(doc_out) if doc_out else np.zeros(0, "u4")
+        colcat = (np.concatenate(col_out) if with_col and col_out else None)
+        return segcat, doccat, colcat, total
+
     def decode_term(self, seg, term_idx, df):
         import numpy as np
 

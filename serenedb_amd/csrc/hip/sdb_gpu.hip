@@ -465,6 +465,8 @@ struct WindowArgs {
   long long flo, fhi;     // inclusive
   uint32_t nbuckets;
   unsigned long long* bucket_out;  // [2*nbuckets]: count, sum (i64 bits)
+  uint32_t dcache_n;  // staged descriptors per term (host-shrunk so the
+                      // cache fits LDS for wide plans; <= SDB_DESC_CACHE)
 };
 
 // Persistent-range window kernel: each workgroup owns a CONTIGUOUS range of
@@ -546,10 +548,10 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     if (a.fcol)
       for (uint32_t i = tid; i < 2 * a.nbuckets; i += SDB_NTHREADS)
         lbuck[i] = 0;
-    // stage this window's descriptors: term t's next SDB_DESC_CACHE descs
+    // stage this window's descriptors: term t's next dcache_n descs
     // from its cursor, as coalesced u32 reads (7 words per desc)
     {
-      const uint32_t words_per_term = SDB_DESC_CACHE * 7u;
+      const uint32_t words_per_term = a.dcache_n * 7u;
       for (uint32_t i = tid; i < a.nterms * words_per_term;
            i += SDB_NTHREADS) {
         const uint32_t t = i / words_per_term;
@@ -560,7 +562,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                             ? te.desc_end - b0
                                             : 0);
         if (wrd < avail * 7u) {
-          ((uint32_t*)&dcache[t * SDB_DESC_CACHE])[wrd] =
+          ((uint32_t*)&dcache[t * a.dcache_n])[wrd] =
             ((const uint32_t*)&a.desc[b0])[wrd];
         }
       }
@@ -577,13 +579,26 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       const TermDev te = terms[tid];
       const uint32_t cur0w = cursors[tid];
       float ub = 0.0f;
-      for (uint32_t i = 0; i < SDB_DESC_CACHE; ++i) {
+      uint32_t i = 0;
+      for (; i < a.dcache_n; ++i) {
         if (te.desc_begin + cur0w + i >= te.desc_end) break;
-        const SdbBlockDesc d = dcache[tid * SDB_DESC_CACHE + i];
+        const SdbBlockDesc d = dcache[tid * a.dcache_n + i];
         if (d.prev_doc >= hi) break;
         const float u =
           score_one(a.scorer, te.num, te.nc, te.nl, d.max_freq, d.min_norm);
         ub = u > ub ? u : ub;
+      }
+      // a term denser than dcache_n blocks/window: the bound MUST cover the
+      // unstaged tail too or wub underestimates and WAND over-prunes
+      if (i == a.dcache_n) {
+        for (uint64_t bb = te.desc_begin + cur0w + i; bb < te.desc_end;
+             ++bb) {
+          const SdbBlockDesc d = a.desc[bb];
+          if (d.prev_doc >= hi) break;
+          const float u = score_one(a.scorer, te.num, te.nc, te.nl,
+                                    d.max_freq, d.min_norm);
+          ub = u > ub ? u : ub;
+        }
       }
       wub[tid] = ub;
     }
@@ -609,8 +624,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       uint64_t b = te.desc_begin + cur0 + wave;
       while (b < dend) {
         const uint32_t rel = (uint32_t)(b - te.desc_begin) - cur0;
-        const SdbBlockDesc d = rel < SDB_DESC_CACHE
-                                 ? dcache[t * SDB_DESC_CACHE + rel]
+        const SdbBlockDesc d = rel < a.dcache_n
+                                 ? dcache[t * a.dcache_n + rel]
                                  : a.desc[b];
         if (d.prev_doc >= hi) break;  // first doc > hi
         if (a.wand) {
@@ -648,8 +663,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           const uint64_t b2 = b + SDB_NWAVES;
           if (b2 < dend) {
             const uint32_t rel2 = (uint32_t)(b2 - te.desc_begin) - cur0;
-            const SdbBlockDesc d2 = rel2 < SDB_DESC_CACHE
-                                      ? dcache[t * SDB_DESC_CACHE + rel2]
+            const SdbBlockDesc d2 = rel2 < a.dcache_n
+                                      ? dcache[t * a.dcache_n + rel2]
                                       : a.desc[b2];
             if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
                 try_block_fused2(pl, d, d2, lane, a.norm_stream, lo, hi,
@@ -668,8 +683,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           const uint64_t nb = b + SDB_NWAVES;
           if (nb < dend) {
             const uint32_t nrel = (uint32_t)(nb - te.desc_begin) - cur0;
-            const SdbBlockDesc dn = nrel < SDB_DESC_CACHE
-                                      ? dcache[t * SDB_DESC_CACHE + nrel]
+            const SdbBlockDesc dn = nrel < a.dcache_n
+                                      ? dcache[t * a.dcache_n + nrel]
                                       : a.desc[nb];
             if (dn.prev_doc < hi) {
               const uint32_t* pfp = (const uint32_t*)(
@@ -727,8 +742,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       const uint32_t cur0 = cur;
       while (te.desc_begin + cur < te.desc_end) {
         const uint32_t rel = cur - cur0;
-        const uint32_t last = rel < SDB_DESC_CACHE
-                                ? dcache[t * SDB_DESC_CACHE + rel].last_doc
+        const uint32_t last = rel < a.dcache_n
+                                ? dcache[t * a.dcache_n + rel].last_doc
                                 : a.desc[te.desc_begin + cur].last_doc;
         if (last > hi) break;
         ++cur;
@@ -1204,6 +1219,10 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       idf[t] = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
                             ((double)g_dwt[t] + 0.5));
       num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
+      // BM1 (k == 0, bm25.cpp:112-140 Bm1Score + :333-336): without a
+      // filter boost every score is 0, so no hit beats the collector's
+      // FLT_MIN threshold; matches still count
+      if (k1 == 0.0f) num[t] = 0.0f;
       smax += num[t] > 0 ? num[t] : 0.0f;
     } else {
       // TFIDF::collect (tfidf.cpp:148-151)
@@ -1236,15 +1255,23 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_buckets, 0, 8 * 2 * SDB_MAX_BUCKETS,
                             ctx->stream));
 
-  const size_t lds_bytes = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
+  const size_t lds_fixed = SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS +
                            SDB_NWAVES * 384 * 4 + SDB_HIST_BINS * 4 +
                            (2 + SDB_NWAVES + 2 * SDB_MAX_TERMS) * 4 +
-                           8 * 2 * SDB_MAX_BUCKETS +
-                           sizeof(SdbBlockDesc) * SDB_DESC_CACHE *
-                             plan->nterms;
-  if (lds_bytes > 160 * 1024)  // LDS/CU cap: ~14 terms at the default
-    return SDB_ERR_INVALID;    // window size (shrink the desc
-                               // cache for wider plans in a later round)
+                           8 * 2 * SDB_MAX_BUCKETS;
+  // desc-cache depth shrinks for wide plans so the LDS always fits: 32
+  // descs/term up to 14 terms, down to 14 descs/term at SDB_MAX_TERMS=32
+  // (blocks beyond the cache fall back to global desc reads in-kernel)
+  uint32_t dcache_n = SDB_DESC_CACHE;
+  {
+    const size_t room = 160 * 1024 - lds_fixed;
+    const uint32_t fit = (uint32_t)(
+      room / (sizeof(SdbBlockDesc) * plan->nterms));
+    if (fit < dcache_n) dcache_n = fit;
+  }
+  if (dcache_n == 0) return SDB_ERR_INVALID;  // unreachable at nterms<=32
+  const size_t lds_bytes =
+    lds_fixed + sizeof(SdbBlockDesc) * dcache_n * plan->nterms;
 
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   for (uint32_t s = 0; s < nsegs; ++s) {
@@ -1282,6 +1309,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.k = k;
     a.smax = smax;
     a.seg_idx = s;
+    a.dcache_n = dcache_n;
     a.gthresh = ctx->d_gthresh;
     a.ghist = ctx->d_ghist;
     a.cands = ctx->d_cands;

@@ -508,6 +508,97 @@ def test_eight_term_disjunction(ctx):
     check_parity(ctx, blob, list(range(8)), [1.0] * 8, 100, min_match=3)
 
 
+def test_wide_plans(ctx):
+    """20- and 32-term plans (the LDS desc cache shrinks per term count;
+    blocks past the cache fall back to global desc reads)."""
+    rng = np.random.default_rng(71)
+    sels20 = [float(s) for s in rng.uniform(0.004, 0.05, 20)]
+    blob, _, _ = make_corpus(72, 300_000, sels20)
+    boosts = [float(b) for b in rng.uniform(0.25, 4.0, 20)]
+    check_parity(ctx, blob, list(range(20)), boosts, 250)
+    check_parity(ctx, blob, list(range(20)), [1.0] * 20, 100, min_match=5)
+
+    sels32 = [float(s) for s in rng.uniform(0.003, 0.04, 32)]
+    blob32, _, _ = make_corpus(73, 200_000, sels32)
+    check_parity(ctx, blob32, list(range(32)), [1.0] * 32, 500)
+    check_parity(ctx, blob32, list(range(32)), [1.0] * 32, 50, min_match=12)
+    # > SDB_MAX_TERMS is rejected, not silently truncated
+    import ctypes as CT
+    seg = ctx.load_segment(blob32)
+    hits = (sa.SdbScoreDoc * 10)()
+    n = CT.c_uint32(0)
+    tm = CT.c_uint64(0)
+    plan = ctx._make_plan([0] * 33, [1.0] * 33, 1, 1.2, 0.75, None)
+    rc = sa.gpu().sdb_gpu_execute_topk(
+        ctx._ctx, (CT.c_void_p * 1)(CT.c_void_p(seg.value)), 1,
+        CT.byref(plan), 10, hits, CT.byref(n), CT.byref(tm))
+    assert rc == -1  # SDB_ERR_INVALID
+
+
+def test_wand_dense_term_exactness(ctx):
+    """WAND with a term denser than the desc cache depth (>32 blocks per
+    24576-doc window at sel 0.25): the window bound must cover the unstaged
+    descriptor tail, or pruning drops true hits."""
+    blob, _, _ = make_corpus(74, 400_000, [0.25, 0.01])
+    seg = ctx.load_segment(blob)
+    for k in (10, 100):
+        base, bt = ctx.execute_topk([seg], [0, 1], [1.0, 3.0], k)
+        wand, _wt = ctx.execute_topk([seg], [0, 1], [1.0, 3.0], k, wand=True)
+        assert bt >= _wt  # wand counts only visited matches
+        np.testing.assert_array_equal(base["doc"], wand["doc"])
+        np.testing.assert_array_equal(
+            base["score"].view(np.uint32), wand["score"].view(np.uint32))
+
+
+def test_bm1_parity(ctx):
+    """BM1 (k1=0): empty top-k with exact match counting on both sides."""
+    blob, _, _ = make_corpus(75, 200_000, [0.05, 0.02])
+    seg = ctx.load_segment(blob)
+    hits, total = ctx.execute_topk([seg], [0, 1], [1.0, 1.0], 100, k1=0.0)
+    ohits, ototal = po.execute_topk([blob], [0, 1], [1.0, 1.0], 100, k1=0.0)
+    assert len(hits) == 0 and len(ohits) == 0
+    assert total == ototal > 0
+
+
+def test_match_docs_multi_segment(ctx):
+    """streaming scan across segments: per-segment loop (the reference's
+    worker claims segments one at a time), (segment, doc) ascending, column
+    gather per segment, cap cuts emission but not total_matches."""
+    seed = 76
+    sels = [0.05, 0.02]
+    n1, n2 = 150_000, 120_000
+    b1 = sa.build_synth_segment(seed, 1, n1, sels)
+    b2 = sa.build_synth_segment(seed, n1 + 1, n1 + n2, sels)
+    rng = np.random.default_rng(8)
+    col1 = rng.integers(0, 1 << 40, n1 + 1).astype(np.int64)
+    col2 = rng.integers(0, 1 << 40, n2 + 1).astype(np.int64)
+    s1 = ctx.load_segment(b1)
+    s2 = ctx.load_segment(b2)
+    ctx.attach_column(s1, col1)
+    ctx.attach_column(s2, col2)
+    segs_out, docs, vals, total = ctx.execute_match_docs_multi(
+        [s1, s2], [0, 1], [1.0, 1.0], n1 + n2, with_col=True)
+    exp = []
+    tot = 0
+    for si, (blob, col) in enumerate(((b1, col1), (b2, col2))):
+        od, ov, ot = po.execute_match_docs(blob, [0, 1], [1.0, 1.0],
+                                           n1 + n2, col=col)
+        tot += ot
+        exp.extend((si, int(d), int(v)) for d, v in zip(od, ov))
+    assert total == tot
+    got = list(zip(segs_out.tolist(), docs.tolist(), vals.tolist()))
+    assert got == exp
+    # cap smaller than the first segment's matches: emission truncated,
+    # total still covers every segment
+    cap = len([e for e in exp if e[0] == 0]) // 2
+    segs_c, docs_c, _, total_c = ctx.execute_match_docs_multi(
+        [s1, s2], [0, 1], [1.0, 1.0], cap, with_col=False)
+    assert total_c == tot
+    assert len(docs_c) == cap
+    assert got[:cap] == list(zip(segs_c.tolist(), docs_c.tolist(),
+                                 [e[2] for e in exp[:cap]]))
+
+
 def test_hybrid_multi_segment(ctx):
     """hybrid across two resident segments: bucket aggregates accumulate
     and the merged top-k equals per-segment oracle runs merged with global

@@ -157,6 +157,37 @@ def test_topk_exact_vs_brute(seed, mm):
         assert h["score"] == scores[int(h["doc"])], int(h["doc"])
 
 
+def test_bm1_zero_scores():
+    """BM1 (k=0, bm25.cpp:112-140 Bm1Score + :333-336 dispatch): without a
+    filter boost every score is memset to 0, so nothing beats the
+    collector's FLT_MIN threshold -> empty top-k, matches still counted."""
+    doc_count = 30_000
+    sels = [0.1, 0.05]
+    blob, postings, norms = synth_corpus(48, doc_count, sels)
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 50, k1=0.0)
+    _, _, nmatch = brute_topk(postings, norms, doc_count, sels, 50)
+    assert len(hits) == 0
+    assert total == nmatch
+
+
+def test_wide_plan_vs_brute():
+    """32-term disjunction and min-match on the oracle side (the GPU wide
+    -plan parity test leans on this being right)."""
+    doc_count = 20_000
+    rng = np.random.default_rng(7)
+    sels = [float(s) for s in rng.uniform(0.005, 0.06, 32)]
+    blob, postings, norms = synth_corpus(46, doc_count, sels)
+    for mm in (1, 12):
+        hits, total = po.execute_topk([blob], list(range(32)), [1.0] * 32,
+                                      80, min_match=mm)
+        order, scores, nmatch = brute_topk(postings, norms, doc_count, sels,
+                                           80, min_match=mm)
+        assert total == nmatch
+        assert [int(h["doc"]) for h in hits] == [int(d) for d in order]
+        for h in hits:
+            assert h["score"] == scores[int(h["doc"])]
+
+
 def test_mech_equals_exact():
     doc_count = 30_000
     blob, postings, norms = synth_corpus(7, doc_count, [0.08, 0.03, 0.01])
