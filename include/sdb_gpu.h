@@ -169,6 +169,52 @@ int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
 int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
                         uint32_t* docs, uint32_t* freqs);
 
+/* ---- hybrid: top-k AND pushed column predicates ----
+ * MaybeWrapColFilter semantics (duckdb_search_full_scan.cpp TableFilter
+ * wrap; index/table_filter_iterator.hpp:104-312 ColFilterChain): matches
+ * survive only if EVERY attached-column predicate passes; the analytics
+ * consumer's per-bucket COUNT/SUM aggregates over the survivors.
+ * Columns attach to numbered slots (up to 4); the original
+ * sdb_gpu_segment_attach_column is slot 0. */
+#define SDB_MAX_FILTER_COLS 4
+
+typedef enum SdbPredOp {
+  SDB_PRED_NONE = 0,
+  SDB_PRED_LT = 1,  /* col < v */
+  SDB_PRED_GE = 2,  /* col >= v */
+  SDB_PRED_BETWEEN = 3, /* lo <= col <= hi */
+} SdbPredOp;
+
+int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                  const int64_t* data);
+int sdb_gpu_segment_attach_column_slot(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                       uint32_t slot, const int64_t* data);
+
+typedef struct SdbHybridPred {
+  uint32_t slot;  /* attached column slot */
+  SdbPredOp op;   /* LT / GE / BETWEEN (table_filter_iterator.hpp ops) */
+  int64_t lo, hi; /* BETWEEN inclusive; LT/GE use lo */
+} SdbHybridPred;
+
+/* Single-predicate hybrid (BASELINE configs[3]): col0 BETWEEN [flo,fhi],
+ * buckets over its value span. */
+int sdb_gpu_execute_topk_hybrid(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                                uint32_t nsegs, const SdbQueryPlan* plan,
+                                uint32_t k, int64_t flo, int64_t fhi,
+                                uint32_t nbuckets, int64_t* bucket_count,
+                                int64_t* bucket_sum, SdbScoreDoc* hits,
+                                uint32_t* out_count, uint64_t* total_matches);
+
+/* Predicate-chain hybrid: preds[0] must be BETWEEN (it defines the bucket
+ * span); preds[1..] are additional AND-ed predicates on any attached slot.
+ * bucket_count/bucket_sum aggregate preds[0]'s column over full survivors. */
+int sdb_gpu_execute_topk_hybrid_chain(
+  SdbGpuCtx* ctx, SdbGpuSegment* const* segs, uint32_t nsegs,
+  const SdbQueryPlan* plan, uint32_t k, const SdbHybridPred* preds,
+  uint32_t npreds, uint32_t nbuckets, int64_t* bucket_count,
+  int64_t* bucket_sum, SdbScoreDoc* hits, uint32_t* out_count,
+  uint64_t* total_matches);
+
 /* ---- columnar scan -> filter -> hash aggregate ---- */
 
 /* Column layout: dense device-resident columns, or FoR/bitpack row groups
@@ -192,13 +238,6 @@ typedef struct SdbGpuTable SdbGpuTable;
 int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
                        uint32_t ncols, uint64_t rows, SdbGpuTable** out);
 int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab);
-
-typedef enum SdbPredOp {
-  SDB_PRED_NONE = 0,
-  SDB_PRED_LT = 1,  /* col < v */
-  SDB_PRED_GE = 2,  /* col >= v */
-  SDB_PRED_BETWEEN = 3, /* lo <= col <= hi */
-} SdbPredOp;
 
 typedef struct SdbPredSpec {
   uint32_t col;

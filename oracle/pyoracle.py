@@ -334,6 +334,39 @@ def execute_topk_hybrid(blob, term_idx, boosts, k, col, flo, fhi, nbuckets,
     return _hits_to_np(hits, out_count.value), total.value, bcnt, bsum
 
 
+def execute_topk_hybrid_chain(blob, term_idx, boosts, k, cols, ops, los,
+                              his, nbuckets, min_match=1, k1=1.2, b=0.75):
+    """Predicate-chain hybrid: cols[0] carries the BETWEEN bucket span
+    (ops[0] must be 3); cols[1..] AND-narrow (1=LT 2=GE 3=BETWEEN)."""
+    lib().o_set_scorer(0)  # these entry points are BM25-only
+    buf = np.frombuffer(blob, dtype=np.uint8)
+    ti = _u32arr(term_idx)
+    bo = np.ascontiguousarray(boosts, dtype=np.float32)
+    ncols = len(cols)
+    keep = [np.ascontiguousarray(c, dtype=np.int64) for c in cols]
+    PI64 = C.POINTER(C.c_int64)
+    col_ptrs = (PI64 * ncols)(*[c.ctypes.data_as(PI64) for c in keep])
+    ops_arr = (C.c_int * ncols)(*[int(o) for o in ops])
+    los_arr = np.ascontiguousarray(los, dtype=np.int64)
+    his_arr = np.ascontiguousarray(his, dtype=np.int64)
+    bcnt = np.zeros(nbuckets, dtype=np.int64)
+    bsum = np.zeros(nbuckets, dtype=np.int64)
+    hits = (OScoreDoc * k)()
+    out_count = C.c_uint32(0)
+    total = C.c_uint64(0)
+    rc = lib().o_execute_topk_hybrid_chain(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)),
+        ti.ctypes.data_as(PU32), bo.ctypes.data_as(C.POINTER(C.c_float)),
+        C.c_uint32(len(ti)), C.c_uint32(min_match), C.c_float(k1),
+        C.c_float(b), C.c_uint32(k), col_ptrs, ops_arr,
+        los_arr.ctypes.data_as(PI64), his_arr.ctypes.data_as(PI64),
+        C.c_uint32(ncols), C.c_uint32(nbuckets),
+        bcnt.ctypes.data_as(PI64), bsum.ctypes.data_as(PI64),
+        hits, C.byref(out_count), C.byref(total))
+    assert rc == 0, rc
+    return _hits_to_np(hits, out_count.value), total.value, bcnt, bsum
+
+
 def execute_match_docs(blob, term_idx, boosts, cap, col=None, min_match=1,
                        k1=1.2, b=0.75):
     """Streaming match emission (doc-ascending) + optional column gather."""

@@ -542,6 +542,12 @@ typedef struct {
   uint32_t nbuckets;  /* bucket = (col-lo)*nbuckets/(hi-lo+1) */
   int64_t* bucket_count;
   int64_t* bucket_sum;
+  /* extra AND-ed chain predicates (ColFilterChain,
+   * table_filter_iterator.hpp:104-312); ops: 1=LT 2=GE 3=BETWEEN */
+  uint32_t nextra;
+  const int64_t* xcol[3];
+  int xop[3];
+  int64_t xlo[3], xhi[3];
 } OHybrid;
 
 static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
@@ -604,6 +610,17 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
           const int64_t v = hy->col[doc];
           if (v < hy->lo || v > hy->hi) {
             --matches; /* popcount above pre-counted this doc */
+            continue;
+          }
+          int pass = 1;
+          for (uint32_t x = 0; x < hy->nextra; ++x) {
+            const int64_t xv = hy->xcol[x][doc];
+            if (hy->xop[x] == 1) pass &= xv < hy->xlo[x];
+            else if (hy->xop[x] == 2) pass &= xv >= hy->xlo[x];
+            else pass &= (xv >= hy->xlo[x]) && (xv <= hy->xhi[x]);
+          }
+          if (!pass) {
+            --matches;
             continue;
           }
           const uint64_t span = (uint64_t)(hy->hi - hy->lo) + 1;
@@ -784,15 +801,18 @@ int o_execute_topk(const OSegBlob* segs, uint32_t nsegs,
  * duckdb_search_full_scan.cpp:1900-1912, table_filter_iterator.hpp:229-312,
  * aggregate consumer external). col indexed by LOCAL doc id of the single
  * provided segment. */
-int o_execute_topk_hybrid(const void* blob, uint64_t size,
-                          const uint32_t* term_idx, const float* boosts,
-                          uint32_t nterms, uint32_t min_match, float k1,
-                          float b, uint64_t g_dwf, const uint64_t* g_dwt,
-                          uint64_t g_ttf, uint32_t k, const int64_t* col,
-                          int64_t flo, int64_t fhi, uint32_t nbuckets,
-                          int64_t* bucket_count, int64_t* bucket_sum,
-                          OScoreDoc* hits, uint32_t* out_count,
-                          uint64_t* total_matches) {
+static int o_hybrid_impl(const void* blob, uint64_t size,
+                         const uint32_t* term_idx, const float* boosts,
+                         uint32_t nterms, uint32_t min_match, float k1,
+                         float b, uint64_t g_dwf, const uint64_t* g_dwt,
+                         uint64_t g_ttf, uint32_t k, const int64_t* col,
+                         int64_t flo, int64_t fhi, uint32_t nbuckets,
+                         const int64_t* const* chain_cols,
+                         const int* chain_ops, const int64_t* chain_los,
+                         const int64_t* chain_his, uint32_t ncols_chain,
+                         int64_t* bucket_count, int64_t* bucket_sum,
+                         OScoreDoc* hits, uint32_t* out_count,
+                         uint64_t* total_matches) {
   SdbSegmentView v;
   int rc = o_segment_parse(blob, size, &v);
   if (rc) return rc;
@@ -813,7 +833,21 @@ int o_execute_topk_hybrid(const void* blob, uint64_t size,
   if (rc) { free(cur); free(auto_dwt); return rc; }
   memset(bucket_count, 0, sizeof(int64_t) * nbuckets);
   memset(bucket_sum, 0, sizeof(int64_t) * nbuckets);
-  OHybrid hy = {col, flo, fhi, nbuckets, bucket_count, bucket_sum};
+  OHybrid hy;
+  memset(&hy, 0, sizeof(hy));
+  hy.col = col;
+  hy.lo = flo;
+  hy.hi = fhi;
+  hy.nbuckets = nbuckets;
+  hy.bucket_count = bucket_count;
+  hy.bucket_sum = bucket_sum;
+  for (uint32_t x = 0; x + 1 < ncols_chain; ++x) {
+    hy.xcol[x] = chain_cols[x + 1];
+    hy.xop[x] = chain_ops[x + 1];
+    hy.xlo[x] = chain_los[x + 1];
+    hy.xhi[x] = chain_his[x + 1];
+  }
+  hy.nextra = ncols_chain ? ncols_chain - 1 : 0;
   OCandVec cands = {0, 0, 0};
   uint64_t matches =
     o_exec_range(cur, nterms, v.norms, min_match ? min_match : 1, 1,
@@ -831,6 +865,37 @@ int o_execute_topk_hybrid(const void* blob, uint64_t size,
   *total_matches = matches;
   free(cands.v);
   return 0;
+}
+
+int o_execute_topk_hybrid(const void* blob, uint64_t size,
+                          const uint32_t* term_idx, const float* boosts,
+                          uint32_t nterms, uint32_t min_match, float k1,
+                          float b, uint64_t g_dwf, const uint64_t* g_dwt,
+                          uint64_t g_ttf, uint32_t k, const int64_t* col,
+                          int64_t flo, int64_t fhi, uint32_t nbuckets,
+                          int64_t* bucket_count, int64_t* bucket_sum,
+                          OScoreDoc* hits, uint32_t* out_count,
+                          uint64_t* total_matches) {
+  return o_hybrid_impl(blob, size, term_idx, boosts, nterms, min_match, k1,
+                       b, g_dwf, g_dwt, g_ttf, k, col, flo, fhi, nbuckets,
+                       NULL, NULL, NULL, NULL, 0, bucket_count, bucket_sum,
+                       hits, out_count, total_matches);
+}
+
+/* Chain form: cols[0] carries the BETWEEN (los[0]..his[0]) bucket span
+ * (ops[0] must be 3); cols[1..ncols) AND-narrow with ops LT/GE/BETWEEN. */
+int o_execute_topk_hybrid_chain(
+  const void* blob, uint64_t size, const uint32_t* term_idx,
+  const float* boosts, uint32_t nterms, uint32_t min_match, float k1,
+  float b, uint32_t k, const int64_t* const* cols, const int* ops,
+  const int64_t* los, const int64_t* his, uint32_t ncols, uint32_t nbuckets,
+  int64_t* bucket_count, int64_t* bucket_sum, OScoreDoc* hits,
+  uint32_t* out_count, uint64_t* total_matches) {
+  if (ncols == 0 || ncols > 4 || ops[0] != 3) return -1;
+  return o_hybrid_impl(blob, size, term_idx, boosts, nterms, min_match, k1,
+                       b, 0, NULL, 0, k, cols[0], los[0], his[0], nbuckets,
+                       cols, ops, los, his, ncols, bucket_count, bucket_sum,
+                       hits, out_count, total_matches);
 }
 
 /* STREAMING scan: emit every matching doc id ascending, plus (optionally)

@@ -355,12 +355,13 @@ class GpuContext:
                             dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value
 
-    def attach_column(self, seg, col):
+    def attach_column(self, seg, col, slot=0):
         import numpy as np
 
         col = np.ascontiguousarray(col, dtype=np.int64)
-        rc = self._lib.sdb_gpu_segment_attach_column(
-            self._ctx, seg, col.ctypes.data_as(C.POINTER(C.c_int64)))
+        rc = self._lib.sdb_gpu_segment_attach_column_slot(
+            self._ctx, seg, C.c_uint32(slot),
+            col.ctypes.data_as(C.POINTER(C.c_int64)))
         if rc != 0:
             raise RuntimeError(f"attach_column rc={rc}")
 
@@ -386,6 +387,43 @@ class GpuContext:
             hits, C.byref(out_count), C.byref(total))
         if rc != 0:
             raise RuntimeError(f"sdb_gpu_execute_topk_hybrid rc={rc}")
+        n = out_count.value
+        dt = np.dtype([("score", "f4"), ("doc", "u4"), ("segment", "u4")])
+        res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),
+                            dtype=dt).copy() if n else np.zeros(0, dtype=dt)
+        return res, total.value, bcnt, bsum
+
+    def execute_topk_hybrid_chain(self, segs, term_idx, boosts, k, preds,
+                                  nbuckets, min_match=1, k1=1.2, b=0.75,
+                                  global_stats=None):
+        """preds: list of (slot, op, lo, hi); preds[0] must be BETWEEN
+        (op=3) and defines the bucket span (ColFilterChain semantics,
+        table_filter_iterator.hpp:104-312)."""
+        import numpy as np
+
+        class _HP(C.Structure):
+            _fields_ = [("slot", C.c_uint32), ("op", C.c_int),
+                        ("lo", C.c_int64), ("hi", C.c_int64)]
+
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b,
+                               global_stats)
+        seg_arr = (C.c_void_p * len(segs))(
+            *[C.c_void_p(s.value) for s in segs])
+        parr = (_HP * len(preds))(
+            *[_HP(s, o, int(lo), int(hi)) for s, o, lo, hi in preds])
+        hits = (SdbScoreDoc * k)()
+        out_count = C.c_uint32(0)
+        total = C.c_uint64(0)
+        bcnt = np.zeros(nbuckets, dtype=np.int64)
+        bsum = np.zeros(nbuckets, dtype=np.int64)
+        PI64 = C.POINTER(C.c_int64)
+        rc = self._lib.sdb_gpu_execute_topk_hybrid_chain(
+            self._ctx, seg_arr, len(segs), C.byref(plan), C.c_uint32(k),
+            parr, C.c_uint32(len(preds)), C.c_uint32(nbuckets),
+            bcnt.ctypes.data_as(PI64), bsum.ctypes.data_as(PI64),
+            hits, C.byref(out_count), C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_topk_hybrid_chain rc={rc}")
         n = out_count.value
         dt = np.dtype([("score", "f4"), ("doc", "u4"), ("segment", "u4")])
         res = np.frombuffer(C.string_at(hits, C.sizeof(SdbScoreDoc) * n),

@@ -75,7 +75,9 @@ struct SdbGpuSegment {
   SdbBlockDesc* desc;   // device
   uint8_t* payload;     // device
   uint32_t* norms;      // device, doc_count+1
-  long long* fcol;      // device, doc_count+1 (hybrid filter column) or null
+  long long* fcols[4];  // device, doc_count+1 per attached filter-column
+                        // slot (SDB_MAX_FILTER_COLS) or null; slot 0 is
+                        // the classic hybrid/bucket column
   SdbTermEntry* terms_host;  // host copy of term table
   SdbSegHeader hdr;     // host copy
 };
@@ -462,7 +464,13 @@ struct WindowArgs {
   // per-bucket COUNT/SUM (TableFilterDocIterator semantics,
   // index/table_filter_iterator.hpp:104-312)
   const long long* fcol;  // device, indexed by doc; NULL = no filter
-  long long flo, fhi;     // inclusive
+  long long flo, fhi;     // inclusive (preds[0]: BETWEEN, the bucket span)
+  // extra AND-ed chain predicates (ColFilterChain,
+  // table_filter_iterator.hpp:104-312): ops use SdbPredOp values
+  uint32_t nfx;
+  const long long* fxc[3];
+  int fxop[3];
+  long long fxlo[3], fxhi[3];
   uint32_t nbuckets;
   unsigned long long* bucket_out;  // [2*nbuckets]: count, sum (i64 bits)
   uint32_t dcache_n;  // staged descriptors per term (host-shrunk so the
@@ -787,6 +795,17 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         if (a.fcol) {
           const long long vv = a.fcol[lo + off];
           if (vv < a.flo || vv > a.fhi) {
+            cwin[off] = 0;
+            continue;
+          }
+          bool pass = true;  // AND chain over the extra predicates
+          for (uint32_t x = 0; x < a.nfx; ++x) {
+            const long long xv = a.fxc[x][lo + off];
+            if (a.fxop[x] == 1) pass &= xv < a.fxlo[x];
+            else if (a.fxop[x] == 2) pass &= xv >= a.fxlo[x];
+            else pass &= (xv >= a.fxlo[x]) & (xv <= a.fxhi[x]);
+          }
+          if (!pass) {
             cwin[off] = 0;
             continue;
           }
@@ -1152,7 +1171,8 @@ int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
   hipFree(seg->desc);
   hipFree(seg->payload);
   hipFree(seg->norms);
-  if (seg->fcol) hipFree(seg->fcol);
+  for (int i = 0; i < 4; ++i)
+    if (seg->fcols[i]) hipFree(seg->fcols[i]);
   std::free(seg->terms_host);
   delete seg;
   return SDB_OK;
@@ -1160,8 +1180,9 @@ int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
 
 static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                           uint32_t nsegs, const SdbQueryPlan* plan,
-                          uint32_t k, int hybrid, long long h_flo,
-                          long long h_fhi, uint32_t h_nbuckets,
+                          uint32_t k, int hybrid,
+                          const SdbHybridPred* hpreds, uint32_t nhp,
+                          uint32_t h_nbuckets,
                           int64_t* bucket_count, int64_t* bucket_sum,
                           SdbScoreDoc* hits, uint32_t* out_count,
                           uint64_t* total_matches, int count_only = 0) {
@@ -1170,9 +1191,19 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     return SDB_ERR_INVALID;
   /* hits == NULL: candidate-only mode (streaming match emission reads the
    * device candidate buffer itself; no host select) */
-  if (hybrid)
-    for (uint32_t s = 0; s < nsegs; ++s)
-      if (!segs[s]->fcol) return SDB_ERR_INVALID;
+  if (hybrid) {
+    if (!hpreds || nhp == 0 || nhp > SDB_MAX_FILTER_COLS ||
+        hpreds[0].op != SDB_PRED_BETWEEN)  // preds[0] defines bucket span
+      return SDB_ERR_INVALID;
+    for (uint32_t x = 0; x < nhp; ++x) {
+      if (hpreds[x].slot >= 4 ||
+          (hpreds[x].op != SDB_PRED_LT && hpreds[x].op != SDB_PRED_GE &&
+           hpreds[x].op != SDB_PRED_BETWEEN))
+        return SDB_ERR_INVALID;
+      for (uint32_t s = 0; s < nsegs; ++s)
+        if (!segs[s]->fcols[hpreds[x].slot]) return SDB_ERR_INVALID;
+    }
+  }
   for (uint32_t i = 0; i < plan->nterms; ++i)  // dup terms would double
     for (uint32_t j = i + 1; j < plan->nterms; ++j)  // count match tallies
       if (plan->terms[i].term_idx == plan->terms[j].term_idx)
@@ -1291,9 +1322,16 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                              sizeof(TermDev) * plan->nterms,
                              hipMemcpyHostToDevice, ctx->stream));
     WindowArgs a{};
-    a.fcol = hybrid ? seg->fcol : nullptr;
-    a.flo = h_flo;
-    a.fhi = h_fhi;
+    a.fcol = hybrid ? seg->fcols[hpreds[0].slot] : nullptr;
+    a.flo = hybrid ? hpreds[0].lo : 0;
+    a.fhi = hybrid ? hpreds[0].hi : 0;
+    a.nfx = hybrid ? nhp - 1 : 0;
+    for (uint32_t x = 0; hybrid && x + 1 < nhp; ++x) {
+      a.fxc[x] = seg->fcols[hpreds[x + 1].slot];
+      a.fxop[x] = (int)hpreds[x + 1].op;
+      a.fxlo[x] = hpreds[x + 1].lo;
+      a.fxhi[x] = hpreds[x + 1].hi;
+    }
     a.nbuckets = h_nbuckets;
     a.bucket_out = ctx->d_buckets;
     a.desc = seg->desc;
@@ -1428,7 +1466,8 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                          uint32_t nsegs, const SdbQueryPlan* plan, uint32_t k,
                          SdbScoreDoc* hits, uint32_t* out_count,
                          uint64_t* total_matches) {
-  return exec_topk_impl(ctx, segs, nsegs, plan, k, 0, 0, 0, 0, nullptr,
+  return exec_topk_impl(ctx, segs, nsegs, plan, k, 0, nullptr, 0, 0,
+                        nullptr,
                         nullptr, hits, out_count, total_matches);
 }
 
@@ -1442,7 +1481,8 @@ int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   SdbQueryPlan p = *plan;
   p.wand = 0;
   uint32_t n = 0;
-  return exec_topk_impl(ctx, segs, nsegs, &p, 1, 0, 0, 0, 0, nullptr,
+  return exec_topk_impl(ctx, segs, nsegs, &p, 1, 0, nullptr, 0, 0,
+                        nullptr,
                         nullptr, /*hits=*/nullptr, &n, total_matches,
                         /*count_only=*/1);
 }
@@ -1450,16 +1490,21 @@ int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
 // Attach the hybrid filter column (i64[doc_count+1], index 0 unused) to a
 // resident segment — the analytics column the reference's MaybeWrapColFilter
 // pushes into the scan (duckdb_search_full_scan.cpp:1900-1912).
-int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
-                                  const int64_t* data) {
-  if (!ctx || !seg || !data) return SDB_ERR_INVALID;
-  if (!seg->fcol)
-    HIP_CHECK(
-      hipMalloc(&seg->fcol, 8ull * ((uint64_t)seg->hdr.doc_count + 1)));
-  HIP_CHECK(hipMemcpy(seg->fcol, data,
+int sdb_gpu_segment_attach_column_slot(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                       uint32_t slot, const int64_t* data) {
+  if (!ctx || !seg || !data || slot >= 4) return SDB_ERR_INVALID;
+  if (!seg->fcols[slot])
+    HIP_CHECK(hipMalloc(&seg->fcols[slot],
+                        8ull * ((uint64_t)seg->hdr.doc_count + 1)));
+  HIP_CHECK(hipMemcpy(seg->fcols[slot], data,
                       8ull * ((uint64_t)seg->hdr.doc_count + 1),
                       hipMemcpyHostToDevice));
   return SDB_OK;
+}
+
+int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                  const int64_t* data) {
+  return sdb_gpu_segment_attach_column_slot(ctx, seg, 0, data);
 }
 
 // Hybrid: BM25 top-k AND col BETWEEN [flo,fhi] + per-bucket COUNT/SUM over
@@ -1472,8 +1517,23 @@ int sdb_gpu_execute_topk_hybrid(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                                 uint32_t* out_count,
                                 uint64_t* total_matches) {
   if (nbuckets == 0 || nbuckets > SDB_MAX_BUCKETS) return SDB_ERR_INVALID;
-  return exec_topk_impl(ctx, segs, nsegs, plan, k, 1, flo, fhi, nbuckets,
+  const SdbHybridPred p0 = {0, SDB_PRED_BETWEEN, flo, fhi};
+  return exec_topk_impl(ctx, segs, nsegs, plan, k, 1, &p0, 1, nbuckets,
                         bucket_count, bucket_sum, hits, out_count,
+                        total_matches);
+}
+
+// Predicate-chain hybrid (ColFilterChain, table_filter_iterator.hpp:104-312):
+// preds[0] BETWEEN defines the bucket span; preds[1..] AND-narrow further.
+int sdb_gpu_execute_topk_hybrid_chain(
+  SdbGpuCtx* ctx, SdbGpuSegment* const* segs, uint32_t nsegs,
+  const SdbQueryPlan* plan, uint32_t k, const SdbHybridPred* preds,
+  uint32_t npreds, uint32_t nbuckets, int64_t* bucket_count,
+  int64_t* bucket_sum, SdbScoreDoc* hits, uint32_t* out_count,
+  uint64_t* total_matches) {
+  if (nbuckets == 0 || nbuckets > SDB_MAX_BUCKETS) return SDB_ERR_INVALID;
+  return exec_topk_impl(ctx, segs, nsegs, plan, k, 1, preds, npreds,
+                        nbuckets, bucket_count, bucket_sum, hits, out_count,
                         total_matches);
 }
 
@@ -1492,8 +1552,8 @@ int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
   // run with a k no window can reach: threshold stays 0, all matches append
   uint32_t dummy_n = 0;
   SdbGpuSegment* segs1[1] = {seg};
-  int rc = exec_topk_impl(ctx, segs1, 1, plan, 0xFFFFFFFFu, 0, 0, 0, 0,
-                          nullptr, nullptr, /*hits=*/nullptr, &dummy_n,
+  int rc = exec_topk_impl(ctx, segs1, 1, plan, 0xFFFFFFFFu, 0, nullptr, 0,
+                          0, nullptr, nullptr, /*hits=*/nullptr, &dummy_n,
                           total_matches);
   if (rc) return rc;
   // all candidates are still on the device; re-read and order by doc
@@ -1508,7 +1568,7 @@ int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
             });
   const uint64_t n = std::min<uint64_t>(cap, cands.size());
   for (uint64_t i = 0; i < n; ++i) docs_out[i] = cands[i].doc;
-  if (col_out && seg->fcol && n) {
+  if (col_out && seg->fcols[0] && n) {
     // device gather: upload the doc-ordered hit ids, one gather kernel,
     // one D2H of the values (HitBatcher's dense/scatter gather analogue)
     uint32_t* d_docs;
@@ -1518,7 +1578,7 @@ int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
     HIP_CHECK(hipMemcpy(d_docs, docs_out, 4 * n, hipMemcpyHostToDevice));
     const uint32_t nb = (uint32_t)((n + 255) / 256);
     hipLaunchKernelGGL(gather_col_kernel, dim3(nb), dim3(256), 0,
-                       ctx->stream, d_docs, seg->fcol, d_vals, n);
+                       ctx->stream, d_docs, seg->fcols[0], d_vals, n);
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipMemcpyAsync(col_out, d_vals, 8 * n, hipMemcpyDeviceToHost,
                              ctx->stream));

@@ -550,6 +550,60 @@ def test_wand_dense_term_exactness(ctx):
             base["score"].view(np.uint32), wand["score"].view(np.uint32))
 
 
+def test_hybrid_chain_parity(ctx):
+    """Predicate-chain hybrid vs oracle: 2- and 3-pred chains over distinct
+    column slots (BETWEEN + GE + LT), plus vacuous chain == single-pred."""
+    doc_count = 300_000
+    blob, _, _ = make_corpus(77, doc_count, [0.08, 0.04, 0.02])
+    rng = np.random.default_rng(11)
+    span = 1 << 30
+    col0 = rng.integers(0, span, doc_count + 1).astype(np.int64)
+    col1 = rng.integers(0, span, doc_count + 1).astype(np.int64)
+    col2 = rng.integers(0, span, doc_count + 1).astype(np.int64)
+    seg = ctx.load_segment(blob)
+    ctx.attach_column(seg, col0, slot=0)
+    ctx.attach_column(seg, col1, slot=1)
+    ctx.attach_column(seg, col2, slot=2)
+    flo, fhi = int(span * 0.1), int(span * 0.9) - 1
+    g1 = int(span * 0.4)
+    l2 = int(span * 0.7)
+    nb = 32
+    for preds, ops, los, his in (
+        ([(0, 3, flo, fhi), (1, 2, g1, 0)], [3, 2], [flo, g1], [fhi, 0]),
+        ([(0, 3, flo, fhi), (1, 2, g1, 0), (2, 1, l2, 0)],
+         [3, 2, 1], [flo, g1, l2], [fhi, 0, 0]),
+    ):
+        hits, total, bcnt, bsum = ctx.execute_topk_hybrid_chain(
+            [seg], [0, 1, 2], [1.0] * 3, 500, preds, nb)
+        cols = [col0, col1, col2][:len(preds)]
+        ohits, ototal, ocnt, osum = po.execute_topk_hybrid_chain(
+            blob, [0, 1, 2], [1.0] * 3, 500, cols, ops, los, his, nb)
+        assert total == ototal
+        np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+        np.testing.assert_array_equal(
+            hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
+        np.testing.assert_array_equal(bcnt, ocnt)
+        np.testing.assert_array_equal(bsum, osum)
+    # vacuous extra pred == single-pred hybrid (same GPU box, same inputs)
+    h1, t1, c1, s1 = ctx.execute_topk_hybrid(
+        [seg], [0, 1, 2], [1.0] * 3, 200, flo, fhi, nb)
+    h2, t2, c2, s2 = ctx.execute_topk_hybrid_chain(
+        [seg], [0, 1, 2], [1.0] * 3, 200, [(0, 3, flo, fhi), (1, 2, 0, 0)],
+        nb)
+    assert t1 == t2
+    np.testing.assert_array_equal(h1["doc"], h2["doc"])
+    np.testing.assert_array_equal(c1, c2)
+    np.testing.assert_array_equal(s1, s2)
+    # unattached slot / non-BETWEEN primary rejected
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        ctx.execute_topk_hybrid_chain([seg], [0], [1.0], 10,
+                                      [(3, 3, 0, 10)], nb)
+    with _pytest.raises(RuntimeError):
+        ctx.execute_topk_hybrid_chain([seg], [0], [1.0], 10,
+                                      [(0, 2, 0, 0)], nb)
+
+
 def test_bm1_parity(ctx):
     """BM1 (k1=0): empty top-k with exact match counting on both sides."""
     blob, _, _ = make_corpus(75, 200_000, [0.05, 0.02])

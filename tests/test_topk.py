@@ -157,6 +157,49 @@ def test_topk_exact_vs_brute(seed, mm):
         assert h["score"] == scores[int(h["doc"])], int(h["doc"])
 
 
+def test_hybrid_chain_vs_brute():
+    """Predicate-chain hybrid (ColFilterChain AND semantics): oracle chain
+    vs independent numpy evaluation, and a vacuous extra predicate must
+    reproduce the single-predicate hybrid exactly."""
+    doc_count = 25_000
+    sels = [0.1, 0.05]
+    blob, postings, norms = synth_corpus(49, doc_count, sels)
+    rng = np.random.default_rng(10)
+    span = 1 << 20
+    col0 = rng.integers(0, span, doc_count + 1).astype(np.int64)
+    col1 = rng.integers(0, span, doc_count + 1).astype(np.int64)
+    flo, fhi = int(span * 0.2), int(span * 0.8) - 1
+    glim = int(span * 0.5)
+    nb = 16
+    hits, total, bcnt, bsum = po.execute_topk_hybrid_chain(
+        blob, [0, 1], [1.0, 1.0], 50, [col0, col1], [3, 2],
+        [flo, glim], [fhi, 0], nb)
+    # numpy: survivors = matches AND col0 BETWEEN AND col1 >= glim
+    order, scores, _ = brute_topk(postings, norms, doc_count, sels, 10**9)
+    surv = [d for d in order
+            if flo <= col0[d] <= fhi and col1[d] >= glim]
+    assert total == len(surv)
+    assert [int(h["doc"]) for h in hits] == surv[:50]
+    ecnt = np.zeros(nb, dtype=np.int64)
+    esum = np.zeros(nb, dtype=np.int64)
+    for d in surv:
+        b_ = min((col0[d] - flo) * nb // (fhi - flo + 1), nb - 1)
+        ecnt[b_] += 1
+        esum[b_] += col0[d]
+    np.testing.assert_array_equal(bcnt, ecnt)
+    np.testing.assert_array_equal(bsum, esum)
+    # vacuous extra pred == single-pred hybrid
+    h1, t1, c1, s1 = po.execute_topk_hybrid(
+        blob, [0, 1], [1.0, 1.0], 50, col0, flo, fhi, nb)
+    h2, t2, c2, s2 = po.execute_topk_hybrid_chain(
+        blob, [0, 1], [1.0, 1.0], 50, [col0, col1], [3, 2],
+        [flo, 0], [fhi, 0], nb)
+    assert t1 == t2
+    np.testing.assert_array_equal(h1["doc"], h2["doc"])
+    np.testing.assert_array_equal(c1, c2)
+    np.testing.assert_array_equal(s1, s2)
+
+
 def test_bm1_zero_scores():
     """BM1 (k=0, bm25.cpp:112-140 Bm1Score + :333-336 dispatch): without a
     filter boost every score is memset to 0, so nothing beats the
