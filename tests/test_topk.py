@@ -20,9 +20,12 @@ from oracle import pyoracle as po
 GOLDEN = os.path.join(os.path.dirname(__file__), "golden")
 
 
-def build_term_corpus(docs_fields):
+def build_term_corpus(docs_fields, unit_features=False):
     """docs_fields: list (per doc, 1-based ids implied) of token lists.
-    Returns (blob, vocab dict term->idx, norms)."""
+    Returns (blob, vocab dict term->idx, norms). unit_features mirrors the
+    reference's test_query corpora indexed WITHOUT freq/norm features
+    (bm25_test.cpp:465-500 StringField insert): freq=1 per posting and a
+    constant norm, making BM25 rank by idf sums with doc-order ties."""
     vocab = {}
     for f in docs_fields:
         for t in f:
@@ -30,7 +33,7 @@ def build_term_corpus(docs_fields):
     postings = [[] for _ in vocab]
     for d, f in enumerate(docs_fields, start=1):
         for t in sorted(set(f), key=lambda x: vocab[x]):
-            postings[vocab[t]].append((d, f.count(t)))
+            postings[vocab[t]].append((d, 1 if unit_features else f.count(t)))
     plist = []
     for pl in postings:
         docs = np.array([d for d, _ in pl], dtype=np.uint32)
@@ -38,7 +41,7 @@ def build_term_corpus(docs_fields):
         plist.append((docs, freqs))
     norms = np.zeros(len(docs_fields) + 1, dtype=np.uint32)
     for d, f in enumerate(docs_fields, start=1):
-        norms[d] = len(f)
+        norms[d] = 1 if unit_features else len(f)
     blob = sa.build_segment(len(docs_fields), plist, norms)
     return blob, vocab, norms
 
@@ -71,21 +74,22 @@ def numpy_bm25_topk(docs_fields, terms, k1=1.2, b=0.75, fp64=False):
     return order, scores
 
 
-@pytest.mark.parametrize("case_idx", [0, 1, 2, 3, 4])
+@pytest.mark.parametrize("case_idx", [0, 1, 2, 3, 4, 5, 6, 7])
 def test_golden_rank_order(case_idx):
     g = json.load(open(os.path.join(GOLDEN,
                                     "bm25_simple_sequential_order.json")))
     docs_fields = g["docs"]
     case = g["cases"][case_idx]
     scorer = case.get("scorer", "bm25")
-    blob, vocab, _ = build_term_corpus(docs_fields)
+    unit = case.get("features") == "no_freq_no_norm"
+    blob, vocab, _ = build_term_corpus(docs_fields, unit_features=unit)
     term_idx = [vocab[t] for t in case["terms"]]
     boosts = [1.0] * len(term_idx)
     hits, total = po.execute_topk([blob], term_idx, boosts, k=8,
                                   scorer=scorer)
     got_seq = [int(h["doc"]) - 1 for h in hits]  # seq = doc-1 in this corpus
     assert got_seq == case["expected_seq_order"], case["cite"]
-    if scorer == "bm25":
+    if scorer == "bm25" and not unit:
         # mechanics path agrees
         mhits, mtotal = po.execute_topk_mech([blob], term_idx, boosts, k=8)
         assert total == mtotal
