@@ -604,6 +604,30 @@ def test_hybrid_chain_parity(ctx):
                                       [(0, 2, 0, 0)], nb)
 
 
+def test_wand_pruning_fires_and_stays_exact(ctx):
+    """Block-max pruning on the shape it exists for (skewed term
+    frequencies, the BMW paper's motivating case): rare freq spikes set the
+    k-th threshold; every block whose max_freq descriptor bound falls below
+    it is skipped without decoding. Must skip real work (visited < total)
+    and still return the exact top-k."""
+    doc_count = 20_000_000
+    docs = np.arange(1, doc_count + 1, 10, dtype=np.uint32)  # 2M postings
+    freqs = np.ones(len(docs), dtype=np.uint32)
+    freqs[::1009] = 200  # spike ~1 posting per 8 blocks
+    norms = sa.synth_norms(78, doc_count)
+    blob = sa.build_segment(doc_count, [(docs, freqs)], norms)
+    seg = ctx.load_segment(blob)
+    k = 10
+    base, total = ctx.execute_topk([seg], [0], [1.0], k)
+    wand, visited = ctx.execute_topk([seg], [0], [1.0], k, wand=True)
+    np.testing.assert_array_equal(base["doc"], wand["doc"])
+    np.testing.assert_array_equal(
+        base["score"].view(np.uint32), wand["score"].view(np.uint32))
+    # spike blocks are ~1/8 of all blocks; everything else prunes once the
+    # threshold locks onto the spike scores
+    assert visited < total // 2, (visited, total)
+
+
 def test_bm1_parity(ctx):
     """BM1 (k1=0): empty top-k with exact match counting on both sides."""
     blob, _, _ = make_corpus(75, 200_000, [0.05, 0.02])
