@@ -229,6 +229,57 @@ def decode_col_i64(blob, rows):
     return out
 
 
+def encode_col_str(values):
+    """Dictionary-encode a string column: sorted-unique dictionary + i64
+    codes. The SORTED dictionary is what makes string predicates map to
+    contiguous code ranges, so the GPU scan path (sdb_gpu_scan_agg over
+    i64/FoR columns) covers string GROUP BY / WHERE with zero new device
+    code — the dictionary-vector approach of the reference's analytics
+    engine (un-vendored DuckDB fork; result-level parity per SURVEY.md
+    §8c). Returns (codes int64 array, dictionary list)."""
+    import numpy as np
+
+    dictionary = sorted(set(values))
+    index = {s: i for i, s in enumerate(dictionary)}
+    codes = np.fromiter((index[v] for v in values), dtype=np.int64,
+                        count=len(values))
+    return codes, dictionary
+
+
+def str_pred_to_code(dictionary, op, lo=None, hi=None):
+    """Translate a string predicate into an (op, ilo, ihi) triple over the
+    sorted dictionary's codes, directly usable as an SdbPredSpec:
+      op 'eq'      lo               -> BETWEEN [c, c] (empty if absent)
+      op 'between' lo, hi inclusive -> BETWEEN code range
+      op 'prefix'  lo               -> BETWEEN over the prefix span
+      op 'lt' / 'ge' lo             -> LT / GE boundary code
+    An unsatisfiable predicate returns BETWEEN (1, 0), which passes no row."""
+    import bisect
+
+    n = len(dictionary)
+
+    def left(s):
+        return bisect.bisect_left(dictionary, s)
+
+    EMPTY = (3, 1, 0)  # SDB_PRED_BETWEEN with lo > hi: matches nothing
+    if op == "eq":
+        i = left(lo)
+        return (3, i, i) if i < n and dictionary[i] == lo else EMPTY
+    if op == "between":
+        a = left(lo)
+        z = bisect.bisect_right(dictionary, hi) - 1
+        return (3, a, z) if a <= z else EMPTY
+    if op == "prefix":
+        a = left(lo)
+        z = bisect.bisect_left(dictionary, lo + "\U0010ffff") - 1
+        return (3, a, z) if a <= z else EMPTY
+    if op == "lt":
+        return (1, left(lo), 0)  # codes < first code >= lo
+    if op == "ge":
+        return (2, left(lo), 0)
+    raise ValueError(op)
+
+
 def bm25_stats(docs_with_field, docs_with_term, total_term_freq, k=1.2,
                b=0.75):
     idf = C.c_float(0)
