@@ -110,6 +110,16 @@ struct ScanArgs {
   uint32_t st_lds_off[3];   // u32 offset of each stage buffer in smem
 };
 
+// branchless funnel-shift extraction: value = bits [bit, bit+width) of the
+// packed stream = v_alignbit_b32 of the two covering words (width <= 32 so
+// one alignbit covers any in-word offset), masked. The +1 word read is
+// unconditional; table_load over-allocates 4 pad bytes so the final value's
+// slack read stays in bounds (its bits are masked out).
+__device__ __forceinline__ uint32_t bits_at(const uint32_t* w, uint64_t bit) {
+  const uint64_t wi = bit >> 5;
+  return __builtin_amdgcn_alignbit(w[wi + 1], w[wi], (uint32_t)bit & 31u);
+}
+
 // desc fields are read per call from desc[rg]: the address is wave-uniform
 // so the loads scalarize/broadcast from cache (a register-cached desc array
 // indexed by a runtime agg index would spill to scratch — guide rule 20:
@@ -120,16 +130,12 @@ __device__ __forceinline__ int64_t col_read(const ColRef& c, uint32_t rg,
   const SdbColGroupDescDev& d = c.desc[rg];
   if (d.width == 0) return d.base;
   const uint32_t* w = (const uint32_t*)c.data + d.word_off;
-  const uint64_t bit = (r - r0) * d.width;
-  uint64_t v = w[bit >> 5] >> (bit & 31);
-  if ((bit & 31) + d.width > 32)
-    v |= (uint64_t)w[(bit >> 5) + 1] << (32 - (bit & 31));
-  const uint64_t mask =
-    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
-  return d.base + (int64_t)(v & mask);
+  const uint32_t mask =
+    d.width >= 32 ? 0xFFFFFFFFu : ((1u << d.width) - 1u);
+  return d.base + (int64_t)(bits_at(w, (r - r0) * d.width) & mask);
 }
 
-// paired FoR extraction: values r and r+1 from <=3 payload words
+// paired FoR extraction: values r and r+1
 __device__ __forceinline__ void col_read2(const ColRef& c, uint32_t rg,
                                           uint64_t r0, uint64_t r,
                                           int64_t& x0, int64_t& x1) {
@@ -147,22 +153,10 @@ __device__ __forceinline__ void col_read2(const ColRef& c, uint32_t rg,
   }
   const uint32_t* w = (const uint32_t*)c.data + d.word_off;
   const uint64_t bit0 = (r - r0) * d.width;
-  const uint64_t w0i = bit0 >> 5;
-  const uint32_t w0 = w[w0i], w1 = w[w0i + 1], w2 = w[w0i + 2];
-  const uint64_t mask =
-    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
-  const uint32_t sh0 = (uint32_t)(bit0 & 31);
-  uint64_t v0 = (uint64_t)w0 >> sh0;
-  if (sh0 + d.width > 32) v0 |= (uint64_t)w1 << (32 - sh0);
-  const uint64_t bit1 = bit0 + d.width;
-  const uint32_t rel = (uint32_t)((bit1 >> 5) - w0i);  // 0, 1 or 2
-  const uint32_t wa = rel == 0 ? w0 : (rel == 1 ? w1 : w2);
-  const uint32_t wb = rel == 0 ? w1 : w2;
-  const uint32_t sh1 = (uint32_t)(bit1 & 31);
-  uint64_t v1 = (uint64_t)wa >> sh1;
-  if (sh1 + d.width > 32) v1 |= (uint64_t)wb << (32 - sh1);
-  x0 = d.base + (int64_t)(v0 & mask);
-  x1 = d.base + (int64_t)(v1 & mask);
+  const uint32_t mask =
+    d.width >= 32 ? 0xFFFFFFFFu : ((1u << d.width) - 1u);
+  x0 = d.base + (int64_t)(bits_at(w, bit0) & mask);
+  x1 = d.base + (int64_t)(bits_at(w, bit0 + d.width) & mask);
 }
 
 // staged analogue of col_read2: the chunk's packed words sit in LDS (sw),
@@ -180,22 +174,10 @@ __device__ __forceinline__ void col_read2_lds(const SdbColGroupDescDev* desc,
     return;
   }
   const uint64_t bit0 = (uint64_t)lr * d.width;
-  const uint64_t w0i = bit0 >> 5;
-  const uint32_t w0 = sw[w0i], w1 = sw[w0i + 1], w2 = sw[w0i + 2];
-  const uint64_t mask =
-    d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);
-  const uint32_t sh0 = (uint32_t)(bit0 & 31);
-  uint64_t v0 = (uint64_t)w0 >> sh0;
-  if (sh0 + d.width > 32) v0 |= (uint64_t)w1 << (32 - sh0);
-  const uint64_t bit1 = bit0 + d.width;
-  const uint32_t rel = (uint32_t)((bit1 >> 5) - w0i);  // 0, 1 or 2
-  const uint32_t wa = rel == 0 ? w0 : (rel == 1 ? w1 : w2);
-  const uint32_t wb = rel == 0 ? w1 : w2;
-  const uint32_t sh1 = (uint32_t)(bit1 & 31);
-  uint64_t v1 = (uint64_t)wa >> sh1;
-  if (sh1 + d.width > 32) v1 |= (uint64_t)wb << (32 - sh1);
-  x0 = d.base + (int64_t)(v0 & mask);
-  x1 = d.base + (int64_t)(v1 & mask);
+  const uint32_t mask =
+    d.width >= 32 ? 0xFFFFFFFFu : ((1u << d.width) - 1u);
+  x0 = d.base + (int64_t)(bits_at(sw, bit0) & mask);
+  x1 = d.base + (int64_t)(bits_at(sw, bit0 + d.width) & mask);
 }
 
 template <int RAW>
@@ -715,7 +697,8 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
         return SDB_ERR_INVALID;
       }
       tab->group_rows = hdr.group_rows;
-      HIP_CHECK(hipMalloc(&tab->cols[c], hdr.size));
+      // +4 pad bytes: bits_at reads one slack word past the final value
+      HIP_CHECK(hipMalloc(&tab->cols[c], hdr.size + 4));
       HIP_CHECK(hipMemcpy(tab->cols[c], cols[c].data, hdr.size,
                           hipMemcpyHostToDevice));
       tab->refs[c].data = (const uint8_t*)tab->cols[c] + hdr.off_payload;
