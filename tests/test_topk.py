@@ -161,6 +161,46 @@ def test_topk_exact_vs_brute(seed, mm):
         assert h["score"] == scores[int(h["doc"])], int(h["doc"])
 
 
+def test_golden_multisegment_order():
+    """Transcribed multi-segment disjunction fixture: the corpus split into
+    even-seq / odd-seq segments, OR of terms {6, 8}, no freq/norm features.
+    Expected {3, 7, 0, 2, 5} for BOTH scorers — bm25_test.cpp:734 and
+    tfidf_test.cpp:759 — which pins the CROSS-SEGMENT stats merge (global
+    df ranks term-8 docs above term-6 docs) and the tie order (the
+    reference's multimap preserves segment-then-doc insertion order; this
+    repo's total order reproduces it)."""
+    g = json.load(open(os.path.join(GOLDEN,
+                                    "bm25_simple_sequential_order.json")))
+    docs_fields = g["docs"]
+    vocab = {}
+    for f in docs_fields:
+        for t in f:
+            vocab.setdefault(t, len(vocab))
+
+    def build(seqs):
+        postings = [[] for _ in vocab]
+        for local, seq in enumerate(seqs, start=1):
+            for t in sorted(set(docs_fields[seq]), key=lambda x: vocab[x]):
+                postings[vocab[t]].append((local, 1))
+        plist = [(np.array([d for d, _ in pl], dtype=np.uint32),
+                  np.array([c for _, c in pl], dtype=np.uint32))
+                 for pl in postings]
+        norms = np.ones(len(seqs) + 1, dtype=np.uint32)
+        norms[0] = 0
+        return sa.build_segment(len(seqs), plist, norms)
+
+    seq0, seq1 = [0, 2, 4, 6], [1, 3, 5, 7]
+    b0, b1 = build(seq0), build(seq1)
+    ti = [vocab["6"], vocab["8"]]
+    for scorer in ("bm25", "tfidf"):
+        hits, total = po.execute_topk([b0, b1], ti, [1.0, 1.0], 8,
+                                      scorer=scorer)
+        seqs = [(seq0 if h["segment"] == 0 else seq1)[int(h["doc"]) - 1]
+                for h in hits]
+        assert seqs == [3, 7, 0, 2, 5], (scorer, seqs)
+        assert total == 5
+
+
 def test_hybrid_chain_vs_brute():
     """Predicate-chain hybrid (ColFilterChain AND semantics): oracle chain
     vs independent numpy evaluation, and a vacuous extra predicate must
