@@ -38,6 +38,10 @@ def _copy_blob(ptr, size):
 
 
 def _load(name):
+    # SDB_GPU_LIB overrides the GPU library path (same-box A/B of two
+    # kernel builds; perf work only, never set in tests/bench defaults)
+    if name == "libsdb_gpu.so" and os.environ.get("SDB_GPU_LIB"):
+        return C.CDLL(os.environ["SDB_GPU_LIB"])
     path = os.path.join(_PKG_DIR, name)
     if not os.path.exists(path):
         raise ImportError(
