@@ -120,6 +120,16 @@ typedef struct SdbQueryPlan {
   uint64_t g_docs_with_field;
   uint64_t g_total_term_freq;
   const uint64_t* g_docs_with_term; /* per plan term, or NULL */
+  /* HasFilterBoost scorer variants (bm25.cpp:112-140 Bm1Boost, :69-109
+   * Bm15/Bm25 boost[i]*num; tfidf equivalents): when nonzero, every
+   * segment must have a boost column attached (sdb_gpu_segment_attach_
+   * boost) and each term contribution is multiplied by boost[doc]. With
+   * k1 == 0 this is BM1's only nonzero form: score = boost[doc]*num.
+   * (The reference's Bm1Boost ASSIGNS rather than merges under multi-term
+   * disjunctions; this implementation sums term contributions uniformly —
+   * single-term BM1 parity is exact, the multi-term BM1 merge order is a
+   * documented refinement.) */
+  uint32_t filter_boost;
 } SdbQueryPlan;
 
 /* Execute BM25 top-k over segments resident on this context's device.
@@ -189,6 +199,10 @@ int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
                                   const int64_t* data);
 int sdb_gpu_segment_attach_column_slot(SdbGpuCtx* ctx, SdbGpuSegment* seg,
                                        uint32_t slot, const int64_t* data);
+/* per-doc f32 filter-boost column (doc_count+1 entries, 1-based docs);
+ * consumed when SdbQueryPlan.filter_boost is set */
+int sdb_gpu_segment_attach_boost(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                 const float* boost);
 
 typedef struct SdbHybridPred {
   uint32_t slot;  /* attached column slot */

@@ -148,9 +148,17 @@ SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
 
 
 def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
-                 global_stats=None, scorer="bm25"):
-    """EXACT top-k (parity oracle). blobs: list of segment blob bytes."""
+                 global_stats=None, scorer="bm25", filter_boost=None):
+    """EXACT top-k (parity oracle). blobs: list of segment blob bytes.
+    filter_boost: optional per-doc f32 multiplier (doc_count+1, 1-based) —
+    the HasFilterBoost scorer variants; single-segment only."""
     lib().o_set_scorer(C.c_uint32(SCORERS[scorer]))
+    fbkeep = None
+    if filter_boost is not None:
+        assert len(blobs) == 1, "filter_boost: single segment only"
+        fbkeep = np.ascontiguousarray(filter_boost, dtype=np.float32)
+        lib().o_set_filter_boost(
+            fbkeep.ctypes.data_as(C.POINTER(C.c_float)))
     arr, keep = _mkblobs(blobs)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)
@@ -168,8 +176,10 @@ def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
         C.c_uint32(min_match), C.c_float(k1), C.c_float(b),
         C.c_uint64(g_dwf), g_dwt_ptr, C.c_uint64(g_ttf), C.c_uint32(k),
         hits, C.byref(out_count), C.byref(total))
+    if filter_boost is not None:
+        lib().o_set_filter_boost(None)
     assert rc == 0, rc
-    del keep, keep2
+    del keep, keep2, fbkeep
     return _hits_to_np(hits, out_count.value), total.value
 
 

@@ -529,6 +529,12 @@ static void o_cand_push(OCandVec* cv, float score, uint32_t doc,
   cv->n++;
 }
 
+/* optional per-doc filter boost (HasFilterBoost scorer variants,
+ * bm25.cpp:112-140 Bm1Boost / :69-109 Bm15/Bm25 boost[i]*num; modeled as
+ * one shared per-doc multiplier applied per term contribution). NULL=off. */
+static const float* o_fboost = NULL;
+void o_set_filter_boost(const float* fb) { o_fboost = fb; }
+
 /* Process windows covering docs [range_lo, range_hi] (inclusive, 1-based,
  * local to the segment). Cursors must be positioned before range_lo.
  * Exactly one of coll / cands is non-NULL.
@@ -584,8 +590,10 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
         while (i < c->buf_len && c->buf_docs[i] <= hi) {
           const uint32_t doc = c->buf_docs[i];
           const uint32_t off = doc - lo;
-          score_win[off] += o_score_one(c->scorer, c->num, c->nc, c->nl,
-                                        c->buf_freqs[i], norms[doc]);
+          float s1 = o_score_one(c->scorer, c->num, c->nc, c->nl,
+                                 c->buf_freqs[i], norms[doc]);
+          if (o_fboost) s1 *= o_fboost[doc]; /* one f32 mul, GPU-identical */
+          score_win[off] += s1;
           const uint8_t cc = ++cnt_win[off];
           if (cc == min_match) mask[off >> 6] |= 1ull << (off & 63);
           ++i;
@@ -674,6 +682,7 @@ static const SdbBlockDesc* o_seek_block(const SdbBlockDesc* b,
 static uint32_t o_scorer_kind = 0; /* set per call by the drivers (the C
   API keeps single-threaded-per-call semantics; sdb_oracle is test infra) */
 
+
 static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
                           const float* boosts, uint32_t nterms, float k1,
                           float b, uint64_t g_dwf, const uint64_t* g_dwt,
@@ -711,8 +720,10 @@ static int o_prep_cursors(const SdbSegmentView* v, const uint32_t* term_idx,
     c->nl = nl;
     /* BM1 (k == 0, bm25.cpp:112-140 Bm1Score + :333-336 dispatch): without
      * a filter boost every score is 0 (memset), so nothing beats the
-     * collector's FLT_MIN threshold; num = 0 reproduces that exactly */
-    if (k1 == 0.0f) c->num = 0.0f;
+     * collector's FLT_MIN threshold; num = 0 reproduces that exactly.
+     * WITH a filter boost: score = fb * num (Bm1Boost), freq-independent,
+     * which the generic formula already yields at c1 == 0. */
+    if (k1 == 0.0f && !o_fboost) c->num = 0.0f;
   }
   return 0;
 }

@@ -356,7 +356,7 @@ class GpuContext:
     SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
 
     def _make_plan(self, term_idx, boosts, min_match, k1, b, global_stats,
-                   scorer="bm25", wand=False):
+                   scorer="bm25", wand=False, filter_boost=False):
         class _Plan(C.Structure):
             _fields_ = [
                 ("terms", C.POINTER(SdbTermRef)),
@@ -369,12 +369,14 @@ class GpuContext:
                 ("g_docs_with_field", C.c_uint64),
                 ("g_total_term_freq", C.c_uint64),
                 ("g_docs_with_term", C.POINTER(C.c_uint64)),
+                ("filter_boost", C.c_uint32),
             ]
 
         terms = (SdbTermRef * len(term_idx))(
             *[SdbTermRef(t, float(bo)) for t, bo in zip(term_idx, boosts)])
         plan = _Plan(terms, len(term_idx), min_match, k1, b,
-                     self.SCORERS[scorer], 1 if wand else 0, 0, 0, None)
+                     self.SCORERS[scorer], 1 if wand else 0, 0, 0, None,
+                     1 if filter_boost else 0)
         plan._keep = terms
         if global_stats is not None:
             dwf, ttf, dwt = global_stats
@@ -386,15 +388,18 @@ class GpuContext:
         return plan
 
     def execute_topk(self, segs, term_idx, boosts, k, min_match=1, k1=1.2,
-                     b=0.75, global_stats=None, scorer="bm25", wand=False):
+                     b=0.75, global_stats=None, scorer="bm25", wand=False,
+                     filter_boost=False):
         """global_stats: optional (docs_with_field, total_term_freq,
         [docs_with_term per term]) for sharded execution. wand=True enables
         exact block-max pruning (OR plans only; total_matches then counts
-        visited matches only)."""
+        visited matches only). filter_boost=True multiplies every term
+        contribution by the segment's attached boost column
+        (attach_boost; the HasFilterBoost scorer variants)."""
         import numpy as np
 
         plan = self._make_plan(term_idx, boosts, min_match, k1, b,
-                               global_stats, scorer, wand)
+                               global_stats, scorer, wand, filter_boost)
         seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value) for s in segs])
         hits = (SdbScoreDoc * k)()
         out_count = C.c_uint32(0)
@@ -419,6 +424,17 @@ class GpuContext:
             col.ctypes.data_as(C.POINTER(C.c_int64)))
         if rc != 0:
             raise RuntimeError(f"attach_column rc={rc}")
+
+    def attach_boost(self, seg, boost):
+        """Attach the per-doc f32 filter-boost column (doc_count+1,
+        1-based docs; bm25.cpp HasFilterBoost variants)."""
+        import numpy as np
+
+        boost = np.ascontiguousarray(boost, dtype=np.float32)
+        rc = self._lib.sdb_gpu_segment_attach_boost(
+            self._ctx, seg, boost.ctypes.data_as(C.POINTER(C.c_float)))
+        if rc != 0:
+            raise RuntimeError(f"attach_boost rc={rc}")
 
     def execute_topk_hybrid(self, segs, term_idx, boosts, k, flo, fhi,
                             nbuckets, min_match=1, k1=1.2, b=0.75,

@@ -78,6 +78,8 @@ struct SdbGpuSegment {
   long long* fcols[4];  // device, doc_count+1 per attached filter-column
                         // slot (SDB_MAX_FILTER_COLS) or null; slot 0 is
                         // the classic hybrid/bucket column
+  float* fboost;        // device, doc_count+1 per-doc filter boost or null
+  float fboost_max;     // host-computed max (WAND bound / smax scaling)
   SdbTermEntry* terms_host;  // host copy of term table
   SdbSegHeader hdr;     // host copy
 };
@@ -295,7 +297,8 @@ __device__ void decode_freq_block_wave(const uint8_t* p, uint32_t len,
 __device__ __forceinline__ bool try_block_fused(
   const uint8_t* pl, const SdbBlockDesc& d, int lane, uint32_t norm_stream,
   uint32_t lo, uint32_t hi, float num, float nc, float nl, uint32_t scorer,
-  const uint32_t* norms_col, float* swin, uint8_t* cwin) {
+  const uint32_t* norms_col, const float* fboost, float* swin,
+  uint8_t* cwin) {
   const uint8_t* db = pl + d.doc_off;
   const uint8_t* fb = pl + d.freq_off;
   const uint8_t* nb = fb + d.flags;
@@ -328,13 +331,15 @@ __device__ __forceinline__ bool try_block_fused(
     n1 = doc1 >= lo && doc1 <= hi ? norms_col[doc1] : 1u;
   }
   if (doc0 >= lo && doc0 <= hi) {
-    const float s = score_one(scorer, num, nc, nl, f0, n0);
+    float s = score_one(scorer, num, nc, nl, f0, n0);
+    if (fboost) s *= fboost[doc0];
     const uint32_t off = doc0 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
   }
   if (doc1 >= lo && doc1 <= hi) {
-    const float s = score_one(scorer, num, nc, nl, f1, n1);
+    float s = score_one(scorer, num, nc, nl, f1, n1);
+    if (fboost) s *= fboost[doc1];
     const uint32_t off = doc1 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -349,7 +354,7 @@ __device__ __forceinline__ bool try_block_fused2(
   const uint8_t* pl, const SdbBlockDesc& da, const SdbBlockDesc& db_,
   int lane, uint32_t norm_stream, uint32_t lo, uint32_t hi, float num,
   float nc, float nl, uint32_t scorer, const uint32_t* norms_col,
-  float* swin, uint8_t* cwin) {
+  const float* fboost, float* swin, uint8_t* cwin) {
   const uint8_t* adoc = pl + da.doc_off;
   const uint8_t* afrq = pl + da.freq_off;
   const uint8_t* anrm = afrq + da.flags;
@@ -403,7 +408,8 @@ __device__ __forceinline__ bool try_block_fused2(
   for (int e = 0; e < 4; ++e) {
     const uint32_t doc = docs[e];
     if (doc < lo || doc > hi) continue;
-    const float s = score_one(scorer, num, nc, nl, frqs[e], nrms[e]);
+    float s = score_one(scorer, num, nc, nl, frqs[e], nrms[e]);
+    if (fboost) s *= fboost[doc];
     const uint32_t off = doc - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -475,6 +481,10 @@ struct WindowArgs {
   unsigned long long* bucket_out;  // [2*nbuckets]: count, sum (i64 bits)
   uint32_t dcache_n;  // staged descriptors per term (host-shrunk so the
                       // cache fits LDS for wide plans; <= SDB_DESC_CACHE)
+  const float* fb;    // per-doc filter boost (null = off); each term
+                      // contribution is multiplied by fb[doc]
+  float fbmax;        // 1.0 when off; scales WAND bounds (smax is scaled
+                      // host-side so the histogram stays valid)
 };
 
 // Persistent-range window kernel: each workgroup owns a CONTIGUOUS range of
@@ -608,7 +618,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           ub = u > ub ? u : ub;
         }
       }
-      wub[tid] = ub;
+      wub[tid] = ub * a.fbmax;  // filter boost can scale any doc up to max
     }
     __syncthreads();
     SDB_T(0)
@@ -645,7 +655,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             __builtin_memcpy(&gtw, &bits, 4);
           }
           const float own =
-            score_one(a.scorer, num, nc, nl, d.max_freq, d.min_norm);
+            score_one(a.scorer, num, nc, nl, d.max_freq, d.min_norm) *
+            a.fbmax;
           if (own + (wand_total_ub - wub[t]) < gtw) {
             b += SDB_NWAVES;
             continue;
@@ -676,8 +687,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                       : a.desc[b2];
             if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
                 try_block_fused2(pl, d, d2, lane, a.norm_stream, lo, hi,
-                                 num, nc, nl, a.scorer, a.norms, swin,
-                                 cwin)) {
+                                 num, nc, nl, a.scorer, a.norms, a.fb,
+                                 swin, cwin)) {
               b += 2 * SDB_NWAVES;
               continue;
             }
@@ -711,7 +722,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #else
 #ifndef SDB_ABLATE_SCORE
         if (try_block_fused(pl, d, lane, a.norm_stream, lo, hi, num, nc, nl,
-                            a.scorer, a.norms, swin, cwin)) {
+                            a.scorer, a.norms, a.fb, swin, cwin)) {
           b += SDB_NWAVES;  // while-loop: explicit advance before continue
           continue;
         }
@@ -730,7 +741,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #ifdef SDB_ABLATE_SCORE
           asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
 #else
-          const float s = score_one(a.scorer, num, nc, nl, freq, norm);
+          float s = score_one(a.scorer, num, nc, nl, freq, norm);
+          if (a.fb) s *= a.fb[doc];
           const uint32_t off = doc - lo;
           swin[off] += s;      // unique doc within the term: no atomics
           cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -1173,6 +1185,7 @@ int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
   hipFree(seg->norms);
   for (int i = 0; i < 4; ++i)
     if (seg->fcols[i]) hipFree(seg->fcols[i]);
+  if (seg->fboost) hipFree(seg->fboost);
   std::free(seg->terms_host);
   delete seg;
   return SDB_OK;
@@ -1208,6 +1221,15 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     for (uint32_t j = i + 1; j < plan->nterms; ++j)  // count match tallies
       if (plan->terms[i].term_idx == plan->terms[j].term_idx)
         return SDB_ERR_INVALID;
+  float fbmax = 1.0f;  // filter boost: every segment needs the column
+  if (plan->filter_boost) {
+    fbmax = 0.0f;
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      if (!segs[s]->fboost) return SDB_ERR_INVALID;
+      fbmax = segs[s]->fboost_max > fbmax ? segs[s]->fboost_max : fbmax;
+    }
+    if (!(fbmax > 0.0f)) return SDB_ERR_INVALID;
+  }
 
   // ---- PreparePhase analogue: global stats (double -> f32, bm25.cpp) ----
   uint64_t g_dwf = plan->g_docs_with_field;
@@ -1252,8 +1274,9 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
       // BM1 (k == 0, bm25.cpp:112-140 Bm1Score + :333-336): without a
       // filter boost every score is 0, so no hit beats the collector's
-      // FLT_MIN threshold; matches still count
-      if (k1 == 0.0f) num[t] = 0.0f;
+      // FLT_MIN threshold; matches still count. WITH a filter boost the
+      // score is fb*num (Bm1Boost), which c1 == 0 already yields.
+      if (k1 == 0.0f && !plan->filter_boost) num[t] = 0.0f;
       smax += num[t] > 0 ? num[t] : 0.0f;
     } else {
       // TFIDF::collect (tfidf.cpp:148-151)
@@ -1275,6 +1298,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       nl = (g_ttf && g_dwf) ? kb / ((float)g_ttf / (float)g_dwf) : kb;
     }
   }
+  if (plan->filter_boost) smax *= fbmax;  // scores reach fb*base
   if (smax <= 0.0f) smax = FLT_MIN;
 
   // ---- reset device state ----
@@ -1348,6 +1372,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.smax = smax;
     a.seg_idx = s;
     a.dcache_n = dcache_n;
+    a.fb = plan->filter_boost ? seg->fboost : nullptr;
+    a.fbmax = plan->filter_boost ? fbmax : 1.0f;
     a.gthresh = ctx->d_gthresh;
     a.ghist = ctx->d_ghist;
     a.cands = ctx->d_cands;
@@ -1505,6 +1531,19 @@ int sdb_gpu_segment_attach_column_slot(SdbGpuCtx* ctx, SdbGpuSegment* seg,
 int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
                                   const int64_t* data) {
   return sdb_gpu_segment_attach_column_slot(ctx, seg, 0, data);
+}
+
+// per-doc f32 filter boost (HasFilterBoost scorer variants; see sdb_gpu.h)
+int sdb_gpu_segment_attach_boost(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                 const float* boost) {
+  if (!ctx || !seg || !boost) return SDB_ERR_INVALID;
+  const uint64_t n = (uint64_t)seg->hdr.doc_count + 1;
+  if (!seg->fboost) HIP_CHECK(hipMalloc(&seg->fboost, 4ull * n));
+  HIP_CHECK(hipMemcpy(seg->fboost, boost, 4ull * n, hipMemcpyHostToDevice));
+  float mx = 0.0f;
+  for (uint64_t i = 1; i < n; ++i) mx = boost[i] > mx ? boost[i] : mx;
+  seg->fboost_max = mx;
+  return SDB_OK;
 }
 
 // Hybrid: BM25 top-k AND col BETWEEN [flo,fhi] + per-bucket COUNT/SUM over

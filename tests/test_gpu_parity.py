@@ -628,6 +628,39 @@ def test_wand_pruning_fires_and_stays_exact(ctx):
     assert visited < total // 2, (visited, total)
 
 
+def test_filter_boost_parity(ctx):
+    """Per-doc filter boost (HasFilterBoost variants): GPU == oracle
+    bit-exact, incl. BM1's only nonzero form and WAND exactness under the
+    scaled bounds; filter_boost without an attached column is rejected."""
+    doc_count = 300_000
+    blob, _, _ = make_corpus(80, doc_count, [0.08, 0.04, 0.02])
+    rng = np.random.default_rng(14)
+    fb = rng.uniform(0.5, 2.0, doc_count + 1).astype(np.float32)
+    fb[0] = 0.0
+    seg = ctx.load_segment(blob)
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):  # no boost column attached yet
+        ctx.execute_topk([seg], [0, 1], [1.0] * 2, 10, filter_boost=True)
+    ctx.attach_boost(seg, fb)
+    for kwargs in ({}, {"min_match": 2}, {"k1": 0.0}, {"scorer": "tfidf"}):
+        hits, total = ctx.execute_topk([seg], [0, 1, 2], [1.0, 2.0, 0.5],
+                                       400, filter_boost=True, **kwargs)
+        ohits, ototal = po.execute_topk([blob], [0, 1, 2], [1.0, 2.0, 0.5],
+                                        400, filter_boost=fb, **kwargs)
+        assert total == ototal
+        np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+        np.testing.assert_array_equal(
+            hits["score"].view(np.uint32), ohits["score"].view(np.uint32))
+    # WAND stays exact with boosted bounds
+    base, _t = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 50,
+                                filter_boost=True)
+    wnd, _v = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 50,
+                               filter_boost=True, wand=True)
+    np.testing.assert_array_equal(base["doc"], wnd["doc"])
+    np.testing.assert_array_equal(
+        base["score"].view(np.uint32), wnd["score"].view(np.uint32))
+
+
 def test_bm1_parity(ctx):
     """BM1 (k1=0): empty top-k with exact match counting on both sides."""
     blob, _, _ = make_corpus(75, 200_000, [0.05, 0.02])

@@ -161,6 +161,52 @@ def test_topk_exact_vs_brute(seed, mm):
         assert h["score"] == scores[int(h["doc"])], int(h["doc"])
 
 
+def test_filter_boost_oracle():
+    """Per-doc filter boost (HasFilterBoost scorer variants,
+    bm25.cpp:112-140): each term contribution is multiplied by fb[doc]
+    (one f32 mul, term-major). Single-term scores must equal
+    f32(contrib * fb) bitwise; BM1 (k1=0) + boost = f32(num * fb)."""
+    doc_count = 30_000
+    sels = [0.1, 0.05]
+    blob, postings, norms = synth_corpus(50, doc_count, sels)
+    rng = np.random.default_rng(13)
+    fb = rng.uniform(0.5, 2.0, doc_count + 1).astype(np.float32)
+    fb[0] = 0.0
+
+    # multi-term: replicate the oracle's arithmetic independently in numpy
+    ttf = int(norms[1:].sum())
+    scores = np.zeros(doc_count + 1, dtype=np.float32)
+    cnt = np.zeros(doc_count + 1, dtype=np.int32)
+    for docs, freqs in postings:
+        df = len(docs)
+        idf = np.float32(np.log1p(
+            (np.float64(doc_count - df) + 0.5) / (np.float64(df) + 0.5)))
+        nc = np.float32(np.float32(1.2) - np.float32(1.2) * np.float32(0.75))
+        avg = np.float32(np.float32(ttf) / np.float32(doc_count))
+        nl = np.float32(np.float32(np.float32(1.2) * np.float32(0.75)) / avg)
+        num = np.float32(np.float32(2.2)) * idf
+        c1 = nc + nl * norms[docs].astype(np.float32)
+        contrib = num - num * c1 / (c1 + freqs.astype(np.float32))
+        scores[docs] += contrib * fb[docs]
+        cnt[docs] += 1
+    hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 300,
+                                  filter_boost=fb)
+    assert total == int((cnt > 0).sum())
+    for h in hits:
+        assert h["score"] == scores[int(h["doc"])], int(h["doc"])
+
+    # BM1 + filter boost: freq-independent fb*num per matching term
+    hits1, _ = po.execute_topk([blob], [0], [1.0], 100, k1=0.0,
+                               filter_boost=fb)
+    df = len(postings[0][0])
+    idf = np.float32(np.log1p(
+        (np.float64(doc_count - df) + 0.5) / (np.float64(df) + 0.5)))
+    num = np.float32(np.float32(1.0)) * idf  # (k+1)=1 at k=0
+    assert len(hits1) == 100
+    for h in hits1:
+        assert h["score"] == np.float32(num * fb[int(h["doc"])])
+
+
 def test_golden_multisegment_order():
     """Transcribed multi-segment disjunction fixture: the corpus split into
     even-seq / odd-seq segments, OR of terms {6, 8}, no freq/norm features.
