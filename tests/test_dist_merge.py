@@ -65,7 +65,8 @@ def _worker(rank, world, port, q):
     dist.destroy_process_group()
 
 
-def test_gloo_sharded_merge_equals_full():
+@pytest.mark.parametrize("world,port", [(2, 29511), (4, 29517)])
+def test_gloo_sharded_merge_equals_full(world, port):
     import serenedb_amd as sa
     from oracle import pyoracle as po
 
@@ -75,9 +76,8 @@ def test_gloo_sharded_merge_equals_full():
 
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29511
-    procs = [ctx.Process(target=_worker, args=(r, 2, port, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, world, port, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     try:

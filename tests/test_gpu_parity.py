@@ -413,6 +413,37 @@ def test_scan_agg_for_staged_edges(ctx):
     np.testing.assert_allclose(usf, gsf, rtol=1e-7, atol=1e-6)
     lib.sdb_gpu_table_free(ctx._ctx, tab)
 
+    # three predicates over FoR columns: preds 0/1 ride the stage slots,
+    # pred 2 exercises the staged kernel's global col_read slow path
+    cols3 = (ColView * 4)(
+        ColView(keys.ctypes.data_as(CT.c_void_p).value, rows, 0),
+        ColView(vb1.ctypes.data_as(CT.c_void_p).value, rows, 2),
+        ColView(v2.ctypes.data_as(CT.c_void_p).value, rows, 1),
+        ColView(vb3.ctypes.data_as(CT.c_void_p).value, rows, 2))
+    tab3 = CT.c_void_p(0)
+    rc = lib.sdb_gpu_table_load(ctx._ctx, cols3, 4, CT.c_uint64(rows),
+                                CT.byref(tab3))
+    assert rc == 0, rc
+    preds3 = (PredSpec * 3)(
+        PredSpec(1, 2, lo, 0, 0, 0),            # v1 >= lo (staged)
+        PredSpec(3, 2, -(1 << 29), 0, 0, 0),    # v3 >= -2^29 (staged)
+        PredSpec(3, 1, 1 << 29, 0, 0, 0))       # v3 < 2^29 (global path)
+    aggs3 = (AggSpec * 2)(AggSpec(0, 0), AggSpec(3, 1))
+    out3 = (AggResult * (ngroups * 2))()
+    rc = lib.sdb_gpu_scan_agg(ctx._ctx, tab3, 0, ngroups, preds3, 3, aggs3,
+                              2, out3, CT.byref(passed))
+    assert rc == 0, rc
+    m3 = (v1 >= lo) & (v3 >= -(1 << 29)) & (v3 < (1 << 29))
+    assert passed.value == int(m3.sum())
+    np.testing.assert_array_equal(
+        np.array([out3[g * 2 + 0].i64 for g in range(ngroups)]),
+        np.bincount(keys[m3], minlength=ngroups))
+    np.testing.assert_array_equal(
+        np.array([out3[g * 2 + 1].i64 for g in range(ngroups)]),
+        np.bincount(keys[m3], weights=v3[m3].astype(np.float64),
+                    minlength=ngroups).astype(np.int64))
+    lib.sdb_gpu_table_free(ctx._ctx, tab3)
+
     # FoR key with the predicate on the key column itself: shared stage
     # slot (add_stage dedup) + agg_src=9 (key-as-value SUM)
     keys_blob = sa.encode_col_i64(keys)
