@@ -195,28 +195,23 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   __syncthreads();
 
   if (RAW) {
-    // raw dense columns: 2 rows/thread with 16-byte loads (the measured
-    // fastest shape: 4.55 TB/s at 1024t/2048b)
-    const uint64_t pair_stride = (uint64_t)gridDim.x * SCAN_NTHREADS * 2u;
+    // raw dense columns: 4 rows/thread with 2x16-byte loads per column
+    // (measured +9% over 2 rows/thread: 1.79 -> 1.64 ms per 400M rows)
+    const uint64_t quad_stride = (uint64_t)gridDim.x * SCAN_NTHREADS * 4u;
     unsigned long long my_passed = 0;
-    const uint64_t rows2 = a.rows & ~1ull;
+    const uint64_t rows4 = a.rows & ~3ull;
     for (uint64_t r =
-           ((uint64_t)blockIdx.x * SCAN_NTHREADS + threadIdx.x) * 2u;
-         r < rows2; r += pair_stride) {
-      bool okv[2] = {true, true};
+           ((uint64_t)blockIdx.x * SCAN_NTHREADS + threadIdx.x) * 4u;
+         r < rows4; r += quad_stride) {
+      bool okv[4] = {true, true, true, true};
       for (uint32_t p = 0; p < a.npreds; ++p) {
-#ifdef SDB_SCAN_NT
-        // streamed-once columns: nontemporal loads keep L2 for reuse data
-        const long long* pp = &((const long long*)a.pred_col[p].data)[r];
-        const int64_t xs[2] = {__builtin_nontemporal_load(pp),
-                               __builtin_nontemporal_load(pp + 1)};
-#else
-        longlong2 x;
+        longlong2 x, y;
         __builtin_memcpy(&x, &((const long long*)a.pred_col[p].data)[r], 16);
-        const int64_t xs[2] = {x.x, x.y};
-#endif
+        __builtin_memcpy(&y, &((const long long*)a.pred_col[p].data)[r + 2],
+                         16);
+        const int64_t xs[4] = {x.x, x.y, y.x, y.y};
 #pragma unroll
-        for (int e = 0; e < 2; ++e) {
+        for (int e = 0; e < 4; ++e) {
           switch (a.pred_op[p]) {
             case SDB_PRED_LT: okv[e] &= xs[e] < a.pred_lo[p]; break;
             case SDB_PRED_GE: okv[e] &= xs[e] >= a.pred_lo[p]; break;
@@ -227,12 +222,13 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
           }
         }
       }
-      if (!okv[0] && !okv[1]) continue;
-      longlong2 kk;
-      __builtin_memcpy(&kk, &((const long long*)a.keys.data)[r], 16);
-      const int64_t ks[2] = {kk.x, kk.y};
+      if (!okv[0] && !okv[1] && !okv[2] && !okv[3]) continue;
+      longlong2 k0, k1;
+      __builtin_memcpy(&k0, &((const long long*)a.keys.data)[r], 16);
+      __builtin_memcpy(&k1, &((const long long*)a.keys.data)[r + 2], 16);
+      const int64_t ks[4] = {k0.x, k0.y, k1.x, k1.y};
 #pragma unroll
-      for (int e = 0; e < 2; ++e) {
+      for (int e = 0; e < 4; ++e) {
         if (!okv[e]) continue;
         ++my_passed;
         const uint32_t g = (uint32_t)ks[e];
@@ -255,8 +251,8 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         }
       }
     }
-    if ((a.rows & 1ull) && blockIdx.x == 0 && threadIdx.x == 0) {
-      const uint64_t r = a.rows - 1;
+    if ((a.rows & 3ull) && blockIdx.x == 0 && threadIdx.x == 0)
+      for (uint64_t r = rows4; r < a.rows; ++r) {
       bool ok = true;
       for (uint32_t p = 0; p < a.npreds; ++p) {
         const int64_t x = ((const long long*)a.pred_col[p].data)[r];
