@@ -979,6 +979,10 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       return;
     }
     uint32_t pos = base + my_excl;
+    // whole-window skip: once the global threshold locks, most windows
+    // contribute zero candidates — their append walk (a full cwin re-scan)
+    // is pure waste. block_total is WG-uniform, so control stays uniform.
+    if (block_total)
     for (uint32_t wb = 4 * tid; wb < wlen; wb += 4 * SDB_NTHREADS) {
       const uint32_t cw = ((const uint32_t*)cwin)[wb >> 2];
       if (cw == 0) continue;
