@@ -701,7 +701,14 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
       // encoded blob: upload whole blob, point refs into it
       SdbColHeaderDev hdr;
       std::memcpy(&hdr, cols[c].data, sizeof(hdr));
-      if (hdr.magic != SDB_COL_MAGIC_DEV || hdr.rows != rows) {
+      // internal consistency (the view carries no external size, so the
+      // declared extents must at least agree with themselves and `rows`)
+      if (hdr.magic != SDB_COL_MAGIC_DEV || hdr.rows != rows ||
+          hdr.group_rows == 0 ||
+          hdr.ngroups != (rows + hdr.group_rows - 1) / hdr.group_rows ||
+          hdr.off_desc > hdr.size || hdr.off_payload > hdr.size ||
+          (uint64_t)hdr.ngroups * sizeof(SdbColGroupDescDev) >
+            hdr.size - hdr.off_desc) {
         table_free_partial(tab);
         return SDB_ERR_INVALID;
       }
@@ -723,10 +730,22 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
         (const SdbColGroupDescDev*)((const uint8_t*)cols[c].data +
                                     hdr.off_desc);
       uint16_t mw = 0;
-      for (uint32_t g = 0; g < hdr.ngroups; ++g)
-        if (hdesc[g].width > mw) mw = hdesc[g].width;
+      const uint64_t payw = (hdr.size - hdr.off_payload) / 4;
+      for (uint32_t g = 0; g < hdr.ngroups; ++g) {
+        const SdbColGroupDescDev& gd = hdesc[g];
+        const uint64_t glen =
+          std::min<uint64_t>(rows, (uint64_t)(g + 1) * hdr.group_rows) -
+          (uint64_t)g * hdr.group_rows;
+        if (gd.width > 32 ||
+            (gd.width && (gd.word_off > payw ||
+                          (glen * gd.width + 31) / 32 > payw - gd.word_off))) {
+          table_free_partial(tab);
+          return SDB_ERR_INVALID;
+        }
+        if (gd.width > mw) mw = gd.width;
+      }
       tab->max_w[c] = mw;
-      tab->paywords[c] = (hdr.size - hdr.off_payload) / 4;
+      tab->paywords[c] = payw;
     } else {
       const size_t esz = cols[c].type == SDB_COL_I64 ? 8 : 4;
       HIP_CHECK(hipMalloc(&tab->cols[c], esz * rows));

@@ -546,14 +546,18 @@ extern "C" int sdb_host_encode_col_i64(const int64_t* vals, uint64_t rows,
     const uint64_t nw = (uint64_t(r1 - r0) * width + 31) / 32;
     const size_t p0 = payload.size();
     payload.resize(p0 + nw, 0u);
-    for (uint64_t r = r0; r < r1; ++r) {
-      const uint64_t v = uint64_t(vals[r] - mn);
-      const uint64_t bit = (r - r0) * width;
-      const uint64_t w = bit >> 5;
-      const uint32_t sh = uint32_t(bit & 31);
-      payload[p0 + w] |= uint32_t(v << sh);
-      if (sh + width > 32) payload[p0 + w + 1] |= uint32_t(v >> (32 - sh));
-    }
+    // width == 0 (constant group): nw == 0, nothing to pack — the old
+    // unconditional loop wrote payload[p0] |= 0 one past the end (a
+    // null-deref for tiny columns; found by tests/test_codec_fuzz.py)
+    if (width)
+      for (uint64_t r = r0; r < r1; ++r) {
+        const uint64_t v = uint64_t(vals[r] - mn);
+        const uint64_t bit = (r - r0) * width;
+        const uint64_t w = bit >> 5;
+        const uint32_t sh = uint32_t(bit & 31);
+        payload[p0 + w] |= uint32_t(v << sh);
+        if (sh + width > 32) payload[p0 + w + 1] |= uint32_t(v >> (32 - sh));
+      }
     desc[g] = d;
   }
   auto align64 = [](uint64_t x) { return (x + 63) & ~63ull; };
@@ -580,10 +584,20 @@ extern "C" int sdb_host_encode_col_i64(const int64_t* vals, uint64_t rows,
 // scalar decode (builder verification + oracle cross-check)
 extern "C" int sdb_host_decode_col_i64(const void* blob, uint64_t size,
                                        int64_t* out, uint64_t rows) {
+  if (!blob || size < sizeof(SdbColHeader)) return -5;
   const SdbColHeader* hdr = (const SdbColHeader*)blob;
-  if (!hdr || hdr->magic != SDB_COL_MAGIC || hdr->rows != rows ||
-      hdr->size > size)
+  if (hdr->magic != SDB_COL_MAGIC || hdr->rows != rows || hdr->size > size ||
+      hdr->group_rows == 0)
     return -5;
+  // every offset/extent must sit inside the declared blob span (a truncated
+  // or corrupted blob must fail loudly, not read out of bounds)
+  const uint64_t ng_need = (rows + hdr->group_rows - 1) / hdr->group_rows;
+  if (hdr->ngroups != ng_need ||
+      hdr->off_desc > hdr->size || hdr->off_payload > hdr->size ||
+      uint64_t(hdr->ngroups) * sizeof(SdbColGroupDesc) >
+        hdr->size - hdr->off_desc)
+    return -5;
+  const uint64_t paywords = (hdr->size - hdr->off_payload) / 4;
   const auto* desc =
     (const SdbColGroupDesc*)((const uint8_t*)blob + hdr->off_desc);
   const auto* pl = (const uint32_t*)((const uint8_t*)blob + hdr->off_payload);
@@ -591,6 +605,11 @@ extern "C" int sdb_host_decode_col_i64(const void* blob, uint64_t size,
     const uint64_t r0 = uint64_t(g) * hdr->group_rows;
     const uint64_t r1 = std::min<uint64_t>(rows, r0 + hdr->group_rows);
     const SdbColGroupDesc d = desc[g];
+    if (d.width > 32) return -5;
+    if (d.width &&
+        (d.word_off > paywords ||
+         ((r1 - r0) * d.width + 31) / 32 > paywords - d.word_off))
+      return -5;
     const uint32_t* w = pl + d.word_off;
     const uint64_t mask =
       d.width >= 32 ? 0xFFFFFFFFull : ((1ull << d.width) - 1);

@@ -1,0 +1,128 @@
+"""Property-based codec fuzzing (hypothesis): the block codec is the parity
+core (FormatTraits128, format_block_128.hpp:51-379 write / :446-636 read),
+so the host encoder and the oracle restatement are cross-checked on
+adversarial shapes the hand-written cases might miss: all-same runs that
+break mid-block, gap spikes that force byte-width jumps inside streamvbyte,
+dense regions that flip to bitset, maximum 32-bit doc ids, freq
+distributions straddling every family boundary.
+
+Both directions: host encode -> oracle decode and oracle encode -> host
+decode must reproduce the input exactly (byte streams may differ only if an
+encoder picks a different legal family — they don't: encoder selection is
+deterministic and mirrored, which test_cross_encoder_bytes pins)."""
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+import serenedb_amd as sa
+from oracle import pyoracle as po
+
+
+def docs_strategy():
+    """ascending doc-id blocks (<=128) built from segments of adversarial
+    delta patterns: all-same runs, unit steps (bitset bait), huge gaps."""
+    seg = st.one_of(
+        st.tuples(st.integers(1, 64), st.integers(1, 4)),          # run of same small delta
+        st.tuples(st.integers(1, 32), st.just(1)),                 # dense run
+        st.tuples(st.integers(1, 4), st.integers(1, 1 << 24)),     # gap spikes
+        st.tuples(st.integers(1, 16), st.integers(1, 1 << 14)),
+    )
+    return st.lists(seg, min_size=1, max_size=8).map(_segments_to_docs)
+
+
+def _segments_to_docs(segs):
+    deltas = []
+    for n, d in segs:
+        deltas.extend([d] * n)
+    deltas = deltas[:128]
+    docs = np.cumsum(np.array(deltas, dtype=np.uint64))
+    docs = docs[docs <= 0xFFFFFFFE]
+    return docs.astype(np.uint32)
+
+
+@st.composite
+def freqs_strategy(draw):
+    n = draw(st.integers(1, 128))
+    kind = draw(st.integers(0, 3))
+    if kind == 0:  # all-same (1..4-byte widths)
+        v = draw(st.sampled_from([1, 2, 255, 256, 65536, 1 << 24]))
+        return np.full(n, v, dtype=np.uint32)
+    if kind == 1:  # small mixed (bitpack bait for full blocks)
+        return np.array(draw(st.lists(st.integers(1, 31), min_size=n,
+                                      max_size=n)), dtype=np.uint32)
+    if kind == 2:  # byte-width straddle (svb tails)
+        return np.array(draw(st.lists(st.sampled_from(
+            [1, 255, 256, 65535, 65536, (1 << 24) - 1, 1 << 24]),
+            min_size=n, max_size=n)), dtype=np.uint32)
+    return np.array(draw(st.lists(st.integers(1, (1 << 31) - 1), min_size=n,
+                                  max_size=n)), dtype=np.uint32)
+
+
+@settings(max_examples=150, deadline=None)
+@given(docs=docs_strategy(), prev_gap=st.integers(0, 1 << 20))
+def test_doc_block_roundtrip_cross(docs, prev_gap):
+    if len(docs) == 0:
+        return
+    prev = int(docs[0]) - 1 if prev_gap > int(docs[0]) - 1 else prev_gap
+    docs = docs + 0  # copy
+    if prev >= int(docs[0]):
+        prev = int(docs[0]) - 1
+    enc_h = sa.encode_doc_block(docs, prev)
+    enc_o = po.encode_doc_block(docs, prev)
+    assert enc_h == enc_o, "encoder family/bytes diverge (host vs oracle)"
+    dec_o, used_o = po.decode_doc_block(enc_h, len(docs), prev)
+    np.testing.assert_array_equal(dec_o, docs)
+    assert used_o == len(enc_h)
+    dec_h, used_h = sa.decode_doc_block(enc_o, len(docs), prev)
+    np.testing.assert_array_equal(dec_h, docs)
+    assert used_h == len(enc_o)
+
+
+@settings(max_examples=150, deadline=None)
+@given(freqs=freqs_strategy())
+def test_freq_block_roundtrip_cross(freqs):
+    enc_h = sa.encode_freq_block(freqs)
+    enc_o = po.encode_freq_block(freqs)
+    assert enc_h == enc_o
+    dec_o, used_o = po.decode_freq_block(enc_h, len(freqs))
+    np.testing.assert_array_equal(dec_o, freqs)
+    assert used_o == len(enc_h)
+    dec_h, used_h = sa.decode_freq_block(enc_o, len(freqs))
+    np.testing.assert_array_equal(dec_h, freqs)
+    assert used_h == len(enc_o)
+
+
+@settings(max_examples=60, deadline=None)
+@given(vals=st.lists(st.integers(-(1 << 31), (1 << 31) - 1), min_size=1,
+                     max_size=4000),
+       base=st.integers(-(1 << 60), 1 << 60),
+       tile=st.sampled_from([1, 7, 1000]))
+def test_col_i64_roundtrip_fuzz(vals, base, tile):
+    # per-group delta must fit u32 (the round-1 codec contract); the base
+    # can sit anywhere in i64
+    arr = np.array(vals * tile, dtype=np.int64)[:200_000] + base
+    blob = sa.encode_col_i64(arr)
+    out = sa.decode_col_i64(blob, len(arr))
+    np.testing.assert_array_equal(out, arr)
+
+
+def test_col_i64_wide_range_rejected():
+    # > 32-bit in-group delta range is rejected loudly, not mis-encoded
+    arr = np.array([0, 1 << 40], dtype=np.int64)
+    with pytest.raises(AssertionError):
+        sa.encode_col_i64(arr)
+
+
+def test_col_i64_malformed_rejected():
+    """table-free validation of the FoR blob: truncation and bad magic are
+    caught host-side before any decode."""
+    arr = np.arange(1000, dtype=np.int64)
+    blob = bytearray(sa.encode_col_i64(arr))
+    bad = bytes(blob[:40])  # truncated below header+desc
+    with pytest.raises(AssertionError):
+        sa.decode_col_i64(bad, 1000)
+    blob2 = bytearray(blob)
+    blob2[0] ^= 0xFF  # corrupt magic
+    with pytest.raises(AssertionError):
+        sa.decode_col_i64(bytes(blob2), 1000)
