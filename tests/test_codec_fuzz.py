@@ -114,6 +114,91 @@ def test_col_i64_wide_range_rejected():
         sa.encode_col_i64(arr)
 
 
+@settings(max_examples=40, deadline=None)
+@given(data=st.data())
+def test_segment_pipeline_fuzz(data):
+    """Whole-segment pipeline on random corpora: build -> oracle decode
+    round trip -> exact top-k vs an independent brute-force scorer. Random
+    postings shapes cross codec-family boundaries inside one segment."""
+    from tests.test_topk import brute_topk  # reuse the independent scorer
+
+    doc_count = data.draw(st.integers(130, 3000))
+    nterms = data.draw(st.integers(1, 4))
+    rng = np.random.default_rng(data.draw(st.integers(0, 1 << 30)))
+    postings = []
+    for _ in range(nterms):
+        style = data.draw(st.integers(0, 3))
+        if style == 0:  # dense run (bitset bait)
+            n = min(doc_count, data.draw(st.integers(1, 400)))
+            start = rng.integers(1, doc_count - n + 2)
+            docs = np.arange(start, start + n, dtype=np.uint32)
+        elif style == 1:  # uniform random
+            n = data.draw(st.integers(1, min(600, doc_count)))
+            docs = np.sort(rng.choice(
+                np.arange(1, doc_count + 1, dtype=np.uint32), n,
+                replace=False))
+        elif style == 2:  # strided (all-same deltas)
+            step = data.draw(st.integers(1, 50))
+            docs = np.arange(1, doc_count + 1, step, dtype=np.uint32)
+        else:  # empty term
+            docs = np.zeros(0, dtype=np.uint32)
+        freqs = rng.integers(1, 200, len(docs)).astype(np.uint32)
+        postings.append((docs, freqs))
+    norms = sa.synth_norms(int(rng.integers(0, 1 << 20)), doc_count)
+    blob = sa.build_segment(doc_count, postings, norms)
+    for t, (docs, freqs) in enumerate(postings):
+        if len(docs) == 0:
+            continue
+        ddocs, dfreqs = po.decode_term(blob, t, len(docs))
+        np.testing.assert_array_equal(ddocs, docs)
+        np.testing.assert_array_equal(dfreqs, freqs)
+    k = data.draw(st.sampled_from([1, 10, 500]))
+    mm = data.draw(st.integers(1, nterms))
+    hits, total = po.execute_topk([blob], list(range(nterms)),
+                                  [1.0] * nterms, k, min_match=mm)
+    sels = [0.0] * nterms  # brute_topk ignores sels content beyond length
+    order, scores, nmatch = brute_topk(postings, norms, doc_count, sels, k,
+                                       min_match=mm)
+    assert total == nmatch
+    assert [int(h["doc"]) for h in hits] == [int(d) for d in order]
+    for h in hits:
+        assert h["score"] == scores[int(h["doc"])]
+
+
+@settings(max_examples=30, deadline=None)
+@given(data=st.data())
+def test_scan_agg_oracle_fuzz(data):
+    """Oracle scan->filter->group-by vs direct numpy on random shapes
+    (group counts, predicate ops, value ranges incl. negatives)."""
+    rows = data.draw(st.integers(1, 30_000))
+    ngroups = data.draw(st.sampled_from([1, 3, 64, 1024]))
+    rng = np.random.default_rng(data.draw(st.integers(0, 1 << 30)))
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(-(1 << 40), 1 << 40, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    op = data.draw(st.sampled_from([1, 2, 3]))
+    lo = int(rng.integers(-(1 << 40), 1 << 40))
+    hi = int(rng.integers(lo, 1 << 40)) if op == 3 else 0
+    ocnt, osi, osf, opassed = po.scan_agg(keys, v1, v2, ngroups, pred_op=op,
+                                          lo=lo, hi=hi)
+    if op == 1:
+        mask = v1 < lo
+    elif op == 2:
+        mask = v1 >= lo
+    else:
+        mask = (v1 >= lo) & (v1 <= hi)
+    assert opassed == int(mask.sum())
+    np.testing.assert_array_equal(
+        ocnt, np.bincount(keys[mask], minlength=ngroups))
+    # int64 sums wrap exactly like the oracle's (use object-free wraparound)
+    esi = np.zeros(ngroups, dtype=np.int64)
+    np.add.at(esi, keys[mask], v1[mask])
+    np.testing.assert_array_equal(osi, esi)
+    esf = np.bincount(keys[mask], weights=v2[mask].astype(np.float64),
+                      minlength=ngroups)
+    np.testing.assert_allclose(osf, esf, rtol=1e-7, atol=1e-9)
+
+
 def test_col_i64_malformed_rejected():
     """table-free validation of the FoR blob: truncation and bad magic are
     caught host-side before any decode."""
