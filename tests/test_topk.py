@@ -161,6 +161,64 @@ def test_topk_exact_vs_brute(seed, mm):
         assert h["score"] == scores[int(h["doc"])], int(h["doc"])
 
 
+def test_wand_block_max_fixture():
+    """Transcribed WAND pruning fixture (formats_15_tests.cpp:901-917
+    LongPostingsWandThreshold60/100): docs 1..10000 with freqs from a
+    default-seeded mt19937 normal draw (committed vectors +
+    tools/gen_wand_freqs.cpp). The reference's wanderator visits exactly
+    the docs of 128-blocks whose block-max freq exceeds the threshold,
+    plus the tail block (its skip list has no tail entry): 1680 at
+    threshold 60 over N(40,7), 16 at threshold 100 over N(50,13). This
+    repo's per-block max_freq descriptors must reproduce those counts
+    (the tail is modeled separately: our descriptors DO bound the tail —
+    a documented refinement that can only prune more, never less)."""
+    import ctypes as CT
+
+    class _Desc(CT.Structure):
+        _fields_ = [("prev_doc", CT.c_uint32), ("last_doc", CT.c_uint32),
+                    ("doc_off", CT.c_uint32), ("freq_off", CT.c_uint32),
+                    ("len", CT.c_uint16), ("flags", CT.c_uint16),
+                    ("max_freq", CT.c_uint32), ("min_norm", CT.c_uint32)]
+
+    class _View(CT.Structure):
+        _fields_ = [("hdr", CT.c_void_p), ("terms", CT.c_void_p),
+                    ("desc", CT.c_void_p), ("norms", CT.c_void_p),
+                    ("payload", CT.c_void_p)]
+
+    class _Term(CT.Structure):
+        _fields_ = [("desc_begin", CT.c_uint64), ("desc_end", CT.c_uint64),
+                    ("payload_begin", CT.c_uint64), ("df", CT.c_uint64),
+                    ("max_freq", CT.c_uint32), ("pad", CT.c_uint32)]
+
+    for fname, thr, expected_ref in (("wand_freqs_n40_7.txt", 60, 1680),
+                                     ("wand_freqs_n50_13.txt", 100, 16)):
+        freqs = np.array([int(x) for x in open(
+            os.path.join(GOLDEN, fname))], dtype=np.uint32)
+        assert len(freqs) == 10000
+        docs = np.arange(1, 10001, dtype=np.uint32)
+        norms = np.ones(10001, dtype=np.uint32)
+        norms[0] = 0
+        blob = sa.build_segment(10000, [(docs, freqs)], norms)
+        buf = np.frombuffer(blob, dtype=np.uint8)
+        v = _View()
+        rc = sa.host().sdb_host_segment_parse(
+            buf.ctypes.data_as(CT.c_void_p), CT.c_uint64(len(buf)),
+            CT.byref(v))
+        assert rc == 0, rc
+        te = CT.cast(v.terms, CT.POINTER(_Term))[0]
+        nblocks = te.desc_end - te.desc_begin
+        darr = CT.cast(v.desc, CT.POINTER(_Desc))
+        visited_ref_model = 0  # reference: tail block always visited
+        for b in range(nblocks):
+            d = darr[te.desc_begin + b]
+            # our descriptor max_freq must equal the true block max
+            exp_max = int(freqs[d.prev_doc:d.last_doc].max())
+            assert d.max_freq == exp_max, (b, d.max_freq, exp_max)
+            if d.len < 128 or d.max_freq > thr:
+                visited_ref_model += d.len
+        assert visited_ref_model == expected_ref, (fname, visited_ref_model)
+
+
 def test_filter_boost_oracle():
     """Per-doc filter boost (HasFilterBoost scorer variants,
     bm25.cpp:112-140): each term contribution is multiplied by fb[doc]
