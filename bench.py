@@ -406,11 +406,18 @@ def bench_scan(args):
         assert rc == 0, rc
         if dist:
             import torch
+            # 24 KB partial-agg merge: COUNT/SUM(i64) slots
+            # merge in the i64 field, SUM(f32->f64) slots in the f64 field
             arr = np.array([out[i].i64 for i in range(ngroups * 3)],
                            dtype=np.int64)
+            farr = np.array([out[i].f64 for i in range(ngroups * 3)],
+                            dtype=np.float64)
             t = torch.from_numpy(arr).cuda()
-            dist.all_reduce(t)  # 24 KB partial-agg merge
+            tf = torch.from_numpy(farr).cuda()
+            dist.all_reduce(t)
+            dist.all_reduce(tf)
             _ = t.cpu().numpy()
+            _ = tf.cpu().numpy()
 
     def sync():
         if dist:

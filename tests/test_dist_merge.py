@@ -65,6 +65,69 @@ def _worker(rank, world, port, q):
     dist.destroy_process_group()
 
 
+def _scan_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from oracle import pyoracle as po
+
+    rows_total, ngroups, seed = 120_000, 64, 21
+    per = rows_total // world
+    lo = rank * per
+    hi = rows_total if rank == world - 1 else (rank + 1) * per
+    rng = np.random.default_rng(seed)  # same stream; slice the shard
+    keys = rng.integers(0, ngroups, rows_total).astype(np.int64)[lo:hi]
+    v1r = rng.integers(0, 1 << 20, rows_total).astype(np.int64)
+    v2r = rng.normal(0, 1, rows_total).astype(np.float32)
+    v1, v2 = v1r[lo:hi], v2r[lo:hi]
+    c = int((1 << 20) * 0.1)
+    cnt, si, sf, passed = po.scan_agg(keys, v1, v2, ngroups, pred_op=1,
+                                      lo=c)
+    ti = torch.from_numpy(np.concatenate(
+        [cnt.astype(np.int64), si.astype(np.int64), [np.int64(passed)]]))
+    tf = torch.from_numpy(sf.astype(np.float64))
+    dist.all_reduce(ti)
+    dist.all_reduce(tf)
+    if rank == 0:
+        q.put((ti.numpy().copy(), tf.numpy().copy()))
+    dist.destroy_process_group()
+
+
+def test_gloo_scan_agg_merge_equals_full():
+    """bench.py's N>1 scan-aggregate merge (sum-allreduce of the group
+    table, i64 and f64 planes) equals the single-shard result."""
+    from oracle import pyoracle as po
+
+    rows_total, ngroups, seed = 120_000, 64, 21
+    rng = np.random.default_rng(seed)
+    keys = rng.integers(0, ngroups, rows_total).astype(np.int64)
+    v1 = rng.integers(0, 1 << 20, rows_total).astype(np.int64)
+    v2 = rng.normal(0, 1, rows_total).astype(np.float32)
+    c = int((1 << 20) * 0.1)
+    fcnt, fsi, fsf, fpassed = po.scan_agg(keys, v1, v2, ngroups, pred_op=1,
+                                          lo=c)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_scan_worker, args=(r, 2, 29523, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        ti, tf = q.get(timeout=300)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    np.testing.assert_array_equal(ti[:ngroups], fcnt)
+    np.testing.assert_array_equal(ti[ngroups:2 * ngroups], fsi)
+    assert ti[-1] == fpassed
+    np.testing.assert_allclose(tf, fsf, rtol=1e-12)
+
+
 @pytest.mark.parametrize("world,port", [(2, 29511), (4, 29517)])
 def test_gloo_sharded_merge_equals_full(world, port):
     import serenedb_amd as sa
