@@ -303,6 +303,16 @@ def test_golden_multisegment_order():
                 for h in hits]
         assert seqs == [3, 7, 0, 2, 5], (scorer, seqs)
         assert total == 5
+    # single-term multi-segment case (bm25_test.cpp:609, tfidf_test.cpp:635
+    # kExpected{0, 2, 5}): term "6" spans both segments, equal scores tie
+    # in (segment, doc) order
+    for scorer in ("bm25", "tfidf"):
+        hits, total = po.execute_topk([b0, b1], [vocab["6"]], [1.0], 8,
+                                      scorer=scorer)
+        seqs = [(seq0 if h["segment"] == 0 else seq1)[int(h["doc"]) - 1]
+                for h in hits]
+        assert seqs == [0, 2, 5], (scorer, seqs)
+        assert total == 3
 
 
 def test_hybrid_chain_vs_brute():
