@@ -1124,19 +1124,19 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
 
 int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   if (!ctx) return SDB_ERR_INVALID;
-  hipFree(ctx->d_cands);
-  hipFree(ctx->d_cand_count);
-  hipFree(ctx->d_total_matches);
-  hipFree(ctx->d_gthresh);
-  hipFree(ctx->d_ghist);
-  hipFree(ctx->d_buckets);
-  hipFree(ctx->d_overflow);
-  hipFree(ctx->d_terms);
-  hipHostFree(ctx->h_counts);
-  hipHostFree(ctx->h_matches);
-  hipEventDestroy(ctx->ev_a);
-  hipEventDestroy(ctx->ev_b);
-  hipStreamDestroy(ctx->stream);
+  (void)hipFree(ctx->d_cands);
+  (void)hipFree(ctx->d_cand_count);
+  (void)hipFree(ctx->d_total_matches);
+  (void)hipFree(ctx->d_gthresh);
+  (void)hipFree(ctx->d_ghist);
+  (void)hipFree(ctx->d_buckets);
+  (void)hipFree(ctx->d_overflow);
+  (void)hipFree(ctx->d_terms);
+  (void)hipHostFree(ctx->h_counts);
+  (void)hipHostFree(ctx->h_matches);
+  (void)hipEventDestroy(ctx->ev_a);
+  (void)hipEventDestroy(ctx->ev_b);
+  (void)hipStreamDestroy(ctx->stream);
   delete ctx;
   return SDB_OK;
 }
@@ -1184,12 +1184,12 @@ int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
 
 int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
   if (!ctx || !seg) return SDB_ERR_INVALID;
-  hipFree(seg->desc);
-  hipFree(seg->payload);
-  hipFree(seg->norms);
+  (void)hipFree(seg->desc);
+  (void)hipFree(seg->payload);
+  (void)hipFree(seg->norms);
   for (int i = 0; i < 4; ++i)
-    if (seg->fcols[i]) hipFree(seg->fcols[i]);
-  if (seg->fboost) hipFree(seg->fboost);
+    if (seg->fcols[i]) (void)hipFree(seg->fcols[i]);
+  if (seg->fboost) (void)hipFree(seg->fboost);
   std::free(seg->terms_host);
   delete seg;
   return SDB_OK;
@@ -1626,8 +1626,8 @@ int sdb_gpu_execute_match_docs(SdbGpuCtx* ctx, SdbGpuSegment* seg,
     HIP_CHECK(hipMemcpyAsync(col_out, d_vals, 8 * n, hipMemcpyDeviceToHost,
                              ctx->stream));
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
-    hipFree(d_docs);
-    hipFree(d_vals);
+    (void)hipFree(d_docs);
+    (void)hipFree(d_vals);
   }
   *out_count = n;
   return SDB_OK;
@@ -1653,8 +1653,8 @@ int sdb_gpu_decode_term(SdbGpuCtx* ctx, SdbGpuSegment* seg, uint32_t term_idx,
   std::vector<uint32_t> tmp_d(cap), tmp_f(cap);
   HIP_CHECK(hipMemcpy(tmp_d.data(), d_docs, cap * 4, hipMemcpyDeviceToHost));
   HIP_CHECK(hipMemcpy(tmp_f.data(), d_freqs, cap * 4, hipMemcpyDeviceToHost));
-  hipFree(d_docs);
-  hipFree(d_freqs);
+  (void)hipFree(d_docs);
+  (void)hipFree(d_freqs);
   uint32_t n = 0;
   for (uint64_t b = 0; b < nblocks; ++b) {
     // need lens: all full except possibly last

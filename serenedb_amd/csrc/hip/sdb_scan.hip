@@ -762,7 +762,7 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
 
 int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab) {
   if (!ctx || !tab) return SDB_ERR_INVALID;
-  for (uint32_t c = 0; c < tab->ncols; ++c) hipFree(tab->cols[c]);
+  for (uint32_t c = 0; c < tab->ncols; ++c) (void)hipFree(tab->cols[c]);
   delete tab;
   return SDB_OK;
 }
@@ -904,8 +904,8 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   HIP_CHECK(hipMemcpyAsync(&h_passed, d_passed, 8, hipMemcpyDeviceToHost,
                            stream));
   HIP_CHECK(hipStreamSynchronize(stream));
-  hipFree(d_out);
-  hipFree(d_passed);
+  (void)hipFree(d_out);
+  (void)hipFree(d_passed);
   for (uint32_t g = 0; g < ngroups; ++g) {
     for (uint32_t q = 0; q < naggs; ++q) {
       const unsigned long long raw = h_out[g * naggs + q];
