@@ -59,7 +59,7 @@ def freqs_strategy(draw):
                                   max_size=n)), dtype=np.uint32)
 
 
-@settings(max_examples=150, deadline=None)
+@settings(max_examples=300, deadline=None)
 @given(docs=docs_strategy(), prev_gap=st.integers(0, 1 << 20))
 def test_doc_block_roundtrip_cross(docs, prev_gap):
     if len(docs) == 0:
@@ -79,7 +79,7 @@ def test_doc_block_roundtrip_cross(docs, prev_gap):
     assert used_h == len(enc_o)
 
 
-@settings(max_examples=150, deadline=None)
+@settings(max_examples=300, deadline=None)
 @given(freqs=freqs_strategy())
 def test_freq_block_roundtrip_cross(freqs):
     enc_h = sa.encode_freq_block(freqs)
@@ -93,7 +93,7 @@ def test_freq_block_roundtrip_cross(freqs):
     assert used_h == len(enc_o)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=120, deadline=None)
 @given(vals=st.lists(st.integers(-(1 << 31), (1 << 31) - 1), min_size=1,
                      max_size=4000),
        base=st.integers(-(1 << 60), 1 << 60),
@@ -114,7 +114,7 @@ def test_col_i64_wide_range_rejected():
         sa.encode_col_i64(arr)
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=80, deadline=None)
 @given(data=st.data())
 def test_segment_pipeline_fuzz(data):
     """Whole-segment pipeline on random corpora: build -> oracle decode
@@ -165,7 +165,7 @@ def test_segment_pipeline_fuzz(data):
         assert h["score"] == scores[int(h["doc"])]
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=60, deadline=None)
 @given(data=st.data())
 def test_scan_agg_oracle_fuzz(data):
     """Oracle scan->filter->group-by vs direct numpy on random shapes
