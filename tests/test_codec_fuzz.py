@@ -199,6 +199,61 @@ def test_scan_agg_oracle_fuzz(data):
     np.testing.assert_allclose(osf, esf, rtol=1e-7, atol=1e-9)
 
 
+@settings(max_examples=40, deadline=None)
+@given(data=st.data())
+def test_hybrid_chain_oracle_fuzz(data):
+    """Oracle hybrid predicate chains vs direct numpy: random chain
+    shapes (1-3 predicates, mixed LT/GE/BETWEEN) over random columns."""
+    doc_count = data.draw(st.integers(2000, 20_000))
+    seed = data.draw(st.integers(0, 1 << 30))
+    rng = np.random.default_rng(seed)
+    sels = [0.15, 0.08]
+    postings = [sa.synth_postings(seed % 1000, doc_count, t, s)
+                for t, s in enumerate(sels)]
+    norms = sa.synth_norms(seed % 1000, doc_count)
+    blob = sa.build_segment(doc_count, postings, norms)
+    span = 1 << 20
+    ncols = data.draw(st.integers(1, 3))
+    cols = [rng.integers(0, span, doc_count + 1).astype(np.int64)
+            for _ in range(ncols)]
+    ops = [3] + [data.draw(st.sampled_from([1, 2, 3]))
+                 for _ in range(ncols - 1)]
+    los = [int(span * 0.2)] + [int(rng.integers(0, span))
+                               for _ in range(ncols - 1)]
+    his = []
+    for i, op in enumerate(ops):
+        his.append(int(rng.integers(los[i], span)) if op == 3 else 0)
+    nb = data.draw(st.sampled_from([1, 16, 128]))
+    hits, total, bcnt, bsum = po.execute_topk_hybrid_chain(
+        blob, [0, 1], [1.0, 1.0], 100, cols, ops, los, his, nb)
+    # numpy expectation
+    matched = np.zeros(doc_count + 1, dtype=bool)
+    for docs, _ in postings:
+        matched[docs] = True
+    mask = matched.copy()
+    for i, op in enumerate(ops):
+        c = cols[i]
+        if op == 1:
+            mask &= c < los[i]
+        elif op == 2:
+            mask &= c >= los[i]
+        else:
+            mask &= (c >= los[i]) & (c <= his[i])
+    assert total == int(mask.sum())
+    surv = np.nonzero(mask)[0]
+    c0 = cols[0]
+    spanb = his[0] - los[0] + 1
+    ecnt = np.zeros(nb, dtype=np.int64)
+    esum = np.zeros(nb, dtype=np.int64)
+    for d in surv:
+        b = min((int(c0[d]) - los[0]) * nb // spanb, nb - 1)
+        ecnt[b] += 1
+        esum[b] += int(c0[d])
+    np.testing.assert_array_equal(bcnt, ecnt)
+    np.testing.assert_array_equal(bsum, esum)
+    assert set(int(h["doc"]) for h in hits) <= set(int(d) for d in surv)
+
+
 def test_col_i64_malformed_rejected():
     """table-free validation of the FoR blob: truncation and bad magic are
     caught host-side before any decode."""
