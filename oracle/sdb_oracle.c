@@ -590,9 +590,12 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
         while (i < c->buf_len && c->buf_docs[i] <= hi) {
           const uint32_t doc = c->buf_docs[i];
           const uint32_t off = doc - lo;
-          float s1 = o_score_one(c->scorer, c->num, c->nc, c->nl,
+          /* filter boost folds into num BEFORE the score form, mirroring
+           * the reference's op order (bm25.cpp: c0 = boost*num, then
+           * c0 - c0*c1/(c1+freq); tfidf boost*num*sqrt(freq)) */
+          const float nm = o_fboost ? c->num * o_fboost[doc] : c->num;
+          float s1 = o_score_one(c->scorer, nm, c->nc, c->nl,
                                  c->buf_freqs[i], norms[doc]);
-          if (o_fboost) s1 *= o_fboost[doc]; /* one f32 mul, GPU-identical */
           score_win[off] += s1;
           const uint8_t cc = ++cnt_win[off];
           if (cc == min_match) mask[off >> 6] |= 1ull << (off & 63);

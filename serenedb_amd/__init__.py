@@ -574,3 +574,97 @@ class GpuContext:
         if rc != 0:
             raise RuntimeError(f"sdb_gpu_decode_term rc={rc}")
         return docs, freqs
+
+    # ---- columnar scan helpers (ctypes mirrors of sdb_gpu.h) ----
+
+    class _ColView(C.Structure):
+        _fields_ = [("data", C.c_void_p), ("rows", C.c_uint64),
+                    ("type", C.c_int)]
+
+    class _PredSpec(C.Structure):
+        _fields_ = [("col", C.c_uint32), ("op", C.c_int),
+                    ("ilo", C.c_int64), ("ihi", C.c_int64),
+                    ("flo", C.c_float), ("fhi", C.c_float)]
+
+    class _AggSpec(C.Structure):
+        _fields_ = [("col", C.c_uint32), ("op", C.c_int)]
+
+    class _AggResult(C.Structure):
+        _fields_ = [("i64", C.c_int64), ("f64", C.c_double)]
+
+    def load_table(self, arrays, codecs=None):
+        """arrays: list of numpy arrays (int64 or float32). codecs: per
+        column, "raw" or "for" (i64 only). Returns an opaque table handle.
+        Keeps the encoded blobs alive for the duration of the call only
+        (table_load uploads them)."""
+        import numpy as np
+
+        n = len(arrays)
+        codecs = codecs or ["raw"] * n
+        views = (self._ColView * n)()
+        keep = []
+        rows = len(arrays[0])
+        for i, (a, enc) in enumerate(zip(arrays, codecs)):
+            if a.dtype == np.float32:
+                a = np.ascontiguousarray(a)
+                keep.append(a)
+                views[i] = self._ColView(
+                    a.ctypes.data_as(C.c_void_p).value, rows, 1)
+            elif enc == "for":
+                blob = encode_col_i64(np.ascontiguousarray(a, np.int64))
+                bv = np.frombuffer(blob, dtype=np.uint8)
+                keep.append(bv)
+                views[i] = self._ColView(
+                    bv.ctypes.data_as(C.c_void_p).value, rows, 2)
+            else:
+                a = np.ascontiguousarray(a, np.int64)
+                keep.append(a)
+                views[i] = self._ColView(
+                    a.ctypes.data_as(C.c_void_p).value, rows, 0)
+        tab = C.c_void_p(0)
+        rc = self._lib.sdb_gpu_table_load(self._ctx, views, C.c_uint32(n),
+                                          C.c_uint64(rows), C.byref(tab))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_table_load rc={rc}")
+        return tab
+
+    def free_table(self, tab):
+        self._lib.sdb_gpu_table_free(self._ctx, tab)
+
+    def scan_agg(self, tab, group_col, ngroups, preds, aggs):
+        """preds: list of (col, op, lo, hi) with SdbPredOp numeric op
+        (1=LT 2=GE 3=BETWEEN 4=EQ); lo/hi int or float (float for f32
+        predicate columns). aggs: list of (col, op) with 0=COUNT
+        1=SUM_I64 2=SUM_F64. Returns (i64 results [ngroups, naggs],
+        f64 results [ngroups, naggs], rows_passed)."""
+        import numpy as np
+
+        np_ = len(preds)
+        pa = (self._PredSpec * max(np_, 1))()
+        for i, (col, op, lo, hi) in enumerate(preds):
+            if isinstance(lo, float) or isinstance(hi, float):
+                pa[i] = self._PredSpec(col, op, 0, 0, lo, hi)
+            else:
+                pa[i] = self._PredSpec(col, op, lo, hi, 0, 0)
+        na = len(aggs)
+        aa = (self._AggSpec * na)(*[self._AggSpec(c, o) for c, o in aggs])
+        out = (self._AggResult * (ngroups * na))()
+        passed = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_scan_agg(
+            self._ctx, tab, C.c_uint32(group_col), C.c_uint32(ngroups), pa,
+            C.c_uint32(np_), aa, C.c_uint32(na), out, C.byref(passed))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_scan_agg rc={rc}")
+        i64 = np.array([[out[g * na + q].i64 for q in range(na)]
+                        for g in range(ngroups)], dtype=np.int64)
+        f64 = np.array([[out[g * na + q].f64 for q in range(na)]
+                        for g in range(ngroups)], dtype=np.float64)
+        return i64, f64, passed.value
+
+    def load_table_i64(self, keys, vals, codec="raw"):
+        return self.load_table([keys, vals], [codec, codec])
+
+    def scan_agg_count_sum(self, tab, ngroups):
+        """COUNT(*) + SUM(col1) grouped by col0, no predicate."""
+        return self.scan_agg(tab, 0, ngroups, [],
+                             [(0, 0), (1, 1)])

