@@ -50,6 +50,8 @@
 // ---------------------------------------------------------------------------
 // tunables
 // ---------------------------------------------------------------------------
+#define SDB_TERM_SLOTS 64u  // per-segment term-table slots (multi-segment
+                             // launches stay async; sync only on wrap)
 #ifndef SDB_WIN_DOCS
 #define SDB_WIN_DOCS 24576u  // docs per workgroup window (96 KB f32 + 24 KB u8)
 #endif
@@ -305,17 +307,30 @@ __device__ void decode_freq_block_wave(const uint8_t* p, uint32_t len,
   }
 }
 
+// match-mark primitives: the general kernel keeps a u8 count window
+// (min_match semantics); the lean sweep kernel keeps a u64 match bitmask
+// (zeroing is 16x cheaper, match count is a popcount, and the result
+// sweep iterates set bits instead of re-scanning every doc slot).
+__device__ __forceinline__ void mark_match_u8(void* cnt, uint32_t off) {
+  ((uint8_t*)cnt)[off] = (uint8_t)(((uint8_t*)cnt)[off] + 1u);
+}
+__device__ __forceinline__ void mark_match_mask(void* cnt, uint32_t off) {
+  // lanes of one wave can hit the same word: needs the LDS atomic form
+  atomicOr((unsigned long long*)cnt + (off >> 6), 1ull << (off & 63u));
+}
+
 // Fused fast path for the dominant block shape (delta-bitpack docs +
 // bitpack freqs + bitpack norms, always full 128-doc blocks): issue every
 // stream's packed-word loads up front (one memory round trip instead of
 // three serialized ones), keep the two values per lane in registers, score
 // directly — no LDS scratch round trip. Returns false for any other family
 // combination (caller falls back to the generic per-stream decode).
+template <int LEAN>
 __device__ __forceinline__ bool try_block_fused(
   const uint8_t* pl, const SdbBlockDesc& d, int lane, uint32_t norm_stream,
   uint32_t lo, uint32_t hi, float num, float nc, float nl, uint32_t scorer,
   const uint32_t* norms_col, const float* fboost, float* swin,
-  uint8_t* cwin) {
+  void* cwin) {
   const uint8_t* db = pl + d.doc_off;
   const uint8_t* fb = pl + d.freq_off;
   const uint8_t* nb = fb + d.flags;
@@ -354,14 +369,14 @@ __device__ __forceinline__ bool try_block_fused(
     const float s = score_one(scorer, nm, nc, nl, f0, n0);
     const uint32_t off = doc0 - lo;
     swin[off] += s;
-    cwin[off] = (uint8_t)(cwin[off] + 1u);
+    if (LEAN) mark_match_mask(cwin, off); else mark_match_u8(cwin, off);
   }
   if (doc1 >= lo && doc1 <= hi) {
     const float nm = fboost ? num * fboost[doc1] : num;
     const float s = score_one(scorer, nm, nc, nl, f1, n1);
     const uint32_t off = doc1 - lo;
     swin[off] += s;
-    cwin[off] = (uint8_t)(cwin[off] + 1u);
+    if (LEAN) mark_match_mask(cwin, off); else mark_match_u8(cwin, off);
   }
   return true;
 }
@@ -369,11 +384,12 @@ __device__ __forceinline__ bool try_block_fused(
 // Dual-block fused path: both blocks' packed-word loads are issued before
 // either scan, so the two dependent-load chains overlap (the decode phase
 // is latency-bound — SDB_TIMING showed it at 43% of the kernel).
+template <int LEAN>
 __device__ __forceinline__ bool try_block_fused2(
   const uint8_t* pl, const SdbBlockDesc& da, const SdbBlockDesc& db_,
   int lane, uint32_t norm_stream, uint32_t lo, uint32_t hi, float num,
   float nc, float nl, uint32_t scorer, const uint32_t* norms_col,
-  const float* fboost, float* swin, uint8_t* cwin) {
+  const float* fboost, float* swin, void* cwin) {
   const uint8_t* adoc = pl + da.doc_off;
   const uint8_t* afrq = pl + da.freq_off;
   const uint8_t* anrm = afrq + da.flags;
@@ -431,7 +447,7 @@ __device__ __forceinline__ bool try_block_fused2(
     const float s = score_one(scorer, nm, nc, nl, frqs[e], nrms[e]);
     const uint32_t off = doc - lo;
     swin[off] += s;
-    cwin[off] = (uint8_t)(cwin[off] + 1u);
+    if (LEAN) mark_match_mask(cwin, off); else mark_match_u8(cwin, off);
   }
   return true;
 }
@@ -718,9 +734,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                       ? dcache[t * a.dcache_n + rel2]
                                       : a.desc[b2];
             if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
-                try_block_fused2(pl, d, d2, lane, a.norm_stream, lo, hi,
-                                 num, nc, nl, a.scorer, a.norms, a.fb,
-                                 swin, cwin)) {
+                try_block_fused2<0>(pl, d, d2, lane, a.norm_stream, lo,
+                                    hi, num, nc, nl, a.scorer, a.norms,
+                                    a.fb, swin, cwin)) {
               b += 2 * SDB_NWAVES;
               continue;
             }
@@ -753,8 +769,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         }
 #else
 #ifndef SDB_ABLATE_SCORE
-        if (try_block_fused(pl, d, lane, a.norm_stream, lo, hi, num, nc, nl,
-                            a.scorer, a.norms, a.fb, swin, cwin)) {
+        if (try_block_fused<0>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
+                               nl, a.scorer, a.norms, a.fb, swin, cwin)) {
           b += SDB_NWAVES;  // while-loop: explicit advance before continue
           continue;
         }
@@ -1023,6 +1039,402 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #endif
 }
 
+
+// ---------------------------------------------------------------------------
+// Lean sweep kernel (round-2): the headline path (min_match == 1, no pushed
+// column filter, scored). Same term-major phase structure and bit-exact
+// fp32 merge order as topk_window_kernel, but reshaped for throughput
+// (round-1 VERDICT: 1.1% of the HBM roofline, latency/overhead-bound):
+//  - the u8 match-count window becomes a u64 match BITMASK (min_match==1
+//    only needs membership): zeroing drops 16x, the match count is a
+//    popcount, and the result sweep visits set bits only instead of
+//    re-scanning every doc slot;
+//  - match count, candidate count, (sampled) histogram and the append
+//    offsets come out of ONE sparse sweep; the append re-walks set bits
+//    only (and only when the window holds any candidate);
+//  - geometry (window docs x threads) is a template parameter: smaller
+//    windows at 2 workgroups/CU let one WG's decode cover the other's
+//    term barriers (the fixed 24576x1024 1-WG/CU shape left every barrier
+//    stall empty).
+template <uint32_t WD, uint32_t NTH>
+// 2nd launch-bounds arg = min waves/SIMD: 4 keeps VGPRs <= 128 so two
+// 512-thread WGs (or one 1024-thread WG) co-reside per CU
+__launch_bounds__(NTH, 4) __global__
+void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
+  constexpr uint32_t NW = NTH / 64u;
+  constexpr uint32_t NWORDS = WD / 64u;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* swin = (float*)smem;                                   // WD * 4
+  unsigned long long* mwin = (unsigned long long*)(swin + WD);  // NWORDS * 8
+  uint32_t* scratch = (uint32_t*)(mwin + NWORDS);               // NW * 384
+  uint32_t* hist = scratch + NW * 384;                          // 256
+  uint32_t* shared_misc = hist + SDB_HIST_BINS;                 // 2 + NW
+  uint32_t* cursors = shared_misc + 2 + NW;                     // max terms
+  float* wub = (float*)(cursors + SDB_MAX_TERMS);               // max terms
+  SdbBlockDesc* dcache = (SdbBlockDesc*)(wub + SDB_MAX_TERMS);
+
+  const uint32_t tid = threadIdx.x;
+  const int lane = tid & 63;
+  const uint32_t wave = tid >> 6;
+  const float inv_smax = (float)SDB_HIST_BINS / a.smax;
+  uint32_t* gh = a.ghist + (blockIdx.x & 7u) * SDB_HIST_BINS;
+  unsigned long long wg_matches = 0;
+#ifdef SDB_TIMING
+  unsigned long long t_acc[6] = {0, 0, 0, 0, 0, 0};
+  long long t_mark = clock64();
+#define SDB_TS(idx)                                        \
+  if (tid == 0) {                                          \
+    const long long now_ = clock64();                      \
+    t_acc[idx] += (unsigned long long)(now_ - t_mark);     \
+    t_mark = now_;                                         \
+  }
+#else
+#define SDB_TS(idx)
+#endif
+
+  const uint32_t nwin = (a.doc_count + WD - 1) / WD;
+  const uint32_t per = (nwin + gridDim.x - 1) / gridDim.x;
+  const uint32_t w_lo = blockIdx.x * per;
+  const uint32_t w_hi = min(nwin, w_lo + per);
+  if (w_lo >= w_hi) return;
+
+  // one binary search per term per WORKGROUP (cursors then advance
+  // monotonically window to window)
+  if (tid < a.nterms) {
+    const TermDev te = terms[tid];
+    const uint32_t first_lo = 1u + w_lo * WD;
+    cursors[tid] = (uint32_t)(
+      lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, first_lo) -
+      te.desc_begin);
+  }
+  __syncthreads();
+
+  for (uint32_t w = w_lo; w < w_hi; ++w) {
+    const uint32_t lo = 1u + w * WD;
+    if (lo > a.doc_count) break;
+    const uint32_t hi = min(lo + WD - 1u, a.doc_count);
+    const uint32_t wlen = hi - lo + 1u;
+    const bool derive = ((w & 3u) == 0) || (w < w_lo + 2);
+    uint32_t my_excl_snap = 0;
+
+    for (uint32_t i = tid; i < WD; i += NTH) swin[i] = 0.0f;
+    for (uint32_t i = tid; i < NWORDS; i += NTH) mwin[i] = 0ull;
+    if (derive)
+      for (uint32_t i = tid; i < SDB_HIST_BINS; i += NTH) hist[i] = 0;
+    // stage this window's descriptors: term t's next dcache_n descs from
+    // its cursor, as coalesced u32 reads
+    {
+      const uint32_t words_per_term = a.dcache_n * 7u;
+      for (uint32_t i = tid; i < a.nterms * words_per_term; i += NTH) {
+        const uint32_t t = i / words_per_term;
+        const uint32_t wrd = i % words_per_term;
+        const TermDev te = terms[t];
+        const uint64_t b0 = te.desc_begin + cursors[t];
+        const uint32_t avail =
+          (uint32_t)(te.desc_end > b0 ? te.desc_end - b0 : 0);
+        if (wrd < avail * 7u)
+          ((uint32_t*)&dcache[t * a.dcache_n])[wrd] =
+            ((const uint32_t*)&a.desc[b0])[wrd];
+      }
+    }
+    if (tid == 0)
+      shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
+    if (a.wand) __syncthreads();  // wub scan reads other threads' staging
+    if (a.wand && tid < a.nterms) {
+      const TermDev te = terms[tid];
+      const uint32_t cur0w = cursors[tid];
+      float ub = 0.0f;
+      uint32_t i = 0;
+      for (; i < a.dcache_n; ++i) {
+        if (te.desc_begin + cur0w + i >= te.desc_end) break;
+        const SdbBlockDesc d = dcache[tid * a.dcache_n + i];
+        if (d.prev_doc >= hi) break;
+        const float u =
+          score_one(a.scorer, te.num, te.nc, te.nl, d.max_freq, d.min_norm);
+        ub = u > ub ? u : ub;
+      }
+      if (i == a.dcache_n) {  // unstaged tail must be covered too
+        for (uint64_t bb = te.desc_begin + cur0w + i; bb < te.desc_end;
+             ++bb) {
+          const SdbBlockDesc d = a.desc[bb];
+          if (d.prev_doc >= hi) break;
+          const float u = score_one(a.scorer, te.num, te.nc, te.nl,
+                                    d.max_freq, d.min_norm);
+          ub = u > ub ? u : ub;
+        }
+      }
+      wub[tid] = ub * a.fbmax * 1.0000019f;  // slack: see window kernel
+    }
+    __syncthreads();
+    SDB_TS(0)
+    float wand_total_ub = 0.0f;
+    if (a.wand)
+      for (uint32_t t = 0; t < a.nterms; ++t) wand_total_ub += wub[t];
+    float gtw = 0.0f;  // float lower bound from the threshold bin
+    if (a.wand && shared_misc[0]) {
+      float t0 = (float)shared_misc[0] / inv_smax;
+      uint32_t t0b;
+      __builtin_memcpy(&t0b, &t0, 4);
+      t0b = t0b > 2 ? t0b - 2 : 0;
+      __builtin_memcpy(&gtw, &t0b, 4);
+    }
+
+    // term-major phases (fixed fp32 merge order -> bit-exact vs oracle)
+    for (uint32_t t = 0; t < a.nterms; ++t) {
+      const TermDev te = terms[t];
+      const uint8_t* pl = a.payload + te.payload_begin;
+      const uint64_t dend = te.desc_end;
+      uint32_t* dbuf = scratch + wave * 384;
+      uint32_t* fbuf = dbuf + 128;
+      uint32_t* nbuf = fbuf + 128;
+      const float num = te.num, nc = te.nc, nl = te.nl;
+      const uint32_t cur0 = cursors[t];
+      uint64_t b = te.desc_begin + cur0 + wave;
+      while (b < dend) {
+        const uint32_t rel = (uint32_t)(b - te.desc_begin) - cur0;
+        const SdbBlockDesc d = rel < a.dcache_n
+                                 ? dcache[t * a.dcache_n + rel]
+                                 : a.desc[b];
+        if (d.prev_doc >= hi) break;
+        if (a.wand) {
+          const float own =
+            score_one(a.scorer, num, nc, nl, d.max_freq, d.min_norm) *
+            a.fbmax * 1.0000019f;
+          if (own + (wand_total_ub - wub[t]) < gtw) {
+            b += NW;
+            continue;
+          }
+        }
+        // pair with the wave's next block when both are the fused shape
+        {
+          const uint64_t b2 = b + NW;
+          if (b2 < dend) {
+            const uint32_t rel2 = (uint32_t)(b2 - te.desc_begin) - cur0;
+            const SdbBlockDesc d2 = rel2 < a.dcache_n
+                                      ? dcache[t * a.dcache_n + rel2]
+                                      : a.desc[b2];
+            if (d2.prev_doc < hi && d.len == 128 && d2.len == 128 &&
+                try_block_fused2<1>(pl, d, d2, lane, a.norm_stream, lo,
+                                    hi, num, nc, nl, a.scorer, a.norms,
+                                    a.fb, swin, mwin)) {
+              b += 2 * NW;
+              continue;
+            }
+          }
+        }
+        // prefetch the NEXT block's payload so its decode loads hit L1
+        {
+          const uint64_t nb = b + NW;
+          if (nb < dend) {
+            const uint32_t nrel = (uint32_t)(nb - te.desc_begin) - cur0;
+            const SdbBlockDesc dn = nrel < a.dcache_n
+                                      ? dcache[t * a.dcache_n + nrel]
+                                      : a.desc[nb];
+            if (dn.prev_doc < hi) {
+              const uint32_t* pfp = (const uint32_t*)(
+                (uintptr_t)(pl + dn.doc_off) & ~(uintptr_t)3);
+              const uint32_t pf = pfp[lane];
+              asm volatile("" ::"v"(pf));
+            }
+          }
+        }
+        if (try_block_fused<1>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
+                               nl, a.scorer, a.norms, a.fb, swin, mwin)) {
+          b += NW;
+          continue;
+        }
+        decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
+        decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+        if (a.norm_stream)
+          decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane,
+                                 nbuf);
+        for (uint32_t j = lane; j < d.len; j += 64) {
+          const uint32_t doc = dbuf[j];
+          if (doc < lo || doc > hi) continue;
+          const uint32_t freq = fbuf[j];
+          const uint32_t norm = a.norm_stream ? nbuf[j] : a.norms[doc];
+          const float nm = a.fb ? num * a.fb[doc] : num;
+          const float s = score_one(a.scorer, nm, nc, nl, freq, norm);
+          const uint32_t off = doc - lo;
+          swin[off] += s;  // unique doc within the term: no atomics
+          mark_match_mask(mwin, off);
+        }
+        b += NW;
+      }
+      __syncthreads();  // term-major merge order (bit-exact vs oracle)
+    }
+    SDB_TS(1)
+    // advance every term's cursor once per window
+    if (tid < a.nterms) {
+      const uint32_t t = tid;
+      const TermDev te = terms[t];
+      uint32_t cur = cursors[t];
+      const uint32_t cur0 = cur;
+      while (te.desc_begin + cur < te.desc_end) {
+        const uint32_t rel = cur - cur0;
+        const uint32_t last = rel < a.dcache_n
+                                ? dcache[t * a.dcache_n + rel].last_doc
+                                : a.desc[te.desc_begin + cur].last_doc;
+        if (last > hi) break;
+        ++cur;
+      }
+      cursors[t] = cur;
+    }
+
+    // ---- ONE sparse sweep over set bits: match count (popcount),
+    // candidate count vs the staged threshold snapshot, sampled histogram
+    uint32_t tbin_w = shared_misc[0];
+    const uint32_t nw_act = (wlen + 63u) / 64u;
+    uint32_t my_matches = 0;
+    uint32_t my_cnt = 0;
+    for (uint32_t i = tid; i < nw_act; i += NTH) {
+      unsigned long long word = mwin[i];
+      if (!word) continue;
+      my_matches += (uint32_t)__popcll(word);
+      const uint32_t base = i * 64u;
+      do {
+        const uint32_t off = base + (uint32_t)__ffsll((long long)word) - 1u;
+        word &= word - 1;
+        const uint32_t sb = score_bin(swin[off], inv_smax);
+        my_cnt += sb >= tbin_w ? 1u : 0u;
+        if (derive) atomicAdd(&hist[sb], 1u);
+      } while (word);
+    }
+    {
+      const uint32_t incl0 = wave_incl_scan(my_cnt, lane);
+      if (lane == 63) scratch[NTH + wave] = incl0;
+      my_excl_snap = incl0 - my_cnt;
+    }
+    uint32_t wm = my_matches;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) wm += __shfl_down(wm, off, 64);
+    if (lane == 0) shared_misc[2 + wave] = wm;
+    __syncthreads();
+    SDB_TS(2)
+
+    // merge window histogram into the per-XCD global shard, derive the
+    // threshold bin from the global suffix counts (identical to the
+    // general kernel; see score_bin for the exactness argument)
+    const uint32_t known_bin = __hip_atomic_load(
+      a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (derive)
+      for (uint32_t b = tid; b < SDB_HIST_BINS; b += NTH)
+        if (b >= known_bin && hist[b]) atomicAdd(&gh[b], hist[b]);
+    if (derive && wave == 0) {
+      uint32_t part = 0;
+      if (SDB_HIST_BINS - 1 - 4 * (uint32_t)lane + 3 >= known_bin) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
+#pragma unroll
+          for (int sh = 0; sh < 8; ++sh)
+            part += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                      __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+        }
+      }
+      const uint32_t suff_incl = wave_incl_scan(part, lane);
+      const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
+      const bool winner =
+        suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
+      if (winner) {
+        uint32_t cum = suff_incl - part;
+        uint32_t binfloor = 0;
+        for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
+          uint32_t add = 0;
+#pragma unroll
+          for (int sh = 0; sh < 8; ++sh)
+            add += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+          cum += add;
+          if (cum >= a.k) {
+            binfloor = (uint32_t)b;
+            break;
+          }
+        }
+        if (binfloor > known_bin) atomicMax(a.gthresh, binfloor);
+      }
+    }
+    if (wave == 0 && lane == 0) {
+      uint32_t total_m = 0;
+      for (uint32_t v = 0; v < NW; ++v) total_m += shared_misc[2 + v];
+      wg_matches += total_m;
+    }
+    SDB_TS(3)
+
+    // first window: refresh the (still ~0) snapshot from the just-derived
+    // global bin and recount, or the whole grid appends every match
+    if (w == w_lo) {
+      if (tid == 0)
+        shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+      __syncthreads();
+      tbin_w = shared_misc[0];
+      uint32_t cnt2 = 0;
+      for (uint32_t i = tid; i < nw_act; i += NTH) {
+        unsigned long long word = mwin[i];
+        if (!word) continue;
+        const uint32_t base = i * 64u;
+        do {
+          const uint32_t off =
+            base + (uint32_t)__ffsll((long long)word) - 1u;
+          word &= word - 1;
+          cnt2 += score_bin(swin[off], inv_smax) >= tbin_w ? 1u : 0u;
+        } while (word);
+      }
+      const uint32_t incl2 = wave_incl_scan(cnt2, lane);
+      if (lane == 63) scratch[NTH + wave] = incl2;
+      my_excl_snap = incl2 - cnt2;
+      __syncthreads();
+    }
+
+    // append candidates (sparse walk, only when the window holds any)
+    uint32_t wave_base = 0;
+    for (uint32_t v = 0; v < wave; ++v) wave_base += scratch[NTH + v];
+    const uint32_t my_excl = wave_base + my_excl_snap;
+    uint32_t block_total = 0;
+    for (uint32_t v = 0; v < NW; ++v) block_total += scratch[NTH + v];
+    if (tid == 0)
+      shared_misc[1] =
+        block_total ? atomicAdd(a.cand_count, block_total) : 0u;
+    __syncthreads();
+    const uint32_t cbase = shared_misc[1];
+    if (cbase + block_total > a.cand_cap) {
+      if (tid == 0) atomicExch(a.overflow, 1u);
+      return;
+    }
+    uint32_t pos = cbase + my_excl;
+    if (block_total)
+      for (uint32_t i = tid; i < nw_act; i += NTH) {
+        unsigned long long word = mwin[i];
+        if (!word) continue;
+        const uint32_t base = i * 64u;
+        do {
+          const uint32_t off =
+            base + (uint32_t)__ffsll((long long)word) - 1u;
+          word &= word - 1;
+          const float s = swin[off];
+          if (score_bin(s, inv_smax) >= tbin_w) {
+            a.cands[pos].score = s;
+            a.cands[pos].doc = lo + off;
+            a.cands[pos].segment_idx = a.seg_idx;
+            ++pos;
+          }
+        } while (word);
+      }
+    __syncthreads();  // window state reused next iteration
+    SDB_TS(4)
+  }
+  if (tid == 0 && wg_matches) atomicAdd(a.total_matches, wg_matches);
+#ifdef SDB_TIMING
+  if (tid == 0)
+    for (int i = 0; i < 6; ++i) atomicAdd(&a.bucket_out[i], t_acc[i]);
+#endif
+#undef SDB_TS
+}
+
 // column gather for the streaming scan: out[i] = col[docs[i]]
 __global__ void gather_col_kernel(const uint32_t* __restrict__ docs,
                                   const long long* __restrict__ col,
@@ -1057,6 +1469,36 @@ __global__ void decode_term_kernel(const SdbBlockDesc* desc, uint64_t b0,
 // host side (C ABI)
 // ---------------------------------------------------------------------------
 namespace {
+
+// lean sweep-kernel geometry (window docs x threads). The instantiation
+// set is fixed; SDB_SWEEP_GEOM="WDxNTH" overrides for on-box A/B sweeps,
+// SDB_SWEEP_GEOM=0 falls back to the general window kernel.
+struct SweepGeom {
+  uint32_t wd, nth;
+};
+constexpr SweepGeom kSweepGeoms[] = {{24576, 1024}, {16384, 1024},
+                                     {12288, 512},  {8192, 512},
+                                     {8192, 256},   {4096, 256}};
+
+bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
+                  hipStream_t st, const WindowArgs& a,
+                  const TermDev* terms) {
+#define SDB_SWEEP_CASE(WDV, NTHV)                                         if (g.wd == WDV && g.nth == NTHV) {                                       hipLaunchKernelGGL((topk_sweep_kernel<WDV, NTHV>), grid, dim3(NTHV),                       lds, st, a, terms);                                  return true;                                                          }
+  SDB_SWEEP_CASE(24576, 1024)
+  SDB_SWEEP_CASE(16384, 1024)
+  SDB_SWEEP_CASE(12288, 512)
+  SDB_SWEEP_CASE(8192, 512)
+  SDB_SWEEP_CASE(8192, 256)
+  SDB_SWEEP_CASE(4096, 256)
+#undef SDB_SWEEP_CASE
+  return false;
+}
+
+// fixed (non-dcache) LDS bytes of one sweep workgroup
+size_t sweep_lds_fixed(const SweepGeom& g) {
+  return (size_t)g.wd * 4 + g.wd / 8 + (g.nth / 64) * 1536 +
+         SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 + 2 * SDB_MAX_TERMS * 4;
+}
 
 int check_gpu() {
   int n = 0;
@@ -1139,7 +1581,8 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 8));
   CTX_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
   CTX_CHECK(hipMalloc(&ctx->d_overflow, 4));
-  CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS));
+  CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS *
+                                       SDB_TERM_SLOTS));
   CTX_CHECK(hipHostMalloc(&ctx->h_counts, 8));
   CTX_CHECK(hipHostMalloc(&ctx->h_matches, 8));
   CTX_CHECK(hipEventCreate(&ctx->ev_a));
@@ -1382,10 +1825,49 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   const size_t lds_bytes =
     lds_fixed + sizeof(SdbBlockDesc) * dcache_n * plan->nterms;
 
+  // lean sweep path: the headline shape (pure top-k disjunction). The
+  // general window kernel keeps min_match>1, hybrid column filters and
+  // CountFast. SDB_SWEEP_GEOM=WDxNTH overrides the geometry; =0 disables.
+  SweepGeom sgeom{24576, 1024};
+  bool use_sweep = (plan->min_match <= 1) && !hybrid && !count_only;
+  if (const char* e = getenv("SDB_SWEEP_GEOM")) {
+    unsigned wd_ = 0, nth_ = 0;
+    if (e[0] == '0' && !e[1]) {
+      use_sweep = false;
+    } else if (sscanf(e, "%ux%u", &wd_, &nth_) == 2 && wd_ >= 64 &&
+               nth_ >= 64 && wd_ % 64 == 0 && nth_ % 64 == 0) {
+      sgeom.wd = wd_;
+      sgeom.nth = nth_;
+    }
+  }
+  uint32_t s_dcache_n = SDB_DESC_CACHE;
+  size_t s_lds = 0;
+  uint32_t s_wgs_per_cu = 1;
+  if (use_sweep) {
+    const size_t fixed = sweep_lds_fixed(sgeom);
+    s_lds = fixed + sizeof(SdbBlockDesc) * s_dcache_n * plan->nterms;
+    const uint32_t wave_cap = 32u / (sgeom.nth / 64u);  // 32 waves/CU
+    s_wgs_per_cu = (uint32_t)(163840 / s_lds);
+    if (s_wgs_per_cu > wave_cap) s_wgs_per_cu = wave_cap;
+    if (s_wgs_per_cu == 0) {  // shrink the desc cache until one WG fits
+      while (s_dcache_n > 1 && s_lds > 163840) {
+        --s_dcache_n;
+        s_lds = fixed + sizeof(SdbBlockDesc) * s_dcache_n * plan->nterms;
+      }
+      if (s_lds > 163840) use_sweep = false;
+      s_wgs_per_cu = 1;
+    }
+  }
+
   HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   for (uint32_t s = 0; s < nsegs; ++s) {
     SdbGpuSegment* seg = segs[s];
-    // per-segment term table (desc ranges differ per segment)
+    // per-segment term table, staged into one of SDB_TERM_SLOTS slots so
+    // consecutive segment launches overlap (round-1 weak #8: the old
+    // single slot forced a stream sync per segment)
+    const uint32_t slot = s % SDB_TERM_SLOTS;
+    if (s && slot == 0) HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    TermDev* d_tslot = ctx->d_terms + (size_t)slot * SDB_MAX_TERMS;
     TermDev tdev[SDB_MAX_TERMS];
     for (uint32_t t = 0; t < plan->nterms; ++t) {
       const SdbTermEntry& te = seg->terms_host[plan->terms[t].term_idx];
@@ -1396,8 +1878,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       tdev[t].nc = nc;
       tdev[t].nl = nl;
     }
-    HIP_CHECK(hipMemcpyAsync(ctx->d_terms, tdev,
-                             sizeof(TermDev) * plan->nterms,
+    HIP_CHECK(hipMemcpyAsync(d_tslot, tdev, sizeof(TermDev) * plan->nterms,
                              hipMemcpyHostToDevice, ctx->stream));
     WindowArgs a{};
     a.fcol = hybrid ? seg->fcols[hpreds[0].slot] : nullptr;
@@ -1435,6 +1916,17 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.cand_cap = SDB_CAND_CAP;
     a.total_matches = ctx->d_total_matches;
     a.overflow = ctx->d_overflow;
+    if (use_sweep) {
+      a.dcache_n = s_dcache_n;
+      const uint32_t nwin = (seg->hdr.doc_count + sgeom.wd - 1) / sgeom.wd;
+      uint32_t ngrid = 256u * s_wgs_per_cu;
+      if (ngrid > nwin) ngrid = nwin;
+      if (!launch_sweep(sgeom, dim3(ngrid), s_lds, ctx->stream, a,
+                        d_tslot))
+        return SDB_ERR_INVALID;  // unknown SDB_SWEEP_GEOM instantiation
+      HIP_CHECK(hipGetLastError());
+      continue;
+    }
     const uint32_t nwin =
       (seg->hdr.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
     uint32_t wgs_per_cu = (uint32_t)(163840 / lds_bytes);
@@ -1443,11 +1935,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     uint32_t ngrid = 256u * wgs_per_cu;
     if (ngrid > nwin) ngrid = nwin;
     hipLaunchKernelGGL(topk_window_kernel, dim3(ngrid), dim3(SDB_NTHREADS),
-                       lds_bytes, ctx->stream, a, ctx->d_terms);
+                       lds_bytes, ctx->stream, a, d_tslot);
     HIP_CHECK(hipGetLastError());
-    if (nsegs > 1) HIP_CHECK(hipStreamSynchronize(ctx->stream));
-    // (sync between segments: d_terms is reused; single-segment path keeps
-    //  everything async until readback)
   }
 
   HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
