@@ -193,6 +193,7 @@ typedef enum SdbPredOp {
   SDB_PRED_LT = 1,  /* col < v */
   SDB_PRED_GE = 2,  /* col >= v */
   SDB_PRED_BETWEEN = 3, /* lo <= col <= hi */
+  SDB_PRED_EQ = 4,  /* col == v (table_filter_iterator.hpp typed compares) */
 } SdbPredOp;
 
 int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
