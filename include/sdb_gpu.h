@@ -286,6 +286,27 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
                      SdbAggResult* out /* ngroups*naggs */,
                      uint64_t* rows_passed);
 
+/* General hash aggregate: ARBITRARY i64 group keys (sparse, any range) —
+ * the north_star "hash-group-by with LDS-staged open-addressed buckets",
+ * replacing the DuckDB PhysicalHashAggregate consumer
+ * (duckdb_search_full_scan.cpp:2141-2171) for keys the dense perfect-hash
+ * kernel (sdb_gpu_scan_agg) cannot take. Per-workgroup open-addressed LDS
+ * table flushed once into a global open-addressed table; same predicate
+ * pushdown and zonemap skips.
+ *
+ * max_groups bounds DISTINCT keys (<= 4M); more distinct keys (or a
+ * pathological probe chain) returns SDB_ERR_OOM. Results: rows 0..
+ * *ngroups_out sorted by key ascending; keys_out/out must hold max_groups
+ * rows. Allocates/frees its device workspace per call (the dense kernel
+ * remains the allocation-free hot path). */
+int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                          uint32_t group_col, uint64_t max_groups,
+                          const SdbPredSpec* preds, uint32_t npreds,
+                          const SdbAggSpec* aggs, uint32_t naggs,
+                          int64_t* keys_out /* max_groups */,
+                          SdbAggResult* out /* max_groups*naggs */,
+                          uint64_t* ngroups_out, uint64_t* rows_passed);
+
 #ifdef __cplusplus
 }
 #endif

@@ -312,6 +312,7 @@ def gpu():
             "sdb_gpu_segment_load", "sdb_gpu_segment_free",
             "sdb_gpu_execute_topk", "sdb_gpu_decode_term",
             "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
+            "sdb_gpu_scan_agg_hash",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
             "sdb_gpu_execute_match_docs", "sdb_gpu_execute_count",
         ):
@@ -668,3 +669,37 @@ class GpuContext:
         """COUNT(*) + SUM(col1) grouped by col0, no predicate."""
         return self.scan_agg(tab, 0, ngroups, [],
                              [(0, 0), (1, 1)])
+
+    def scan_agg_hash(self, tab, group_col, max_groups, preds, aggs):
+        """General hash aggregate (arbitrary i64 keys). Returns
+        (keys[n], i64 results [n, naggs], f64 results [n, naggs],
+        rows_passed), rows sorted by key ascending."""
+        import numpy as np
+
+        np_ = len(preds)
+        pa = (self._PredSpec * max(np_, 1))()
+        for i, (col, op, lo, hi) in enumerate(preds):
+            if isinstance(lo, float) or isinstance(hi, float):
+                pa[i] = self._PredSpec(col, op, 0, 0, lo, hi)
+            else:
+                pa[i] = self._PredSpec(col, op, lo, hi, 0, 0)
+        na = len(aggs)
+        aa = (self._AggSpec * na)(*[self._AggSpec(c, o) for c, o in aggs])
+        keys_out = np.zeros(max_groups, dtype=np.int64)
+        out = (self._AggResult * (max_groups * na))()
+        ng = C.c_uint64(0)
+        passed = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_scan_agg_hash(
+            self._ctx, tab, C.c_uint32(group_col), C.c_uint64(max_groups),
+            pa, C.c_uint32(np_), aa, C.c_uint32(na),
+            keys_out.ctypes.data_as(C.POINTER(C.c_int64)), out,
+            C.byref(ng), C.byref(passed))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_scan_agg_hash rc={rc}")
+        n = ng.value
+        i64 = np.array([[out[g * na + q].i64 for q in range(na)]
+                        for g in range(n)], dtype=np.int64)
+        f64 = np.array([[out[g * na + q].f64 for q in range(na)]
+                        for g in range(n)], dtype=np.float64)
+        return keys_out[:n], i64.reshape(n, na), f64.reshape(n, na), \
+            passed.value

@@ -1180,7 +1180,13 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       __builtin_memcpy(&gtw, &t0b, 4);
     }
 
-    // term-major phases (fixed fp32 merge order -> bit-exact vs oracle)
+    // term-major phases (fixed fp32 merge order -> bit-exact vs oracle).
+    // Cross-phase prefetch: before working term t, issue a non-blocking
+    // touch of term t+1's first block payload for this wave; the value is
+    // consumed at the END of the phase, so the next phase's cold HBM/L3
+    // miss overlaps this phase's work and barrier (phases measured
+    // latency-bound: 62% of the kernel, SDB_TIMING r2).
+    uint32_t pf_acc = 0;
     for (uint32_t t = 0; t < a.nterms; ++t) {
       const TermDev te = terms[t];
       const uint8_t* pl = a.payload + te.payload_begin;
@@ -1191,6 +1197,23 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       const float num = te.num, nc = te.nc, nl = te.nl;
       const uint32_t cur0 = cursors[t];
       uint64_t b = te.desc_begin + cur0 + wave;
+      uint32_t pf_next = 0;
+      if (t + 1 < a.nterms) {
+        const TermDev tn = terms[t + 1];
+        const uint32_t ncur = cursors[t + 1];
+        const uint64_t nb = tn.desc_begin + ncur + wave;
+        if (nb < tn.desc_end) {
+          const SdbBlockDesc dn = wave < a.dcache_n
+                                    ? dcache[(t + 1) * a.dcache_n + wave]
+                                    : a.desc[nb];
+          if (dn.prev_doc < hi) {
+            const uint32_t* pfp = (const uint32_t*)(
+              (uintptr_t)(a.payload + tn.payload_begin + dn.doc_off) &
+              ~(uintptr_t)3);
+            pf_next = pfp[lane] + pfp[lane + 64];  // 512 B line touch
+          }
+        }
+      }
       while (b < dend) {
         const uint32_t rel = (uint32_t)(b - te.desc_begin) - cur0;
         const SdbBlockDesc d = rel < a.dcache_n
@@ -1262,8 +1285,13 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         }
         b += NW;
       }
+      pf_acc ^= pf_next;  // keep the prefetch loads alive to phase end
+#ifdef SDB_TIMING
+      if (t == 0) SDB_TS(5)  // term-0 phase alone (imbalance diagnosis)
+#endif
       __syncthreads();  // term-major merge order (bit-exact vs oracle)
     }
+    asm volatile("" ::"v"(pf_acc));  // waits land here, once per window
     SDB_TS(1)
     // advance every term's cursor once per window
     if (tid < a.nterms) {
@@ -1281,10 +1309,26 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
       cursors[t] = cur;
     }
+    __syncthreads();  // cursors visible to the next-window prefetch below
 
     // ---- ONE sparse sweep over set bits: match count (popcount),
     // candidate count vs the staged threshold snapshot, sampled histogram
     uint32_t tbin_w = shared_misc[0];
+    // next-window prefetch: touch each term's next block payload for this
+    // wave under the sweep/tau/append tail (consumed at window end)
+    uint32_t pf_w = 0;
+    if (w + 1 < w_hi && wave < a.nterms) {
+      const uint32_t t = wave;
+      const TermDev te = terms[t];
+      const uint32_t ncur = cursors[t];  // post-advance (barrier above)
+      const uint64_t nb = te.desc_begin + ncur;
+      if (nb < te.desc_end) {
+        const uint32_t* pfp = (const uint32_t*)(
+          (uintptr_t)(a.payload + te.payload_begin + a.desc[nb].doc_off) &
+          ~(uintptr_t)3);
+        pf_w = pfp[lane] + pfp[lane + 64];
+      }
+    }
     const uint32_t nw_act = (wlen + 63u) / 64u;
     uint32_t my_matches = 0;
     uint32_t my_cnt = 0;
@@ -1424,6 +1468,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
           }
         } while (word);
       }
+    asm volatile("" ::"v"(pf_w));  // next-window prefetch lands
     __syncthreads();  // window state reused next iteration
     SDB_TS(4)
   }
@@ -1478,7 +1523,8 @@ struct SweepGeom {
 };
 constexpr SweepGeom kSweepGeoms[] = {{24576, 1024}, {16384, 1024},
                                      {12288, 512},  {8192, 512},
-                                     {8192, 256},   {4096, 256}};
+                                     {8192, 256},   {4096, 256},
+                                     {24576, 512},  {32768, 512}};
 
 bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
                   hipStream_t st, const WindowArgs& a,
@@ -1490,6 +1536,8 @@ bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
   SDB_SWEEP_CASE(8192, 512)
   SDB_SWEEP_CASE(8192, 256)
   SDB_SWEEP_CASE(4096, 256)
+  SDB_SWEEP_CASE(24576, 512)
+  SDB_SWEEP_CASE(32768, 512)
 #undef SDB_SWEEP_CASE
   return false;
 }
@@ -2011,9 +2059,9 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     hipMemcpy(tdbg, ctx->d_buckets, 48, hipMemcpyDeviceToHost);
     fprintf(stderr,
             "[timing cyc/WG avg] zero+stage=%llu phases=%llu hist=%llu "
-            "tau=%llu append=%llu\n",
+            "tau=%llu append=%llu term0=%llu\n",
             tdbg[0] / 256, tdbg[1] / 256, tdbg[2] / 256, tdbg[3] / 256,
-            tdbg[4] / 256);
+            tdbg[4] / 256, tdbg[5] / 256);
   }
 #endif
   if (hybrid && bucket_count && bucket_sum) {

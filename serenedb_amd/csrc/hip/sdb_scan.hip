@@ -744,6 +744,223 @@ void scan_agg_staged_kernel(ScanArgs a) {
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// General hash aggregate (round-2): arbitrary i64 group keys.
+// north_star: "hash-group-by with LDS-staged open-addressed buckets" — the
+// consumer the reference hands to DuckDB's PhysicalHashAggregate
+// (duckdb_search_full_scan.cpp:2141-2171) for keys the dense perfect-hash
+// kernel cannot take. Per-workgroup open-addressed LDS table (linear
+// probing, 64-bit key CAS), flushed once into a global open-addressed
+// table; rows that miss the LDS probe bound go straight to the global
+// table. Exact integer aggregates; SUM(f32) in f64 (atomic order
+// nondeterministic, as the dense kernel).
+// ---------------------------------------------------------------------------
+
+#define SDB_HKEY_EMPTY 0xFFFFFFFFFFFFFFFFull /* key -1 uses neg_acc */
+
+__device__ __forceinline__ uint64_t mix64(uint64_t x) {
+  // splitmix64 finalizer
+  x ^= x >> 30;
+  x *= 0xbf58476d1ce4e5b9ull;
+  x ^= x >> 27;
+  x *= 0x94d049bb133111ebull;
+  x ^= x >> 31;
+  return x;
+}
+
+struct HashAggArgs {
+  ColRef keys;
+  uint64_t rows;
+  uint32_t group_rows;
+  uint32_t naggs;
+  uint32_t npreds;
+  ColRef pred_col[SCAN_MAX_PREDS];
+  int pred_op[SCAN_MAX_PREDS];
+  int pred_isf32[SCAN_MAX_PREDS];
+  int64_t pred_lo[SCAN_MAX_PREDS];
+  int64_t pred_hi[SCAN_MAX_PREDS];
+  float pred_flo[SCAN_MAX_PREDS], pred_fhi[SCAN_MAX_PREDS];
+  ColRef agg_col[SCAN_MAX_AGGS];
+  int agg_op[SCAN_MAX_AGGS];
+  // global open-addressed table (pow2 capacity, keys init SDB_HKEY_EMPTY)
+  unsigned long long* gkeys;
+  unsigned long long* gacc;  // [cap * naggs]
+  uint32_t gcap_mask;
+  uint32_t max_groups;
+  uint32_t* ginserts;   // distinct-key counter (contract check)
+  uint32_t* goverflow;  // probe bound exceeded
+  unsigned long long* neg_acc;  // [naggs+1]: key == -1 accumulators + flag
+  unsigned long long* rows_passed;
+  uint32_t lds_slots;   // pow2 (0 = LDS staging disabled)
+};
+
+// probe/insert into the global table; returns slot or -1 (overflow).
+__device__ __forceinline__ int64_t hash_global_upsert(
+  const HashAggArgs& a, uint64_t k) {
+  uint32_t h = (uint32_t)mix64(k) & a.gcap_mask;
+  for (uint32_t i = 0; i <= a.gcap_mask; ++i) {
+    const unsigned long long cur = __hip_atomic_load(
+      &a.gkeys[h], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == k) return (int64_t)h;
+    if (cur == SDB_HKEY_EMPTY) {
+      const unsigned long long prev =
+        atomicCAS(&a.gkeys[h], SDB_HKEY_EMPTY, k);
+      if (prev == SDB_HKEY_EMPTY) {
+        atomicAdd(a.ginserts, 1u);
+        return (int64_t)h;
+      }
+      if (prev == k) return (int64_t)h;
+    }
+    h = (h + 1) & a.gcap_mask;
+  }
+  atomicExch(a.goverflow, 1u);
+  return -1;
+}
+
+__device__ __forceinline__ void hash_acc_add(const HashAggArgs& a,
+                                             unsigned long long* acc,
+                                             uint32_t q, uint32_t rg,
+                                             uint64_t r0, uint64_t r,
+                                             bool global_scope) {
+  switch (a.agg_op[q]) {
+    case SDB_AGG_COUNT:
+      atomicAdd(&acc[q], 1ull);
+      break;
+    case SDB_AGG_SUM_I64:
+      atomicAdd(&acc[q],
+                (unsigned long long)col_read(a.agg_col[q], rg, r0, r));
+      break;
+    case SDB_AGG_SUM_F64:
+      atomicAdd((double*)&acc[q],
+                (double)((const float*)a.agg_col[q].data)[r]);
+      break;
+  }
+  (void)global_scope;
+}
+
+__launch_bounds__(SCAN_NTHREADS) __global__
+void scan_agg_hash_kernel(HashAggArgs a) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned long long* lkey = (unsigned long long*)smem;     // lds_slots
+  unsigned long long* lacc = lkey + a.lds_slots;            // slots*naggs
+  const uint32_t S_mask = a.lds_slots - 1;
+  for (uint32_t i = threadIdx.x; i < a.lds_slots * (1 + a.naggs);
+       i += SCAN_NTHREADS)
+    lkey[i] = i < a.lds_slots ? SDB_HKEY_EMPTY : 0ull;
+  __syncthreads();
+
+  const uint32_t n_rowgroups =
+    (uint32_t)((a.rows + a.group_rows - 1) / a.group_rows);
+  unsigned long long my_passed = 0;
+
+  for (uint32_t rg = blockIdx.x; rg < n_rowgroups; rg += gridDim.x) {
+    const uint64_t r0 = (uint64_t)rg * a.group_rows;
+    const uint64_t r1 = min(a.rows, r0 + a.group_rows);
+    bool dead = false;
+    for (uint32_t p = 0; p < a.npreds; ++p) {
+      if (a.pred_col[p].desc && !a.pred_isf32[p]) {
+        const SdbColGroupDescDev& d = a.pred_col[p].desc[rg];
+        switch (a.pred_op[p]) {
+          case SDB_PRED_LT: dead |= d.vmin >= a.pred_lo[p]; break;
+          case SDB_PRED_GE: dead |= d.vmax < a.pred_lo[p]; break;
+          case SDB_PRED_BETWEEN:
+            dead |= (d.vmax < a.pred_lo[p]) | (d.vmin > a.pred_hi[p]);
+            break;
+          case SDB_PRED_EQ:
+            dead |= (d.vmax < a.pred_lo[p]) | (d.vmin > a.pred_lo[p]);
+            break;
+          default: break;
+        }
+      }
+    }
+    if (dead) continue;
+
+    for (uint64_t r = r0 + threadIdx.x; r < r1; r += SCAN_NTHREADS) {
+      bool ok = true;
+      for (uint32_t p = 0; p < a.npreds; ++p) {
+        if (a.pred_isf32[p])
+          ok &= pred_eval_f(a.pred_op[p],
+                            ((const float*)a.pred_col[p].data)[r],
+                            a.pred_flo[p], a.pred_fhi[p]);
+        else
+          ok &= pred_eval_i(a.pred_op[p],
+                            col_read(a.pred_col[p], rg, r0, r),
+                            a.pred_lo[p], a.pred_hi[p]);
+      }
+      if (!ok) continue;
+      ++my_passed;
+      const uint64_t k = (uint64_t)col_read(a.keys, rg, r0, r);
+      if (k == SDB_HKEY_EMPTY) {  // actual key -1: dedicated accumulators
+        if (!__hip_atomic_load(&a.neg_acc[a.naggs], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT))
+          atomicExch(&a.neg_acc[a.naggs], 1ull);
+        for (uint32_t q = 0; q < a.naggs; ++q)
+          hash_acc_add(a, a.neg_acc, q, rg, r0, r, true);
+        continue;
+      }
+      // LDS open-addressed probe (bounded), then global fallback
+      int slot = -1;
+      if (a.lds_slots) {
+        uint32_t h = (uint32_t)mix64(k) & S_mask;
+        for (uint32_t i = 0; i < 32; ++i) {
+          const unsigned long long cur = lkey[h];
+          if (cur == k) {
+            slot = (int)h;
+            break;
+          }
+          if (cur == SDB_HKEY_EMPTY) {
+            const unsigned long long prev =
+              atomicCAS(&lkey[h], SDB_HKEY_EMPTY, k);
+            if (prev == SDB_HKEY_EMPTY || prev == k) {
+              slot = (int)h;
+              break;
+            }
+          }
+          h = (h + 1) & S_mask;
+        }
+      }
+      if (slot >= 0) {
+        hash_acc_add(a, &lacc[(uint32_t)slot * a.naggs], 0, rg, r0, r,
+                     false);
+        for (uint32_t q = 1; q < a.naggs; ++q)
+          hash_acc_add(a, &lacc[(uint32_t)slot * a.naggs], q, rg, r0, r,
+                       false);
+      } else {
+        const int64_t g = hash_global_upsert(a, k);
+        if (g < 0) return;  // table overflow: host reports SDB_ERR_OOM
+        for (uint32_t q = 0; q < a.naggs; ++q)
+          hash_acc_add(a, &a.gacc[(uint64_t)g * a.naggs], q, rg, r0, r,
+                       true);
+      }
+    }
+  }
+
+  unsigned long long wp = my_passed;
+#pragma unroll
+  for (int off = 32; off; off >>= 1) wp += __shfl_down(wp, off, 64);
+  if ((threadIdx.x & 63) == 0 && wp) atomicAdd(a.rows_passed, wp);
+  __syncthreads();
+  // flush the LDS table into the global table
+  for (uint32_t i = threadIdx.x; i < a.lds_slots; i += SCAN_NTHREADS) {
+    const unsigned long long k = lkey[i];
+    if (k == SDB_HKEY_EMPTY) continue;
+    const int64_t g = hash_global_upsert(a, k);
+    if (g < 0) return;
+    for (uint32_t q = 0; q < a.naggs; ++q) {
+      const unsigned long long v = lacc[i * a.naggs + q];
+      if (a.agg_op[q] == SDB_AGG_SUM_F64) {
+        double d;
+        __builtin_memcpy(&d, &v, 8);
+        if (d != 0.0)
+          atomicAdd((double*)&a.gacc[(uint64_t)g * a.naggs + q], d);
+      } else if (v) {
+        atomicAdd(&a.gacc[(uint64_t)g * a.naggs + q], v);
+      }
+    }
+  }
+}
+
 extern "C" {
 
 static void table_free_partial(SdbGpuTable* tab) {
@@ -1046,6 +1263,160 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
     }
   }
   *rows_passed = h_passed;
+  return SDB_OK;
+}
+
+
+int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                          uint32_t group_col, uint64_t max_groups,
+                          const SdbPredSpec* preds, uint32_t npreds,
+                          const SdbAggSpec* aggs, uint32_t naggs,
+                          int64_t* keys_out, SdbAggResult* out,
+                          uint64_t* ngroups_out, uint64_t* rows_passed) {
+  if (!ctx || !tab || !keys_out || !out || !ngroups_out || !rows_passed ||
+      group_col >= tab->ncols || max_groups == 0 ||
+      max_groups > (1u << 22) || naggs == 0 || naggs > SCAN_MAX_AGGS ||
+      npreds > SCAN_MAX_PREDS)
+    return SDB_ERR_INVALID;
+  if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
+
+  hipStream_t stream = ctx->stream;
+  HashAggArgs a{};
+  a.keys = tab->refs[group_col];
+  a.rows = tab->rows;
+  a.group_rows = tab->group_rows;
+  a.naggs = naggs;
+  a.npreds = npreds;
+  for (uint32_t p = 0; p < npreds; ++p) {
+    if (preds[p].col >= tab->ncols) return SDB_ERR_INVALID;
+    if (preds[p].op < SDB_PRED_LT || preds[p].op > SDB_PRED_EQ)
+      return SDB_ERR_INVALID;
+    a.pred_col[p] = tab->refs[preds[p].col];
+    a.pred_op[p] = preds[p].op;
+    a.pred_isf32[p] = tab->types[preds[p].col] == SDB_COL_F32 ? 1 : 0;
+    a.pred_lo[p] = preds[p].ilo;
+    a.pred_hi[p] = preds[p].ihi;
+    a.pred_flo[p] = preds[p].flo;
+    a.pred_fhi[p] = preds[p].fhi;
+  }
+  for (uint32_t q = 0; q < naggs; ++q) {
+    a.agg_op[q] = aggs[q].op;
+    a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT ? ColRef{nullptr, nullptr}
+                                               : tab->refs[aggs[q].col];
+    if (aggs[q].op == SDB_AGG_SUM_I64 &&
+        tab->types[aggs[q].col] == SDB_COL_F32)
+      return SDB_ERR_INVALID;
+    if (aggs[q].op == SDB_AGG_SUM_F64 &&
+        tab->types[aggs[q].col] != SDB_COL_F32)
+      return SDB_ERR_INVALID;
+  }
+  // global table: pow2 capacity >= 2*max_groups (load factor <= 0.5)
+  uint64_t cap = 64;
+  while (cap < 2 * max_groups) cap <<= 1;
+  a.gcap_mask = (uint32_t)(cap - 1);
+  a.max_groups = (uint32_t)max_groups;
+  // LDS: largest pow2 slot count whose (key + naggs accumulators) fit
+  uint32_t slots = 1u << 14;
+  while (slots && (uint64_t)slots * 8 * (1 + naggs) > 150 * 1024)
+    slots >>= 1;
+  a.lds_slots = slots;
+  const size_t lds = (size_t)slots * 8 * (1 + naggs);
+
+  unsigned long long* d_keys = nullptr;
+  unsigned long long* d_acc = nullptr;
+  unsigned long long* d_misc = nullptr;  // [0]=inserts|overflow, then
+                                         // neg_acc[naggs+1], rows_passed
+  const size_t misc_words = 1 + (naggs + 1) + 1;
+#define HASH_CHECK(x)                                        \
+  do {                                                       \
+    hipError_t _e = (x);                                     \
+    if (_e != hipSuccess) {                                  \
+      if (d_keys) (void)hipFree(d_keys);                     \
+      if (d_acc) (void)hipFree(d_acc);                       \
+      if (d_misc) (void)hipFree(d_misc);                     \
+      return _e == hipErrorNoDevice ? SDB_ERR_NO_GPU         \
+             : _e == hipErrorOutOfMemory ? SDB_ERR_OOM       \
+                                         : SDB_ERR_HIP;      \
+    }                                                        \
+  } while (0)
+  HASH_CHECK(hipMalloc(&d_keys, cap * 8));
+  HASH_CHECK(hipMalloc(&d_acc, cap * 8 * naggs));
+  HASH_CHECK(hipMalloc(&d_misc, misc_words * 8));
+  HASH_CHECK(hipMemsetAsync(d_keys, 0xFF, cap * 8, stream));
+  HASH_CHECK(hipMemsetAsync(d_acc, 0, cap * 8 * naggs, stream));
+  HASH_CHECK(hipMemsetAsync(d_misc, 0, misc_words * 8, stream));
+  a.gkeys = d_keys;
+  a.gacc = d_acc;
+  a.ginserts = (uint32_t*)d_misc;
+  a.goverflow = (uint32_t*)d_misc + 1;
+  a.neg_acc = d_misc + 1;
+  a.rows_passed = d_misc + 1 + (naggs + 1);
+
+  uint32_t nblocks =
+    (uint32_t)((tab->rows + a.group_rows - 1) / a.group_rows);
+  if (nblocks > SCAN_MAXB) nblocks = SCAN_MAXB;
+  if (nblocks < 1) nblocks = 1;
+  hipLaunchKernelGGL(scan_agg_hash_kernel, dim3(nblocks),
+                     dim3(SCAN_NTHREADS), lds, stream, a);
+  HASH_CHECK(hipGetLastError());
+
+  std::vector<unsigned long long> h_keys(cap);
+  std::vector<unsigned long long> h_acc(cap * naggs);
+  std::vector<unsigned long long> h_misc(misc_words);
+  HASH_CHECK(hipMemcpyAsync(h_keys.data(), d_keys, cap * 8,
+                            hipMemcpyDeviceToHost, stream));
+  HASH_CHECK(hipMemcpyAsync(h_acc.data(), d_acc, cap * 8 * naggs,
+                            hipMemcpyDeviceToHost, stream));
+  HASH_CHECK(hipMemcpyAsync(h_misc.data(), d_misc, misc_words * 8,
+                            hipMemcpyDeviceToHost, stream));
+  HASH_CHECK(hipStreamSynchronize(stream));
+#undef HASH_CHECK
+  (void)hipFree(d_keys);
+  (void)hipFree(d_acc);
+  (void)hipFree(d_misc);
+
+  const uint32_t inserts = (uint32_t)(h_misc[0] & 0xFFFFFFFFu);
+  const uint32_t overflow = (uint32_t)(h_misc[0] >> 32);
+  const bool neg_present = h_misc[1 + naggs] != 0;
+  const uint64_t distinct = (uint64_t)inserts + (neg_present ? 1 : 0);
+  if (overflow || distinct > max_groups) return SDB_ERR_OOM;
+
+  // compact + sort by key ascending (deterministic result order)
+  std::vector<uint32_t> live;
+  live.reserve(inserts);
+  for (uint64_t i = 0; i < cap; ++i)
+    if (h_keys[i] != SDB_HKEY_EMPTY) live.push_back((uint32_t)i);
+  std::sort(live.begin(), live.end(), [&](uint32_t x, uint32_t y) {
+    return (int64_t)h_keys[x] < (int64_t)h_keys[y];
+  });
+  uint64_t n = 0;
+  auto emit = [&](int64_t key, const unsigned long long* acc) {
+    keys_out[n] = key;
+    for (uint32_t q = 0; q < naggs; ++q) {
+      SdbAggResult* r = &out[n * naggs + q];
+      if (aggs[q].op == SDB_AGG_SUM_F64) {
+        double v;
+        std::memcpy(&v, &acc[q], 8);
+        r->f64 = v;
+        r->i64 = 0;
+      } else {
+        r->i64 = (int64_t)acc[q];
+        r->f64 = 0;
+      }
+    }
+    ++n;
+  };
+  // negative keys sort before -1? No: -1 is the LARGEST negative value,
+  // so emit live keys < -1 first, then -1, then the rest.
+  size_t li = 0;
+  while (li < live.size() && (int64_t)h_keys[live[li]] < -1)
+    emit((int64_t)h_keys[live[li]], &h_acc[(uint64_t)live[li] * naggs]),
+      ++li;
+  if (neg_present) emit(-1, &h_misc[1]);
+  for (; li < live.size(); ++li)
+    emit((int64_t)h_keys[live[li]], &h_acc[(uint64_t)live[li] * naggs]);
+  *ngroups_out = n;
+  *rows_passed = h_misc[1 + naggs + 1];
   return SDB_OK;
 }
 

@@ -937,3 +937,88 @@ def test_dense_group_key_range_rejected(ctx):
         out = ctx.scan_agg_count_sum(tab, ngroups=65)
         assert out is not None
         ctx.free_table(tab)
+
+
+def _np_hash_agg(keys, v1, v2, sel):
+    """independent numpy expected result for the hash aggregate"""
+    ks = keys[sel]
+    uk, inv = np.unique(ks, return_inverse=True)
+    cnt = np.bincount(inv, minlength=len(uk))
+    s1 = np.bincount(inv, weights=v1[sel].astype(np.float64),
+                     minlength=len(uk)).astype(np.int64)
+    # exact integer sums: use object-free path via add.at on int64
+    s1 = np.zeros(len(uk), dtype=np.int64)
+    np.add.at(s1, inv, v1[sel])
+    s2 = np.zeros(len(uk), dtype=np.float64)
+    np.add.at(s2, inv, v2[sel].astype(np.float64))
+    return uk, cnt, s1, s2
+
+
+def test_hash_agg_sparse_keys(ctx):
+    """General hash group-by (north_star's LDS-staged open-addressed
+    buckets; VERDICT #1 missing item): random sparse 64-bit keys, ~100k
+    distinct groups, predicate pushdown — vs an independent numpy
+    aggregation. Raw and FoR key codecs."""
+    rows = 3_000_000
+    rng = np.random.default_rng(99)
+    base = rng.integers(-(1 << 62), 1 << 62, 100_000).astype(np.int64)
+    keys = base[rng.integers(0, len(base), rows)]
+    keys[::997] = -1  # the sentinel-adjacent key must work too
+    v1 = rng.integers(-(1 << 30), 1 << 30, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    c = int((1 << 30) * -0.8)
+    sel = v1 >= c
+    uk, cnt, s1, s2 = _np_hash_agg(keys, v1, v2, sel)
+
+    for enc in ("raw", "for"):
+        tab = ctx.load_table([keys, v1, v2], [enc, enc, "raw"])
+        gkeys, i64, f64, passed = ctx.scan_agg_hash(
+            tab, 0, 200_000, [(1, 2, c, 0)], [(0, 0), (1, 1), (2, 2)])
+        assert passed == int(sel.sum())
+        np.testing.assert_array_equal(gkeys, uk)
+        np.testing.assert_array_equal(i64[:, 0], cnt)
+        np.testing.assert_array_equal(i64[:, 1], s1)
+        np.testing.assert_allclose(f64[:, 2], s2, rtol=1e-7)
+        # distinct keys above max_groups -> loud SDB_ERR_OOM
+        import pytest as _pytest
+        with _pytest.raises(RuntimeError):
+            ctx.scan_agg_hash(tab, 0, 1000, [], [(0, 0)])
+        ctx.free_table(tab)
+
+
+def test_scan_eq_and_f32_predicates(ctx):
+    """EQ + f32 predicate columns (VERDICT #7 / table_filter_iterator
+    typed compares) across the dense and hash aggregate kernels."""
+    rows = 1_000_000
+    ngroups = 256
+    rng = np.random.default_rng(31)
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(0, 1000, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    v2[::101] = np.nan  # NaN fails every compare
+
+    tab = ctx.load_table([keys, v1, v2])
+    # EQ on i64
+    sel = v1 == 123
+    i64, f64, passed = ctx.scan_agg(tab, 0, ngroups,
+                                    [(1, 4, 123, 0)], [(0, 0), (1, 1)])
+    assert passed == int(sel.sum())
+    exp_cnt = np.bincount(keys[sel], minlength=ngroups)
+    np.testing.assert_array_equal(i64[:, 0], exp_cnt)
+    # f32 BETWEEN (NaN rows excluded)
+    fsel = (v2 >= np.float32(-0.5)) & (v2 <= np.float32(0.5))
+    i64b, f64b, passedb = ctx.scan_agg(
+        tab, 0, ngroups, [(2, 3, -0.5, 0.5)], [(0, 0), (2, 2)])
+    assert passedb == int(fsel.sum())
+    np.testing.assert_array_equal(
+        i64b[:, 0], np.bincount(keys[fsel], minlength=ngroups))
+    s2 = np.zeros(ngroups, dtype=np.float64)
+    np.add.at(s2, keys[fsel], v2[fsel].astype(np.float64))
+    np.testing.assert_allclose(f64b[:, 1], s2, rtol=1e-7)
+    # same through the hash kernel
+    gk, hi64, hf64, hp = ctx.scan_agg_hash(
+        tab, 0, ngroups + 8, [(2, 3, -0.5, 0.5)], [(0, 0)])
+    assert hp == passedb
+    live = np.nonzero(np.bincount(keys[fsel], minlength=ngroups))[0]
+    np.testing.assert_array_equal(gk, live)
+    ctx.free_table(tab)
