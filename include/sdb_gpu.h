@@ -204,6 +204,15 @@ int sdb_gpu_segment_attach_column_slot(SdbGpuCtx* ctx, SdbGpuSegment* seg,
  * consumed when SdbQueryPlan.filter_boost is set */
 int sdb_gpu_segment_attach_boost(SdbGpuCtx* ctx, SdbGpuSegment* seg,
                                  const float* boost);
+/* live-document bitmap — the deleted-docs mask the reference wraps around
+ * every scan when a segment has deletes (seg.mask(it),
+ * duckdb_search_full_scan.cpp:1898,2002,2226; Masked count mode :2475).
+ * bit d of mask[d>>6] set = doc d live (docs 1..doc_count; bit 0 of word
+ * 0 unused); (doc_count+64)/64 words. Once attached, every execute on the
+ * segment sees only live docs (hits, counts, WAND thresholds, streaming
+ * emission, hybrid aggregates). NULL detaches. */
+int sdb_gpu_segment_attach_livemask(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                    const uint64_t* mask);
 
 typedef struct SdbHybridPred {
   uint32_t slot;  /* attached column slot */

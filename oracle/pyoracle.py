@@ -148,17 +148,26 @@ SCORERS = {"bm25": 0, "tfidf": 1, "tfidf_norm": 2}
 
 
 def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
-                 global_stats=None, scorer="bm25", filter_boost=None):
+                 global_stats=None, scorer="bm25", filter_boost=None,
+                 live_mask=None):
     """EXACT top-k (parity oracle). blobs: list of segment blob bytes.
     filter_boost: optional per-doc f32 multiplier (doc_count+1, 1-based) —
-    the HasFilterBoost scorer variants; single-segment only."""
+    the HasFilterBoost scorer variants; single-segment only. live_mask:
+    optional uint64 bitmap (bit d of word d>>6 = doc d live; seg.mask
+    deleted-docs analogue); single-segment only."""
     lib().o_set_scorer(C.c_uint32(SCORERS[scorer]))
     fbkeep = None
+    lmkeep = None
     if filter_boost is not None:
         assert len(blobs) == 1, "filter_boost: single segment only"
         fbkeep = np.ascontiguousarray(filter_boost, dtype=np.float32)
         lib().o_set_filter_boost(
             fbkeep.ctypes.data_as(C.POINTER(C.c_float)))
+    if live_mask is not None:
+        assert len(blobs) == 1, "live_mask: single segment only"
+        lmkeep = np.ascontiguousarray(live_mask, dtype=np.uint64)
+        lib().o_set_live_mask(
+            lmkeep.ctypes.data_as(C.POINTER(C.c_uint64)))
     arr, keep = _mkblobs(blobs)
     ti = _u32arr(term_idx)
     bo = np.ascontiguousarray(boosts, dtype=np.float32)
@@ -178,8 +187,10 @@ def execute_topk(blobs, term_idx, boosts, k, min_match=1, k1=1.2, b=0.75,
         hits, C.byref(out_count), C.byref(total))
     if filter_boost is not None:
         lib().o_set_filter_boost(None)
+    if live_mask is not None:
+        lib().o_set_live_mask(None)
     assert rc == 0, rc
-    del keep, keep2, fbkeep
+    del keep, keep2, fbkeep, lmkeep
     return _hits_to_np(hits, out_count.value), total.value
 
 

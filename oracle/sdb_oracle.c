@@ -535,6 +535,12 @@ static void o_cand_push(OCandVec* cv, float score, uint32_t doc,
 static const float* o_fboost = NULL;
 void o_set_filter_boost(const float* fb) { o_fboost = fb; }
 
+/* optional live-document bitmap (deleted-docs mask, seg.mask(it) at
+ * duckdb_search_full_scan.cpp:1898): bit d of word d>>6 = doc d live.
+ * NULL = all live. */
+static const uint64_t* o_live = NULL;
+void o_set_live_mask(const uint64_t* m) { o_live = m; }
+
 /* Process windows covering docs [range_lo, range_hi] (inclusive, 1-based,
  * local to the segment). Cursors must be positioned before range_lo.
  * Exactly one of coll / cands is non-NULL.
@@ -617,6 +623,10 @@ static uint64_t o_exec_range(OCursor* cur, uint32_t nterms,
         const uint32_t bit = (uint32_t)__builtin_ctzll(word);
         word &= word - 1;
         const uint32_t doc = base + bit;
+        if (o_live && !((o_live[doc >> 6] >> (doc & 63u)) & 1ull)) {
+          --matches; /* masked doc: invisible (popcount pre-counted it) */
+          continue;
+        }
         if (hy && hy->col) {
           const int64_t v = hy->col[doc];
           if (v < hy->lo || v > hy->hi) {

@@ -314,6 +314,7 @@ def gpu():
             "sdb_gpu_table_load", "sdb_gpu_table_free", "sdb_gpu_scan_agg",
             "sdb_gpu_scan_agg_hash",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
+            "sdb_gpu_segment_attach_livemask",
             "sdb_gpu_execute_match_docs", "sdb_gpu_execute_count",
         ):
             getattr(lib, f).restype = C.c_int
@@ -425,6 +426,23 @@ class GpuContext:
             col.ctypes.data_as(C.POINTER(C.c_int64)))
         if rc != 0:
             raise RuntimeError(f"attach_column rc={rc}")
+
+    def attach_livemask(self, seg, mask):
+        """Attach (or with mask=None detach) a live-document bitmap —
+        the deleted-docs mask the reference wraps around every scan
+        (seg.mask(it), duckdb_search_full_scan.cpp:1898). uint64 words,
+        bit d of word d>>6 = doc d live."""
+        import numpy as np
+
+        if mask is None:
+            rc = self._lib.sdb_gpu_segment_attach_livemask(self._ctx, seg,
+                                                           None)
+        else:
+            m = np.ascontiguousarray(mask, dtype=np.uint64)
+            rc = self._lib.sdb_gpu_segment_attach_livemask(
+                self._ctx, seg, m.ctypes.data_as(C.POINTER(C.c_uint64)))
+        if rc != 0:
+            raise RuntimeError(f"attach_livemask rc={rc}")
 
     def attach_boost(self, seg, boost):
         """Attach the per-doc f32 filter-boost column (doc_count+1,

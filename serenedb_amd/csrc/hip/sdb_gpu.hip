@@ -84,6 +84,11 @@ struct SdbGpuSegment {
                         // the classic hybrid/bucket column
   float* fboost;        // device, doc_count+1 per-doc filter boost or null
   float fboost_max;     // host-computed max (WAND bound / smax scaling)
+  unsigned long long* live;  // device live-doc bitmap (bit d of word d>>6;
+                             // +1 pad word) or null = all live. The
+                             // reference masks every scan when deletes
+                             // exist (seg.mask(it),
+                             // duckdb_search_full_scan.cpp:1898,2002,2226)
   SdbTermEntry* terms_host;  // host copy of term table
   SdbSegHeader hdr;     // host copy
 };
@@ -516,6 +521,11 @@ struct WindowArgs {
   unsigned long long* bucket_out;  // [2*nbuckets]: count, sum (i64 bits)
   uint32_t dcache_n;  // staged descriptors per term (host-shrunk so the
                       // cache fits LDS for wide plans; <= SDB_DESC_CACHE)
+  const unsigned long long* live;  // live-doc bitmap or null (all live);
+                                   // applied BEFORE match counts,
+                                   // histogram and appends (a masked doc
+                                   // is invisible, Masked count mode
+                                   // duckdb_search_full_scan.cpp:2475)
   const float* fb;    // per-doc filter boost (null = off); each term
                       // contribution is multiplied by fb[doc]
   float fbmax;        // 1.0 when off; scales WAND bounds (smax is scaled
@@ -848,6 +858,13 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         const uint32_t off = base + e;
         if (off >= wlen) break;
         if (((cw >> (8 * e)) & 0xFFu) < mm) continue;
+        if (a.live) {
+          const uint64_t d = (uint64_t)lo + off;
+          if (!((a.live[d >> 6] >> (d & 63u)) & 1ull)) {
+            cwin[off] = 0;  // masked: invisible to count/hist/append
+            continue;
+          }
+        }
         if (a.fcol) {
           const long long vv = a.fcol[lo + off];
           if (vv < a.flo || vv > a.fhi) {
@@ -1334,6 +1351,16 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     uint32_t my_cnt = 0;
     for (uint32_t i = tid; i < nw_act; i += NTH) {
       unsigned long long word = mwin[i];
+      if (word && a.live) {
+        // window word i covers docs lo+64i .. lo+64i+63; lo = 1 + w*WD
+        // and WD % 64 == 0, so the live-bitmap shift is the constant 1
+        const uint64_t d0 = (uint64_t)lo + 64u * i;
+        const unsigned long long lw = (a.live[d0 >> 6] >> (d0 & 63u)) |
+                                      (a.live[(d0 >> 6) + 1]
+                                       << (64u - (d0 & 63u)));
+        word &= lw;
+        mwin[i] = word;  // recount/append walks see the masked word
+      }
       if (!word) continue;
       my_matches += (uint32_t)__popcll(word);
       const uint32_t base = i * 64u;
@@ -1723,8 +1750,30 @@ int sdb_gpu_segment_free(SdbGpuCtx* ctx, SdbGpuSegment* seg) {
   for (int i = 0; i < 4; ++i)
     if (seg->fcols[i]) (void)hipFree(seg->fcols[i]);
   if (seg->fboost) (void)hipFree(seg->fboost);
+  if (seg->live) (void)hipFree(seg->live);
   std::free(seg->terms_host);
   delete seg;
+  return SDB_OK;
+}
+
+// Attach a live-document bitmap (bit d of word d>>6 set = doc d live,
+// docs 1..doc_count; caller passes (doc_count+64)/64 words — the analogue
+// of the reference's deleted-docs mask wrapped around every scan,
+// duckdb_search_full_scan.cpp:1898 seg.mask(it) / Masked count :2475).
+// NULL mask detaches (back to all-live).
+int sdb_gpu_segment_attach_livemask(SdbGpuCtx* ctx, SdbGpuSegment* seg,
+                                    const uint64_t* mask) {
+  if (!ctx || !seg) return SDB_ERR_INVALID;
+  if (!mask) {
+    if (seg->live) (void)hipFree(seg->live);
+    seg->live = nullptr;
+    return SDB_OK;
+  }
+  const uint64_t nwords = ((uint64_t)seg->hdr.doc_count + 64) / 64;
+  if (!seg->live)
+    HIP_CHECK(hipMalloc(&seg->live, 8 * (nwords + 1)));  // +1: funnel pad
+  HIP_CHECK(hipMemcpy(seg->live, mask, 8 * nwords, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemsetAsync(seg->live + nwords, 0, 8, nullptr));
   return SDB_OK;
 }
 
@@ -1957,6 +2006,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.dcache_n = dcache_n;
     a.fb = plan->filter_boost ? seg->fboost : nullptr;
     a.fbmax = plan->filter_boost ? fbmax : 1.0f;
+    a.live = seg->live;
     a.gthresh = ctx->d_gthresh;
     a.ghist = ctx->d_ghist;
     a.cands = ctx->d_cands;

@@ -1022,3 +1022,60 @@ def test_scan_eq_and_f32_predicates(ctx):
     live = np.nonzero(np.bincount(keys[fsel], minlength=ngroups))[0]
     np.testing.assert_array_equal(gk, live)
     ctx.free_table(tab)
+
+
+def test_livemask_parity(ctx):
+    """Deleted-doc live masks (round-1 missing #2; seg.mask(it) at
+    duckdb_search_full_scan.cpp:1898, Masked count :2475): with a mask
+    attached, hits, total_matches, counts, WAND results and streaming
+    emission see LIVE docs only — vs the oracle with the same mask."""
+    doc_count = 500_000
+    blob, postings, _ = make_corpus(55, doc_count, [0.10, 0.05, 0.02])
+    rng = np.random.default_rng(56)
+    # ~25% deleted
+    nwords = (doc_count + 64) // 64
+    mask = rng.integers(0, 1 << 64, nwords, dtype=np.uint64)
+    mask |= rng.integers(0, 1 << 64, nwords, dtype=np.uint64)
+    seg = ctx.load_segment(blob)
+    ctx.attach_livemask(seg, mask)
+
+    def live(d):
+        return (int(mask[d >> 6]) >> (d & 63)) & 1
+
+    for mm in (1, 2, 3):  # lean sweep path (mm=1) and general path
+        hits, total = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 500,
+                                       min_match=mm)
+        ohits, ototal = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 500,
+                                        min_match=mm, live_mask=mask)
+        assert total == ototal
+        np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+        np.testing.assert_array_equal(hits["score"].view(np.uint32),
+                                      ohits["score"].view(np.uint32))
+        assert all(live(int(d)) for d in hits["doc"])
+
+    # masked CountFast == oracle masked count
+    cnt = ctx.execute_count([seg], [0, 1], [1.0, 1.0])
+    _, ocnt = po.execute_topk([blob], [0, 1], [1.0, 1.0], 1,
+                              live_mask=mask)
+    assert cnt == ocnt
+
+    # WAND stays exact under the mask
+    h1, t1 = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 10, wand=True)
+    h0, _ = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 10,
+                            live_mask=mask)
+    np.testing.assert_array_equal(h1["doc"], h0["doc"])
+    np.testing.assert_array_equal(h1["score"].view(np.uint32),
+                                  h0["score"].view(np.uint32))
+
+    # streaming emission: only live docs come out
+    docs, _, tot = ctx.execute_match_docs(seg, [0], [1.0], doc_count)
+    exp = postings[0][0][[bool(live(int(d))) for d in postings[0][0]]]
+    np.testing.assert_array_equal(docs, exp)
+    assert tot == len(exp)
+
+    # detach: everything matches the unmasked oracle again
+    ctx.attach_livemask(seg, None)
+    hits, total = ctx.execute_topk([seg], [0, 1, 2], [1.0] * 3, 100)
+    ohits, ototal = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 100)
+    assert total == ototal
+    np.testing.assert_array_equal(hits["doc"], ohits["doc"])
