@@ -1279,6 +1279,29 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             }
           }
         }
+#ifdef SDB_ABLATE_NOLOAD
+        // perf ablation: fabricate postings without touching the payload
+        // (keeps LDS accumulate + scan shape; results are WRONG)
+        {
+          const uint32_t span = d.last_doc - d.prev_doc;
+          const uint32_t i0 = 2u * (uint32_t)lane;
+#pragma unroll
+          for (uint32_t e = 0; e < 2; ++e) {
+            const uint32_t i = i0 + e;
+            if (i >= d.len) break;
+            const uint32_t doc =
+              d.prev_doc + 1 + (uint32_t)(((uint64_t)i * span) / 128u);
+            if (doc < lo || doc > hi) continue;
+            const float sc = score_one(a.scorer, num, nc, nl, 1 + (i & 7),
+                                       100 + i);
+            const uint32_t off = doc - lo;
+            swin[off] += sc;
+            mark_match_mask(mwin, off);
+          }
+        }
+        b += NW;
+        continue;
+#endif
         if (try_block_fused<1>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
                                nl, a.scorer, a.norms, a.fb, swin, mwin)) {
           b += NW;
@@ -1306,8 +1329,13 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #ifdef SDB_TIMING
       if (t == 0) SDB_TS(5)  // term-0 phase alone (imbalance diagnosis)
 #endif
+#ifndef SDB_ABLATE_NOBARRIER  // perf ablation: cost of term serialization
       __syncthreads();  // term-major merge order (bit-exact vs oracle)
+#endif
     }
+#ifdef SDB_ABLATE_NOBARRIER
+    __syncthreads();
+#endif
     asm volatile("" ::"v"(pf_acc));  // waits land here, once per window
     SDB_TS(1)
     // advance every term's cursor once per window

@@ -961,16 +961,19 @@ def test_hash_agg_sparse_keys(ctx):
     aggregation. Raw and FoR key codecs."""
     rows = 3_000_000
     rng = np.random.default_rng(99)
-    base = rng.integers(-(1 << 62), 1 << 62, 100_000).astype(np.int64)
-    keys = base[rng.integers(0, len(base), rows)]
-    keys[::997] = -1  # the sentinel-adjacent key must work too
+    # raw: full-range 64-bit keys; for: sparse keys within the FoR codec's
+    # 32-bit group-width envelope (the codec is width<=32 by design)
+    base_raw = rng.integers(-(1 << 62), 1 << 62, 100_000).astype(np.int64)
+    base_for = rng.integers(-(1 << 30), 1 << 30, 100_000).astype(np.int64)
     v1 = rng.integers(-(1 << 30), 1 << 30, rows).astype(np.int64)
     v2 = rng.normal(0, 1, rows).astype(np.float32)
     c = int((1 << 30) * -0.8)
     sel = v1 >= c
-    uk, cnt, s1, s2 = _np_hash_agg(keys, v1, v2, sel)
 
-    for enc in ("raw", "for"):
+    for enc, base in (("raw", base_raw), ("for", base_for)):
+        keys = base[rng.integers(0, len(base), rows)]
+        keys[::997] = -1  # the sentinel-adjacent key must work too
+        uk, cnt, s1, s2 = _np_hash_agg(keys, v1, v2, sel)
         tab = ctx.load_table([keys, v1, v2], [enc, enc, "raw"])
         gkeys, i64, f64, passed = ctx.scan_agg_hash(
             tab, 0, 200_000, [(1, 2, c, 0)], [(0, 0), (1, 1), (2, 2)])
