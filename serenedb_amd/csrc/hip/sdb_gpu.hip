@@ -1074,9 +1074,12 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 //    term barriers (the fixed 24576x1024 1-WG/CU shape left every barrier
 //    stall empty).
 template <uint32_t WD, uint32_t NTH>
-// 2nd launch-bounds arg = min waves/SIMD: 4 keeps VGPRs <= 128 so two
-// 512-thread WGs (or one 1024-thread WG) co-reside per CU
-__launch_bounds__(NTH, 4) __global__
+// 2nd launch-bounds arg = min waves/SIMD. A 1024-thread WG forces 4/SIMD
+// (VGPRs <= 128), and at that cap the kernel spills ~65 SGPRs to SCRATCH
+// (212 B/thread private segment measured via rocprofv3) — reloads in the
+// window loop. 512-thread WGs declare 2/SIMD instead: 256 VGPRs, no
+// scratch, trading TLP for a clean register file.
+__launch_bounds__(NTH, NTH == 1024 ? 4 : 2) __global__
 void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   constexpr uint32_t NW = NTH / 64u;
   constexpr uint32_t NWORDS = WD / 64u;
@@ -1246,6 +1249,29 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             continue;
           }
         }
+#ifdef SDB_ABLATE_NOLOAD
+        // perf ablation: fabricate postings with NO payload access at all
+        // (no fused pair, no prefetch; results are WRONG)
+        {
+          const uint32_t span = d.last_doc - d.prev_doc;
+          const uint32_t i0 = 2u * (uint32_t)lane;
+#pragma unroll
+          for (uint32_t e = 0; e < 2; ++e) {
+            const uint32_t i = i0 + e;
+            if (i >= d.len) break;
+            const uint32_t doc =
+              d.prev_doc + 1 + (uint32_t)(((uint64_t)i * span) / 128u);
+            if (doc < lo || doc > hi) continue;
+            const float sc = score_one(a.scorer, num, nc, nl, 1 + (i & 7),
+                                       100 + i);
+            const uint32_t off = doc - lo;
+            swin[off] += sc;
+            mark_match_mask(mwin, off);
+          }
+        }
+        b += NW;
+        continue;
+#endif
         // pair with the wave's next block when both are the fused shape
         {
           const uint64_t b2 = b + NW;
@@ -1279,29 +1305,6 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             }
           }
         }
-#ifdef SDB_ABLATE_NOLOAD
-        // perf ablation: fabricate postings without touching the payload
-        // (keeps LDS accumulate + scan shape; results are WRONG)
-        {
-          const uint32_t span = d.last_doc - d.prev_doc;
-          const uint32_t i0 = 2u * (uint32_t)lane;
-#pragma unroll
-          for (uint32_t e = 0; e < 2; ++e) {
-            const uint32_t i = i0 + e;
-            if (i >= d.len) break;
-            const uint32_t doc =
-              d.prev_doc + 1 + (uint32_t)(((uint64_t)i * span) / 128u);
-            if (doc < lo || doc > hi) continue;
-            const float sc = score_one(a.scorer, num, nc, nl, 1 + (i & 7),
-                                       100 + i);
-            const uint32_t off = doc - lo;
-            swin[off] += sc;
-            mark_match_mask(mwin, off);
-          }
-        }
-        b += NW;
-        continue;
-#endif
         if (try_block_fused<1>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
                                nl, a.scorer, a.norms, a.fb, swin, mwin)) {
           b += NW;
