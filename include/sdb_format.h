@@ -127,6 +127,20 @@ typedef struct SdbSegHeader {
   uint64_t blob_size;
 } SdbSegHeader;
 
+/* Term metadata sidecar for on-disk `.doc` ingestion — exactly the
+ * TermMetaImpl fields the reference's burst-trie terms dict serializes per
+ * term (formats_attributes.hpp:30; PostingsWriterBase::Encode,
+ * formats/posting/writer.hpp:379-412). The burst-trie reader itself is out
+ * of scope this round: callers supply these from a sidecar.
+ */
+typedef struct SdbDocTermMeta {
+  uint32_t docs_count;   /* df */
+  uint64_t total_freq;   /* sum of freqs (meta.freq) */
+  uint64_t doc_start;    /* absolute offset of the term's stream in .doc */
+  uint32_t e_single_doc; /* df==1: doc - doc_limits::min() */
+  uint64_t e_skip_start; /* df>128: skip area offset relative to doc_start */
+} SdbDocTermMeta;
+
 /* Parsed (in-memory) view over a segment blob. */
 typedef struct SdbSegmentView {
   const SdbSegHeader* hdr;

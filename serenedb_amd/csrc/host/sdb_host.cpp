@@ -956,3 +956,263 @@ int sdb_host_topk_select(const SdbScoreDocC* cands, uint64_t n, uint32_t k,
 }
 
 }  // extern "C"
+
+
+/* ------------------------------------------------------------------------- */
+/* On-disk reference ingestion: read the reference's   */
+/* `.doc` postings stream (PostingsWriterBase, formats/posting/writer.hpp)   */
+/* and rebuild it as this project's segment container. The term metadata the */
+/* burst-trie terms dict (.tm) would supply arrives as a caller sidecar      */
+/* (SdbDocTermMeta == the TermMetaImpl fields Encode() serializes,           */
+/* writer.hpp:379-412); the burst-trie itself is SURVEY.md §2 out of scope   */
+/* this round.                                                               */
+/*                                                                           */
+/* File layout restated from the reference:                                  */
+/*  header  : u32 LE magic 0x3fd76c17, vint-length + "iresearch_10_         */
+/*            postings_documents", u32 version (format_utils.cpp:56-64)      */
+/*  per term (TermMeta doc_start): full 128-doc blocks [WriteBlockDelta      */
+/*            docs][WriteBlock freqs] (writer.hpp:618-627); then            */
+/*            df==1        -> nothing (e_single_doc in meta)                 */
+/*            df<=128      -> [root wand: u8 size + bytes][tail delta docs   */
+/*                            + tail freqs]  (EndTerm writer.hpp:346-369:    */
+/*                            write_max_score(0) BEFORE FlushTailDoc)        */
+/*            df>128       -> [tail...][@e_skip_start: root wand][v32        */
+/*                            num_levels][levels n..0: v64 len + bytes]      */
+/*            level-0 entry: v32 absolute block-last doc, v64 doc-ptr delta  */
+/*            (base doc_start), u8 wand size, wand bytes (vint freq [+ vint  */
+/*            norm-freq], wand_writer.hpp:196-206); level>=1 entries append  */
+/*            v64 child position (skip_list.hpp:92-116)                      */
+/*  footer  : u32 LE -magic, u32 alg 0, u64 CRC-32C of file[0..len-8)        */
+/*            (format_utils.cpp WriteFooter; libs/basics/crc.hpp = absl      */
+/*            crc32c = Castagnoli, reflected, init/final-xor 0xffffffff)     */
+/* ------------------------------------------------------------------------- */
+
+static uint32_t g_crc32c_tab[256];
+static bool g_crc32c_ready = false;
+static void crc32c_build_tab() {
+  for (uint32_t i = 0; i < 256; ++i) {
+    uint32_t c = i;
+    for (int k = 0; k < 8; ++k)
+      c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+    g_crc32c_tab[i] = c;
+  }
+  g_crc32c_ready = true;
+}
+static uint32_t crc32c_bytes(const uint8_t* p, size_t n) {
+  if (!g_crc32c_ready) crc32c_build_tab();
+  uint32_t c = 0xFFFFFFFFu;
+  for (size_t i = 0; i < n; ++i)
+    c = g_crc32c_tab[(c ^ p[i]) & 0xFFu] ^ (c >> 8);
+  return c ^ 0xFFFFFFFFu;
+}
+
+struct DocCursor {
+  const uint8_t* p;
+  const uint8_t* end;
+  bool fail = false;
+  uint8_t u8() {
+    if (p >= end) { fail = true; return 0; }
+    return *p++;
+  }
+  uint32_t u32le() {
+    if (end - p < 4) { fail = true; return 0; }
+    uint32_t v;
+    std::memcpy(&v, p, 4);
+    p += 4;
+    return v;  /* LE host assumed (x86/gfx hosts) */
+  }
+  uint64_t vu64() {  /* WriteVarint, data_output.hpp:58 (LEB128) */
+    uint64_t v = 0;
+    for (uint32_t sh = 0; sh < 64; sh += 7) {
+      const uint8_t b = u8();
+      v |= (uint64_t)(b & 0x7F) << sh;
+      if (!(b & 0x80)) return v;
+    }
+    fail = true;
+    return 0;
+  }
+  void skip(uint64_t n) {
+    if ((uint64_t)(end - p) < n) { fail = true; return; }
+    p += n;
+  }
+};
+
+extern "C" {
+
+int sdb_host_ingest_doc(const void* file, uint64_t size,
+                        const SdbDocTermMeta* metas, uint32_t nterms,
+                        uint32_t doc_count, uint32_t has_freq,
+                        const uint32_t* norms /* doc_count+1 */,
+                        void** blob_out, uint64_t* size_out) {
+  if (!file || !metas || !blob_out || !size_out || nterms == 0 ||
+      doc_count == 0)
+    return -1;
+  const uint8_t* f = (const uint8_t*)file;
+  static const char kFmt[] = "iresearch_10_postings_documents";
+  const uint64_t fmt_len = sizeof(kFmt) - 1;
+  const uint64_t hdr_len = 4 + 1 + fmt_len + 4; /* vint(31) = 1 byte */
+  if (size < hdr_len + 16) return -71;
+  /* header */
+  DocCursor h{f, f + size};
+  if (h.u32le() != 0x3fd76c17u) return -72;
+  if (h.vu64() != fmt_len) return -73;
+  if (std::memcmp(h.p, kFmt, fmt_len) != 0) return -73;
+  h.skip(fmt_len);
+  (void)h.u32le(); /* version: recorded formats accept their own range */
+  /* footer (format_utils.cpp ValidateFooter + WriteFooter) */
+  {
+    DocCursor t{f + size - 16, f + size};
+    if (t.u32le() != (uint32_t)(-(int32_t)0x3fd76c17)) return -74;
+    if (t.u32le() != 0) return -75;
+    uint64_t want;
+    std::memcpy(&want, f + size - 8, 8);
+    const uint32_t got = crc32c_bytes(f, size - 8);
+    if (want != (uint64_t)got) return -76;
+  }
+
+  /* decode every term's postings */
+  std::vector<std::vector<uint32_t>> tdocs(nterms), tfreqs(nterms);
+  for (uint32_t t = 0; t < nterms; ++t) {
+    const SdbDocTermMeta& m = metas[t];
+    const uint32_t df = m.docs_count;
+    if (df == 0) continue;
+    tdocs[t].reserve(df);
+    tfreqs[t].reserve(df);
+    if (df == 1) {
+      const uint32_t doc = 1u + m.e_single_doc; /* + doc_limits::min() */
+      if (doc > doc_count) return -77;
+      tdocs[t].push_back(doc);
+      tfreqs[t].push_back(has_freq ? (uint32_t)m.total_freq : 1u);
+      continue;
+    }
+    if (m.doc_start >= size - 16) return -77;
+    DocCursor c{f + m.doc_start, f + size - 16};
+    const uint32_t full_blocks = df / 128u;
+    const uint32_t tail_len = df % 128u;
+    const bool has_skip = df > 128u;
+    uint32_t prev = 0;
+    uint32_t buf_d[128], buf_f[128];
+    /* a block is at most 1 tag + 128*4 values + svb header slack; decode
+     * through a zero-padded copy near the end of the span so a corrupt
+     * length can never read past the caller's buffer */
+    uint8_t safe[704];
+    auto safe_docs = [&](DocCursor& cc, uint32_t len, uint32_t pv,
+                         uint32_t* out) -> bool {
+      const uint64_t rem = (uint64_t)(cc.end - cc.p);
+      uint32_t consumed;
+      if (rem >= sizeof(safe)) {
+        consumed = decode_doc_block(cc.p, len, pv, out);
+      } else {
+        std::memset(safe, 0, sizeof(safe));
+        std::memcpy(safe, cc.p, rem);
+        consumed = decode_doc_block(safe, len, pv, out);
+        if (consumed > rem) return false;
+      }
+      if (!consumed) return false;
+      cc.skip(consumed);
+      return !cc.fail;
+    };
+    auto safe_freqs = [&](DocCursor& cc, uint32_t len,
+                          uint32_t* out) -> bool {
+      const uint64_t rem = (uint64_t)(cc.end - cc.p);
+      uint32_t consumed;
+      if (rem >= sizeof(safe)) {
+        consumed = decode_freq_block(cc.p, len, out);
+      } else {
+        std::memset(safe, 0, sizeof(safe));
+        std::memcpy(safe, cc.p, rem);
+        consumed = decode_freq_block(safe, len, out);
+        if (consumed > rem) return false;
+      }
+      if (!consumed) return false;
+      cc.skip(consumed);
+      return !cc.fail;
+    };
+    auto push_block = [&](uint32_t len) -> bool {
+      uint32_t last = prev;
+      for (uint32_t i = 0; i < len; ++i) {
+        if (buf_d[i] <= last || buf_d[i] > doc_count) return false;
+        last = buf_d[i];
+        tdocs[t].push_back(buf_d[i]);
+        tfreqs[t].push_back(has_freq ? buf_f[i] : 1u);
+      }
+      prev = last;
+      return true;
+    };
+    std::vector<uint64_t> block_end_off(full_blocks);
+    for (uint32_t b = 0; b < full_blocks; ++b) {
+      if (!safe_docs(c, 128, prev, buf_d)) return -78;
+      if (has_freq && !safe_freqs(c, 128, buf_f)) return -78;
+      if (!push_block(128)) return -79;
+      block_end_off[b] = (uint64_t)(c.p - (f + m.doc_start));
+    }
+    auto read_wand_entry = [&](DocCursor& cc) {
+      const uint8_t sz = cc.u8();
+      cc.skip(sz);
+    };
+    if (!has_skip) {
+      read_wand_entry(c); /* root wand BEFORE the tail (EndTerm order) */
+      if (tail_len) {
+        if (!safe_docs(c, tail_len, prev, buf_d)) return -78;
+        if (has_freq && !safe_freqs(c, tail_len, buf_f)) return -78;
+        if (!push_block(tail_len)) return -79;
+      }
+    } else {
+      if (tail_len) {
+        if (!safe_docs(c, tail_len, prev, buf_d)) return -78;
+        if (has_freq && !safe_freqs(c, tail_len, buf_f)) return -78;
+        if (!push_block(tail_len)) return -79;
+      }
+      /* the tail must end exactly where the meta says the skip area is */
+      if ((uint64_t)(c.p - (f + m.doc_start)) != m.e_skip_start) return -80;
+      DocCursor sk{f + m.doc_start + m.e_skip_start, f + size - 16};
+      read_wand_entry(sk); /* root */
+      const uint64_t num_levels = sk.vu64();
+      if (num_levels == 0 || num_levels > 10) return -81;
+      /* levels are written n..0; walk down to level 0 and validate it */
+      const uint8_t* lvl0 = nullptr;
+      uint64_t lvl0_len = 0;
+      for (uint64_t l = 0; l < num_levels; ++l) {
+        const uint64_t len = sk.vu64();
+        if (sk.fail) return -81;
+        if (l + 1 == num_levels) { /* last written = level 0 */
+          lvl0 = sk.p;
+          lvl0_len = len;
+        }
+        sk.skip(len);
+      }
+      if (sk.fail || !lvl0) return -81;
+      /* level-0 entries: one per full block except the last flush without
+       * a following doc (writer.hpp:733 skip cadence; no terminal skip) */
+      const uint32_t nentries = tail_len ? full_blocks : full_blocks - 1;
+      DocCursor e{lvl0, lvl0 + lvl0_len};
+      uint64_t ptr = 0;
+      for (uint32_t i = 0; i < nentries; ++i) {
+        const uint32_t entry_doc = (uint32_t)e.vu64();
+        ptr += e.vu64();
+        read_wand_entry(e);
+        if (e.fail) return -82;
+        /* cross-checks against the decoded blocks */
+        if (entry_doc != tdocs[t][(size_t)(i + 1) * 128 - 1]) return -83;
+        if (ptr != block_end_off[i]) return -84;
+      }
+    }
+    if (tdocs[t].size() != df) return -85;
+  }
+
+  /* rebuild as this project's container (norm streams, descriptors, WAND
+   * bounds recomputed exactly as the synthetic builder does) */
+  std::vector<uint32_t> df(nterms);
+  std::vector<const uint32_t*> dptr(nterms), fptr(nterms);
+  for (uint32_t t = 0; t < nterms; ++t) {
+    df[t] = (uint32_t)tdocs[t].size();
+    dptr[t] = tdocs[t].data();
+    fptr[t] = tfreqs[t].data();
+  }
+  return sdb_host_build_segment(doc_count, nterms, df.data(),
+                                (const uint32_t**)dptr.data(),
+                                (const uint32_t**)fptr.data(), norms,
+                                blob_out, size_out);
+}
+
+}  /* extern "C" */
