@@ -1538,6 +1538,420 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #undef SDB_TS
 }
 
+
+// ---------------------------------------------------------------------------
+// Per-WAVE window kernel (round-2, second reshape). Diagnosis that led
+// here (gpurun_out/r2_abl_1b.log + SQ PMC): the workgroup-window kernels
+// are NOT bandwidth- or issue-bound — with every payload load ablated
+// away they still run at ~58% of their time, and waves park 78% of
+// cycles. The WG-wide machinery (barrier convoys, WG-wide zero/sweep
+// passes, serialized per-phase latencies) is the bound.
+//
+// This kernel has NO __syncthreads at all: each WAVE owns a contiguous
+// doc range and processes it with PRIVATE per-wave LDS state, so wave
+// lockstep replaces every barrier and independent waves cover each
+// other's stalls:
+//   - per term, a RING of decoded (doc,score) postings (filled
+//     wave-cooperatively a 128-block at a time with the fused register
+//     decode; the ring decouples block boundaries from accumulation, so
+//     nothing is ever decoded twice);
+//   - a small (SDB_PW_SUBW-doc) score window + match bitmask, filled
+//     term-major from the rings (fixed fp32 merge order -> still
+//     bit-exact vs the oracle);
+//   - the same bin-space threshold machinery (score_bin), with a
+//     per-wave LDS histogram flushed into the global shards on a cadence
+//     and the suffix-count derivation run wave-locally.
+// A small SEED launch (strided 1%-ish doc sample, histogram only, no
+// appends) locks a valid threshold bin before the main grid starts, so
+// early sub-windows do not flood the candidate buffer — the bound stays
+// provable: sample counts are a subset of global counts.
+// ---------------------------------------------------------------------------
+
+#ifndef SDB_PW_SUBW
+#define SDB_PW_SUBW 1024u  // docs per sub-window (per-wave LDS f32 window)
+#endif
+#define SDB_PW_RING 256u   // ring entries per term (2 full blocks)
+#define SDB_PW_TERMS 4u    // wave path handles <= 4 terms (headline shape)
+#define SDB_PW_NTH 256u    // threads per WG = 4 independent waves
+
+// per-wave LDS footprint (u64-aligned)
+#define SDB_PW_WAVE_LDS_BYTES                                            \
+  (SDB_PW_SUBW * 4 /*swin*/ + (SDB_PW_SUBW / 64) * 8 /*mask*/ +          \
+   SDB_PW_TERMS * SDB_PW_RING * 8 /*rings*/ + 384 * 4 /*scratch*/ +      \
+   SDB_HIST_BINS * 4 /*hist*/)
+
+struct PwState {  // per-(wave,term) wave-uniform state
+  uint64_t cur;       // next block (absolute desc index)
+  uint64_t dend;      // term desc end
+  uint64_t pbase;     // payload begin
+  uint32_t ring_head; // logical head
+  uint32_t ring_cnt;  // entries in ring
+  uint32_t cov;       // all postings with doc <= cov are in the ring
+  float num, nc, nl;
+};
+
+// wave-cooperative: decode block `d` of term state `st`, score, and append
+// postings inside [range_lo, range_hi] to the ring. Returns appended count.
+__device__ __forceinline__ uint32_t pw_fill_block(
+  const WindowArgs& a, PwState& st, const SdbBlockDesc& d, int lane,
+  uint32_t range_lo, uint32_t range_hi, unsigned long long* ring,
+  uint32_t* scratch) {
+  const uint8_t* pl = a.payload + st.pbase;
+  uint32_t doc0 = 0, doc1 = 0;
+  float s0 = 0.f, s1 = 0.f;
+  bool have01 = false;
+  if (d.len == 128) {
+    const uint8_t* db = pl + d.doc_off;
+    const uint8_t* fb = pl + d.freq_off;
+    const uint8_t* nb = fb + d.flags;
+    const uint32_t dtag = db[0], ftag = fb[0];
+    const uint32_t ntag = a.norm_stream ? nb[0] : SDB_E_BITPACK_01;
+    if (dtag >= SDB_DE_DELTA_BITPACK_02 && ftag >= SDB_E_BITPACK_01 &&
+        ntag >= SDB_E_BITPACK_01) {
+      const uint32_t dbits = dtag - SDB_DE_DELTA_BITPACK_02 + 2;
+      const uint32_t fbits = ftag - SDB_E_BITPACK_01 + 1;
+      const uint32_t nbits = ntag - SDB_E_BITPACK_01 + 1;
+      const uint32_t i0 = 2u * (uint32_t)lane, i1 = i0 + 1;
+      const uint32_t dd0 = extract_packed(db + 1, dbits, i0);
+      const uint32_t dd1 = extract_packed(db + 1, dbits, i1);
+      const uint32_t f0 = extract_packed(fb + 1, fbits, i0);
+      const uint32_t f1 = extract_packed(fb + 1, fbits, i1);
+      uint32_t n0 = 0, n1 = 0;
+      if (a.norm_stream) {
+        n0 = extract_packed(nb + 1, nbits, i0);
+        n1 = extract_packed(nb + 1, nbits, i1);
+      }
+      const uint32_t pair = dd0 + dd1;
+      const uint32_t incl = wave_incl_scan(pair, lane);
+      const uint32_t excl = incl - pair;
+      doc0 = d.prev_doc + excl + dd0;
+      doc1 = d.prev_doc + excl + pair;
+      if (!a.norm_stream) {
+        n0 = doc0 <= a.doc_count ? a.norms[doc0] : 1u;
+        n1 = doc1 <= a.doc_count ? a.norms[doc1] : 1u;
+      }
+      const float nm0 = a.fb && doc0 <= a.doc_count
+                          ? st.num * a.fb[doc0] : st.num;
+      const float nm1 = a.fb && doc1 <= a.doc_count
+                          ? st.num * a.fb[doc1] : st.num;
+      s0 = score_one(a.scorer, nm0, st.nc, st.nl, f0, n0);
+      s1 = score_one(a.scorer, nm1, st.nc, st.nl, f1, n1);
+      have01 = true;
+    }
+  }
+  if (!have01) {
+    // generic families through the per-wave scratch (no cross-wave use)
+    uint32_t* dbuf = scratch;
+    uint32_t* fbuf = scratch + 128;
+    uint32_t* nbuf = scratch + 256;
+    decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
+    decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
+    if (a.norm_stream)
+      decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane, nbuf);
+    const uint32_t i0 = 2u * (uint32_t)lane, i1 = i0 + 1;
+    doc0 = i0 < d.len ? dbuf[i0] : 0xFFFFFFFFu;
+    doc1 = i1 < d.len ? dbuf[i1] : 0xFFFFFFFFu;
+    const uint32_t f0 = i0 < d.len ? fbuf[i0] : 0;
+    const uint32_t f1 = i1 < d.len ? fbuf[i1] : 0;
+    const uint32_t n0 =
+      i0 < d.len ? (a.norm_stream ? nbuf[i0] : a.norms[doc0]) : 1u;
+    const uint32_t n1 =
+      i1 < d.len ? (a.norm_stream ? nbuf[i1] : a.norms[doc1]) : 1u;
+    const float nm0 =
+      a.fb && i0 < d.len ? st.num * a.fb[doc0] : st.num;
+    const float nm1 =
+      a.fb && i1 < d.len ? st.num * a.fb[doc1] : st.num;
+    s0 = score_one(a.scorer, nm0, st.nc, st.nl, f0, n0);
+    s1 = score_one(a.scorer, nm1, st.nc, st.nl, f1, n1);
+  }
+  // range filter + order-preserving wave compaction into the ring
+  const bool k0 = doc0 >= range_lo && doc0 <= range_hi &&
+                  2u * (uint32_t)lane < d.len;
+  const bool k1 = doc1 >= range_lo && doc1 <= range_hi &&
+                  2u * (uint32_t)lane + 1 < d.len;
+  const uint32_t cnt = (k0 ? 1u : 0u) + (k1 ? 1u : 0u);
+  const uint32_t incl = wave_incl_scan(cnt, lane);
+  const uint32_t excl = incl - cnt;
+  const uint32_t total = __shfl(incl, 63, 64);
+  uint32_t pos = st.ring_head + st.ring_cnt + excl;
+  if (k0) {
+    ring[pos & (SDB_PW_RING - 1)] =
+      ((unsigned long long)__float_as_uint(s0) << 32) | doc0;
+    ++pos;
+  }
+  if (k1)
+    ring[pos & (SDB_PW_RING - 1)] =
+      ((unsigned long long)__float_as_uint(s1) << 32) | doc1;
+  return total;
+}
+
+__launch_bounds__(SDB_PW_NTH, 2) __global__
+void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
+                      uint32_t nwaves_total, uint32_t range_docs,
+                      uint32_t range_stride, int seed_mode) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const uint32_t tid = threadIdx.x;
+  const int lane = tid & 63;
+  const uint32_t wave_in_wg = tid >> 6;
+  const uint32_t gwave = blockIdx.x * (SDB_PW_NTH / 64u) + wave_in_wg;
+  if (gwave >= nwaves_total) return;
+
+  char* wbase = smem + (size_t)wave_in_wg * SDB_PW_WAVE_LDS_BYTES;
+  float* swin = (float*)wbase;
+  unsigned long long* mwin = (unsigned long long*)(swin + SDB_PW_SUBW);
+  unsigned long long* rings = mwin + SDB_PW_SUBW / 64;
+  uint32_t* scratch = (uint32_t*)(rings + SDB_PW_TERMS * SDB_PW_RING);
+  uint32_t* hist = scratch + 384;
+
+  // wave's doc range: seed mode samples strided slices; main mode tiles
+  // the doc space contiguously
+  const uint64_t r_lo64 = 1ull + (uint64_t)gwave * range_stride;
+  if (r_lo64 > a.doc_count) return;
+  const uint32_t range_lo = (uint32_t)r_lo64;
+  const uint32_t range_hi = (uint32_t)min(
+    (uint64_t)a.doc_count, r_lo64 + range_docs - 1);
+
+  const float inv_smax = (float)SDB_HIST_BINS / a.smax;
+  uint32_t* gh = a.ghist + ((blockIdx.x + wave_in_wg) & 7u) * SDB_HIST_BINS;
+
+  for (uint32_t b = lane; b < SDB_HIST_BINS; b += 64) hist[b] = 0;
+
+  // per-term state (arrays force-unrolled so state stays in registers)
+  PwState st[SDB_PW_TERMS];
+#pragma unroll
+  for (uint32_t t = 0; t < SDB_PW_TERMS; ++t) {
+    st[t] = PwState{};
+    if (t < a.nterms) {
+      const TermDev te = terms[t];
+      st[t].dend = te.desc_end;
+      st[t].pbase = te.payload_begin;
+      st[t].num = te.num;
+      st[t].nc = te.nc;
+      st[t].nl = te.nl;
+      st[t].cur =
+        lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, range_lo);
+      // cov invariant: every posting in (consumed, cov] sits in the ring;
+      // nothing ringed yet
+      st[t].cov = st[t].cur >= te.desc_end ? range_hi : range_lo - 1;
+    } else {
+      st[t].cov = range_hi;
+      st[t].ring_cnt = 0;
+    }
+  }
+
+  unsigned long long wv_matches = 0;
+  uint32_t tbin = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+  uint32_t subw_idx = 0;
+  bool overflowed = false;
+
+  for (uint32_t sub_lo = range_lo; sub_lo <= range_hi;
+       ++subw_idx) {
+    uint32_t sub_hi = min(sub_lo + SDB_PW_SUBW - 1u, range_hi);
+    // ensure every term's ring covers sub_hi (or its blocks are done);
+    // a full ring bounds the sub-window instead
+#pragma unroll
+    for (uint32_t t = 0; t < SDB_PW_TERMS; ++t) {
+      if (t >= a.nterms) continue;
+      while (st[t].cov < sub_hi &&
+             st[t].ring_cnt + 128 <= SDB_PW_RING) {
+        if (st[t].cur >= st[t].dend) {
+          st[t].cov = range_hi;
+          break;
+        }
+        const SdbBlockDesc d = a.desc[st[t].cur];
+        if (d.prev_doc >= range_hi) {  // block fully beyond the range
+          st[t].cur = st[t].dend;
+          st[t].cov = range_hi;
+          break;
+        }
+        const uint32_t added =
+          pw_fill_block(a, st[t], d, lane, range_lo, range_hi,
+                        rings + t * SDB_PW_RING, scratch);
+        st[t].ring_cnt += added;
+        st[t].cov = min(d.last_doc, range_hi);
+        ++st[t].cur;
+      }
+      if (st[t].cov < sub_hi) sub_hi = st[t].cov;  // ring-full clamp
+    }
+    const uint32_t sub_len = sub_hi - sub_lo + 1u;
+    const uint32_t nwords = (sub_len + 63u) / 64u;
+
+    // zero the window (wide stores)
+    {
+      float4 z{0.f, 0.f, 0.f, 0.f};
+      float4* sw4 = (float4*)swin;
+      for (uint32_t i = lane; i < (sub_len + 3u) / 4u; i += 64) sw4[i] = z;
+      for (uint32_t i = lane; i < nwords; i += 64) mwin[i] = 0ull;
+    }
+
+    // term-major scatter from the rings (fixed merge order: bit-exact)
+#pragma unroll
+    for (uint32_t t = 0; t < SDB_PW_TERMS; ++t) {
+      if (t >= a.nterms) continue;
+      unsigned long long* ring = rings + t * SDB_PW_RING;
+      // count entries with doc <= sub_hi (entries are doc-sorted)
+      uint32_t n = 0;
+      while (n < st[t].ring_cnt) {
+        const uint32_t m = min(64u, st[t].ring_cnt - n);
+        uint32_t doc = 0;
+        if ((uint32_t)lane < m)
+          doc = (uint32_t)ring[(st[t].ring_head + n + lane) &
+                               (SDB_PW_RING - 1)];
+        const unsigned long long gt =
+          __ballot((uint32_t)lane < m && doc > sub_hi);
+        if (gt) {
+          n += (uint32_t)__ffsll((long long)gt) - 1u;
+          break;
+        }
+        n += m;
+      }
+      // scatter them
+      for (uint32_t base = 0; base < n; base += 64) {
+        if (base + lane < n) {
+          const unsigned long long e =
+            ring[(st[t].ring_head + base + lane) & (SDB_PW_RING - 1)];
+          const uint32_t doc = (uint32_t)e;
+          const uint32_t off = doc - sub_lo;
+          float sc;
+          const uint32_t sb32 = (uint32_t)(e >> 32);
+          __builtin_memcpy(&sc, &sb32, 4);
+          swin[off] += sc;  // distinct docs within a term
+          atomicOr(&mwin[off >> 6], 1ull << (off & 63u));
+        }
+      }
+      st[t].ring_head += n;
+      st[t].ring_cnt -= n;
+    }
+
+    // live-doc mask (general shift; sub_lo is arbitrary here)
+    if (a.live) {
+      for (uint32_t i = lane; i < nwords; i += 64) {
+        const uint64_t d0 = (uint64_t)sub_lo + 64u * i;
+        const uint32_t sh = (uint32_t)(d0 & 63u);
+        unsigned long long lw = a.live[d0 >> 6] >> sh;
+        if (sh) lw |= a.live[(d0 >> 6) + 1] << (64u - sh);
+        if (mwin[i]) mwin[i] &= lw;
+      }
+    }
+
+    // sweep: matches, histogram, candidate emission
+    uint32_t my_matches = 0, my_emit = 0;
+    unsigned long long word = lane < (int)nwords ? mwin[lane] : 0ull;
+    my_matches = (uint32_t)__popcll(word);
+    {
+      unsigned long long w2 = word;
+      while (w2) {
+        const uint32_t off =
+          64u * lane + (uint32_t)__ffsll((long long)w2) - 1u;
+        w2 &= w2 - 1;
+        const uint32_t sb = score_bin(swin[off], inv_smax);
+        atomicAdd(&hist[sb], 1u);
+        my_emit += (!seed_mode && sb >= tbin) ? 1u : 0u;
+      }
+    }
+    uint32_t wm = my_matches;
+#pragma unroll
+    for (int o = 32; o; o >>= 1) wm += __shfl_down(wm, o, 64);
+    wv_matches += __shfl(wm, 0, 64);
+    if (!seed_mode) {
+      const uint32_t incl = wave_incl_scan(my_emit, lane);
+      const uint32_t total = __shfl(incl, 63, 64);
+      if (total) {
+        uint32_t cbase = 0;
+        if (lane == 0) cbase = atomicAdd(a.cand_count, total);
+        cbase = __shfl(cbase, 0, 64);
+        if (cbase + total > a.cand_cap) {
+          if (lane == 0) atomicExch(a.overflow, 1u);
+          overflowed = true;
+          break;
+        }
+        uint32_t pos = cbase + incl - my_emit;
+        unsigned long long w3 = word;
+        while (w3) {
+          const uint32_t off =
+            64u * lane + (uint32_t)__ffsll((long long)w3) - 1u;
+          w3 &= w3 - 1;
+          const float sc = swin[off];
+          if (score_bin(sc, inv_smax) >= tbin) {
+            a.cands[pos].score = sc;
+            a.cands[pos].doc = sub_lo + off;
+            a.cands[pos].segment_idx = a.seg_idx;
+            ++pos;
+          }
+        }
+      }
+    }
+
+    // histogram flush + threshold refresh/derive on a cadence (early and
+    // then sparse; all wave-local, no synchronization)
+    const bool flush =
+      (subw_idx & 63u) == 7u || subw_idx == 1 || subw_idx == 3 ||
+      sub_hi >= range_hi;
+    if (flush) {
+      for (uint32_t b = lane; b < SDB_HIST_BINS; b += 64) {
+        const uint32_t v = hist[b];
+        if (v) {
+          atomicAdd(&gh[b], v);
+          hist[b] = 0;
+        }
+      }
+      // derive the threshold bin from the global suffix counts (same
+      // proof as the WG kernels; see score_bin)
+      const uint32_t known_bin = __hip_atomic_load(
+        a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      uint32_t part = 0;
+      {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
+          if (b >= known_bin) {
+#pragma unroll
+            for (int sh = 0; sh < 8; ++sh)
+              part += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                        __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT);
+          }
+        }
+      }
+      const uint32_t suff_incl = wave_incl_scan(part, lane);
+      const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
+      const bool winner =
+        suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
+      if (winner) {
+        uint32_t cum = suff_incl - part;
+        uint32_t binfloor = 0;
+        for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
+          uint32_t add = 0;
+#pragma unroll
+          for (int sh = 0; sh < 8; ++sh)
+            add += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+          cum += add;
+          if (cum >= a.k) {
+            binfloor = (uint32_t)b;
+            break;
+          }
+        }
+        if (binfloor > known_bin) atomicMax(a.gthresh, binfloor);
+      }
+      const uint32_t nt = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                            __HIP_MEMORY_SCOPE_AGENT);
+      tbin = nt > tbin ? nt : tbin;
+    } else if ((subw_idx & 15u) == 15u) {
+      const uint32_t nt = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                            __HIP_MEMORY_SCOPE_AGENT);
+      tbin = nt > tbin ? nt : tbin;
+    }
+
+    if (sub_hi >= range_hi) break;
+    sub_lo = sub_hi + 1u;
+  }
+  if (!overflowed && !seed_mode && lane == 0 && wv_matches)
+    atomicAdd(a.total_matches, wv_matches);
+}
+
 // column gather for the streaming scan: out[i] = col[docs[i]]
 __global__ void gather_col_kernel(const uint32_t* __restrict__ docs,
                                   const long long* __restrict__ col,
@@ -1684,7 +2098,7 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_cand_count, 4));
   CTX_CHECK(hipMalloc(&ctx->d_total_matches, 8));
   CTX_CHECK(hipMalloc(&ctx->d_gthresh, 4));
-  CTX_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 8));
+  CTX_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 16));
   CTX_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
   CTX_CHECK(hipMalloc(&ctx->d_overflow, 4));
   CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS *
@@ -1930,7 +2344,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   HIP_CHECK(hipMemsetAsync(ctx->d_cand_count, 0, 4, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_total_matches, 0, 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_gthresh, 0, 4, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(ctx->d_ghist, 0, 4 * SDB_HIST_BINS * 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(ctx->d_ghist, 0, 4 * SDB_HIST_BINS * 16,
+                           ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_overflow, 0, 4, ctx->stream));
   HIP_CHECK(hipMemsetAsync(ctx->d_buckets, 0, 8 * 2 * SDB_MAX_BUCKETS,
                             ctx->stream));
@@ -1958,7 +2373,17 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   // CountFast. SDB_SWEEP_GEOM=WDxNTH overrides the geometry; =0 disables.
   SweepGeom sgeom{24576, 1024};
   bool use_sweep = (plan->min_match <= 1) && !hybrid && !count_only;
+  // per-wave path: the headline shape (WAND keeps the sweep kernel's
+  // window-level bounds; SDB_TOPK_PATH=sweep|general overrides for A/B)
+  bool use_wave =
+    use_sweep && plan->nterms <= SDB_PW_TERMS && !plan->wand;
+  if (const char* e = getenv("SDB_TOPK_PATH")) {
+    if (!strcmp(e, "sweep")) use_wave = false;
+    if (!strcmp(e, "general")) { use_wave = false; use_sweep = false; }
+  }
   if (const char* e = getenv("SDB_SWEEP_GEOM")) {
+    // explicit sweep-geometry benching: takes the wave path out of play
+    use_wave = false;
     unsigned wd_ = 0, nth_ = 0;
     if (e[0] == '0' && !e[1]) {
       use_sweep = false;
@@ -2045,6 +2470,37 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.cand_cap = SDB_CAND_CAP;
     a.total_matches = ctx->d_total_matches;
     a.overflow = ctx->d_overflow;
+    if (use_wave) {
+      // per-wave kernel (no barriers): a strided-sample SEED launch locks
+      // a valid threshold bin first (its histogram lives in its own ghist
+      // shards so sampled docs never double-count), then the main grid
+      // tiles the doc space
+      const uint32_t nwaves = (uint32_t)std::min<uint64_t>(
+        4096, ((uint64_t)seg->hdr.doc_count + SDB_PW_SUBW - 1) /
+                SDB_PW_SUBW);
+      const uint32_t rdocs =
+        (uint32_t)(((uint64_t)seg->hdr.doc_count + nwaves - 1) / nwaves);
+      const size_t pw_lds = (size_t)(SDB_PW_NTH / 64) *
+                            SDB_PW_WAVE_LDS_BYTES;
+      if (s == 0 && k != 0xFFFFFFFFu && nwaves > 64) {
+        WindowArgs sa = a;
+        sa.ghist = ctx->d_ghist + 8 * SDB_HIST_BINS;  // seed-only shards
+        const uint32_t seed_waves = 256;
+        const uint32_t seed_docs = 4096;
+        const uint32_t seed_stride = (uint32_t)std::max<uint64_t>(
+          seed_docs, seg->hdr.doc_count / seed_waves);
+        hipLaunchKernelGGL(topk_wave_kernel,
+                           dim3((seed_waves + 3) / 4), dim3(SDB_PW_NTH),
+                           pw_lds, ctx->stream, sa, d_tslot, seed_waves,
+                           seed_docs, seed_stride, /*seed=*/1);
+        HIP_CHECK(hipGetLastError());
+      }
+      hipLaunchKernelGGL(topk_wave_kernel, dim3((nwaves + 3) / 4),
+                         dim3(SDB_PW_NTH), pw_lds, ctx->stream, a,
+                         d_tslot, nwaves, rdocs, rdocs, /*seed=*/0);
+      HIP_CHECK(hipGetLastError());
+      continue;
+    }
     if (use_sweep) {
       a.dcache_n = s_dcache_n;
       const uint32_t nwin = (seg->hdr.doc_count + sgeom.wd - 1) / sgeom.wd;

@@ -61,7 +61,10 @@ def main():
       seg = ctx.load_segment(blob)
       lib = sa.gpu()
       for geom in args.geoms.split(","):
-          os.environ["SDB_SWEEP_GEOM"] = geom
+          if geom == "wave":
+              os.environ.pop("SDB_SWEEP_GEOM", None)
+          else:
+              os.environ["SDB_SWEEP_GEOM"] = geom
           for _ in range(args.warmup):
               hits, total = ctx.execute_topk([seg], term_idx, boosts, args.k,
                                              wand=args.wand)
