@@ -1744,6 +1744,18 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
                                     __HIP_MEMORY_SCOPE_AGENT);
   uint32_t subw_idx = 0;
   bool overflowed = false;
+#ifdef SDB_TIMING
+  unsigned long long pw_acc[6] = {0, 0, 0, 0, 0, 0};
+  long long pw_mark = clock64();
+#define SDB_TW(idx)                                          \
+  if (lane == 0) {                                           \
+    const long long now_ = clock64();                        \
+    pw_acc[idx] += (unsigned long long)(now_ - pw_mark);     \
+    pw_mark = now_;                                          \
+  }
+#else
+#define SDB_TW(idx)
+#endif
 
   for (uint32_t sub_lo = range_lo; sub_lo <= range_hi;
        ++subw_idx) {
@@ -1765,15 +1777,31 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
           st[t].cov = range_hi;
           break;
         }
+        // issue the NEXT descriptor's load under this block's decode
+        SdbBlockDesc dn{};
+        const bool have_dn = st[t].cur + 1 < st[t].dend;
+        if (have_dn) dn = a.desc[st[t].cur + 1];
         const uint32_t added =
           pw_fill_block(a, st[t], d, lane, range_lo, range_hi,
                         rings + t * SDB_PW_RING, scratch);
         st[t].ring_cnt += added;
         st[t].cov = min(d.last_doc, range_hi);
         ++st[t].cur;
+        while (have_dn && st[t].cov < sub_hi &&
+               st[t].ring_cnt + 128 <= SDB_PW_RING &&
+               dn.prev_doc < range_hi) {
+          const uint32_t add2 =
+            pw_fill_block(a, st[t], dn, lane, range_lo, range_hi,
+                          rings + t * SDB_PW_RING, scratch);
+          st[t].ring_cnt += add2;
+          st[t].cov = min(dn.last_doc, range_hi);
+          ++st[t].cur;
+          break;  // one carried block, then re-enter the outer loop
+        }
       }
       if (st[t].cov < sub_hi) sub_hi = st[t].cov;  // ring-full clamp
     }
+    SDB_TW(0)
     const uint32_t sub_len = sub_hi - sub_lo + 1u;
     const uint32_t nwords = (sub_len + 63u) / 64u;
 
@@ -1784,6 +1812,7 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       for (uint32_t i = lane; i < (sub_len + 3u) / 4u; i += 64) sw4[i] = z;
       for (uint32_t i = lane; i < nwords; i += 64) mwin[i] = 0ull;
     }
+    SDB_TW(4)
 
     // term-major scatter from the rings (fixed merge order: bit-exact)
 #pragma unroll
@@ -1823,6 +1852,7 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       st[t].ring_head += n;
       st[t].ring_cnt -= n;
     }
+    SDB_TW(1)
 
     // live-doc mask (general shift; sub_lo is arbitrary here)
     if (a.live) {
@@ -1835,18 +1865,29 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       }
     }
 
-    // sweep: matches, histogram, candidate emission
+    // sweep: matches, histogram, candidate emission. Each mask word is
+    // split across 4 lanes (lane&15 selects the word, lane>>4 its 16-bit
+    // quarter) so all 64 lanes share the serial bit walks; the histogram
+    // is sampled 1/4 of sub-windows (a subset of true counts keeps the
+    // threshold derivation valid, as in the WG kernels) plus the first
+    // few so the bound can move early.
+    const bool histing = seed_mode || (subw_idx & 3u) == 0 || subw_idx < 8;
     uint32_t my_matches = 0, my_emit = 0;
-    unsigned long long word = lane < (int)nwords ? mwin[lane] : 0ull;
+    const uint32_t wq = (uint32_t)lane >> 4;   // quarter 0..3
+    const uint32_t wi = (uint32_t)lane & 15u;  // word 0..15
+    unsigned long long word =
+      wi < nwords
+        ? (mwin[wi] >> (16u * wq)) & 0xFFFFull
+        : 0ull;
     my_matches = (uint32_t)__popcll(word);
     {
       unsigned long long w2 = word;
       while (w2) {
-        const uint32_t off =
-          64u * lane + (uint32_t)__ffsll((long long)w2) - 1u;
+        const uint32_t off = 64u * wi + 16u * wq +
+                             (uint32_t)__ffsll((long long)w2) - 1u;
         w2 &= w2 - 1;
         const uint32_t sb = score_bin(swin[off], inv_smax);
-        atomicAdd(&hist[sb], 1u);
+        if (histing) atomicAdd(&hist[sb], 1u);
         my_emit += (!seed_mode && sb >= tbin) ? 1u : 0u;
       }
     }
@@ -1869,8 +1910,8 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
         uint32_t pos = cbase + incl - my_emit;
         unsigned long long w3 = word;
         while (w3) {
-          const uint32_t off =
-            64u * lane + (uint32_t)__ffsll((long long)w3) - 1u;
+          const uint32_t off = 64u * wi + 16u * wq +
+                               (uint32_t)__ffsll((long long)w3) - 1u;
           w3 &= w3 - 1;
           const float sc = swin[off];
           if (score_bin(sc, inv_smax) >= tbin) {
@@ -1883,6 +1924,7 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       }
     }
 
+    SDB_TW(2)
     // histogram flush + threshold refresh/derive on a cadence (early and
     // then sparse; all wave-local, no synchronization)
     const bool flush =
@@ -1945,11 +1987,17 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       tbin = nt > tbin ? nt : tbin;
     }
 
+    SDB_TW(3)
     if (sub_hi >= range_hi) break;
     sub_lo = sub_hi + 1u;
   }
   if (!overflowed && !seed_mode && lane == 0 && wv_matches)
     atomicAdd(a.total_matches, wv_matches);
+#ifdef SDB_TIMING
+  if (lane == 0 && !seed_mode)
+    for (int i = 0; i < 6; ++i) atomicAdd(&a.bucket_out[i], pw_acc[i]);
+#endif
+#undef SDB_TW
 }
 
 // column gather for the streaming scan: out[i] = col[docs[i]]
