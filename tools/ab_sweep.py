@@ -53,7 +53,10 @@ def main():
         libspecs.append((nm, pth))
     for libname, libpath in libspecs:
       if libpath is None:
-          os.environ.pop("SDB_GPU_LIB", None)
+          # leave an externally exported SDB_GPU_LIB alone unless this
+          # run itself is comparing libs
+          if len(libspecs) > 1:
+              os.environ.pop("SDB_GPU_LIB", None)
       else:
           os.environ["SDB_GPU_LIB"] = libpath
       sa._gpu = None  # re-resolve the GPU library for this arm

@@ -40,7 +40,8 @@ def run_pass(counter, tag, script):
     outdir = os.path.join(REPO, "gpurun_out", f"pmc_{tag}_{counter}")
     os.makedirs(outdir, exist_ok=True)
     env = dict(os.environ, TMPDIR="/tmp")
-    cmd = ["rocprofv3", "--pmc", counter, "-d", outdir, "-o", "t", "--",
+    cmd = ["rocprofv3", "--pmc", counter, "-d", outdir, "-o", "t",
+           "--output-format", "csv", "--",
            sys.executable, "-c", script]
     subprocess.run(cmd, check=True, cwd="/tmp", env=env,
                    stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
