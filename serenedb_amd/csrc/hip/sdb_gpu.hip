@@ -1592,10 +1592,14 @@ struct PwState {  // per-(wave,term) wave-uniform state
 
 // wave-cooperative: decode block `d` of term state `st`, score, and append
 // postings inside [range_lo, range_hi] to the ring. Returns appended count.
-__device__ __forceinline__ uint32_t pw_fill_block(
+__device__ __noinline__ uint32_t pw_fill_block(
   const WindowArgs& a, PwState& st, const SdbBlockDesc& d, int lane,
   uint32_t range_lo, uint32_t range_hi, unsigned long long* ring,
   uint32_t* scratch) {
+  // NOINLINE: the per-term loops would otherwise instantiate this ~500-
+  // instruction body 4-8x and blow the instruction cache — the measured
+  // 15k cycles/block fill cost (gpurun_out/r2_wavetime_1b.log) is I$-
+  // refetch, not memory or issue.
   const uint8_t* pl = a.payload + st.pbase;
   uint32_t doc0 = 0, doc1 = 0;
   float s0 = 0.f, s1 = 0.f;
@@ -1683,6 +1687,59 @@ __device__ __forceinline__ uint32_t pw_fill_block(
     ring[pos & (SDB_PW_RING - 1)] =
       ((unsigned long long)__float_as_uint(s1) << 32) | doc1;
   return total;
+}
+
+// flush the per-wave histogram into the global shards and derive the
+// threshold bin from the global suffix counts (same proof as the WG
+// kernels; see score_bin)
+__device__ __noinline__ void pw_flush_derive(const WindowArgs& a,
+                                             uint32_t* hist, uint32_t* gh,
+                                             int lane, uint32_t& tbin) {
+  for (uint32_t b = lane; b < SDB_HIST_BINS; b += 64) {
+    const uint32_t v = hist[b];
+    if (v) {
+      atomicAdd(&gh[b], v);
+      hist[b] = 0;
+    }
+  }
+  const uint32_t known_bin = __hip_atomic_load(
+    a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  uint32_t part = 0;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
+    if (b >= known_bin) {
+#pragma unroll
+      for (int sh = 0; sh < 8; ++sh)
+        part += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                  __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
+  const uint32_t suff_incl = wave_incl_scan(part, lane);
+  const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
+  const bool winner = suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
+  if (winner) {
+    uint32_t cum = suff_incl - part;
+    uint32_t binfloor = 0;
+    for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
+      uint32_t add = 0;
+#pragma unroll
+      for (int sh = 0; sh < 8; ++sh)
+        add += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
+                                 __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT);
+      cum += add;
+      if (cum >= a.k) {
+        binfloor = (uint32_t)b;
+        break;
+      }
+    }
+    if (binfloor > known_bin) atomicMax(a.gthresh, binfloor);
+  }
+  const uint32_t nt = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT);
+  tbin = nt > tbin ? nt : tbin;
 }
 
 __launch_bounds__(SDB_PW_NTH, 2) __global__
@@ -1777,27 +1834,12 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
           st[t].cov = range_hi;
           break;
         }
-        // issue the NEXT descriptor's load under this block's decode
-        SdbBlockDesc dn{};
-        const bool have_dn = st[t].cur + 1 < st[t].dend;
-        if (have_dn) dn = a.desc[st[t].cur + 1];
         const uint32_t added =
           pw_fill_block(a, st[t], d, lane, range_lo, range_hi,
                         rings + t * SDB_PW_RING, scratch);
         st[t].ring_cnt += added;
         st[t].cov = min(d.last_doc, range_hi);
         ++st[t].cur;
-        while (have_dn && st[t].cov < sub_hi &&
-               st[t].ring_cnt + 128 <= SDB_PW_RING &&
-               dn.prev_doc < range_hi) {
-          const uint32_t add2 =
-            pw_fill_block(a, st[t], dn, lane, range_lo, range_hi,
-                          rings + t * SDB_PW_RING, scratch);
-          st[t].ring_cnt += add2;
-          st[t].cov = min(dn.last_doc, range_hi);
-          ++st[t].cur;
-          break;  // one carried block, then re-enter the outer loop
-        }
       }
       if (st[t].cov < sub_hi) sub_hi = st[t].cov;  // ring-full clamp
     }
@@ -1926,61 +1968,13 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
 
     SDB_TW(2)
     // histogram flush + threshold refresh/derive on a cadence (early and
-    // then sparse; all wave-local, no synchronization)
+    // then sparse; all wave-local, no synchronization; noinline keeps the
+    // cold body out of the hot loop's instruction footprint)
     const bool flush =
       (subw_idx & 63u) == 7u || subw_idx == 1 || subw_idx == 3 ||
       sub_hi >= range_hi;
     if (flush) {
-      for (uint32_t b = lane; b < SDB_HIST_BINS; b += 64) {
-        const uint32_t v = hist[b];
-        if (v) {
-          atomicAdd(&gh[b], v);
-          hist[b] = 0;
-        }
-      }
-      // derive the threshold bin from the global suffix counts (same
-      // proof as the WG kernels; see score_bin)
-      const uint32_t known_bin = __hip_atomic_load(
-        a.gthresh, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      uint32_t part = 0;
-      {
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const uint32_t b = SDB_HIST_BINS - 4 * lane - 4 + j;
-          if (b >= known_bin) {
-#pragma unroll
-            for (int sh = 0; sh < 8; ++sh)
-              part += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
-                                        __ATOMIC_RELAXED,
-                                        __HIP_MEMORY_SCOPE_AGENT);
-          }
-        }
-      }
-      const uint32_t suff_incl = wave_incl_scan(part, lane);
-      const uint32_t suff_prev = __shfl_up(suff_incl, 1, 64);
-      const bool winner =
-        suff_incl >= a.k && (lane == 0 || suff_prev < a.k);
-      if (winner) {
-        uint32_t cum = suff_incl - part;
-        uint32_t binfloor = 0;
-        for (int b = (int)(SDB_HIST_BINS - 1 - 4 * lane);; --b) {
-          uint32_t add = 0;
-#pragma unroll
-          for (int sh = 0; sh < 8; ++sh)
-            add += __hip_atomic_load(&a.ghist[sh * SDB_HIST_BINS + b],
-                                     __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-          cum += add;
-          if (cum >= a.k) {
-            binfloor = (uint32_t)b;
-            break;
-          }
-        }
-        if (binfloor > known_bin) atomicMax(a.gthresh, binfloor);
-      }
-      const uint32_t nt = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
-                                            __HIP_MEMORY_SCOPE_AGENT);
-      tbin = nt > tbin ? nt : tbin;
+      pw_flush_derive(a, hist, gh, lane, tbin);
     } else if ((subw_idx & 15u) == 15u) {
       const uint32_t nt = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
                                             __HIP_MEMORY_SCOPE_AGENT);
