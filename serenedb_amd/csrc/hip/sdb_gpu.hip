@@ -1592,14 +1592,10 @@ struct PwState {  // per-(wave,term) wave-uniform state
 
 // wave-cooperative: decode block `d` of term state `st`, score, and append
 // postings inside [range_lo, range_hi] to the ring. Returns appended count.
-__device__ __noinline__ uint32_t pw_fill_block(
+__device__ __forceinline__ uint32_t pw_fill_block(
   const WindowArgs& a, PwState& st, const SdbBlockDesc& d, int lane,
   uint32_t range_lo, uint32_t range_hi, unsigned long long* ring,
   uint32_t* scratch) {
-  // NOINLINE: the per-term loops would otherwise instantiate this ~500-
-  // instruction body 4-8x and blow the instruction cache — the measured
-  // 15k cycles/block fill cost (gpurun_out/r2_wavetime_1b.log) is I$-
-  // refetch, not memory or issue.
   const uint8_t* pl = a.payload + st.pbase;
   uint32_t doc0 = 0, doc1 = 0;
   float s0 = 0.f, s1 = 0.f;
@@ -1643,6 +1639,18 @@ __device__ __noinline__ uint32_t pw_fill_block(
       have01 = true;
     }
   }
+#ifdef SDB_PW_ABLATE_NOLOAD
+  // perf ablation: synthetic postings, no payload/desc-dependent loads
+  {
+    const uint32_t span = d.last_doc - d.prev_doc;
+    const uint32_t i0 = 2u * (uint32_t)lane;
+    doc0 = d.prev_doc + 1 + (uint32_t)(((uint64_t)i0 * span) / 128u);
+    doc1 = d.prev_doc + 1 + (uint32_t)(((uint64_t)(i0 + 1) * span) / 128u);
+    s0 = score_one(a.scorer, st.num, st.nc, st.nl, 1 + (i0 & 7), 100);
+    s1 = score_one(a.scorer, st.num, st.nc, st.nl, 1 + ((i0 + 1) & 7), 100);
+    have01 = true;
+  }
+#endif
   if (!have01) {
     // generic families through the per-wave scratch (no cross-wave use)
     uint32_t* dbuf = scratch;
@@ -1692,7 +1700,7 @@ __device__ __noinline__ uint32_t pw_fill_block(
 // flush the per-wave histogram into the global shards and derive the
 // threshold bin from the global suffix counts (same proof as the WG
 // kernels; see score_bin)
-__device__ __noinline__ void pw_flush_derive(const WindowArgs& a,
+__device__ void pw_flush_derive(const WindowArgs& a,
                                              uint32_t* hist, uint32_t* gh,
                                              int lane, uint32_t& tbin) {
   for (uint32_t b = lane; b < SDB_HIST_BINS; b += 64) {
@@ -1860,6 +1868,30 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
 #pragma unroll
     for (uint32_t t = 0; t < SDB_PW_TERMS; ++t) {
       if (t >= a.nterms) continue;
+#ifdef SDB_PW_ABLATE_NOSCATTER
+      // perf ablation: drop ring contents without accumulating
+      {
+        unsigned long long* ring0 = rings + t * SDB_PW_RING;
+        uint32_t n = 0;
+        while (n < st[t].ring_cnt) {
+          const uint32_t m = min(64u, st[t].ring_cnt - n);
+          uint32_t doc = 0;
+          if ((uint32_t)lane < m)
+            doc = (uint32_t)ring0[(st[t].ring_head + n + lane) &
+                                  (SDB_PW_RING - 1)];
+          const unsigned long long gt =
+            __ballot((uint32_t)lane < m && doc > sub_hi);
+          if (gt) {
+            n += (uint32_t)__ffsll((long long)gt) - 1u;
+            break;
+          }
+          n += m;
+        }
+        st[t].ring_head += n;
+        st[t].ring_cnt -= n;
+        continue;
+      }
+#endif
       unsigned long long* ring = rings + t * SDB_PW_RING;
       // count entries with doc <= sub_hi (entries are doc-sorted)
       uint32_t n = 0;
