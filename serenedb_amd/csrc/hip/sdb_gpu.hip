@@ -150,6 +150,24 @@ __device__ __forceinline__ float score_one(uint32_t scorer, float num,
   return num - num * c1 / (c1 + (float)freq);
 }
 
+// Load up to three stream tag bytes through the VECTOR memory path and
+// broadcast. The pointers are wave-uniform, so a plain p[0] scalarizes
+// into an SMEM load — and SMEM shares the lgkmcnt FIFO with LDS ops, so
+// every later LDS-dependent wait in the same span drains behind a
+// ~900-cycle scalar fetch (the systemic stall measured on every kernel
+// this round). A lane-varying address forces global_load (vmcnt domain).
+__device__ __forceinline__ void load_tags3(const uint8_t* p0,
+                                           const uint8_t* p1,
+                                           const uint8_t* p2, int lane,
+                                           uint32_t* t0, uint32_t* t1,
+                                           uint32_t* t2) {
+  const uint8_t* tp = lane == 1 ? p1 : (lane == 2 ? p2 : p0);
+  const uint32_t v = tp[0];
+  *t0 = __shfl(v, 0, 64);
+  *t1 = __shfl(v, 1, 64);
+  *t2 = __shfl(v, 2, 64);
+}
+
 // vertical-layout delta extract: value i of a 128-block, b bits
 // (simdcomp d1 layout — see include/sdb_format.h)
 __device__ __forceinline__ uint32_t extract_packed(const uint8_t* base,
@@ -339,8 +357,9 @@ __device__ __forceinline__ bool try_block_fused(
   const uint8_t* db = pl + d.doc_off;
   const uint8_t* fb = pl + d.freq_off;
   const uint8_t* nb = fb + d.flags;
-  const uint32_t dtag = db[0], ftag = fb[0];
-  const uint32_t ntag = norm_stream ? nb[0] : SDB_E_BITPACK_01;
+  uint32_t dtag, ftag, ntag;
+  load_tags3(db, fb, nb, lane, &dtag, &ftag, &ntag);  // no SMEM (poison)
+  if (!norm_stream) ntag = SDB_E_BITPACK_01;
   if (dtag < SDB_DE_DELTA_BITPACK_02 || ftag < SDB_E_BITPACK_01 ||
       ntag < SDB_E_BITPACK_01)
     return false;
@@ -401,10 +420,20 @@ __device__ __forceinline__ bool try_block_fused2(
   const uint8_t* bdoc = pl + db_.doc_off;
   const uint8_t* bfrq = pl + db_.freq_off;
   const uint8_t* bnrm = bfrq + db_.flags;
-  const uint32_t adt = adoc[0], aft = afrq[0];
-  const uint32_t bdt = bdoc[0], bft = bfrq[0];
-  const uint32_t ant = norm_stream ? anrm[0] : SDB_E_BITPACK_01;
-  const uint32_t bnt = norm_stream ? bnrm[0] : SDB_E_BITPACK_01;
+  // six tag bytes through the vector path (see load_tags3)
+  const uint8_t* tp = lane == 1   ? afrq
+                      : lane == 2 ? anrm
+                      : lane == 3 ? bdoc
+                      : lane == 4 ? bfrq
+                      : lane == 5 ? bnrm
+                                  : adoc;
+  const uint32_t tv = tp[0];
+  const uint32_t adt = __shfl(tv, 0, 64), aft = __shfl(tv, 1, 64);
+  const uint32_t ant0 = __shfl(tv, 2, 64);
+  const uint32_t bdt = __shfl(tv, 3, 64), bft = __shfl(tv, 4, 64);
+  const uint32_t bnt0 = __shfl(tv, 5, 64);
+  const uint32_t ant = norm_stream ? ant0 : SDB_E_BITPACK_01;
+  const uint32_t bnt = norm_stream ? bnt0 : SDB_E_BITPACK_01;
   if (adt < SDB_DE_DELTA_BITPACK_02 || aft < SDB_E_BITPACK_01 ||
       ant < SDB_E_BITPACK_01 || bdt < SDB_DE_DELTA_BITPACK_02 ||
       bft < SDB_E_BITPACK_01 || bnt < SDB_E_BITPACK_01 || !norm_stream)
@@ -1091,11 +1120,17 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* shared_misc = hist + SDB_HIST_BINS;                 // 2 + NW
   uint32_t* cursors = shared_misc + 2 + NW;                     // max terms
   float* wub = (float*)(cursors + SDB_MAX_TERMS);               // max terms
-  SdbBlockDesc* dcache = (SdbBlockDesc*)(wub + SDB_MAX_TERMS);
+  TermDev* tstage = (TermDev*)(wub + SDB_MAX_TERMS);            // max terms
+  SdbBlockDesc* dcache = (SdbBlockDesc*)(tstage + SDB_MAX_TERMS);
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
   const uint32_t wave = tid >> 6;
+  // stage the term table through the vector path once: per-phase reads
+  // of the global term table are wave-uniform SMEM loads whose lgkmcnt
+  // sharing with LDS ops stalls every later LDS wait (see load_tags3)
+  for (uint32_t i = tid; i < a.nterms * (sizeof(TermDev) / 4); i += NTH)
+    ((uint32_t*)tstage)[i] = ((const uint32_t*)terms)[i];
   const float inv_smax = (float)SDB_HIST_BINS / a.smax;
   uint32_t* gh = a.ghist + (blockIdx.x & 7u) * SDB_HIST_BINS;
   unsigned long long wg_matches = 0;
@@ -1120,8 +1155,9 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 
   // one binary search per term per WORKGROUP (cursors then advance
   // monotonically window to window)
+  __syncthreads();  // term table staged
   if (tid < a.nterms) {
-    const TermDev te = terms[tid];
+    const TermDev te = tstage[tid];
     const uint32_t first_lo = 1u + w_lo * WD;
     cursors[tid] = (uint32_t)(
       lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, first_lo) -
@@ -1148,7 +1184,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       for (uint32_t i = tid; i < a.nterms * words_per_term; i += NTH) {
         const uint32_t t = i / words_per_term;
         const uint32_t wrd = i % words_per_term;
-        const TermDev te = terms[t];
+        const TermDev te = tstage[t];
         const uint64_t b0 = te.desc_begin + cursors[t];
         const uint32_t avail =
           (uint32_t)(te.desc_end > b0 ? te.desc_end - b0 : 0);
@@ -1162,7 +1198,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
                                          __HIP_MEMORY_SCOPE_AGENT);
     if (a.wand) __syncthreads();  // wub scan reads other threads' staging
     if (a.wand && tid < a.nterms) {
-      const TermDev te = terms[tid];
+      const TermDev te = tstage[tid];
       const uint32_t cur0w = cursors[tid];
       float ub = 0.0f;
       uint32_t i = 0;
@@ -1208,7 +1244,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // latency-bound: 62% of the kernel, SDB_TIMING r2).
     uint32_t pf_acc = 0;
     for (uint32_t t = 0; t < a.nterms; ++t) {
-      const TermDev te = terms[t];
+      const TermDev te = tstage[t];
       const uint8_t* pl = a.payload + te.payload_begin;
       const uint64_t dend = te.desc_end;
       uint32_t* dbuf = scratch + wave * 384;
@@ -1219,7 +1255,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       uint64_t b = te.desc_begin + cur0 + wave;
       uint32_t pf_next = 0;
       if (t + 1 < a.nterms) {
-        const TermDev tn = terms[t + 1];
+        const TermDev tn = tstage[t + 1];
         const uint32_t ncur = cursors[t + 1];
         const uint64_t nb = tn.desc_begin + ncur + wave;
         if (nb < tn.desc_end) {
@@ -1344,7 +1380,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     // advance every term's cursor once per window
     if (tid < a.nterms) {
       const uint32_t t = tid;
-      const TermDev te = terms[t];
+      const TermDev te = tstage[t];
       uint32_t cur = cursors[t];
       const uint32_t cur0 = cur;
       while (te.desc_begin + cur < te.desc_end) {
@@ -1367,7 +1403,7 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     uint32_t pf_w = 0;
     if (w + 1 < w_hi && wave < a.nterms) {
       const uint32_t t = wave;
-      const TermDev te = terms[t];
+      const TermDev te = tstage[t];
       const uint32_t ncur = cursors[t];  // post-advance (barrier above)
       const uint64_t nb = te.desc_begin + ncur;
       if (nb < te.desc_end) {
@@ -1575,14 +1611,17 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #define SDB_PW_NTH 256u    // threads per WG = 4 independent waves
 
 // per-wave LDS footprint (u64-aligned)
+#define SDB_PW_DSTAGE 8u  // LDS-staged descriptors per term per wave
 #define SDB_PW_WAVE_LDS_BYTES                                            \
   (SDB_PW_SUBW * 4 /*swin*/ + (SDB_PW_SUBW / 64) * 8 /*mask*/ +          \
    SDB_PW_TERMS * SDB_PW_RING * 8 /*rings*/ + 384 * 4 /*scratch*/ +      \
-   SDB_HIST_BINS * 4 /*hist*/)
+   SDB_HIST_BINS * 4 /*hist*/ +                                          \
+   SDB_PW_TERMS * SDB_PW_DSTAGE * 28 /*desc stage*/ + 32 /*pad*/)
 
 struct PwState {  // per-(wave,term) wave-uniform state
   uint64_t cur;       // next block (absolute desc index)
   uint64_t dend;      // term desc end
+  uint64_t dsbase;    // desc index of LDS stage slot 0 (~0 = none)
   uint64_t pbase;     // payload begin
   uint32_t ring_head; // logical head
   uint32_t ring_cnt;  // entries in ring
@@ -1604,8 +1643,9 @@ __device__ __forceinline__ uint32_t pw_fill_block(
     const uint8_t* db = pl + d.doc_off;
     const uint8_t* fb = pl + d.freq_off;
     const uint8_t* nb = fb + d.flags;
-    const uint32_t dtag = db[0], ftag = fb[0];
-    const uint32_t ntag = a.norm_stream ? nb[0] : SDB_E_BITPACK_01;
+    uint32_t dtag, ftag, ntag;
+    load_tags3(db, fb, nb, lane, &dtag, &ftag, &ntag);
+    if (!a.norm_stream) ntag = SDB_E_BITPACK_01;
     if (dtag >= SDB_DE_DELTA_BITPACK_02 && ftag >= SDB_E_BITPACK_01 &&
         ntag >= SDB_E_BITPACK_01) {
       const uint32_t dbits = dtag - SDB_DE_DELTA_BITPACK_02 + 2;
@@ -1767,6 +1807,10 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
   unsigned long long* rings = mwin + SDB_PW_SUBW / 64;
   uint32_t* scratch = (uint32_t*)(rings + SDB_PW_TERMS * SDB_PW_RING);
   uint32_t* hist = scratch + 384;
+  // descriptors staged through the VECTOR path: a direct a.desc[cur] read
+  // is wave-uniform and scalarizes to SMEM, whose lgkmcnt sharing with
+  // LDS ops was the kernel-wide stall (see load_tags3)
+  SdbBlockDesc* dstage = (SdbBlockDesc*)(hist + SDB_HIST_BINS);
 
   // wave's doc range: seed mode samples strided slices; main mode tiles
   // the doc space contiguously
@@ -1795,6 +1839,7 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
       st[t].nl = te.nl;
       st[t].cur =
         lower_bound_last_doc(a.desc, te.desc_begin, te.desc_end, range_lo);
+      st[t].dsbase = ~0ull;
       // cov invariant: every posting in (consumed, cov] sits in the ring;
       // nothing ringed yet
       st[t].cov = st[t].cur >= te.desc_end ? range_hi : range_lo - 1;
@@ -1836,7 +1881,19 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
           st[t].cov = range_hi;
           break;
         }
-        const SdbBlockDesc d = a.desc[st[t].cur];
+        if (st[t].cur < st[t].dsbase ||
+            st[t].cur >= st[t].dsbase + SDB_PW_DSTAGE) {
+          const uint32_t nst = (uint32_t)min(
+            (uint64_t)SDB_PW_DSTAGE, st[t].dend - st[t].cur);
+          const uint32_t words = nst * 7u;
+          if ((uint32_t)lane < words)
+            ((uint32_t*)(dstage + t * SDB_PW_DSTAGE))[lane] =
+              ((const uint32_t*)&a.desc[st[t].cur])[lane];
+          st[t].dsbase = st[t].cur;
+        }
+        const SdbBlockDesc d =
+          dstage[t * SDB_PW_DSTAGE +
+                 (uint32_t)(st[t].cur - st[t].dsbase)];
         if (d.prev_doc >= range_hi) {  // block fully beyond the range
           st[t].cur = st[t].dend;
           st[t].cov = range_hi;
@@ -2091,7 +2148,8 @@ bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
 // fixed (non-dcache) LDS bytes of one sweep workgroup
 size_t sweep_lds_fixed(const SweepGeom& g) {
   return (size_t)g.wd * 4 + g.wd / 8 + (g.nth / 64) * 1536 +
-         SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 + 2 * SDB_MAX_TERMS * 4;
+         SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 + 2 * SDB_MAX_TERMS * 4 +
+         sizeof(TermDev) * SDB_MAX_TERMS;
 }
 
 int check_gpu() {
