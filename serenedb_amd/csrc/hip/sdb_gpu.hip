@@ -1121,7 +1121,9 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   uint32_t* cursors = shared_misc + 2 + NW;                     // max terms
   float* wub = (float*)(cursors + SDB_MAX_TERMS);               // max terms
   TermDev* tstage = (TermDev*)(wub + SDB_MAX_TERMS);            // max terms
-  SdbBlockDesc* dcache = (SdbBlockDesc*)(tstage + SDB_MAX_TERMS);
+  unsigned long long* lbuck =
+    (unsigned long long*)(tstage + SDB_MAX_TERMS);  // 2 * max buckets
+  SdbBlockDesc* dcache = (SdbBlockDesc*)(lbuck + 2 * SDB_MAX_BUCKETS);
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
@@ -1177,6 +1179,8 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     for (uint32_t i = tid; i < NWORDS; i += NTH) mwin[i] = 0ull;
     if (derive)
       for (uint32_t i = tid; i < SDB_HIST_BINS; i += NTH) hist[i] = 0;
+    if (a.fcol)
+      for (uint32_t i = tid; i < 2 * a.nbuckets; i += NTH) lbuck[i] = 0;
     // stage this window's descriptors: term t's next dcache_n descs from
     // its cursor, as coalesced u32 reads
     {
@@ -1428,6 +1432,37 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         word &= lw;
         mwin[i] = word;  // recount/append walks see the masked word
       }
+      if (word && a.fcol) {
+        // hybrid: pushed column predicates narrow the matches
+        // (TableFilterDocIterator / ColFilterChain semantics); survivors
+        // feed the per-bucket COUNT/SUM
+        unsigned long long w0 = word;
+        while (w0) {
+          const uint32_t bit = (uint32_t)__ffsll((long long)w0) - 1u;
+          w0 &= w0 - 1;
+          const uint32_t doc = lo + 64u * i + bit;
+          const long long vv = a.fcol[doc];
+          bool pass = vv >= a.flo && vv <= a.fhi;
+          for (uint32_t x = 0; pass && x < a.nfx; ++x) {
+            const long long xv = a.fxc[x][doc];
+            if (a.fxop[x] == 1) pass = xv < a.fxlo[x];
+            else if (a.fxop[x] == 2) pass = xv >= a.fxlo[x];
+            else pass = (xv >= a.fxlo[x]) & (xv <= a.fxhi[x]);
+          }
+          if (!pass) {
+            word &= ~(1ull << bit);
+            continue;
+          }
+          const unsigned long long span =
+            (unsigned long long)(a.fhi - a.flo) + 1ull;
+          uint32_t bkt = (uint32_t)(
+            ((unsigned long long)(vv - a.flo) * a.nbuckets) / span);
+          if (bkt >= a.nbuckets) bkt = a.nbuckets - 1;
+          atomicAdd(&lbuck[2 * bkt], 1ull);
+          atomicAdd(&lbuck[2 * bkt + 1], (unsigned long long)vv);
+        }
+        mwin[i] = word;
+      }
       if (!word) continue;
       my_matches += (uint32_t)__popcll(word);
       const uint32_t base = i * 64u;
@@ -1450,6 +1485,10 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
     if (lane == 0) shared_misc[2 + wave] = wm;
     __syncthreads();
     SDB_TS(2)
+
+    if (a.fcol)
+      for (uint32_t i = tid; i < 2 * a.nbuckets; i += NTH)
+        if (lbuck[i]) atomicAdd(&a.bucket_out[i], lbuck[i]);
 
     // merge window histogram into the per-XCD global shard, derive the
     // threshold bin from the global suffix counts (identical to the
@@ -1606,11 +1645,16 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
 #ifndef SDB_PW_SUBW
 #define SDB_PW_SUBW 1024u  // docs per sub-window (per-wave LDS f32 window)
 #endif
-#define SDB_PW_RING 256u   // ring entries per term (2 full blocks)
 #define SDB_PW_TERMS 4u    // wave path handles <= 4 terms (headline shape)
 #define SDB_PW_NTH 256u    // threads per WG = 4 independent waves
 
 // per-wave LDS footprint (u64-aligned)
+#ifndef SDB_PW_RING
+#define SDB_PW_RING 256u   // ring entries per term (pow2; >= 128 + leftover)
+#endif
+#ifndef SDB_PW_MINW
+#define SDB_PW_MINW 2      // min waves/SIMD the kernel is compiled for
+#endif
 #define SDB_PW_DSTAGE 8u  // LDS-staged descriptors per term per wave
 #define SDB_PW_WAVE_LDS_BYTES                                            \
   (SDB_PW_SUBW * 4 /*swin*/ + (SDB_PW_SUBW / 64) * 8 /*mask*/ +          \
@@ -1790,7 +1834,7 @@ __device__ void pw_flush_derive(const WindowArgs& a,
   tbin = nt > tbin ? nt : tbin;
 }
 
-__launch_bounds__(SDB_PW_NTH, 2) __global__
+__launch_bounds__(SDB_PW_NTH, SDB_PW_MINW) __global__
 void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
                       uint32_t nwaves_total, uint32_t range_docs,
                       uint32_t range_stride, int seed_mode) {
@@ -2004,18 +2048,26 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
     // few so the bound can move early.
     const bool histing = seed_mode || (subw_idx & 3u) == 0 || subw_idx < 8;
     uint32_t my_matches = 0, my_emit = 0;
-    const uint32_t wq = (uint32_t)lane >> 4;   // quarter 0..3
-    const uint32_t wi = (uint32_t)lane & 15u;  // word 0..15
+    // each lane owns a K-bit slice of the whole window bitmask (K =
+    // SUBW/64 divides 64, so a slice never straddles words) — all 64
+    // lanes share the serial bit walks at every SUBW
+    constexpr uint32_t K = SDB_PW_SUBW / 64u;
+    static_assert(K >= 1 && K <= 64 && (64 % K) == 0,
+                  "lane slice must divide a mask word");
+    const uint32_t bit0 = (uint32_t)lane * K;
     unsigned long long word =
-      wi < nwords
-        ? (mwin[wi] >> (16u * wq)) & 0xFFFFull
+      bit0 < sub_len + 63u  // slices past nwords read zeroed words? no:
+        ? (bit0 >> 6) < nwords
+            ? (mwin[bit0 >> 6] >> (bit0 & 63u)) &
+                (K == 64 ? ~0ull : ((1ull << K) - 1ull))
+            : 0ull
         : 0ull;
     my_matches = (uint32_t)__popcll(word);
     {
       unsigned long long w2 = word;
       while (w2) {
-        const uint32_t off = 64u * wi + 16u * wq +
-                             (uint32_t)__ffsll((long long)w2) - 1u;
+        const uint32_t off =
+          bit0 + (uint32_t)__ffsll((long long)w2) - 1u;
         w2 &= w2 - 1;
         const uint32_t sb = score_bin(swin[off], inv_smax);
         if (histing) atomicAdd(&hist[sb], 1u);
@@ -2041,8 +2093,8 @@ void topk_wave_kernel(WindowArgs a, const TermDev* __restrict__ terms,
         uint32_t pos = cbase + incl - my_emit;
         unsigned long long w3 = word;
         while (w3) {
-          const uint32_t off = 64u * wi + 16u * wq +
-                               (uint32_t)__ffsll((long long)w3) - 1u;
+          const uint32_t off =
+            bit0 + (uint32_t)__ffsll((long long)w3) - 1u;
           w3 &= w3 - 1;
           const float sc = swin[off];
           if (score_bin(sc, inv_smax) >= tbin) {
@@ -2149,7 +2201,7 @@ bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
 size_t sweep_lds_fixed(const SweepGeom& g) {
   return (size_t)g.wd * 4 + g.wd / 8 + (g.nth / 64) * 1536 +
          SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 + 2 * SDB_MAX_TERMS * 4 +
-         sizeof(TermDev) * SDB_MAX_TERMS;
+         sizeof(TermDev) * SDB_MAX_TERMS + 16 * SDB_MAX_BUCKETS;
 }
 
 int check_gpu() {
@@ -2504,7 +2556,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   // general window kernel keeps min_match>1, hybrid column filters and
   // CountFast. SDB_SWEEP_GEOM=WDxNTH overrides the geometry; =0 disables.
   SweepGeom sgeom{24576, 1024};
-  bool use_sweep = (plan->min_match <= 1) && !hybrid && !count_only;
+  bool use_sweep = (plan->min_match <= 1) && !count_only;
   // per-wave path: the headline shape (WAND keeps the sweep kernel's
   // window-level bounds; SDB_TOPK_PATH=sweep|general overrides for A/B)
   bool use_wave =
