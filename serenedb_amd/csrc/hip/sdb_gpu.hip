@@ -2583,8 +2583,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   bool use_sweep = (plan->min_match <= 1) && !count_only;
   // per-wave path: the headline shape (WAND keeps the sweep kernel's
   // window-level bounds; SDB_TOPK_PATH=sweep|general overrides for A/B)
-  bool use_wave =
-    use_sweep && plan->nterms <= SDB_PW_TERMS && !plan->wand;
+  bool use_wave = use_sweep && !hybrid &&
+                  plan->nterms <= SDB_PW_TERMS && !plan->wand;
   if (const char* e = getenv("SDB_TOPK_PATH")) {
     if (!strcmp(e, "sweep")) use_wave = false;
     if (!strcmp(e, "general")) { use_wave = false; use_sweep = false; }
