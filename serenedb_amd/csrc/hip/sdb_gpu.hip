@@ -2581,11 +2581,16 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   // CountFast. SDB_SWEEP_GEOM=WDxNTH overrides the geometry; =0 disables.
   SweepGeom sgeom{24576, 1024};
   bool use_sweep = (plan->min_match <= 1) && !count_only;
-  // per-wave path: the headline shape (WAND keeps the sweep kernel's
-  // window-level bounds; SDB_TOPK_PATH=sweep|general overrides for A/B)
-  bool use_wave = use_sweep && !hybrid &&
-                  plan->nterms <= SDB_PW_TERMS && !plan->wand;
+  // The per-wave (barrier-free) kernel is an EXPERIMENT kept behind
+  // SDB_TOPK_PATH=wave: across every measured variant it trails the
+  // sweep kernel (r2 A/B logs under profiles/), parked ~78% of wave
+  // cycles on the per-block fetch chain at 8 waves/CU. The sweep kernel
+  // ships as the default top-k path.
+  bool use_wave = false;
   if (const char* e = getenv("SDB_TOPK_PATH")) {
+    if (!strcmp(e, "wave"))
+      use_wave = use_sweep && !hybrid && plan->nterms <= SDB_PW_TERMS &&
+                 !plan->wand;
     if (!strcmp(e, "sweep")) use_wave = false;
     if (!strcmp(e, "general")) { use_wave = false; use_sweep = false; }
   }

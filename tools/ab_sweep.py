@@ -66,7 +66,9 @@ def main():
       for geom in args.geoms.split(","):
           if geom == "wave":
               os.environ.pop("SDB_SWEEP_GEOM", None)
+              os.environ["SDB_TOPK_PATH"] = "wave"
           else:
+              os.environ.pop("SDB_TOPK_PATH", None)
               os.environ["SDB_SWEEP_GEOM"] = geom
           for _ in range(args.warmup):
               hits, total = ctx.execute_topk([seg], term_idx, boosts, args.k,
