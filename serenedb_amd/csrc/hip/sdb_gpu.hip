@@ -1197,29 +1197,10 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             ((const uint32_t*)&a.desc[b0])[wrd];
       }
     }
-    // cross-term payload prefetch: each wave touches its phase-entry
-    // block of EVERY term here, so the four phases' cold fetches overlap
-    // in ONE memory round trip instead of serializing one per phase
-    // (phases measured 62% of the kernel and latency-bound). Descriptor
-    // words come via a 7-lane vector load + shuffles (no SMEM — see
-    // load_tags3).
-    uint32_t pf0 = 0;
-    for (uint32_t t = 0; t < a.nterms; ++t) {
-      const TermDev te = tstage[t];
-      const uint64_t b = te.desc_begin + cursors[t] + wave;
-      if (b < te.desc_end) {
-        const uint32_t* dw = (const uint32_t*)&a.desc[b];
-        const uint32_t w7 = lane < 7 ? dw[lane] : 0u;
-        const uint32_t prevd = __shfl(w7, 0, 64);
-        const uint32_t doc_off = __shfl(w7, 2, 64);
-        if (prevd < hi) {
-          const uint32_t* pfp = (const uint32_t*)(
-            (uintptr_t)(a.payload + te.payload_begin + doc_off) &
-            ~(uintptr_t)3);
-          pf0 ^= pfp[lane] + pfp[lane + 64];
-        }
-      }
-    }
+    // (a window-start cross-term payload prefetch was tried here and
+    // REVERTED: +5% kernel time at 1B — the 7-lane desc read + shuffle
+    // chain and the extra VMEM pressure cost more than the overlap won;
+    // gpurun_out/r2_pf_1b.log)
     if (tid == 0)
       shared_misc[0] = __hip_atomic_load(a.gthresh, __ATOMIC_RELAXED,
                                          __HIP_MEMORY_SCOPE_AGENT);
@@ -1263,7 +1244,6 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       __builtin_memcpy(&gtw, &t0b, 4);
     }
 
-    asm volatile("" ::"v"(pf0));  // prefetch lands once, covers 4 phases
     // term-major phases (fixed fp32 merge order -> bit-exact vs oracle).
     // Cross-phase prefetch: before working term t, issue a non-blocking
     // touch of term t+1's first block payload for this wave; the value is
