@@ -153,3 +153,71 @@ def test_gloo_sharded_merge_equals_full(world, port):
     assert total == ftotal
     np.testing.assert_array_equal(docs.astype(np.uint32), fhits["doc"])
     np.testing.assert_array_equal(scores, fhits["score"])
+
+
+def _hash_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    rows_total, seed = 90_000, 33
+    per = rows_total // world
+    lo = rank * per
+    hi = rows_total if rank == world - 1 else (rank + 1) * per
+    rng = np.random.default_rng(seed)
+    base = rng.integers(-(1 << 60), 1 << 60, 5000).astype(np.int64)
+    keys = base[rng.integers(0, len(base), rows_total)][lo:hi]
+    vals = rng.integers(-1000, 1000, rows_total).astype(np.int64)[lo:hi]
+    # per-rank "hash aggregate" rows (what sdb_gpu_scan_agg_hash returns:
+    # keys ascending + per-key aggregates)
+    uk, inv = np.unique(keys, return_inverse=True)
+    cnt = np.bincount(inv, minlength=len(uk)).astype(np.int64)
+    s = np.zeros(len(uk), dtype=np.int64)
+    np.add.at(s, inv, vals)
+    # the N>1 merge: variable-size all_gather (gloo object collective),
+    # concat + groupby on every rank — the SURVEY §8e "gather + host
+    # merge" path for open hash tables
+    rows = [None] * world
+    dist.all_gather_object(rows, (uk, cnt, s))
+    allk = np.concatenate([r[0] for r in rows])
+    allc = np.concatenate([r[1] for r in rows])
+    alls = np.concatenate([r[2] for r in rows])
+    mk, minv = np.unique(allk, return_inverse=True)
+    mc = np.zeros(len(mk), dtype=np.int64)
+    ms = np.zeros(len(mk), dtype=np.int64)
+    np.add.at(mc, minv, allc)
+    np.add.at(ms, minv, alls)
+    if rank == 0:
+        q.put((mk, mc, ms))
+    dist.destroy_process_group()
+
+
+def test_gloo_hash_agg_merge_equals_full():
+    """Distributed merge of per-rank HASH-aggregate tables (arbitrary
+    sparse i64 keys -> gather + merge by key) equals the
+    single-node aggregation."""
+    rows_total, seed = 90_000, 33
+    rng = np.random.default_rng(seed)
+    base = rng.integers(-(1 << 60), 1 << 60, 5000).astype(np.int64)
+    keys = base[rng.integers(0, len(base), rows_total)]
+    vals = rng.integers(-1000, 1000, rows_total).astype(np.int64)
+    uk, inv = np.unique(keys, return_inverse=True)
+    fc = np.bincount(inv, minlength=len(uk)).astype(np.int64)
+    fs = np.zeros(len(uk), dtype=np.int64)
+    np.add.at(fs, inv, vals)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_hash_worker, args=(r, 2, 29531, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        mk, mc, ms = q.get(timeout=300)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    np.testing.assert_array_equal(mk, uk)
+    np.testing.assert_array_equal(mc, fc)
+    np.testing.assert_array_equal(ms, fs)
