@@ -156,6 +156,17 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                          SdbScoreDoc* hits, uint32_t* out_count,
                          uint64_t* total_matches);
 
+/* Pipelined batch: nq queries of the SAME plan, each fully re-executed
+ * (nothing cached between queries); query q+1's kernels overlap query
+ * q's candidate readback and exact host select — the production QPS
+ * shape of RunTopKScan's worker loop (duckdb_search_full_scan.cpp:
+ * 1925-2000). hits: nq*k entries; out_counts/totals: nq entries.
+ * Results of every query are identical to sdb_gpu_execute_topk's. */
+int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                               uint32_t nsegs, const SdbQueryPlan* plan,
+                               uint32_t k, uint32_t nq, SdbScoreDoc* hits,
+                               uint32_t* out_counts, uint64_t* totals);
+
 /* CountFast — exact match count without scoring (docs-only decode;
  * DecideScanMode Count/CountFast, duckdb_search_full_scan.cpp:972). */
 int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,

@@ -315,6 +315,7 @@ def gpu():
             "sdb_gpu_scan_agg_hash",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
             "sdb_gpu_segment_attach_livemask",
+            "sdb_gpu_execute_topk_batch",
             "sdb_gpu_execute_match_docs", "sdb_gpu_execute_count",
         ):
             getattr(lib, f).restype = C.c_int
@@ -758,3 +759,39 @@ def ingest_doc(doc_file, term_metas, doc_count, norms=None, has_freq=True):
     out = _copy_blob(blob, size.value)
     h.sdb_host_blob_free(blob)
     return out
+
+    def execute_topk_batch(self, segs, term_idx, boosts, k, nq,
+                           min_match=1, k1=1.2, b=0.75, global_stats=None,
+                           scorer="bm25", wand=False, filter_boost=False,
+                           all_hits=False):
+        """Pipelined batch of nq identical-plan queries (each fully
+        re-executed); returns (hits — list of nq structured arrays when
+        all_hits else just the last query's array — and totals [nq])."""
+        import numpy as np
+
+        plan, keep = self._make_plan(term_idx, boosts, min_match, k1, b,
+                                     global_stats, scorer, wand,
+                                     filter_boost)
+        seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value)
+                                             for s in segs])
+        hits = (SdbScoreDoc * (nq * k))()
+        counts = (C.c_uint32 * nq)()
+        totals = (C.c_uint64 * nq)()
+        rc = self._lib.sdb_gpu_execute_topk_batch(
+            self._ctx, seg_arr, C.c_uint32(len(segs)), C.byref(plan),
+            C.c_uint32(k), C.c_uint32(nq), hits, counts, totals)
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_topk_batch rc={rc}")
+        del keep
+
+        def conv(q):
+            n = counts[q]
+            res = np.zeros(n, dtype=[("score", "f4"), ("doc", "u4"),
+                                     ("segment", "u4")])
+            for i in range(n):
+                h = hits[q * k + i]
+                res[i] = (h.score, h.doc, h.segment_idx)
+            return res
+
+        out = ([conv(q) for q in range(nq)] if all_hits else conv(nq - 1))
+        return out, [int(totals[q]) for q in range(nq)]

@@ -64,6 +64,7 @@
 #define SDB_DESC_CACHE 32u  // staged descriptors per term per window
 #define SDB_MAX_BUCKETS 128u  // hybrid group-by buckets
 #define SDB_CAND_CAP (64u * 1024u * 1024u)  // 64M candidates (768 MB)
+#define SDB_PIN_CANDS (1u << 21)  // pinned candidate staging (24 MB)
 
 #define HIP_CHECK(x)                        \
   do {                                      \
@@ -2250,6 +2251,12 @@ const char* sdb_gpu_version(void) { return "sdb_gpu 0.1 gfx950"; }
 // free whatever a partially-constructed context owns (error paths)
 static void ctx_free_partial(SdbGpuCtx* ctx) {
   if (ctx->d_cands) (void)hipFree(ctx->d_cands);
+  if (ctx->d_cands2) (void)hipFree(ctx->d_cands2);
+  if (ctx->d_ghist2) (void)hipFree(ctx->d_ghist2);
+  if (ctx->d_qmisc) (void)hipFree(ctx->d_qmisc);
+  if (ctx->h_qmisc) (void)hipHostFree(ctx->h_qmisc);
+  if (ctx->h_cands_pin) (void)hipHostFree(ctx->h_cands_pin);
+  if (ctx->copy_stream) (void)hipStreamDestroy(ctx->copy_stream);
   if (ctx->d_cand_count) (void)hipFree(ctx->d_cand_count);
   if (ctx->d_total_matches) (void)hipFree(ctx->d_total_matches);
   if (ctx->d_gthresh) (void)hipFree(ctx->d_gthresh);
@@ -2282,6 +2289,7 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   auto* ctx = new SdbGpuCtx{};
   ctx->device = device;
   CTX_CHECK(hipStreamCreate(&ctx->stream));
+  CTX_CHECK(hipStreamCreate(&ctx->copy_stream));
   CTX_CHECK(hipMalloc(&ctx->d_cands, sizeof(SdbScoreDoc) * (size_t)SDB_CAND_CAP));
   CTX_CHECK(hipMalloc(&ctx->d_cand_count, 4));
   CTX_CHECK(hipMalloc(&ctx->d_total_matches, 8));
@@ -2295,12 +2303,32 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipHostMalloc(&ctx->h_matches, 8));
   CTX_CHECK(hipEventCreate(&ctx->ev_a));
   CTX_CHECK(hipEventCreate(&ctx->ev_b));
+  // pipelined-batch state: a SECOND query-state set so query q+1's
+  // kernels enqueue while q's results are read back and selected; a
+  // dedicated copy stream keeps D2H reads off the kernel stream
+  CTX_CHECK(hipMalloc(&ctx->d_cands2, sizeof(SdbScoreDoc) *
+                                        (size_t)SDB_CAND_CAP));
+  CTX_CHECK(hipMalloc(&ctx->d_ghist2, 4 * SDB_HIST_BINS * 16));
+  CTX_CHECK(hipMalloc(&ctx->d_qmisc, 2 * 64));
+  CTX_CHECK(hipHostMalloc(&ctx->h_qmisc, 2 * 64));
+  CTX_CHECK(hipHostMalloc(&ctx->h_cands_pin, SDB_PIN_CANDS *
+                                               sizeof(SdbScoreDoc)));
+  CTX_CHECK(hipEventCreate(&ctx->ev_q[0]));
+  CTX_CHECK(hipEventCreate(&ctx->ev_q[1]));
   *out = ctx;
   return SDB_OK;
 }
 
 int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   if (!ctx) return SDB_ERR_INVALID;
+  (void)hipFree(ctx->d_cands2);
+  (void)hipFree(ctx->d_ghist2);
+  (void)hipFree(ctx->d_qmisc);
+  (void)hipHostFree(ctx->h_qmisc);
+  (void)hipHostFree(ctx->h_cands_pin);
+  (void)hipEventDestroy(ctx->ev_q[0]);
+  (void)hipEventDestroy(ctx->ev_q[1]);
+  (void)hipStreamDestroy(ctx->copy_stream);
   (void)hipFree(ctx->d_cands);
   (void)hipFree(ctx->d_cand_count);
   (void)hipFree(ctx->d_total_matches);
@@ -2410,6 +2438,101 @@ int sdb_gpu_segment_attach_livemask(SdbGpuCtx* ctx, SdbGpuSegment* seg,
   return SDB_OK;
 }
 
+struct PlanStats {
+  float num[SDB_MAX_TERMS];
+  float nc, nl, smax, fbmax;
+};
+
+// PreparePhase analogue: global stats (double -> f32, bm25.cpp:288-306);
+// shared by the single-query and the pipelined batch executors.
+static int prep_plan_stats(SdbGpuSegment* const* segs, uint32_t nsegs,
+                           const SdbQueryPlan* plan, int hybrid,
+                           PlanStats* out);
+
+static int prep_plan_stats(SdbGpuSegment* const* segs, uint32_t nsegs,
+                           const SdbQueryPlan* plan, int hybrid,
+                           PlanStats* out) {
+  (void)hybrid;
+  float fbmax = 1.0f;  // filter boost: every segment needs the column
+  if (plan->filter_boost) {
+    fbmax = 0.0f;
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      if (!segs[s]->fboost) return SDB_ERR_INVALID;
+      fbmax = segs[s]->fboost_max > fbmax ? segs[s]->fboost_max : fbmax;
+    }
+    if (!(fbmax > 0.0f)) return SDB_ERR_INVALID;
+  }
+  uint64_t g_dwf = plan->g_docs_with_field;
+  uint64_t g_ttf = plan->g_total_term_freq;
+  std::vector<uint64_t> g_dwt(plan->nterms, 0);
+  if (g_dwf == 0) {
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      g_dwf += segs[s]->hdr.docs_with_field;
+      g_ttf += segs[s]->hdr.total_term_freq;
+      for (uint32_t t = 0; t < plan->nterms; ++t) {
+        const uint32_t ti = plan->terms[t].term_idx;
+        if (ti >= segs[s]->hdr.nterms) return SDB_ERR_INVALID;
+        g_dwt[t] += segs[s]->terms_host[ti].df;
+      }
+    }
+  } else {
+    if (!plan->g_docs_with_term) return SDB_ERR_INVALID;
+    for (uint32_t t = 0; t < plan->nterms; ++t)
+      g_dwt[t] = plan->g_docs_with_term[t];
+  }
+  const float k1 = plan->k1, b = plan->b;
+  const uint32_t scorer = plan->scorer;
+  float nc = 0.0f, nl = 0.0f;
+  float smax = 0.0f;
+  for (uint32_t t = 0; t < plan->nterms; ++t) {
+    if (g_dwt[t] == 0) {
+      out->num[t] = 0.0f;
+      continue;
+    }
+    uint32_t term_max_freq = 1;
+    for (uint32_t s = 0; s < nsegs; ++s) {
+      const uint32_t ti = plan->terms[t].term_idx;
+      term_max_freq =
+        std::max(term_max_freq, segs[s]->terms_host[ti].max_freq);
+    }
+    if (scorer == SDB_SCORER_BM25) {
+      // BM25::collect (bm25.cpp:288-306)
+      const float idf = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
+                                     ((double)g_dwt[t] + 0.5));
+      out->num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf;
+      // BM1 (k == 0, bm25.cpp:112-140): without a filter boost every
+      // score is 0; matches still count
+      if (k1 == 0.0f && !plan->filter_boost) out->num[t] = 0.0f;
+      smax += out->num[t] > 0 ? out->num[t] : 0.0f;
+    } else {
+      // TFIDF::collect (tfidf.cpp:148-151)
+      const float idf = (float)log1p(((double)g_dwf + 1.0) /
+                                     ((double)g_dwt[t] + 1.0));
+      out->num[t] = plan->terms[t].boost * idf;
+      const float ub = out->num[t] * sqrtf((float)term_max_freq);
+      smax += ub > 0 ? ub : 0.0f;
+    }
+  }
+  if (scorer == SDB_SCORER_BM25) {
+    const float kb = k1 * b;
+    if (b == 0.0f) {
+      nc = k1;
+      nl = 0.0f;
+    } else {
+      nc = k1 - kb;
+      nl = (g_ttf && g_dwf) ? kb / ((float)g_ttf / (float)g_dwf) : kb;
+    }
+  }
+  if (plan->filter_boost) smax *= fbmax;  // scores reach fb*base
+  // smax <= 0: all scores are 0; 1.0 keeps inv_smax finite (bin 0)
+  if (smax <= 0.0f) smax = 1.0f;
+  out->nc = nc;
+  out->nl = nl;
+  out->smax = smax;
+  out->fbmax = plan->filter_boost ? fbmax : 1.0f;
+  return SDB_OK;
+}
+
 static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                           uint32_t nsegs, const SdbQueryPlan* plan,
                           uint32_t k, int hybrid,
@@ -2440,93 +2563,20 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     for (uint32_t j = i + 1; j < plan->nterms; ++j)  // count match tallies
       if (plan->terms[i].term_idx == plan->terms[j].term_idx)
         return SDB_ERR_INVALID;
-    // negative (or NaN) boosts break every non-negative-score assumption:
-    // the histogram binning, the bin-threshold monotonicity argument and
-    // the WAND upper bounds
+    // negative (or NaN) boosts break every non-negative-score assumption
+    // 
     if (!(plan->terms[i].boost >= 0.0f)) return SDB_ERR_INVALID;
   }
-  float fbmax = 1.0f;  // filter boost: every segment needs the column
-  if (plan->filter_boost) {
-    fbmax = 0.0f;
-    for (uint32_t s = 0; s < nsegs; ++s) {
-      if (!segs[s]->fboost) return SDB_ERR_INVALID;
-      fbmax = segs[s]->fboost_max > fbmax ? segs[s]->fboost_max : fbmax;
-    }
-    if (!(fbmax > 0.0f)) return SDB_ERR_INVALID;
+  PlanStats ps;
+  {
+    const int prc = prep_plan_stats(segs, nsegs, plan, hybrid, &ps);
+    if (prc) return prc;
   }
-
-  // ---- PreparePhase analogue: global stats (double -> f32, bm25.cpp) ----
-  uint64_t g_dwf = plan->g_docs_with_field;
-  uint64_t g_ttf = plan->g_total_term_freq;
-  std::vector<uint64_t> g_dwt(plan->nterms, 0);
-  if (g_dwf == 0) {
-    for (uint32_t s = 0; s < nsegs; ++s) {
-      g_dwf += segs[s]->hdr.docs_with_field;
-      g_ttf += segs[s]->hdr.total_term_freq;
-      for (uint32_t t = 0; t < plan->nterms; ++t) {
-        const uint32_t ti = plan->terms[t].term_idx;
-        if (ti >= segs[s]->hdr.nterms) return SDB_ERR_INVALID;
-        g_dwt[t] += segs[s]->terms_host[ti].df;
-      }
-    }
-  } else {
-    if (!plan->g_docs_with_term) return SDB_ERR_INVALID;
-    for (uint32_t t = 0; t < plan->nterms; ++t)
-      g_dwt[t] = plan->g_docs_with_term[t];
-  }
-  const float k1 = plan->k1, b = plan->b;
   const uint32_t scorer = plan->scorer;
-  std::vector<float> idf(plan->nterms), num(plan->nterms);
-  float nc = 0.0f, nl = 0.0f;
-  float smax = 0.0f;
-  for (uint32_t t = 0; t < plan->nterms; ++t) {
-    if (g_dwt[t] == 0) {
-      idf[t] = 0.0f;
-      num[t] = 0.0f;
-      continue;
-    }
-    uint32_t term_max_freq = 1;
-    for (uint32_t s = 0; s < nsegs; ++s) {
-      const uint32_t ti = plan->terms[t].term_idx;
-      term_max_freq =
-        std::max(term_max_freq, segs[s]->terms_host[ti].max_freq);
-    }
-    if (scorer == SDB_SCORER_BM25) {
-      // BM25::collect (bm25.cpp:288-306)
-      idf[t] = (float)log1p(((double)(g_dwf - g_dwt[t]) + 0.5) /
-                            ((double)g_dwt[t] + 0.5));
-      num[t] = plan->terms[t].boost * (k1 + 1.0f) * idf[t];
-      // BM1 (k == 0, bm25.cpp:112-140 Bm1Score + :333-336): without a
-      // filter boost every score is 0, so no hit beats the collector's
-      // FLT_MIN threshold; matches still count. WITH a filter boost the
-      // score is fb*num (Bm1Boost), which c1 == 0 already yields.
-      if (k1 == 0.0f && !plan->filter_boost) num[t] = 0.0f;
-      smax += num[t] > 0 ? num[t] : 0.0f;
-    } else {
-      // TFIDF::collect (tfidf.cpp:148-151)
-      idf[t] = (float)log1p(((double)g_dwf + 1.0) /
-                            ((double)g_dwt[t] + 1.0));
-      num[t] = plan->terms[t].boost * idf[t];
-      // score <= num*sqrt(max_freq) (norm >= 1 only helps the bound)
-      const float ub = num[t] * sqrtf((float)term_max_freq);
-      smax += ub > 0 ? ub : 0.0f;
-    }
-  }
-  if (scorer == SDB_SCORER_BM25) {
-    const float kb = k1 * b;
-    if (b == 0.0f) {
-      nc = k1;
-      nl = 0.0f;
-    } else {
-      nc = k1 - kb;
-      nl = (g_ttf && g_dwf) ? kb / ((float)g_ttf / (float)g_dwf) : kb;
-    }
-  }
-  if (plan->filter_boost) smax *= fbmax;  // scores reach fb*base
-  // smax <= 0 means every score is 0 (nothing beats the > FLT_MIN
-  // acceptance); 1.0 keeps inv_smax finite so 0-scores bin to 0 cleanly
-  // (256/FLT_MIN would be +inf and 0*inf = NaN in the binning)
-  if (smax <= 0.0f) smax = 1.0f;
+  const float* num = ps.num;
+  const float nc = ps.nc, nl = ps.nl;
+  float smax = ps.smax;
+  const float fbmax = ps.fbmax;
 
   // ---- reset device state ----
   HIP_CHECK(hipMemsetAsync(ctx->d_cand_count, 0, 4, ctx->stream));
@@ -2818,6 +2868,280 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   return exec_topk_impl(ctx, segs, nsegs, plan, k, 0, nullptr, 0, 0,
                         nullptr,
                         nullptr, hits, out_count, total_matches);
+}
+
+/* Pipelined batch execution: nq queries of the SAME plan, each fully
+ * re-executed (decode -> score -> threshold -> exact select; nothing is
+ * cached between queries). Two query-state sets alternate so query q+1's
+ * kernels are already enqueued while q's candidates are read back on a
+ * separate copy stream and selected on the host — the production QPS
+ * shape of RunTopKScan, where worker threads keep the executor busy
+ * while results drain (duckdb_search_full_scan.cpp:1925-2000).
+ * hits: nq*k entries; out_counts/totals: nq entries. Non-hybrid,
+ * min_match semantics as execute_topk. */
+int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                               uint32_t nsegs, const SdbQueryPlan* plan,
+                               uint32_t k, uint32_t nq, SdbScoreDoc* hits,
+                               uint32_t* out_counts, uint64_t* totals) {
+  if (!ctx || !segs || !plan || !hits || !out_counts || !totals ||
+      nq == 0 || plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS ||
+      k == 0 || k == 0xFFFFFFFFu)
+    return SDB_ERR_INVALID;
+  for (uint32_t i = 0; i < plan->nterms; ++i) {
+    for (uint32_t j = i + 1; j < plan->nterms; ++j)
+      if (plan->terms[i].term_idx == plan->terms[j].term_idx)
+        return SDB_ERR_INVALID;
+    if (!(plan->terms[i].boost >= 0.0f)) return SDB_ERR_INVALID;
+  }
+  PlanStats ps;
+  {
+    const int prc = prep_plan_stats(segs, nsegs, plan, 0, &ps);
+    if (prc) return prc;
+  }
+  const float smax = ps.smax;
+  const float inv_smax = (float)SDB_HIST_BINS / smax;
+
+  // path/geometry decisions identical to exec_topk_impl
+  SweepGeom sgeom{24576, 1024};
+  bool use_sweep = plan->min_match <= 1;
+  bool use_wave = false;
+  if (const char* e = getenv("SDB_TOPK_PATH")) {
+    if (!strcmp(e, "wave"))
+      use_wave = use_sweep && plan->nterms <= SDB_PW_TERMS && !plan->wand;
+    if (!strcmp(e, "general")) use_sweep = false;
+  }
+  if (const char* e = getenv("SDB_SWEEP_GEOM")) {
+    use_wave = false;
+    unsigned wd_ = 0, nth_ = 0;
+    if (e[0] == '0' && !e[1]) use_sweep = false;
+    else if (sscanf(e, "%ux%u", &wd_, &nth_) == 2 && wd_ >= 64 &&
+             nth_ >= 64 && wd_ % 64 == 0 && nth_ % 64 == 0) {
+      sgeom.wd = wd_;
+      sgeom.nth = nth_;
+    }
+  }
+  const size_t lds_fixed_gen =
+    SDB_WIN_DOCS * sizeof(float) + SDB_WIN_DOCS + SDB_NWAVES * 384 * 4 +
+    SDB_HIST_BINS * 4 + (2 + SDB_NWAVES + 2 * SDB_MAX_TERMS) * 4 +
+    sizeof(TermDev) * SDB_MAX_TERMS + 8 * 2 * SDB_MAX_BUCKETS;
+  uint32_t gen_dcache_n = SDB_DESC_CACHE;
+  {
+    const size_t room = 160 * 1024 - lds_fixed_gen;
+    const uint32_t fit =
+      (uint32_t)(room / (sizeof(SdbBlockDesc) * plan->nterms));
+    if (fit < gen_dcache_n) gen_dcache_n = fit;
+  }
+  uint32_t s_dcache_n = SDB_DESC_CACHE;
+  size_t s_lds = 0;
+  uint32_t s_wgs_per_cu = 1;
+  if (use_sweep) {
+    const size_t fixed = sweep_lds_fixed(sgeom);
+    s_lds = fixed + sizeof(SdbBlockDesc) * s_dcache_n * plan->nterms;
+    const uint32_t wave_cap = 32u / (sgeom.nth / 64u);
+    s_wgs_per_cu = (uint32_t)(163840 / s_lds);
+    if (s_wgs_per_cu > wave_cap) s_wgs_per_cu = wave_cap;
+    if (s_wgs_per_cu == 0) {
+      while (s_dcache_n > 1 && s_lds > 163840) {
+        --s_dcache_n;
+        s_lds = fixed + sizeof(SdbBlockDesc) * s_dcache_n * plan->nterms;
+      }
+      if (s_lds > 163840) use_sweep = false;
+      s_wgs_per_cu = 1;
+    }
+  }
+
+  // per-set device state inside d_qmisc (64 B apart):
+  //  +0 gthresh(u32) +8 cand_count(u32) +16 overflow(u32) +24 total(u64)
+  auto set_ptrs = [&](int qs, WindowArgs& a) {
+    unsigned char* base = ctx->d_qmisc + 64 * qs;
+    a.gthresh = (uint32_t*)(base + 0);
+    a.cand_count = (uint32_t*)(base + 8);
+    a.overflow = (uint32_t*)(base + 16);
+    a.total_matches = (unsigned long long*)(base + 24);
+    a.ghist = qs ? ctx->d_ghist2 : ctx->d_ghist;
+    a.cands = qs ? ctx->d_cands2 : ctx->d_cands;
+  };
+
+  double total_kernel_ms = 0.0;
+  HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+
+  auto enqueue = [&](uint32_t q) -> int {
+    const int qs = (int)(q & 1u);
+    HIP_CHECK(hipMemsetAsync(ctx->d_qmisc + 64 * qs, 0, 64, ctx->stream));
+    HIP_CHECK(hipMemsetAsync(qs ? ctx->d_ghist2 : ctx->d_ghist, 0,
+                             4 * SDB_HIST_BINS * 16, ctx->stream));
+    for (uint32_t sg = 0; sg < nsegs; ++sg) {
+      SdbGpuSegment* seg = segs[sg];
+      const uint32_t slot = (q * nsegs + sg) % SDB_TERM_SLOTS;
+      if ((q * nsegs + sg) >= SDB_TERM_SLOTS && slot == 0)
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      TermDev* d_tslot = ctx->d_terms + (size_t)slot * SDB_MAX_TERMS;
+      TermDev tdev[SDB_MAX_TERMS];
+      for (uint32_t t = 0; t < plan->nterms; ++t) {
+        const SdbTermEntry& te = seg->terms_host[plan->terms[t].term_idx];
+        tdev[t].desc_begin = te.desc_begin;
+        tdev[t].desc_end = te.desc_end;
+        tdev[t].payload_begin = te.payload_begin;
+        tdev[t].num = ps.num[t];
+        tdev[t].nc = ps.nc;
+        tdev[t].nl = ps.nl;
+      }
+      HIP_CHECK(hipMemcpyAsync(d_tslot, tdev,
+                               sizeof(TermDev) * plan->nterms,
+                               hipMemcpyHostToDevice, ctx->stream));
+      WindowArgs a{};
+      a.desc = seg->desc;
+      a.payload = seg->payload;
+      a.norms = seg->norms;
+      a.doc_count = seg->hdr.doc_count;
+      a.scorer = plan->scorer;
+      a.wand = (plan->wand && plan->min_match <= 1) ? 1u : 0u;
+      a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
+      a.nterms = plan->nterms;
+      a.min_match = plan->min_match ? plan->min_match : 1;
+      a.k = k;
+      a.smax = smax;
+      a.seg_idx = sg;
+      a.cand_cap = SDB_CAND_CAP;
+      a.nbuckets = 0;
+      a.bucket_out = ctx->d_buckets;
+      a.fb = plan->filter_boost ? seg->fboost : nullptr;
+      a.fbmax = ps.fbmax;
+      a.live = seg->live;
+      set_ptrs(qs, a);
+      if (use_wave) {
+        const uint32_t nwaves = (uint32_t)std::min<uint64_t>(
+          4096, ((uint64_t)seg->hdr.doc_count + SDB_PW_SUBW - 1) /
+                  SDB_PW_SUBW);
+        const uint32_t rdocs =
+          (uint32_t)(((uint64_t)seg->hdr.doc_count + nwaves - 1) / nwaves);
+        const size_t pw_lds =
+          (size_t)(SDB_PW_NTH / 64) * SDB_PW_WAVE_LDS_BYTES;
+        if (sg == 0 && nwaves > 64) {
+          WindowArgs sa = a;
+          sa.ghist = a.ghist + 8 * SDB_HIST_BINS;
+          hipLaunchKernelGGL(topk_wave_kernel, dim3(64), dim3(SDB_PW_NTH),
+                             pw_lds, ctx->stream, sa, d_tslot, 256, 4096,
+                             (uint32_t)std::max<uint64_t>(
+                               4096, seg->hdr.doc_count / 256),
+                             1);
+          HIP_CHECK(hipGetLastError());
+        }
+        hipLaunchKernelGGL(topk_wave_kernel, dim3((nwaves + 3) / 4),
+                           dim3(SDB_PW_NTH), pw_lds, ctx->stream, a,
+                           d_tslot, nwaves, rdocs, rdocs, 0);
+        HIP_CHECK(hipGetLastError());
+        continue;
+      }
+      if (use_sweep) {
+        a.dcache_n = s_dcache_n;
+        const uint32_t nwin =
+          (seg->hdr.doc_count + sgeom.wd - 1) / sgeom.wd;
+        uint32_t ngrid = 256u * s_wgs_per_cu;
+        if (ngrid > nwin) ngrid = nwin;
+        if (!launch_sweep(sgeom, dim3(ngrid), s_lds, ctx->stream, a,
+                          d_tslot))
+          return SDB_ERR_INVALID;
+        HIP_CHECK(hipGetLastError());
+        continue;
+      }
+      a.dcache_n = gen_dcache_n;
+      const uint32_t nwin =
+        (seg->hdr.doc_count + SDB_WIN_DOCS - 1) / SDB_WIN_DOCS;
+      uint32_t wgs_per_cu =
+        (uint32_t)(163840 /
+                   (lds_fixed_gen +
+                    sizeof(SdbBlockDesc) * gen_dcache_n * plan->nterms));
+      if (wgs_per_cu < 1) wgs_per_cu = 1;
+      if (wgs_per_cu > 4) wgs_per_cu = 4;
+      uint32_t ngrid = 256u * wgs_per_cu;
+      if (ngrid > nwin) ngrid = nwin;
+      hipLaunchKernelGGL(topk_window_kernel, dim3(ngrid),
+                         dim3(SDB_NTHREADS),
+                         lds_fixed_gen +
+                           sizeof(SdbBlockDesc) * gen_dcache_n *
+                             plan->nterms,
+                         ctx->stream, a, d_tslot);
+      HIP_CHECK(hipGetLastError());
+    }
+    // readback the tiny per-query state to its pinned mirror, then mark
+    HIP_CHECK(hipMemcpyAsync(ctx->h_qmisc + 64 * qs,
+                             ctx->d_qmisc + 64 * qs, 64,
+                             hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHECK(hipEventRecord(ctx->ev_q[qs], ctx->stream));
+    return SDB_OK;
+  };
+
+  auto harvest = [&](uint32_t q) -> int {
+    const int qs = (int)(q & 1u);
+    HIP_CHECK(hipEventSynchronize(ctx->ev_q[qs]));
+    const unsigned char* m = ctx->h_qmisc + 64 * qs;
+    uint32_t final_bin, ncand, ovf;
+    uint64_t total;
+    std::memcpy(&final_bin, m + 0, 4);
+    std::memcpy(&ncand, m + 8, 4);
+    std::memcpy(&ovf, m + 16, 4);
+    std::memcpy(&total, m + 24, 8);
+    if (ovf) return SDB_ERR_OOM;
+    SdbScoreDoc* csrc = qs ? ctx->d_cands2 : ctx->d_cands;
+    std::vector<SdbScoreDoc> heap_c;
+    SdbScoreDoc* cands;
+    if (ncand <= SDB_PIN_CANDS) {
+      HIP_CHECK(hipMemcpyAsync(ctx->h_cands_pin, csrc,
+                               sizeof(SdbScoreDoc) * ncand,
+                               hipMemcpyDeviceToHost, ctx->copy_stream));
+      HIP_CHECK(hipStreamSynchronize(ctx->copy_stream));
+      cands = ctx->h_cands_pin;
+    } else {
+      heap_c.resize(ncand);
+      HIP_CHECK(hipMemcpyAsync(heap_c.data(), csrc,
+                               sizeof(SdbScoreDoc) * ncand,
+                               hipMemcpyDeviceToHost, ctx->copy_stream));
+      HIP_CHECK(hipStreamSynchronize(ctx->copy_stream));
+      cands = heap_c.data();
+    }
+    // exact final select (same rule as exec_topk_impl)
+    size_t n = 0;
+    for (size_t i = 0; i < ncand; ++i) {
+      const float sc = cands[i].score;
+      uint32_t sb = (uint32_t)(sc * inv_smax);
+      if (sb >= SDB_HIST_BINS) sb = SDB_HIST_BINS - 1;
+      if (sb >= final_bin && sc > FLT_MIN) cands[n++] = cands[i];
+    }
+    auto cmp = [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
+      if (x.score != y.score) return x.score > y.score;
+      if (x.segment_idx != y.segment_idx)
+        return x.segment_idx < y.segment_idx;
+      return x.doc < y.doc;
+    };
+    const size_t kk = std::min<size_t>(k, n);
+    if (kk < n) std::nth_element(cands, cands + kk, cands + n, cmp);
+    std::sort(cands, cands + kk, cmp);
+    std::copy(cands, cands + kk, hits + (size_t)q * k);
+    out_counts[q] = (uint32_t)kk;
+    totals[q] = total;
+    return SDB_OK;
+  };
+
+  int rc = enqueue(0);
+  if (rc) return rc;
+  for (uint32_t q = 1; q < nq; ++q) {
+    rc = enqueue(q);
+    if (rc) return rc;
+    rc = harvest(q - 1);
+    if (rc) return rc;
+  }
+  rc = harvest(nq - 1);
+  if (rc) return rc;
+  HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  {
+    float ms = 0.0f;
+    HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_a, ctx->ev_b));
+    total_kernel_ms = (double)ms;
+  }
+  ctx->last_kernel_ms = total_kernel_ms / nq;
+  return SDB_OK;
 }
 
 // CountFast: exact match count without scoring (docs-only decode —

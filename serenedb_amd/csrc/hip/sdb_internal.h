@@ -27,6 +27,16 @@ struct SdbGpuCtx {
   uint32_t* h_counts;  // pinned: [cand_count, overflow]
   unsigned long long* h_matches;
   hipEvent_t ev_a, ev_b;   // bracket the window kernels of one execute call
+  // pipelined-batch state (sdb_gpu_execute_topk_batch): second query-state
+  // set + dedicated copy stream + pinned readback
+  hipStream_t copy_stream;
+  SdbScoreDoc* d_cands2;
+  uint32_t* d_ghist2;
+  unsigned char* d_qmisc;   // 2 x 64 B: {gthresh u32, pad, cand_count u32,
+                            //  overflow u32, total u64} per set
+  unsigned char* h_qmisc;   // pinned mirror
+  SdbScoreDoc* h_cands_pin; // pinned candidate staging (SDB_PIN_CANDS)
+  hipEvent_t ev_q[2];
   double last_kernel_ms;   // read back via sdb_gpu_last_kernel_ms
   // breakdown of the last execute_topk (sdb_gpu_last_stats)
   unsigned last_ncand;

@@ -1082,3 +1082,27 @@ def test_livemask_parity(ctx):
     ohits, ototal = po.execute_topk([blob], [0, 1, 2], [1.0] * 3, 100)
     assert total == ototal
     np.testing.assert_array_equal(hits["doc"], ohits["doc"])
+
+
+def test_batch_pipelined_equals_single(ctx):
+    """sdb_gpu_execute_topk_batch (pipelined QPS shape) must return, for
+    EVERY query in the batch, exactly what the single-query entry
+    returns — nothing is cached or shared between queries beyond the
+    resident segment."""
+    blob, _, _ = make_corpus(61, 400_000, [0.1, 0.05, 0.02, 0.01])
+    seg = ctx.load_segment(blob)
+    single, stotal = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, 777)
+    batch, btotals = ctx.execute_topk_batch([seg], [0, 1, 2, 3], [1.0] * 4,
+                                            777, 5, all_hits=True)
+    assert btotals == [stotal] * 5
+    for q in range(5):
+        np.testing.assert_array_equal(batch[q]["doc"], single["doc"])
+        np.testing.assert_array_equal(batch[q]["score"].view(np.uint32),
+                                      single["score"].view(np.uint32))
+    # min_match / AND plans route through the general kernel in batch too
+    s2, t2 = ctx.execute_topk([seg], [0, 1], [1.0] * 2, 50, min_match=2)
+    b2, bt2 = ctx.execute_topk_batch([seg], [0, 1], [1.0] * 2, 50, 3,
+                                     min_match=2, all_hits=True)
+    assert bt2 == [t2] * 3
+    np.testing.assert_array_equal(b2[0]["doc"], s2["doc"])
+    np.testing.assert_array_equal(b2[2]["doc"], s2["doc"])
