@@ -724,42 +724,6 @@ class GpuContext:
             passed.value
 
 
-def ingest_doc(doc_file, term_metas, doc_count, norms=None, has_freq=True):
-    """Read a reference-format `.doc` postings file (+ sidecar term metas:
-    list of dicts with docs_count, total_freq, doc_start, e_single_doc,
-    e_skip_start) and rebuild it as a segment blob. norms: uint32
-    [doc_count+1] or None."""
-    import numpy as np
-
-    class _TM(C.Structure):
-        _fields_ = [("docs_count", C.c_uint32), ("total_freq", C.c_uint64),
-                    ("doc_start", C.c_uint64), ("e_single_doc", C.c_uint32),
-                    ("e_skip_start", C.c_uint64)]
-
-    n = len(term_metas)
-    tm = (_TM * n)(*[_TM(m["docs_count"], m.get("total_freq", 0),
-                         m.get("doc_start", 0), m.get("e_single_doc", 0),
-                         m.get("e_skip_start", 0)) for m in term_metas])
-    buf = np.frombuffer(bytes(doc_file), dtype=np.uint8)
-    norm_ptr = None
-    if norms is not None:
-        norms = np.ascontiguousarray(norms, dtype=np.uint32)
-        norm_ptr = norms.ctypes.data_as(C.POINTER(C.c_uint32))
-    blob = C.c_void_p(0)
-    size = C.c_uint64(0)
-    h = host()
-    h.sdb_host_ingest_doc.restype = C.c_int
-    rc = h.sdb_host_ingest_doc(
-        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)), tm,
-        C.c_uint32(n), C.c_uint32(doc_count),
-        C.c_uint32(1 if has_freq else 0), norm_ptr,
-        C.byref(blob), C.byref(size))
-    if rc != 0:
-        raise ValueError(f"sdb_host_ingest_doc rc={rc}")
-    out = _copy_blob(blob, size.value)
-    h.sdb_host_blob_free(blob)
-    return out
-
     def execute_topk_batch(self, segs, term_idx, boosts, k, nq,
                            min_match=1, k1=1.2, b=0.75, global_stats=None,
                            scorer="bm25", wand=False, filter_boost=False,
@@ -795,3 +759,40 @@ def ingest_doc(doc_file, term_metas, doc_count, norms=None, has_freq=True):
 
         out = ([conv(q) for q in range(nq)] if all_hits else conv(nq - 1))
         return out, [int(totals[q]) for q in range(nq)]
+
+
+def ingest_doc(doc_file, term_metas, doc_count, norms=None, has_freq=True):
+    """Read a reference-format `.doc` postings file (+ sidecar term metas:
+    list of dicts with docs_count, total_freq, doc_start, e_single_doc,
+    e_skip_start) and rebuild it as a segment blob. norms: uint32
+    [doc_count+1] or None."""
+    import numpy as np
+
+    class _TM(C.Structure):
+        _fields_ = [("docs_count", C.c_uint32), ("total_freq", C.c_uint64),
+                    ("doc_start", C.c_uint64), ("e_single_doc", C.c_uint32),
+                    ("e_skip_start", C.c_uint64)]
+
+    n = len(term_metas)
+    tm = (_TM * n)(*[_TM(m["docs_count"], m.get("total_freq", 0),
+                         m.get("doc_start", 0), m.get("e_single_doc", 0),
+                         m.get("e_skip_start", 0)) for m in term_metas])
+    buf = np.frombuffer(bytes(doc_file), dtype=np.uint8)
+    norm_ptr = None
+    if norms is not None:
+        norms = np.ascontiguousarray(norms, dtype=np.uint32)
+        norm_ptr = norms.ctypes.data_as(C.POINTER(C.c_uint32))
+    blob = C.c_void_p(0)
+    size = C.c_uint64(0)
+    h = host()
+    h.sdb_host_ingest_doc.restype = C.c_int
+    rc = h.sdb_host_ingest_doc(
+        buf.ctypes.data_as(C.c_void_p), C.c_uint64(len(buf)), tm,
+        C.c_uint32(n), C.c_uint32(doc_count),
+        C.c_uint32(1 if has_freq else 0), norm_ptr,
+        C.byref(blob), C.byref(size))
+    if rc != 0:
+        raise ValueError(f"sdb_host_ingest_doc rc={rc}")
+    out = _copy_blob(blob, size.value)
+    h.sdb_host_blob_free(blob)
+    return out
