@@ -733,9 +733,8 @@ class GpuContext:
         all_hits else just the last query's array — and totals [nq])."""
         import numpy as np
 
-        plan, keep = self._make_plan(term_idx, boosts, min_match, k1, b,
-                                     global_stats, scorer, wand,
-                                     filter_boost)
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b,
+                               global_stats, scorer, wand, filter_boost)
         seg_arr = (C.c_void_p * len(segs))(*[C.c_void_p(s.value)
                                              for s in segs])
         hits = (SdbScoreDoc * (nq * k))()
@@ -746,7 +745,6 @@ class GpuContext:
             C.c_uint32(k), C.c_uint32(nq), hits, counts, totals)
         if rc != 0:
             raise RuntimeError(f"sdb_gpu_execute_topk_batch rc={rc}")
-        del keep
 
         def conv(q):
             n = counts[q]
