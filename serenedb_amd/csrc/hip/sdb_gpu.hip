@@ -2309,8 +2309,8 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_cands2, sizeof(SdbScoreDoc) *
                                         (size_t)SDB_CAND_CAP));
   CTX_CHECK(hipMalloc(&ctx->d_ghist2, 4 * SDB_HIST_BINS * 16));
-  CTX_CHECK(hipMalloc(&ctx->d_qmisc, 2 * 64));
-  CTX_CHECK(hipHostMalloc(&ctx->h_qmisc, 2 * 64));
+  CTX_CHECK(hipMalloc(&ctx->d_qmisc, 2 * 512));
+  CTX_CHECK(hipHostMalloc(&ctx->h_qmisc, 2 * 512));
   CTX_CHECK(hipHostMalloc(&ctx->h_cands_pin, SDB_PIN_CANDS *
                                                sizeof(SdbScoreDoc)));
   CTX_CHECK(hipEventCreate(&ctx->ev_q[0]));
@@ -2950,14 +2950,16 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     }
   }
 
-  // per-set device state inside d_qmisc (64 B apart):
-  //  +0 gthresh(u32) +8 cand_count(u32) +16 overflow(u32) +24 total(u64)
+  // per-set device state inside d_qmisc. Each atomically-updated word
+  // sits on its OWN 128 B line: the round-1 zero-workspace experiment
+  // showed gthresh/cand_count sharing a line costs measurably (the line
+  // ping-pongs between XCDs on every atomic).
   auto set_ptrs = [&](int qs, WindowArgs& a) {
-    unsigned char* base = ctx->d_qmisc + 64 * qs;
+    unsigned char* base = ctx->d_qmisc + 512 * qs;
     a.gthresh = (uint32_t*)(base + 0);
-    a.cand_count = (uint32_t*)(base + 8);
-    a.overflow = (uint32_t*)(base + 16);
-    a.total_matches = (unsigned long long*)(base + 24);
+    a.cand_count = (uint32_t*)(base + 128);
+    a.overflow = (uint32_t*)(base + 256);
+    a.total_matches = (unsigned long long*)(base + 384);
     a.ghist = qs ? ctx->d_ghist2 : ctx->d_ghist;
     a.cands = qs ? ctx->d_cands2 : ctx->d_cands;
   };
@@ -2967,7 +2969,8 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
 
   auto enqueue = [&](uint32_t q) -> int {
     const int qs = (int)(q & 1u);
-    HIP_CHECK(hipMemsetAsync(ctx->d_qmisc + 64 * qs, 0, 64, ctx->stream));
+    HIP_CHECK(hipMemsetAsync(ctx->d_qmisc + 512 * qs, 0, 512,
+                             ctx->stream));
     HIP_CHECK(hipMemsetAsync(qs ? ctx->d_ghist2 : ctx->d_ghist, 0,
                              4 * SDB_HIST_BINS * 16, ctx->stream));
     for (uint32_t sg = 0; sg < nsegs; ++sg) {
@@ -3065,8 +3068,8 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       HIP_CHECK(hipGetLastError());
     }
     // readback the tiny per-query state to its pinned mirror, then mark
-    HIP_CHECK(hipMemcpyAsync(ctx->h_qmisc + 64 * qs,
-                             ctx->d_qmisc + 64 * qs, 64,
+    HIP_CHECK(hipMemcpyAsync(ctx->h_qmisc + 512 * qs,
+                             ctx->d_qmisc + 512 * qs, 512,
                              hipMemcpyDeviceToHost, ctx->stream));
     HIP_CHECK(hipEventRecord(ctx->ev_q[qs], ctx->stream));
     return SDB_OK;
@@ -3075,13 +3078,13 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   auto harvest = [&](uint32_t q) -> int {
     const int qs = (int)(q & 1u);
     HIP_CHECK(hipEventSynchronize(ctx->ev_q[qs]));
-    const unsigned char* m = ctx->h_qmisc + 64 * qs;
+    const unsigned char* m = ctx->h_qmisc + 512 * qs;
     uint32_t final_bin, ncand, ovf;
     uint64_t total;
     std::memcpy(&final_bin, m + 0, 4);
-    std::memcpy(&ncand, m + 8, 4);
-    std::memcpy(&ovf, m + 16, 4);
-    std::memcpy(&total, m + 24, 8);
+    std::memcpy(&ncand, m + 128, 4);
+    std::memcpy(&ovf, m + 256, 4);
+    std::memcpy(&total, m + 384, 8);
     if (ovf) return SDB_ERR_OOM;
     SdbScoreDoc* csrc = qs ? ctx->d_cands2 : ctx->d_cands;
     std::vector<SdbScoreDoc> heap_c;
