@@ -2255,6 +2255,7 @@ static void ctx_free_partial(SdbGpuCtx* ctx) {
   if (ctx->d_ghist2) (void)hipFree(ctx->d_ghist2);
   if (ctx->d_qmisc) (void)hipFree(ctx->d_qmisc);
   if (ctx->h_qmisc) (void)hipHostFree(ctx->h_qmisc);
+  if (ctx->h_terms_pin) (void)hipHostFree(ctx->h_terms_pin);
   if (ctx->h_cands_pin) (void)hipHostFree(ctx->h_cands_pin);
   if (ctx->copy_stream) (void)hipStreamDestroy(ctx->copy_stream);
   if (ctx->d_cand_count) (void)hipFree(ctx->d_cand_count);
@@ -2311,6 +2312,9 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_ghist2, 4 * SDB_HIST_BINS * 16));
   CTX_CHECK(hipMalloc(&ctx->d_qmisc, 2 * 512));
   CTX_CHECK(hipHostMalloc(&ctx->h_qmisc, 2 * 512));
+  CTX_CHECK(hipHostMalloc(&ctx->h_terms_pin,
+                          sizeof(TermDev) * SDB_MAX_TERMS *
+                            SDB_TERM_SLOTS));
   CTX_CHECK(hipHostMalloc(&ctx->h_cands_pin, SDB_PIN_CANDS *
                                                sizeof(SdbScoreDoc)));
   CTX_CHECK(hipEventCreate(&ctx->ev_q[0]));
@@ -2326,6 +2330,7 @@ int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   (void)hipFree(ctx->d_qmisc);
   (void)hipHostFree(ctx->h_qmisc);
   (void)hipHostFree(ctx->h_cands_pin);
+  (void)hipHostFree(ctx->h_terms_pin);
   (void)hipEventDestroy(ctx->ev_q[0]);
   (void)hipEventDestroy(ctx->ev_q[1]);
   (void)hipStreamDestroy(ctx->copy_stream);
@@ -2664,7 +2669,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     const uint32_t slot = s % SDB_TERM_SLOTS;
     if (s && slot == 0) HIP_CHECK(hipStreamSynchronize(ctx->stream));
     TermDev* d_tslot = ctx->d_terms + (size_t)slot * SDB_MAX_TERMS;
-    TermDev tdev[SDB_MAX_TERMS];
+    TermDev* tdev = ctx->h_terms_pin + (size_t)slot * SDB_MAX_TERMS;
     for (uint32_t t = 0; t < plan->nterms; ++t) {
       const SdbTermEntry& te = seg->terms_host[plan->terms[t].term_idx];
       tdev[t].desc_begin = te.desc_begin;
@@ -2979,7 +2984,7 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       if ((q * nsegs + sg) >= SDB_TERM_SLOTS && slot == 0)
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
       TermDev* d_tslot = ctx->d_terms + (size_t)slot * SDB_MAX_TERMS;
-      TermDev tdev[SDB_MAX_TERMS];
+      TermDev* tdev = ctx->h_terms_pin + (size_t)slot * SDB_MAX_TERMS;
       for (uint32_t t = 0; t < plan->nterms; ++t) {
         const SdbTermEntry& te = seg->terms_host[plan->terms[t].term_idx];
         tdev[t].desc_begin = te.desc_begin;

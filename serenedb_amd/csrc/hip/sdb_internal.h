@@ -35,6 +35,10 @@ struct SdbGpuCtx {
   unsigned char* d_qmisc;   // 2 x 64 B: {gthresh u32, pad, cand_count u32,
                             //  overflow u32, total u64} per set
   unsigned char* h_qmisc;   // pinned mirror
+  TermDev* h_terms_pin;     // pinned term-table staging (SDB_TERM_SLOTS
+                            // x SDB_MAX_TERMS): pageable-source async
+                            // copies block the host until the stream
+                            // drains, which serialized the batch pipeline
   SdbScoreDoc* h_cands_pin; // pinned candidate staging (SDB_PIN_CANDS)
   hipEvent_t ev_q[2];
   double last_kernel_ms;   // read back via sdb_gpu_last_kernel_ms
