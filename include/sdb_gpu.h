@@ -205,6 +205,10 @@ typedef enum SdbPredOp {
   SDB_PRED_GE = 2,  /* col >= v */
   SDB_PRED_BETWEEN = 3, /* lo <= col <= hi */
   SDB_PRED_EQ = 4,  /* col == v (table_filter_iterator.hpp typed compares) */
+  /* bare null checks, evaluated on the validity plane alone
+   * (table_filter_iterator.hpp:65-67 NullCheckKind) */
+  SDB_PRED_ISNULL = 5,
+  SDB_PRED_NOTNULL = 6,
 } SdbPredOp;
 
 int sdb_gpu_segment_attach_column(SdbGpuCtx* ctx, SdbGpuSegment* seg,
@@ -273,6 +277,17 @@ typedef struct SdbGpuTable SdbGpuTable;
 int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
                        uint32_t ncols, uint64_t rows, SdbGpuTable** out);
 int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab);
+
+/* Column validity (null) bitmap — SQL three-valued logic as the
+ * reference's pushed filters implement it (null_semantics_fuzz.py;
+ * table_filter_iterator.hpp NullCheckKind): a comparison predicate drops
+ * NULL rows; SDB_PRED_ISNULL / SDB_PRED_NOTNULL evaluate the validity
+ * plane alone; SUM aggregates skip NULL values while COUNT(*) counts the
+ * row. bit r of bits[r>>6] set = row r VALID; (rows+63)/64 words. NULL
+ * detaches (all valid). Validity on the GROUP-KEY column is rejected at
+ * scan time (NULL-group semantics not implemented). */
+int sdb_gpu_table_attach_validity(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                                  uint32_t col, const uint64_t* bits);
 
 typedef struct SdbPredSpec {
   uint32_t col;

@@ -315,7 +315,7 @@ def gpu():
             "sdb_gpu_scan_agg_hash",
             "sdb_gpu_segment_attach_column", "sdb_gpu_execute_topk_hybrid",
             "sdb_gpu_segment_attach_livemask",
-            "sdb_gpu_execute_topk_batch",
+            "sdb_gpu_execute_topk_batch", "sdb_gpu_table_attach_validity",
             "sdb_gpu_execute_match_docs", "sdb_gpu_execute_count",
         ):
             getattr(lib, f).restype = C.c_int
@@ -650,6 +650,24 @@ class GpuContext:
 
     def free_table(self, tab):
         self._lib.sdb_gpu_table_free(self._ctx, tab)
+
+    def attach_validity(self, tab, col, bits):
+        """Attach (mask=None detaches) a column validity bitmap: bit r of
+        word r>>6 set = row r valid (NOT NULL). SQL three-valued logic:
+        comparisons drop NULL rows, ISNULL/NOTNULL (ops 5/6) evaluate the
+        plane alone, SUM skips NULLs, COUNT(*) counts rows."""
+        import numpy as np
+
+        if bits is None:
+            rc = self._lib.sdb_gpu_table_attach_validity(
+                self._ctx, tab, C.c_uint32(col), None)
+        else:
+            b = np.ascontiguousarray(bits, dtype=np.uint64)
+            rc = self._lib.sdb_gpu_table_attach_validity(
+                self._ctx, tab, C.c_uint32(col),
+                b.ctypes.data_as(C.POINTER(C.c_uint64)))
+        if rc != 0:
+            raise RuntimeError(f"attach_validity rc={rc}")
 
     def scan_agg(self, tab, group_col, ngroups, preds, aggs):
         """preds: list of (col, op, lo, hi) with SdbPredOp numeric op

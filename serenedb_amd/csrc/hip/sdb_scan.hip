@@ -85,6 +85,7 @@ struct SdbGpuTable {
   // key scribble past the LDS accumulators (ADVICE r1 / VERDICT weak #4)
   int64_t col_min[16], col_max[16];
   uint8_t has_minmax[16];
+  unsigned long long* valid[16];  // device validity bitmaps or null
   uint32_t ncols;
   uint64_t rows;
   uint32_t group_rows;   // shared by every FoR column (0 if none)
@@ -128,9 +129,11 @@ struct ScanArgs {
   int64_t pred_lo[SCAN_MAX_PREDS];
   int64_t pred_hi[SCAN_MAX_PREDS];
   float pred_flo[SCAN_MAX_PREDS], pred_fhi[SCAN_MAX_PREDS];
+  const unsigned long long* pred_valid[SCAN_MAX_PREDS];  // null = all valid
   ColRef agg_col[SCAN_MAX_AGGS];
   int agg_op[SCAN_MAX_AGGS];
   int agg_src[SCAN_MAX_AGGS];  // 0=own col_read, 1+p=pred p's value, 9=key
+  const unsigned long long* agg_valid[SCAN_MAX_AGGS];  // SUM skips nulls
   unsigned long long* out;
   unsigned long long* rows_passed;
   // --- staged-FoR variant only (scan_agg_staged_kernel) ---
@@ -164,6 +167,19 @@ __device__ __forceinline__ bool pred_eval_i(int op, int64_t x, int64_t lo,
     default: return true;
   }
 }
+__device__ __forceinline__ bool valid_at(const unsigned long long* v,
+                                         uint64_t r) {
+  return !v || ((v[r >> 6] >> (r & 63u)) & 1ull);
+}
+// predicate with SQL three-valued logic over an optional validity plane:
+// comparisons fail on NULL; ISNULL/NOTNULL evaluate the plane alone
+__device__ __forceinline__ bool pred_eval_iv(
+  int op, const unsigned long long* vb, uint64_t r, int64_t x, int64_t lo,
+  int64_t hi) {
+  if (op == SDB_PRED_ISNULL) return vb && !valid_at(vb, r);
+  if (op == SDB_PRED_NOTNULL) return valid_at(vb, r);
+  return valid_at(vb, r) && pred_eval_i(op, x, lo, hi);
+}
 __device__ __forceinline__ bool pred_eval_f(int op, float x, float lo,
                                             float hi) {
   switch (op) {
@@ -173,6 +189,14 @@ __device__ __forceinline__ bool pred_eval_f(int op, float x, float lo,
     case SDB_PRED_EQ: return x == lo;
     default: return true;
   }
+}
+
+__device__ __forceinline__ bool pred_eval_fv(
+  int op, const unsigned long long* vb, uint64_t r, float x, float lo,
+  float hi) {
+  if (op == SDB_PRED_ISNULL) return vb && !valid_at(vb, r);
+  if (op == SDB_PRED_NOTNULL) return valid_at(vb, r);
+  return valid_at(vb, r) && pred_eval_f(op, x, lo, hi);
 }
 
 // branchless funnel-shift extraction: value = bits [bit, bit+width) of the
@@ -270,8 +294,8 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
           const float fs[4] = {f.x, f.y, f.z, f.w};
 #pragma unroll
           for (int e = 0; e < 4; ++e)
-            okv[e] &= pred_eval_f(a.pred_op[p], fs[e], a.pred_flo[p],
-                                  a.pred_fhi[p]);
+            okv[e] &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r + e,
+                                   fs[e], a.pred_flo[p], a.pred_fhi[p]);
           continue;
         }
         longlong2 x, y;
@@ -281,8 +305,8 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         const int64_t xs[4] = {x.x, x.y, y.x, y.y};
 #pragma unroll
         for (int e = 0; e < 4; ++e)
-          okv[e] &= pred_eval_i(a.pred_op[p], xs[e], a.pred_lo[p],
-                                a.pred_hi[p]);
+          okv[e] &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r + e,
+                                 xs[e], a.pred_lo[p], a.pred_hi[p]);
       }
       if (!okv[0] && !okv[1] && !okv[2] && !okv[3]) continue;
       longlong2 k0, k1;
@@ -296,6 +320,9 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         const uint32_t g = (uint32_t)ks[e];
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[g * a.naggs + q];
+          if (a.agg_op[q] != SDB_AGG_COUNT &&
+              !valid_at(a.agg_valid[q], r + e))
+            continue;  // SUM skips NULL values; COUNT(*) counts the row
           switch (a.agg_op[q]) {
             case SDB_AGG_COUNT:
               atomicAdd(slot, 1ull);
@@ -318,19 +345,22 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
       bool ok = true;
       for (uint32_t p = 0; p < a.npreds; ++p) {
         if (a.pred_isf32[p])
-          ok &= pred_eval_f(a.pred_op[p],
-                            ((const float*)a.pred_col[p].data)[r],
-                            a.pred_flo[p], a.pred_fhi[p]);
+          ok &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r,
+                             ((const float*)a.pred_col[p].data)[r],
+                             a.pred_flo[p], a.pred_fhi[p]);
         else
-          ok &= pred_eval_i(a.pred_op[p],
-                            ((const long long*)a.pred_col[p].data)[r],
-                            a.pred_lo[p], a.pred_hi[p]);
+          ok &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r,
+                             ((const long long*)a.pred_col[p].data)[r],
+                             a.pred_lo[p], a.pred_hi[p]);
       }
       if (ok) {
         ++my_passed;
         const uint32_t g = (uint32_t)((const long long*)a.keys.data)[r];
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[g * a.naggs + q];
+          if (a.agg_op[q] != SDB_AGG_COUNT &&
+              !valid_at(a.agg_valid[q], r))
+            continue;
           switch (a.agg_op[q]) {
             case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
             case SDB_AGG_SUM_I64:
@@ -404,8 +434,9 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
           const float* fc = (const float*)a.pred_col[p].data;
 #pragma unroll
           for (int e = 0; e < 2; ++e)
-            okv[e] &= pred_eval_f(a.pred_op[p], fc[r + e], a.pred_flo[p],
-                                  a.pred_fhi[p]);
+            okv[e] &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r + e,
+                                   fc[r + e], a.pred_flo[p],
+                                   a.pred_fhi[p]);
           continue;
         }
         int64_t x0, x1;
@@ -415,20 +446,21 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         const int64_t xs[2] = {x0, x1};
 #pragma unroll
         for (int e = 0; e < 2; ++e)
-          okv[e] &= pred_eval_i(a.pred_op[p], xs[e], a.pred_lo[p],
-                                a.pred_hi[p]);
+          okv[e] &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r + e,
+                                 xs[e], a.pred_lo[p], a.pred_hi[p]);
       }
       for (uint32_t p = 2; p < a.npreds; ++p) {  // rare: >2 predicates
 #pragma unroll
         for (int e = 0; e < 2; ++e) {
           if (a.pred_isf32[p])
-            okv[e] &= pred_eval_f(a.pred_op[p],
-                                  ((const float*)a.pred_col[p].data)[r + e],
-                                  a.pred_flo[p], a.pred_fhi[p]);
+            okv[e] &= pred_eval_fv(
+              a.pred_op[p], a.pred_valid[p], r + e,
+              ((const float*)a.pred_col[p].data)[r + e], a.pred_flo[p],
+              a.pred_fhi[p]);
           else
-            okv[e] &= pred_eval_i(a.pred_op[p],
-                                  col_read(a.pred_col[p], rg, r0, r + e),
-                                  a.pred_lo[p], a.pred_hi[p]);
+            okv[e] &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r + e,
+                                   col_read(a.pred_col[p], rg, r0, r + e),
+                                   a.pred_lo[p], a.pred_hi[p]);
         }
       }
       if (!okv[0] && !okv[1]) continue;
@@ -442,6 +474,9 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         const uint32_t grp = (uint32_t)ks[e];
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
+          if (a.agg_op[q] != SDB_AGG_COUNT &&
+              !valid_at(a.agg_valid[q], r + e))
+            continue;
           switch (a.agg_op[q]) {
             case SDB_AGG_COUNT:
               atomicAdd(slot, 1ull);
@@ -471,21 +506,25 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
       int64_t pv0s = 0, pv1s = 0;
       for (uint32_t p = 0; p < a.npreds; ++p) {
         if (a.pred_isf32[p]) {
-          ok &= pred_eval_f(a.pred_op[p],
-                            ((const float*)a.pred_col[p].data)[r],
-                            a.pred_flo[p], a.pred_fhi[p]);
+          ok &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r,
+                             ((const float*)a.pred_col[p].data)[r],
+                             a.pred_flo[p], a.pred_fhi[p]);
           continue;
         }
         const int64_t x = col_read(a.pred_col[p], rg, r0, r);
         if (p == 0) pv0s = x;
         else if (p == 1) pv1s = x;
-        ok &= pred_eval_i(a.pred_op[p], x, a.pred_lo[p], a.pred_hi[p]);
+        ok &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r, x,
+                           a.pred_lo[p], a.pred_hi[p]);
       }
       if (ok) {
         ++my_passed;
         const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
+          if (a.agg_op[q] != SDB_AGG_COUNT &&
+              !valid_at(a.agg_valid[q], r))
+            continue;
           switch (a.agg_op[q]) {
             case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
             case SDB_AGG_SUM_I64: {
@@ -608,8 +647,9 @@ void scan_agg_staged_kernel(ScanArgs a) {
             const float* fc = (const float*)a.pred_col[p].data;
 #pragma unroll
             for (int e = 0; e < 2; ++e)
-              okv[e] &= pred_eval_f(a.pred_op[p], fc[r + e],
-                                    a.pred_flo[p], a.pred_fhi[p]);
+              okv[e] &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r + e,
+                                     fc[r + e], a.pred_flo[p],
+                                     a.pred_fhi[p]);
             continue;
           }
           int64_t x0, x1;
@@ -623,20 +663,22 @@ void scan_agg_staged_kernel(ScanArgs a) {
           const int64_t xs[2] = {x0, x1};
 #pragma unroll
           for (int e = 0; e < 2; ++e)
-            okv[e] &= pred_eval_i(a.pred_op[p], xs[e], a.pred_lo[p],
-                                  a.pred_hi[p]);
+            okv[e] &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r + e,
+                                   xs[e], a.pred_lo[p], a.pred_hi[p]);
         }
         for (uint32_t p = 2; p < a.npreds; ++p) {  // rare: >2 predicates
 #pragma unroll
           for (int e = 0; e < 2; ++e) {
             if (a.pred_isf32[p])
-              okv[e] &= pred_eval_f(
-                a.pred_op[p], ((const float*)a.pred_col[p].data)[r + e],
-                a.pred_flo[p], a.pred_fhi[p]);
+              okv[e] &= pred_eval_fv(
+                a.pred_op[p], a.pred_valid[p], r + e,
+                ((const float*)a.pred_col[p].data)[r + e], a.pred_flo[p],
+                a.pred_fhi[p]);
             else
-              okv[e] &= pred_eval_i(a.pred_op[p],
-                                    col_read(a.pred_col[p], rg, r0, r + e),
-                                    a.pred_lo[p], a.pred_hi[p]);
+              okv[e] &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r + e,
+                                     col_read(a.pred_col[p], rg, r0,
+                                              r + e),
+                                     a.pred_lo[p], a.pred_hi[p]);
           }
         }
         if (!okv[0] && !okv[1]) continue;
@@ -654,6 +696,9 @@ void scan_agg_staged_kernel(ScanArgs a) {
           const uint32_t grp = (uint32_t)ks[e];
           for (uint32_t q = 0; q < a.naggs; ++q) {
             unsigned long long* slot = &acc[grp * a.naggs + q];
+            if (a.agg_op[q] != SDB_AGG_COUNT &&
+                !valid_at(a.agg_valid[q], r + e))
+              continue;
             switch (a.agg_op[q]) {
               case SDB_AGG_COUNT:
                 atomicAdd(slot, 1ull);
@@ -689,21 +734,25 @@ void scan_agg_staged_kernel(ScanArgs a) {
       int64_t pv0s = 0, pv1s = 0;
       for (uint32_t p = 0; p < a.npreds; ++p) {
         if (a.pred_isf32[p]) {
-          ok &= pred_eval_f(a.pred_op[p],
-                            ((const float*)a.pred_col[p].data)[r],
-                            a.pred_flo[p], a.pred_fhi[p]);
+          ok &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r,
+                             ((const float*)a.pred_col[p].data)[r],
+                             a.pred_flo[p], a.pred_fhi[p]);
           continue;
         }
         const int64_t x = col_read(a.pred_col[p], rg, r0, r);
         if (p == 0) pv0s = x;
         else if (p == 1) pv1s = x;
-        ok &= pred_eval_i(a.pred_op[p], x, a.pred_lo[p], a.pred_hi[p]);
+        ok &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r, x,
+                           a.pred_lo[p], a.pred_hi[p]);
       }
       if (ok) {
         ++my_passed;
         const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
+          if (a.agg_op[q] != SDB_AGG_COUNT &&
+              !valid_at(a.agg_valid[q], r))
+            continue;
           switch (a.agg_op[q]) {
             case SDB_AGG_COUNT: atomicAdd(slot, 1ull); break;
             case SDB_AGG_SUM_I64: {
@@ -781,8 +830,10 @@ struct HashAggArgs {
   int64_t pred_lo[SCAN_MAX_PREDS];
   int64_t pred_hi[SCAN_MAX_PREDS];
   float pred_flo[SCAN_MAX_PREDS], pred_fhi[SCAN_MAX_PREDS];
+  const unsigned long long* pred_valid[SCAN_MAX_PREDS];
   ColRef agg_col[SCAN_MAX_AGGS];
   int agg_op[SCAN_MAX_AGGS];
+  const unsigned long long* agg_valid[SCAN_MAX_AGGS];
   // global open-addressed table (pow2 capacity, keys init SDB_HKEY_EMPTY)
   unsigned long long* gkeys;
   unsigned long long* gacc;  // [cap * naggs]
@@ -823,6 +874,8 @@ __device__ __forceinline__ void hash_acc_add(const HashAggArgs& a,
                                              uint32_t q, uint32_t rg,
                                              uint64_t r0, uint64_t r,
                                              bool global_scope) {
+  if (a.agg_op[q] != SDB_AGG_COUNT && !valid_at(a.agg_valid[q], r))
+    return;  // SUM skips NULL values; COUNT(*) counts the row
   switch (a.agg_op[q]) {
     case SDB_AGG_COUNT:
       atomicAdd(&acc[q], 1ull);
@@ -880,13 +933,13 @@ void scan_agg_hash_kernel(HashAggArgs a) {
       bool ok = true;
       for (uint32_t p = 0; p < a.npreds; ++p) {
         if (a.pred_isf32[p])
-          ok &= pred_eval_f(a.pred_op[p],
-                            ((const float*)a.pred_col[p].data)[r],
-                            a.pred_flo[p], a.pred_fhi[p]);
+          ok &= pred_eval_fv(a.pred_op[p], a.pred_valid[p], r,
+                             ((const float*)a.pred_col[p].data)[r],
+                             a.pred_flo[p], a.pred_fhi[p]);
         else
-          ok &= pred_eval_i(a.pred_op[p],
-                            col_read(a.pred_col[p], rg, r0, r),
-                            a.pred_lo[p], a.pred_hi[p]);
+          ok &= pred_eval_iv(a.pred_op[p], a.pred_valid[p], r,
+                             col_read(a.pred_col[p], rg, r0, r),
+                             a.pred_lo[p], a.pred_hi[p]);
       }
       if (!ok) continue;
       ++my_passed;
@@ -1079,8 +1132,28 @@ int sdb_gpu_table_load(SdbGpuCtx* ctx, const SdbColumnView* cols,
 
 int sdb_gpu_table_free(SdbGpuCtx* ctx, SdbGpuTable* tab) {
   if (!ctx || !tab) return SDB_ERR_INVALID;
-  for (uint32_t c = 0; c < tab->ncols; ++c) (void)hipFree(tab->cols[c]);
+  for (uint32_t c = 0; c < tab->ncols; ++c) {
+    (void)hipFree(tab->cols[c]);
+    if (tab->valid[c]) (void)hipFree(tab->valid[c]);
+  }
   delete tab;
+  return SDB_OK;
+}
+
+int sdb_gpu_table_attach_validity(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                                  uint32_t col, const uint64_t* bits) {
+  if (!ctx || !tab || col >= tab->ncols) return SDB_ERR_INVALID;
+  if (!bits) {
+    if (tab->valid[col]) (void)hipFree(tab->valid[col]);
+    tab->valid[col] = nullptr;
+    return SDB_OK;
+  }
+  const uint64_t nwords = (tab->rows + 63) / 64;
+  if (!tab->valid[col])
+    HIP_CHECK(hipMalloc(&tab->valid[col], 8 * (nwords + 1)));
+  HIP_CHECK(hipMemcpy(tab->valid[col], bits, 8 * nwords,
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemsetAsync(tab->valid[col] + nwords, 0xFF, 8, nullptr));
   return SDB_OK;
 }
 
@@ -1093,6 +1166,7 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
       naggs > SCAN_MAX_AGGS || npreds > SCAN_MAX_PREDS)
     return SDB_ERR_INVALID;
   if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
+  if (tab->valid[group_col]) return SDB_ERR_INVALID;  // no NULL groups
   // dense-key contract: keys must be provably inside [0, ngroups) or the
   // perfect-hash accumulate acc[key*naggs+q] corrupts LDS. Both column
   // kinds carry a load-time range (FoR zonemaps / raw-column reduction).
@@ -1112,7 +1186,7 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   a.npreds = npreds;
   for (uint32_t p = 0; p < npreds; ++p) {
     if (preds[p].col >= tab->ncols) return SDB_ERR_INVALID;
-    if (preds[p].op < SDB_PRED_LT || preds[p].op > SDB_PRED_EQ)
+    if (preds[p].op < SDB_PRED_LT || preds[p].op > SDB_PRED_NOTNULL)
       return SDB_ERR_INVALID;
     a.pred_col[p] = tab->refs[preds[p].col];
     a.pred_op[p] = preds[p].op;
@@ -1121,11 +1195,14 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
     a.pred_hi[p] = preds[p].ihi;
     a.pred_flo[p] = preds[p].flo;
     a.pred_fhi[p] = preds[p].fhi;
+    a.pred_valid[p] = tab->valid[preds[p].col];
   }
   for (uint32_t q = 0; q < naggs; ++q) {
     a.agg_op[q] = aggs[q].op;
     a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT ? ColRef{nullptr, nullptr}
                                                : tab->refs[aggs[q].col];
+    a.agg_valid[q] =
+      aggs[q].op == SDB_AGG_COUNT ? nullptr : tab->valid[aggs[q].col];
     a.agg_src[q] = 0;
     if (aggs[q].op == SDB_AGG_SUM_I64) {
       if (aggs[q].col == group_col) a.agg_src[q] = 9;
@@ -1279,6 +1356,7 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
       npreds > SCAN_MAX_PREDS)
     return SDB_ERR_INVALID;
   if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
+  if (tab->valid[group_col]) return SDB_ERR_INVALID;  // no NULL groups
 
   hipStream_t stream = ctx->stream;
   HashAggArgs a{};
@@ -1289,7 +1367,7 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
   a.npreds = npreds;
   for (uint32_t p = 0; p < npreds; ++p) {
     if (preds[p].col >= tab->ncols) return SDB_ERR_INVALID;
-    if (preds[p].op < SDB_PRED_LT || preds[p].op > SDB_PRED_EQ)
+    if (preds[p].op < SDB_PRED_LT || preds[p].op > SDB_PRED_NOTNULL)
       return SDB_ERR_INVALID;
     a.pred_col[p] = tab->refs[preds[p].col];
     a.pred_op[p] = preds[p].op;
@@ -1298,11 +1376,14 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
     a.pred_hi[p] = preds[p].ihi;
     a.pred_flo[p] = preds[p].flo;
     a.pred_fhi[p] = preds[p].fhi;
+    a.pred_valid[p] = tab->valid[preds[p].col];
   }
   for (uint32_t q = 0; q < naggs; ++q) {
     a.agg_op[q] = aggs[q].op;
     a.agg_col[q] = aggs[q].op == SDB_AGG_COUNT ? ColRef{nullptr, nullptr}
                                                : tab->refs[aggs[q].col];
+    a.agg_valid[q] =
+      aggs[q].op == SDB_AGG_COUNT ? nullptr : tab->valid[aggs[q].col];
     if (aggs[q].op == SDB_AGG_SUM_I64 &&
         tab->types[aggs[q].col] == SDB_COL_F32)
       return SDB_ERR_INVALID;

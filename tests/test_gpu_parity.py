@@ -1106,3 +1106,84 @@ def test_batch_pipelined_equals_single(ctx):
     assert bt2 == [t2] * 3
     np.testing.assert_array_equal(b2[0]["doc"], s2["doc"])
     np.testing.assert_array_equal(b2[2]["doc"], s2["doc"])
+
+
+def test_validity_null_semantics(ctx):
+    """Column validity (null) masks — SQL three-valued logic as the
+    reference's pushed filters implement it (tests/fuzz/
+    null_semantics_fuzz.py; table_filter_iterator.hpp NullCheckKind):
+    comparisons drop NULL rows, IS [NOT] NULL evaluates the validity
+    plane alone, SUM skips NULL values while COUNT(*) counts the row —
+    vs an independent numpy evaluation, on dense, FoR and hash paths."""
+    rows = 800_000
+    ngroups = 128
+    rng = np.random.default_rng(71)
+    keys = rng.integers(0, ngroups, rows).astype(np.int64)
+    v1 = rng.integers(0, 1000, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    valid1 = rng.random(rows) > 0.3   # 30% NULLs in v1
+    valid2 = rng.random(rows) > 0.2   # 20% NULLs in v2
+
+    def words(v):
+        w = np.zeros((rows + 63) // 64, dtype=np.uint64)
+        idx = np.nonzero(v)[0]
+        np.bitwise_or.at(w, idx // 64,
+                         (np.uint64(1) << (idx % 64).astype(np.uint64)))
+        return w
+
+    for enc in ("raw", "for"):
+        tab = ctx.load_table([keys, v1, v2], [enc, enc, "raw"])
+        ctx.attach_validity(tab, 1, words(valid1))
+        ctx.attach_validity(tab, 2, words(valid2))
+
+        # (a) comparison drops NULLs: v1 < 500
+        sel = valid1 & (v1 < 500)
+        i64, f64, passed = ctx.scan_agg(tab, 0, ngroups, [(1, 1, 500, 0)],
+                                        [(0, 0), (1, 1), (2, 2)])
+        assert passed == int(sel.sum())
+        np.testing.assert_array_equal(
+            i64[:, 0], np.bincount(keys[sel], minlength=ngroups))
+        # SUM(v1) over selected rows (all valid there); SUM(v2) skips
+        # v2-NULLs among the selected rows
+        s1 = np.zeros(ngroups, dtype=np.int64)
+        np.add.at(s1, keys[sel], v1[sel])
+        np.testing.assert_array_equal(i64[:, 1], s1)
+        sel2 = sel & valid2
+        s2 = np.zeros(ngroups, dtype=np.float64)
+        np.add.at(s2, keys[sel2], v2[sel2].astype(np.float64))
+        np.testing.assert_allclose(f64[:, 2], s2, rtol=1e-7)
+
+        # (b) IS NULL / IS NOT NULL on the validity plane alone
+        i64b, _, pb = ctx.scan_agg(tab, 0, ngroups, [(1, 5, 0, 0)],
+                                   [(0, 0)])
+        assert pb == int((~valid1).sum())
+        np.testing.assert_array_equal(
+            i64b[:, 0], np.bincount(keys[~valid1], minlength=ngroups))
+        _, _, pnn = ctx.scan_agg(tab, 0, ngroups, [(1, 6, 0, 0)], [(0, 0)])
+        assert pnn == int(valid1.sum())
+        # ISNULL on a column WITHOUT validity selects nothing
+        _, _, p0 = ctx.scan_agg(tab, 0, ngroups, [(0, 5, 0, 0)], [(0, 0)])
+        assert p0 == 0
+
+        # (c) f32 predicate drops NULLs
+        fsel = valid2 & (v2 >= np.float32(0.0))
+        _, _, pf = ctx.scan_agg(tab, 0, ngroups, [(2, 2, 0.0, 0.0)],
+                                [(0, 0)])
+        assert pf == int(fsel.sum())
+
+        # (d) hash path honors the same semantics
+        gk, hi64, _, hp = ctx.scan_agg_hash(
+            tab, 0, ngroups + 8, [(1, 5, 0, 0)], [(0, 0)])
+        assert hp == pb
+        live = np.nonzero(np.bincount(keys[~valid1],
+                                      minlength=ngroups))[0]
+        np.testing.assert_array_equal(gk, live)
+
+        # (e) validity on the group key is rejected
+        import pytest as _pytest
+        ctx.attach_validity(tab, 0, words(valid1))
+        with _pytest.raises(RuntimeError):
+            ctx.scan_agg(tab, 0, ngroups, [], [(0, 0)])
+        ctx.attach_validity(tab, 0, None)  # detach restores
+        ctx.scan_agg(tab, 0, ngroups, [], [(0, 0)])
+        ctx.free_table(tab)
