@@ -2439,7 +2439,10 @@ int sdb_gpu_segment_attach_livemask(SdbGpuCtx* ctx, SdbGpuSegment* seg,
   if (!seg->live)
     HIP_CHECK(hipMalloc(&seg->live, 8 * (nwords + 1)));  // +1: funnel pad
   HIP_CHECK(hipMemcpy(seg->live, mask, 8 * nwords, hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemsetAsync(seg->live + nwords, 0, 8, nullptr));
+  // synchronous: the pad word must be visible before any later kernel on
+  // the context stream reads it (an async memset on the null stream has
+  // no ordering against ctx->stream)
+  HIP_CHECK(hipMemset(seg->live + nwords, 0, 8));
   return SDB_OK;
 }
 

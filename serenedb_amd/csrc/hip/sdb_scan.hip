@@ -1153,7 +1153,9 @@ int sdb_gpu_table_attach_validity(SdbGpuCtx* ctx, SdbGpuTable* tab,
     HIP_CHECK(hipMalloc(&tab->valid[col], 8 * (nwords + 1)));
   HIP_CHECK(hipMemcpy(tab->valid[col], bits, 8 * nwords,
                       hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemsetAsync(tab->valid[col] + nwords, 0xFF, 8, nullptr));
+  // synchronous for the same null-stream-ordering reason as the live
+  // mask's pad word
+  HIP_CHECK(hipMemset(tab->valid[col] + nwords, 0xFF, 8));
   return SDB_OK;
 }
 

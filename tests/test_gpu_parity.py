@@ -1187,3 +1187,21 @@ def test_validity_null_semantics(ctx):
         ctx.attach_validity(tab, 0, None)  # detach restores
         ctx.scan_agg(tab, 0, ngroups, [], [(0, 0)])
         ctx.free_table(tab)
+
+
+def test_wave_kernel_path_parity(ctx):
+    """The experimental barrier-free per-wave kernel (SDB_TOPK_PATH=wave)
+    keeps full parity even though it is no longer the default path."""
+    import os
+    blob, _, _ = make_corpus(81, 600_000, [0.1, 0.05, 0.02, 0.01])
+    seg = ctx.load_segment(blob)
+    base, btotal = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, 500)
+    os.environ["SDB_TOPK_PATH"] = "wave"
+    try:
+        w, wtotal = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, 500)
+    finally:
+        os.environ.pop("SDB_TOPK_PATH", None)
+    assert wtotal == btotal
+    np.testing.assert_array_equal(w["doc"], base["doc"])
+    np.testing.assert_array_equal(w["score"].view(np.uint32),
+                                  base["score"].view(np.uint32))
