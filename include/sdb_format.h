@@ -90,7 +90,13 @@ typedef struct SdbBlockDesc {
   uint32_t doc_off;  /* byte offset of doc-block payload (incl. tag byte) */
   uint32_t freq_off; /* byte offset of freq-block payload (incl. tag byte) */
   uint16_t len;      /* docs in this block: 128, or 1..127 for the tail */
-  uint16_t flags;    /* reserved */
+  uint16_t flags;    /* v2: bit0=1 -> the fused shape (128 docs, all three
+                      * streams bitpacked) and bits1-5/6-10/11-15 carry the
+                      * doc/freq/norm bit widths, so the decode issues every
+                      * packed-word load without first fetching the payload
+                      * tag bytes (one full memory round trip per block);
+                      * bit0=0 -> bits1+ = freq-block byte size (norm_off =
+                      * freq_off + (flags>>1)) */
   uint32_t max_freq; /* max freq in block  (WAND bound, round-2 pruning) */
   uint32_t min_norm; /* min norm over block's docs (WAND bound) */
 } SdbBlockDesc;

@@ -359,7 +359,8 @@ uint32_t o_encode_freq_block(const uint32_t* in, uint32_t len, uint8_t* out) {
 int o_segment_parse(const void* blob, uint64_t size, SdbSegmentView* out) {
   if (!blob || size < sizeof(SdbSegHeader)) return -5;
   const SdbSegHeader* hdr = (const SdbSegHeader*)blob;
-  if (hdr->magic != SDB_SEG_MAGIC || hdr->version < 1 || hdr->version > 2 ||
+  if (hdr->magic != SDB_SEG_MAGIC || hdr->version < 1 ||
+      hdr->version > 3 || hdr->version == 2 ||
       hdr->blob_size > size)
     return -5;
   const uint8_t* base = (const uint8_t*)blob;

@@ -355,38 +355,30 @@ __device__ __forceinline__ bool try_block_fused(
   uint32_t lo, uint32_t hi, float num, float nc, float nl, uint32_t scorer,
   const uint32_t* norms_col, const float* fboost, float* swin,
   void* cwin) {
+  // v2 descriptors carry the fused-shape bit + all three bit widths, so
+  // every packed-word load below issues with NO payload-tag fetch first
+  // (one whole memory round trip per block saved; sdb_format.h flags)
+  if (!(d.flags & 1u) || !norm_stream) return false;
+  const uint32_t dbits = (d.flags >> 1) & 31u;
+  const uint32_t fbits = (d.flags >> 6) & 31u;
+  const uint32_t nbits = (d.flags >> 11) & 31u;
   const uint8_t* db = pl + d.doc_off;
   const uint8_t* fb = pl + d.freq_off;
-  const uint8_t* nb = fb + d.flags;
-  uint32_t dtag, ftag, ntag;
-  load_tags3(db, fb, nb, lane, &dtag, &ftag, &ntag);  // no SMEM (poison)
-  if (!norm_stream) ntag = SDB_E_BITPACK_01;
-  if (dtag < SDB_DE_DELTA_BITPACK_02 || ftag < SDB_E_BITPACK_01 ||
-      ntag < SDB_E_BITPACK_01)
-    return false;
-  const uint32_t dbits = dtag - SDB_DE_DELTA_BITPACK_02 + 2;
-  const uint32_t fbits = ftag - SDB_E_BITPACK_01 + 1;
-  const uint32_t nbits = ntag - SDB_E_BITPACK_01 + 1;
+  const uint8_t* nb = fb + 1 + 16u * fbits;  // bitpack size = 1 + 16*bits
   const uint32_t i0 = 2u * lane, i1 = i0 + 1;
   // all loads issue here, before any cross-lane dependency
   const uint32_t dd0 = extract_packed(db + 1, dbits, i0);
   const uint32_t dd1 = extract_packed(db + 1, dbits, i1);
   const uint32_t f0 = extract_packed(fb + 1, fbits, i0);
   const uint32_t f1 = extract_packed(fb + 1, fbits, i1);
-  uint32_t n0 = 0, n1 = 0;
-  if (norm_stream) {
-    n0 = extract_packed(nb + 1, nbits, i0);
-    n1 = extract_packed(nb + 1, nbits, i1);
-  }
+  const uint32_t n0 = extract_packed(nb + 1, nbits, i0);
+  const uint32_t n1 = extract_packed(nb + 1, nbits, i1);
+  (void)norms_col;
   const uint32_t pair = dd0 + dd1;
   const uint32_t incl = wave_incl_scan(pair, lane);
   const uint32_t excl = incl - pair;
   const uint32_t doc0 = d.prev_doc + excl + dd0;
   const uint32_t doc1 = d.prev_doc + excl + pair;
-  if (!norm_stream) {
-    n0 = doc0 >= lo && doc0 <= hi ? norms_col[doc0] : 1u;
-    n1 = doc1 >= lo && doc1 <= hi ? norms_col[doc1] : 1u;
-  }
   // filter boost folds into num BEFORE the score form, mirroring the
   // reference's op order (bm25.cpp: c0 = boost*num, then c0 - c0*c1/(c1+f))
   if (doc0 >= lo && doc0 <= hi) {
@@ -415,44 +407,32 @@ __device__ __forceinline__ bool try_block_fused2(
   int lane, uint32_t norm_stream, uint32_t lo, uint32_t hi, float num,
   float nc, float nl, uint32_t scorer, const uint32_t* norms_col,
   const float* fboost, float* swin, void* cwin) {
+  if (!(da.flags & 1u) || !(db_.flags & 1u) || !norm_stream)
+    return false;  // widths come from the descriptors (sdb_format.h)
+  const uint32_t adb = (da.flags >> 1) & 31u, afb = (da.flags >> 6) & 31u;
+  const uint32_t anb = (da.flags >> 11) & 31u;
+  const uint32_t bdb = (db_.flags >> 1) & 31u, bfb = (db_.flags >> 6) & 31u;
+  const uint32_t bnb = (db_.flags >> 11) & 31u;
   const uint8_t* adoc = pl + da.doc_off;
   const uint8_t* afrq = pl + da.freq_off;
-  const uint8_t* anrm = afrq + da.flags;
+  const uint8_t* anrm = afrq + 1 + 16u * afb;
   const uint8_t* bdoc = pl + db_.doc_off;
   const uint8_t* bfrq = pl + db_.freq_off;
-  const uint8_t* bnrm = bfrq + db_.flags;
-  // six tag bytes through the vector path (see load_tags3)
-  const uint8_t* tp = lane == 1   ? afrq
-                      : lane == 2 ? anrm
-                      : lane == 3 ? bdoc
-                      : lane == 4 ? bfrq
-                      : lane == 5 ? bnrm
-                                  : adoc;
-  const uint32_t tv = tp[0];
-  const uint32_t adt = __shfl(tv, 0, 64), aft = __shfl(tv, 1, 64);
-  const uint32_t ant0 = __shfl(tv, 2, 64);
-  const uint32_t bdt = __shfl(tv, 3, 64), bft = __shfl(tv, 4, 64);
-  const uint32_t bnt0 = __shfl(tv, 5, 64);
-  const uint32_t ant = norm_stream ? ant0 : SDB_E_BITPACK_01;
-  const uint32_t bnt = norm_stream ? bnt0 : SDB_E_BITPACK_01;
-  if (adt < SDB_DE_DELTA_BITPACK_02 || aft < SDB_E_BITPACK_01 ||
-      ant < SDB_E_BITPACK_01 || bdt < SDB_DE_DELTA_BITPACK_02 ||
-      bft < SDB_E_BITPACK_01 || bnt < SDB_E_BITPACK_01 || !norm_stream)
-    return false;
+  const uint8_t* bnrm = bfrq + 1 + 16u * bfb;
   const uint32_t i0 = 2u * lane, i1 = i0 + 1;
-  // every load issues here
-  const uint32_t a_d0 = extract_packed(adoc + 1, adt - SDB_DE_DELTA_BITPACK_02 + 2, i0);
-  const uint32_t a_d1 = extract_packed(adoc + 1, adt - SDB_DE_DELTA_BITPACK_02 + 2, i1);
-  const uint32_t b_d0 = extract_packed(bdoc + 1, bdt - SDB_DE_DELTA_BITPACK_02 + 2, i0);
-  const uint32_t b_d1 = extract_packed(bdoc + 1, bdt - SDB_DE_DELTA_BITPACK_02 + 2, i1);
-  const uint32_t a_f0 = extract_packed(afrq + 1, aft - SDB_E_BITPACK_01 + 1, i0);
-  const uint32_t a_f1 = extract_packed(afrq + 1, aft - SDB_E_BITPACK_01 + 1, i1);
-  const uint32_t b_f0 = extract_packed(bfrq + 1, bft - SDB_E_BITPACK_01 + 1, i0);
-  const uint32_t b_f1 = extract_packed(bfrq + 1, bft - SDB_E_BITPACK_01 + 1, i1);
-  const uint32_t a_n0 = extract_packed(anrm + 1, ant - SDB_E_BITPACK_01 + 1, i0);
-  const uint32_t a_n1 = extract_packed(anrm + 1, ant - SDB_E_BITPACK_01 + 1, i1);
-  const uint32_t b_n0 = extract_packed(bnrm + 1, bnt - SDB_E_BITPACK_01 + 1, i0);
-  const uint32_t b_n1 = extract_packed(bnrm + 1, bnt - SDB_E_BITPACK_01 + 1, i1);
+  // every load issues here — no tag fetch precedes them
+  const uint32_t a_d0 = extract_packed(adoc + 1, adb, i0);
+  const uint32_t a_d1 = extract_packed(adoc + 1, adb, i1);
+  const uint32_t b_d0 = extract_packed(bdoc + 1, bdb, i0);
+  const uint32_t b_d1 = extract_packed(bdoc + 1, bdb, i1);
+  const uint32_t a_f0 = extract_packed(afrq + 1, afb, i0);
+  const uint32_t a_f1 = extract_packed(afrq + 1, afb, i1);
+  const uint32_t b_f0 = extract_packed(bfrq + 1, bfb, i0);
+  const uint32_t b_f1 = extract_packed(bfrq + 1, bfb, i1);
+  const uint32_t a_n0 = extract_packed(anrm + 1, anb, i0);
+  const uint32_t a_n1 = extract_packed(anrm + 1, anb, i1);
+  const uint32_t b_n0 = extract_packed(bnrm + 1, bnb, i0);
+  const uint32_t b_n1 = extract_packed(bnrm + 1, bnb, i1);
   // two independent wave scans (shfl chains interleave)
   const uint32_t a_pair = a_d0 + a_d1;
   const uint32_t b_pair = b_d0 + b_d1;
@@ -818,8 +798,8 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
         decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
         if (a.norm_stream)  // v2: norm block follows the freq block
-          decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane,
-                                 nbuf);
+          decode_freq_block_wave(pl + d.freq_off + (d.flags >> 1),
+                                 d.len, lane, nbuf);
 #endif
         for (uint32_t j = lane; j < d.len; j += 64) {
           const uint32_t doc = dbuf[j];
@@ -1358,8 +1338,8 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
         decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
         if (a.norm_stream)
-          decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane,
-                                 nbuf);
+          decode_freq_block_wave(pl + d.freq_off + (d.flags >> 1),
+                                 d.len, lane, nbuf);
         for (uint32_t j = lane; j < d.len; j += 64) {
           const uint32_t doc = dbuf[j];
           if (doc < lo || doc > hi) continue;
@@ -1688,18 +1668,14 @@ __device__ __forceinline__ uint32_t pw_fill_block(
   uint32_t doc0 = 0, doc1 = 0;
   float s0 = 0.f, s1 = 0.f;
   bool have01 = false;
-  if (d.len == 128) {
+  if ((d.flags & 1u) && a.norm_stream) {
     const uint8_t* db = pl + d.doc_off;
     const uint8_t* fb = pl + d.freq_off;
-    const uint8_t* nb = fb + d.flags;
-    uint32_t dtag, ftag, ntag;
-    load_tags3(db, fb, nb, lane, &dtag, &ftag, &ntag);
-    if (!a.norm_stream) ntag = SDB_E_BITPACK_01;
-    if (dtag >= SDB_DE_DELTA_BITPACK_02 && ftag >= SDB_E_BITPACK_01 &&
-        ntag >= SDB_E_BITPACK_01) {
-      const uint32_t dbits = dtag - SDB_DE_DELTA_BITPACK_02 + 2;
-      const uint32_t fbits = ftag - SDB_E_BITPACK_01 + 1;
-      const uint32_t nbits = ntag - SDB_E_BITPACK_01 + 1;
+    {
+      const uint32_t dbits = (d.flags >> 1) & 31u;
+      const uint32_t fbits = (d.flags >> 6) & 31u;
+      const uint32_t nbits = (d.flags >> 11) & 31u;
+      const uint8_t* nb = fb + 1 + 16u * fbits;
       const uint32_t i0 = 2u * (uint32_t)lane, i1 = i0 + 1;
       const uint32_t dd0 = extract_packed(db + 1, dbits, i0);
       const uint32_t dd1 = extract_packed(db + 1, dbits, i1);
@@ -1748,7 +1724,8 @@ __device__ __forceinline__ uint32_t pw_fill_block(
     decode_doc_block_wave(pl + d.doc_off, d.len, d.prev_doc, lane, dbuf);
     decode_freq_block_wave(pl + d.freq_off, d.len, lane, fbuf);
     if (a.norm_stream)
-      decode_freq_block_wave(pl + d.freq_off + d.flags, d.len, lane, nbuf);
+      decode_freq_block_wave(pl + d.freq_off + (d.flags >> 1), d.len,
+                             lane, nbuf);
     const uint32_t i0 = 2u * (uint32_t)lane, i1 = i0 + 1;
     doc0 = i0 < d.len ? dbuf[i0] : 0xFFFFFFFFu;
     doc1 = i1 < d.len ? dbuf[i1] : 0xFFFFFFFFu;
@@ -2221,7 +2198,8 @@ int parse_blob(const void* blob, size_t size, SdbSegHeader* hdr_out) {
   SdbSegHeader hdr;
   std::memcpy(&hdr, blob, sizeof(hdr));
   if (hdr.magic != SDB_SEG_MAGIC) return -52;
-  if (hdr.version < 1 || hdr.version > 2) return -53;
+  if (hdr.version < 1 || hdr.version > 3 || hdr.version == 2)
+    return -53;  // v2's flags layout was retired with v3 (sdb_format.h)
   if (hdr.blob_size > size) return -54;
   // section-extent checks against the blob (the staging memcpy
   // path below must never read past the caller's span; same checks the
@@ -2704,7 +2682,7 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     a.scorer = scorer;
     a.wand = (plan->wand && (plan->min_match <= 1) && !hybrid) ? 1u : 0u;
     a.count_only = count_only;
-    a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
+    a.norm_stream = seg->hdr.version >= 3 ? 1u : 0u;
     a.nterms = plan->nterms;
     a.min_match = plan->min_match ? plan->min_match : 1;
     a.k = k;
@@ -3007,7 +2985,7 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       a.doc_count = seg->hdr.doc_count;
       a.scorer = plan->scorer;
       a.wand = (plan->wand && plan->min_match <= 1) ? 1u : 0u;
-      a.norm_stream = seg->hdr.version >= 2 ? 1u : 0u;
+      a.norm_stream = seg->hdr.version >= 3 ? 1u : 0u;
       a.nterms = plan->nterms;
       a.min_match = plan->min_match ? plan->min_match : 1;
       a.k = k;

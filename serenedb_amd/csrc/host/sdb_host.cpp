@@ -766,20 +766,36 @@ int sdb_host_build_segment(uint32_t doc_count, uint32_t nterms,
       bd.freq_off = uint32_t(payload.size() - te.payload_begin);
       sz = encode_freq_block(tf + pos, len, buf);
       payload.insert(payload.end(), buf, buf + sz);
+      const uint8_t ftag = buf[0];
       /* v2: per-block norm stream right after the freq block, encoded with
-       * the same non-delta families; flags = freq-block byte size so
-       * norm_off = freq_off + flags without a parse. This materializes the
-       * norm-column gather at index build time: query results
-       * are identical (parity tests compare against the column-reading
-       * oracle bit-for-bit), HBM traffic drops from a scattered 4 B/posting
+       * the same non-delta families. This materializes the norm-column
+       * gather at index build time: query results are
+       * identical (parity tests compare against the column-reading oracle
+       * bit-for-bit), HBM traffic drops from a scattered 4 B/posting
        * gather to ~1.5 B/posting of sequential payload. */
-      bd.flags = uint16_t(sz);
+      uint8_t ntag;
       {
         uint32_t nvals[kBlock];
         for (uint32_t i = 0; i < len; ++i)
           nvals[i] = norms ? norms[td[pos + i]] : 1u;
         const uint32_t nsz = encode_freq_block(nvals, len, buf);
         payload.insert(payload.end(), buf, buf + nsz);
+        ntag = buf[0];
+      }
+      /* flags (sdb_format.h): for the fused shape carry all three bit
+       * widths so the GPU decode needs no payload-tag fetch before
+       * issuing its packed-word loads; otherwise carry the freq-block
+       * size (norm_off derivation) with bit0 clear */
+      const uint8_t dtag2 = payload[te.payload_begin + bd.doc_off];
+      if (len == kBlock && dtag2 >= SDB_DE_DELTA_BITPACK_02 &&
+          ftag >= SDB_E_BITPACK_01 && ntag >= SDB_E_BITPACK_01) {
+        const uint32_t dbits = dtag2 - SDB_DE_DELTA_BITPACK_02 + 2;
+        const uint32_t fbits = ftag - SDB_E_BITPACK_01 + 1;
+        const uint32_t nbits = ntag - SDB_E_BITPACK_01 + 1;
+        bd.flags = uint16_t(1u | (dbits << 1) | (fbits << 6) |
+                            (nbits << 11));
+      } else {
+        bd.flags = uint16_t(sz << 1);
       }
       uint32_t mf = 0, mn = 0xFFFFFFFFu;
       for (uint32_t i = 0; i < len; ++i) {
@@ -801,7 +817,8 @@ int sdb_host_build_segment(uint32_t doc_count, uint32_t nterms,
   auto align64 = [](uint64_t x) { return (x + 63) & ~63ull; };
   SdbSegHeader hdr{};
   hdr.magic = SDB_SEG_MAGIC;
-  hdr.version = 2; /* v2 = per-block norm streams (flags = freq size) */
+  hdr.version = 3; /* v3 = per-block norm streams + width-carrying
+                  * descriptor flags (sdb_format.h) */
   hdr.nterms = nterms;
   hdr.doc_count = doc_count;
   hdr.docs_with_field = doc_count;
@@ -891,7 +908,8 @@ int sdb_host_segment_parse(const void* blob, uint64_t size,
   if (!blob || size < sizeof(SdbSegHeader)) return -5;
   const auto* hdr = static_cast<const SdbSegHeader*>(blob);
   if (hdr->magic != SDB_SEG_MAGIC || hdr->version < 1 ||
-      hdr->version > 2 || hdr->blob_size > size)
+      hdr->version > 3 || hdr->version == 2 /* retired flags layout */ ||
+      hdr->blob_size > size)
     return -5;
   const auto* base = static_cast<const uint8_t*>(blob);
   out->hdr = hdr;

@@ -208,7 +208,9 @@ def test_block_every_bitwidth_roundtrip():
 
 
 def test_v2_norm_stream_matches_column():
-    """v2 segments embed per-block norm streams (flags = freq-block size);
+    """v2 segments embed per-block norm streams (flags carries either the
+    packed bit widths for the fused shape or the freq-block size —
+    sdb_format.h);
     decoded values must equal the norm column entries for the block's docs —
     the index-build-time materialization of the reference's norm-column
     gather (DESIGN.md)."""
@@ -223,7 +225,7 @@ def test_v2_norm_stream_matches_column():
     blob = sa.build_segment(doc_count, [(docs, freqs)], norms)
     buf = np.frombuffer(blob, dtype=np.uint8)
     hdr = np.frombuffer(blob[:16], dtype=np.uint32)
-    assert hdr[2] == 2, "expect format v2"
+    assert hdr[2] == 3, "expect format v3"
 
     class _View(CT.Structure):
         _fields_ = [("hdr", CT.c_void_p), ("terms", CT.c_void_p),
@@ -244,8 +246,17 @@ def test_v2_norm_stream_matches_column():
         d = struct.unpack("<IIIIHHII", blob[off_desc + 28 * b:
                                             off_desc + 28 * b + 28])
         prev, last, doc_off, freq_off, length, flags = d[:6]
-        norm_payload = blob[off_payload + freq_off + flags:
-                            off_payload + freq_off + flags + 600]
+        if flags & 1:  # fused: freq-block size = 1 + 16*fbits
+            fsize = 1 + 16 * ((flags >> 6) & 31)
+            # the carried widths must match the payload tags
+            assert blob[off_payload + doc_off] - 8 + 2 == (flags >> 1) & 31
+            assert blob[off_payload + freq_off] - 5 + 1 == (flags >> 6) & 31
+            assert (blob[off_payload + freq_off + fsize] - 5 + 1 ==
+                    (flags >> 11) & 31)
+        else:
+            fsize = flags >> 1
+        norm_payload = blob[off_payload + freq_off + fsize:
+                            off_payload + freq_off + fsize + 600]
         dec, _ = po.decode_freq_block(norm_payload, length)
         np.testing.assert_array_equal(dec, norms[docs[pos:pos + length]])
         pos += length
@@ -283,4 +294,8 @@ def test_segment_blob_format_stability():
         "and bump the format version)")
 
 
-GOLDEN_BLOB_SHA = "f5687e1d73b2af986569ff30951718cbde6b4c6b399b92bb0d151bbbd4bb0d62"
+# v3 (round 2): descriptor flags carry the fused-shape bit + packed bit
+# widths (sdb_format.h) — deliberate container change, version bumped 2->3;
+# the PAYLOAD block encodings (the reference-pinned bytes) are unchanged
+# and separately pinned by the codec round-trip + simdcomp tests above.
+GOLDEN_BLOB_SHA = "15b95522b8b8eb2146c1609af94acef00ceaf542e02dcadb8eea8de819ea4fe7"
