@@ -897,7 +897,7 @@ void scan_agg_hash_kernel(HashAggArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned long long* lkey = (unsigned long long*)smem;     // lds_slots
   unsigned long long* lacc = lkey + a.lds_slots;            // slots*naggs
-  const uint32_t S_mask = a.lds_slots - 1;
+  const uint32_t S_mask = a.lds_slots ? a.lds_slots - 1 : 0;
   for (uint32_t i = threadIdx.x; i < a.lds_slots * (1 + a.naggs);
        i += SCAN_NTHREADS)
     lkey[i] = i < a.lds_slots ? SDB_HKEY_EMPTY : 0ull;
@@ -1398,12 +1398,17 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
   while (cap < 2 * max_groups) cap <<= 1;
   a.gcap_mask = (uint32_t)(cap - 1);
   a.max_groups = (uint32_t)max_groups;
-  // LDS: largest pow2 slot count whose (key + naggs accumulators) fit
+  // LDS: largest pow2 slot count whose (key + naggs accumulators) fit.
+  // When the distinct-key bound dwarfs the LDS table, nearly every row
+  // would burn a full probe chain before falling through to the global
+  // table anyway (measured 3.3G rows/s at 100k groups vs 67.9G dense):
+  // skip LDS staging entirely and upsert straight into the global table.
   uint32_t slots = 1u << 14;
   while (slots && (uint64_t)slots * 8 * (1 + naggs) > 150 * 1024)
     slots >>= 1;
+  if (max_groups > (uint64_t)slots * 2) slots = 0;
   a.lds_slots = slots;
-  const size_t lds = (size_t)slots * 8 * (1 + naggs);
+  const size_t lds = slots ? (size_t)slots * 8 * (1 + naggs) : 16;
 
   unsigned long long* d_keys = nullptr;
   unsigned long long* d_acc = nullptr;
