@@ -349,21 +349,32 @@ __device__ __forceinline__ void mark_match_mask(void* cnt, uint32_t off) {
 // three serialized ones), keep the two values per lane in registers, score
 // directly — no LDS scratch round trip. Returns false for any other family
 // combination (caller falls back to the generic per-stream decode).
+#ifdef SDB_SWEEP_STAGE
+// LDS-DMA payload staging (sweep kernel): per (wave, term<=4) slot holding
+// one fused block's raw payload bytes, filled with global_load_lds_dwordx4
+// during the PREVIOUS window's phases so next window's first-block decode
+// reads LDS instead of riding an exposed HBM round trip.
+#define SDB_STG_TERMS 4
+#define SDB_STG_CAP 352  // 22 x 16 B; fused span 3+16*(db+fb+nb) + align pad
+#endif
+
 template <int LEAN>
 __device__ __forceinline__ bool try_block_fused(
-  const uint8_t* pl, const SdbBlockDesc& d, int lane, uint32_t norm_stream,
-  uint32_t lo, uint32_t hi, float num, float nc, float nl, uint32_t scorer,
+  const uint8_t* db0 /* doc-block base: global payload or an LDS-staged
+                        copy (the three streams are contiguous) */,
+  const SdbBlockDesc& d, int lane, uint32_t norm_stream, uint32_t lo,
+  uint32_t hi, float num, float nc, float nl, uint32_t scorer,
   const uint32_t* norms_col, const float* fboost, float* swin,
   void* cwin) {
-  // v2 descriptors carry the fused-shape bit + all three bit widths, so
+  // v3 descriptors carry the fused-shape bit + all three bit widths, so
   // every packed-word load below issues with NO payload-tag fetch first
   // (one whole memory round trip per block saved; sdb_format.h flags)
   if (!(d.flags & 1u) || !norm_stream) return false;
   const uint32_t dbits = (d.flags >> 1) & 31u;
   const uint32_t fbits = (d.flags >> 6) & 31u;
   const uint32_t nbits = (d.flags >> 11) & 31u;
-  const uint8_t* db = pl + d.doc_off;
-  const uint8_t* fb = pl + d.freq_off;
+  const uint8_t* db = db0;
+  const uint8_t* fb = db0 + (d.freq_off - d.doc_off);
   const uint8_t* nb = fb + 1 + 16u * fbits;  // bitpack size = 1 + 16*bits
   const uint32_t i0 = 2u * lane, i1 = i0 + 1;
   // all loads issue here, before any cross-lane dependency
@@ -789,8 +800,9 @@ void topk_window_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
         }
 #else
 #ifndef SDB_ABLATE_SCORE
-        if (try_block_fused<0>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
-                               nl, a.scorer, a.norms, a.fb, swin, cwin)) {
+        if (try_block_fused<0>(pl + d.doc_off, d, lane, a.norm_stream, lo,
+                               hi, num, nc, nl, a.scorer, a.norms, a.fb,
+                               swin, cwin)) {
           b += SDB_NWAVES;  // while-loop: explicit advance before continue
           continue;
         }
@@ -1105,10 +1117,21 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
   unsigned long long* lbuck =
     (unsigned long long*)(tstage + SDB_MAX_TERMS);  // 2 * max buckets
   SdbBlockDesc* dcache = (SdbBlockDesc*)(lbuck + 2 * SDB_MAX_BUCKETS);
+#ifdef SDB_SWEEP_STAGE
+  // slot metadata: (align<<28)|block-index-within-term, ~0u = empty; the
+  // slot content is ALWAYS the payload of exactly the block the index
+  // names, so a stale entry that still matches the cursor is a valid hit
+  uint32_t* smeta = (uint32_t*)&dcache[a.nterms * a.dcache_n];
+  uint8_t* stg = (uint8_t*)(((uintptr_t)(smeta + NW * SDB_STG_TERMS) + 15) &
+                            ~(uintptr_t)15);
+#endif
 
   const uint32_t tid = threadIdx.x;
   const int lane = tid & 63;
   const uint32_t wave = tid >> 6;
+#ifdef SDB_SWEEP_STAGE
+  if (tid < NW * SDB_STG_TERMS) smeta[tid] = ~0u;
+#endif
   // stage the term table through the vector path once: per-phase reads
   // of the global term table are wave-uniform SMEM loads whose lgkmcnt
   // sharing with LDS ops stalls every later LDS wait (see load_tags3)
@@ -1226,6 +1249,11 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       }
     }
     __syncthreads();
+#ifdef SDB_SWEEP_STAGE
+    // the previous window's global_load_lds writes must be visible before
+    // any staged-slot read below (normally already drained at a barrier)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#endif
     SDB_TS(0)
     float wand_total_ub = 0.0f;
     if (a.wand)
@@ -1288,6 +1316,67 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             continue;
           }
         }
+#ifdef SDB_SWEEP_STAGE
+        bool stg_hit = false;
+        if (rel == wave && t < SDB_STG_TERMS) {
+          // this wave's FIRST block of the phase: consume the staged copy
+          // if the slot holds exactly this block, then re-arm the slot
+          // with next window's first block (slot is wave-private)
+          uint32_t* mslot = &smeta[wave * SDB_STG_TERMS + t];
+          const uint32_t mv = *mslot;
+          if (((mv ^ (cur0 + wave)) & 0x0FFFFFFFu) == 0 && mv != ~0u) {
+            const uint8_t* db0 =
+              stg + (wave * SDB_STG_TERMS + t) * SDB_STG_CAP + (mv >> 28);
+            if (try_block_fused<1>(db0, d, lane, a.norm_stream, lo, hi,
+                                   num, nc, nl, a.scorer, a.norms, a.fb,
+                                   swin, mwin))
+              stg_hit = true;
+          }
+          // re-stage: predict this term's post-window cursor from the
+          // staged last_docs (wave-parallel ballot), pick next first block
+          uint32_t nmeta = ~0u;
+          const uint64_t rem64 = dend - (te.desc_begin + cur0);
+          const uint32_t avail =
+            rem64 < a.dcache_n ? (uint32_t)rem64 : a.dcache_n;
+          uint32_t lastd = 0xFFFFFFFFu;
+          if ((uint32_t)lane < avail)
+            lastd = dcache[t * a.dcache_n + lane].last_doc;
+          const unsigned long long inwin = __ballot(lastd <= hi);
+          const uint32_t adv = (uint32_t)__builtin_ctzll(~inwin);
+          const uint32_t nrel = adv + wave;
+          if (adv < avail && nrel < avail) {
+            const SdbBlockDesc dn = dcache[t * a.dcache_n + nrel];
+            if (dn.flags & 1u) {
+              const uint32_t fbits = (dn.flags >> 6) & 31u;
+              const uint32_t nbits = (dn.flags >> 11) & 31u;
+              const uint32_t span = (uint32_t)(dn.freq_off - dn.doc_off) +
+                                    2u + 16u * (fbits + nbits);
+              const uint32_t al = (uint32_t)(dn.doc_off & 15u);
+              const uint32_t tot = al + span;
+              // clamp: the 16 B-aligned copy may read up to 15 B past the
+              // span — only safe when another block follows in this term
+              if (tot <= SDB_STG_CAP &&
+                  te.desc_begin + cur0 + nrel + 1 < dend) {
+                const uint8_t* gsrc = (const uint8_t*)(
+                  (uintptr_t)(pl + dn.doc_off) & ~(uintptr_t)15);
+                uint8_t* ldst =
+                  stg + (wave * SDB_STG_TERMS + t) * SDB_STG_CAP;
+                const uint32_t nv = (tot + 15u) >> 4;
+                if ((uint32_t)lane < nv)
+                  __builtin_amdgcn_global_load_lds(
+                    (const uint32_t*)(gsrc + 16 * lane), (uint32_t*)ldst,
+                    16, 0, 0);
+                nmeta = (al << 28) | ((cur0 + nrel) & 0x0FFFFFFFu);
+              }
+            }
+          }
+          if (lane == 0) *mslot = nmeta;
+          if (stg_hit) {
+            b += NW;
+            continue;
+          }
+        }
+#endif
 #ifdef SDB_ABLATE_NOLOAD
         // perf ablation: fabricate postings with NO payload access at all
         // (no fused pair, no prefetch; results are WRONG)
@@ -1344,8 +1433,9 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
             }
           }
         }
-        if (try_block_fused<1>(pl, d, lane, a.norm_stream, lo, hi, num, nc,
-                               nl, a.scorer, a.norms, a.fb, swin, mwin)) {
+        if (try_block_fused<1>(pl + d.doc_off, d, lane, a.norm_stream, lo,
+                               hi, num, nc, nl, a.scorer, a.norms, a.fb,
+                               swin, mwin)) {
           b += NW;
           continue;
         }
@@ -1372,7 +1462,16 @@ void topk_sweep_kernel(WindowArgs a, const TermDev* __restrict__ terms) {
       if (t == 0) SDB_TS(5)  // term-0 phase alone (imbalance diagnosis)
 #endif
 #ifndef SDB_ABLATE_NOBARRIER  // perf ablation: cost of term serialization
+#ifdef SDB_SWEEP_STAGE
+      // raw barrier + lgkmcnt-only wait: __syncthreads() would drain the
+      // in-flight global_load_lds with vmcnt(0) at EVERY phase boundary
+      // (guide: pipelining across barriers); LDS visibility for the
+      // term-major merge order needs only lgkmcnt
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+#else
       __syncthreads();  // term-major merge order (bit-exact vs oracle)
+#endif
 #endif
     }
 #ifdef SDB_ABLATE_NOBARRIER
@@ -2195,9 +2294,16 @@ bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
 
 // fixed (non-dcache) LDS bytes of one sweep workgroup
 size_t sweep_lds_fixed(const SweepGeom& g) {
-  return (size_t)g.wd * 4 + g.wd / 8 + (g.nth / 64) * 1536 +
-         SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 + 2 * SDB_MAX_TERMS * 4 +
-         sizeof(TermDev) * SDB_MAX_TERMS + 16 * SDB_MAX_BUCKETS;
+  size_t b = (size_t)g.wd * 4 + g.wd / 8 + (g.nth / 64) * 1536 +
+             SDB_HIST_BINS * 4 + (2 + g.nth / 64) * 4 +
+             2 * SDB_MAX_TERMS * 4 + sizeof(TermDev) * SDB_MAX_TERMS +
+             16 * SDB_MAX_BUCKETS;
+#ifdef SDB_SWEEP_STAGE
+  b += (g.nth / 64) * SDB_STG_TERMS * 4   // smeta
+       + 16                               // alignment pad
+       + (g.nth / 64) * SDB_STG_TERMS * SDB_STG_CAP;
+#endif
+  return b;
 }
 
 int check_gpu() {
