@@ -2447,7 +2447,7 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_overflow, 4));
   CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS *
                                        SDB_TERM_SLOTS));
-  CTX_CHECK(hipHostMalloc(&ctx->h_counts, 8));
+  CTX_CHECK(hipHostMalloc(&ctx->h_counts, 16));  // count, ovf, final bin
   CTX_CHECK(hipHostMalloc(&ctx->h_matches, 8));
   CTX_CHECK(hipEventCreate(&ctx->ev_a));
   CTX_CHECK(hipEventCreate(&ctx->ev_b));
@@ -2929,6 +2929,8 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                            hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipMemcpyAsync(ctx->h_counts + 1, ctx->d_overflow, 4,
                            hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(ctx->h_counts + 2, ctx->d_gthresh, 4,
+                           hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipMemcpyAsync(ctx->h_matches, ctx->d_total_matches, 8,
                            hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
@@ -2947,10 +2949,21 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     return SDB_OK;
   }
   const auto t_rb0 = std::chrono::steady_clock::now();
-  std::vector<SdbScoreDoc> cands(ncand);
-  if (ncand) {
-    HIP_CHECK(hipMemcpy(cands.data(), ctx->d_cands,
+  // pinned staging when it fits (the common case): async DMA instead of
+  // the pageable bounce path; filter + select run in place on the pinned
+  // buffer. Oversized candidate sets fall back to a pageable vector.
+  std::vector<SdbScoreDoc> cands_heap;
+  SdbScoreDoc* cands = ctx->h_cands_pin;
+  if (ncand && ncand <= SDB_PIN_CANDS) {
+    HIP_CHECK(hipMemcpyAsync(ctx->h_cands_pin, ctx->d_cands,
+                             sizeof(SdbScoreDoc) * ncand,
+                             hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  } else if (ncand) {
+    cands_heap.resize(ncand);
+    HIP_CHECK(hipMemcpy(cands_heap.data(), ctx->d_cands,
                         sizeof(SdbScoreDoc) * ncand, hipMemcpyDeviceToHost));
+    cands = cands_heap.data();
   }
   ctx->last_readback_ms =
     std::chrono::duration<double, std::milli>(
@@ -2962,18 +2975,17 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   // windows appended against a weaker bin; every true top-k member has bin
   // >= the final bin — see score_bin — so this drops no top-k member),
   // then exact select under (score desc, seg, doc). The binning expression
-  // is the identical f32 multiply+truncate the kernel used.
-  uint32_t final_bin = 0;
-  HIP_CHECK(hipMemcpy(&final_bin, ctx->d_gthresh, 4, hipMemcpyDeviceToHost));
+  // is the identical f32 multiply+truncate the kernel used; the final bin
+  // came back batched with the counts (h_counts[2]), no extra sync copy.
+  const uint32_t final_bin = ctx->h_counts[2];
   const float inv_smax = (float)SDB_HIST_BINS / smax;
   size_t n = 0;
-  for (size_t i = 0; i < cands.size(); ++i) {
+  for (size_t i = 0; i < ncand; ++i) {
     const float s = cands[i].score;
     uint32_t sb = (uint32_t)(s * inv_smax);
     if (sb >= SDB_HIST_BINS) sb = SDB_HIST_BINS - 1;
     if (sb >= final_bin && s > FLT_MIN) cands[n++] = cands[i];
   }
-  cands.resize(n);
   const float gtau_final =
     final_bin ? (float)final_bin / inv_smax : 0.0f;  // diagnostic only
   auto cmp = [](const SdbScoreDoc& x, const SdbScoreDoc& y) {
@@ -2981,11 +2993,10 @@ static int exec_topk_impl(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     if (x.segment_idx != y.segment_idx) return x.segment_idx < y.segment_idx;
     return x.doc < y.doc;
   };
-  const size_t kk = std::min<size_t>(k, cands.size());
-  if (kk < cands.size())
-    std::nth_element(cands.begin(), cands.begin() + kk, cands.end(), cmp);
-  std::sort(cands.begin(), cands.begin() + kk, cmp);
-  std::copy(cands.begin(), cands.begin() + kk, hits);
+  const size_t kk = std::min<size_t>(k, n);
+  if (kk < n) std::nth_element(cands, cands + kk, cands + n, cmp);
+  std::sort(cands, cands + kk, cmp);
+  std::copy(cands, cands + kk, hits);
   *out_count = (uint32_t)kk;
   *total_matches = *ctx->h_matches;
 #ifdef SDB_TIMING

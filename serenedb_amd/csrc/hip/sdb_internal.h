@@ -24,7 +24,7 @@ struct SdbGpuCtx {
   unsigned long long* d_buckets;  // hybrid bucket aggregates [2*128]
   TermDev* d_terms;
   uint32_t* d_overflow;
-  uint32_t* h_counts;  // pinned: [cand_count, overflow]
+  uint32_t* h_counts;  // pinned: [cand_count, overflow, final_bin]
   unsigned long long* h_matches;
   hipEvent_t ev_a, ev_b;   // bracket the window kernels of one execute call
   // pipelined-batch state (sdb_gpu_execute_topk_batch): second query-state
