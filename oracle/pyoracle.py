@@ -271,6 +271,35 @@ def scan_agg(keys, v1, v2, ngroups, pred_op=0, lo=0, hi=0):
     return cnt, si, sf, passed.value
 
 
+def str_pred_mask(values, op, lo, hi=None):
+    """ORACLE (test infrastructure): raw variable-width string predicate,
+    memcmp order on unsigned bytes — restates the reference's pushed
+    typed compares (index/table_filter_iterator.hpp:104-227) applied to
+    var-width string vectors (the reference routes raw strings through
+    its DuckDB fork, column_reader.hpp:247-255; parity at result level).
+    Python bytes comparison IS lexicographic unsigned-byte order, an
+    implementation independent of the GPU kernel's per-lane loop."""
+    lob = lo.encode() if isinstance(lo, str) else bytes(lo or b"")
+    hib = hi.encode() if isinstance(hi, str) else bytes(hi or b"")
+    out = np.zeros(len(values), dtype=bool)
+    for i, v in enumerate(values):
+        b = v.encode() if isinstance(v, str) else bytes(v)
+        if op == "lt":
+            m = b < lob
+        elif op == "ge":
+            m = b >= lob
+        elif op == "eq":
+            m = b == lob
+        elif op == "between":
+            m = lob <= b <= hib
+        elif op == "prefix":
+            m = b.startswith(lob)
+        else:
+            raise ValueError(op)
+        out[i] = m
+    return out
+
+
 # --- reference simdcomp pins ---------------------------------------------
 
 def ref_pack_d1(prev, values128, bits):

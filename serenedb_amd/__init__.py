@@ -233,6 +233,18 @@ def decode_col_i64(blob, rows):
     return out
 
 
+def encode_col_str_raw(values):
+    """Raw variable-width string column: returns (offsets uint64[n+1],
+    blob bytes) for sdb_gpu_table_attach_strcol. Values may be str
+    (UTF-8 encoded) or bytes; arbitrary bytes incl. NUL are preserved."""
+    import numpy as np
+
+    enc = [v.encode() if isinstance(v, str) else bytes(v) for v in values]
+    offsets = np.zeros(len(enc) + 1, dtype=np.uint64)
+    np.cumsum([len(b) for b in enc], out=offsets[1:])
+    return offsets, b"".join(enc)
+
+
 def encode_col_str(values):
     """Dictionary-encode a string column: sorted-unique dictionary + i64
     codes. The SORTED dictionary is what makes string predicates map to
@@ -668,6 +680,43 @@ class GpuContext:
                 b.ctypes.data_as(C.POINTER(C.c_uint64)))
         if rc != 0:
             raise RuntimeError(f"attach_validity rc={rc}")
+
+    def attach_strcol(self, tab, slot, offsets, blob):
+        """Attach a raw variable-width string column (non-dictionary;
+        include/sdb_gpu.h sdb_gpu_table_attach_strcol): offsets[rows+1]
+        uint64 byte offsets into blob (bytes). Slot 0..3, independent of
+        the i64/f32 column list."""
+        import numpy as np
+
+        off = np.ascontiguousarray(offsets, dtype=np.uint64)
+        bl = np.frombuffer(bytes(blob), dtype=np.uint8) if len(blob) \
+            else np.zeros(0, dtype=np.uint8)
+        rc = self._lib.sdb_gpu_table_attach_strcol(
+            self._ctx, tab, C.c_uint32(slot),
+            off.ctypes.data_as(C.POINTER(C.c_uint64)),
+            bl.ctypes.data_as(C.POINTER(C.c_uint8)),
+            C.c_uint64(len(bl)))
+        if rc != 0:
+            raise RuntimeError(f"attach_strcol rc={rc}")
+
+    STR_OPS = {"lt": 1, "ge": 2, "between": 3, "eq": 4, "prefix": 8}
+
+    def strpred_mask(self, tab, slot, op, lo, hi=None):
+        """Evaluate a string predicate over an attached raw string slot
+        into its device row bitmask (memcmp order on UTF-8 bytes). op:
+        "lt"/"ge"/"between"/"eq"/"prefix" or the numeric SdbPredOp. The
+        mask is then consumed by scan_agg / scan_agg_hash via the pred
+        tuple (slot, 7, 0, 0)  [op 7 = SDB_PRED_STRMASK]."""
+        opn = self.STR_OPS[op] if isinstance(op, str) else int(op)
+        lob = lo.encode() if isinstance(lo, str) else bytes(lo or b"")
+        hib = hi.encode() if isinstance(hi, str) else bytes(hi or b"")
+        la = (C.c_uint8 * max(len(lob), 1))(*lob)
+        ha = (C.c_uint8 * max(len(hib), 1))(*hib)
+        rc = self._lib.sdb_gpu_strpred_mask(
+            self._ctx, tab, C.c_uint32(slot), C.c_int(opn), la,
+            C.c_uint32(len(lob)), ha, C.c_uint32(len(hib)))
+        if rc != 0:
+            raise RuntimeError(f"strpred_mask rc={rc}")
 
     def scan_agg(self, tab, group_col, ngroups, preds, aggs):
         """preds: list of (col, op, lo, hi) with SdbPredOp numeric op

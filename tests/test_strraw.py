@@ -1,0 +1,185 @@
+"""Raw variable-width string columns (VERDICT r1 missing #5; SURVEY.md
+§8f row 3 second half — the dictionary path shipped round 1).
+
+Storage: offsets[rows+1] + byte blob attached to a table slot
+(sdb_gpu_table_attach_strcol). Predicates (LT/GE/BETWEEN/EQ lexicographic
+on unsigned bytes = memcmp order, PREFIX starts-with) evaluate once on
+device into a row bitmask (sdb_gpu_strpred_mask); scans consume the mask
+through pred (slot, 7=SDB_PRED_STRMASK, 0, 0). Oracle: pyoracle.
+str_pred_mask (python-bytes comparison — an independent implementation).
+"""
+import numpy as np
+import pytest
+
+import serenedb_amd as sa
+from oracle import pyoracle as po
+
+
+def make_raw_strings(seed, n):
+    """mixed-length strings incl. empties, NUL bytes, shared prefixes"""
+    rng = np.random.default_rng(seed)
+    vocab = []
+    for c in "abcdefgh":
+        for i in range(40):
+            vocab.append(f"{c}{i:03d}")
+            vocab.append(f"{c}{i:03d}-suffix")
+    vocab += ["", "a", "a\x00b", "a\x00", "zz", "zzzz", "\x00lead"]
+    return [vocab[i] for i in rng.integers(0, len(vocab), n)]
+
+
+CASES = [
+    ("eq", "c017", None),
+    ("eq", "c017-suffix", None),
+    ("eq", "", None),            # empty string rows
+    ("eq", "a\x00b", None),      # embedded NUL
+    ("eq", "zzzzz", None),       # absent
+    ("prefix", "d0", None),
+    ("prefix", "", None),        # everything
+    ("prefix", "a\x00", None),   # NUL inside prefix
+    ("lt", "c000", None),
+    ("lt", "", None),            # nothing (no string < "")
+    ("ge", "f031", None),
+    ("between", "b000", "c999"),
+    ("between", "x", "y"),       # empty range
+    ("between", "a", "a\x00b"),  # NUL-byte upper bound
+]
+
+
+def naive_mask(vals, op, lo, hi):
+    """second independent CPU evaluation (string-space, not bytes-space
+    where possible) to pin the oracle itself"""
+    out = np.zeros(len(vals), dtype=bool)
+    for i, v in enumerate(vals):
+        if op == "eq":
+            out[i] = v == lo
+        elif op == "prefix":
+            out[i] = v.startswith(lo)
+        elif op == "lt":
+            out[i] = v.encode() < lo.encode()
+        elif op == "ge":
+            out[i] = v.encode() >= lo.encode()
+        else:
+            out[i] = lo.encode() <= v.encode() <= hi.encode()
+    return out
+
+
+def test_oracle_str_pred_matches_naive():
+    vals = make_raw_strings(7, 4000)
+    for op, lo, hi in CASES:
+        np.testing.assert_array_equal(
+            po.str_pred_mask(vals, op, lo, hi), naive_mask(vals, op, lo, hi),
+            err_msg=f"{op} {lo!r} {hi!r}")
+
+
+def test_encode_col_str_raw_roundtrip():
+    vals = make_raw_strings(8, 1000)
+    off, blob = sa.encode_col_str_raw(vals)
+    assert off[0] == 0 and off[-1] == len(blob)
+    for i, v in enumerate(vals):
+        assert blob[int(off[i]):int(off[i + 1])] == v.encode()
+
+
+@pytest.mark.gpu
+def test_strpred_mask_gpu_parity():
+    """device mask == oracle mask for every op/edge case; mask consumed
+    by scan_agg (dense) and scan_agg_hash, aggregates exact vs numpy."""
+    rows = 200_000
+    vals = make_raw_strings(11, rows)
+    rng = np.random.default_rng(12)
+    keys = rng.integers(0, 64, rows).astype(np.int64)
+    v1 = rng.integers(-1 << 30, 1 << 30, rows).astype(np.int64)
+    v2 = rng.normal(0, 1, rows).astype(np.float32)
+    off, blob = sa.encode_col_str_raw(vals)
+
+    ctx = sa.GpuContext(0)
+    tab = ctx.load_table([keys, v1, v2])
+    ctx.attach_strcol(tab, 0, off, blob)
+
+    for op, lo, hi in CASES:
+        ctx.strpred_mask(tab, 0, op, lo, hi)
+        exp = po.str_pred_mask(vals, op, lo, hi)
+        i64, f64, passed = ctx.scan_agg(
+            tab, 0, 64, [(0, 7, 0, 0)], [(0, 0), (1, 1), (2, 2)])
+        assert passed == int(exp.sum()), (op, lo, hi)
+        exp_cnt = np.bincount(keys[exp], minlength=64)
+        exp_si = np.bincount(keys[exp], weights=v1[exp].astype(np.float64),
+                             minlength=64)
+        np.testing.assert_array_equal(i64[:, 0], exp_cnt, err_msg=str(op))
+        # SUM_I64 exact (wrap-around): recompute exactly in python ints
+        exp_si_exact = np.zeros(64, dtype=np.int64)
+        np.add.at(exp_si_exact, keys[exp], v1[exp])
+        np.testing.assert_array_equal(i64[:, 1], exp_si_exact,
+                                      err_msg=str(op))
+        exp_sf = np.bincount(keys[exp], weights=v2[exp].astype(np.float64),
+                             minlength=64)
+        np.testing.assert_allclose(f64[:, 2], exp_sf, rtol=1e-10,
+                                   atol=1e-7, err_msg=str(op))
+        del exp_si
+
+    # string pred AND numeric pred together
+    ctx.strpred_mask(tab, 0, "prefix", "c", None)
+    exp = po.str_pred_mask(vals, "prefix", "c", None) & (v1 >= 0)
+    i64, _, passed = ctx.scan_agg(tab, 0, 64, [(0, 7, 0, 0), (1, 2, 0, 0)],
+                                  [(0, 0)])
+    assert passed == int(exp.sum())
+    np.testing.assert_array_equal(
+        i64[:, 0], np.bincount(keys[exp], minlength=64))
+
+    # hash-aggregate path consumes the same mask
+    kh, hi64, _, hpassed = ctx.scan_agg_hash(tab, 0, 64, [(0, 7, 0, 0)],
+                                             [(0, 0)])
+    exp = po.str_pred_mask(vals, "prefix", "c", None)
+    assert hpassed == int(exp.sum())
+    cnt = np.bincount(keys[exp], minlength=64)
+    present = np.flatnonzero(cnt)
+    np.testing.assert_array_equal(kh, present)
+    np.testing.assert_array_equal(hi64[:, 0], cnt[present])
+
+    # string mask + validity on the agg column (SUM skips NULLs)
+    valid = rng.integers(0, 2, rows).astype(bool)
+    vb = np.zeros((rows + 63) // 64, dtype=np.uint64)
+    idx = np.flatnonzero(valid)
+    np.bitwise_or.at(vb, idx >> 6, np.uint64(1) << (idx & 63).astype(np.uint64))
+    ctx.attach_validity(tab, 1, vb)
+    i64, _, passed = ctx.scan_agg(tab, 0, 64, [(0, 7, 0, 0)],
+                                  [(0, 0), (1, 1)])
+    assert passed == int(exp.sum())  # COUNT(*) counts masked rows
+    exp_si = np.zeros(64, dtype=np.int64)
+    sel = exp & valid
+    np.add.at(exp_si, keys[sel], v1[sel])
+    np.testing.assert_array_equal(i64[:, 1], exp_si)
+    ctx.attach_validity(tab, 1, None)
+
+    ctx.free_table(tab)
+
+
+@pytest.mark.gpu
+def test_strcol_error_paths():
+    rows = 1000
+    vals = make_raw_strings(13, rows)
+    rng = np.random.default_rng(14)
+    keys = rng.integers(0, 8, rows).astype(np.int64)
+    off, blob = sa.encode_col_str_raw(vals)
+    ctx = sa.GpuContext(0)
+    tab = ctx.load_table([keys])
+
+    # scan with STRMASK before any mask computed -> rejected
+    ctx.attach_strcol(tab, 0, off, blob)
+    with pytest.raises(RuntimeError):
+        ctx.scan_agg(tab, 0, 8, [(1, 7, 0, 0)], [(0, 0)])  # slot 1 unset
+    # strpred on an unattached slot -> rejected
+    with pytest.raises(RuntimeError):
+        ctx.strpred_mask(tab, 2, "eq", "x")
+    # literal over 63 bytes -> rejected
+    with pytest.raises(RuntimeError):
+        ctx.strpred_mask(tab, 0, "eq", "x" * 64)
+    # non-monotone offsets -> rejected at attach
+    bad = off.copy()
+    if rows > 2:
+        bad[1], bad[2] = bad[2] + 1, bad[1]
+    with pytest.raises(RuntimeError):
+        ctx.attach_strcol(tab, 1, bad, blob)
+    # offsets not spanning the blob -> rejected
+    with pytest.raises(RuntimeError):
+        ctx.attach_strcol(tab, 1, off, blob + b"extra")
+    ctx.free_table(tab)
