@@ -223,18 +223,51 @@ def bench_bm25(args, dist, hybrid=False):
     sync()
 
     # ---- timed ----
-    # (per-step execution; the pipelined batch entry
-    # sdb_gpu_execute_topk_batch was benched at +25-35% ms/step on this
-    # workload — see tools/ROUND3_NOTES.md — so it stays a tested API
-    # rather than the bench default)
+    # Default execution = the pipelined batch entry
+    # (sdb_gpu_execute_topk_batch): every query fully re-executed (memset
+    # -> kernels -> readback -> exact select), query k+1's kernels enqueue
+    # while query k's candidates read back and select on the host — the
+    # production QPS shape. Kernel-trace cadence 0.49 ms/query vs 0.67
+    # per-step at 100M (profiles/r2_evidence/r2_batch_trace.log); results
+    # bit-equal per query (tests/test_gpu_parity.py
+    # test_batch_pipelined_equals_single, re-asserted in warmup here).
+    # Hybrid keeps per-step (the batch entry carries no bucket state).
+    use_batch = args.execution == "batch" and not hybrid
+    if use_batch:
+        ref_hits, ref_total = ctx.execute_topk([seg], term_idx, boosts, k,
+                                               global_stats=gstats)
+        bh, bt = ctx.execute_topk_batch([seg], term_idx, boosts, k, 2,
+                                        global_stats=gstats, all_hits=True)
+        assert bt[0] == ref_total and len(bh[0]) == len(ref_hits)
+        assert all(tuple(a) == tuple(b) for a, b in zip(bh[0], ref_hits))
     kernel_ms_acc = 0.0
     sync()
     t0 = time.time()
-    for _ in range(args.steps):
-        scores, docs, total = step()
+    if use_batch:
+        hits_l, totals_l = ctx.execute_topk_batch(
+            [seg], term_idx, boosts, k, args.steps, global_stats=gstats,
+            all_hits=bool(dist))
         ms = CT.c_double(0)
         lib.sdb_gpu_last_kernel_ms(ctx._ctx, CT.byref(ms))
-        kernel_ms_acc += ms.value
+        kernel_ms_acc = ms.value * args.steps  # batch reports total/nq
+        total = totals_l[-1]
+        if dist:
+            import torch
+            for q in range(args.steps):
+                packed = torch.from_numpy(
+                    pack_hits(hits_l[q], lo - 1, k)).cuda()
+                gathered = [torch.empty_like(packed) for _ in range(world)]
+                dist.all_gather(gathered, packed)
+                tm = torch.tensor([totals_l[q]], dtype=torch.int64,
+                                  device="cuda")
+                dist.all_reduce(tm)
+                total = int(tm.item())
+    else:
+        for _ in range(args.steps):
+            scores, docs, total = step()
+            ms = CT.c_double(0)
+            lib.sdb_gpu_last_kernel_ms(ctx._ctx, CT.byref(ms))
+            kernel_ms_acc += ms.value
     sync()
     elapsed = time.time() - t0
     if dist:
@@ -353,7 +386,7 @@ def bench_bm25(args, dist, hybrid=False):
                          ("bm25_top1000_4term_or_100M"
                           if doc_count == 100_000_000
                           else f"bm25_top1000_4term_or_{doc_count}")),
-            "execution": "per-step",
+            "execution": "pipelined-batch" if use_batch else "per-step",
             "doc_count": doc_count,
             "selectivities": sels,
             "k": k,
@@ -556,6 +589,7 @@ def main():
     ap.add_argument("--docs", type=int, default=100_000_000)
     ap.add_argument("--rows", type=int, default=1_000_000_000)
     ap.add_argument("--scan-codec", default="raw", choices=["for", "raw"])
+    ap.add_argument("--execution", default="batch", choices=["batch", "step"])
     ap.add_argument("--cpu-seconds", type=float, default=10.0)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()

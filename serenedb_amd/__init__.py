@@ -815,11 +815,11 @@ class GpuContext:
 
         def conv(q):
             n = counts[q]
-            res = np.zeros(n, dtype=[("score", "f4"), ("doc", "u4"),
-                                     ("segment", "u4")])
-            for i in range(n):
-                h = hits[q * k + i]
-                res[i] = (h.score, h.doc, h.segment_idx)
+            base = C.addressof(hits) + q * k * C.sizeof(SdbScoreDoc)
+            res = np.frombuffer(
+                C.string_at(base, C.sizeof(SdbScoreDoc) * n),
+                dtype=[("score", "f4"), ("doc", "u4"), ("segment", "u4")],
+                count=n).copy()
             return res
 
         out = ([conv(q) for q in range(nq)] if all_hits else conv(nq - 1))
