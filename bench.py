@@ -232,21 +232,36 @@ def bench_bm25(args, dist, hybrid=False):
     # bit-equal per query (tests/test_gpu_parity.py
     # test_batch_pipelined_equals_single, re-asserted in warmup here).
     # Hybrid keeps per-step (the batch entry carries no bucket state).
-    use_batch = args.execution == "batch" and not hybrid
-    if use_batch:
+    use_batch = args.execution == "batch"
+    if use_batch and not hybrid:
         ref_hits, ref_total = ctx.execute_topk([seg], term_idx, boosts, k,
                                                global_stats=gstats)
         bh, bt = ctx.execute_topk_batch([seg], term_idx, boosts, k, 2,
                                         global_stats=gstats, all_hits=True)
         assert bt[0] == ref_total and len(bh[0]) == len(ref_hits)
         assert all(tuple(a) == tuple(b) for a, b in zip(bh[0], ref_hits))
+    elif use_batch:
+        rh, rt, rc_, rs = ctx.execute_topk_hybrid(
+            [seg], term_idx, boosts, k, flo, fhi, nbuckets,
+            global_stats=gstats)
+        bh, bt, bc, bs = ctx.execute_topk_hybrid_batch(
+            [seg], term_idx, boosts, k, flo, fhi, nbuckets, 2,
+            global_stats=gstats, all_hits=True)
+        assert bt[0] == rt and len(bh[0]) == len(rh)
+        assert all(tuple(a) == tuple(b) for a, b in zip(bh[0], rh))
+        assert (bc[0] == rc_).all() and (bs[0] == rs).all()
     kernel_ms_acc = 0.0
     sync()
     t0 = time.time()
     if use_batch:
-        hits_l, totals_l = ctx.execute_topk_batch(
-            [seg], term_idx, boosts, k, args.steps, global_stats=gstats,
-            all_hits=bool(dist))
+        if hybrid:
+            hits_l, totals_l, bc, bs = ctx.execute_topk_hybrid_batch(
+                [seg], term_idx, boosts, k, flo, fhi, nbuckets,
+                args.steps, global_stats=gstats, all_hits=bool(dist))
+        else:
+            hits_l, totals_l = ctx.execute_topk_batch(
+                [seg], term_idx, boosts, k, args.steps,
+                global_stats=gstats, all_hits=bool(dist))
         ms = CT.c_double(0)
         lib.sdb_gpu_last_kernel_ms(ctx._ctx, CT.byref(ms))
         kernel_ms_acc = ms.value * args.steps  # batch reports total/nq
@@ -262,6 +277,11 @@ def bench_bm25(args, dist, hybrid=False):
                                   device="cuda")
                 dist.all_reduce(tm)
                 total = int(tm.item())
+                if hybrid:
+                    bb = torch.from_numpy(
+                        np.concatenate([bc[q], bs[q]])).cuda()
+                    dist.all_reduce(bb)  # partial-agg merge
+                    _ = bb.cpu().numpy()
     else:
         for _ in range(args.steps):
             scores, docs, total = step()

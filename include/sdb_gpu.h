@@ -167,6 +167,17 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                                uint32_t k, uint32_t nq, SdbScoreDoc* hits,
                                uint32_t* out_counts, uint64_t* totals);
 
+/* pipelined-batch form of the hybrid entry (same per-query semantics as
+ * sdb_gpu_execute_topk_hybrid): bucket_counts/bucket_sums hold
+ * nq * nbuckets rows, query-major. Bucket state is double-buffered by
+ * query parity so query k+1's kernels overlap query k's readback. */
+int sdb_gpu_execute_topk_hybrid_batch(
+  SdbGpuCtx* ctx, SdbGpuSegment* const* segs, uint32_t nsegs,
+  const SdbQueryPlan* plan, uint32_t k, int64_t flo, int64_t fhi,
+  uint32_t nbuckets, uint32_t nq, int64_t* bucket_counts,
+  int64_t* bucket_sums, SdbScoreDoc* hits, uint32_t* out_counts,
+  uint64_t* totals);
+
 /* CountFast — exact match count without scoring (docs-only decode;
  * DecideScanMode Count/CountFast, duckdb_search_full_scan.cpp:972). */
 int sdb_gpu_execute_count(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,

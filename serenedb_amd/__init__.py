@@ -496,6 +496,44 @@ class GpuContext:
                             dtype=dt).copy() if n else np.zeros(0, dtype=dt)
         return res, total.value, bcnt, bsum
 
+    def execute_topk_hybrid_batch(self, segs, term_idx, boosts, k, flo,
+                                  fhi, nbuckets, nq, min_match=1, k1=1.2,
+                                  b=0.75, global_stats=None,
+                                  all_hits=False):
+        """Pipelined batch of nq identical hybrid queries (each fully
+        re-executed; per-query semantics == execute_topk_hybrid).
+        Returns (hits, totals[nq], bcnt [nq, nbuckets], bsum [nq,
+        nbuckets])."""
+        import numpy as np
+
+        plan = self._make_plan(term_idx, boosts, min_match, k1, b,
+                               global_stats)
+        seg_arr = (C.c_void_p * len(segs))(
+            *[C.c_void_p(s.value) for s in segs])
+        hits = (SdbScoreDoc * (nq * k))()
+        counts = (C.c_uint32 * nq)()
+        totals = (C.c_uint64 * nq)()
+        bcnt = np.zeros((nq, nbuckets), dtype=np.int64)
+        bsum = np.zeros((nq, nbuckets), dtype=np.int64)
+        PI64 = C.POINTER(C.c_int64)
+        rc = self._lib.sdb_gpu_execute_topk_hybrid_batch(
+            self._ctx, seg_arr, len(segs), C.byref(plan), C.c_uint32(k),
+            C.c_int64(flo), C.c_int64(fhi), C.c_uint32(nbuckets),
+            C.c_uint32(nq), bcnt.ctypes.data_as(PI64),
+            bsum.ctypes.data_as(PI64), hits, counts, totals)
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_execute_topk_hybrid_batch rc={rc}")
+        dt = np.dtype([("score", "f4"), ("doc", "u4"), ("segment", "u4")])
+
+        def conv(q):
+            n = counts[q]
+            base = C.addressof(hits) + q * k * C.sizeof(SdbScoreDoc)
+            return np.frombuffer(C.string_at(base, 12 * n), dtype=dt,
+                                 count=n).copy()
+
+        out = ([conv(q) for q in range(nq)] if all_hits else conv(nq - 1))
+        return out, [int(totals[q]) for q in range(nq)], bcnt, bsum
+
     def execute_topk_hybrid_chain(self, segs, term_idx, boosts, k, preds,
                                   nbuckets, min_match=1, k1=1.2, b=0.75,
                                   global_stats=None):

@@ -2443,7 +2443,9 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
   CTX_CHECK(hipMalloc(&ctx->d_total_matches, 8));
   CTX_CHECK(hipMalloc(&ctx->d_gthresh, 4));
   CTX_CHECK(hipMalloc(&ctx->d_ghist, 4 * SDB_HIST_BINS * 16));
-  CTX_CHECK(hipMalloc(&ctx->d_buckets, 8 * 2 * SDB_MAX_BUCKETS));
+  // x2: the batch pipeline double-buffers bucket state by query
+  // parity (per-step uses offset 0 only)
+  CTX_CHECK(hipMalloc(&ctx->d_buckets, 2 * 8 * 2 * SDB_MAX_BUCKETS));
   CTX_CHECK(hipMalloc(&ctx->d_overflow, 4));
   CTX_CHECK(hipMalloc(&ctx->d_terms, sizeof(TermDev) * SDB_MAX_TERMS *
                                        SDB_TERM_SLOTS));
@@ -3045,14 +3047,31 @@ int sdb_gpu_execute_topk(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
  * while results drain (duckdb_search_full_scan.cpp:1925-2000).
  * hits: nq*k entries; out_counts/totals: nq entries. Non-hybrid,
  * min_match semantics as execute_topk. */
-int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
-                               uint32_t nsegs, const SdbQueryPlan* plan,
-                               uint32_t k, uint32_t nq, SdbScoreDoc* hits,
-                               uint32_t* out_counts, uint64_t* totals) {
+static int exec_topk_batch_impl(
+  SdbGpuCtx* ctx, SdbGpuSegment* const* segs, uint32_t nsegs,
+  const SdbQueryPlan* plan, uint32_t k, uint32_t nq,
+  const SdbHybridPred* hpreds, uint32_t nhp, uint32_t h_nbuckets,
+  int64_t* bucket_counts, int64_t* bucket_sums, SdbScoreDoc* hits,
+  uint32_t* out_counts, uint64_t* totals) {
   if (!ctx || !segs || !plan || !hits || !out_counts || !totals ||
       nq == 0 || plan->nterms == 0 || plan->nterms > SDB_MAX_TERMS ||
       k == 0 || k == 0xFFFFFFFFu)
     return SDB_ERR_INVALID;
+  const bool hybrid = nhp > 0;
+  if (hybrid) {  // same contract as exec_topk_impl's hybrid arm
+    if (!hpreds || nhp > SDB_MAX_FILTER_COLS ||
+        hpreds[0].op != SDB_PRED_BETWEEN || h_nbuckets == 0 ||
+        h_nbuckets > SDB_MAX_BUCKETS || !bucket_counts || !bucket_sums)
+      return SDB_ERR_INVALID;
+    for (uint32_t x = 0; x < nhp; ++x) {
+      if (hpreds[x].slot >= 4 ||
+          (hpreds[x].op != SDB_PRED_LT && hpreds[x].op != SDB_PRED_GE &&
+           hpreds[x].op != SDB_PRED_BETWEEN))
+        return SDB_ERR_INVALID;
+      for (uint32_t sx = 0; sx < nsegs; ++sx)
+        if (!segs[sx]->fcols[hpreds[x].slot]) return SDB_ERR_INVALID;
+    }
+  }
   for (uint32_t i = 0; i < plan->nterms; ++i) {
     for (uint32_t j = i + 1; j < plan->nterms; ++j)
       if (plan->terms[i].term_idx == plan->terms[j].term_idx)
@@ -3073,7 +3092,8 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   bool use_wave = false;
   if (const char* e = getenv("SDB_TOPK_PATH")) {
     if (!strcmp(e, "wave"))
-      use_wave = use_sweep && plan->nterms <= SDB_PW_TERMS && !plan->wand;
+      use_wave = use_sweep && plan->nterms <= SDB_PW_TERMS &&
+                 !plan->wand && !hybrid;
     if (!strcmp(e, "general")) use_sweep = false;
   }
   if (const char* e = getenv("SDB_SWEEP_GEOM")) {
@@ -3139,6 +3159,10 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
                              ctx->stream));
     HIP_CHECK(hipMemsetAsync(qs ? ctx->d_ghist2 : ctx->d_ghist, 0,
                              4 * SDB_HIST_BINS * 16, ctx->stream));
+    if (hybrid)
+      HIP_CHECK(hipMemsetAsync(
+        ctx->d_buckets + (size_t)qs * 2 * SDB_MAX_BUCKETS, 0,
+        8 * 2 * SDB_MAX_BUCKETS, ctx->stream));
     for (uint32_t sg = 0; sg < nsegs; ++sg) {
       SdbGpuSegment* seg = segs[sg];
       const uint32_t slot = (q * nsegs + sg) % SDB_TERM_SLOTS;
@@ -3164,7 +3188,7 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       a.norms = seg->norms;
       a.doc_count = seg->hdr.doc_count;
       a.scorer = plan->scorer;
-      a.wand = (plan->wand && plan->min_match <= 1) ? 1u : 0u;
+      a.wand = (plan->wand && plan->min_match <= 1 && !hybrid) ? 1u : 0u;
       a.norm_stream = seg->hdr.version >= 3 ? 1u : 0u;
       a.nterms = plan->nterms;
       a.min_match = plan->min_match ? plan->min_match : 1;
@@ -3172,8 +3196,18 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
       a.smax = smax;
       a.seg_idx = sg;
       a.cand_cap = SDB_CAND_CAP;
-      a.nbuckets = 0;
-      a.bucket_out = ctx->d_buckets;
+      a.fcol = hybrid ? seg->fcols[hpreds[0].slot] : nullptr;
+      a.flo = hybrid ? hpreds[0].lo : 0;
+      a.fhi = hybrid ? hpreds[0].hi : 0;
+      a.nfx = hybrid ? nhp - 1 : 0;
+      for (uint32_t x = 0; hybrid && x + 1 < nhp; ++x) {
+        a.fxc[x] = seg->fcols[hpreds[x + 1].slot];
+        a.fxop[x] = (int)hpreds[x + 1].op;
+        a.fxlo[x] = hpreds[x + 1].lo;
+        a.fxhi[x] = hpreds[x + 1].hi;
+      }
+      a.nbuckets = hybrid ? h_nbuckets : 0;
+      a.bucket_out = ctx->d_buckets + (size_t)qs * 2 * SDB_MAX_BUCKETS;
       a.fb = plan->filter_boost ? seg->fboost : nullptr;
       a.fbmax = ps.fbmax;
       a.live = seg->live;
@@ -3289,6 +3323,17 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
     std::copy(cands, cands + kk, hits + (size_t)q * k);
     out_counts[q] = (uint32_t)kk;
     totals[q] = total;
+    if (hybrid) {
+      unsigned long long hb[2 * SDB_MAX_BUCKETS];
+      HIP_CHECK(hipMemcpyAsync(
+        hb, ctx->d_buckets + (size_t)qs * 2 * SDB_MAX_BUCKETS,
+        8ull * 2 * h_nbuckets, hipMemcpyDeviceToHost, ctx->copy_stream));
+      HIP_CHECK(hipStreamSynchronize(ctx->copy_stream));
+      for (uint32_t i = 0; i < h_nbuckets; ++i) {
+        bucket_counts[(size_t)q * h_nbuckets + i] = (int64_t)hb[2 * i];
+        bucket_sums[(size_t)q * h_nbuckets + i] = (int64_t)hb[2 * i + 1];
+      }
+    }
     return SDB_OK;
   };
 
@@ -3311,6 +3356,28 @@ int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
   }
   ctx->last_kernel_ms = total_kernel_ms / nq;
   return SDB_OK;
+}
+
+int sdb_gpu_execute_topk_batch(SdbGpuCtx* ctx, SdbGpuSegment* const* segs,
+                               uint32_t nsegs, const SdbQueryPlan* plan,
+                               uint32_t k, uint32_t nq, SdbScoreDoc* hits,
+                               uint32_t* out_counts, uint64_t* totals) {
+  return exec_topk_batch_impl(ctx, segs, nsegs, plan, k, nq, nullptr, 0, 0,
+                              nullptr, nullptr, hits, out_counts, totals);
+}
+
+// pipelined hybrid batch: per-query 2*nbuckets bucket planes double-
+// buffered by query parity; bucket_counts/bucket_sums hold nq*nbuckets
+int sdb_gpu_execute_topk_hybrid_batch(
+  SdbGpuCtx* ctx, SdbGpuSegment* const* segs, uint32_t nsegs,
+  const SdbQueryPlan* plan, uint32_t k, int64_t flo, int64_t fhi,
+  uint32_t nbuckets, uint32_t nq, int64_t* bucket_counts,
+  int64_t* bucket_sums, SdbScoreDoc* hits, uint32_t* out_counts,
+  uint64_t* totals) {
+  const SdbHybridPred p0 = {0, SDB_PRED_BETWEEN, flo, fhi};
+  return exec_topk_batch_impl(ctx, segs, nsegs, plan, k, nq, &p0, 1,
+                              nbuckets, bucket_counts, bucket_sums, hits,
+                              out_counts, totals);
 }
 
 // CountFast: exact match count without scoring (docs-only decode —

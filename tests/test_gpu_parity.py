@@ -1106,6 +1106,24 @@ def test_batch_pipelined_equals_single(ctx):
     assert bt2 == [t2] * 3
     np.testing.assert_array_equal(b2[0]["doc"], s2["doc"])
     np.testing.assert_array_equal(b2[2]["doc"], s2["doc"])
+    # hybrid batch: per-query hits AND bucket planes equal the single
+    # hybrid entry (bucket state double-buffered by query parity)
+    rng = np.random.default_rng(62)
+    col = rng.integers(0, 1 << 20, 400_000 + 1).astype(np.int64)
+    ctx.attach_column(seg, col)
+    flo, fhi = 1000, (1 << 19)
+    sh, st, sc, ss = ctx.execute_topk_hybrid([seg], [0, 1, 2, 3],
+                                             [1.0] * 4, 777, flo, fhi, 64)
+    bh, bt, bc, bs = ctx.execute_topk_hybrid_batch(
+        [seg], [0, 1, 2, 3], [1.0] * 4, 777, flo, fhi, 64, 5,
+        all_hits=True)
+    assert bt == [st] * 5
+    for q in range(5):
+        np.testing.assert_array_equal(bh[q]["doc"], sh["doc"])
+        np.testing.assert_array_equal(bh[q]["score"].view(np.uint32),
+                                      sh["score"].view(np.uint32))
+        np.testing.assert_array_equal(bc[q], sc)
+        np.testing.assert_array_equal(bs[q], ss)
 
 
 def test_validity_null_semantics(ctx):
