@@ -71,6 +71,35 @@ def test_oracle_str_pred_matches_naive():
             err_msg=f"{op} {lo!r} {hi!r}")
 
 
+def test_str_pred_fuzz_random_bytes():
+    """hypothesis fuzz: oracle mask == naive bytes-space evaluation for
+    random byte strings (any bytes, incl. NUL and 0xFF) and random
+    literals, all five ops."""
+    from hypothesis import given, settings, strategies as st
+
+    bstr = st.binary(min_size=0, max_size=12)
+
+    @settings(max_examples=200, deadline=None)
+    @given(vals=st.lists(bstr, min_size=1, max_size=40), lo=bstr, hi=bstr,
+           op=st.sampled_from(["lt", "ge", "eq", "between", "prefix"]))
+    def check(vals, lo, hi, op):
+        got = po.str_pred_mask(vals, op, lo, hi)
+        for i, v in enumerate(vals):
+            if op == "lt":
+                exp = v < lo
+            elif op == "ge":
+                exp = v >= lo
+            elif op == "eq":
+                exp = v == lo
+            elif op == "between":
+                exp = lo <= v <= hi
+            else:
+                exp = v.startswith(lo)
+            assert got[i] == exp, (v, op, lo, hi)
+
+    check()
+
+
 def test_encode_col_str_raw_roundtrip():
     vals = make_raw_strings(8, 1000)
     off, blob = sa.encode_col_str_raw(vals)
