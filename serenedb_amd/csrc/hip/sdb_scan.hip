@@ -279,7 +279,12 @@ __device__ __forceinline__ void col_read2_lds(const SdbColGroupDescDev* desc,
   x1 = d.base + (int64_t)(bits_at(sw, bit0 + d.width) & mask);
 }
 
-template <int RAW>
+// ASHAPE: compile-time aggregate-shape dispatch (DuckDB-style operator
+// specialization). 0 = generic runtime loop; 1 = the common 3-agg shape
+// [COUNT(*), SUM(i64), SUM(f32->f64)] with the per-row per-agg op/src
+// switches straight-lined (the FoR walker is instruction-issue-bound —
+// tools/ROUND2_NOTES.md). Any other shape falls back to ASHAPE=0.
+template <int RAW, int ASHAPE>
 __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned long long* acc = (unsigned long long*)smem;  // ngroups*naggs
@@ -482,6 +487,24 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         if (!okv[e]) continue;
         ++my_passed;
         const uint32_t grp = (uint32_t)ks[e];
+        if (ASHAPE == 1) {  // straight-lined 3-agg shape (dead if generic)
+          unsigned long long* s3 = &acc[grp * 3u];
+          atomicAdd(s3, 1ull);
+          if (valid_at(a.agg_valid[1], r + e)) {
+            int64_t x;
+            switch (a.agg_src[1]) {
+              case 1: x = pva[e]; break;
+              case 2: x = pvb[e]; break;
+              case 9: x = ks[e]; break;
+              default: x = col_read(a.agg_col[1], rg, r0, r + e); break;
+            }
+            atomicAdd(s3 + 1, (unsigned long long)x);
+          }
+          if (valid_at(a.agg_valid[2], r + e))
+            atomicAdd((double*)(s3 + 2),
+                      (double)((const float*)a.agg_col[2].data)[r + e]);
+          continue;
+        }
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
           if (a.agg_op[q] != SDB_AGG_COUNT &&
@@ -530,6 +553,23 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
       if (ok) {
         ++my_passed;
         const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
+        if (ASHAPE == 1) {  // straight-lined 3-agg shape
+          unsigned long long* s3 = &acc[grp * 3u];
+          atomicAdd(s3, 1ull);
+          if (valid_at(a.agg_valid[1], r)) {
+            int64_t x;
+            switch (a.agg_src[1]) {
+              case 1: x = pv0s; break;
+              case 2: x = pv1s; break;
+              case 9: x = (int64_t)grp; break;
+              default: x = col_read(a.agg_col[1], rg, r0, r); break;
+            }
+            atomicAdd(s3 + 1, (unsigned long long)x);
+          }
+          if (valid_at(a.agg_valid[2], r))
+            atomicAdd((double*)(s3 + 2),
+                      (double)((const float*)a.agg_col[2].data)[r]);
+        } else
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
           if (a.agg_op[q] != SDB_AGG_COUNT &&
@@ -582,6 +622,7 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
 // are staged cooperatively (one coalesced burst per column), and all pair
 // extraction reads hit LDS. Zonemap skips, predicate fast path, and the
 // agg_src decode-once dedup are identical to the unstaged walker.
+template <int ASHAPE>
 __launch_bounds__(SCAN_NTHREADS) __global__
 void scan_agg_staged_kernel(ScanArgs a) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -704,6 +745,29 @@ void scan_agg_staged_kernel(ScanArgs a) {
           if (!okv[e]) continue;
           ++my_passed;
           const uint32_t grp = (uint32_t)ks[e];
+          if (ASHAPE == 1) {  // straight-lined 3-agg shape
+            unsigned long long* s3 = &acc[grp * 3u];
+            atomicAdd(s3, 1ull);
+            if (valid_at(a.agg_valid[1], r + e)) {
+              int64_t x;
+              switch (a.agg_src[1]) {
+                case 1: x = pva[e]; break;
+                case 2: x = pvb[e]; break;
+                case 9: x = ks[e]; break;
+                default: x = col_read(a.agg_col[1], rg, r0, r + e); break;
+              }
+              atomicAdd(s3 + 1, (unsigned long long)x);
+            }
+            if (valid_at(a.agg_valid[2], r + e)) {
+              const float* fc = (const float*)a.agg_col[2].data;
+              const float fv = fc == a.f32_pay
+                                 ? ((const float*)(sw +
+                                                   a.f32_lds_off))[lr + e]
+                                 : fc[r + e];
+              atomicAdd((double*)(s3 + 2), (double)fv);
+            }
+            continue;
+          }
           for (uint32_t q = 0; q < a.naggs; ++q) {
             unsigned long long* slot = &acc[grp * a.naggs + q];
             if (a.agg_op[q] != SDB_AGG_COUNT &&
@@ -758,6 +822,23 @@ void scan_agg_staged_kernel(ScanArgs a) {
       if (ok) {
         ++my_passed;
         const uint32_t grp = (uint32_t)col_read(a.keys, rg, r0, r);
+        if (ASHAPE == 1) {  // straight-lined 3-agg shape
+          unsigned long long* s3 = &acc[grp * 3u];
+          atomicAdd(s3, 1ull);
+          if (valid_at(a.agg_valid[1], r)) {
+            int64_t x;
+            switch (a.agg_src[1]) {
+              case 1: x = pv0s; break;
+              case 2: x = pv1s; break;
+              case 9: x = (int64_t)grp; break;
+              default: x = col_read(a.agg_col[1], rg, r0, r); break;
+            }
+            atomicAdd(s3 + 1, (unsigned long long)x);
+          }
+          if (valid_at(a.agg_valid[2], r))
+            atomicAdd((double*)(s3 + 2),
+                      (double)((const float*)a.agg_col[2].data)[r]);
+        } else
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[grp * a.naggs + q];
           if (a.agg_op[q] != SDB_AGG_COUNT &&
@@ -1461,14 +1542,25 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
     }
   }
 
-  if (staged)
-    hipLaunchKernelGGL(scan_agg_staged_kernel, dim3(nblocks),
+  // compile-time agg-shape dispatch: the common [COUNT, SUM_I64,
+  // SUM_F64] shape gets the straight-lined FoR accumulate bodies
+  const bool shape3 = naggs == 3 && aggs[0].op == SDB_AGG_COUNT &&
+                      aggs[1].op == SDB_AGG_SUM_I64 &&
+                      aggs[2].op == SDB_AGG_SUM_F64;
+  if (staged && shape3)
+    hipLaunchKernelGGL((scan_agg_staged_kernel<1>), dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
+  else if (staged)
+    hipLaunchKernelGGL((scan_agg_staged_kernel<0>), dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
+  else if (any_for && shape3)
+    hipLaunchKernelGGL((scan_agg_kernel<0, 1>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   else if (any_for)
-    hipLaunchKernelGGL((scan_agg_kernel<0>), dim3(nblocks),
+    hipLaunchKernelGGL((scan_agg_kernel<0, 0>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   else
-    hipLaunchKernelGGL((scan_agg_kernel<1>), dim3(nblocks),
+    hipLaunchKernelGGL((scan_agg_kernel<1, 0>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   HIP_CHECK_CLEAN(hipGetLastError());
   std::vector<unsigned long long> h_out(nslots);
