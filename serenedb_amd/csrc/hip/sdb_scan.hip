@@ -333,6 +333,18 @@ __launch_bounds__(SCAN_NTHREADS) __global__ void scan_agg_kernel(ScanArgs a) {
         if (!okv[e]) continue;
         ++my_passed;
         const uint32_t g = (uint32_t)ks[e];
+        if (ASHAPE == 1) {  // straight-lined 3-agg shape (dead if generic)
+          unsigned long long* s3 = &acc[g * 3u];
+          atomicAdd(s3, 1ull);
+          if (valid_at(a.agg_valid[1], r + e))
+            atomicAdd(s3 + 1, (unsigned long long)((const long long*)
+                                                     a.agg_col[1]
+                                                       .data)[r + e]);
+          if (valid_at(a.agg_valid[2], r + e))
+            atomicAdd((double*)(s3 + 2),
+                      (double)((const float*)a.agg_col[2].data)[r + e]);
+          continue;
+        }
         for (uint32_t q = 0; q < a.naggs; ++q) {
           unsigned long long* slot = &acc[g * a.naggs + q];
           if (a.agg_op[q] != SDB_AGG_COUNT &&
@@ -1546,7 +1558,8 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   // SUM_F64] shape gets the straight-lined FoR accumulate bodies
   const bool shape3 = naggs == 3 && aggs[0].op == SDB_AGG_COUNT &&
                       aggs[1].op == SDB_AGG_SUM_I64 &&
-                      aggs[2].op == SDB_AGG_SUM_F64;
+                      aggs[2].op == SDB_AGG_SUM_F64 &&
+                      !getenv("SDB_SCAN_NOSHAPE");
   if (staged && shape3)
     hipLaunchKernelGGL((scan_agg_staged_kernel<1>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
@@ -1558,6 +1571,9 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
                        dim3(SCAN_NTHREADS), lds, stream, a);
   else if (any_for)
     hipLaunchKernelGGL((scan_agg_kernel<0, 0>), dim3(nblocks),
+                       dim3(SCAN_NTHREADS), lds, stream, a);
+  else if (shape3)
+    hipLaunchKernelGGL((scan_agg_kernel<1, 1>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   else
     hipLaunchKernelGGL((scan_agg_kernel<1, 0>), dim3(nblocks),

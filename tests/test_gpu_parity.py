@@ -1106,6 +1106,20 @@ def test_batch_pipelined_equals_single(ctx):
     assert bt2 == [t2] * 3
     np.testing.assert_array_equal(b2[0]["doc"], s2["doc"])
     np.testing.assert_array_equal(b2[2]["doc"], s2["doc"])
+    # MULTI-SEGMENT batch: term-slot rotation walks q*nsegs+sg — two
+    # segments per query must still match the single-query entry
+    blob2, _, _ = make_corpus(63, 300_000, [0.1, 0.05, 0.02, 0.01])
+    seg2 = ctx.load_segment(blob2)
+    sm, tm = ctx.execute_topk([seg, seg2], [0, 1, 2, 3], [1.0] * 4, 500)
+    bm, btm = ctx.execute_topk_batch([seg, seg2], [0, 1, 2, 3], [1.0] * 4,
+                                     500, 4, all_hits=True)
+    assert btm == [tm] * 4
+    for q in range(4):
+        np.testing.assert_array_equal(bm[q]["doc"], sm["doc"])
+        np.testing.assert_array_equal(bm[q]["segment"], sm["segment"])
+        np.testing.assert_array_equal(bm[q]["score"].view(np.uint32),
+                                      sm["score"].view(np.uint32))
+
     # hybrid batch: per-query hits AND bucket planes equal the single
     # hybrid entry (bucket state double-buffered by query parity)
     rng = np.random.default_rng(62)
