@@ -1572,10 +1572,11 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
   else if (any_for)
     hipLaunchKernelGGL((scan_agg_kernel<0, 0>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
-  else if (shape3)
-    hipLaunchKernelGGL((scan_agg_kernel<1, 1>), dim3(nblocks),
-                       dim3(SCAN_NTHREADS), lds, stream, a);
   else
+    // RAW stays generic: the straight-lined 3-agg arm measured 183G vs
+    // 234G rows/s on the same box (r2_raw_shape_ab.log) — the runtime
+    // agg loop bounds the e-unroll better there; only the
+    // instruction-bound FoR walkers keep ASHAPE=1
     hipLaunchKernelGGL((scan_agg_kernel<1, 0>), dim3(nblocks),
                        dim3(SCAN_NTHREADS), lds, stream, a);
   HIP_CHECK_CLEAN(hipGetLastError());
