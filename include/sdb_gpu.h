@@ -328,6 +328,22 @@ int sdb_gpu_strpred_mask(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t slot,
                          SdbPredOp op, const uint8_t* lo, uint32_t lo_len,
                          const uint8_t* hi, uint32_t hi_len);
 
+/* FSST-style compressed string slot (SURVEY.md 8f row 3 "dict/FSST"):
+ * rows hold code streams — code c < nsym expands to symbol bytes
+ * symbols[sym_offsets[c]..sym_offsets[c+1]) (1..8 bytes each, nsym <=
+ * 254, total symbol bytes <= 2048); code 255 escapes the next literal
+ * byte. offsets[rows+1] index the ENCODED blob. Predicates on such a
+ * slot (same sdb_gpu_strpred_mask call) decode on the fly against the
+ * plain-byte literal — semantics identical to the uncompressed slot. */
+int sdb_gpu_table_attach_strcol_fsst(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                                     uint32_t slot,
+                                     const uint64_t* offsets,
+                                     const uint8_t* enc_blob,
+                                     uint64_t enc_len,
+                                     const uint8_t* symbols,
+                                     const uint32_t* sym_offsets,
+                                     uint32_t nsym);
+
 typedef struct SdbPredSpec {
   uint32_t col;
   SdbPredOp op;

@@ -300,6 +300,28 @@ def str_pred_mask(values, op, lo, hi=None):
     return out
 
 
+def fsst_decode(enc, symbols):
+    """ORACLE (test infrastructure): FSST-style decode — code c < nsym
+    expands to symbols[c]; 255 escapes the next literal byte. Raises on
+    malformed streams (out-of-range code, escape at end)."""
+    out = bytearray()
+    i = 0
+    enc = bytes(enc)
+    while i < len(enc):
+        c = enc[i]
+        if c == 255:
+            if i + 1 >= len(enc):
+                raise ValueError("escape at end of stream")
+            out.append(enc[i + 1])
+            i += 2
+        else:
+            if c >= len(symbols):
+                raise ValueError(f"code {c} out of range")
+            out += symbols[c]
+            i += 1
+    return bytes(out)
+
+
 # --- reference simdcomp pins ---------------------------------------------
 
 def ref_pack_d1(prev, values128, bits):

@@ -245,6 +245,61 @@ def encode_col_str_raw(values):
     return offsets, b"".join(enc)
 
 
+def encode_col_str_fsst(values, sample_bytes=1 << 18):
+    """FSST-style compression for a raw string column (SURVEY.md 8f row
+    3): builds a symbol table (<=254 symbols of 1..8 bytes, greedy by
+    gain over a sample), encodes every string as code bytes (255 escapes
+    a literal byte), longest-match, deterministic. Returns (offsets
+    uint64[n+1] into the ENCODED blob, enc_blob bytes, symbols
+    list[bytes]) for GpuContext.attach_strcol_fsst. Decode restatement
+    lives in oracle/pyoracle.fsst_decode."""
+    import numpy as np
+    from collections import Counter
+
+    enc_in = [v.encode() if isinstance(v, str) else bytes(v)
+              for v in values]
+    # sample substrings, gain = (len-1 or the escape byte saved) * count
+    gains = Counter()
+    seen = 0
+    for b in enc_in:
+        if seen > sample_bytes:
+            break
+        seen += len(b)
+        for i in range(len(b)):
+            for L in range(1, 9):
+                if i + L > len(b):
+                    break
+                gains[b[i:i + L]] += L - 1 if L > 1 else 0
+        for ch in b:
+            gains[bytes([ch])] += 1  # 1-byte symbol saves the escape
+    ranked = sorted(gains.items(), key=lambda kv: (-kv[1], kv[0]))
+    symbols = [sym for sym, g in ranked[:254] if g > 0]
+    symbols.sort(key=lambda x: (-len(x), x))  # longest-match by scan
+    by_first = {}
+    for code, sym in enumerate(symbols):
+        by_first.setdefault(sym[0], []).append((sym, code))
+
+    out = bytearray()
+    offsets = np.zeros(len(enc_in) + 1, dtype=np.uint64)
+    for r, b in enumerate(enc_in):
+        i = 0
+        while i < len(b):
+            best = None
+            for sym, code in by_first.get(b[i], ()):
+                if b.startswith(sym, i):
+                    best = (sym, code)
+                    break  # symbols sorted longest-first
+            if best:
+                out.append(best[1])
+                i += len(best[0])
+            else:
+                out.append(255)
+                out.append(b[i])
+                i += 1
+        offsets[r + 1] = len(out)
+    return offsets, bytes(out), symbols
+
+
 def encode_col_str(values):
     """Dictionary-encode a string column: sorted-unique dictionary + i64
     codes. The SORTED dictionary is what makes string predicates map to
@@ -736,6 +791,30 @@ class GpuContext:
             C.c_uint64(len(bl)))
         if rc != 0:
             raise RuntimeError(f"attach_strcol rc={rc}")
+
+    def attach_strcol_fsst(self, tab, slot, offsets, enc_blob, symbols):
+        """Attach an FSST-style compressed string column (see
+        encode_col_str_fsst). Predicates via the same strpred_mask call
+        decode on the fly — semantics identical to the raw slot."""
+        import numpy as np
+
+        off = np.ascontiguousarray(offsets, dtype=np.uint64)
+        bl = np.frombuffer(bytes(enc_blob), dtype=np.uint8) if \
+            len(enc_blob) else np.zeros(0, dtype=np.uint8)
+        sym_off = np.zeros(len(symbols) + 1, dtype=np.uint32)
+        np.cumsum([len(x) for x in symbols], out=sym_off[1:])
+        sym = np.frombuffer(b"".join(bytes(x) for x in symbols),
+                            dtype=np.uint8) if symbols else \
+            np.zeros(0, dtype=np.uint8)
+        rc = self._lib.sdb_gpu_table_attach_strcol_fsst(
+            self._ctx, tab, C.c_uint32(slot),
+            off.ctypes.data_as(C.POINTER(C.c_uint64)),
+            bl.ctypes.data_as(C.POINTER(C.c_uint8)), C.c_uint64(len(bl)),
+            sym.ctypes.data_as(C.POINTER(C.c_uint8)),
+            sym_off.ctypes.data_as(C.POINTER(C.c_uint32)),
+            C.c_uint32(len(symbols)))
+        if rc != 0:
+            raise RuntimeError(f"attach_strcol_fsst rc={rc}")
 
     STR_OPS = {"lt": 1, "ge": 2, "between": 3, "eq": 4, "prefix": 8}
 

@@ -100,6 +100,31 @@ def test_str_pred_fuzz_random_bytes():
     check()
 
 
+def test_fsst_roundtrip_and_fuzz():
+    """FSST-style encode -> oracle decode round trip: fixed corpus plus
+    hypothesis fuzz over random byte strings (incl. 0xFF bytes, which
+    must be escaped)."""
+    from hypothesis import given, settings, strategies as st
+
+    vals = make_raw_strings(21, 3000) + ["\xff\xff", "\xffx", "x\xff"]
+    off, blob, syms = sa.encode_col_str_fsst(vals)
+    assert len(syms) <= 254 and all(1 <= len(x) <= 8 for x in syms)
+    assert len(blob) < sum(len(v.encode()) for v in vals)  # compresses
+    for i, v in enumerate(vals):
+        assert po.fsst_decode(blob[int(off[i]):int(off[i + 1])],
+                              syms) == v.encode(), v
+
+    @settings(max_examples=100, deadline=None)
+    @given(vs=st.lists(st.binary(min_size=0, max_size=24), min_size=1,
+                       max_size=30))
+    def fuzz(vs):
+        o, b, sy = sa.encode_col_str_fsst(vs)
+        for i, v in enumerate(vs):
+            assert po.fsst_decode(b[int(o[i]):int(o[i + 1])], sy) == v
+
+    fuzz()
+
+
 def test_encode_col_str_raw_roundtrip():
     vals = make_raw_strings(8, 1000)
     off, blob = sa.encode_col_str_raw(vals)
@@ -144,6 +169,19 @@ def test_strpred_mask_gpu_parity():
         np.testing.assert_allclose(f64[:, 2], exp_sf, rtol=1e-10,
                                    atol=1e-7, err_msg=str(op))
         del exp_si
+
+    # FSST-compressed slot: same strings on slot 1, every op's mask and
+    # aggregates must equal the raw slot's (decode-on-the-fly predicate)
+    foff, fblob, fsyms = sa.encode_col_str_fsst(vals)
+    assert len(fblob) < len(blob)  # actually compressed
+    ctx.attach_strcol_fsst(tab, 1, foff, fblob, fsyms)
+    for op, lo, hi in CASES:
+        ctx.strpred_mask(tab, 0, op, lo, hi)
+        i64r, _, pr = ctx.scan_agg(tab, 0, 64, [(0, 7, 0, 0)], [(0, 0)])
+        ctx.strpred_mask(tab, 1, op, lo, hi)
+        i64f, _, pf = ctx.scan_agg(tab, 0, 64, [(1, 7, 0, 0)], [(0, 0)])
+        assert pr == pf, (op, lo, hi)
+        np.testing.assert_array_equal(i64r, i64f, err_msg=str((op, lo)))
 
     # string pred AND numeric pred together
     ctx.strpred_mask(tab, 0, "prefix", "c", None)
@@ -211,4 +249,13 @@ def test_strcol_error_paths():
     # offsets not spanning the blob -> rejected
     with pytest.raises(RuntimeError):
         ctx.attach_strcol(tab, 1, off, blob + b"extra")
+    # FSST symbol-table validation: oversized/empty symbols rejected
+    foff, fblob, fsyms = sa.encode_col_str_fsst(vals)
+    with pytest.raises(RuntimeError):
+        ctx.attach_strcol_fsst(tab, 1, foff, fblob, [b"123456789"])  # >8
+    with pytest.raises(RuntimeError):
+        ctx.attach_strcol_fsst(tab, 1, foff, fblob, [b""])  # empty
+    with pytest.raises(RuntimeError):
+        ctx.attach_strcol_fsst(tab, 1, foff, fblob,
+                               [bytes([i & 0xFF]) for i in range(255)])
     ctx.free_table(tab)
