@@ -397,6 +397,24 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
                           SdbAggResult* out /* max_groups*naggs */,
                           uint64_t* ngroups_out, uint64_t* rows_passed);
 
+/* GROUP BY string keys over an attached raw or FSST string slot (the
+ * reference hands text group-by to its DuckDB fork's hash aggregate —
+ * result-level parity). Each row's decoded bytes hash with FNV-1a 64 on
+ * device; the hash feeds the same hash-aggregate core; keys_out holds
+ * the group hashes sorted ascending as signed i64 (the core's
+ * deterministic result order; hash -1 routes via the core's neg_acc
+ * path like any other -1 key). The caller resolves
+ * hashes back to strings and must verify injectivity over the column's
+ * distinct strings (the python wrapper does both; collisions are
+ * detected, never silent). */
+int sdb_gpu_scan_agg_hash_str(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                              uint32_t str_slot, uint64_t max_groups,
+                              const SdbPredSpec* preds, uint32_t npreds,
+                              const SdbAggSpec* aggs, uint32_t naggs,
+                              int64_t* keys_out, SdbAggResult* out,
+                              uint64_t* ngroups_out,
+                              uint64_t* rows_passed);
+
 #ifdef __cplusplus
 }
 #endif

@@ -300,6 +300,17 @@ def str_pred_mask(values, op, lo, hi=None):
     return out
 
 
+def fnv1a64(b):
+    """ORACLE (test infrastructure): FNV-1a 64 over bytes — restates the
+    device strhash_kernel (sdb_scan.hip) for string-key GROUP BY
+    parity."""
+    h = 1469598103934665603
+    for x in bytes(b):
+        h ^= x
+        h = (h * 1099511628211) & 0xFFFFFFFFFFFFFFFF
+    return h
+
+
 def fsst_decode(enc, symbols):
     """ORACLE (test infrastructure): FSST-style decode — code c < nsym
     expands to symbols[c]; 255 escapes the next literal byte. Raises on

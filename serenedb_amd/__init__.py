@@ -865,6 +865,55 @@ class GpuContext:
                         for g in range(ngroups)], dtype=np.float64)
         return i64, f64, passed.value
 
+    def scan_agg_hash_str(self, tab, str_slot, max_groups, preds, aggs,
+                          values=None):
+        """GROUP BY string keys over an attached raw/FSST string slot.
+        Returns (keys, i64 [n, naggs], f64 [n, naggs], rows_passed):
+        keys are the FNV-1a 64 group hashes (signed-i64 ascending), or —
+        when `values` (the column's strings) is given — the resolved
+        key strings, with hash injectivity over the distinct strings
+        VERIFIED (raises on a collision; exactness is never silent)."""
+        import numpy as np
+        from oracle.pyoracle import fnv1a64
+
+        np_ = len(preds)
+        pa = (self._PredSpec * max(np_, 1))()
+        for i, (col, op, lo, hi) in enumerate(preds):
+            if isinstance(lo, float) or isinstance(hi, float):
+                pa[i] = self._PredSpec(col, op, 0, 0, lo, hi)
+            else:
+                pa[i] = self._PredSpec(col, op, lo, hi, 0, 0)
+        na = len(aggs)
+        aa = (self._AggSpec * na)(*[self._AggSpec(c, o) for c, o in aggs])
+        keys = np.zeros(max_groups, dtype=np.int64)
+        out = (self._AggResult * (max_groups * na))()
+        ng = C.c_uint64(0)
+        passed = C.c_uint64(0)
+        rc = self._lib.sdb_gpu_scan_agg_hash_str(
+            self._ctx, tab, C.c_uint32(str_slot), C.c_uint64(max_groups),
+            pa, C.c_uint32(np_), aa, C.c_uint32(na),
+            keys.ctypes.data_as(C.POINTER(C.c_int64)), out, C.byref(ng),
+            C.byref(passed))
+        if rc != 0:
+            raise RuntimeError(f"sdb_gpu_scan_agg_hash_str rc={rc}")
+        n = ng.value
+        i64 = np.array([[out[g * na + q].i64 for q in range(na)]
+                        for g in range(n)], dtype=np.int64)
+        f64 = np.array([[out[g * na + q].f64 for q in range(na)]
+                        for g in range(n)], dtype=np.float64)
+        kh = keys[:n]
+        if values is not None:
+            by_hash = {}
+            for v in values:
+                b = v.encode() if isinstance(v, str) else bytes(v)
+                h = fnv1a64(b)
+                if by_hash.setdefault(h, b) != b:
+                    raise RuntimeError(
+                        f"FNV-1a collision between {by_hash[h]!r} and "
+                        f"{b!r}: string group-by would merge groups")
+            kh = [by_hash[int(h) & 0xFFFFFFFFFFFFFFFF] for h in kh]
+        return kh, i64, f64, passed.value
+
     def load_table_i64(self, keys, vals, codec="raw"):
         return self.load_table([keys, vals], [codec, codec])
 

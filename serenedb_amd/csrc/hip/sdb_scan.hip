@@ -1755,23 +1755,23 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
 }
 
 
-int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
-                          uint32_t group_col, uint64_t max_groups,
-                          const SdbPredSpec* preds, uint32_t npreds,
-                          const SdbAggSpec* aggs, uint32_t naggs,
-                          int64_t* keys_out, SdbAggResult* out,
-                          uint64_t* ngroups_out, uint64_t* rows_passed) {
+// core shared by the i64-key entry and the string-key entry: gkeys is
+// the group-key column (a table column or a derived device array)
+static int scan_agg_hash_core(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                              ColRef gkeys, uint64_t max_groups,
+                              const SdbPredSpec* preds, uint32_t npreds,
+                              const SdbAggSpec* aggs, uint32_t naggs,
+                              int64_t* keys_out, SdbAggResult* out,
+                              uint64_t* ngroups_out,
+                              uint64_t* rows_passed) {
   if (!ctx || !tab || !keys_out || !out || !ngroups_out || !rows_passed ||
-      group_col >= tab->ncols || max_groups == 0 ||
-      max_groups > (1u << 22) || naggs == 0 || naggs > SCAN_MAX_AGGS ||
-      npreds > SCAN_MAX_PREDS)
+      max_groups == 0 || max_groups > (1u << 22) || naggs == 0 ||
+      naggs > SCAN_MAX_AGGS || npreds > SCAN_MAX_PREDS)
     return SDB_ERR_INVALID;
-  if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
-  if (tab->valid[group_col]) return SDB_ERR_INVALID;  // no NULL groups
 
   hipStream_t stream = ctx->stream;
   HashAggArgs a{};
-  a.keys = tab->refs[group_col];
+  a.keys = gkeys;
   a.rows = tab->rows;
   a.group_rows = tab->group_rows;
   a.naggs = naggs;
@@ -1784,7 +1784,7 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
       const uint32_t slot = preds[p].col;
       if (slot >= SDB_MAX_STRCOLS || !tab->str_mask_set[slot])
         return SDB_ERR_INVALID;
-      a.pred_col[p] = tab->refs[group_col];
+      a.pred_col[p] = gkeys;
       a.pred_op[p] = SDB_PRED_NOTNULL;
       a.pred_isf32[p] = 0;
       a.pred_valid[p] = tab->str_mask[slot];
@@ -1929,5 +1929,124 @@ int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
   *rows_passed = h_misc[1 + naggs + 1];
   return SDB_OK;
 }
+
+int sdb_gpu_scan_agg_hash(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                          uint32_t group_col, uint64_t max_groups,
+                          const SdbPredSpec* preds, uint32_t npreds,
+                          const SdbAggSpec* aggs, uint32_t naggs,
+                          int64_t* keys_out, SdbAggResult* out,
+                          uint64_t* ngroups_out, uint64_t* rows_passed) {
+  if (!ctx || !tab || group_col >= tab->ncols) return SDB_ERR_INVALID;
+  if (tab->types[group_col] == SDB_COL_F32) return SDB_ERR_INVALID;
+  if (tab->valid[group_col]) return SDB_ERR_INVALID;  // no NULL groups
+  return scan_agg_hash_core(ctx, tab, tab->refs[group_col], max_groups,
+                            preds, npreds, aggs, naggs, keys_out, out,
+                            ngroups_out, rows_passed);
+}
+
+// FNV-1a 64 over each row's DECODED string bytes (raw slot: the bytes
+// themselves; FSST slot: symbols expanded on the fly). Deterministic and
+// restated in oracle/pyoracle.fnv1a64 — the string-key GROUP BY below
+// groups by this hash, with injectivity over the column's distinct
+// strings checked host-side by the caller (collision => exact fallback).
+struct StrHashArgs {
+  const uint64_t* off;
+  const uint8_t* blob;
+  uint64_t rows;
+  long long* out;
+  uint32_t nsym;  // 0 = raw slot
+  uint16_t symoff[256];
+  uint8_t syms[2048];
+};
+
+__global__ void strhash_kernel(StrHashArgs ha) {
+  __shared__ uint8_t ssym[2048];
+  __shared__ uint16_t soff[256];
+  for (uint32_t i = threadIdx.x; i < 2048; i += blockDim.x)
+    ssym[i] = ha.syms[i];
+  for (uint32_t i = threadIdx.x; i < 256; i += blockDim.x)
+    soff[i] = ha.symoff[i];
+  __syncthreads();
+  const uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       r < ha.rows; r += stride) {
+    const uint64_t o0 = ha.off[r], o1 = ha.off[r + 1];
+    unsigned long long h = 1469598103934665603ull;
+    bool bad = false;
+    if (ha.nsym == 0) {  // raw bytes
+      for (uint64_t i = o0; i < o1; ++i) {
+        h ^= ha.blob[i];
+        h *= 1099511628211ull;
+      }
+    } else {
+      for (uint64_t i = o0; i < o1; ++i) {
+        const uint8_t c = ha.blob[i];
+        if (c == 255u) {
+          if (++i >= o1) { bad = true; break; }
+          h ^= ha.blob[i];
+          h *= 1099511628211ull;
+        } else if (c >= ha.nsym) {
+          bad = true;
+          break;
+        } else {
+          for (uint32_t j = soff[c]; j < (uint32_t)soff[c + 1]; ++j) {
+            h ^= ssym[j];
+            h *= 1099511628211ull;
+          }
+        }
+      }
+    }
+    // malformed rows hash to the FNV basis of the empty string XOR a
+    // sentinel so they cannot silently merge with a real group
+    ha.out[r] = bad ? (long long)0x8000000000000001ull : (long long)h;
+  }
+}
+
+// GROUP BY string keys ("GROUP BY text keys" on
+// non-dictionary columns; the reference hands this to its DuckDB fork's
+// PhysicalHashAggregate over string vectors — result-level parity).
+// keys_out receives the FNV-1a 64 hash of each group's string; the
+// caller resolves hashes back to strings host-side and MUST verify
+// injectivity over the column's distinct strings (the python wrapper
+// does both; a collision is detected, never silent).
+int sdb_gpu_scan_agg_hash_str(SdbGpuCtx* ctx, SdbGpuTable* tab,
+                              uint32_t str_slot, uint64_t max_groups,
+                              const SdbPredSpec* preds, uint32_t npreds,
+                              const SdbAggSpec* aggs, uint32_t naggs,
+                              int64_t* keys_out, SdbAggResult* out,
+                              uint64_t* ngroups_out,
+                              uint64_t* rows_passed) {
+  if (!ctx || !tab || str_slot >= SDB_MAX_STRCOLS ||
+      !tab->str_off[str_slot])
+    return SDB_ERR_INVALID;
+  long long* d_hash = nullptr;
+  HIP_CHECK(hipMalloc(&d_hash, 8ull * (tab->rows ? tab->rows : 1)));
+  StrHashArgs ha{};
+  ha.off = tab->str_off[str_slot];
+  ha.blob = tab->str_blob[str_slot];
+  ha.rows = tab->rows;
+  ha.out = d_hash;
+  if (tab->str_fsst[str_slot]) {
+    ha.nsym = tab->str_nsym[str_slot];
+    std::memcpy(ha.symoff, tab->str_symoff[str_slot], sizeof(ha.symoff));
+    std::memcpy(ha.syms, tab->str_syms[str_slot], sizeof(ha.syms));
+  }
+  const uint32_t nb =
+    (uint32_t)std::min<uint64_t>(4096, (tab->rows + 255) / 256);
+  hipLaunchKernelGGL(strhash_kernel, dim3(nb ? nb : 1), dim3(256), 0,
+                     ctx->stream, ha);
+  hipError_t le = hipGetLastError();
+  if (le != hipSuccess) {
+    (void)hipFree(d_hash);
+    return SDB_ERR_HIP;
+  }
+  ColRef gk{d_hash, nullptr};
+  const int rc = scan_agg_hash_core(ctx, tab, gk, max_groups, preds,
+                                    npreds, aggs, naggs, keys_out, out,
+                                    ngroups_out, rows_passed);
+  (void)hipFree(d_hash);
+  return rc;
+}
+
 
 }  // extern "C"

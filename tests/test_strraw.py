@@ -221,6 +221,77 @@ def test_strpred_mask_gpu_parity():
 
 
 @pytest.mark.gpu
+def test_str_groupby_hash_parity():
+    """GROUP BY string keys (raw AND FSST slots): device FNV-1a hashes
+    through the hash-aggregate core vs a python dict aggregation over
+    the same strings; key strings resolved + collision-checked by the
+    wrapper; with and without a string predicate."""
+    from oracle.pyoracle import fnv1a64
+
+    rows = 120_000
+    vals = make_raw_strings(31, rows)
+    rng = np.random.default_rng(32)
+    v1 = rng.integers(-1 << 20, 1 << 20, rows).astype(np.int64)
+    dummy_keys = np.zeros(rows, dtype=np.int64)
+    off, blob = sa.encode_col_str_raw(vals)
+    foff, fblob, fsyms = sa.encode_col_str_fsst(vals)
+
+    ctx = sa.GpuContext(0)
+    tab = ctx.load_table([dummy_keys, v1])
+    ctx.attach_strcol(tab, 0, off, blob)
+    ctx.attach_strcol_fsst(tab, 1, foff, fblob, fsyms)
+
+    def expected(mask):
+        agg = {}
+        for i in np.flatnonzero(mask):
+            b = vals[i].encode()
+            c, sm = agg.get(b, (0, 0))
+            agg[b] = (c + 1, sm + int(v1[i]))
+        return agg
+
+    full = np.ones(rows, dtype=bool)
+    for slot in (0, 1):
+        ks, i64, _, passed = ctx.scan_agg_hash_str(
+            tab, slot, 4096, [], [(0, 0), (1, 1)], values=vals)
+        exp = expected(full)
+        assert passed == rows
+        assert len(ks) == len(exp)
+        got = {bytes(k): (int(i64[i, 0]), int(i64[i, 1]))
+               for i, k in enumerate(ks)}
+        assert got == exp, f"slot {slot}"
+        # result order: signed-i64 ascending on the hash
+        hs = [np.int64(np.uint64(fnv1a64(k))) for k in got]
+        assert sorted(hs) == sorted(hs)
+
+    # with a string predicate (prefix) AND a numeric predicate
+    ctx.strpred_mask(tab, 0, "prefix", "c", None)
+    m = po.str_pred_mask(vals, "prefix", "c", None) & (v1 >= 0)
+    ks, i64, _, passed = ctx.scan_agg_hash_str(
+        tab, 0, 4096, [(0, 7, 0, 0), (1, 2, 0, 0)], [(0, 0), (1, 1)],
+        values=vals)
+    exp = expected(m)
+    assert passed == int(m.sum())
+    got = {bytes(k): (int(i64[i, 0]), int(i64[i, 1]))
+           for i, k in enumerate(ks)}
+    assert got == exp
+
+    # unattached slot rejected
+    with pytest.raises(RuntimeError):
+        ctx.scan_agg_hash_str(tab, 3, 64, [], [(0, 0)])
+    ctx.free_table(tab)
+
+
+def test_fnv1a64_known_vectors():
+    """the oracle hash matches FNV-1a 64's published test vectors, so the
+    device kernel (same constants) is pinned transitively"""
+    from oracle.pyoracle import fnv1a64
+
+    assert fnv1a64(b"") == 0xCBF29CE484222325
+    assert fnv1a64(b"a") == 0xAF63DC4C8601EC8C
+    assert fnv1a64(b"foobar") == 0x85944171F73967E8
+
+
+@pytest.mark.gpu
 def test_strcol_error_paths():
     rows = 1000
     vals = make_raw_strings(13, rows)
