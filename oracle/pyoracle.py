@@ -304,7 +304,7 @@ def fnv1a64(b):
     """ORACLE (test infrastructure): FNV-1a 64 over bytes — restates the
     device strhash_kernel (sdb_scan.hip) for string-key GROUP BY
     parity."""
-    h = 1469598103934665603
+    h = 14695981039346656037
     for x in bytes(b):
         h ^= x
         h = (h * 1099511628211) & 0xFFFFFFFFFFFFFFFF

@@ -1971,7 +1971,7 @@ __global__ void strhash_kernel(StrHashArgs ha) {
   for (uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
        r < ha.rows; r += stride) {
     const uint64_t o0 = ha.off[r], o1 = ha.off[r + 1];
-    unsigned long long h = 1469598103934665603ull;
+    unsigned long long h = 14695981039346656037ull;
     bool bad = false;
     if (ha.nsym == 0) {  // raw bytes
       for (uint64_t i = o0; i < o1; ++i) {
