@@ -192,6 +192,85 @@ def _hash_worker(rank, world, port, q):
     dist.destroy_process_group()
 
 
+def _strkey_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    from oracle.pyoracle import fnv1a64
+
+    rows_total, seed = 60_000, 35
+    per = rows_total // world
+    lo = rank * per
+    hi = rows_total if rank == world - 1 else (rank + 1) * per
+    rng = np.random.default_rng(seed)
+    vocab = [f"term-{i:04d}" for i in range(800)]
+    vals_s = [vocab[i] for i in rng.integers(0, len(vocab), rows_total)]
+    vals = rng.integers(-1000, 1000, rows_total).astype(np.int64)
+    # per-rank string-key hash aggregate (what scan_agg_hash_str returns:
+    # FNV-1a hash keys signed-ascending + per-key aggregates); ranks
+    # shard rows, the merge is identical to the i64 hash-agg merge
+    # because the keys ARE i64 hashes
+    keys = np.array([np.int64(np.uint64(fnv1a64(v.encode())))
+                     for v in vals_s[lo:hi]])
+    mv = vals[lo:hi]
+    uk, inv = np.unique(keys, return_inverse=True)
+    cnt = np.bincount(inv, minlength=len(uk)).astype(np.int64)
+    sm = np.zeros(len(uk), dtype=np.int64)
+    np.add.at(sm, inv, mv)
+    rows = [None] * world
+    dist.all_gather_object(rows, (uk, cnt, sm))
+    allk = np.concatenate([r[0] for r in rows])
+    allc = np.concatenate([r[1] for r in rows])
+    alls = np.concatenate([r[2] for r in rows])
+    mk, minv = np.unique(allk, return_inverse=True)
+    mc = np.zeros(len(mk), dtype=np.int64)
+    ms = np.zeros(len(mk), dtype=np.int64)
+    np.add.at(mc, minv, allc)
+    np.add.at(ms, minv, alls)
+    if rank == 0:
+        q.put((mk, mc, ms))
+    dist.destroy_process_group()
+
+
+def test_gloo_strkey_groupby_merge_equals_full():
+    """Distributed merge of per-rank STRING-key group-bys: the keys are
+    FNV-1a 64 hashes (what sdb_gpu_scan_agg_hash_str returns), so the
+    merge is the same gather + merge-by-key as the i64 hash aggregate;
+    equals the single-node string aggregation resolved back through the
+    hash."""
+    from oracle.pyoracle import fnv1a64
+
+    rows_total, seed = 60_000, 35
+    rng = np.random.default_rng(seed)
+    vocab = [f"term-{i:04d}" for i in range(800)]
+    vals_s = [vocab[i] for i in rng.integers(0, len(vocab), rows_total)]
+    vals = rng.integers(-1000, 1000, rows_total).astype(np.int64)
+    agg = {}
+    for v, x in zip(vals_s, vals):
+        c, sm = agg.get(v, (0, 0))
+        agg[v] = (c + 1, sm + int(x))
+    exp = sorted(
+        (np.int64(np.uint64(fnv1a64(k.encode()))), c, sm)
+        for k, (c, sm) in agg.items())
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_strkey_worker, args=(r, 2, 29537, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        mk, mc, ms = q.get(timeout=300)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    np.testing.assert_array_equal(mk, [e[0] for e in exp])
+    np.testing.assert_array_equal(mc, [e[1] for e in exp])
+    np.testing.assert_array_equal(ms, [e[2] for e in exp])
+
+
 def test_gloo_hash_agg_merge_equals_full():
     """Distributed merge of per-rank HASH-aggregate tables (arbitrary
     sparse i64 keys -> gather + merge by key) equals the
