@@ -1120,6 +1120,17 @@ def test_batch_pipelined_equals_single(ctx):
         np.testing.assert_array_equal(bm[q]["score"].view(np.uint32),
                                       sm["score"].view(np.uint32))
 
+    # WAND plans through the batch (plan.wand propagates per segment)
+    sw, tw = ctx.execute_topk([seg], [0, 1, 2, 3], [1.0] * 4, 10,
+                              wand=True)
+    bw, btw = ctx.execute_topk_batch([seg], [0, 1, 2, 3], [1.0] * 4, 10,
+                                     3, wand=True, all_hits=True)
+    assert btw == [tw] * 3
+    for q in range(3):
+        np.testing.assert_array_equal(bw[q]["doc"], sw["doc"])
+        np.testing.assert_array_equal(bw[q]["score"].view(np.uint32),
+                                      sw["score"].view(np.uint32))
+
     # hybrid batch: per-query hits AND bucket planes equal the single
     # hybrid entry (bucket state double-buffered by query parity)
     rng = np.random.default_rng(62)

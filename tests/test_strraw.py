@@ -278,6 +278,10 @@ def test_str_groupby_hash_parity():
     # unattached slot rejected
     with pytest.raises(RuntimeError):
         ctx.scan_agg_hash_str(tab, 3, 64, [], [(0, 0)])
+    # more distinct strings than max_groups -> loud SDB_ERR_OOM, never a
+    # silently truncated result
+    with pytest.raises(RuntimeError):
+        ctx.scan_agg_hash_str(tab, 0, 8, [], [(0, 0)])
     ctx.free_table(tab)
 
 
