@@ -2397,6 +2397,8 @@ const char* sdb_gpu_version(void) { return "sdb_gpu 0.1 gfx950"; }
 
 // free whatever a partially-constructed context owns (error paths)
 static void ctx_free_partial(SdbGpuCtx* ctx) {
+  if (ctx->d_scan_out) (void)hipFree(ctx->d_scan_out);
+  if (ctx->d_scan_passed) (void)hipFree(ctx->d_scan_passed);
   if (ctx->d_cands) (void)hipFree(ctx->d_cands);
   if (ctx->d_cands2) (void)hipFree(ctx->d_cands2);
   if (ctx->d_ghist2) (void)hipFree(ctx->d_ghist2);
@@ -2474,6 +2476,8 @@ int sdb_gpu_ctx_create(int device, SdbGpuCtx** out) {
 
 int sdb_gpu_ctx_destroy(SdbGpuCtx* ctx) {
   if (!ctx) return SDB_ERR_INVALID;
+  (void)hipFree(ctx->d_scan_out);
+  (void)hipFree(ctx->d_scan_passed);
   (void)hipFree(ctx->d_cands2);
   (void)hipFree(ctx->d_ghist2);
   (void)hipFree(ctx->d_qmisc);

@@ -16,6 +16,11 @@ struct TermDev {
 struct SdbGpuCtx {
   int device;
   hipStream_t stream;
+  // cached dense-scan workspace: the dense scan_agg hot path is
+  // allocation-free across calls (grown on demand, freed with the ctx)
+  unsigned long long* d_scan_out;
+  uint64_t scan_out_cap;  // slots
+  unsigned long long* d_scan_passed;
   SdbScoreDoc* d_cands;
   uint32_t* d_cand_count;
   unsigned long long* d_total_matches;

@@ -1612,23 +1612,28 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
       return SDB_ERR_INVALID;
   }
   const uint32_t nslots = ngroups * naggs;
-  unsigned long long* d_out = nullptr;
-  unsigned long long* d_passed = nullptr;
-  // every HIP failure past the first hipMalloc frees both device buffers
-  // before returning (ADVICE r1 leak)
+  // ctx-cached workspace: the dense scan hot path allocates nothing per
+  // call (grown on demand; buffers owned and freed by the context)
 #define HIP_CHECK_CLEAN(x)                                   \
   do {                                                       \
     hipError_t _e = (x);                                     \
     if (_e != hipSuccess) {                                  \
-      if (d_out) (void)hipFree(d_out);                       \
-      if (d_passed) (void)hipFree(d_passed);                 \
       return _e == hipErrorNoDevice ? SDB_ERR_NO_GPU         \
              : _e == hipErrorOutOfMemory ? SDB_ERR_OOM       \
                                          : SDB_ERR_HIP;      \
     }                                                        \
   } while (0)
-  HIP_CHECK_CLEAN(hipMalloc(&d_out, 8ull * nslots));
-  HIP_CHECK_CLEAN(hipMalloc(&d_passed, 8));
+  if (ctx->scan_out_cap < nslots) {
+    if (ctx->d_scan_out) (void)hipFree(ctx->d_scan_out);
+    ctx->d_scan_out = nullptr;
+    ctx->scan_out_cap = 0;
+    HIP_CHECK_CLEAN(hipMalloc(&ctx->d_scan_out, 8ull * nslots));
+    ctx->scan_out_cap = nslots;
+  }
+  if (!ctx->d_scan_passed)
+    HIP_CHECK_CLEAN(hipMalloc(&ctx->d_scan_passed, 8));
+  unsigned long long* d_out = ctx->d_scan_out;
+  unsigned long long* d_passed = ctx->d_scan_passed;
   HIP_CHECK_CLEAN(hipMemsetAsync(d_out, 0, 8ull * nslots, stream));
   HIP_CHECK_CLEAN(hipMemsetAsync(d_passed, 0, 8, stream));
   a.out = d_out;
@@ -1733,8 +1738,7 @@ int sdb_gpu_scan_agg(SdbGpuCtx* ctx, SdbGpuTable* tab, uint32_t group_col,
                                  hipMemcpyDeviceToHost, stream));
   HIP_CHECK_CLEAN(hipStreamSynchronize(stream));
 #undef HIP_CHECK_CLEAN
-  (void)hipFree(d_out);
-  (void)hipFree(d_passed);
+  // d_out/d_passed are ctx-cached; freed with the context
   for (uint32_t g = 0; g < ngroups; ++g) {
     for (uint32_t q = 0; q < naggs; ++q) {
       const unsigned long long raw = h_out[g * naggs + q];
