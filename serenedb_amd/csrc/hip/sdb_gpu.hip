@@ -2320,10 +2320,9 @@ namespace {
 struct SweepGeom {
   uint32_t wd, nth;
 };
-constexpr SweepGeom kSweepGeoms[] = {{24576, 1024}, {16384, 1024},
-                                     {12288, 512},  {8192, 512},
-                                     {8192, 256},   {4096, 256},
-                                     {24576, 512},  {32768, 512}};
+// compiled geometry set (see SDB_SWEEP_CASE below): 24576x1024 default,
+// 16384x1024, 12288x512, 8192x512, 8192x256, 4096x256, 24576x512,
+// 32768x512 — selected at launch by SDB_SWEEP_GEOM
 
 bool launch_sweep(const SweepGeom& g, dim3 grid, size_t lds,
                   hipStream_t st, const WindowArgs& a,
